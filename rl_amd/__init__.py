@@ -1,0 +1,13 @@
+"""rl_amd — MI355X-native reinforcement-learning framework.
+
+A from-scratch framework with the capabilities of pytorch/rl (TorchRL),
+built MI355X-first: PyTorch-ROCm for the module layer, hand-written
+HIP/CDNA4 kernels for the hot ops (GAE/TD(λ)/V-trace scans, prioritized
+sum-trees, fused actor epilogues, fused recurrent scans), and RCCL over
+xGMI for every collective.
+"""
+__version__ = "0.1.0"
+
+from ._utils import logger, timeit, seed_everything, set_profiling_enabled
+
+__all__ = ["logger", "timeit", "seed_everything", "set_profiling_enabled"]

@@ -1,0 +1,294 @@
+"""Core runtime utilities for rl_amd.
+
+MI355X-native re-design of the reference's runtime layer
+(cf. pytorch/rl torchrl/_utils.py:48-1838): logger, ``timeit`` accumulation,
+seeding, env-var flags, multiprocessing helpers and profiling gates.
+"""
+from __future__ import annotations
+
+import collections
+import contextlib
+import logging
+import math
+import os
+import sys
+import time
+import traceback
+from typing import Any, Callable, Dict, Iterator, Optional
+
+import numpy as np
+import torch
+
+__all__ = [
+    "logger",
+    "timeit",
+    "seed_everything",
+    "strtobool",
+    "get_binary_env_var",
+    "VERBOSE",
+    "set_profiling_enabled",
+    "profiling_enabled",
+    "record_function",
+    "prod",
+    "_ProcessNoWarn",
+    "_check_for_faulty_process",
+    "hip_is_available",
+    "device_of",
+]
+
+
+# --------------------------------------------------------------------------- #
+# Logging
+# --------------------------------------------------------------------------- #
+def _make_logger() -> logging.Logger:
+    log = logging.getLogger("rl_amd")
+    if not log.handlers:
+        handler = logging.StreamHandler(sys.stderr)
+        handler.setFormatter(
+            logging.Formatter("%(asctime)s [%(name)s][%(levelname)s] %(message)s")
+        )
+        log.addHandler(handler)
+    log.setLevel(os.environ.get("RL_AMD_LOGLEVEL", "INFO").upper())
+    return log
+
+
+logger = _make_logger()
+
+
+def strtobool(val: str) -> bool:
+    """Convert a string representation of truth to ``True`` or ``False``."""
+    val = val.lower()
+    if val in ("y", "yes", "t", "true", "on", "1"):
+        return True
+    if val in ("n", "no", "f", "false", "off", "0"):
+        return False
+    raise ValueError(f"invalid truth value {val!r}")
+
+
+def get_binary_env_var(key: str, default: bool = False) -> bool:
+    val = os.environ.get(key)
+    if val is None:
+        return default
+    try:
+        return strtobool(val)
+    except ValueError:
+        return default
+
+
+VERBOSE = get_binary_env_var("VERBOSE", False)
+
+
+# --------------------------------------------------------------------------- #
+# timeit — accumulating timer (reference torchrl/_utils.py:221)
+# --------------------------------------------------------------------------- #
+class timeit:
+    """Context-manager / decorator that accumulates wall-clock timings globally.
+
+    Usage::
+
+        with timeit("collect"):
+            ...
+        timeit.print()
+    """
+
+    _REG: Dict[str, list] = collections.defaultdict(lambda: [0.0, 0.0, 0])
+
+    def __init__(self, name: str):
+        self.name = name
+
+    def __call__(self, fn: Callable) -> Callable:
+        def wrapper(*args, **kwargs):
+            with timeit(self.name):
+                return fn(*args, **kwargs)
+
+        wrapper.__name__ = getattr(fn, "__name__", "wrapped")
+        return wrapper
+
+    def __enter__(self):
+        self.t0 = time.perf_counter()
+        return self
+
+    def __exit__(self, *exc):
+        dt = time.perf_counter() - self.t0
+        entry = self._REG[self.name]
+        entry[0] = entry[0] + dt  # total
+        entry[2] = entry[2] + 1  # count
+        entry[1] = entry[0] / entry[2]  # mean
+
+    @classmethod
+    def todict(cls, percall: bool = True) -> Dict[str, float]:
+        if percall:
+            return {k: v[1] for k, v in cls._REG.items()}
+        return {k: v[0] for k, v in cls._REG.items()}
+
+    @classmethod
+    def erase(cls) -> None:
+        cls._REG.clear()
+
+    @classmethod
+    def print(cls, prefix: str = "") -> None:  # noqa: A003
+        total = sum(v[0] for v in cls._REG.values())
+        for k, (tot, mean, count) in sorted(
+            cls._REG.items(), key=lambda kv: -kv[1][0]
+        ):
+            pct = 100.0 * tot / total if total else 0.0
+            logger.info(
+                f"{prefix}{k}: total={tot:.4f}s mean={mean * 1e3:.3f}ms "
+                f"count={count} ({pct:.1f}%)"
+            )
+
+
+# --------------------------------------------------------------------------- #
+# Profiling gates (reference torchrl/_utils.py:433-515)
+# --------------------------------------------------------------------------- #
+_PROFILING_ENABLED = get_binary_env_var("RL_AMD_PROFILING", False)
+
+
+def set_profiling_enabled(mode: bool) -> None:
+    global _PROFILING_ENABLED
+    _PROFILING_ENABLED = bool(mode)
+
+
+def profiling_enabled() -> bool:
+    return _PROFILING_ENABLED
+
+
+@contextlib.contextmanager
+def record_function(name: str) -> Iterator[None]:
+    """``torch.profiler.record_function`` range, emitted only when profiling
+    is enabled so the hot path stays free of profiler overhead."""
+    if _PROFILING_ENABLED:
+        with torch.profiler.record_function(name):
+            yield
+    else:
+        yield
+
+
+# --------------------------------------------------------------------------- #
+# Seeding
+# --------------------------------------------------------------------------- #
+def seed_everything(seed: int) -> int:
+    import random
+
+    random.seed(seed)
+    np.random.seed(seed % (2**32))
+    torch.manual_seed(seed)
+    if torch.cuda.is_available():
+        torch.cuda.manual_seed_all(seed)
+    return seed
+
+
+def seed_generator(seed: int) -> int:
+    """Next seed in a deterministic chain (for per-worker seeding)."""
+    max_seed_val = (2**32) - 1
+    rng = np.random.default_rng(seed)
+    return int(rng.integers(0, max_seed_val))
+
+
+# --------------------------------------------------------------------------- #
+# Math / misc
+# --------------------------------------------------------------------------- #
+def prod(seq) -> int:
+    return int(math.prod(seq))
+
+
+def hip_is_available() -> bool:
+    """True when a ROCm GPU is visible (torch.cuda IS HIP on ROCm builds)."""
+    return torch.cuda.is_available()
+
+
+def device_of(x: Any) -> Optional[torch.device]:
+    if isinstance(x, torch.Tensor):
+        return x.device
+    dev = getattr(x, "device", None)
+    if dev is not None:
+        return torch.device(dev)
+    return None
+
+
+def expand_as_right(t: torch.Tensor, dest: torch.Tensor) -> torch.Tensor:
+    """Expand ``t`` on the right to match ``dest``'s ndim, then broadcast."""
+    while t.dim() < dest.dim():
+        t = t.unsqueeze(-1)
+    return t.expand_as(dest)
+
+
+def expand_right(t: torch.Tensor, shape) -> torch.Tensor:
+    while t.dim() < len(shape):
+        t = t.unsqueeze(-1)
+    return t.expand(shape)
+
+
+# --------------------------------------------------------------------------- #
+# Multiprocessing helpers (reference torchrl/_utils.py:48-124, :520)
+# --------------------------------------------------------------------------- #
+import multiprocessing as _mp  # noqa: E402
+
+
+class _ProcessNoWarn(_mp.get_context("spawn").Process):
+    """Spawn-context Process that silences warnings in the child and
+    forwards the parent's relevant env vars."""
+
+    def __init__(self, *args, num_threads: int | None = None, **kwargs):
+        self.filter_warnings_subprocess = True
+        self.num_threads = num_threads
+        super().__init__(*args, **kwargs)
+
+    def run(self):
+        if self.num_threads is not None:
+            torch.set_num_threads(self.num_threads)
+        if self.filter_warnings_subprocess:
+            import warnings
+
+            with warnings.catch_warnings():
+                warnings.simplefilter("ignore")
+                return super().run()
+        return super().run()
+
+
+def _check_for_faulty_process(processes) -> None:
+    """Raise if any worker process died; terminate the rest first
+    (reference torchrl/_utils.py:520)."""
+    terminate = False
+    for p in processes:
+        if not p.is_alive():
+            terminate = True
+            break
+    if terminate:
+        for p in processes:
+            if p.is_alive():
+                p.terminate()
+        raise RuntimeError(
+            "At least one simulation process failed; tearing the others down. "
+            "Check the tracebacks above for the root cause."
+        )
+
+
+class _ErrorCatcher:
+    """Wrap a worker target so exceptions propagate with traceback text."""
+
+    def __init__(self, fn):
+        self.fn = fn
+
+    def __call__(self, *args, **kwargs):
+        try:
+            return self.fn(*args, **kwargs)
+        except Exception:
+            traceback.print_exc()
+            raise
+
+
+def implement_for(module_name: str, from_version: str = None, to_version: str = None):
+    """Lightweight version-dispatch decorator (reference uses pyvers'
+    ``implement_for``, torchrl/_utils.py:29).  Here a no-op pass-through that
+    only gates on importability of ``module_name``."""
+
+    def deco(fn):
+        try:
+            __import__(module_name.split(".")[0])
+            fn._implement_for_available = True
+        except ImportError:
+            fn._implement_for_available = False
+        return fn
+
+    return deco
