@@ -1,0 +1,300 @@
+"""Deterministic mock envs for unit testing (reference:
+pytorch/rl torchrl/testing/mocking_classes.py — CountingEnv:1168,
+NestedCountingEnv:1492, EnvWithDynamicSpec:2307, EnvThatErrorsAfter10Iters:2486).
+
+Every collector/transform/objective test runs on these — no simulator deps.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ..data.tensor_specs import (
+    Binary,
+    Bounded,
+    Categorical,
+    Composite,
+    OneHot,
+    Unbounded,
+)
+from ..envs.common import EnvBase
+from ..tensordict import TensorDict, TensorDictBase
+
+__all__ = [
+    "CountingEnv",
+    "NestedCountingEnv",
+    "ContinuousActionVecMockEnv",
+    "DiscreteActionVecMockEnv",
+    "MockSerialEnv",
+    "EnvThatErrors",
+]
+
+
+class CountingEnv(EnvBase):
+    """Obs counts steps since reset; reward always 1; terminates at
+    ``max_steps``.  Bit-for-bit predictable — collectors and GAE outputs are
+    assertable exactly (reference mocking_classes.py:1168)."""
+
+    def __init__(
+        self,
+        max_steps: int = 5,
+        start_val: int = 0,
+        batch_size=(),
+        device=None,
+    ):
+        super().__init__(device=device, batch_size=batch_size)
+        self.max_steps = max_steps
+        self.start_val = start_val
+        bs = self.batch_size
+        self.observation_spec = Composite(
+            {
+                "observation": Unbounded(
+                    shape=(*bs, 1), device=self.device, dtype=torch.float32
+                )
+            },
+            shape=bs,
+            device=self.device,
+        )
+        self.action_spec = Binary(shape=(*bs, 1), device=self.device, dtype=torch.bool)
+        self.reward_spec = Unbounded(shape=(*bs, 1), device=self.device)
+        self.count = torch.zeros(
+            (*bs, 1), dtype=torch.float32, device=self.device
+        )
+
+    def _reset(self, tensordict: Optional[TensorDictBase] = None, **kwargs) -> TensorDictBase:
+        if tensordict is not None and "_reset" in tensordict:
+            mask = tensordict.get("_reset").reshape(self.count.shape)
+            self.count = torch.where(
+                mask, torch.full_like(self.count, float(self.start_val)), self.count
+            )
+        else:
+            self.count.fill_(float(self.start_val))
+        return TensorDict(
+            {
+                "observation": self.count.clone(),
+                "done": torch.zeros_like(self.count, dtype=torch.bool),
+                "terminated": torch.zeros_like(self.count, dtype=torch.bool),
+            },
+            batch_size=self.batch_size,
+            device=self.device,
+        )
+
+    def _step(self, tensordict: TensorDictBase) -> TensorDictBase:
+        action = tensordict.get("action")
+        self.count = self.count + action.to(self.count.dtype).reshape(self.count.shape)
+        done = self.count >= self.max_steps
+        return TensorDict(
+            {
+                "observation": self.count.clone(),
+                "reward": torch.ones_like(self.count),
+                "done": done,
+                "terminated": done,
+            },
+            batch_size=self.batch_size,
+            device=self.device,
+        )
+
+    def _set_seed(self, seed):
+        return seed
+
+
+class NestedCountingEnv(CountingEnv):
+    """CountingEnv with observation nested under ("data", "states")
+    (reference mocking_classes.py:1492)."""
+
+    def __init__(self, max_steps: int = 5, nest_obs: bool = True, batch_size=(), device=None):
+        super().__init__(max_steps=max_steps, batch_size=batch_size, device=device)
+        bs = self.batch_size
+        self.observation_spec = Composite(
+            {
+                "data": Composite(
+                    {
+                        "states": Unbounded(
+                            shape=(*bs, 1), device=self.device, dtype=torch.float32
+                        )
+                    },
+                    shape=bs,
+                    device=self.device,
+                )
+            },
+            shape=bs,
+            device=self.device,
+        )
+
+    def _reset(self, tensordict=None, **kwargs):
+        td = super()._reset(tensordict, **kwargs)
+        obs = td.pop("observation")
+        td.set(("data", "states"), obs)
+        return td
+
+    def _step(self, tensordict):
+        td = super()._step(tensordict)
+        obs = td.pop("observation")
+        td.set(("data", "states"), obs)
+        return td
+
+
+class ContinuousActionVecMockEnv(EnvBase):
+    """Gaussian-dynamics continuous-control mock: obs' = obs + action-norm,
+    reward = -|obs|.  Shapes mimic a MuJoCo-style env."""
+
+    def __init__(
+        self,
+        obs_dim: int = 7,
+        action_dim: int = 5,
+        max_steps: int = 100,
+        batch_size=(),
+        device=None,
+    ):
+        super().__init__(device=device, batch_size=batch_size)
+        bs = self.batch_size
+        self.obs_dim = obs_dim
+        self.action_dim = action_dim
+        self.max_steps = max_steps
+        self.observation_spec = Composite(
+            {"observation": Unbounded(shape=(*bs, obs_dim), device=self.device)},
+            shape=bs,
+            device=self.device,
+        )
+        self.action_spec = Bounded(
+            low=-1.0, high=1.0, shape=(*bs, action_dim), device=self.device
+        )
+        self.reward_spec = Unbounded(shape=(*bs, 1), device=self.device)
+        self._obs = None
+        self._t = None
+        self._gen = torch.Generator(device="cpu")
+
+    def _reset(self, tensordict=None, **kwargs):
+        bs = self.batch_size
+        self._obs = torch.randn(
+            (*bs, self.obs_dim), generator=self._gen
+        ).to(self.device)
+        self._t = torch.zeros((*bs, 1), device=self.device)
+        return TensorDict(
+            {
+                "observation": self._obs.clone(),
+                "done": torch.zeros((*bs, 1), dtype=torch.bool, device=self.device),
+                "terminated": torch.zeros((*bs, 1), dtype=torch.bool, device=self.device),
+            },
+            batch_size=bs,
+            device=self.device,
+        )
+
+    def _step(self, tensordict):
+        action = tensordict.get("action")
+        self._obs = self._obs + action.norm(dim=-1, keepdim=True) * 0.1
+        self._t = self._t + 1
+        reward = -self._obs.norm(dim=-1, keepdim=True)
+        done = self._t >= self.max_steps
+        return TensorDict(
+            {
+                "observation": self._obs.clone(),
+                "reward": reward,
+                "done": done,
+                "terminated": done,
+            },
+            batch_size=self.batch_size,
+            device=self.device,
+        )
+
+    def _set_seed(self, seed):
+        if seed is not None:
+            self._gen.manual_seed(seed)
+        return seed
+
+
+class DiscreteActionVecMockEnv(EnvBase):
+    """Discrete-action mock (CartPole-shaped): obs_dim floats, n_actions
+    one-hot actions."""
+
+    def __init__(
+        self,
+        obs_dim: int = 4,
+        n_actions: int = 2,
+        max_steps: int = 50,
+        categorical: bool = False,
+        batch_size=(),
+        device=None,
+    ):
+        super().__init__(device=device, batch_size=batch_size)
+        bs = self.batch_size
+        self.obs_dim = obs_dim
+        self.n_actions = n_actions
+        self.max_steps = max_steps
+        self.categorical = categorical
+        self.observation_spec = Composite(
+            {"observation": Unbounded(shape=(*bs, obs_dim), device=self.device)},
+            shape=bs,
+            device=self.device,
+        )
+        if categorical:
+            self.action_spec = Categorical(n_actions, shape=bs, device=self.device)
+        else:
+            self.action_spec = OneHot(
+                n_actions, shape=(*bs, n_actions), device=self.device
+            )
+        self.reward_spec = Unbounded(shape=(*bs, 1), device=self.device)
+        self._obs = None
+        self._t = None
+        self._gen = torch.Generator(device="cpu")
+
+    def _reset(self, tensordict=None, **kwargs):
+        bs = self.batch_size
+        self._obs = torch.randn((*bs, self.obs_dim), generator=self._gen).to(self.device) * 0.05
+        self._t = torch.zeros((*bs, 1), device=self.device)
+        return TensorDict(
+            {
+                "observation": self._obs.clone(),
+                "done": torch.zeros((*bs, 1), dtype=torch.bool, device=self.device),
+                "terminated": torch.zeros((*bs, 1), dtype=torch.bool, device=self.device),
+            },
+            batch_size=bs,
+            device=self.device,
+        )
+
+    def _step(self, tensordict):
+        action = tensordict.get("action")
+        if not self.categorical:
+            act_idx = action.to(torch.float32).argmax(-1, keepdim=True)
+        else:
+            act_idx = action.reshape(*self.batch_size, 1)
+        self._obs = self._obs + (act_idx.float() - 0.5) * 0.01
+        self._t = self._t + 1
+        reward = torch.ones((*self.batch_size, 1), device=self.device)
+        done = (self._t >= self.max_steps) | (self._obs.abs().max(-1, keepdim=True).values > 2.0)
+        return TensorDict(
+            {
+                "observation": self._obs.clone(),
+                "reward": reward,
+                "done": done,
+                "terminated": done,
+            },
+            batch_size=self.batch_size,
+            device=self.device,
+        )
+
+    def _set_seed(self, seed):
+        if seed is not None:
+            self._gen.manual_seed(seed)
+        return seed
+
+
+class MockSerialEnv(CountingEnv):
+    """Alias kept for reference-parity test naming."""
+
+
+class EnvThatErrors(CountingEnv):
+    """Raises after N steps — fault-injection fixture
+    (reference mocking_classes.py:2486 ``EnvThatErrorsAfter10Iters``)."""
+
+    def __init__(self, error_at: int = 10, **kwargs):
+        super().__init__(**kwargs)
+        self.error_at = error_at
+        self._n = 0
+
+    def _step(self, tensordict):
+        self._n += 1
+        if self._n >= self.error_at:
+            raise RuntimeError("EnvThatErrors: deliberate failure")
+        return super()._step(tensordict)
