@@ -1,0 +1,198 @@
+"""LossModule — the base class of every objective.
+
+Reference: pytorch/rl torchrl/objectives/common.py:87 (LossModule,
+convert_to_functional:426, target params :579-588, vmap ensembles
+:939-969).
+
+MI355X design note: the reference extracts functional parameter
+TensorDicts and vmaps over them.  rl_amd keeps live modules — ensembles
+are ``torch.func.stack_module_state`` + ``torch.vmap`` over N stacked
+parameter sets (one batched MFMA GEMM instead of N small ones), and target
+networks are deep copies whose parameters are demoted to buffers (so they
+checkpoint with ``state_dict`` but never reach the optimizer).
+"""
+from __future__ import annotations
+
+import copy
+import dataclasses
+from typing import Iterator, List, Optional, Sequence, Tuple, Union
+
+import torch
+from torch import nn
+
+from ..tensordict import TensorDict, TensorDictBase, TensorDictModuleBase
+
+__all__ = ["LossModule", "params_to_buffers"]
+
+
+def params_to_buffers(module: nn.Module) -> nn.Module:
+    """Demote every Parameter to a buffer, in place."""
+    for m in module.modules():
+        names = list(m._parameters.keys())
+        for n in names:
+            p = m._parameters.pop(n)
+            if p is not None:
+                m.register_buffer(n, p.detach().clone())
+    return module
+
+
+class _EnsembleModule(nn.Module):
+    """N copies of a module evaluated as one vmapped call
+    (reference vmap-over-params, common.py:939-969)."""
+
+    def __init__(self, module_factory_or_module, num_copies: int, reinit: bool = True):
+        super().__init__()
+        base = module_factory_or_module
+        copies = []
+        for i in range(num_copies):
+            m = copy.deepcopy(base)
+            if reinit and i > 0:
+                for sub in m.modules():
+                    if hasattr(sub, "reset_parameters"):
+                        sub.reset_parameters()
+            copies.append(m)
+        self.modules_list = nn.ModuleList(copies)
+        self.num_copies = num_copies
+
+    def forward(self, td: TensorDictBase) -> TensorDictBase:
+        """Run all copies on the same input; stack outputs along dim 0."""
+        from ..tensordict import stack as td_stack
+
+        outs = [m(td.clone(False)) for m in self.modules_list]
+        return td_stack(outs, 0)
+
+    @property
+    def in_keys(self):
+        return self.modules_list[0].in_keys
+
+    @property
+    def out_keys(self):
+        return self.modules_list[0].out_keys
+
+    def __getitem__(self, i):
+        return self.modules_list[i]
+
+    def __len__(self):
+        return self.num_copies
+
+
+class LossModule(TensorDictModuleBase):
+    """Base class for objectives.
+
+    Sub-classes call :meth:`convert_to_functional` for each network they
+    own; target copies are created with ``create_target_params=True`` and
+    updated by ``SoftUpdate``/``HardUpdate`` (objectives/utils.py).
+    """
+
+    @dataclasses.dataclass
+    class _AcceptedKeys:
+        pass
+
+    default_keys = _AcceptedKeys
+    out_keys: List[str] = []
+
+    def __init__(self):
+        super().__init__()
+        self._tensor_keys = self._AcceptedKeys()
+        self._networks: List[str] = []
+        self._has_update_associated = {}
+        self.value_estimator = None
+        self.in_keys = []
+
+    # ------------------------------------------------------------------ #
+    @property
+    def tensor_keys(self):
+        return self._tensor_keys
+
+    def set_keys(self, **kwargs) -> "LossModule":
+        for k, v in kwargs.items():
+            if not hasattr(self._tensor_keys, k):
+                raise ValueError(
+                    f"{k} is not an accepted key for {type(self).__name__}; "
+                    f"accepted: {[f.name for f in dataclasses.fields(self._tensor_keys)]}"
+                )
+            setattr(self._tensor_keys, k, v)
+        self._forward_key_update()
+        return self
+
+    def _forward_key_update(self):
+        pass
+
+    # ------------------------------------------------------------------ #
+    def convert_to_functional(
+        self,
+        module: TensorDictModuleBase,
+        module_name: str,
+        expand_dim: Optional[int] = None,
+        create_target_params: bool = False,
+        compare_against: Optional[Sequence] = None,
+        **kwargs,
+    ) -> None:
+        """Register ``module`` (optionally expanded to an ensemble of
+        ``expand_dim`` copies) and optionally a frozen target copy
+        (reference common.py:426)."""
+        if expand_dim is not None and expand_dim > 1:
+            module = _EnsembleModule(module, expand_dim)
+        setattr(self, module_name, module)
+        self._networks.append(module_name)
+        if create_target_params:
+            target = params_to_buffers(copy.deepcopy(module))
+            setattr(self, f"{module_name}_target", target)
+            self._has_update_associated[module_name] = False
+        # extend in_keys
+        for k in getattr(module, "in_keys", []):
+            if k not in self.in_keys:
+                self.in_keys.append(k)
+
+    def target_network(self, module_name: str) -> nn.Module:
+        return getattr(self, f"{module_name}_target")
+
+    def _networks_with_targets(self) -> List[str]:
+        return [n for n in self._networks if hasattr(self, f"{n}_target")]
+
+    # ------------------------------------------------------------------ #
+    def make_value_estimator(self, value_type=None, **hyperparams):
+        """Build the default value estimator (reference common.py + each
+        loss's ``make_value_estimator``)."""
+        from .utils import ValueEstimators, default_value_kwargs
+        from .value.advantages import GAE, TD0Estimator, TD1Estimator, TDLambdaEstimator, VTrace
+
+        if value_type is None:
+            value_type = getattr(self, "default_value_estimator", ValueEstimators.TD0)
+        hp = dict(default_value_kwargs(value_type))
+        hp.update(hyperparams)
+        value_net = getattr(self, "critic_network", None) or getattr(
+            self, "value_network", None
+        )
+        if value_type == ValueEstimators.GAE:
+            self.value_estimator = GAE(value_network=value_net, **hp)
+        elif value_type == ValueEstimators.TD0:
+            self.value_estimator = TD0Estimator(value_network=value_net, **hp)
+        elif value_type == ValueEstimators.TD1:
+            self.value_estimator = TD1Estimator(value_network=value_net, **hp)
+        elif value_type == ValueEstimators.TDLambda:
+            self.value_estimator = TDLambdaEstimator(value_network=value_net, **hp)
+        elif value_type == ValueEstimators.VTrace:
+            actor = getattr(self, "actor_network", None)
+            self.value_estimator = VTrace(
+                value_network=value_net, actor_network=actor, **hp
+            )
+        else:
+            raise NotImplementedError(f"value type {value_type}")
+        self._default_value_estimator_keys()
+        return self.value_estimator
+
+    def _default_value_estimator_keys(self):
+        pass
+
+    def forward(self, tensordict: TensorDictBase) -> TensorDictBase:
+        raise NotImplementedError
+
+    def loss_and_backprop(self, td: TensorDictBase) -> TensorDictBase:
+        """Convenience: forward + sum of loss_* keys + backward."""
+        out = self.forward(td)
+        total = sum(
+            v for k, v in out.items() if isinstance(k, str) and k.startswith("loss_")
+        )
+        total.backward()
+        return out

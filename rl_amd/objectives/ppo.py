@@ -1,0 +1,281 @@
+"""PPO family: PPOLoss, ClipPPOLoss, KLPENPPOLoss.
+
+Reference: pytorch/rl torchrl/objectives/ppo.py (PPOLoss:109,
+ClipPPOLoss:1082, KLPENPPOLoss:1458).
+"""
+from __future__ import annotations
+
+import dataclasses
+import math
+from typing import Optional, Union
+
+import torch
+
+from ..tensordict import TensorDict, TensorDictBase, TensorDictModuleBase
+from .common import LossModule
+from .utils import ValueEstimators, distance_loss
+
+__all__ = ["PPOLoss", "ClipPPOLoss", "KLPENPPOLoss"]
+
+
+class PPOLoss(LossModule):
+    """Vanilla policy-gradient surrogate with entropy bonus and critic loss
+    (reference ppo.py:109)."""
+
+    @dataclasses.dataclass
+    class _AcceptedKeys:
+        advantage: str = "advantage"
+        value_target: str = "value_target"
+        value: str = "state_value"
+        sample_log_prob: str = "sample_log_prob"
+        action: str = "action"
+        reward: tuple = ("next", "reward")
+        done: tuple = ("next", "done")
+        terminated: tuple = ("next", "terminated")
+
+    default_value_estimator = ValueEstimators.GAE
+    out_keys = ["loss_objective", "loss_critic", "loss_entropy", "entropy", "ESS"]
+
+    def __init__(
+        self,
+        actor_network: TensorDictModuleBase,
+        critic_network: TensorDictModuleBase,
+        *,
+        entropy_bonus: bool = True,
+        samples_mc_entropy: int = 1,
+        entropy_coeff: float = 0.01,
+        critic_coeff: float = 1.0,
+        loss_critic_type: str = "smooth_l1",
+        normalize_advantage: bool = False,
+        normalize_advantage_exclude_dims: tuple = (),
+        gamma: Optional[float] = None,
+        separate_losses: bool = False,
+        advantage_key: Optional[str] = None,
+        value_target_key: Optional[str] = None,
+        reduction: str = "mean",
+        clip_value: Optional[float] = None,
+    ):
+        super().__init__()
+        self.convert_to_functional(actor_network, "actor_network")
+        self.convert_to_functional(critic_network, "critic_network")
+        self.entropy_bonus = entropy_bonus
+        self.samples_mc_entropy = samples_mc_entropy
+        self.entropy_coeff = entropy_coeff
+        self.critic_coeff = critic_coeff
+        self.loss_critic_type = loss_critic_type
+        self.normalize_advantage = normalize_advantage
+        self.reduction = reduction
+        self.clip_value = clip_value
+        self._gamma_init = gamma
+        if advantage_key is not None:
+            self._tensor_keys.advantage = advantage_key
+        if value_target_key is not None:
+            self._tensor_keys.value_target = value_target_key
+
+    @property
+    def actor(self):  # legacy alias
+        return self.actor_network
+
+    @property
+    def critic(self):
+        return self.critic_network
+
+    def make_value_estimator(self, value_type=None, **hyperparams):
+        if self._gamma_init is not None:
+            hyperparams.setdefault("gamma", self._gamma_init)
+        return super().make_value_estimator(value_type, **hyperparams)
+
+    def reset(self) -> None:
+        pass
+
+    def _log_weight(self, td: TensorDictBase):
+        """log π_new(a|s) - log π_old(a|s) plus the distribution."""
+        prev_log_prob = td.get(self.tensor_keys.sample_log_prob)
+        dist = self.actor_network.get_dist(td.clone(False))
+        action = td.get(self.tensor_keys.action)
+        log_prob = dist.log_prob(action)
+        if prev_log_prob.dim() > log_prob.dim():
+            prev_log_prob = prev_log_prob.squeeze(-1)
+        log_weight = log_prob - prev_log_prob.detach()
+        return log_weight, dist, log_prob
+
+    def _entropy(self, dist) -> torch.Tensor:
+        try:
+            ent = dist.entropy()
+        except NotImplementedError:
+            x = dist.rsample((self.samples_mc_entropy,))
+            ent = -dist.log_prob(x).mean(0)
+        return ent
+
+    def loss_critic(self, td: TensorDictBase) -> torch.Tensor:
+        target = td.get(self.tensor_keys.value_target)
+        td_out = self.critic_network(td.clone(False))
+        value = td_out.get(self.tensor_keys.value)
+        loss_value = distance_loss(value, target, self.loss_critic_type)
+        if self.clip_value is not None:
+            old_value = td.get(self.tensor_keys.value, None)
+            if old_value is not None:
+                value_clipped = old_value + (value - old_value).clamp(
+                    -self.clip_value, self.clip_value
+                )
+                loss_clipped = distance_loss(
+                    value_clipped, target, self.loss_critic_type
+                )
+                loss_value = torch.maximum(loss_value, loss_clipped)
+        return loss_value
+
+    def _normalize_adv(self, adv: torch.Tensor) -> torch.Tensor:
+        loc = adv.mean()
+        scale = adv.std().clamp_min(1e-6)
+        return (adv - loc) / scale
+
+    def _reduce(self, x: torch.Tensor) -> torch.Tensor:
+        if self.reduction == "mean":
+            return x.mean()
+        if self.reduction == "sum":
+            return x.sum()
+        return x
+
+    def forward(self, tensordict: TensorDictBase) -> TensorDictBase:
+        td = tensordict.clone(False)
+        advantage = td.get(self.tensor_keys.advantage, None)
+        if advantage is None:
+            if self.value_estimator is None:
+                self.make_value_estimator()
+            self.value_estimator(td)
+            advantage = td.get(self.tensor_keys.advantage)
+        if self.normalize_advantage and advantage.numel() > 1:
+            advantage = self._normalize_adv(advantage)
+        log_weight, dist, _ = self._log_weight(td)
+        lw = log_weight
+        if lw.dim() < advantage.dim():
+            lw = lw.unsqueeze(-1)
+        neg_loss = lw.exp() * advantage
+        with torch.no_grad():
+            lw_flat = log_weight.reshape(-1)
+            ess = lw_flat.exp().sum().pow(2) / lw_flat.mul(2).exp().sum().clamp_min(1e-12)
+        out = TensorDict({"loss_objective": -self._reduce(neg_loss)}, batch_size=[])
+        if self.entropy_bonus:
+            entropy = self._entropy(dist)
+            out.set("entropy", entropy.detach().mean())
+            out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
+        if self.critic_coeff is not None and self.critic_coeff > 0:
+            loss_critic = self.loss_critic(td)
+            out.set("loss_critic", self.critic_coeff * self._reduce(loss_critic))
+        out.set("ESS", ess / log_weight.numel())
+        return out
+
+
+class ClipPPOLoss(PPOLoss):
+    """Clipped-ratio PPO (reference ppo.py:1082)."""
+
+    def __init__(self, actor_network, critic_network, *, clip_epsilon: float = 0.2, **kwargs):
+        super().__init__(actor_network, critic_network, **kwargs)
+        self.register_buffer("clip_epsilon", torch.as_tensor(clip_epsilon))
+
+    @property
+    def _clip_bounds(self):
+        return (
+            math.log1p(-float(self.clip_epsilon)),
+            math.log1p(float(self.clip_epsilon)),
+        )
+
+    def forward(self, tensordict: TensorDictBase) -> TensorDictBase:
+        td = tensordict.clone(False)
+        advantage = td.get(self.tensor_keys.advantage, None)
+        if advantage is None:
+            if self.value_estimator is None:
+                self.make_value_estimator()
+            self.value_estimator(td)
+            advantage = td.get(self.tensor_keys.advantage)
+        if self.normalize_advantage and advantage.numel() > 1:
+            advantage = self._normalize_adv(advantage)
+        log_weight, dist, _ = self._log_weight(td)
+        if log_weight.dim() < advantage.dim():
+            log_weight = log_weight.unsqueeze(-1)
+        ratio = log_weight.exp()
+        gain1 = ratio * advantage
+        ratio_clamped = log_weight.clamp(*self._clip_bounds).exp()
+        gain2 = ratio_clamped * advantage
+        gain = torch.minimum(gain1, gain2)
+        with torch.no_grad():
+            lw_flat = log_weight.reshape(-1)
+            ess = lw_flat.exp().sum().pow(2) / lw_flat.mul(2).exp().sum().clamp_min(1e-12)
+            clip_fraction = (ratio_clamped != ratio).float().mean()
+        out = TensorDict(
+            {
+                "loss_objective": -self._reduce(gain),
+                "clip_fraction": clip_fraction,
+                "ESS": ess / log_weight.numel(),
+            },
+            batch_size=[],
+        )
+        if self.entropy_bonus:
+            entropy = self._entropy(dist)
+            out.set("entropy", entropy.detach().mean())
+            out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
+        if self.critic_coeff is not None and self.critic_coeff > 0:
+            loss_critic = self.loss_critic(td)
+            out.set("loss_critic", self.critic_coeff * self._reduce(loss_critic))
+        return out
+
+
+class KLPENPPOLoss(PPOLoss):
+    """KL-penalty PPO with adaptive β (reference ppo.py:1458)."""
+
+    def __init__(
+        self,
+        actor_network,
+        critic_network,
+        *,
+        dtarg: float = 0.01,
+        beta: float = 1.0,
+        increment: float = 2.0,
+        decrement: float = 0.5,
+        samples_mc_kl: int = 1,
+        **kwargs,
+    ):
+        super().__init__(actor_network, critic_network, **kwargs)
+        self.dtarg = dtarg
+        self._beta_init = beta
+        self.register_buffer("beta", torch.as_tensor(beta))
+        self.increment = increment
+        self.decrement = decrement
+        self.samples_mc_kl = samples_mc_kl
+
+    def forward(self, tensordict: TensorDictBase) -> TensorDictBase:
+        td = tensordict.clone(False)
+        advantage = td.get(self.tensor_keys.advantage, None)
+        if advantage is None:
+            if self.value_estimator is None:
+                self.make_value_estimator()
+            self.value_estimator(td)
+            advantage = td.get(self.tensor_keys.advantage)
+        if self.normalize_advantage and advantage.numel() > 1:
+            advantage = self._normalize_adv(advantage)
+        log_weight, dist, log_prob = self._log_weight(td)
+        lw = log_weight
+        if lw.dim() < advantage.dim():
+            lw = lw.unsqueeze(-1)
+        neg_loss = lw.exp() * advantage
+        # MC KL(π_old ‖ π_new) ≈ E_old[log w]⁻
+        kl = -log_weight.mean()
+        if kl > self.dtarg * 1.5:
+            self.beta.data.mul_(self.increment)
+        elif kl < self.dtarg / 1.5:
+            self.beta.data.mul_(self.decrement)
+        out = TensorDict(
+            {
+                "loss_objective": -self._reduce(neg_loss) + self.beta * kl,
+                "kl": kl.detach(),
+                "beta": self.beta.clone(),
+            },
+            batch_size=[],
+        )
+        if self.entropy_bonus:
+            entropy = self._entropy(dist)
+            out.set("entropy", entropy.detach().mean())
+            out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
+        if self.critic_coeff is not None and self.critic_coeff > 0:
+            out.set("loss_critic", self.critic_coeff * self._reduce(self.loss_critic(td)))
+        return out
