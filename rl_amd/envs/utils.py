@@ -180,6 +180,11 @@ class RandomPolicy:
         self.action_key = action_key
 
     def __call__(self, td: TensorDictBase) -> TensorDictBase:
+        from ..data.tensor_specs import Composite
+
         # spec shape already includes env batch dims
-        td.set(self.action_key, self.action_spec.rand())
+        if isinstance(self.action_spec, Composite):
+            td.update(self.action_spec.rand())
+        else:
+            td.set(self.action_key, self.action_spec.rand())
         return td

@@ -102,6 +102,17 @@ def _shape_of_index(batch_size: torch.Size, idx) -> torch.Size:
         start, stop, step = idx.indices(batch_size[0])
         n = max(0, (stop - start + (step - 1 if step > 0 else step + 1)) // step)
         return torch.Size([n, *batch_size[1:]])
+    if isinstance(idx, torch.Tensor) and idx.dtype == torch.bool:
+        # boolean mask over the leading dims: result is [n_true, *rest]
+        n = int(idx.sum())
+        return torch.Size([n, *batch_size[idx.dim():]])
+    if isinstance(idx, tuple) and any(
+        isinstance(i, torch.Tensor) and i.dtype == torch.bool for i in idx
+    ):
+        # rare path: bool mask inside a tuple — use a real (1-byte) dummy
+        return torch.zeros(batch_size, dtype=torch.bool, device="cpu")[
+            tuple(i.cpu() if isinstance(i, torch.Tensor) else i for i in idx)
+        ].shape
     # general path: meta-tensor indexing
     t = torch.empty(batch_size, device="meta")
     if isinstance(idx, tuple):

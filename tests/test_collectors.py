@@ -1,0 +1,217 @@
+"""Collector tests: single-process hot loop, RB integration, multiprocess."""
+import pytest
+import torch
+
+from rl_amd.collectors import (
+    AsyncCollector,
+    Collector,
+    MultiAsyncCollector,
+    MultiSyncCollector,
+    split_trajectories,
+)
+from rl_amd.data import LazyTensorStorage, TensorDictReplayBuffer
+from rl_amd.envs.transforms import InitTracker, StepCounter, TransformedEnv
+from rl_amd.modules import MLP, ProbabilisticActor, TanhNormal, NormalParamExtractor
+from rl_amd.tensordict import TensorDict, TensorDictModule
+from rl_amd.testing import ContinuousActionVecMockEnv, CountingEnv
+
+
+def make_env():
+    return ContinuousActionVecMockEnv(batch_size=[2], max_steps=10)
+
+
+def make_policy(obs_dim=7, act_dim=5):
+    net = torch.nn.Sequential(
+        MLP(in_features=obs_dim, out_features=2 * act_dim, num_cells=[16]),
+        NormalParamExtractor(),
+    )
+    mod = TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"])
+    return ProbabilisticActor(
+        mod, in_keys=["loc", "scale"], distribution_class=TanhNormal, return_log_prob=True
+    )
+
+
+class TestCollector:
+    def test_frames_exact(self):
+        col = Collector(make_env(), frames_per_batch=40, total_frames=120)
+        frames = sum(b.numel() for b in col)
+        assert frames == 120
+        col.shutdown()
+
+    def test_batch_shape(self):
+        col = Collector(
+            ContinuousActionVecMockEnv(batch_size=[4]), frames_per_batch=64, total_frames=64
+        )
+        batch = next(iter(col))
+        assert tuple(batch.batch_size) == (4, 16)
+        assert ("next", "observation") in batch.keys(True, True)
+        col.shutdown()
+
+    def test_unbatched_env(self):
+        col = Collector(CountingEnv(max_steps=4), frames_per_batch=12, total_frames=12)
+        batch = next(iter(col))
+        assert tuple(batch.batch_size) == (12,)
+        col.shutdown()
+
+    def test_policy(self):
+        col = Collector(
+            make_env(), make_policy(), frames_per_batch=20, total_frames=20
+        )
+        batch = next(iter(col))
+        assert "sample_log_prob" in batch
+        assert batch["action"].abs().max() <= 1.0
+        col.shutdown()
+
+    def test_traj_ids_advance(self):
+        col = Collector(
+            CountingEnv(max_steps=3, batch_size=[2]), frames_per_batch=24, total_frames=24
+        )
+        batch = next(iter(col))
+        ids = batch.get(("collector", "traj_ids"))
+        # 12 steps per env, 3 steps per traj → several distinct ids
+        assert ids.max() > 1
+        col.shutdown()
+
+    def test_reset_at_each_iter(self):
+        env = CountingEnv(max_steps=100, batch_size=[2])
+        col = Collector(
+            env, frames_per_batch=8, total_frames=16, reset_at_each_iter=True
+        )
+        batches = list(col)
+        first_obs = batches[1].get("observation")[:, 0]
+        assert (first_obs == 0).all()
+        col.shutdown()
+
+    def test_replay_buffer_mode(self):
+        rb = TensorDictReplayBuffer(storage=LazyTensorStorage(500), batch_size=16)
+        col = Collector(
+            make_env(), frames_per_batch=40, total_frames=80, replay_buffer=rb
+        )
+        outs = [b for b in col]
+        assert all(b is None for b in outs)
+        assert len(rb) == 80
+        col.shutdown()
+
+    def test_init_random_frames(self):
+        col = Collector(
+            make_env(),
+            make_policy(),
+            frames_per_batch=20,
+            total_frames=40,
+            init_random_frames=20,
+        )
+        b1 = next(iter(col))
+        assert "sample_log_prob" not in b1 or b1.get("sample_log_prob", None) is None
+        col.shutdown()
+
+    def test_state_dict_roundtrip(self):
+        pol = make_policy()
+        col = Collector(make_env(), pol, frames_per_batch=20, total_frames=40)
+        next(iter(col))
+        sd = col.state_dict()
+        assert sd["frames"] == 20
+        col2 = Collector(make_env(), make_policy(), frames_per_batch=20, total_frames=40)
+        col2.load_state_dict(sd)
+        assert col2._frames == 20
+        col.shutdown()
+        col2.shutdown()
+
+    def test_update_policy_weights(self):
+        pol = make_policy()
+        col = Collector(make_env(), make_policy(), frames_per_batch=20, total_frames=20)
+        col.update_policy_weights_(pol)
+        for p1, p2 in zip(col.policy.parameters(), pol.parameters()):
+            assert torch.allclose(p1, p2)
+        col.shutdown()
+
+    def test_transformed_env(self):
+        env = TransformedEnv(make_env(), StepCounter())
+        col = Collector(env, frames_per_batch=20, total_frames=20)
+        batch = next(iter(col))
+        assert ("next", "step_count") in batch.keys(True, True)
+        col.shutdown()
+
+    def test_start_background(self):
+        rb = TensorDictReplayBuffer(storage=LazyTensorStorage(500), batch_size=8)
+        col = Collector(
+            make_env(), frames_per_batch=20, total_frames=100, replay_buffer=rb
+        )
+        col.start()
+        import time
+
+        t0 = time.time()
+        while len(rb) < 20 and time.time() - t0 < 10:
+            time.sleep(0.05)
+        assert len(rb) >= 20
+        col.async_shutdown()
+        col.shutdown()
+
+
+class TestSplitTrajs:
+    def test_split_shapes(self):
+        col = Collector(
+            CountingEnv(max_steps=3, batch_size=[2]),
+            frames_per_batch=20,
+            total_frames=20,
+            split_trajs=True,
+        )
+        batch = next(iter(col))
+        assert "mask" in batch
+        assert batch.batch_size[0] >= 2  # several trajectories split out
+        # masked-out steps are zero
+        col.shutdown()
+
+
+class TestMultiCollectors:
+    @pytest.mark.parametrize("cls", [MultiSyncCollector, MultiAsyncCollector])
+    def test_frames(self, cls):
+        col = cls([make_env] * 2, frames_per_batch=40, total_frames=80)
+        frames = 0
+        for b in col:
+            frames += b.numel()
+        assert frames == 80
+        col.shutdown()
+
+    def test_sync_batch_shape(self):
+        col = MultiSyncCollector([make_env] * 2, frames_per_batch=40, total_frames=40)
+        b = next(iter(col))
+        # stack of 2 workers × [2 envs, 10 steps]
+        assert tuple(b.batch_size) == (2, 2, 10)
+        col.shutdown()
+
+    def test_sync_cat_results(self):
+        col = MultiSyncCollector(
+            [make_env] * 2, frames_per_batch=40, total_frames=40, cat_results=0
+        )
+        b = next(iter(col))
+        assert tuple(b.batch_size) == (4, 10)
+        col.shutdown()
+
+    def test_with_policy(self):
+        col = MultiSyncCollector(
+            [make_env] * 2, make_policy(), frames_per_batch=40, total_frames=40
+        )
+        b = next(iter(col))
+        assert "sample_log_prob" in b.keys(True, True)
+        col.shutdown()
+
+    def test_update_weights(self):
+        pol = make_policy()
+        col = MultiSyncCollector(
+            [make_env] * 2, pol, frames_per_batch=40, total_frames=40
+        )
+        col.update_policy_weights_()
+        b = next(iter(col))
+        assert b is not None
+        col.shutdown()
+
+    def test_async_collector_single(self):
+        col = AsyncCollector(make_env, frames_per_batch=20, total_frames=40)
+        frames = sum(b.numel() for b in col)
+        assert frames == 40
+        col.shutdown()
+
+    def test_seed(self):
+        col = MultiSyncCollector([make_env] * 2, frames_per_batch=40, total_frames=40)
+        col.set_seed(42)
+        col.shutdown()
