@@ -1,0 +1,175 @@
+// pybind bindings for the rl_amd native extension (`rl_amd._C`).
+//
+// Mirrors the reference's pybind module `torchrl._torchrl`
+// (pytorch/rl torchrl/csrc/pybind.cpp:21-38): CPU segment trees +
+// device kernels + safetanh — here the device side is HIP/CDNA4 and the
+// value scans are first-class.
+
+#include <torch/extension.h>
+
+#include <vector>
+
+#include "segment_tree_cpu.h"
+
+#ifdef RL_AMD_WITH_HIP
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+extern "C" {
+void launch_gae_f32(const float*, const float*, const float*, const bool*,
+                    const bool*, float*, float*, long, long, float, float,
+                    void*);
+void launch_gae_bf16(const void*, const void*, const void*, const bool*,
+                     const bool*, void*, void*, long, long, float, float,
+                     void*);
+void launch_revscan_f32(const float*, const float*, float*, long, long, void*);
+void launch_revscan_bf16(const void*, const void*, void*, long, long, void*);
+void launch_vtrace_f32(const float*, const float*, const float*, const float*,
+                       const float*, const bool*, const bool*, float*, float*,
+                       long, long, float, float, float, void*);
+void launch_tree_scan_f64(const double*, const double*, long*, long, long,
+                          long, void*);
+void launch_tree_update_f64(double*, double*, int*, const long*, const double*,
+                            long, long, int, void*);
+}
+
+static void check_gae_args(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on a HIP device");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// rows = product of batch dims; tensors viewed [rows, T]
+std::vector<torch::Tensor> gae(torch::Tensor reward, torch::Tensor value,
+                               torch::Tensor next_value, torch::Tensor done,
+                               torch::Tensor terminated, double gamma,
+                               double lmbda) {
+  check_gae_args(reward, "reward");
+  auto sizes = reward.sizes();
+  long T = sizes[sizes.size() - 1];
+  long B = reward.numel() / T;
+  auto adv = torch::empty_like(reward);
+  auto vt = torch::empty_like(reward);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (reward.scalar_type() == torch::kFloat32) {
+    launch_gae_f32(reward.data_ptr<float>(), value.data_ptr<float>(),
+                   next_value.data_ptr<float>(), done.data_ptr<bool>(),
+                   terminated.data_ptr<bool>(), adv.data_ptr<float>(),
+                   vt.data_ptr<float>(), B, T, (float)gamma, (float)lmbda,
+                   (void*)stream);
+  } else if (reward.scalar_type() == torch::kBFloat16) {
+    launch_gae_bf16(reward.data_ptr(), value.data_ptr(), next_value.data_ptr(),
+                    done.data_ptr<bool>(), terminated.data_ptr<bool>(),
+                    adv.data_ptr(), vt.data_ptr(), B, T, (float)gamma,
+                    (float)lmbda, (void*)stream);
+  } else {
+    TORCH_CHECK(false, "gae: dtype must be float32 or bfloat16");
+  }
+  return {adv, vt};
+}
+
+torch::Tensor revscan(torch::Tensor a, torch::Tensor b) {
+  check_gae_args(a, "a");
+  auto sizes = a.sizes();
+  long T = sizes[sizes.size() - 1];
+  long B = a.numel() / T;
+  auto y = torch::empty_like(b);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (a.scalar_type() == torch::kFloat32) {
+    launch_revscan_f32(a.data_ptr<float>(), b.data_ptr<float>(),
+                       y.data_ptr<float>(), B, T, (void*)stream);
+  } else if (a.scalar_type() == torch::kBFloat16) {
+    launch_revscan_bf16(a.data_ptr(), b.data_ptr(), y.data_ptr(), B, T,
+                        (void*)stream);
+  } else {
+    TORCH_CHECK(false, "revscan: dtype must be float32 or bfloat16");
+  }
+  return y;
+}
+
+std::vector<torch::Tensor> vtrace(torch::Tensor log_pi, torch::Tensor log_mu,
+                                  torch::Tensor reward, torch::Tensor value,
+                                  torch::Tensor next_value, torch::Tensor done,
+                                  torch::Tensor terminated, double gamma,
+                                  double rho_thresh, double c_thresh) {
+  check_gae_args(reward, "reward");
+  TORCH_CHECK(reward.scalar_type() == torch::kFloat32,
+              "vtrace: fp32 only (cast inputs)");
+  auto sizes = reward.sizes();
+  long T = sizes[sizes.size() - 1];
+  long B = reward.numel() / T;
+  auto adv = torch::empty_like(reward);
+  auto vs = torch::empty_like(reward);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_vtrace_f32(log_pi.data_ptr<float>(), log_mu.data_ptr<float>(),
+                    reward.data_ptr<float>(), value.data_ptr<float>(),
+                    next_value.data_ptr<float>(), done.data_ptr<bool>(),
+                    terminated.data_ptr<bool>(), adv.data_ptr<float>(),
+                    vs.data_ptr<float>(), B, T, (float)gamma, (float)rho_thresh,
+                    (float)c_thresh, (void*)stream);
+  return {adv, vs};
+}
+
+// Device segment trees: fused inverse-CDF descent over a sum tree in HBM.
+torch::Tensor tree_scan_lower_bound(torch::Tensor tree, torch::Tensor mass,
+                                    long size, long capacity) {
+  TORCH_CHECK(tree.is_cuda() && mass.is_cuda(), "tree/mass must be on device");
+  TORCH_CHECK(tree.scalar_type() == torch::kFloat64, "tree must be f64");
+  auto out = torch::empty({mass.numel()},
+                          torch::dtype(torch::kLong).device(mass.device()));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_tree_scan_f64(tree.data_ptr<double>(), mass.data_ptr<double>(),
+                       out.data_ptr<long>(), mass.numel(), size, capacity,
+                       (void*)stream);
+  return out;
+}
+
+// Fused leaf-scatter + two-pass path recompute; `cnt` is a persistent
+// int32 workspace of 2*size zeros (self-restoring across calls).
+void tree_update(torch::Tensor sum_tree, torch::Tensor min_tree,
+                 torch::Tensor cnt, torch::Tensor index, torch::Tensor value,
+                 long size, bool with_min) {
+  TORCH_CHECK(sum_tree.is_cuda(), "tree must be on device");
+  TORCH_CHECK(cnt.scalar_type() == torch::kInt32, "cnt workspace must be i32");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_tree_update_f64(sum_tree.data_ptr<double>(),
+                         with_min ? min_tree.data_ptr<double>() : nullptr,
+                         cnt.data_ptr<int>(), index.data_ptr<long>(),
+                         value.data_ptr<double>(), index.numel(), size,
+                         with_min ? 1 : 0, (void*)stream);
+}
+#endif  // RL_AMD_WITH_HIP
+
+// ---------------------------------------------------------------------------
+// safetanh / safeatanh (reference torchrl/csrc/utils.cpp:9-48)
+// ---------------------------------------------------------------------------
+torch::Tensor safetanh_fwd(torch::Tensor x, double eps) {
+  auto out = x.tanh();
+  return out.clamp(-1.0 + eps, 1.0 - eps);
+}
+
+torch::Tensor safeatanh_fwd(torch::Tensor y, double eps) {
+  return y.clamp(-1.0 + eps, 1.0 - eps).atanh();
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "rl_amd native ops (CPU segment trees + CDNA4 HIP kernels)";
+  m.attr("COMPILED_WITH_HIP") =
+#ifdef RL_AMD_WITH_HIP
+      true;
+#else
+      false;
+#endif
+
+  rl_amd::define_segment_trees(m);
+  m.def("safetanh", &safetanh_fwd, "clamped tanh");
+  m.def("safeatanh", &safeatanh_fwd, "clamped atanh");
+
+#ifdef RL_AMD_WITH_HIP
+  m.def("gae", &gae, "fused GAE scan (HIP)");
+  m.def("revscan", &revscan, "generic reverse linear-recurrence scan (HIP)");
+  m.def("vtrace", &vtrace, "fused V-trace (HIP)");
+  m.def("tree_scan_lower_bound", &tree_scan_lower_bound,
+        "sum-tree inverse-CDF descent (HIP)");
+  m.def("tree_update", &tree_update, "segment-tree leaf update+recompute (HIP)");
+#endif
+}

@@ -256,20 +256,34 @@ class GAE(ValueEstimatorBase):
             raise RuntimeError("GAE expects a tensordict with a time dimension")
         value, next_value = self._call_value_nets(tensordict)
         reward, done, terminated = self._get_done_terminated_reward(tensordict)
-        fn = (
-            F.vec_generalized_advantage_estimate
-            if self.vectorized
-            else F.generalized_advantage_estimate
-        )
-        adv, value_target = fn(
-            float(self.gamma),
-            float(self.lmbda),
-            value,
-            next_value,
-            reward,
-            done,
-            terminated,
-        )
+        if reward.is_cuda and self.vectorized and not self.differentiable:
+            # fused single-kernel HIP scan (rl_amd/csrc/value_scan.hip)
+            from ... import ops
+
+            adv, value_target = ops.gae(
+                float(self.gamma),
+                float(self.lmbda),
+                value,
+                next_value,
+                reward,
+                done,
+                terminated,
+            )
+        else:
+            fn = (
+                F.vec_generalized_advantage_estimate
+                if self.vectorized
+                else F.generalized_advantage_estimate
+            )
+            adv, value_target = fn(
+                float(self.gamma),
+                float(self.lmbda),
+                value,
+                next_value,
+                reward,
+                done,
+                terminated,
+            )
         if self.average_gae:
             loc = adv.mean()
             scale = adv.std().clamp_min(1e-6)

@@ -1,0 +1,185 @@
+"""rl_amd.ops — dispatch layer over the native extension (rl_amd._C).
+
+On a HIP device the fused CDNA4 kernels run; on CPU the torch scan
+implementations (objectives/value/functional.py) are the fallback.  If a
+GPU is present but the extension is missing, ops raise — GPU runs must
+never silently fall back to eager torch (that would be an unmeasured
+perf regression masquerading as success).
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+_C = None
+_C_ERR: Optional[str] = None
+try:
+    from rl_amd import _C  # type: ignore
+except Exception as e:  # pragma: no cover
+    _C_ERR = repr(e)
+
+HAS_EXT = _C is not None
+HAS_HIP_EXT = bool(HAS_EXT and getattr(_C, "COMPILED_WITH_HIP", False))
+
+_ALLOW_FALLBACK = os.environ.get("RL_AMD_ALLOW_EAGER_FALLBACK", "0") == "1"
+
+
+def _require_ext():
+    if not HAS_HIP_EXT and not _ALLOW_FALLBACK:
+        raise RuntimeError(
+            "rl_amd._C HIP extension not available on a GPU device "
+            f"(import error: {_C_ERR}). Build it with "
+            "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950), "
+            "or set RL_AMD_ALLOW_EAGER_FALLBACK=1 to run the slow torch path."
+        )
+    return HAS_HIP_EXT
+
+
+def ext_module():
+    return _C
+
+
+def _flat_time(x: torch.Tensor) -> torch.Tensor:
+    """[*, T, 1] or [*, T] → contiguous [*, T] view."""
+    if x.shape[-1] == 1 and x.dim() >= 2:
+        x = x.squeeze(-1)
+    return x.contiguous()
+
+
+def gae(
+    gamma: float,
+    lmbda: float,
+    state_value: torch.Tensor,
+    next_state_value: torch.Tensor,
+    reward: torch.Tensor,
+    done: torch.Tensor,
+    terminated: torch.Tensor,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Fused GAE: returns (advantage, value_target) with the input shape."""
+    if reward.is_cuda and _require_ext():
+        shape = reward.shape
+        r = _flat_time(reward)
+        v = _flat_time(state_value).to(r.dtype)
+        nv = _flat_time(next_state_value).to(r.dtype)
+        d = _flat_time(done)
+        tm = _flat_time(terminated)
+        adv, vt = _C.gae(r, v, nv, d, tm, float(gamma), float(lmbda))
+        return adv.reshape(shape), vt.reshape(shape)
+    from ..objectives.value import functional as F
+
+    return F.vec_generalized_advantage_estimate(
+        gamma, lmbda, state_value, next_state_value, reward, done, terminated
+    )
+
+
+def revscan(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """y[t] = b[t] + a[t]·y[t+1] over the last dim."""
+    if a.is_cuda and _require_ext():
+        return _C.revscan(a.contiguous(), b.contiguous())
+    from ..objectives.value.functional import _reverse_scan
+
+    return _reverse_scan(b.unsqueeze(-1), a.unsqueeze(-1)).squeeze(-1)
+
+
+def vtrace(
+    gamma: float,
+    log_pi: torch.Tensor,
+    log_mu: torch.Tensor,
+    state_value: torch.Tensor,
+    next_state_value: torch.Tensor,
+    reward: torch.Tensor,
+    done: torch.Tensor,
+    terminated: torch.Tensor,
+    rho_thresh: float = 1.0,
+    c_thresh: float = 1.0,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if reward.is_cuda and _require_ext():
+        shape = reward.shape
+        adv, vs = _C.vtrace(
+            _flat_time(log_pi).float(),
+            _flat_time(log_mu).float(),
+            _flat_time(reward).float(),
+            _flat_time(state_value).float(),
+            _flat_time(next_state_value).float(),
+            _flat_time(done),
+            _flat_time(terminated),
+            float(gamma),
+            float(rho_thresh),
+            float(c_thresh),
+        )
+        return adv.reshape(shape), vs.reshape(shape)
+    from ..objectives.value import functional as F
+
+    return F.vec_vtrace_advantage_estimate(
+        gamma,
+        log_pi,
+        log_mu,
+        state_value,
+        next_state_value,
+        reward,
+        done,
+        terminated,
+        rho_thresh,
+        c_thresh,
+    )
+
+
+def safetanh(x: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if HAS_EXT:
+        return _C.safetanh(x, eps)
+    from ..modules.distributions.continuous import safetanh as py_safetanh
+
+    return py_safetanh(x, eps)
+
+
+class DeviceSumTree:
+    """HBM-resident sum(+min) segment tree pair driven by the fused HIP
+    kernels (scan descent + two-pass update).  Numerics validated against
+    the torch trees in tests/test_ops.py."""
+
+    def __init__(self, capacity: int, device, with_min: bool = True):
+        _require_ext()
+        size = 1
+        while size < capacity:
+            size *= 2
+        self.capacity = capacity
+        self.size = size
+        self.device = torch.device(device)
+        self.sum_tree = torch.zeros(2 * size, dtype=torch.float64, device=self.device)
+        self.min_tree = torch.full(
+            (2 * size,), float("inf"), dtype=torch.float64, device=self.device
+        )
+        self.with_min = with_min
+        self._cnt = torch.zeros(2 * size, dtype=torch.int32, device=self.device)
+
+    def update(self, index: torch.Tensor, value: torch.Tensor) -> None:
+        index = index.to(self.device, torch.long).reshape(-1)
+        value = value.to(self.device, torch.float64).reshape(-1)
+        if index.numel() == 0:
+            return
+        # last-writer-wins dedup (same semantics as the serialized
+        # reference write, cuda_segment_tree.cu:27-37)
+        order = torch.arange(index.numel(), device=self.device)
+        winner = torch.full((self.size,), -1, dtype=torch.long, device=self.device)
+        winner.scatter_reduce_(0, index, order, reduce="amax")
+        sel = winner[index] == order
+        index = index[sel]
+        value = value[sel]
+        _C.tree_update(
+            self.sum_tree, self.min_tree, self._cnt, index, value, self.size, self.with_min
+        )
+
+    def total(self) -> torch.Tensor:
+        return self.sum_tree[1]
+
+    def min(self) -> torch.Tensor:
+        return self.min_tree[1]
+
+    def scan_lower_bound(self, mass: torch.Tensor) -> torch.Tensor:
+        mass = mass.to(self.device, torch.float64).reshape(-1)
+        return _C.tree_scan_lower_bound(self.sum_tree, mass, self.size, self.capacity)
+
+    def get(self, index: torch.Tensor) -> torch.Tensor:
+        return self.sum_tree[self.size + index.to(self.device, torch.long)]
