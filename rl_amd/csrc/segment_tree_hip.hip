@@ -70,7 +70,10 @@ __global__ void tree_set_leaves_kernel(double* __restrict__ sum_tree,
   if (with_min) min_tree[node] = value[g];
 }
 
-// pass 1: mark ancestor counts
+// pass 1: mark arrival counts.  FIRST arrival at a node continues upward,
+// later arrivals stop — so cnt[P] == number of distinct child paths that
+// will arrive at P during resolve (1 or 2), and every node on any updated
+// path is marked exactly once per arriving path.
 __global__ void tree_mark_kernel(int* __restrict__ cnt,
                                  const long* __restrict__ index, const long n,
                                  const long size) {
@@ -78,12 +81,15 @@ __global__ void tree_mark_kernel(int* __restrict__ cnt,
   if (g >= n) return;
   long node = (size + index[g]) >> 1;
   while (node >= 1) {
-    atomicAdd(&cnt[node], 1);
+    const int prev = atomicAdd(&cnt[node], 1);
+    if (prev > 0) return;  // someone already marked this node and above
     node >>= 1;
   }
 }
 
-// pass 2: last arrival recomputes
+// pass 2: LAST arrival recomputes and continues; earlier arrivals retire.
+// Mirrors the mark gating, so exactly cnt[P] threads decrement each node
+// and every counter returns to zero.
 __global__ void tree_resolve_kernel(double* __restrict__ sum_tree,
                                     double* __restrict__ min_tree,
                                     int* __restrict__ cnt,
@@ -94,9 +100,9 @@ __global__ void tree_resolve_kernel(double* __restrict__ sum_tree,
   if (g >= n) return;
   long node = (size + index[g]) >> 1;
   while (node >= 1) {
-    __threadfence();
     const int prev = atomicSub(&cnt[node], 1);
-    if (prev > 1) return;  // a later-arriving sibling path finishes this node
+    if (prev > 1) return;  // the other child path finishes this node
+    __threadfence();  // acquire: sibling subtree writes are now visible
     const double l = sum_tree[2 * node];
     const double r = sum_tree[2 * node + 1];
     sum_tree[node] = l + r;
@@ -105,6 +111,7 @@ __global__ void tree_resolve_kernel(double* __restrict__ sum_tree,
       const double rm = min_tree[2 * node + 1];
       min_tree[node] = lm < rm ? lm : rm;
     }
+    __threadfence();  // release: make this node visible before parent's atomic
     node >>= 1;
   }
 }

@@ -178,6 +178,32 @@ class PrioritizedSampler(Sampler):
             self.max_capacity, device=self.device, dtype=self.dtype
         )
         self._max_priority = 1.0
+        self._device_tree = None
+
+    def _maybe_promote(self, storage) -> None:
+        """Move the trees into HBM (fused HIP kernels) once the storage is
+        seen to live on a HIP device (reference selects CUDA trees the same
+        way, prioritized.py:289-336)."""
+        if self._device_tree is not None:
+            return
+        dev = getattr(storage, "device", None)
+        if dev is None or torch.device(dev).type != "cuda":
+            self._device_tree = False  # sentinel: stay on torch trees
+            return
+        try:
+            from ...ops import DeviceSumTree, HAS_HIP_EXT
+
+            if not HAS_HIP_EXT:
+                self._device_tree = False
+                return
+            tree = DeviceSumTree(self.max_capacity, device=dev, with_min=True)
+            vals = self._sum_tree.dump_values()
+            nz = vals.nonzero().reshape(-1)
+            if nz.numel():
+                tree.update(nz.to(dev), vals[nz].to(dev))
+            self._device_tree = tree
+        except Exception:
+            self._device_tree = False
 
     @property
     def default_priority(self) -> float:
@@ -191,6 +217,9 @@ class PrioritizedSampler(Sampler):
         val = torch.full(
             (index.numel(),), self.default_priority, dtype=self.dtype
         )
+        if self._device_tree:
+            self._device_tree.update(index, val)
+            return
         self._sum_tree.update(index, val)
         self._min_tree.update(index, val)
 
@@ -198,6 +227,19 @@ class PrioritizedSampler(Sampler):
         n = len(storage)
         if n == 0:
             raise RuntimeError("cannot sample from an empty storage")
+        self._maybe_promote(storage)
+        if self._device_tree:
+            tree = self._device_tree
+            p_sum = tree.total()
+            p_min = tree.min()
+            mass = (
+                torch.rand(batch_size, device=tree.device, dtype=torch.float64)
+                * p_sum
+            )
+            index = tree.scan_lower_bound(mass).clamp_max(n - 1)
+            p = tree.get(index)
+            weight = (p / p_min).pow(-self.beta)
+            return index, {"_weight": weight.to(torch.float32)}
         p_sum = self._sum_tree.query(0, n)
         p_min = self._min_tree.query(0, n)
         if p_sum <= 0:
@@ -232,6 +274,9 @@ class PrioritizedSampler(Sampler):
                 priority = priority.min(-1).values
         self._max_priority = max(self._max_priority, float(priority.max()))
         p_alpha = (priority + self.eps).pow(self.alpha)
+        if self._device_tree:
+            self._device_tree.update(index, p_alpha)
+            return
         self._sum_tree.update(index, p_alpha)
         self._min_tree.update(index, p_alpha)
 
@@ -241,13 +286,21 @@ class PrioritizedSampler(Sampler):
         ))
 
     def state_dict(self):
+        if self._device_tree:
+            leaves = self._device_tree.sum_tree[
+                self._device_tree.size : self._device_tree.size + self.max_capacity
+            ].cpu()
+            sum_values = min_values = leaves
+        else:
+            sum_values = self._sum_tree.dump_values()
+            min_values = self._min_tree.dump_values()
         return {
             "alpha": self.alpha,
             "beta": self.beta,
             "eps": self.eps,
             "_max_priority": self._max_priority,
-            "sum_values": self._sum_tree.dump_values(),
-            "min_values": self._min_tree.dump_values(),
+            "sum_values": sum_values,
+            "min_values": min_values,
         }
 
     def load_state_dict(self, sd):
