@@ -1,0 +1,2 @@
+from .pendulum import PendulumEnv
+from .synthetic import AntVec, HalfCheetahVec, HumanoidVec, SyntheticMuJoCoEnv

@@ -1,0 +1,9 @@
+from .loggers import (
+    CSVLogger,
+    Logger,
+    MLFlowLogger,
+    TensorboardLogger,
+    WandbLogger,
+    generate_exp_name,
+    get_logger,
+)

@@ -1,0 +1,241 @@
+"""Trainer orchestration tests: hooks, algorithm trainers end-to-end,
+checkpoint/resume."""
+import os
+
+import pytest
+import torch
+
+from rl_amd.checkpoint import Checkpoint, CheckpointRotation, GlobalRNGState
+from rl_amd.collectors import Collector
+from rl_amd.data import LazyTensorStorage, TensorDictReplayBuffer
+from rl_amd.modules import (
+    MLP,
+    EGreedyModule,
+    NormalParamExtractor,
+    ProbabilisticActor,
+    QValueActor,
+    TanhNormal,
+    ValueOperator,
+)
+from rl_amd.record import CSVLogger
+from rl_amd.tensordict import TensorDict, TensorDictModule, TensorDictSequential
+from rl_amd.testing import ContinuousActionVecMockEnv, DiscreteActionVecMockEnv
+from rl_amd.trainers import (
+    DQNTrainer,
+    PPOTrainer,
+    SACTrainer,
+    Trainer,
+)
+
+
+def make_cont_actor_critic(obs_dim=7, act_dim=5):
+    net = torch.nn.Sequential(
+        MLP(in_features=obs_dim, out_features=2 * act_dim, num_cells=[32]),
+        NormalParamExtractor(),
+    )
+    actor = ProbabilisticActor(
+        TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"]),
+        in_keys=["loc", "scale"],
+        distribution_class=TanhNormal,
+        return_log_prob=True,
+    )
+    critic = ValueOperator(
+        MLP(in_features=obs_dim, out_features=1, num_cells=[32]),
+        in_keys=["observation"],
+    )
+    return actor, critic
+
+
+def make_q_actor(obs_dim=4, n_act=2):
+    env = DiscreteActionVecMockEnv(batch_size=[4])
+    return QValueActor(
+        MLP(in_features=obs_dim, out_features=n_act, num_cells=[32]),
+        spec=env.action_spec,
+    )
+
+
+class TestTrainers:
+    def test_dqn_end_to_end(self, tmp_path):
+        env = DiscreteActionVecMockEnv(batch_size=[4], max_steps=20)
+        qnet = make_q_actor()
+        eg = EGreedyModule(spec=env.action_spec, eps_init=0.5)
+        policy = TensorDictSequential(qnet, eg)
+        col = Collector(env, policy, frames_per_batch=64, total_frames=256)
+        tr = DQNTrainer(
+            value_network=qnet,
+            collector=col,
+            total_frames=256,
+            batch_size=32,
+            buffer_size=500,
+            progress_bar=False,
+            logger=CSVLogger("dqn", log_dir=str(tmp_path)),
+        )
+        tr.train()
+        assert tr.collected_frames == 256
+        assert tr._optim_count > 0
+        assert "loss" in tr._log_cache
+
+    def test_ppo_end_to_end(self):
+        env = ContinuousActionVecMockEnv(batch_size=[4], max_steps=50)
+        actor, critic = make_cont_actor_critic()
+        col = Collector(env, actor, frames_per_batch=128, total_frames=256)
+        tr = PPOTrainer(
+            actor=actor,
+            critic=critic,
+            collector=col,
+            total_frames=256,
+            minibatch_size=64,
+            num_epochs=2,
+            progress_bar=False,
+        )
+        tr.train()
+        assert tr.collected_frames == 256
+        assert any(k.startswith("loss") for k in tr._log_cache)
+
+    def test_sac_end_to_end(self):
+        env = ContinuousActionVecMockEnv(batch_size=[2], max_steps=50)
+        actor, _ = make_cont_actor_critic()
+        qnet = ValueOperator(
+            MLP(in_features=7 + 5, out_features=1, num_cells=[32]),
+            in_keys=["observation", "action"],
+        )
+        col = Collector(env, actor, frames_per_batch=32, total_frames=64)
+        tr = SACTrainer(
+            actor=actor,
+            qvalue=qnet,
+            collector=col,
+            total_frames=64,
+            batch_size=16,
+            buffer_size=500,
+            optim_steps_per_batch=2,
+            progress_bar=False,
+        )
+        tr.train()
+        assert tr.collected_frames == 64
+
+    def test_trainer_checkpoint(self, tmp_path):
+        env = DiscreteActionVecMockEnv(batch_size=[4], max_steps=20)
+        qnet = make_q_actor()
+        col = Collector(env, qnet, frames_per_batch=64, total_frames=128)
+        save_file = str(tmp_path / "trainer_ckpt")
+        tr = DQNTrainer(
+            value_network=qnet,
+            collector=col,
+            total_frames=128,
+            batch_size=32,
+            buffer_size=500,
+            progress_bar=False,
+            save_trainer_file=save_file,
+        )
+        tr.train()
+        tr.save_trainer(force_save=True)
+        assert os.path.exists(os.path.join(save_file, "manifest.json"))
+        # resume into a fresh trainer
+        env2 = DiscreteActionVecMockEnv(batch_size=[4], max_steps=20)
+        qnet2 = make_q_actor()
+        col2 = Collector(env2, qnet2, frames_per_batch=64, total_frames=128)
+        tr2 = DQNTrainer(
+            value_network=qnet2,
+            collector=col2,
+            total_frames=128,
+            batch_size=32,
+            buffer_size=500,
+            progress_bar=False,
+            save_trainer_file=save_file,
+        )
+        tr2.load_from_file(save_file)
+        assert tr2.collected_frames == 128
+
+
+class TestCheckpoint:
+    def test_manifest_format(self, tmp_path):
+        import json
+
+        ckpt = Checkpoint()
+        ckpt.register(torch.nn.Linear(3, 2), "model")
+        ckpt.register(GlobalRNGState(), "rng")
+        path = str(tmp_path / "ck")
+        ckpt.save(path)
+        with open(os.path.join(path, "manifest.json")) as f:
+            manifest = json.load(f)
+        assert manifest["format"] == "torchrl.checkpoint"
+        assert manifest["version"] == 1
+        assert "model" in manifest["components"]
+
+    def test_roundtrip_module(self, tmp_path):
+        m1 = torch.nn.Linear(3, 2)
+        ckpt = Checkpoint().register(m1, "model")
+        path = str(tmp_path / "ck")
+        ckpt.save(path)
+        m2 = torch.nn.Linear(3, 2)
+        Checkpoint().register(m2, "model").load(path)
+        assert torch.allclose(m1.weight, m2.weight)
+
+    def test_zip_archive(self, tmp_path):
+        m1 = torch.nn.Linear(3, 2)
+        ckpt = Checkpoint().register(m1, "model")
+        path = str(tmp_path / "ck.zip")
+        ckpt.save(path)
+        assert os.path.exists(path)
+        m2 = torch.nn.Linear(3, 2)
+        Checkpoint().register(m2, "model").load(path)
+        assert torch.allclose(m1.weight, m2.weight)
+
+    def test_rng_state_roundtrip(self, tmp_path):
+        rng = GlobalRNGState()
+        ckpt = Checkpoint().register(rng, "rng")
+        path = str(tmp_path / "ck")
+        torch.manual_seed(0)
+        ckpt.save(path)
+        a = torch.randn(3)
+        Checkpoint().register(GlobalRNGState(), "rng").load(path)
+        b = torch.randn(3)
+        assert torch.allclose(a, b)
+
+    def test_replay_buffer_in_checkpoint(self, tmp_path):
+        rb = TensorDictReplayBuffer(storage=LazyTensorStorage(50), batch_size=4)
+        rb.extend(TensorDict({"x": torch.randn(10, 2)}, batch_size=[10]))
+        ckpt = Checkpoint().register(rb, "buffer")
+        path = str(tmp_path / "ck")
+        ckpt.save(path)
+        rb2 = TensorDictReplayBuffer(storage=LazyTensorStorage(50), batch_size=4)
+        Checkpoint().register(rb2, "buffer").load(path)
+        assert len(rb2) == 10
+
+    def test_rotation(self, tmp_path):
+        m = torch.nn.Linear(2, 2)
+        ckpt = Checkpoint().register(m, "model")
+        rot = CheckpointRotation(ckpt, str(tmp_path / "rots"), keep_last=2)
+        for i, metric in enumerate([1.0, 3.0, 2.0, 0.5]):
+            rot.step(i, metric=metric)
+        dirs = sorted(os.listdir(str(tmp_path / "rots")))
+        assert "ckpt_best" in dirs
+        assert len([d for d in dirs if d != "ckpt_best"]) == 2
+
+    def test_rotation_best_metric(self, tmp_path):
+        m = torch.nn.Linear(2, 2)
+        ckpt = Checkpoint().register(m, "model")
+        rot = CheckpointRotation(ckpt, str(tmp_path / "rots"), keep_last=1)
+        with torch.no_grad():
+            m.weight.fill_(1.0)
+        rot.step(0, metric=5.0)
+        with torch.no_grad():
+            m.weight.fill_(2.0)
+        rot.step(1, metric=1.0)
+        # best should still hold weight=1
+        m2 = torch.nn.Linear(2, 2)
+        Checkpoint().register(m2, "model").load(rot.best_path)
+        assert (m2.weight == 1.0).all()
+
+
+class TestLoggers:
+    def test_csv_logger(self, tmp_path):
+        log = CSVLogger("exp1", log_dir=str(tmp_path))
+        log.log_scalar("reward", 1.5, step=10)
+        log.log_scalar("reward", 2.5, step=20)
+        log.log_hparams({"lr": 0.001})
+        path = os.path.join(log.experiment_dir, "scalars", "reward.csv")
+        with open(path) as f:
+            lines = f.read().strip().split("\n")
+        assert len(lines) == 2
+        assert lines[0] == "10,1.5"
