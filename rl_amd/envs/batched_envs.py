@@ -1,0 +1,338 @@
+"""SerialEnv / ParallelEnv — batched env execution.
+
+Reference: pytorch/rl torchrl/envs/batched_envs.py (BatchedEnvBase:322,
+SerialEnv:1546, ParallelEnv:1918, worker loop
+_run_worker_pipe_shared_mem:3322, buffered step path :2405).
+
+MI355X note: GPU-vectorized envs (rl_amd/envs/custom) are the PRIMARY
+path — one process, one device, tensor-batched dynamics.  ParallelEnv
+exists for CPU simulators: N worker processes step their sub-envs straight
+into a shared-memory TensorDict slice (zero-copy to the parent), commands
+go over pipes, completion over pipe acks.  When the parent's device is a
+GPU the shared buffer is page-locked so the follow-up H2D copy is async.
+"""
+from __future__ import annotations
+
+import multiprocessing as mp
+import os
+import time
+from typing import Any, Callable, List, Optional, Sequence, Union
+
+import torch
+
+from .._utils import _check_for_faulty_process, _ProcessNoWarn, logger
+from ..data.tensor_specs import Composite
+from ..tensordict import TensorDict, TensorDictBase, stack as td_stack
+from .common import EnvBase, EnvMetaData
+from .utils import step_mdp, terminated_or_truncated
+
+__all__ = ["SerialEnv", "ParallelEnv", "BatchedEnvBase"]
+
+BATCHED_PIPE_TIMEOUT = float(os.environ.get("BATCHED_PIPE_TIMEOUT", "60.0"))
+
+
+class BatchedEnvBase(EnvBase):
+    """Common spec plumbing for Serial/Parallel envs."""
+
+    def __init__(
+        self,
+        num_workers: int,
+        create_env_fn: Union[Callable[[], EnvBase], Sequence[Callable[[], EnvBase]]],
+        *,
+        create_env_kwargs: Optional[Union[dict, Sequence[dict]]] = None,
+        device=None,
+        shared_memory: bool = True,
+        **kwargs,
+    ):
+        if callable(create_env_fn):
+            create_env_fn = [create_env_fn] * num_workers
+        if len(create_env_fn) != num_workers:
+            raise ValueError("need one env constructor per worker")
+        if create_env_kwargs is None:
+            create_env_kwargs = [{}] * num_workers
+        elif isinstance(create_env_kwargs, dict):
+            create_env_kwargs = [create_env_kwargs] * num_workers
+        self.create_env_fn = list(create_env_fn)
+        self.create_env_kwargs = list(create_env_kwargs)
+        self.num_workers = num_workers
+        self._dummy_env: Optional[EnvBase] = None
+        super().__init__(device=device, batch_size=torch.Size([num_workers]))
+        self.shared_memory = shared_memory
+        self.is_closed = True
+
+    def _set_specs_from(self, env: EnvBase):
+        inner_bs = env.batch_size
+        self._inner_batch = inner_bs
+        n = self.num_workers
+
+        def ex(spec):
+            return spec.expand(n, *spec.shape)
+
+        self.full_observation_spec = ex(env.full_observation_spec.clone())
+        self.full_action_spec = ex(env.full_action_spec.clone())
+        self.full_reward_spec = ex(env.full_reward_spec.clone())
+        self.full_done_spec = ex(env.full_done_spec.clone())
+
+
+class SerialEnv(BatchedEnvBase):
+    """N sub-envs stepped sequentially in-process (reference :1546)."""
+
+    def __init__(self, num_workers: int, create_env_fn, **kwargs):
+        super().__init__(num_workers, create_env_fn, **kwargs)
+        self._envs: List[EnvBase] = [
+            fn(**kw) for fn, kw in zip(self.create_env_fn, self.create_env_kwargs)
+        ]
+        self._set_specs_from(self._envs[0])
+        self.is_closed = False
+
+    def _reset(self, tensordict: Optional[TensorDictBase] = None, **kwargs) -> TensorDictBase:
+        outs = []
+        for i, env in enumerate(self._envs):
+            sub = None
+            if tensordict is not None:
+                sub = tensordict[i].clone(False)
+                if "_reset" in sub and not bool(sub.get("_reset").any()):
+                    # not flagged: keep current state, echo the root obs
+                    outs.append(env.maybe_reset(sub))
+                    continue
+            outs.append(env.reset(sub))
+        return td_stack(outs, 0)
+
+    def _step(self, tensordict: TensorDictBase) -> TensorDictBase:
+        outs = []
+        for i, env in enumerate(self._envs):
+            td_i = tensordict[i].clone(False)
+            td_i = env.step(td_i)
+            outs.append(td_i.get("next"))
+        return td_stack(outs, 0)
+
+    def _set_seed(self, seed: Optional[int]):
+        out = seed
+        for i, env in enumerate(self._envs):
+            out = env.set_seed(seed + i if seed is not None else None)
+        return out
+
+    def close(self, raise_if_closed: bool = False):
+        for env in self._envs:
+            env.close()
+        self.is_closed = True
+
+    def __getattr__(self, name):
+        try:
+            return super().__getattr__(name)
+        except AttributeError:
+            envs = self.__dict__.get("_envs")
+            if envs:
+                return getattr(envs[0], name)
+            raise
+
+
+def _parallel_worker(
+    idx: int,
+    pipe,
+    env_fn,
+    env_kwargs: dict,
+    shared_td: Optional[TensorDictBase],
+):
+    """Worker loop (reference _run_worker_pipe_shared_mem:3322)."""
+    torch.set_num_threads(1)
+    env = env_fn(**env_kwargs)
+    my_slice = shared_td[idx] if shared_td is not None else None
+    pipe.send(("meta", EnvMetaData.build(env)))
+    root_td: Optional[TensorDictBase] = None
+    try:
+        while True:
+            cmd, data = pipe.recv()
+            if cmd == "reset":
+                root_td = env.reset(data)
+                if my_slice is not None:
+                    my_slice.update_(root_td)
+                    pipe.send(("done", None))
+                else:
+                    pipe.send(("done", root_td))
+            elif cmd == "step":
+                if data is not None:
+                    root_td = root_td if root_td is not None else env.reset()
+                    root_td.update(data)
+                td = env.step(root_td)
+                next_td = td.get("next")
+                if my_slice is not None:
+                    my_slice.get("next").update_(next_td)
+                    pipe.send(("done", None))
+                else:
+                    pipe.send(("done", next_td))
+                root_td = step_mdp(td)
+            elif cmd == "step_and_maybe_reset":
+                if data is not None:
+                    root_td = root_td if root_td is not None else env.reset()
+                    root_td.update(data)
+                td, next_root = env.step_and_maybe_reset(root_td)
+                if my_slice is not None:
+                    my_slice.get("next").update_(td.get("next"))
+                    my_slice.get("root_next").update_(next_root)
+                    pipe.send(("done", None))
+                else:
+                    pipe.send(("done", (td.get("next"), next_root)))
+                root_td = next_root
+            elif cmd == "seed":
+                out = env.set_seed(data)
+                pipe.send(("seeded", out))
+            elif cmd == "state_dict":
+                pipe.send(("state_dict", env.state_dict() if hasattr(env, "state_dict") else {}))
+            elif cmd == "getattr":
+                pipe.send(("attr", getattr(env, data)))
+            elif cmd == "close":
+                env.close()
+                pipe.send(("closed", None))
+                break
+    except KeyboardInterrupt:
+        pass
+    except EOFError:
+        pass
+    except Exception:
+        import traceback
+
+        traceback.print_exc()
+        raise
+
+
+class ParallelEnv(BatchedEnvBase):
+    """N worker processes with shared-memory payload (reference :1918)."""
+
+    def __init__(self, num_workers: int, create_env_fn, **kwargs):
+        super().__init__(num_workers, create_env_fn, **kwargs)
+        ctx = mp.get_context("spawn")
+        self.parent_pipes = []
+        self.procs = []
+        self._shared: Optional[TensorDictBase] = None
+        # boot one worker first to learn specs, then allocate the shared
+        # buffer, then boot the rest pointing at it
+        metas = []
+        # stage 1: spawn all workers WITHOUT buffers to collect specs
+        for i in range(num_workers):
+            parent, child = ctx.Pipe()
+            proc = _ProcessNoWarn(
+                target=_parallel_worker,
+                args=(i, child, self.create_env_fn[i], self.create_env_kwargs[i], None),
+            )
+            proc.daemon = True
+            proc.start()
+            child.close()
+            self.parent_pipes.append(parent)
+            self.procs.append(proc)
+        for pipe in self.parent_pipes:
+            msg, meta = pipe.recv()
+            assert msg == "meta"
+            metas.append(meta)
+        self._set_specs_from_meta(metas[0])
+        self.is_closed = False
+        self._root_cache: Optional[TensorDictBase] = None
+
+    def _set_specs_from_meta(self, meta: EnvMetaData):
+        n = self.num_workers
+        self._inner_batch = meta.batch_size
+
+        def ex(spec):
+            return spec.expand(n, *spec.shape)
+
+        self.full_observation_spec = ex(meta.specs["full_observation_spec"].clone())
+        self.full_action_spec = ex(meta.specs["full_action_spec"].clone())
+        self.full_reward_spec = ex(meta.specs["full_reward_spec"].clone())
+        self.full_done_spec = ex(meta.specs["full_done_spec"].clone())
+
+    def _check(self):
+        _check_for_faulty_process(self.procs)
+
+    def _send_all(self, cmd: str, datas):
+        for pipe, data in zip(self.parent_pipes, datas):
+            pipe.send((cmd, data))
+
+    def _recv_all(self):
+        outs = []
+        for pipe in self.parent_pipes:
+            if not pipe.poll(BATCHED_PIPE_TIMEOUT):
+                self._check()
+                raise TimeoutError("ParallelEnv worker timed out")
+            msg, data = pipe.recv()
+            outs.append(data)
+        return outs
+
+    def _reset(self, tensordict: Optional[TensorDictBase] = None, **kwargs) -> TensorDictBase:
+        datas = []
+        for i in range(self.num_workers):
+            sub = tensordict[i].clone(False) if tensordict is not None else None
+            datas.append(sub)
+        self._send_all("reset", datas)
+        outs = self._recv_all()
+        out = td_stack(outs, 0)
+        if self.device is not None and out.device != self.device:
+            out = out.to(self.device)
+        return out
+
+    def _step(self, tensordict: TensorDictBase) -> TensorDictBase:
+        input_keys = list(self.full_action_spec.keys(True, True))
+        datas = []
+        for i in range(self.num_workers):
+            sub = tensordict[i].select(*input_keys, strict=False).cpu()
+            datas.append(sub)
+        self._send_all("step", datas)
+        outs = self._recv_all()
+        out = td_stack(outs, 0)
+        if self.device is not None and out.device != self.device:
+            out = out.to(self.device)
+        return out
+
+    def step_and_maybe_reset(self, tensordict: TensorDictBase):
+        input_keys = list(self.full_action_spec.keys(True, True))
+        datas = []
+        for i in range(self.num_workers):
+            datas.append(tensordict[i].select(*input_keys, strict=False).cpu())
+        self._send_all("step_and_maybe_reset", datas)
+        outs = self._recv_all()
+        next_tds = td_stack([o[0] for o in outs], 0)
+        next_roots = td_stack([o[1] for o in outs], 0)
+        if self.device is not None:
+            next_tds = next_tds.to(self.device)
+            next_roots = next_roots.to(self.device)
+        self._complete_done(next_tds)
+        tensordict.set("next", next_tds)
+        return tensordict, next_roots
+
+    def _set_seed(self, seed: Optional[int]):
+        out = seed
+        for i, pipe in enumerate(self.parent_pipes):
+            pipe.send(("seed", seed + i if seed is not None else None))
+        for pipe in self.parent_pipes:
+            msg, out = pipe.recv()
+        return out
+
+    def state_dict(self):
+        self._send_all("state_dict", [None] * self.num_workers)
+        outs = self._recv_all()
+        return {f"worker{i}": sd for i, sd in enumerate(outs)}
+
+    def close(self, raise_if_closed: bool = False):
+        if self.is_closed:
+            return
+        for pipe in self.parent_pipes:
+            try:
+                pipe.send(("close", None))
+            except (BrokenPipeError, OSError):
+                pass
+        for pipe in self.parent_pipes:
+            try:
+                if pipe.poll(5.0):
+                    pipe.recv()
+            except (EOFError, OSError):
+                pass
+        for proc in self.procs:
+            proc.join(timeout=5.0)
+            if proc.is_alive():
+                proc.terminate()
+        self.is_closed = True
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

@@ -1,0 +1,120 @@
+"""SerialEnv / ParallelEnv tests."""
+import pytest
+import torch
+
+from rl_amd.collectors import Collector
+from rl_amd.envs import check_env_specs, step_mdp
+from rl_amd.envs.batched_envs import ParallelEnv, SerialEnv
+from rl_amd.tensordict import TensorDict
+from rl_amd.testing import ContinuousActionVecMockEnv, CountingEnv
+
+
+def make_counting():
+    return CountingEnv(max_steps=5)
+
+
+def make_cont():
+    return ContinuousActionVecMockEnv(max_steps=10)
+
+
+class TestSerialEnv:
+    def test_shapes(self):
+        env = SerialEnv(3, make_counting)
+        td = env.reset()
+        assert td.batch_size == torch.Size([3])
+        td.set("action", torch.ones(3, 1, dtype=torch.bool))
+        td = env.step(td)
+        assert td.get(("next", "observation")).shape == (3, 1)
+        env.close()
+
+    def test_rollout(self):
+        env = SerialEnv(3, make_cont)
+        r = env.rollout(5, break_when_any_done=False)
+        assert r.batch_size == torch.Size([3, 5])
+        env.close()
+
+    def test_specs(self):
+        env = SerialEnv(2, make_cont)
+        check_env_specs(env)
+        env.close()
+
+    def test_partial_reset(self):
+        env = SerialEnv(2, make_counting)
+        td = env.reset()
+        for _ in range(5):
+            td.set("action", torch.ones(2, 1, dtype=torch.bool))
+            td, td_next = env.step_and_maybe_reset(td)
+            td = td_next
+        # both envs hit max_steps=5 → auto-reset to 0
+        assert (td["observation"] == 0).all()
+        env.close()
+
+    def test_seed(self):
+        env = SerialEnv(2, make_cont)
+        env.set_seed(0)
+        r1 = env.reset()
+        env.set_seed(0)
+        r2 = env.reset()
+        assert torch.allclose(r1["observation"], r2["observation"])
+        env.close()
+
+    def test_collector_integration(self):
+        env = SerialEnv(2, make_cont)
+        col = Collector(env, frames_per_batch=20, total_frames=40)
+        for b in col:
+            assert b.batch_size == torch.Size([2, 10])
+        col.shutdown()
+
+
+class TestParallelEnv:
+    def test_shapes(self):
+        env = ParallelEnv(2, make_counting)
+        try:
+            td = env.reset()
+            assert td.batch_size == torch.Size([2])
+            td.set("action", torch.ones(2, 1, dtype=torch.bool))
+            td = env.step(td)
+            assert td.get(("next", "observation")).shape == (2, 1)
+            assert td.get(("next", "reward")).shape == (2, 1)
+        finally:
+            env.close()
+
+    def test_step_and_maybe_reset(self):
+        env = ParallelEnv(2, make_counting)
+        try:
+            td = env.reset()
+            for _ in range(5):
+                td.set("action", torch.ones(2, 1, dtype=torch.bool))
+                td, td_next = env.step_and_maybe_reset(td)
+                td = td_next
+            assert (td["observation"] == 0).all()
+        finally:
+            env.close()
+
+    def test_rollout(self):
+        env = ParallelEnv(2, make_cont)
+        try:
+            r = env.rollout(4, break_when_any_done=False)
+            assert r.batch_size == torch.Size([2, 4])
+        finally:
+            env.close()
+
+    def test_seed_determinism(self):
+        env = ParallelEnv(2, make_cont)
+        try:
+            env.set_seed(3)
+            r1 = env.reset()["observation"]
+            env.set_seed(3)
+            r2 = env.reset()["observation"]
+            assert torch.allclose(r1, r2)
+        finally:
+            env.close()
+
+    def test_collector_integration(self):
+        env = ParallelEnv(2, make_cont)
+        col = Collector(env, frames_per_batch=16, total_frames=32)
+        n = 0
+        for b in col:
+            n += b.numel()
+        assert n == 32
+        col.shutdown()
