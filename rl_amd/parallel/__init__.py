@@ -1,0 +1,11 @@
+from .comm import (
+    GradAllReducer,
+    all_reduce_grads,
+    broadcast_tensordict,
+    init_distributed,
+    irecv_tensordict,
+    isend_tensordict,
+    recv_tensordict,
+    rendezvous_store,
+    send_tensordict,
+)

@@ -1,0 +1,200 @@
+"""Weight-sync schemes: publishing learner weights to collectors/workers.
+
+Reference: pytorch/rl torchrl/weight_update/weight_sync_schemes.py
+(WeightStrategy:145, WeightSyncScheme:346), _shared.py (shared-mem
+transport :59,327), _mp.py (pipe transport :18), _distributed.py
+(torch.distributed transport :36,596), _noupdate.py:13.
+
+A scheme pairs a SENDER (learner side) with RECEIVERS (worker side) over a
+transport.  On MI355X the distributed transport is a packed RCCL broadcast
+over xGMI (one collective for all leaves).
+"""
+from __future__ import annotations
+
+import multiprocessing as mp
+from typing import Any, Callable, Dict, List, Optional, Sequence, Union
+
+import torch
+
+from ..tensordict import TensorDict, TensorDictBase
+
+__all__ = [
+    "WeightStrategy",
+    "WeightSyncScheme",
+    "SharedMemWeightSyncScheme",
+    "MultiProcessWeightSyncScheme",
+    "DistributedWeightSyncScheme",
+    "NoWeightSyncScheme",
+]
+
+
+class WeightStrategy:
+    """Extract / apply weights in a chosen format (reference :145)."""
+
+    def __init__(self, extract_as: str = "tensordict"):
+        if extract_as not in ("tensordict", "state_dict"):
+            raise ValueError("extract_as must be tensordict or state_dict")
+        self.extract_as = extract_as
+
+    def extract(self, model) -> Union[TensorDictBase, dict]:
+        if isinstance(model, TensorDictBase):
+            return model
+        if self.extract_as == "tensordict":
+            return TensorDict.from_module(model).apply(
+                lambda t: t.detach()
+            )
+        return {k: v.detach() for k, v in model.state_dict().items()}
+
+    def apply(self, model, weights) -> None:
+        if isinstance(weights, TensorDictBase):
+            weights.to_module(model)
+        elif isinstance(weights, dict):
+            model.load_state_dict(weights)
+        else:
+            raise TypeError(f"cannot apply weights of type {type(weights)}")
+
+
+class WeightSyncScheme:
+    """ABC (reference :346): create sender/receiver pairs per model_id."""
+
+    def __init__(self, strategy: str = "tensordict"):
+        self.strategy = WeightStrategy(strategy)
+
+    def connect(self, model) -> None:
+        """Learner-side registration."""
+        self.model = model
+
+    def send(self, weights=None) -> None:
+        raise NotImplementedError
+
+    def receive(self, model) -> bool:
+        """Worker-side poll; returns True if new weights were applied."""
+        raise NotImplementedError
+
+
+class NoWeightSyncScheme(WeightSyncScheme):
+    """No-op (reference _noupdate.py:13)."""
+
+    def send(self, weights=None) -> None:
+        pass
+
+    def receive(self, model) -> bool:
+        return False
+
+
+class SharedMemWeightSyncScheme(WeightSyncScheme):
+    """Learner writes into a shared-memory TensorDict; workers copy out
+    (reference _shared.py:59,327).  Zero-pickle: one ``update_`` on each
+    side."""
+
+    def __init__(self, strategy: str = "tensordict"):
+        super().__init__(strategy)
+        self._shared: Optional[TensorDictBase] = None
+        self._version = mp.Value("L", 0)
+        self._seen = 0
+
+    def connect(self, model) -> "SharedMemWeightSyncScheme":
+        super().connect(model)
+        weights = self.strategy.extract(model)
+        if isinstance(weights, dict):
+            weights = TensorDict(
+                {k.replace(".", "․"): v for k, v in weights.items()},
+                batch_size=[],
+            )
+        self._shared = weights.clone().cpu().share_memory_()
+        return self
+
+    @property
+    def shared_weights(self) -> TensorDictBase:
+        return self._shared
+
+    def send(self, weights=None) -> None:
+        src = weights if weights is not None else self.strategy.extract(self.model)
+        if isinstance(src, dict):
+            for k, v in src.items():
+                self._shared.get(k.replace(".", "․")).copy_(v.detach().cpu())
+        else:
+            self._shared.update_(src.cpu() if src.device else src)
+        with self._version.get_lock():
+            self._version.value += 1
+
+    def receive(self, model) -> bool:
+        v = self._version.value
+        if v == self._seen:
+            return False
+        self._seen = v
+        weights = self._shared
+        self.strategy.apply(model, weights)
+        return True
+
+
+class MultiProcessWeightSyncScheme(WeightSyncScheme):
+    """Pipe transport: pickled state through an mp.Pipe per worker
+    (reference _mp.py:18,516)."""
+
+    def __init__(self, strategy: str = "state_dict"):
+        super().__init__(strategy)
+        self.pipes: List = []
+
+    def add_worker(self):
+        parent, child = mp.get_context("spawn").Pipe()
+        self.pipes.append(parent)
+        return child
+
+    def send(self, weights=None) -> None:
+        src = weights if weights is not None else self.strategy.extract(self.model)
+        if isinstance(src, TensorDictBase):
+            src = src.cpu()
+        else:
+            src = {k: v.cpu() for k, v in src.items()}
+        for pipe in self.pipes:
+            pipe.send(src)
+
+    @staticmethod
+    def receive_from(child_pipe, model, strategy: Optional[WeightStrategy] = None) -> bool:
+        strategy = strategy or WeightStrategy("state_dict")
+        updated = False
+        while child_pipe.poll():
+            weights = child_pipe.recv()
+            strategy.apply(model, weights)
+            updated = True
+        return updated
+
+    def receive(self, model) -> bool:
+        raise RuntimeError("use receive_from(child_pipe, model) in the worker")
+
+
+class DistributedWeightSyncScheme(WeightSyncScheme):
+    """torch.distributed transport: packed broadcast from rank ``src``
+    (reference _distributed.py:36,596; on ROCm this is ONE RCCL broadcast
+    over xGMI for the whole parameter set)."""
+
+    def __init__(self, strategy: str = "tensordict", src: int = 0, group=None):
+        super().__init__(strategy)
+        self.src = src
+        self.group = group
+
+    def send(self, weights=None) -> None:
+        from ..parallel.comm import broadcast_tensordict
+
+        src_w = weights if weights is not None else self.strategy.extract(self.model)
+        if isinstance(src_w, dict):
+            src_w = TensorDict(
+                {k.replace(".", "․"): v for k, v in src_w.items()}, batch_size=[]
+            )
+        broadcast_tensordict(src_w, src=self.src, group=self.group)
+
+    def receive(self, model) -> bool:
+        from ..parallel.comm import broadcast_tensordict
+
+        weights = self.strategy.extract(model)
+        if isinstance(weights, dict):
+            weights = TensorDict(
+                {k.replace(".", "․"): v for k, v in weights.items()}, batch_size=[]
+            )
+        broadcast_tensordict(weights, src=self.src, group=self.group)
+        if isinstance(weights, TensorDictBase):
+            # leaves were updated in place via the module's own tensors when
+            # extract_as == tensordict (from_module shares storage)
+            pass
+        return True
