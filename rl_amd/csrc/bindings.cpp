@@ -31,6 +31,13 @@ void launch_tree_scan_f64(const double*, const double*, long*, long, long,
                           long, void*);
 void launch_tree_update_f64(double*, double*, int*, const long*, const double*,
                             long, long, int, void*);
+int gru_fused_lds_bytes(int);
+int lstm_fused_lds_bytes(int);
+void launch_gru_fused(const float*, const float*, const float*, const bool*,
+                      const float*, float*, float*, int, int, int, void*);
+void launch_lstm_fused(const float*, const float*, const bool*, const float*,
+                       const float*, float*, float*, float*, int, int, int,
+                       void*);
 }
 
 static void check_gae_args(const torch::Tensor& t, const char* name) {
@@ -137,6 +144,49 @@ void tree_update(torch::Tensor sum_tree, torch::Tensor min_tree,
                          value.data_ptr<double>(), index.numel(), size,
                          with_min ? 1 : 0, (void*)stream);
 }
+// Fused GRU forward scan: gates_x = x@W_ih + b_ih precomputed outside
+// (one hipBLASLt GEMM over B*T); returns (ys, h_final).
+std::vector<torch::Tensor> gru_fused(torch::Tensor gates_x, torch::Tensor w_hh,
+                                     torch::Tensor bias_hh,
+                                     torch::Tensor is_init,
+                                     torch::Tensor h0) {
+  TORCH_CHECK(gates_x.is_cuda() && gates_x.is_contiguous());
+  long B = gates_x.size(0), T = gates_x.size(1);
+  long H = gates_x.size(2) / 3;
+  TORCH_CHECK(gru_fused_lds_bytes(H) <= 160 * 1024,
+              "H too large for the fused GRU LDS budget");
+  auto ys = torch::empty({B, T, H}, gates_x.options());
+  auto h_out = torch::empty({B, H}, gates_x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_gru_fused(gates_x.data_ptr<float>(), w_hh.data_ptr<float>(),
+                   bias_hh.data_ptr<float>(), is_init.data_ptr<bool>(),
+                   h0.numel() ? h0.data_ptr<float>() : nullptr,
+                   ys.data_ptr<float>(), h_out.data_ptr<float>(), B, T, H,
+                   (void*)stream);
+  return {ys, h_out};
+}
+
+std::vector<torch::Tensor> lstm_fused(torch::Tensor gates_x,
+                                      torch::Tensor w_hh,
+                                      torch::Tensor is_init, torch::Tensor h0,
+                                      torch::Tensor c0) {
+  TORCH_CHECK(gates_x.is_cuda() && gates_x.is_contiguous());
+  long B = gates_x.size(0), T = gates_x.size(1);
+  long H = gates_x.size(2) / 4;
+  TORCH_CHECK(lstm_fused_lds_bytes(H) <= 160 * 1024,
+              "H too large for the fused LSTM LDS budget");
+  auto ys = torch::empty({B, T, H}, gates_x.options());
+  auto h_out = torch::empty({B, H}, gates_x.options());
+  auto c_out = torch::empty({B, H}, gates_x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_lstm_fused(gates_x.data_ptr<float>(), w_hh.data_ptr<float>(),
+                    is_init.data_ptr<bool>(),
+                    h0.numel() ? h0.data_ptr<float>() : nullptr,
+                    c0.numel() ? c0.data_ptr<float>() : nullptr,
+                    ys.data_ptr<float>(), h_out.data_ptr<float>(),
+                    c_out.data_ptr<float>(), B, T, H, (void*)stream);
+  return {ys, h_out, c_out};
+}
 #endif  // RL_AMD_WITH_HIP
 
 // ---------------------------------------------------------------------------
@@ -171,5 +221,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tree_scan_lower_bound", &tree_scan_lower_bound,
         "sum-tree inverse-CDF descent (HIP)");
   m.def("tree_update", &tree_update, "segment-tree leaf update+recompute (HIP)");
+  m.def("gru_fused", &gru_fused, "fused GRU forward scan with resets (HIP)");
+  m.def("lstm_fused", &lstm_fused, "fused LSTM forward scan with resets (HIP)");
 #endif
 }

@@ -19,3 +19,13 @@ from .exploration import (
     EGreedyWrapper,
     OrnsteinUhlenbeckProcessModule,
 )
+from .rnn import (
+    GRUCell,
+    GRUModule,
+    LSTMCell,
+    LSTMModule,
+    gru_scan,
+    lstm_scan,
+    recurrent_mode,
+    set_recurrent_mode,
+)

@@ -183,3 +183,39 @@ class DeviceSumTree:
 
     def get(self, index: torch.Tensor) -> torch.Tensor:
         return self.sum_tree[self.size + index.to(self.device, torch.long)]
+
+
+def gru_fused(cell, x: torch.Tensor, is_init: torch.Tensor, h0: Optional[torch.Tensor] = None):
+    """Fused GRU forward over [B, T, F] with per-step resets.
+
+    Matches rnn.gru_scan numerics (tests/test_rnn.py); forward-only."""
+    _require_ext()
+    B, T = x.shape[0], x.shape[1]
+    H = cell.hidden_size
+    gates_x = (x.float() @ cell.weight_ih.T.float() + cell.bias_ih.float()).contiguous()
+    ii = is_init.squeeze(-1) if is_init.dim() == 3 else is_init
+    ys, h = _C.gru_fused(
+        gates_x,
+        cell.weight_hh.detach().float().contiguous(),
+        cell.bias_hh.detach().float().contiguous(),
+        ii.contiguous(),
+        h0.float().contiguous() if h0 is not None else torch.Tensor(),
+    )
+    return ys.to(x.dtype), h.to(x.dtype)
+
+
+def lstm_fused(cell, x: torch.Tensor, is_init: torch.Tensor, h0=None, c0=None):
+    _require_ext()
+    H = cell.hidden_size
+    gates_x = (
+        x.float() @ cell.weight_ih.T.float() + cell.bias_ih.float() + cell.bias_hh.float()
+    ).contiguous()
+    ii = is_init.squeeze(-1) if is_init.dim() == 3 else is_init
+    ys, h, c = _C.lstm_fused(
+        gates_x,
+        cell.weight_hh.detach().float().contiguous(),
+        ii.contiguous(),
+        h0.float().contiguous() if h0 is not None else torch.Tensor(),
+        c0.float().contiguous() if c0 is not None else torch.Tensor(),
+    )
+    return ys.to(x.dtype), h.to(x.dtype), c.to(x.dtype)
