@@ -303,10 +303,17 @@ class EnvBase(nn.Module):
 
     def step(self, tensordict: TensorDictBase) -> TensorDictBase:
         """Run one transition.  Writes the result under ``tensordict["next"]``
-        and returns the same root (reference torchrl/envs/common.py:2340)."""
+        and returns the same root (reference torchrl/envs/common.py:2340).
+
+        Pre-existing ``"next"`` entries (e.g. recurrent state written by the
+        policy) are preserved and merged."""
         next_td = self._step(tensordict)
         self._complete_done(next_td)
-        tensordict.set("next", next_td)
+        existing = tensordict.get("next", None)
+        if existing is not None and not existing.is_empty():
+            existing.update(next_td)
+        else:
+            tensordict.set("next", next_td)
         return tensordict
 
     def _complete_done(self, td: TensorDictBase) -> None:
