@@ -10,3 +10,4 @@ from .models import (
     DuelingMlpDQNet,
     NormalParamExtractor,
 )
+from .multiagent import Mixer, MultiAgentConvNet, MultiAgentMLP, QMixer, VDNMixer

@@ -14,3 +14,8 @@ from .utils import (
     hold_out_net,
 )
 from .value import GAE, TD0Estimator, TD1Estimator, TDLambdaEstimator, ValueEstimatorBase, VTrace
+from .cql import CQLLoss, DiscreteCQLLoss
+from .iql import DiscreteIQLLoss, IQLLoss
+from .imitation import BCLoss, DTLoss, GAILLoss, OnlineDTLoss, RNDLoss
+from .multiagent import IPPOLoss, MAPPOLoss, QMixerLoss
+from .redq import CrossQLoss, REDQLoss
