@@ -355,11 +355,24 @@ class EnvBase(nn.Module):
             return self.reset(tensordict)
         return tensordict
 
+    # Vectorized envs whose _reset applies a "_reset" mask with pure tensor
+    # ops set this True: step_and_maybe_reset then resets UNCONDITIONALLY
+    # (masked), avoiding the bool(done.any()) host-device sync per step —
+    # the difference between a stall and a fully queued HIP stream.
+    _supports_masked_reset: bool = False
+
     def step_and_maybe_reset(self, tensordict: TensorDictBase):
         """step + partial auto-reset; returns ``(td_with_next, next_root)``
         (reference torchrl/envs/common.py:4090)."""
         tensordict = self.step(tensordict)
         next_root = step_mdp(tensordict)
+        if self._supports_masked_reset:
+            done = next_root.get("done", None)
+            if done is None:
+                return tensordict, next_root
+            next_root.set("_reset", done)
+            next_root = self.reset(next_root)
+            return tensordict, next_root
         any_done = terminated_or_truncated(next_root, key="_reset")
         if any_done:
             next_root = self.reset(next_root)

@@ -25,6 +25,7 @@ DEFAULT_Y = 1.0
 class PendulumEnv(EnvBase):
     metadata = {"render_modes": []}
     batch_locked = False
+    _supports_masked_reset = True
 
     def __init__(
         self,

@@ -28,6 +28,8 @@ __all__ = ["SyntheticMuJoCoEnv", "HalfCheetahVec", "HumanoidVec", "AntVec"]
 class SyntheticMuJoCoEnv(EnvBase):
     """Batched synthetic control env with MuJoCo-style IO shapes."""
 
+    _supports_masked_reset = True
+
     OBS_DIM = 17
     ACT_DIM = 6
 
@@ -73,9 +75,16 @@ class SyntheticMuJoCoEnv(EnvBase):
 
     def _reset(self, tensordict=None, **kwargs) -> TensorDictBase:
         bs = self.batch_size
-        new_state = torch.randn(
-            (*bs, self.obs_dim), generator=self._gen
-        ).to(self.device, self.dtype) * 0.1
+        if self.device.type == "cuda":
+            # device-side RNG: masked per-step resets stay fully on-GPU
+            new_state = (
+                torch.randn((*bs, self.obs_dim), device=self.device, dtype=self.dtype)
+                * 0.1
+            )
+        else:
+            new_state = torch.randn(
+                (*bs, self.obs_dim), generator=self._gen
+            ).to(self.device, self.dtype) * 0.1
         new_t = torch.zeros((*bs, 1), device=self.device)
         if tensordict is not None and "_reset" in tensordict and self._state is not None:
             mask = tensordict.get("_reset").reshape(*bs, 1)

@@ -36,6 +36,8 @@ class CountingEnv(EnvBase):
     ``max_steps``.  Bit-for-bit predictable — collectors and GAE outputs are
     assertable exactly (reference mocking_classes.py:1168)."""
 
+    _supports_masked_reset = True
+
     def __init__(
         self,
         max_steps: int = 5,
