@@ -9,3 +9,8 @@ from .utils import (
     step_mdp,
     terminated_or_truncated,
 )
+from .batched_envs import BatchedEnvBase, ParallelEnv, SerialEnv
+from .custom import AntVec, HalfCheetahVec, HumanoidVec, PendulumEnv, SyntheticMuJoCoEnv
+from .gym_like import GymLikeEnv, default_info_dict_reader
+from .model_based import DreamerEnv, ModelBasedEnvBase
+from . import transforms

@@ -49,3 +49,5 @@ from .tensordict_module import (
     set_recurrent_mode,
 )
 from .models import Mixer, MultiAgentConvNet, MultiAgentMLP, QMixer, VDNMixer
+from .models import ConsistentDropout, NoisyLazyLinear, NoisyLinear, gSDEModule, reset_noise
+from .planners import CEMPlanner, MPCPlannerBase, MPPIPlanner

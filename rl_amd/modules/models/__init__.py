@@ -11,3 +11,4 @@ from .models import (
     NormalParamExtractor,
 )
 from .multiagent import Mixer, MultiAgentConvNet, MultiAgentMLP, QMixer, VDNMixer
+from .exploration import ConsistentDropout, NoisyLazyLinear, NoisyLinear, gSDEModule, reset_noise
