@@ -8,3 +8,5 @@ from .multi import (
     _Interruptor,
 )
 from .utils import split_trajectories
+from .distributed import DistributedCollector, DistributedSyncCollector
+from .evaluator import Evaluator
