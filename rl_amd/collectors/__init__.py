@@ -10,3 +10,4 @@ from .multi import (
 from .utils import split_trajectories
 from .distributed import DistributedCollector, DistributedSyncCollector
 from .evaluator import Evaluator
+from .llm import LLMCollector

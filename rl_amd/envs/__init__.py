@@ -14,3 +14,4 @@ from .custom import AntVec, HalfCheetahVec, HumanoidVec, PendulumEnv, SyntheticM
 from .gym_like import GymLikeEnv, default_info_dict_reader
 from .model_based import DreamerEnv, ModelBasedEnvBase
 from . import transforms
+from .llm import ChatEnv

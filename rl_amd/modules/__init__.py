@@ -52,3 +52,4 @@ from .models import Mixer, MultiAgentConvNet, MultiAgentMLP, QMixer, VDNMixer
 from .models import ConsistentDropout, NoisyLazyLinear, NoisyLinear, gSDEModule, reset_noise
 from .planners import CEMPlanner, MPCPlannerBase, MPPIPlanner
 from .inference_server import InferenceServer, PolicyClient
+from .llm import LLMWrapperBase, TransformersWrapper

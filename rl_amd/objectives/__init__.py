@@ -19,3 +19,4 @@ from .iql import DiscreteIQLLoss, IQLLoss
 from .imitation import BCLoss, DTLoss, GAILLoss, OnlineDTLoss, RNDLoss
 from .multiagent import IPPOLoss, MAPPOLoss, QMixerLoss
 from .redq import CrossQLoss, REDQLoss
+from .llm import CISPOLoss, DAPO, GRPOLoss, MCAdvantage, SFTLoss
