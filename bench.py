@@ -5,9 +5,11 @@ HalfCheetah-v4 shapes (obs 17 / act 6), synthetic dynamics + random-init
 weights (no network for datasets), bf16 compute, GAE on the fused HIP scan.
 
 One full PPO iteration per step:
-  rollout T env steps x 4096 envs (policy sample + env step, on-device)
-  -> GAE (bf16 HIP scan) -> 1 epoch of 4 minibatch ClipPPO updates
-  -> optimizer step (+ DDP all-reduce over RCCL when world_size > 1).
+  rollout T env steps x 4096 envs (policy sample + env step, on-device,
+  written into a pre-allocated [B, T] HBM store; with --graph the whole
+  T-step rollout is captured ONCE as a hipGraph and replayed per iter)
+  -> GAE (fused HIP scan) -> 1 epoch of 4 minibatch ClipPPO updates
+  -> optimizer step (+ flat RCCL all-reduce when world_size > 1).
 
 Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W
@@ -28,11 +30,21 @@ import torch
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 from rl_amd.envs.custom.synthetic import HalfCheetahVec
-from rl_amd.envs.utils import ExplorationType, set_exploration_type, step_mdp
+from rl_amd.envs.utils import ExplorationType, set_exploration_type
 from rl_amd.modules import MLP, NormalParamExtractor, ProbabilisticActor, TanhNormal, ValueOperator
 from rl_amd.objectives import ClipPPOLoss
 from rl_amd.objectives.value.advantages import GAE
-from rl_amd.tensordict import TensorDict, TensorDictModule, stack as td_stack
+from rl_amd.tensordict import TensorDict, TensorDictModule
+
+STORE_KEYS = [
+    "observation",
+    "action",
+    "sample_log_prob",
+    ("next", "observation"),
+    ("next", "reward"),
+    ("next", "done"),
+    ("next", "terminated"),
+]
 
 
 def parse_args():
@@ -46,10 +58,13 @@ def parse_args():
     p.add_argument("--epochs", type=int, default=1)
     p.add_argument("--hidden", type=int, default=64)
     p.add_argument("--device", type=str, default=None)
+    p.add_argument("--graph", action="store_true", help="hipGraph-capture the rollout")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
+    p.set_defaults(graph=True)
     return p.parse_args()
 
 
-def build(args, device, dtype):
+def build(args, device):
     env = HalfCheetahVec(batch_size=[args.envs], device=device, dtype=torch.float32)
     obs_dim, act_dim = env.obs_dim, env.act_dim
     actor_net = torch.nn.Sequential(
@@ -101,11 +116,10 @@ def main():
 
     distributed = world > 1
     if distributed:
-        backend = "nccl" if cuda else "gloo"
-        torch.distributed.init_process_group(backend=backend)
+        torch.distributed.init_process_group(backend="nccl" if cuda else "gloo")
 
     torch.manual_seed(1234 + rank)
-    env, actor, critic = build(args, device, dtype)
+    env, actor, critic = build(args, device)
     env.set_seed(1234 + rank)
 
     params = list(actor.parameters()) + list(critic.parameters())
@@ -114,35 +128,72 @@ def main():
                            critic_coeff=0.5, normalize_advantage=True)
     gae = GAE(gamma=0.99, lmbda=0.95, value_network=critic, vectorized=True)
 
-    if distributed:
-        # wrap the parameter set for gradient all-reduce: simplest robust
-        # form — manual bucketed all-reduce after backward (overlap comes
-        # from RCCL stream concurrency; DDP module wrapping splits the
-        # TensorDict API, so keep it manual here)
-        pass
-
-    T = args.horizon
-    B = args.envs
+    T, B = args.horizon, args.envs
     frames_per_step = T * B
-
     autocast = torch.autocast(device_type="cuda", dtype=dtype, enabled=cuda)
+    obs_dim, act_dim = env.obs_dim, env.act_dim
 
-    def one_step(carrier):
-        # ---- rollout
-        snaps = []
+    # pre-allocated [B, T] rollout store resident in HBM
+    store = TensorDict(
+        {
+            "observation": torch.zeros(B, T, obs_dim, device=device),
+            "action": torch.zeros(B, T, act_dim, device=device),
+            "sample_log_prob": torch.zeros(B, T, device=device),
+            "next": {
+                "observation": torch.zeros(B, T, obs_dim, device=device),
+                "reward": torch.zeros(B, T, 1, device=device),
+                "done": torch.zeros(B, T, 1, dtype=torch.bool, device=device),
+                "terminated": torch.zeros(B, T, 1, dtype=torch.bool, device=device),
+            },
+        },
+        batch_size=[B, T],
+        device=device,
+    )
+
+    carrier0 = env.reset()
+    entry_obs = carrier0.get("observation")  # static entry buffer
+
+    def rollout_body():
+        carrier = TensorDict(
+            {"observation": entry_obs}, batch_size=[B], device=device
+        )
         with torch.no_grad(), set_exploration_type(ExplorationType.RANDOM):
-            for _ in range(T):
+            for t in range(T):
                 with autocast:
                     carrier = actor(carrier)
                 carrier.set("action", carrier.get("action").float())
                 carrier, next_root = env.step_and_maybe_reset(carrier)
-                snaps.append(carrier.clone(False))
+                for k in STORE_KEYS:
+                    col = store.get(k)[:, t]
+                    col.copy_(carrier.get(k).reshape(col.shape))
                 carrier = next_root
-        batch = td_stack(snaps, 1)  # [B, T]
-        # ---- advantage (bf16 HIP scan on device)
+        # close the loop: replays start from the final observation
+        entry_obs.copy_(carrier.get("observation"))
+
+    rollout = rollout_body
+    if args.graph and cuda:
+        env.enable_capture_mode(True)
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    rollout_body()
+            torch.cuda.current_stream().wait_stream(side)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                rollout_body()
+            rollout = g.replay
+        except Exception as e:  # capture unsupported → eager fallback
+            print(f"[bench] hipGraph capture failed ({e!r}); running eager", file=sys.stderr)
+            rollout = rollout_body
+            args.graph = False
+
+    def one_step():
+        rollout()
+        batch = store
         with torch.no_grad(), autocast:
             gae(batch)
-        # ---- PPO epochs
         flat = batch.reshape(-1)
         n = flat.batch_size[0]
         mb = n // args.minibatches
@@ -175,12 +226,9 @@ def main():
                                 off += k
                 torch.nn.utils.clip_grad_norm_(params, 1.0)
                 optim.step()
-        return carrier
 
-    carrier = env.reset()
-    # warmup
     for _ in range(args.warmup):
-        carrier = one_step(carrier)
+        one_step()
 
     if distributed:
         torch.distributed.barrier()
@@ -188,7 +236,7 @@ def main():
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        carrier = one_step(carrier)
+        one_step()
     if cuda:
         torch.cuda.synchronize()
     if distributed:
@@ -226,6 +274,7 @@ def main():
                 "n_envs_per_gpu": B,
                 "ppo_epochs": args.epochs,
                 "minibatches": args.minibatches,
+                "hip_graph": bool(args.graph and cuda),
             },
         }
         print(json.dumps(result))

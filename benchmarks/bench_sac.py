@@ -1,0 +1,148 @@
+"""Secondary benchmark — SAC with a 1M-transition prioritized replay
+buffer resident in HBM (BASELINE.json config 3: "SAC Humanoid-v4,
+1M-transition PrioritizedReplayBuffer resident in HBM on 1 MI355X").
+
+Metric: SAC optimizer samples/sec.  One step = sample 256 from the PER
+(device sum-tree inverse-CDF), SAC fwd+bwd over twin Q nets, optimizer
+step, priority update (device tree scatter+recompute), plus collecting
+256 fresh frames from a Humanoid-shaped vec env into the buffer.
+
+Usage: python benchmarks/bench_sac.py [--steps K] [--warmup W]
+Prints one JSON line like bench.py.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from rl_amd.data import LazyTensorStorage, TensorDictPrioritizedReplayBuffer
+from rl_amd.envs.custom.synthetic import HumanoidVec
+from rl_amd.envs.utils import ExplorationType, set_exploration_type
+from rl_amd.modules import MLP, NormalParamExtractor, ProbabilisticActor, TanhNormal, ValueOperator
+from rl_amd.objectives import SACLoss, SoftUpdate
+from rl_amd.tensordict import TensorDict, TensorDictModule, stack as td_stack
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=256)
+    p.add_argument("--buffer", type=int, default=1_000_000)
+    p.add_argument("--envs", type=int, default=256)
+    p.add_argument("--utd", type=int, default=4, help="optim steps per env step batch")
+    args = p.parse_args()
+
+    cuda = torch.cuda.is_available()
+    device = torch.device("cuda:0" if cuda else "cpu")
+    torch.manual_seed(0)
+
+    env = HumanoidVec(batch_size=[args.envs], device=device)
+    obs_dim, act_dim = env.obs_dim, env.act_dim
+    actor_net = torch.nn.Sequential(
+        MLP(in_features=obs_dim, out_features=2 * act_dim, num_cells=[256, 256], device=device),
+        NormalParamExtractor(),
+    )
+    actor = ProbabilisticActor(
+        TensorDictModule(actor_net, in_keys=["observation"], out_keys=["loc", "scale"]),
+        in_keys=["loc", "scale"],
+        distribution_class=TanhNormal,
+        return_log_prob=True,
+    )
+    qnet = ValueOperator(
+        MLP(in_features=obs_dim + act_dim, out_features=1, num_cells=[256, 256], device=device),
+        in_keys=["observation", "action"],
+    )
+    loss = SACLoss(actor, qnet, num_qvalue_nets=2)
+    loss.make_value_estimator()
+    loss = loss.to(device)
+    optim = torch.optim.Adam(loss.parameters(), lr=3e-4)
+    updater = SoftUpdate(loss, tau=0.005)
+
+    rb = TensorDictPrioritizedReplayBuffer(
+        storage=LazyTensorStorage(args.buffer, device=device),
+        batch_size=args.batch,
+        alpha=0.7,
+        beta=0.5,
+    )
+
+    carrier = env.reset()
+
+    def collect():
+        nonlocal carrier
+        with torch.no_grad(), set_exploration_type(ExplorationType.RANDOM):
+            carrier = actor(carrier)
+            carrier.set("action", carrier.get("action").float())
+            carrier, nxt = env.step_and_maybe_reset(carrier)
+            rb.extend(carrier.select(
+                "observation", "action", "sample_log_prob",
+                ("next", "observation"), ("next", "reward"),
+                ("next", "done"), ("next", "terminated"),
+            ))
+            carrier = nxt
+
+    def train_step():
+        for _ in range(args.utd):
+            batch = rb.sample()
+            out = loss(batch)
+            total = out.get("loss_actor") + out.get("loss_qvalue") + out.get("loss_alpha")
+            optim.zero_grad(set_to_none=True)
+            total.backward()
+            optim.step()
+            updater.step()
+            rb.update_tensordict_priority(batch)
+
+    # prefill
+    while len(rb) < 4 * args.batch:
+        collect()
+
+    for _ in range(args.warmup):
+        collect()
+        train_step()
+    if cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        collect()
+        train_step()
+    if cuda:
+        torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    samples = args.steps * args.utd * args.batch
+    print(
+        json.dumps(
+            {
+                "metric": "sac_per_samples_per_sec",
+                "value": samples / dt,
+                "unit": "samples/s",
+                "n_gpus": 1,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": dt / args.steps * 1000,
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "fp32",
+                "data": "synthetic",
+                "config": {
+                    "model": "sac_humanoid_mlp256x256",
+                    "global_batch": args.batch,
+                    "buffer_size": args.buffer,
+                    "utd": args.utd,
+                    "parallelism": "dp1",
+                    "buffer_device": str(device),
+                },
+            }
+        )
+    )
+
+
+if __name__ == "__main__":
+    main()

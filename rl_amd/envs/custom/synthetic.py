@@ -72,6 +72,12 @@ class SyntheticMuJoCoEnv(EnvBase):
         self._state: Optional[torch.Tensor] = None
         self._t: Optional[torch.Tensor] = None
         self._gen = torch.Generator(device="cpu")
+        self._capture_safe = False
+
+    def enable_capture_mode(self, mode: bool = True) -> "SyntheticMuJoCoEnv":
+        """In-place state updates for hipGraph capture."""
+        self._capture_safe = mode
+        return self
 
     def _reset(self, tensordict=None, **kwargs) -> TensorDictBase:
         bs = self.batch_size
@@ -88,8 +94,12 @@ class SyntheticMuJoCoEnv(EnvBase):
         new_t = torch.zeros((*bs, 1), device=self.device)
         if tensordict is not None and "_reset" in tensordict and self._state is not None:
             mask = tensordict.get("_reset").reshape(*bs, 1)
-            self._state = torch.where(mask, new_state, self._state)
-            self._t = torch.where(mask, new_t, self._t)
+            if self._capture_safe:
+                self._state.copy_(torch.where(mask, new_state, self._state))
+                self._t.copy_(torch.where(mask, new_t, self._t))
+            else:
+                self._state = torch.where(mask, new_state, self._state)
+                self._t = torch.where(mask, new_t, self._t)
         else:
             self._state = new_state
             self._t = new_t
@@ -107,8 +117,12 @@ class SyntheticMuJoCoEnv(EnvBase):
         action = tensordict.get("action").to(self.dtype).clamp(-1, 1)
         s = self._state
         drive = action @ self.B
-        self._state = torch.tanh(s @ self.A + drive)
-        self._t = self._t + 1
+        new_state = torch.tanh(s @ self.A + drive)
+        # in-place state update keeps the buffers stable — required for
+        # hipGraph capture of the rollout loop (replays rewrite the same
+        # memory; a rebinding would read stale state on replay)
+        self._state = self._state.copy_(new_state) if self._capture_safe else new_state
+        self._t = self._t.add_(1) if self._capture_safe else self._t + 1
         # forward-progress reward: first obs component is "velocity"
         vel = self._state[..., :1]
         ctrl_cost = 0.1 * action.pow(2).sum(-1, keepdim=True)
