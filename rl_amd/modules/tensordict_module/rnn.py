@@ -149,6 +149,7 @@ def lstm_scan(
     gx_all = x @ cell.weight_ih.T + cell.bias_ih
     h, c = h0, c0
     hs = []
+    cs = []
     for t in range(T):
         mask = is_init[:, t].unsqueeze(-1).to(x.dtype)
         h = h * (1 - mask)
@@ -160,7 +161,8 @@ def lstm_scan(
         c = f * c + i * g
         h = o * c.tanh()
         hs.append(h)
-    return torch.stack(hs, 1), h, c
+        cs.append(c)
+    return torch.stack(hs, 1), h, torch.stack(cs, 1)
 
 
 def gru_scan(
@@ -274,8 +276,10 @@ class LSTMModule(_RNNModuleBase):
             if h0 is not None and h0.dim() == 3:
                 h0 = h0[:, 0]
                 c0 = c0[:, 0]
-            ys, h, c = lstm_scan(self.cell, x, is_init, h0, c0)
+            ys, h, cs = lstm_scan(self.cell, x, is_init, h0, c0)
             td.set(self.out_key, ys)
+            td.set(("next", "recurrent_state_h"), ys)
+            td.set(("next", "recurrent_state_c"), cs)
             return td
         # single-step
         h = td.get("recurrent_state_h", None)
@@ -327,6 +331,7 @@ class GRUModule(_RNNModuleBase):
             else:
                 ys, h = gru_scan(self.cell, x, is_init, h0)
             td.set(self.out_key, ys)
+            td.set(("next", "recurrent_state"), ys)
             return td
         h = td.get("recurrent_state", None)
         if h is None:
