@@ -130,7 +130,11 @@ def main():
 
     T, B = args.horizon, args.envs
     frames_per_step = T * B
-    autocast = torch.autocast(device_type="cuda", dtype=dtype, enabled=cuda)
+    # cache_enabled=False: the autocast weight cache allocates during
+    # hipGraph capture, which is forbidden mid-capture
+    autocast = torch.autocast(
+        device_type="cuda", dtype=dtype, enabled=cuda, cache_enabled=False
+    )
     obs_dim, act_dim = env.obs_dim, env.act_dim
 
     # pre-allocated [B, T] rollout store resident in HBM
