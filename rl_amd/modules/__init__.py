@@ -53,3 +53,14 @@ from .models import ConsistentDropout, NoisyLazyLinear, NoisyLinear, gSDEModule,
 from .planners import CEMPlanner, MPCPlannerBase, MPPIPlanner
 from .inference_server import InferenceServer, PolicyClient
 from .llm import LLMWrapperBase, TransformersWrapper
+from .models import (
+    DTActor,
+    DecisionTransformer,
+    DreamerActor,
+    ObsDecoder,
+    ObsEncoder,
+    RSSMPosterior,
+    RSSMPrior,
+    RSSMRollout,
+    WorldModelWrapper,
+)

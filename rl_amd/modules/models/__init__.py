@@ -12,3 +12,16 @@ from .models import (
 )
 from .multiagent import Mixer, MultiAgentConvNet, MultiAgentMLP, QMixer, VDNMixer
 from .exploration import ConsistentDropout, NoisyLazyLinear, NoisyLinear, gSDEModule, reset_noise
+from .model_based import (
+    DreamerActor,
+    ObsDecoder,
+    ObsEncoder,
+    RSSMPosterior,
+    RSSMPrior,
+    RSSMRollout,
+    WorldModelWrapper,
+)
+try:
+    from .decision_transformer import DecisionTransformer, DTActor
+except ImportError:  # transformers not installed
+    DecisionTransformer = DTActor = None

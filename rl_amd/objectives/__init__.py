@@ -20,3 +20,4 @@ from .imitation import BCLoss, DTLoss, GAILLoss, OnlineDTLoss, RNDLoss
 from .multiagent import IPPOLoss, MAPPOLoss, QMixerLoss
 from .redq import CrossQLoss, REDQLoss
 from .llm import CISPOLoss, DAPO, GRPOLoss, MCAdvantage, SFTLoss
+from .dreamer import DreamerActorLoss, DreamerModelLoss, DreamerValueLoss, WorldModelLoss
