@@ -21,3 +21,4 @@ from .multiagent import IPPOLoss, MAPPOLoss, QMixerLoss
 from .redq import CrossQLoss, REDQLoss
 from .llm import CISPOLoss, DAPO, GRPOLoss, MCAdvantage, SFTLoss
 from .dreamer import DreamerActorLoss, DreamerModelLoss, DreamerValueLoss, WorldModelLoss
+from .tqc import TQCLoss

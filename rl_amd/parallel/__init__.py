@@ -9,3 +9,4 @@ from .comm import (
     rendezvous_store,
     send_tensordict,
 )
+from .mailbox import CommandChannel, Mailbox, RequestReply

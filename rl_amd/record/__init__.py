@@ -7,3 +7,4 @@ from .loggers import (
     generate_exp_name,
     get_logger,
 )
+from .recorder import Every, LoggerMonitor, PixelRenderTransform, TensorDictRecorder, VideoRecorder
