@@ -15,3 +15,5 @@ from .gym_like import GymLikeEnv, default_info_dict_reader
 from .model_based import DreamerEnv, ModelBasedEnvBase
 from . import transforms
 from .llm import ChatEnv
+from .async_envs import AsyncEnvPool
+from .env_creator import EnvCreator, env_creator
