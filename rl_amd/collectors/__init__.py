@@ -11,3 +11,9 @@ from .utils import split_trajectories
 from .distributed import DistributedCollector, DistributedSyncCollector
 from .evaluator import Evaluator
 from .llm import LLMCollector
+from .weight_update import (
+    MultiProcessedWeightUpdater,
+    RemoteModuleWeightUpdater,
+    VanillaWeightUpdater,
+    WeightUpdaterBase,
+)

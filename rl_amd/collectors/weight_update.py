@@ -1,0 +1,86 @@
+"""Legacy (v1) weight-updater API kept for reference parity.
+
+Reference: pytorch/rl torchrl/collectors/weight_update.py
+(WeightUpdaterBase, VanillaWeightUpdater, MultiProcessedWeightUpdater,
+RemoteModuleWeightUpdater) — v1 API retained alongside the v2 schemes in
+rl_amd.weight_update.
+"""
+from __future__ import annotations
+
+from typing import Callable, Dict, Iterable, Optional
+
+import torch
+
+from ..tensordict import TensorDict, TensorDictBase
+
+__all__ = [
+    "WeightUpdaterBase",
+    "VanillaWeightUpdater",
+    "MultiProcessedWeightUpdater",
+    "RemoteModuleWeightUpdater",
+]
+
+
+class WeightUpdaterBase:
+    """Push/pull policy weights into a collector's workers."""
+
+    collector = None
+
+    def register_collector(self, collector) -> "WeightUpdaterBase":
+        self.collector = collector
+        return self
+
+    def _get_server_weights(self) -> TensorDictBase:
+        raise NotImplementedError
+
+    def _push_weights(self, weights: TensorDictBase) -> None:
+        raise NotImplementedError
+
+    def update_weights(self, weights: Optional[TensorDictBase] = None) -> None:
+        if weights is None:
+            weights = self._get_server_weights()
+        self._push_weights(weights)
+
+    __call__ = update_weights
+
+
+class VanillaWeightUpdater(WeightUpdaterBase):
+    """Single-process: copy the train policy's weights into the collector
+    policy."""
+
+    def __init__(self, weight_getter: Optional[Callable] = None, policy_weights: Optional[TensorDictBase] = None):
+        self.weight_getter = weight_getter
+        self.policy_weights = policy_weights
+
+    def _get_server_weights(self):
+        if self.weight_getter is not None:
+            src = self.weight_getter()
+            if isinstance(src, TensorDictBase):
+                return src
+            return TensorDict.from_module(src).apply(lambda t: t.detach())
+        return self.policy_weights
+
+    def _push_weights(self, weights):
+        with torch.no_grad():
+            weights.to_module(self.collector.policy)
+
+
+class MultiProcessedWeightUpdater(WeightUpdaterBase):
+    """Broadcast a weight copy to every worker pipe (v1 analog of the
+    pipe scheme)."""
+
+    def __init__(self, get_server_weights: Callable, policy_weights: Optional[Dict] = None):
+        self.get_server_weights = get_server_weights
+
+    def _get_server_weights(self):
+        src = self.get_server_weights()
+        if isinstance(src, TensorDictBase):
+            return src
+        return TensorDict.from_module(src).apply(lambda t: t.detach().cpu())
+
+    def _push_weights(self, weights):
+        self.collector.update_policy_weights_(weights)
+
+
+class RemoteModuleWeightUpdater(MultiProcessedWeightUpdater):
+    """Alias for parity: remote-module flavored pipe updater."""

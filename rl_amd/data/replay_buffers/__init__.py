@@ -32,3 +32,12 @@ from .writers import (
     TensorDictRoundRobinWriter,
     Writer,
 )
+from .extras import (
+    CompressedListStorage,
+    ConsumingSampler,
+    LinearScheduler,
+    ParameterScheduler,
+    PromptGroupSampler,
+    StalenessAwareSampler,
+    StepScheduler,
+)

@@ -44,6 +44,8 @@ __all__ = [
     "LRSchedulerHook",
     "EarlyStopping",
     "SelectKeys",
+    "UTDRHook",
+    "LogValidationReward",
 ]
 
 REPLAY_BUFFER_CLASS_ERROR = "hook expects a replay buffer instance"
@@ -595,3 +597,55 @@ class EarlyStopping(TrainerHookBase):
 
     def register(self, trainer: Trainer, name: str = "early_stopping"):
         trainer.register_op("post_steps", self, name)
+
+
+class UTDRHook(TrainerHookBase):
+    """Log the update-to-data ratio (reference :3024)."""
+
+    def __init__(self, trainer: Trainer):
+        self.trainer = trainer
+
+    def __call__(self, batch=None):
+        frames = max(1, self.trainer.collected_frames)
+        return {"utd_ratio": self.trainer._optim_count / frames}
+
+    def register(self, trainer: Trainer, name: str = "utdr"):
+        trainer.register_op("post_steps_log", self, name)
+
+
+class LogValidationReward(TrainerHookBase):
+    """Periodic eval-env rollouts logged as validation reward
+    (reference :2530)."""
+
+    def __init__(
+        self,
+        *,
+        record_interval: int,
+        record_frames: int,
+        environment,
+        policy_exploration=None,
+        log_keyname: str = "r_evaluation",
+    ):
+        self.record_interval = record_interval
+        self.record_frames = record_frames
+        self.environment = environment
+        self.policy_exploration = policy_exploration
+        self.log_keyname = log_keyname
+        self._count = 0
+
+    def __call__(self, batch=None):
+        self._count += 1
+        if self._count % self.record_interval:
+            return None
+        from ..envs.utils import ExplorationType, set_exploration_type
+
+        with set_exploration_type(ExplorationType.DETERMINISTIC), torch.no_grad():
+            rollout = self.environment.rollout(
+                self.record_frames,
+                policy=self.policy_exploration,
+                break_when_any_done=True,
+            )
+        return {self.log_keyname: rollout.get(("next", "reward")).sum().item()}
+
+    def register(self, trainer: Trainer, name: str = "log_validation_reward"):
+        trainer.register_op("post_steps_log", self, name)
