@@ -189,6 +189,9 @@ def main():
                 rollout_body()
             rollout = g.replay
         except Exception as e:  # capture unsupported → eager fallback
+            import traceback
+
+            traceback.print_exc(file=sys.stderr)
             print(f"[bench] hipGraph capture failed ({e!r}); running eager", file=sys.stderr)
             rollout = rollout_body
             args.graph = False

@@ -115,16 +115,26 @@ class TanhNormal(D.TransformedDistribution):
         self.upscale = upscale
         if tanh_loc:
             loc = (loc / upscale).tanh() * upscale
-        low = torch.as_tensor(low, device=loc.device, dtype=loc.dtype)
-        high = torch.as_tensor(high, device=loc.device, dtype=loc.dtype)
-        self.low = low
-        self.high = high
-        self.non_trivial_bounds = bool((low != -1.0).any() or (high != 1.0).any())
+        # keep python-number bounds as floats: tensorizing them would issue
+        # a blocking H2D copy on EVERY distribution construction (once per
+        # env step in a collector) and break hipGraph capture
+        if isinstance(low, (int, float)) and isinstance(high, (int, float)):
+            self.low = float(low)
+            self.high = float(high)
+            self.non_trivial_bounds = low != -1.0 or high != 1.0
+        else:
+            low = torch.as_tensor(low, device=loc.device, dtype=loc.dtype)
+            high = torch.as_tensor(high, device=loc.device, dtype=loc.dtype)
+            self.low = low
+            self.high = high
+            self.non_trivial_bounds = bool((low != -1.0).any() or (high != 1.0).any())
         base = D.Independent(D.Normal(loc, scale), event_dims)
         transforms = [D.TanhTransform(cache_size=1)]
         if self.non_trivial_bounds:
             transforms.append(
-                D.AffineTransform(loc=(high + low) / 2, scale=(high - low) / 2)
+                D.AffineTransform(
+                    loc=(self.high + self.low) / 2, scale=(self.high - self.low) / 2
+                )
             )
         super().__init__(base, transforms)
 

@@ -339,17 +339,19 @@ class NormalParamExtractor(nn.Module):
         self.scale_mapping = scale_mapping
         self.scale_lb = scale_lb
         if scale_mapping.startswith("biased_softplus"):
-            self._bias = float(scale_mapping.rsplit("_", 1)[-1]) if "_" in scale_mapping[len("biased_softplus"):] else 1.0
+            bias = float(scale_mapping.rsplit("_", 1)[-1]) if "_" in scale_mapping[len("biased_softplus"):] else 1.0
+            import math
+
+            self._inv_softplus_bias = math.log(math.expm1(bias))
         else:
-            self._bias = None
+            self._inv_softplus_bias = None
 
     def forward(self, x):
         loc, scale = x.chunk(2, -1)
         if self.scale_mapping == "exp":
             scale = scale.clamp(-20.0, 2.0).exp()
-        elif self._bias is not None:
-            inv_softplus_bias = torch.tensor(self._bias).expm1().log().item()
-            scale = torch.nn.functional.softplus(scale + inv_softplus_bias)
+        elif self._inv_softplus_bias is not None:
+            scale = torch.nn.functional.softplus(scale + self._inv_softplus_bias)
         else:  # softplus
             scale = torch.nn.functional.softplus(scale)
         return loc, scale.clamp_min(self.scale_lb)
