@@ -26,7 +26,7 @@
 #include <hip/hip_bf16.h>
 
 #define THREADS 256
-#define B_TILE 32
+#define B_TILE 8
 
 namespace {
 
