@@ -1,2 +1,3 @@
 from .pendulum import PendulumEnv
 from .synthetic import AntVec, HalfCheetahVec, HumanoidVec, SyntheticMuJoCoEnv
+from .tictactoe import TicTacToeEnv
