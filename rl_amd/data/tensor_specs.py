@@ -389,6 +389,8 @@ class OneHot(TensorSpec):
         return torch.nn.functional.one_hot(idx, self.n).to(self.dtype)
 
     def to_categorical(self, val: torch.Tensor) -> torch.Tensor:
+        if val.dtype == torch.bool:
+            val = val.to(torch.uint8)
         return val.argmax(-1)
 
     def encode(self, val, *, ignore_device=False):
@@ -444,6 +446,8 @@ class MultiOneHot(TensorSpec):
         return torch.cat(outs, -1)
 
     def to_categorical(self, val) -> torch.Tensor:
+        if val.dtype == torch.bool:
+            val = val.to(torch.uint8)
         outs = []
         start = 0
         for n in self.nvec:

@@ -94,6 +94,20 @@ def unravel_key(key: NestedKey) -> NestedKey:
     return flat
 
 
+def _normalize_index(idx, nb: int):
+    """Anchor an index to the BATCH dims: expand Ellipsis against nb dims
+    so leaf tensors (which have extra feature dims) are indexed only on
+    their leading batch dims."""
+    if idx is Ellipsis:
+        return tuple(slice(None) for _ in range(nb))
+    if isinstance(idx, tuple) and any(i is Ellipsis for i in idx):
+        pos = idx.index(Ellipsis)
+        n_named = len(idx) - 1
+        fill = tuple(slice(None) for _ in range(nb - n_named))
+        return idx[:pos] + fill + idx[pos + 1 :]
+    return idx
+
+
 def _shape_of_index(batch_size: torch.Size, idx) -> torch.Size:
     """Shape of ``empty(batch_size)[idx]`` without allocating (meta tensor)."""
     if isinstance(idx, int):
@@ -485,6 +499,7 @@ class TensorDict(TensorDictBase):
         return self._index(index)
 
     def _index(self, index) -> "TensorDict":
+        index = _normalize_index(index, len(self._batch_size))
         new_bs = _shape_of_index(self._batch_size, index)
         out = {}
         for k, v in self._data.items():
@@ -501,6 +516,7 @@ class TensorDict(TensorDictBase):
             self.set(index, value)
             return
         if isinstance(value, TensorDictBase):
+            index = _normalize_index(index, len(self._batch_size))
             for k in value.keys(True, True):
                 v = value.get(k)
                 if isinstance(v, NonTensorData):
