@@ -17,3 +17,5 @@ from .weight_update import (
     VanillaWeightUpdater,
     WeightUpdaterBase,
 )
+from .async_batched import AsyncBatchedCollector
+from .profiling import ProfileConfig, ProfilerHook, enable_profile
