@@ -43,15 +43,19 @@ class AsyncBatchedCollector(BaseCollector):
         n = self.pool.num_envs
         reset_td = self.pool.reset()
         # submit a first step for every env
+        def _rand_actions(td, ids):
+            acts = td_stack(
+                [self.pool._envs[i].rand_action(td[j].clone(False)) for j, i in enumerate(ids)],
+                0,
+            )
+            td.update(acts)
+            return td
+
         with set_exploration_type(self.exploration_type), torch.no_grad():
             if self.policy is not None:
                 reset_td = self.policy(reset_td)
             else:
-                for i in range(n):
-                    env = self.pool._envs[i]
-                    reset_td[i : i + 1].update(
-                        env.rand_action(reset_td[i].clone(False)).unsqueeze(0)
-                    )
+                reset_td = _rand_actions(reset_td, list(range(n)))
             self.pool.async_step_send(reset_td)
             collected = []
             collected_frames_in_batch = 0
@@ -71,11 +75,7 @@ class AsyncBatchedCollector(BaseCollector):
                 if self.policy is not None:
                     nxt = self.policy(nxt)
                 else:
-                    for j, i in enumerate(ids):
-                        env = self.pool._envs[i]
-                        nxt[j : j + 1].update(
-                            env.rand_action(nxt[j].clone(False)).unsqueeze(0)
-                        )
+                    nxt = _rand_actions(nxt, ids)
                 self.pool.async_step_send(nxt, env_ids=ids)
                 if collected_frames_in_batch >= self.frames_per_batch:
                     from ..tensordict import cat as td_cat
