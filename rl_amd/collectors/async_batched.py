@@ -83,6 +83,10 @@ class AsyncBatchedCollector(BaseCollector):
                     yield td_cat(collected, 0)
                     collected = []
                     collected_frames_in_batch = 0
+            if collected:
+                from ..tensordict import cat as td_cat
+
+                yield td_cat(collected, 0)
 
     def shutdown(self, timeout=None):
         if not self.closed:
