@@ -174,7 +174,11 @@ class QValueModule(TensorDictModuleBase):
                 values = values.masked_fill(~mask, -3.4e38)
         idx = values.argmax(-1)
         if self.action_space == "one_hot":
-            action = torch.nn.functional.one_hot(idx, values.shape[-1]).to(torch.int64)
+            action = torch.nn.functional.one_hot(idx, values.shape[-1])
+            # match the env's declared action dtype (OneHot defaults to bool)
+            action = action.to(
+                self.spec.dtype if self.spec is not None else torch.int64
+            )
         else:
             action = idx
         chosen = values.gather(-1, idx.unsqueeze(-1))
