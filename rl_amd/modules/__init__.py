@@ -64,3 +64,5 @@ from .models import (
     RSSMRollout,
     WorldModelWrapper,
 )
+from .tensordict_module import DecisionTransformerInferenceWrapper, LMHeadActorValueOperator
+from .tensordict_module import SafeModule, SafeProbabilisticModule, SafeSequential

@@ -7,4 +7,4 @@ from .continuous import (
     safeatanh,
     safetanh,
 )
-from .discrete import MaskedCategorical, MaskedOneHotCategorical, OneHotCategorical, Ordinal
+from .discrete import LLMMaskedCategorical, MaskedCategorical, MaskedOneHotCategorical, OneHotCategorical, Ordinal

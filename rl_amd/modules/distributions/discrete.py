@@ -11,7 +11,7 @@ from typing import Optional
 import torch
 from torch import distributions as D
 
-__all__ = ["OneHotCategorical", "MaskedCategorical", "Ordinal", "MaskedOneHotCategorical"]
+__all__ = ["OneHotCategorical", "MaskedCategorical", "Ordinal", "MaskedOneHotCategorical", "LLMMaskedCategorical"]
 
 
 class OneHotCategorical(D.Categorical):
@@ -119,3 +119,21 @@ def _ordinal_logits(scores: torch.Tensor) -> torch.Tensor:
     cum = torch.cumsum(log_sig, -1)
     rev = torch.flip(torch.cumsum(torch.flip(log_one_minus, (-1,)), -1), (-1,)) - log_one_minus
     return cum + rev
+
+
+class LLMMaskedCategorical(MaskedCategorical):
+    """Token-level masked categorical for LLM losses
+    (reference discrete.py:708): mask shape [*, T, V] over vocab per
+    position; log_prob reduces over the sequence with the mask."""
+
+    def __init__(self, logits: torch.Tensor, mask: torch.Tensor, **kwargs):
+        super().__init__(logits=logits, mask=mask, **kwargs)
+        self._token_mask = mask
+
+    def log_prob_token(self, tokens: torch.Tensor) -> torch.Tensor:
+        return super().log_prob(tokens)
+
+    def log_prob(self, tokens: torch.Tensor) -> torch.Tensor:
+        lp = super().log_prob(tokens)
+        valid = self._token_mask.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
+        return (lp * valid.to(lp.dtype)).sum(-1)

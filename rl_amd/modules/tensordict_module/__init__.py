@@ -29,3 +29,5 @@ from .rnn import (
     recurrent_mode,
     set_recurrent_mode,
 )
+from .actors import DecisionTransformerInferenceWrapper, LMHeadActorValueOperator
+from .common import SafeModule, SafeProbabilisticModule, SafeSequential

@@ -424,3 +424,72 @@ class MultiStepActorWrapper(TensorDictModuleBase):
         self._ptr += 1
         td.set(self.action_key, act)
         return td
+
+
+class DecisionTransformerInferenceWrapper(TensorDictModuleBase):
+    """Maintain the (R, s, a) context window for DT rollouts
+    (reference actors.py:1844): keeps the last ``inference_context`` steps,
+    feeds them to the DT actor, emits the action for the current step."""
+
+    def __init__(
+        self,
+        policy: TensorDictModuleBase,
+        *,
+        inference_context: int = 5,
+        observation_key: str = "observation",
+        action_key: str = "action",
+        return_to_go_key: str = "return_to_go",
+    ):
+        super().__init__()
+        self.policy = policy
+        self.inference_context = inference_context
+        self.observation_key = observation_key
+        self.action_key = action_key
+        self.return_to_go_key = return_to_go_key
+        self.in_keys = [observation_key, return_to_go_key]
+        self.out_keys = [action_key]
+        self._obs_hist = None
+        self._act_hist = None
+        self._rtg_hist = None
+
+    def reset(self):
+        self._obs_hist = self._act_hist = self._rtg_hist = None
+
+    def forward(self, td: TensorDictBase) -> TensorDictBase:
+        obs = td.get(self.observation_key).unsqueeze(-2)
+        rtg = td.get(self.return_to_go_key).unsqueeze(-2)
+        if self._obs_hist is None:
+            self._obs_hist = obs
+            self._rtg_hist = rtg
+            self._act_hist = torch.zeros(
+                *obs.shape[:-2], 1, self._infer_act_dim(td), device=obs.device
+            )
+        else:
+            self._obs_hist = torch.cat([self._obs_hist, obs], -2)[..., -self.inference_context :, :]
+            self._rtg_hist = torch.cat([self._rtg_hist, rtg], -2)[..., -self.inference_context :, :]
+            pad_act = torch.zeros_like(self._act_hist[..., :1, :])
+            self._act_hist = torch.cat([self._act_hist, pad_act], -2)[..., -self.inference_context :, :]
+        seq = td.clone(False)
+        seq.set(self.observation_key, self._obs_hist)
+        seq.set(self.action_key, self._act_hist)
+        seq.set(self.return_to_go_key, self._rtg_hist)
+        out = self.policy(seq)
+        act_seq = out.get(self.action_key)
+        action = act_seq[..., -1, :]
+        self._act_hist = torch.cat([self._act_hist[..., :-1, :], action.unsqueeze(-2)], -2)
+        td.set(self.action_key, action)
+        return td
+
+    def _infer_act_dim(self, td):
+        for m in self.policy.modules():
+            if hasattr(m, "action_head"):
+                return m.action_head.out_features
+        raise RuntimeError("cannot infer action dim; pass a DTActor-style policy")
+
+
+class LMHeadActorValueOperator(ActorValueOperator):
+    """Language-model trunk with separate LM head (policy) and value head
+    (reference actors.py:2235)."""
+
+    def __init__(self, base_model: TensorDictModuleBase, lm_head: TensorDictModuleBase, value_head: TensorDictModuleBase):
+        super().__init__(base_model, lm_head, value_head)
