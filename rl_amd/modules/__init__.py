@@ -66,3 +66,4 @@ from .models import (
 )
 from .tensordict_module import DecisionTransformerInferenceWrapper, LMHeadActorValueOperator
 from .tensordict_module import SafeModule, SafeProbabilisticModule, SafeSequential
+from .distributions import LLMMaskedCategorical
