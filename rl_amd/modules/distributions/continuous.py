@@ -74,7 +74,11 @@ class IndependentNormal(D.Independent):
         self.upscale = upscale
         if tanh_loc:
             loc = (loc / upscale).tanh() * upscale
-        super().__init__(D.Normal(loc, scale, **kwargs), event_dims)
+        super().__init__(
+            D.Normal(loc, scale, validate_args=False, **kwargs),
+            event_dims,
+            validate_args=False,
+        )
 
     @property
     def mode(self):
@@ -128,7 +132,12 @@ class TanhNormal(D.TransformedDistribution):
             self.low = low
             self.high = high
             self.non_trivial_bounds = bool((low != -1.0).any() or (high != 1.0).any())
-        base = D.Independent(D.Normal(loc, scale), event_dims)
+        # validate_args=False everywhere: torch's arg validation runs a
+        # device->host sync (torch._is_all_true) per construction — one
+        # blocking sync per env step and a hipGraph-capture blocker
+        base = D.Independent(
+            D.Normal(loc, scale, validate_args=False), event_dims, validate_args=False
+        )
         transforms = [D.TanhTransform(cache_size=1)]
         if self.non_trivial_bounds:
             transforms.append(
@@ -136,7 +145,7 @@ class TanhNormal(D.TransformedDistribution):
                     loc=(self.high + self.low) / 2, scale=(self.high - self.low) / 2
                 )
             )
-        super().__init__(base, transforms)
+        super().__init__(base, transforms, validate_args=False)
 
     @property
     def root_dist(self) -> D.Normal:

@@ -18,6 +18,7 @@ class OneHotCategorical(D.Categorical):
     """Categorical that samples one-hot vectors (reference discrete.py:65)."""
 
     def __init__(self, logits: Optional[torch.Tensor] = None, probs: Optional[torch.Tensor] = None, **kwargs):
+        kwargs.setdefault("validate_args", False)
         super().__init__(probs=probs, logits=logits, **kwargs)
         self.num_samples = self._num_events
 
@@ -74,6 +75,7 @@ class MaskedCategorical(D.Categorical):
         if probs is not None and mask is not None:
             probs = probs * mask
             probs = probs / probs.sum(-1, keepdim=True).clamp_min(1e-10)
+        kwargs.setdefault("validate_args", False)
         super().__init__(probs=probs, logits=logits, **kwargs)
 
     @property
