@@ -47,5 +47,17 @@ with torch.no_grad():
     results["gru_fused_B256_T200_H128_ms"] = timeit_gpu(lambda: ops.gru_fused(cell, x, ii), n=20)
     results["gru_scan_B256_T200_H128_ms"] = timeit_gpu(lambda: gru_scan(cell, x, ii), n=20)
 
+x4 = torch.randn(4096, 64, 64, device="cuda")
+ii4 = torch.rand(4096, 64, device="cuda") < 0.02
+with torch.no_grad():
+    results["gru_fused_B4096_T64_H128_ms"] = timeit_gpu(lambda: ops.gru_fused(cell, x4, ii4), n=20)
+    results["gru_scan_B4096_T64_H128_ms"] = timeit_gpu(lambda: gru_scan(cell, x4, ii4), n=20)
+from rl_amd.modules import LSTMCell
+from rl_amd.modules.tensordict_module.rnn import lstm_scan
+lcell = LSTMCell(64, 128, device="cuda")
+with torch.no_grad():
+    results["lstm_fused_B4096_T64_H128_ms"] = timeit_gpu(lambda: ops.lstm_fused(lcell, x4, ii4), n=20)
+    results["lstm_scan_B4096_T64_H128_ms"] = timeit_gpu(lambda: lstm_scan(lcell, x4, ii4), n=20)
+
 print(json.dumps(results, indent=1))
 open("gpurun_out/kernel_microbench.json", "w").write(json.dumps(results, indent=1))
