@@ -32,6 +32,11 @@ void launch_tree_scan_f64(const double*, const double*, long*, long, long,
 void launch_tree_update_f64(double*, double*, int*, const long*, const double*,
                             long, long, int, void*);
 int gru_fused_lds_bytes(int);
+int fused_actor_lds_bytes(int, int, int, int);
+void launch_fused_actor(const float*, const float*, const float*, const float*,
+                        const float*, const float*, const float*, const float*,
+                        float*, float*, float*, float*, int, int, int, int,
+                        int, float, float, void*);
 int lstm_fused_lds_bytes(int);
 void launch_gru_fused(const float*, const float*, const float*, const bool*,
                       const float*, float*, float*, int, int, int, void*);
@@ -187,6 +192,35 @@ std::vector<torch::Tensor> lstm_fused(torch::Tensor gates_x,
                     c_out.data_ptr<float>(), B, T, H, (void*)stream);
   return {ys, h_out, c_out};
 }
+// Fused actor: MLP(tanh)x2 + heads + TanhNormal sample + log-prob.
+std::vector<torch::Tensor> fused_actor(
+    torch::Tensor obs, torch::Tensor w1, torch::Tensor b1, torch::Tensor w2,
+    torch::Tensor b2, torch::Tensor w3, torch::Tensor b3, torch::Tensor eps,
+    double inv_softplus_bias, double scale_lb, bool want_loc_scale) {
+  TORCH_CHECK(obs.is_cuda() && obs.is_contiguous(), "obs must be device+contig");
+  long B = obs.size(0), O = obs.size(1);
+  long H1 = w1.size(0), H2 = w2.size(0), A = w3.size(0) / 2;
+  TORCH_CHECK(fused_actor_lds_bytes(O, H1, H2, A) <= 160 * 1024,
+              "fused actor exceeds the LDS budget; fall back to eager");
+  auto action = torch::empty({B, A}, obs.options());
+  auto logp = torch::empty({B}, obs.options());
+  torch::Tensor loc, scale;
+  if (want_loc_scale) {
+    loc = torch::empty({B, A}, obs.options());
+    scale = torch::empty({B, A}, obs.options());
+  }
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_fused_actor(
+      obs.data_ptr<float>(), w1.data_ptr<float>(), b1.data_ptr<float>(),
+      w2.data_ptr<float>(), b2.data_ptr<float>(), w3.data_ptr<float>(),
+      b3.data_ptr<float>(), eps.data_ptr<float>(), action.data_ptr<float>(),
+      logp.data_ptr<float>(),
+      want_loc_scale ? loc.data_ptr<float>() : nullptr,
+      want_loc_scale ? scale.data_ptr<float>() : nullptr, B, O, H1, H2, A,
+      (float)inv_softplus_bias, (float)scale_lb, (void*)stream);
+  if (want_loc_scale) return {action, logp, loc, scale};
+  return {action, logp};
+}
 #endif  // RL_AMD_WITH_HIP
 
 // ---------------------------------------------------------------------------
@@ -223,5 +257,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tree_update", &tree_update, "segment-tree leaf update+recompute (HIP)");
   m.def("gru_fused", &gru_fused, "fused GRU forward scan with resets (HIP)");
   m.def("lstm_fused", &lstm_fused, "fused LSTM forward scan with resets (HIP)");
+  m.def("fused_actor", &fused_actor,
+        "fused MLP+TanhNormal actor forward (HIP)");
 #endif
 }
