@@ -219,3 +219,65 @@ def lstm_fused(cell, x: torch.Tensor, is_init: torch.Tensor, h0=None, c0=None):
         c0.float().contiguous() if c0 is not None else torch.Tensor(),
     )
     return ys.to(x.dtype), h.to(x.dtype), c.to(x.dtype)
+
+
+class FusedTanhNormalActor(torch.nn.Module):
+    """Drop-in rollout policy: MLP(tanh)x2 + heads + TanhNormal sample +
+    log-prob as ONE kernel (rl_amd/csrc/fused_actor.hip).
+
+    Wraps the SAME parameters as the eager
+    ``Sequential(MLP[Linear,Tanh,Linear,Tanh,Linear], NormalParamExtractor)``
+    actor, so training updates flow through unchanged; only the no-grad
+    rollout forward uses the fused kernel.  Falls back to the eager actor
+    when grads are required or off-GPU.
+    """
+
+    def __init__(self, eager_actor, *, in_key: str = "observation", scale_lb: float = 1e-4):
+        super().__init__()
+        self.eager_actor = eager_actor
+        self.in_key = in_key
+        self.scale_lb = scale_lb
+        # locate the Linear layers + extractor inside the eager actor
+        import torch.nn as nn
+
+        linears = [m for m in eager_actor.modules() if isinstance(m, nn.Linear)]
+        if len(linears) != 3:
+            raise ValueError("FusedTanhNormalActor expects exactly 3 Linear layers")
+        self.linears = linears
+        from ..modules.models.models import NormalParamExtractor
+
+        extractors = [m for m in eager_actor.modules() if isinstance(m, NormalParamExtractor)]
+        self.inv_softplus_bias = (
+            extractors[0]._inv_softplus_bias if extractors else 0.5413248546129181
+        )
+        self.in_keys = [in_key]
+        self.out_keys = ["action", "sample_log_prob", "loc", "scale"]
+
+    def forward(self, td):
+        obs = td.get(self.in_key)
+        if not (obs.is_cuda and not torch.is_grad_enabled() and HAS_HIP_EXT):
+            return self.eager_actor(td)
+        w1, w2, w3 = self.linears
+        A = w3.out_features // 2
+        eps = torch.randn(obs.shape[0], A, device=obs.device)
+        action, logp, loc, scale = _C.fused_actor(
+            obs.contiguous().float(),
+            w1.weight.detach(), w1.bias.detach(),
+            w2.weight.detach(), w2.bias.detach(),
+            w3.weight.detach(), w3.bias.detach(),
+            eps,
+            float(self.inv_softplus_bias),
+            float(self.scale_lb),
+            True,
+        )
+        td.set("action", action)
+        td.set("sample_log_prob", logp)
+        td.set("loc", loc)
+        td.set("scale", scale)
+        return td
+
+    def get_dist(self, td):
+        return self.eager_actor.get_dist(td)
+
+    def parameters(self, recurse: bool = True):
+        return self.eager_actor.parameters(recurse)
