@@ -60,7 +60,10 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--graph", action="store_true", help="hipGraph-capture the rollout")
     p.add_argument("--no-graph", dest="graph", action="store_false")
-    p.set_defaults(graph=True)
+    p.add_argument("--fused-actor", action="store_true",
+                   help="single-kernel MLP+TanhNormal rollout policy")
+    p.add_argument("--no-fused-actor", dest="fused_actor", action="store_false")
+    p.set_defaults(graph=True, fused_actor=True)
     return p.parse_args()
 
 
@@ -154,6 +157,16 @@ def main():
         device=device,
     )
 
+    rollout_policy = actor
+    if args.fused_actor and cuda:
+        try:
+            from rl_amd.ops import FusedTanhNormalActor
+
+            rollout_policy = FusedTanhNormalActor(actor)
+        except Exception as e:
+            print(f"[bench] fused actor unavailable ({e!r}); eager policy", file=sys.stderr)
+            args.fused_actor = False
+
     carrier0 = env.reset()
     entry_obs = carrier0.get("observation")  # static entry buffer
 
@@ -163,8 +176,11 @@ def main():
         )
         with torch.no_grad(), set_exploration_type(ExplorationType.RANDOM):
             for t in range(T):
-                with autocast:
-                    carrier = actor(carrier)
+                if args.fused_actor and cuda:
+                    carrier = rollout_policy(carrier)
+                else:
+                    with autocast:
+                        carrier = rollout_policy(carrier)
                 carrier.set("action", carrier.get("action").float())
                 carrier, next_root = env.step_and_maybe_reset(carrier)
                 for k in STORE_KEYS:
@@ -282,6 +298,7 @@ def main():
                 "ppo_epochs": args.epochs,
                 "minibatches": args.minibatches,
                 "hip_graph": bool(args.graph and cuda),
+                "fused_actor": bool(args.fused_actor and cuda),
             },
         }
         print(json.dumps(result))
