@@ -31,6 +31,8 @@ __all__ = [
     "HashTransform",
     "TokenizerTransform",
     "ConditionalPolicySwitch",
+    "AutoResetTransform",
+    "VecGymEnvTransform",
 ]
 
 
@@ -384,3 +386,35 @@ class ConditionalPolicySwitch(Transform):
             m = m.unsqueeze(-1)
         td.set("action", torch.where(m.expand_as(action), alt_action, action))
         return td
+
+
+class AutoResetTransform(Transform):
+    """Adapter for auto-resetting envs (reference _env.py:2011 +
+    _misc.py:383 VecGymEnvTransform): libs that reset themselves on done
+    return the POST-reset observation in the done step's `next`; this
+    transform stashes it and re-emits it as the reset observation so the
+    rl_amd step/reset contract holds."""
+
+    def __init__(self, replace: str = "stash", obs_keys: Sequence = ("observation",)):
+        super().__init__()
+        self.obs_keys = [unravel_key(k) for k in obs_keys]
+        self._stash: dict = {}
+
+    def _step(self, td, next_td):
+        done = next_td.get("done", None)
+        if done is not None and bool(done.any()):
+            for k in self.obs_keys:
+                val = next_td.get(k, None)
+                if val is not None:
+                    self._stash[k] = val.clone()
+        return next_td
+
+    def _reset(self, td, td_reset):
+        if self._stash:
+            for k, v in self._stash.items():
+                td_reset.set(k, v)
+            self._stash = {}
+        return td_reset
+
+
+VecGymEnvTransform = AutoResetTransform
