@@ -22,3 +22,4 @@ from .redq import CrossQLoss, REDQLoss
 from .llm import CISPOLoss, DAPO, GRPOLoss, MCAdvantage, SFTLoss
 from .dreamer import DreamerActorLoss, DreamerModelLoss, DreamerValueLoss, WorldModelLoss
 from .tqc import TQCLoss
+from .act import ACTLoss, DiffusionActor, DiffusionBCLoss
