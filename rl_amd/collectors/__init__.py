@@ -20,3 +20,4 @@ from .weight_update import (
 from .async_batched import AsyncBatchedCollector
 from .profiling import ProfileConfig, ProfilerHook, enable_profile
 from .graph import GraphedRollout
+from .rpc import RPCCollector

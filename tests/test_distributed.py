@@ -192,3 +192,30 @@ def test_mp_pipe_weight_sync():
     scheme.send()
     assert MultiProcessWeightSyncScheme.receive_from(child, worker)
     assert (worker.weight == 0.25).all()
+
+
+def _rpc_collector_main(q):
+    from rl_amd.collectors import RPCCollector
+
+    col = RPCCollector(
+        [_env_fn, _env_fn],
+        None,
+        frames_per_batch=40,
+        total_frames=80,
+    )
+    frames = 0
+    for batch in col:
+        frames += batch.numel()
+    col.shutdown()
+    q.put(frames)
+
+
+@pytest.mark.timeout(240)
+def test_rpc_collector():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    p = ctx.Process(target=_rpc_collector_main, args=(q,))
+    p.start()
+    frames = q.get(timeout=200)
+    p.join(30)
+    assert frames == 80
