@@ -63,7 +63,10 @@ def parse_args():
     p.add_argument("--fused-actor", action="store_true",
                    help="single-kernel MLP+TanhNormal rollout policy")
     p.add_argument("--no-fused-actor", dest="fused_actor", action="store_false")
-    p.set_defaults(graph=True, fused_actor=True)
+    p.add_argument("--full-graph", dest="full_graph", action="store_true",
+                   help="capture rollout+GAE+PPO update as one hipGraph")
+    p.add_argument("--no-full-graph", dest="full_graph", action="store_false")
+    p.set_defaults(graph=True, fused_actor=True, full_graph=True)
     return p.parse_args()
 
 
@@ -212,8 +215,10 @@ def main():
             rollout = rollout_body
             args.graph = False
 
-    def one_step():
-        rollout()
+    # full-step graph capture needs stable grad buffers
+    zero_set_to_none = not (args.full_graph and cuda and not distributed)
+
+    def update_phase():
         batch = store
         with torch.no_grad(), autocast:
             gae(batch)
@@ -232,7 +237,7 @@ def main():
                         + out.get("loss_critic")
                         + out.get("loss_entropy")
                     )
-                optim.zero_grad(set_to_none=True)
+                optim.zero_grad(set_to_none=zero_set_to_none)
                 total.backward()
                 if distributed:
                     with torch.no_grad():
@@ -250,8 +255,36 @@ def main():
                 torch.nn.utils.clip_grad_norm_(params, 1.0)
                 optim.step()
 
+    def one_step():
+        rollout()
+        update_phase()
+
+    step_fn = one_step
+    if args.full_graph and cuda and not distributed:
+        # capture EVERYTHING (rollout + GAE + fwd/bwd/Adam): grads must be
+        # pre-allocated and kept (set_to_none=False) so buffers are static
+        try:
+            env.enable_capture_mode(True)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    one_step()
+            torch.cuda.current_stream().wait_stream(side)
+            g_full = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g_full):
+                one_step()
+            step_fn = g_full.replay
+        except Exception as e:
+            import traceback
+
+            traceback.print_exc(file=sys.stderr)
+            print(f"[bench] full-step capture failed ({e!r}); rollout-graph only", file=sys.stderr)
+            args.full_graph = False
+            step_fn = one_step
+
     for _ in range(args.warmup):
-        one_step()
+        step_fn()
 
     if distributed:
         torch.distributed.barrier()
@@ -259,7 +292,7 @@ def main():
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        one_step()
+        step_fn()
     if cuda:
         torch.cuda.synchronize()
     if distributed:
@@ -299,6 +332,7 @@ def main():
                 "minibatches": args.minibatches,
                 "hip_graph": bool(args.graph and cuda),
                 "fused_actor": bool(args.fused_actor and cuda),
+                "full_step_graph": bool(args.full_graph and cuda and not distributed),
             },
         }
         print(json.dumps(result))
