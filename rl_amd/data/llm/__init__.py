@@ -1,1 +1,2 @@
 from .history import ContentBase, History
+from .datasets import PairwisePreferenceDataset, PromptDataset, PromptTensorDictLoader
