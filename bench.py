@@ -259,6 +259,12 @@ def main():
         rollout()
         update_phase()
 
+    def one_step_eager_rollout():
+        # used inside full-step capture: a graph cannot replay another
+        # graph, so the rollout body runs inline
+        rollout_body()
+        update_phase()
+
     step_fn = one_step
     if args.full_graph and cuda and not distributed:
         # capture EVERYTHING (rollout + GAE + fwd/bwd/Adam): grads must be
@@ -269,11 +275,11 @@ def main():
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
                 for _ in range(3):
-                    one_step()
+                    one_step_eager_rollout()
             torch.cuda.current_stream().wait_stream(side)
             g_full = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g_full):
-                one_step()
+                one_step_eager_rollout()
             step_fn = g_full.replay
         except Exception as e:
             import traceback
