@@ -127,6 +127,15 @@ class SerialEnv(BatchedEnvBase):
             raise
 
 
+def _copy_into(dst: TensorDictBase, src: TensorDictBase) -> None:
+    """Copy matching leaves of src into dst in place (dst is a shared-
+    memory slice: only pre-existing keys stay shared)."""
+    for k in dst.keys(True, True):
+        v = src.get(k, None)
+        if v is not None and isinstance(v, torch.Tensor):
+            dst.get(k).copy_(v)
+
+
 def _parallel_worker(
     idx: int,
     pipe,
@@ -140,36 +149,47 @@ def _parallel_worker(
     my_slice = shared_td[idx] if shared_td is not None else None
     pipe.send(("meta", EnvMetaData.build(env)))
     root_td: Optional[TensorDictBase] = None
+    action_keys = env.full_action_spec.keys(True, True) or ["action"]
     try:
         while True:
             cmd, data = pipe.recv()
-            if cmd == "reset":
+            if cmd == "set_buffer":
+                # shared-memory payload buffer: this worker owns row idx
+                my_slice = data[idx]
+                pipe.send(("buffer_set", None))
+            elif cmd == "reset":
                 root_td = env.reset(data)
                 if my_slice is not None:
-                    my_slice.update_(root_td)
+                    _copy_into(my_slice.get("root_next"), root_td)
                     pipe.send(("done", None))
                 else:
                     pipe.send(("done", root_td))
             elif cmd == "step":
+                root_td = root_td if root_td is not None else env.reset()
                 if data is not None:
-                    root_td = root_td if root_td is not None else env.reset()
                     root_td.update(data)
+                elif my_slice is not None:
+                    for k in action_keys:
+                        root_td.set(k, my_slice.get(k).clone())
                 td = env.step(root_td)
                 next_td = td.get("next")
                 if my_slice is not None:
-                    my_slice.get("next").update_(next_td)
+                    _copy_into(my_slice.get("next"), next_td)
                     pipe.send(("done", None))
                 else:
                     pipe.send(("done", next_td))
                 root_td = step_mdp(td)
             elif cmd == "step_and_maybe_reset":
+                root_td = root_td if root_td is not None else env.reset()
                 if data is not None:
-                    root_td = root_td if root_td is not None else env.reset()
                     root_td.update(data)
+                elif my_slice is not None:
+                    for k in action_keys:
+                        root_td.set(k, my_slice.get(k).clone())
                 td, next_root = env.step_and_maybe_reset(root_td)
                 if my_slice is not None:
-                    my_slice.get("next").update_(td.get("next"))
-                    my_slice.get("root_next").update_(next_root)
+                    _copy_into(my_slice.get("next"), td.get("next"))
+                    _copy_into(my_slice.get("root_next"), next_root)
                     pipe.send(("done", None))
                 else:
                     pipe.send(("done", (td.get("next"), next_root)))
@@ -227,6 +247,39 @@ class ParallelEnv(BatchedEnvBase):
         self._set_specs_from_meta(metas[0])
         self.is_closed = False
         self._root_cache: Optional[TensorDictBase] = None
+        if self.shared_memory:
+            self._setup_shared_buffer()
+
+    def _setup_shared_buffer(self):
+        """Allocate the shared-memory payload TensorDict (reference
+        _create_td :1272, share :1414-1424): one [N]-row buffer holding
+        action inputs, the step's ``next`` outputs and the post-reset
+        ``root_next`` roots.  Workers write their row in place; pipes
+        carry only 2-byte acks."""
+        from ..tensordict import TensorDict as _TD
+
+        n = self.num_workers
+        buf = _TD({}, batch_size=[n])
+        for k in self.full_action_spec.keys(True, True):
+            buf.set(k, self.full_action_spec[k].zero())
+        nxt = _TD({}, batch_size=[n])
+        root = _TD({}, batch_size=[n])
+        for k in self.full_observation_spec.keys(True, True):
+            nxt.set(k, self.full_observation_spec[k].zero())
+            root.set(k, self.full_observation_spec[k].zero())
+        for k in self.full_reward_spec.keys(True, True):
+            nxt.set(k, self.full_reward_spec[k].zero())
+        for k in self.full_done_spec.keys(True, True):
+            nxt.set(k, self.full_done_spec[k].zero())
+            root.set(k, self.full_done_spec[k].zero())
+        buf.set("next", nxt)
+        buf.set("root_next", root)
+        buf.share_memory_()
+        self._shared = buf
+        self._send_all("set_buffer", [buf] * n)
+        for pipe in self.parent_pipes:
+            msg, _ = pipe.recv()
+            assert msg == "buffer_set"
 
     def _set_specs_from_meta(self, meta: EnvMetaData):
         n = self.num_workers
@@ -264,33 +317,55 @@ class ParallelEnv(BatchedEnvBase):
             datas.append(sub)
         self._send_all("reset", datas)
         outs = self._recv_all()
-        out = td_stack(outs, 0)
+        if self._shared is not None:
+            out = self._shared.get("root_next").clone()
+        else:
+            out = td_stack(outs, 0)
         if self.device is not None and out.device != self.device:
             out = out.to(self.device)
         return out
 
+    def _write_actions(self, tensordict: TensorDictBase) -> None:
+        for k in self.full_action_spec.keys(True, True):
+            val = tensordict.get(k, None)
+            if val is not None:
+                self._shared.get(k).copy_(val.cpu() if val.device.type != "cpu" else val)
+
     def _step(self, tensordict: TensorDictBase) -> TensorDictBase:
         input_keys = list(self.full_action_spec.keys(True, True))
-        datas = []
-        for i in range(self.num_workers):
-            sub = tensordict[i].select(*input_keys, strict=False).cpu()
-            datas.append(sub)
-        self._send_all("step", datas)
-        outs = self._recv_all()
-        out = td_stack(outs, 0)
+        if self._shared is not None:
+            self._write_actions(tensordict)
+            self._send_all("step", [None] * self.num_workers)
+            self._recv_all()
+            out = self._shared.get("next").clone()
+        else:
+            datas = []
+            for i in range(self.num_workers):
+                sub = tensordict[i].select(*input_keys, strict=False).cpu()
+                datas.append(sub)
+            self._send_all("step", datas)
+            outs = self._recv_all()
+            out = td_stack(outs, 0)
         if self.device is not None and out.device != self.device:
             out = out.to(self.device)
         return out
 
     def step_and_maybe_reset(self, tensordict: TensorDictBase):
         input_keys = list(self.full_action_spec.keys(True, True))
-        datas = []
-        for i in range(self.num_workers):
-            datas.append(tensordict[i].select(*input_keys, strict=False).cpu())
-        self._send_all("step_and_maybe_reset", datas)
-        outs = self._recv_all()
-        next_tds = td_stack([o[0] for o in outs], 0)
-        next_roots = td_stack([o[1] for o in outs], 0)
+        if self._shared is not None:
+            self._write_actions(tensordict)
+            self._send_all("step_and_maybe_reset", [None] * self.num_workers)
+            self._recv_all()
+            next_tds = self._shared.get("next").clone()
+            next_roots = self._shared.get("root_next").clone()
+        else:
+            datas = []
+            for i in range(self.num_workers):
+                datas.append(tensordict[i].select(*input_keys, strict=False).cpu())
+            self._send_all("step_and_maybe_reset", datas)
+            outs = self._recv_all()
+            next_tds = td_stack([o[0] for o in outs], 0)
+            next_roots = td_stack([o[1] for o in outs], 0)
         if self.device is not None:
             next_tds = next_tds.to(self.device)
             next_roots = next_roots.to(self.device)

@@ -118,3 +118,37 @@ class TestParallelEnv:
             n += b.numel()
         assert n == 32
         col.shutdown()
+
+
+class TestParallelEnvSharedMemory:
+    def test_shared_buffer_active(self):
+        env = ParallelEnv(2, make_counting, shared_memory=True)
+        try:
+            assert env._shared is not None
+            assert env._shared.is_shared()
+            td = env.reset()
+            td.set("action", torch.ones(2, 1, dtype=torch.bool))
+            td, nxt = env.step_and_maybe_reset(td)
+            assert (td.get(("next", "observation")) == 1).all()
+        finally:
+            env.close()
+
+    def test_no_buffer_mode_matches(self):
+        env_sh = ParallelEnv(2, make_counting, shared_memory=True)
+        env_pk = ParallelEnv(2, make_counting, shared_memory=False)
+        try:
+            td1 = env_sh.reset()
+            td2 = env_pk.reset()
+            for _ in range(4):
+                act = torch.ones(2, 1, dtype=torch.bool)
+                td1.set("action", act)
+                td2.set("action", act.clone())
+                td1, n1 = env_sh.step_and_maybe_reset(td1)
+                td2, n2 = env_pk.step_and_maybe_reset(td2)
+                assert torch.equal(
+                    td1.get(("next", "observation")), td2.get(("next", "observation"))
+                )
+                td1, td2 = n1, n2
+        finally:
+            env_sh.close()
+            env_pk.close()
