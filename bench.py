@@ -129,7 +129,11 @@ def main():
     env.set_seed(1234 + rank)
 
     params = list(actor.parameters()) + list(critic.parameters())
-    optim = torch.optim.Adam(params, lr=3e-4)
+    optim = torch.optim.Adam(
+        params,
+        lr=3e-4,
+        capturable=bool(args.full_graph and cuda and world == 1),
+    )
     loss_mod = ClipPPOLoss(actor, critic, clip_epsilon=0.2, entropy_coeff=0.01,
                            critic_coeff=0.5, normalize_advantage=True)
     gae = GAE(gamma=0.99, lmbda=0.95, value_network=critic, vectorized=True)
