@@ -10,3 +10,4 @@ from .comm import (
     send_tensordict,
 )
 from .mailbox import CommandChannel, Mailbox, RequestReply
+from .replay_service import ReplayBufferClient, ReplayBufferService

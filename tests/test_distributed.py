@@ -219,3 +219,70 @@ def test_rpc_collector():
     frames = q.get(timeout=200)
     p.join(30)
     assert frames == 80
+
+
+def _replay_service_producer(conn, q):
+    from rl_amd.parallel import ReplayBufferClient
+    from rl_amd.tensordict import TensorDict
+
+    client = ReplayBufferClient(conn)
+    client.extend(TensorDict({"x": torch.arange(10).float().unsqueeze(-1)}, batch_size=[10]))
+    q.put(len(client))
+
+
+@pytest.mark.timeout(120)
+def test_replay_buffer_service_cross_process():
+    from rl_amd.data import LazyTensorStorage, TensorDictReplayBuffer
+    from rl_amd.parallel import ReplayBufferService
+
+    rb = TensorDictReplayBuffer(storage=LazyTensorStorage(100), batch_size=4)
+    service = ReplayBufferService(rb)
+    conn = service.make_client_conn()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    p = ctx.Process(target=_replay_service_producer, args=(conn, q))
+    p.start()
+    n = q.get(timeout=60)
+    p.join(30)
+    assert n == 10
+    assert len(rb) == 10
+    s = rb.sample()
+    assert s["x"].shape == (4, 1)
+    service.shutdown()
+
+
+def _learner_group_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from rl_amd.trainers import LearnerGroup
+
+    torch.manual_seed(100 + rank)  # DIFFERENT init per rank
+    model = torch.nn.Linear(4, 2)
+    group = LearnerGroup(model)  # broadcast_init syncs to rank0 weights
+    x = torch.full((2, 4), float(rank + 1))
+    model(x).sum().backward()
+    group.finalize_grads()
+    q.put((rank, model.weight.sum().item(), model.weight.grad.sum().item()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_learner_group_sync():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_learner_group_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = {}
+    for _ in range(2):
+        rank, wsum, gsum = q.get(timeout=60)
+        outs[rank] = (wsum, gsum)
+    for p in procs:
+        p.join(30)
+    # weights synced at init, grads averaged
+    assert outs[0][0] == pytest.approx(outs[1][0])
+    assert outs[0][1] == pytest.approx(outs[1][1])
