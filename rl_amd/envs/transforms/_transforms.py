@@ -230,12 +230,32 @@ class TensorDictPrimer(Transform):
     """Seed extra keys (e.g. RNN hidden state) into reset output and spec
     (reference _env.py:211)."""
 
-    def __init__(self, primers: Optional[dict] = None, default_value: float = 0.0, **kwargs):
+    def __init__(
+        self,
+        primers: Optional[dict] = None,
+        default_value: float = 0.0,
+        random: bool = False,
+        **kwargs,
+    ):
         super().__init__()
         if primers is None:
             primers = kwargs
         self.primers = {unravel_key(k): v for k, v in primers.items()}
         self.default_value = default_value
+        self.random = random
+
+    def _expanded_shape(self, spec, td) -> tuple:
+        # primer specs are batch-free: prepend the env batch dims
+        return (*td.batch_size, *spec.shape)
+
+    def _make_value(self, spec, td):
+        shape = self._expanded_shape(spec, td)
+        if self.random:
+            return torch.randn(shape, device=td.device, dtype=spec.dtype)
+        val = torch.zeros(shape, device=td.device, dtype=spec.dtype)
+        if self.default_value:
+            val = val + self.default_value
+        return val
 
     def _reset(self, td, td_reset):
         for key, spec in self.primers.items():
@@ -244,10 +264,7 @@ class TensorDictPrimer(Transform):
             ):
                 td_reset.set(key, td.get(key))
             elif key not in td_reset:
-                val = spec.zero()
-                if self.default_value:
-                    val = val + self.default_value
-                td_reset.set(key, val)
+                td_reset.set(key, self._make_value(spec, td_reset))
         return td_reset
 
     def _step(self, td, next_td):
@@ -255,12 +272,15 @@ class TensorDictPrimer(Transform):
             if key not in next_td and key in td:
                 next_td.set(key, td.get(key))
             elif key not in next_td:
-                next_td.set(key, self.primers[key].zero())
+                next_td.set(key, self._make_value(self.primers[key], next_td))
         return next_td
 
     def transform_observation_spec(self, spec):
         for key, s in self.primers.items():
-            spec[key] = s
+            try:
+                spec[key] = s.expand(*spec.shape, *s.shape)
+            except (NotImplementedError, RuntimeError):
+                spec[key] = s
         return spec
 
 
