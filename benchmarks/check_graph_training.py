@@ -17,8 +17,7 @@ from rl_amd.tensordict import TensorDictModule
 
 
 def main(iters: int = 60, envs: int = 1024, T: int = 32):
-    assert torch.cuda.is_available()
-    device = torch.device("cuda")
+    device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
     torch.manual_seed(0)
     env = PendulumEnv(batch_size=[envs], device=device)
     net = torch.nn.Sequential(
@@ -62,12 +61,17 @@ def main(iters: int = 60, envs: int = 1024, T: int = 32):
         rewards.append(r)
         if i % 10 == 0:
             print(f"iter {i}: mean reward {r:.4f}")
-    early = sum(rewards[:10]) / 10
-    late = sum(rewards[-10:]) / 10
+    w = max(1, len(rewards) // 6)
+    early = sum(rewards[:w]) / w
+    late = sum(rewards[-w:]) / w
     print(f"early {early:.4f} -> late {late:.4f}")
     assert late > early + 0.3, "graphed PPO failed to improve Pendulum reward"
     print("LEARNING CHECK PASSED")
 
 
 if __name__ == "__main__":
-    main()
+    import sys
+    kw = {}
+    if len(sys.argv) > 1:
+        kw = dict(iters=int(sys.argv[1]), envs=int(sys.argv[2]), T=int(sys.argv[3]))
+    main(**kw)

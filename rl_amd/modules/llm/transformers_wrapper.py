@@ -124,7 +124,11 @@ class TransformersWrapper(LLMWrapperBase):
             lps.append(logp.gather(-1, tok.unsqueeze(-1)).squeeze(-1))
         log_probs = torch.stack(lps, 1) if lps else torch.zeros_like(resp_tokens, dtype=torch.float)
         resp_texts = self.tokenizer.batch_decode(resp_tokens, skip_special_tokens=True)
-        td.set_non_tensor("text_response", resp_texts)
+        td.set_non_tensor("text_response", resp_texts if len(td.batch_size) else resp_texts[0])
+        if not td.batch_size:
+            # scalar-batch dialog: drop the singleton generation batch dim
+            resp_tokens = resp_tokens.squeeze(0)
+            log_probs = log_probs.squeeze(0)
         td.set("tokens_response", resp_tokens.cpu() if td.device is None else resp_tokens)
         td.set("log_probs", log_probs.cpu() if td.device is None else log_probs)
         # update history
@@ -151,6 +155,8 @@ class TransformersWrapper(LLMWrapperBase):
         resp = td.get("tokens_response")
         if self.device is not None:
             resp = resp.to(self.device)
+        if resp.dim() == 1:  # scalar-batch dialog
+            resp = resp.unsqueeze(0)
         full = torch.cat([ids, resp], 1)
         full_mask = torch.cat([mask, torch.ones_like(resp)], 1)
         logits = self.model(input_ids=full, attention_mask=full_mask).logits

@@ -1284,10 +1284,33 @@ def stack(tds: Sequence[TensorDict], dim: int = 0) -> TensorDict:
     for k in _common_keys(tds):
         vals = [td.get(k) for td in tds]
         if isinstance(vals[0], NonTensorData):
-            out.set(k, vals[0])
+            out.set(k, _merge_non_tensor(vals))
         elif isinstance(vals[0], torch.Tensor):
             out.set(k, torch.stack(vals, dim=dim))
     return out
+
+
+def _merge_non_tensor(vals: Sequence[NonTensorData]) -> NonTensorData:
+    """Combine non-tensor leaves across stacked/concatenated tds:
+    identical values collapse to one; dicts of batch-lists (the History
+    convention: {"roles": [[...]], "contents": [[...]]}) concatenate
+    their outer lists; anything else becomes a list of values."""
+    datas = [v.data if isinstance(v, NonTensorData) else v for v in vals]
+    try:
+        if all(d == datas[0] for d in datas[1:]):
+            return NonTensorData(datas[0])
+    except Exception:
+        pass
+    if all(isinstance(d, dict) for d in datas) and all(
+        set(d.keys()) == set(datas[0].keys()) for d in datas[1:]
+    ) and all(
+        isinstance(v, list) for d in datas for v in d.values()
+    ):
+        merged = {
+            key: [item for d in datas for item in d[key]] for key in datas[0]
+        }
+        return NonTensorData(merged)
+    return NonTensorData(datas)
 
 
 lazy_stack = stack
@@ -1311,7 +1334,7 @@ def cat(tds: Sequence[TensorDict], dim: int = 0) -> TensorDict:
     for k in _common_keys(tds):
         vals = [td.get(k) for td in tds]
         if isinstance(vals[0], NonTensorData):
-            out.set(k, vals[0])
+            out.set(k, _merge_non_tensor(vals))
         elif isinstance(vals[0], torch.Tensor):
             out.set(k, torch.cat(vals, dim=dim))
     return out
