@@ -37,9 +37,11 @@ def main(iters: int = 60, envs: int = 1024, T: int = 32):
     )
     loss_mod = ClipPPOLoss(actor, critic, critic_coeff=0.5, normalize_advantage=True)
     gae = GAE(gamma=0.99, lmbda=0.95, value_network=critic)
-    optim = torch.optim.Adam(loss_mod.parameters(), lr=3e-4)
+    optim = torch.optim.Adam(loss_mod.parameters(), lr=1e-3)
     gr = GraphedRollout(env, actor, horizon=T).initialize()
     print("captured:", gr.captured)
+    if device.type == "cuda":
+        assert gr.captured, "rollout graph capture failed on GPU — the check must exercise the captured path"
     rewards = []
     for i in range(iters):
         batch = gr.collect()

@@ -65,6 +65,13 @@ class PendulumEnv(EnvBase):
         self._th: Optional[torch.Tensor] = None
         self._thdot: Optional[torch.Tensor] = None
         self._gen = torch.Generator(device="cpu")
+        self._capture_safe = False
+
+    def enable_capture_mode(self, mode: bool = True) -> "PendulumEnv":
+        """In-place state updates + device-side RNG so the step/reset
+        loop is hipGraph-capturable (no host syncs, stable buffers)."""
+        self._capture_safe = mode
+        return self
 
     def _obs(self) -> torch.Tensor:
         return torch.stack(
@@ -86,22 +93,33 @@ class PendulumEnv(EnvBase):
         bs = self.batch_size
         high_th = DEFAULT_X
         high_thdot = DEFAULT_Y
-        new_th = (
-            torch.rand(bs or (), generator=self._gen).to(self.device) * 2 * high_th
-            - high_th
-        )
-        new_thdot = (
-            torch.rand(bs or (), generator=self._gen).to(self.device) * 2 * high_thdot
-            - high_thdot
-        )
+        if self.device is not None and self.device.type == "cuda":
+            # device-side RNG (default CUDA generator is graph-safe)
+            new_th = torch.rand(bs or (), device=self.device) * 2 * high_th - high_th
+            new_thdot = (
+                torch.rand(bs or (), device=self.device) * 2 * high_thdot - high_thdot
+            )
+        else:
+            new_th = (
+                torch.rand(bs or (), generator=self._gen).to(self.device) * 2 * high_th
+                - high_th
+            )
+            new_thdot = (
+                torch.rand(bs or (), generator=self._gen).to(self.device) * 2 * high_thdot
+                - high_thdot
+            )
         if (
             tensordict is not None
             and "_reset" in tensordict
             and self._th is not None
         ):
             mask = tensordict.get("_reset").reshape(bs or ())
-            self._th = torch.where(mask, new_th, self._th)
-            self._thdot = torch.where(mask, new_thdot, self._thdot)
+            if self._capture_safe:
+                self._th.copy_(torch.where(mask, new_th, self._th))
+                self._thdot.copy_(torch.where(mask, new_thdot, self._thdot))
+            else:
+                self._th = torch.where(mask, new_th, self._th)
+                self._thdot = torch.where(mask, new_thdot, self._thdot)
         else:
             self._th = new_th
             self._thdot = new_thdot
@@ -118,12 +136,19 @@ class PendulumEnv(EnvBase):
         newthdot = thdot + (3 * g / (2 * l) * th.sin() + 3.0 / (m * l**2) * u) * dt
         newthdot = newthdot.clamp(-self.max_speed, self.max_speed)
         newth = th + newthdot * dt
-        self._th = newth
-        self._thdot = newthdot
+        if self._capture_safe:
+            # stable buffers: hipGraph replays rewrite the same memory
+            self._th.copy_(newth)
+            self._thdot.copy_(newthdot)
+        else:
+            self._th = newth
+            self._thdot = newthdot
         bs = self.batch_size
         return self._make_td({"reward": -costs.reshape(*bs, 1)})
 
     def _set_seed(self, seed: Optional[int]):
         if seed is not None:
             self._gen.manual_seed(seed)
+            if self.device is not None and self.device.type == "cuda":
+                torch.cuda.manual_seed(seed)
         return seed

@@ -6,6 +6,7 @@ from .models import (
     DdpgCnnQNet,
     DdpgMlpActor,
     DdpgMlpQNet,
+    DistributionalDQNnet,
     DuelingCnnDQNet,
     DuelingMlpDQNet,
     NormalParamExtractor,

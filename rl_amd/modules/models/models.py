@@ -261,6 +261,24 @@ class DuelingCnnDQNet(nn.Module):
         return val + adv - adv.mean(-1, keepdim=True)
 
 
+class DistributionalDQNnet(nn.Module):
+    """Log-softmax head for distributional (C51-style) Q-networks
+    (reference models.py DistributionalDQNnet): normalizes the atom
+    dimension of a ``[*, n_atoms, n_actions]`` logit tensor so
+    :class:`~rl_amd.modules.DistributionalQValueModule` receives proper
+    log-probabilities."""
+
+    def __init__(self, in_keys=None, out_keys=None):
+        super().__init__()
+        self.in_keys = in_keys
+        self.out_keys = out_keys
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.dim() < 2:
+            raise ValueError("DistributionalDQNnet expects [*, n_atoms, n_actions]")
+        return torch.log_softmax(x, dim=-2)
+
+
 class DdpgMlpActor(nn.Module):
     """DDPG MLP actor (reference models.py:1207)."""
 
