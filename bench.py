@@ -66,7 +66,10 @@ def parse_args():
     p.add_argument("--full-graph", dest="full_graph", action="store_true",
                    help="capture rollout+GAE+PPO update as one hipGraph")
     p.add_argument("--no-full-graph", dest="full_graph", action="store_false")
-    p.set_defaults(graph=True, fused_actor=True, full_graph=True)
+    p.add_argument("--splitk", dest="splitk", action="store_true",
+                   help="split-K HIP wgrad kernel in the update backward")
+    p.add_argument("--no-splitk", dest="splitk", action="store_false")
+    p.set_defaults(graph=True, fused_actor=True, full_graph=True, splitk=True)
     return p.parse_args()
 
 
@@ -127,6 +130,18 @@ def main():
     torch.manual_seed(1234 + rank)
     env, actor, critic = build(args, device)
     env.set_seed(1234 + rank)
+
+    if args.splitk and cuda:
+        # minibatch backward wgrad via the split-K kernel (csrc/wgrad.hip):
+        # hipBLASLt runs these skinny [H,16k]x[16k,H] reductions on one
+        # workgroup (measured 101us, 12.9% of the step in the r11 profile)
+        from rl_amd.ops import HAS_HIP_EXT, convert_linears_to_splitk
+
+        if HAS_HIP_EXT:
+            convert_linears_to_splitk(actor)
+            convert_linears_to_splitk(critic)
+        else:
+            args.splitk = False
 
     params = list(actor.parameters()) + list(critic.parameters())
     optim = torch.optim.Adam(
@@ -342,6 +357,7 @@ def main():
                 "minibatches": args.minibatches,
                 "hip_graph": bool(args.graph and cuda),
                 "fused_actor": bool(args.fused_actor and cuda),
+                "splitk_wgrad": bool(args.splitk and cuda),
                 "full_step_graph": bool(args.full_graph and cuda and not distributed),
             },
         }

@@ -38,6 +38,8 @@ void launch_fused_actor(const float*, const float*, const float*, const float*,
                         float*, float*, float*, float*, int, int, int, int,
                         int, float, float, void*);
 int lstm_fused_lds_bytes(int);
+void launch_wgrad_splitk(const void*, const void*, float*, float*, long,
+                         int, int, void*);
 void launch_gru_fused(const float*, const float*, const float*, const bool*,
                       const float*, float*, float*, int, int, int, void*);
 void launch_lstm_fused(const float*, const float*, const bool*, const float*,
@@ -221,6 +223,28 @@ std::vector<torch::Tensor> fused_actor(
   if (want_loc_scale) return {action, logp, loc, scale};
   return {action, logp};
 }
+// Split-K weight gradient: dW = dY^T X (+ dBias), bf16 in / fp32 accum.
+std::vector<torch::Tensor> wgrad_splitk(torch::Tensor dy, torch::Tensor x,
+                                        bool want_bias) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda(), "wgrad_splitk: device tensors");
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 &&
+                  x.scalar_type() == torch::kBFloat16,
+              "wgrad_splitk: bf16 inputs");
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous(), "contiguous inputs");
+  TORCH_CHECK(dy.size(0) == x.size(0), "K mismatch");
+  long K = dy.size(0);
+  int N = (int)dy.size(1), M = (int)x.size(1);
+  auto opts = dy.options().dtype(torch::kFloat32);
+  auto dw = torch::zeros({N, M}, opts);
+  torch::Tensor db;
+  if (want_bias) db = torch::zeros({N}, opts);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_wgrad_splitk(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(),
+                      want_bias ? db.data_ptr<float>() : nullptr, K, N, M,
+                      (void*)stream);
+  if (want_bias) return {dw, db};
+  return {dw};
+}
 #endif  // RL_AMD_WITH_HIP
 
 // ---------------------------------------------------------------------------
@@ -250,6 +274,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
 #ifdef RL_AMD_WITH_HIP
   m.def("gae", &gae, "fused GAE scan (HIP)");
+  m.def("wgrad_splitk", &wgrad_splitk,
+        "split-K skinny weight gradient dW=dY^T X (HIP)");
   m.def("revscan", &revscan, "generic reverse linear-recurrence scan (HIP)");
   m.def("vtrace", &vtrace, "fused V-trace (HIP)");
   m.def("tree_scan_lower_bound", &tree_scan_lower_bound,

@@ -281,3 +281,73 @@ class FusedTanhNormalActor(torch.nn.Module):
 
     def parameters(self, recurse: bool = True):
         return self.eager_actor.parameters(recurse)
+
+
+# --------------------------------------------------------------------------- #
+# Split-K weight gradient for skinny linear layers
+# --------------------------------------------------------------------------- #
+class _SplitKLinearFn(torch.autograd.Function):
+    """F.linear with the weight gradient computed by the split-K HIP
+    kernel (csrc/wgrad.hip).
+
+    Why: measured on MI355X (rocprofv3, PPO bench r11), hipBLASLt runs
+    the [64,16384]x[16384,64] wgrad of a 16k-row minibatch on ONE
+    workgroup (no split-K) at ~101 us — 12.9% of the training step.  The
+    split-K kernel spreads K over ~128 workgroups with fp32 atomic
+    accumulation and fuses the bias gradient.
+    """
+
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, x, weight, bias):
+        x = x.contiguous()
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        return torch.nn.functional.linear(x, weight, bias)
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        dx = (dy @ weight).reshape(x.shape)
+        if dy2.is_cuda and dy2.dtype == torch.bfloat16 and HAS_HIP_EXT:
+            outs = _C.wgrad_splitk(dy2, x2, ctx.has_bias)
+            dw = outs[0].to(weight.dtype)
+            db = outs[1].to(dy.dtype) if ctx.has_bias else None
+        else:
+            dw = (dy2.t().float() @ x2.float()).to(weight.dtype)
+            db = dy2.float().sum(0).to(dy.dtype) if ctx.has_bias else None
+        return dx, dw, db
+
+
+class SplitKLinear(torch.nn.Linear):
+    """nn.Linear whose backward uses the split-K wgrad kernel on
+    bf16/cuda inputs (eager semantics elsewhere)."""
+
+    def forward(self, x):
+        if x.is_cuda and HAS_HIP_EXT:
+            return _SplitKLinearFn.apply(x, self.weight, self.bias)
+        return super().forward(x)
+
+
+def convert_linears_to_splitk(module: torch.nn.Module) -> torch.nn.Module:
+    """Swap every nn.Linear in ``module`` for a weight-sharing
+    :class:`SplitKLinear` (in place)."""
+    for name, child in module.named_children():
+        if type(child) is torch.nn.Linear:
+            new = SplitKLinear(
+                child.in_features,
+                child.out_features,
+                bias=child.bias is not None,
+                device=child.weight.device,
+                dtype=child.weight.dtype,
+            )
+            new.weight = child.weight
+            new.bias = child.bias
+            setattr(module, name, new)
+        else:
+            convert_linears_to_splitk(child)
+    return module
