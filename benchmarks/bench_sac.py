@@ -38,6 +38,10 @@ def main():
     p.add_argument("--buffer", type=int, default=1_000_000)
     p.add_argument("--envs", type=int, default=256)
     p.add_argument("--utd", type=int, default=4, help="optim steps per env step batch")
+    p.add_argument("--graph", dest="graph", action="store_true",
+                   help="hipGraph-capture the SAC update (loss fwd/bwd + Adam + soft-update)")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
+    p.set_defaults(graph=True)
     args = p.parse_args()
 
     cuda = torch.cuda.is_available()
@@ -63,7 +67,8 @@ def main():
     loss = SACLoss(actor, qnet, num_qvalue_nets=2)
     loss.make_value_estimator()
     loss = loss.to(device)
-    optim = torch.optim.Adam(loss.parameters(), lr=3e-4)
+    use_graph = bool(args.graph and cuda)
+    optim = torch.optim.Adam(loss.parameters(), lr=3e-4, capturable=use_graph)
     updater = SoftUpdate(loss, tau=0.005)
 
     rb = TensorDictPrioritizedReplayBuffer(
@@ -88,20 +93,59 @@ def main():
             ))
             carrier = nxt
 
+    def update_body(batch):
+        out = loss(batch)
+        total = out.get("loss_actor") + out.get("loss_qvalue") + out.get("loss_alpha")
+        optim.zero_grad(set_to_none=not use_graph)
+        total.backward()
+        optim.step()
+        updater.step()
+
+    # hipGraph-captured update: the SAC update at batch 256 is pure launch
+    # overhead (hundreds of ~5us kernels); a single graph replay removes
+    # it.  PER sample / priority write stay eager (buffer length grows).
+    static_batch = None
+    graph = None
+
+    def capture_update():
+        nonlocal static_batch, graph
+        static_batch = rb.sample().clone()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                update_body(static_batch)
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            update_body(static_batch)
+        graph = g
+
     def train_step():
         for _ in range(args.utd):
             batch = rb.sample()
-            out = loss(batch)
-            total = out.get("loss_actor") + out.get("loss_qvalue") + out.get("loss_alpha")
-            optim.zero_grad(set_to_none=True)
-            total.backward()
-            optim.step()
-            updater.step()
-            rb.update_tensordict_priority(batch)
+            if graph is not None:
+                for k in list(static_batch.keys(True, True)):
+                    static_batch.get(k).copy_(batch.get(k))
+                graph.replay()
+                rb.update_tensordict_priority(static_batch)
+            else:
+                update_body(batch)
+                rb.update_tensordict_priority(batch)
 
     # prefill
     while len(rb) < 4 * args.batch:
         collect()
+
+    if use_graph:
+        try:
+            capture_update()
+        except Exception:
+            import traceback
+
+            traceback.print_exc()
+            graph = None
+            use_graph = False
 
     for _ in range(args.warmup):
         collect()
@@ -138,6 +182,7 @@ def main():
                     "utd": args.utd,
                     "parallelism": "dp1",
                     "buffer_device": str(device),
+                    "update_graph": bool(graph is not None),
                 },
             }
         )
