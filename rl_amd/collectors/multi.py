@@ -46,7 +46,8 @@ class _Interruptor:
     (reference collectors/_constants.py)."""
 
     def __init__(self):
-        self._stop = mp.Value("b", False)
+        # spawn context: the flag crosses into spawn-context workers
+        self._stop = mp.get_context("spawn").Value("b", False)
 
     def start_collection(self):
         with self._stop.get_lock():
@@ -321,6 +322,18 @@ class MultiSyncCollector(_MultiCollectorBase):
                 ):
                     self.interruptor.stop_collection()
             ordered = [results[i] for i in sorted(results)]
+            if self.interruptor is not None:
+                # preempted stragglers return fewer time steps; right-pad
+                # them (zeros) so the worker batches stack
+                t_max = max(b.batch_size[-1] for b in ordered)
+                from ..tensordict import pad as td_pad
+
+                ordered = [
+                    b
+                    if b.batch_size[-1] == t_max
+                    else td_pad(b, [0, 0] * (len(b.batch_size) - 1) + [0, t_max - b.batch_size[-1]])
+                    for b in ordered
+                ]
             cat_results = self.cat_results
             if cat_results in (None, "stack"):
                 out = td_stack(ordered, 0)

@@ -274,3 +274,47 @@ class TestGraphedRollout:
         a = out.get("action")[:, 0].clamp(-1, 1)
         expected = torch.tanh(o @ env.A + a @ env.B)
         assert torch.allclose(expected, out.get(("next", "observation"))[:, 0], atol=1e-4)
+
+
+def _make_slow_env():
+    import time as _time
+
+    from rl_amd.testing import ContinuousActionVecMockEnv
+
+    class SlowEnv(ContinuousActionVecMockEnv):
+        def _step(self, td):
+            _time.sleep(0.05)
+            return super()._step(td)
+
+    return SlowEnv(batch_size=[2], max_steps=10)
+
+
+def _make_fast_env():
+    from rl_amd.testing import ContinuousActionVecMockEnv
+
+    return ContinuousActionVecMockEnv(batch_size=[2], max_steps=10)
+
+
+class TestPreemption:
+    @pytest.mark.timeout(180)
+    def test_preemptive_threshold_pads_stragglers(self):
+        """One slow worker + one fast: with preemptive_threshold=0.5 the
+        fast worker's completion interrupts the slow one; its partial
+        batch is zero-padded so the stack still shapes up (reference
+        behavior: _multi_sync preemption)."""
+        col = MultiSyncCollector(
+            [_make_fast_env, _make_slow_env],
+            frames_per_batch=80,
+            total_frames=160,
+            preemptive_threshold=0.5,
+        )
+        try:
+            n = 0
+            for batch in col:
+                # [workers=2, envs=2, T<=20]
+                assert batch.batch_size[0] == 2
+                assert batch.batch_size[-1] <= 20
+                n += 1
+            assert n == 2
+        finally:
+            col.shutdown()
