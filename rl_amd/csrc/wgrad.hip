@@ -117,7 +117,7 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_splitk_kernel(
 
 }  // namespace
 
-void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
+extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
                          float* dbias, long K, int N, int M, void* stream) {
   const int slabs = (int)((K + K_SLAB - 1) / K_SLAB);
   dim3 grid(slabs, (N + 63) / 64, (M + 63) / 64);
