@@ -2,3 +2,4 @@ from .pendulum import PendulumEnv
 from .synthetic import AntVec, HalfCheetahVec, HumanoidVec, SyntheticMuJoCoEnv
 from .tictactoe import TicTacToeEnv
 from .trading import TradingEnv
+from .vla import ToyVLAEnv

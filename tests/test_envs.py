@@ -260,3 +260,60 @@ class TestTransforms:
         env.append_transform(StepCounter())
         td = env.reset()
         assert "step_count" in td
+
+
+class TestToyVLAEnv:
+    def test_echo_mode_schema(self):
+        from rl_amd.envs import ToyVLAEnv, check_env_specs
+
+        env = ToyVLAEnv(batch_size=[2], seed=0)
+        check_env_specs(env)
+        td = env.reset()
+        assert td["observation", "image"].shape == (2, 3, 16, 16)
+        assert td["observation", "image"].dtype == torch.uint8
+        td.set("action", 0.5 * torch.ones(2, 4))
+        td = env.step(td)
+        # the state echoes the executed action
+        assert (td["next", "observation", "state"][:, :4] == 0.5).all()
+        # effort penalty
+        assert (td["next", "reward"] < 0).all()
+        assert not td["next", "done"].any()
+
+    def test_tracking_oracle_succeeds(self):
+        from rl_amd.envs import ToyVLAEnv
+
+        env = ToyVLAEnv(action_dim=2, state_dim=4, success_steps=2, seed=0)
+        td = env.reset()
+        for _ in range(2):
+            td.set("action", td["observation", "state"][..., 2:4])
+            td = env.step(td)["next"].exclude("reward")
+        assert bool(td["success"].item()) and bool(td["terminated"].item())
+
+    def test_tracking_random_rarely_succeeds(self):
+        from rl_amd.envs import ToyVLAEnv
+
+        env = ToyVLAEnv(batch_size=[8], action_dim=4, state_dim=8, success_steps=3, seed=1)
+        r = env.rollout(10, break_when_any_done=False)
+        assert not r["next", "success"].all()
+
+    def test_group_repeats(self):
+        from rl_amd.envs import ToyVLAEnv
+
+        env = ToyVLAEnv(action_dim=2, state_dim=4, success_steps=1, group_repeats=2, seed=0)
+        targets, gids = [], []
+        for _ in range(4):
+            td = env.reset()
+            targets.append(td["observation", "state"][..., 2:4].clone())
+            gids.append(int(td["group_id"].item()))
+        assert torch.equal(targets[0], targets[1])
+        assert not torch.equal(targets[1], targets[2])
+        assert gids == [0, 0, 1, 1]
+
+    def test_from_pixels_render(self):
+        from rl_amd.envs import ToyVLAEnv
+
+        env = ToyVLAEnv(batch_size=[2], success_steps=3, state_dim=8, from_pixels=True, seed=0)
+        td = env.reset()
+        assert td["pixels"].shape == (2, 64, 64, 3)
+        # target marker drawn in green channel
+        assert (td["pixels"][..., 1] > 0).any()
