@@ -17,5 +17,5 @@ from . import transforms
 from .llm import ChatEnv
 from .async_envs import AsyncEnvPool
 from .env_creator import EnvCreator, env_creator
-from .custom import TicTacToeEnv, ToyVLAEnv
+from .custom import ChessEnv, TicTacToeEnv, ToyVLAEnv
 from . import libs

@@ -1,4 +1,5 @@
 from .pendulum import PendulumEnv
+from .chess import ChessEnv
 from .synthetic import AntVec, HalfCheetahVec, HumanoidVec, SyntheticMuJoCoEnv
 from .tictactoe import TicTacToeEnv
 from .trading import TradingEnv

@@ -317,3 +317,23 @@ class TestToyVLAEnv:
         assert td["pixels"].shape == (2, 64, 64, 3)
         # target marker drawn in green channel
         assert (td["pixels"][..., 1] > 0).any()
+
+
+class TestChessEnv:
+    def test_gated_import_error(self):
+        """chess is not installed in this image: the env must raise a
+        clear ImportError (reference gating pattern)."""
+        import importlib.util
+
+        from rl_amd.envs import ChessEnv
+
+        if importlib.util.find_spec("chess") is None:
+            with pytest.raises(ImportError, match="chess"):
+                ChessEnv()
+        else:
+            env = ChessEnv()
+            td = env.reset()
+            assert td["action_mask"].sum() == 20  # legal openings
+            td.set("action", torch.tensor([0]))
+            td = env.step(td)
+            assert not td["next", "done"].item()
