@@ -19,7 +19,7 @@ from .iql import DiscreteIQLLoss, IQLLoss
 from .imitation import BCLoss, DTLoss, GAILLoss, OnlineDTLoss, RNDLoss
 from .multiagent import IPPOLoss, MAPPOLoss, QMixerLoss
 from .redq import CrossQLoss, REDQLoss
-from .llm import CISPOLoss, DAPO, GRPOLoss, MCAdvantage, SFTLoss
+from .llm import CISPOLoss, DAPO, DistillationLoss, GRPOLoss, MCAdvantage, SFTLoss
 from .dreamer import DreamerActorLoss, DreamerModelLoss, DreamerValueLoss, WorldModelLoss
 from .tqc import TQCLoss
 from .act import ACTLoss, DiffusionActor, DiffusionBCLoss

@@ -1,1 +1,1 @@
-from .grpo import CISPOLoss, DAPO, GRPOLoss, MCAdvantage, SFTLoss
+from .grpo import CISPOLoss, DAPO, DistillationLoss, GRPOLoss, MCAdvantage, SFTLoss

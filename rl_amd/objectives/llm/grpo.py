@@ -174,3 +174,71 @@ class SFTLoss(LossModule):
         else:
             loss = nll.mean() if self.reduction == "mean" else nll.sum()
         return TensorDict({"loss_sft": loss}, batch_size=[])
+
+
+class DistillationLoss(LossModule):
+    """Token-level knowledge distillation for LLM policies
+    (reference torchrl/objectives/llm/distillation.py:105).
+
+    Minimizes a KL between student and teacher per-token log-probs with
+    the k3 estimator (same approximation as the GRPO/SFT KL
+    regularizers).  Student log-probs come from running
+    ``actor_network`` in log-prob mode; teacher log-probs are read from
+    the tensordict (``ref_log_probs``) — written by an offline scoring
+    pass or a teacher wrapper.
+
+    ``kl_direction``:
+      * ``"reverse"`` (default): KL(student || teacher) — on-policy
+        (tokens sampled from the student).
+      * ``"forward"``: KL(teacher || student) — teacher-generated data.
+
+    Gradients flow through the student log-probs only.
+    """
+
+    @dataclasses.dataclass
+    class _AcceptedKeys:
+        log_probs: str = "log_probs"
+        ref_log_probs: str = "ref_log_probs"
+        mask: str = "attention_mask_response"
+
+    def __init__(
+        self,
+        actor_network: TensorDictModuleBase,
+        *,
+        kl_direction: str = "reverse",
+        coeff: float = 1.0,
+    ):
+        if kl_direction not in ("reverse", "forward"):
+            raise ValueError("kl_direction must be 'reverse' or 'forward'")
+        super().__init__()
+        self.actor_network = actor_network
+        self.kl_direction = kl_direction
+        self.coeff = coeff
+
+    def forward(self, tensordict: TensorDictBase) -> TensorDictBase:
+        keys = self.tensor_keys
+        td = tensordict.clone(False)
+        teacher_lp = td.get(keys.ref_log_probs)
+        self.actor_network.generate = False
+        out = self.actor_network(td)
+        student_lp = out.get(keys.log_probs)
+        # k3 estimator: KL(p||q) ~ E_p[ exp(lq-lp) - 1 - (lq-lp) ]
+        if self.kl_direction == "reverse":
+            lr = teacher_lp.detach() - student_lp
+        else:
+            lr = student_lp - teacher_lp.detach()
+            # forward KL on teacher samples: gradient still through student
+        kl = lr.exp() - 1 - lr
+        mask = td.get(keys.mask, None)
+        if mask is not None:
+            kl = kl * mask
+            loss = kl.sum() / mask.sum().clamp_min(1)
+        else:
+            loss = kl.mean()
+        return TensorDict(
+            {
+                "loss_distill": self.coeff * loss,
+                "kl_to_teacher": loss.detach(),
+            },
+            batch_size=[],
+        )

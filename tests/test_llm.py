@@ -191,3 +191,42 @@ class TestEndToEndGRPO:
             not torch.equal(a, b) for a, b in zip(w0, lm.parameters())
         )
         assert changed, "GRPO loop did not update the policy weights"
+
+
+class TestDistillation:
+    def test_distill_zero_when_matching(self, lm, tok):
+        from rl_amd.objectives import DistillationLoss
+
+        # build a rollout and use the SAME model as teacher: KL ~ 0
+        gen = TransformersWrapper(lm, tokenizer=tok, generate=True, max_new_tokens=4)
+        h = History.from_text(["q"] * 2)
+        td = TensorDict({}, batch_size=[2])
+        td.set_non_tensor("history", {"roles": h.roles, "contents": h.contents})
+        td = gen(td)
+        hist = td.get_non_tensor("history")
+        td.set_non_tensor("history", {"roles": [r[:-1] for r in hist["roles"]], "contents": [c[:-1] for c in hist["contents"]]})
+        actor = TransformersWrapper(lm, tokenizer=tok, generate=False)
+        scored = actor(td.clone(False))
+        td.set("ref_log_probs", scored.get("log_probs").detach())
+        loss = DistillationLoss(actor)
+        out = loss(td)
+        assert out.get("loss_distill").abs() < 1e-4  # same model → KL ≈ 0
+        out.get("loss_distill").backward()
+
+    def test_distill_nonzero_and_directional(self, lm, tok):
+        from rl_amd.objectives import DistillationLoss
+
+        gen = TransformersWrapper(lm, tokenizer=tok, generate=True, max_new_tokens=4)
+        h = History.from_text(["q"] * 2)
+        td = TensorDict({}, batch_size=[2])
+        td.set_non_tensor("history", {"roles": h.roles, "contents": h.contents})
+        td = gen(td)
+        hist = td.get_non_tensor("history")
+        td.set_non_tensor("history", {"roles": [r[:-1] for r in hist["roles"]], "contents": [c[:-1] for c in hist["contents"]]})
+        actor = TransformersWrapper(lm, tokenizer=tok, generate=False)
+        scored = actor(td.clone(False))
+        td.set("ref_log_probs", scored.get("log_probs").detach() - 0.5)  # shifted teacher
+        for d in ("reverse", "forward"):
+            out = DistillationLoss(actor, kl_direction=d)(td.clone(False))
+            assert torch.isfinite(out.get("loss_distill"))
+            assert out.get("loss_distill") > 0
