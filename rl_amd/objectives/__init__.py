@@ -21,5 +21,11 @@ from .multiagent import IPPOLoss, MAPPOLoss, QMixerLoss
 from .redq import CrossQLoss, REDQLoss
 from .llm import CISPOLoss, DAPO, DistillationLoss, GRPOLoss, MCAdvantage, SFTLoss
 from .dreamer import DreamerActorLoss, DreamerModelLoss, DreamerValueLoss, WorldModelLoss
+from .dreamer_v3 import (
+    DreamerV3ActorLoss,
+    DreamerV3ModelLoss,
+    DreamerV3ValueLoss,
+    categorical_kl_terms,
+)
 from .tqc import TQCLoss
 from .act import ACTLoss, DiffusionActor, DiffusionBCLoss
