@@ -21,6 +21,7 @@ from .multiagent import IPPOLoss, MAPPOLoss, QMixerLoss
 from .redq import CrossQLoss, REDQLoss
 from .llm import CISPOLoss, DAPO, DistillationLoss, GRPOLoss, MCAdvantage, SFTLoss
 from .dreamer import DreamerActorLoss, DreamerModelLoss, DreamerValueLoss, WorldModelLoss
+from .pilco import ExponentialQuadraticCost
 from .dreamer_v3 import (
     DreamerV3ActorLoss,
     DreamerV3ModelLoss,

@@ -402,3 +402,44 @@ class TestDreamerV3:
         out = loss(td)
         out.get("loss_value").backward()
         assert torch.isfinite(out.get("loss_value"))
+
+
+class TestPILCO:
+    def test_zero_variance_matches_pointwise(self):
+        from rl_amd.objectives import ExponentialQuadraticCost
+        from rl_amd.tensordict import TensorDict
+
+        m = torch.randn(16, 3)
+        S = torch.zeros(16, 3, 3)
+        td = TensorDict({"observation": {"mean": m, "var": S}}, batch_size=[16])
+        loss = ExponentialQuadraticCost(reduction="none")
+        cost = loss(td).get("loss_cost")
+        expected = 1 - torch.exp(-0.5 * m.pow(2).sum(-1))
+        assert torch.allclose(cost, expected, atol=1e-4)
+
+    def test_variance_increases_cost_at_target(self):
+        from rl_amd.objectives import ExponentialQuadraticCost
+        from rl_amd.tensordict import TensorDict
+
+        m = torch.zeros(1, 2)  # exactly on target
+        tight = TensorDict({"observation": {"mean": m, "var": torch.zeros(1, 2, 2)}}, batch_size=[1])
+        loose = TensorDict(
+            {"observation": {"mean": m, "var": 2.0 * torch.eye(2).unsqueeze(0)}},
+            batch_size=[1],
+        )
+        loss = ExponentialQuadraticCost()
+        c0 = loss(tight).get("loss_cost")
+        c1 = loss(loose).get("loss_cost")
+        assert c0 < 1e-4
+        assert c1 > c0  # uncertainty near the target raises expected cost
+
+    def test_gradients_flow(self):
+        from rl_amd.objectives import ExponentialQuadraticCost
+        from rl_amd.tensordict import TensorDict
+
+        m = torch.randn(8, 3, requires_grad=True)
+        S = 0.1 * torch.eye(3).expand(8, 3, 3)
+        td = TensorDict({"observation": {"mean": m, "var": S}}, batch_size=[8])
+        loss = ExponentialQuadraticCost()(td).get("loss_cost")
+        loss.backward()
+        assert m.grad is not None and torch.isfinite(m.grad).all()
