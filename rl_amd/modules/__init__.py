@@ -68,3 +68,12 @@ from .models import (
 from .tensordict_module import DecisionTransformerInferenceWrapper, LMHeadActorValueOperator
 from .tensordict_module import SafeModule, SafeProbabilisticModule, SafeSequential
 from .distributions import LLMMaskedCategorical
+from .models import (
+    BatchRenorm1d,
+    ConsistentDropout,
+    ConsistentDropoutModule,
+    Squeeze2dLayer,
+    SqueezeLayer,
+    SymExpTwoHot,
+)
+from . import functional

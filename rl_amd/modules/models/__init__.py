@@ -26,3 +26,11 @@ try:
     from .decision_transformer import DecisionTransformer, DTActor
 except ImportError:  # transformers not installed
     DecisionTransformer = DTActor = None
+from .extras import (
+    BatchRenorm1d,
+    ConsistentDropout,
+    ConsistentDropoutModule,
+    Squeeze2dLayer,
+    SqueezeLayer,
+    SymExpTwoHot,
+)

@@ -1,0 +1,49 @@
+"""Model-extras tests: BatchRenorm1d, SymExpTwoHot, ConsistentDropout."""
+import pytest
+import torch
+
+
+
+
+class TestModelExtras:
+    def test_batchrenorm_warmup_and_running(self):
+        from rl_amd.modules import BatchRenorm1d
+
+        brn = BatchRenorm1d(4, warmup_steps=2)
+        x = torch.randn(64, 4) * 2 + 3
+        for _ in range(5):
+            brn(x)
+        brn.eval()
+        y = brn(x)
+        # eval path normalizes with running stats learned during training
+        assert y.mean().abs() < 0.5
+        assert (y.std(0) - 1).abs().max() < 0.5
+
+    def test_symexp_twohot_head(self):
+        from rl_amd.modules import SymExpTwoHot
+        from rl_amd.modules.functional import symlog, two_hot_encode, default_bins
+
+        head = SymExpTwoHot(255)
+        for target in (0.0, 5.0, -17.0):
+            enc = two_hot_encode(symlog(torch.tensor([target])), default_bins(255))
+            # logits proportional to the two-hot distribution recover the value
+            v = head((enc + 1e-9).log() * 50)
+            assert abs(v.item() - target) < 0.3, (target, v.item())
+
+    def test_consistent_dropout_resample_on_init(self):
+        from rl_amd.modules import ConsistentDropoutModule
+        from rl_amd.tensordict import TensorDict
+
+        torch.manual_seed(3)
+        cd = ConsistentDropoutModule(0.5)
+        cd.train()
+        td = TensorDict(
+            {"observation": torch.ones(4, 32), "is_init": torch.zeros(4, 1, dtype=torch.bool)},
+            batch_size=[4],
+        )
+        a = cd(td.clone())["observation"]
+        b = cd(td.clone())["observation"]
+        assert torch.equal(a, b)
+        td.set("is_init", torch.ones(4, 1, dtype=torch.bool))
+        c = cd(td.clone())["observation"]
+        assert not torch.equal(a, c)
