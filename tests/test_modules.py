@@ -9,9 +9,9 @@ class TestModelExtras:
     def test_batchrenorm_warmup_and_running(self):
         from rl_amd.modules import BatchRenorm1d
 
-        brn = BatchRenorm1d(4, warmup_steps=2)
+        brn = BatchRenorm1d(4, warmup_steps=2, momentum=0.5)
         x = torch.randn(64, 4) * 2 + 3
-        for _ in range(5):
+        for _ in range(20):
             brn(x)
         brn.eval()
         y = brn(x)
