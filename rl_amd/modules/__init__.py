@@ -77,3 +77,11 @@ from .models import (
     SymExpTwoHot,
 )
 from . import functional
+from .models import (
+    DreamerV3BlockGRU,
+    DreamerV3MLP,
+    DreamerV3RMSNorm,
+    RSSMPosteriorV3,
+    RSSMPriorV3,
+    RSSMRolloutV3,
+)

@@ -34,3 +34,12 @@ from .extras import (
     SqueezeLayer,
     SymExpTwoHot,
 )
+from .dreamer_v3 import (
+    DreamerV3BlockGRU,
+    DreamerV3BlockLinear,
+    DreamerV3MLP,
+    DreamerV3RMSNorm,
+    RSSMPosteriorV3,
+    RSSMPriorV3,
+    RSSMRolloutV3,
+)
