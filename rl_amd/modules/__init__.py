@@ -85,3 +85,4 @@ from .models import (
     RSSMPriorV3,
     RSSMRolloutV3,
 )
+from .models import GPT2RewardModel, OnlineDTActor, RewardModel

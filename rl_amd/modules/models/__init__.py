@@ -43,3 +43,5 @@ from .dreamer_v3 import (
     RSSMPriorV3,
     RSSMRolloutV3,
 )
+from .decision_transformer import OnlineDTActor
+from .llm_models import GPT2RewardModel, RewardModel

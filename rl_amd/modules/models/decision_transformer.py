@@ -97,3 +97,27 @@ class DTActor(nn.Module):
     def forward(self, observation, action, return_to_go):
         h = self.transformer(observation, action, return_to_go)
         return torch.tanh(self.action_head(h))
+
+
+class OnlineDTActor(nn.Module):
+    """Online Decision Transformer actor (reference models.py:1507):
+    Gaussian head over the DT hidden state — returns (mu, std) for the
+    stochastic policy of Zheng et al. 2022."""
+
+    def __init__(self, state_dim: int, action_dim: int, transformer_config: Optional[dict] = None, device=None):
+        super().__init__()
+        self.transformer = DecisionTransformer(
+            state_dim, action_dim, transformer_config, device=device
+        )
+        h = self.transformer.hidden_dim
+        self.mean_head = nn.Linear(h, action_dim, device=device)
+        self.logstd_head = nn.Linear(h, action_dim, device=device)
+        # logstd bounds (paper appendix): keep std in a sane range
+        self.log_std_min = -20.0
+        self.log_std_max = 2.0
+
+    def forward(self, observation, action, return_to_go):
+        h = self.transformer(observation, action, return_to_go)
+        mu = torch.tanh(self.mean_head(h))
+        log_std = self.logstd_head(h).clamp(self.log_std_min, self.log_std_max)
+        return mu, log_std.exp()

@@ -100,3 +100,35 @@ class TestDreamerV3Models:
         gru = DreamerV3BlockGRU(12, 16, num_blocks=4)
         h = gru(torch.randn(5, 12), torch.zeros(5, 16))
         assert h.shape == (5, 16)
+
+
+class TestRewardModelAndDT:
+    def test_online_dt_actor(self):
+        from rl_amd.modules import OnlineDTActor
+
+        m = OnlineDTActor(4, 2)
+        mu, std = m(torch.randn(3, 10, 4), torch.randn(3, 10, 2), torch.randn(3, 10, 1))
+        assert mu.shape == (3, 10, 2)
+        assert (std > 0).all() and (std < 10).all()
+        (mu.sum() + std.sum()).backward()
+
+    def test_reward_model_pairwise(self):
+        transformers = pytest.importorskip("transformers")
+        from rl_amd.modules import RewardModel
+        from rl_amd.testing.llm_mocks import make_tiny_lm
+
+        torch.manual_seed(0)
+        rm = RewardModel(model=make_tiny_lm())
+        ids = torch.randint(1, 250, (4, 12))
+        mask = torch.ones_like(ids)
+        mask[:, 9:] = 0
+        rewards, end = rm(ids, mask)
+        assert rewards.shape == (4, 12) and end.shape == (4, 1)
+        # end score is the reward at the last unmasked position
+        assert torch.allclose(end.squeeze(-1), rewards[:, 8])
+        loss = RewardModel.compute_reward_loss(end[:2], end[2:])
+        loss.backward()
+        assert torch.isfinite(loss)
+        # ordering: equal scores → loss = log 2
+        l0 = RewardModel.compute_reward_loss(torch.zeros(3, 1), torch.zeros(3, 1))
+        assert l0 == pytest.approx(0.6931, abs=1e-3)
