@@ -86,3 +86,4 @@ from .models import (
     RSSMRolloutV3,
 )
 from .models import GPT2RewardModel, OnlineDTActor, RewardModel
+from .models import ExactGPRegressor, GPWorldModel, RBFController

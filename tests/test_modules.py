@@ -132,3 +132,56 @@ class TestRewardModelAndDT:
         # ordering: equal scores → loss = log 2
         l0 = RewardModel.compute_reward_loss(torch.zeros(3, 1), torch.zeros(3, 1))
         assert l0 == pytest.approx(0.6931, abs=1e-3)
+
+
+class TestPILCOModels:
+    def test_gp_regressor_fits_smooth_fn(self):
+        from rl_amd.modules import ExactGPRegressor
+
+        torch.manual_seed(0)
+        X = torch.linspace(-3, 3, 40).unsqueeze(-1)
+        y = torch.sin(X.squeeze(-1))
+        gp = ExactGPRegressor(1).fit(X, y, iters=80)
+        Xq = torch.linspace(-2.5, 2.5, 11).unsqueeze(-1)
+        mean, var = gp.predict(Xq)
+        assert (mean - torch.sin(Xq.squeeze(-1))).abs().max() < 0.1
+        assert (var > 0).all()
+        # extrapolation is more uncertain than interpolation
+        _, var_far = gp.predict(torch.tensor([[10.0]]))
+        assert var_far.item() > var.max().item()
+
+    def test_gp_world_model_on_linear_dynamics(self):
+        from rl_amd.modules import GPWorldModel
+        from rl_amd.tensordict import TensorDict
+
+        torch.manual_seed(0)
+        obs = torch.randn(60, 2)
+        act = torch.randn(60, 1)
+        nxt = obs + 0.1 * torch.cat([act, -act], -1)  # known dynamics
+        ds = TensorDict(
+            {"observation": obs, "action": act, "next": {"observation": nxt}},
+            batch_size=[60],
+        )
+        wm = GPWorldModel(2, 1).fit(ds, iters=60)
+        mean, var = wm.predict(obs[:5], act[:5])
+        assert (mean - nxt[:5]).abs().max() < 0.05
+        td = TensorDict(
+            {
+                "observation": {"mean": obs[:4], "var": torch.zeros(4, 2, 2)},
+                "action": {"mean": act[:4]},
+            },
+            batch_size=[4],
+        )
+        wm(td)
+        assert td[("next", "observation", "mean")].shape == (4, 2)
+        assert td[("next", "observation", "var")].shape == (4, 2, 2)
+
+    def test_rbf_controller_bounds(self):
+        from rl_amd.modules import RBFController
+
+        ctrl = RBFController(3, 2, max_action=0.7)
+        u = ctrl(torch.randn(16, 3) * 10)
+        assert u.shape == (16, 2)
+        assert u.abs().max() <= 0.7
+        u.sum().backward()
+        assert ctrl.weights.grad is not None
