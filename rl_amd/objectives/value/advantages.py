@@ -140,20 +140,21 @@ class TD0Estimator(ValueEstimatorBase):
     def __init__(self, *, gamma: float, value_network=None, shifted: bool = False, differentiable: bool = False, skip_existing=None, device=None, **kwargs):
         super().__init__(value_network, shifted, differentiable, skip_existing, device)
         self.register_buffer("gamma", torch.as_tensor(gamma))
+        self._gamma_float = float(gamma)
 
     def value_estimate(self, td, next_value=None, **kwargs):
         if next_value is None:
             _, next_value = self._call_value_nets(td)
         reward, done, terminated = self._get_done_terminated_reward(td)
         return F.td0_return_estimate(
-            float(self.gamma), next_value, reward, terminated, done=done
+            self._gamma_float, next_value, reward, terminated, done=done
         )
 
     def forward(self, tensordict: TensorDictBase, **kwargs) -> TensorDictBase:
         value, next_value = self._call_value_nets(tensordict)
         reward, done, terminated = self._get_done_terminated_reward(tensordict)
         target = F.td0_return_estimate(
-            float(self.gamma), next_value, reward, terminated, done=done
+            self._gamma_float, next_value, reward, terminated, done=done
         )
         adv = target - value
         tensordict.set(self.tensor_keys.advantage, adv)
@@ -167,6 +168,7 @@ class TD1Estimator(ValueEstimatorBase):
     def __init__(self, *, gamma: float, value_network=None, shifted: bool = False, differentiable: bool = False, skip_existing=None, device=None, vectorized: bool = True, **kwargs):
         super().__init__(value_network, shifted, differentiable, skip_existing, device)
         self.register_buffer("gamma", torch.as_tensor(gamma))
+        self._gamma_float = float(gamma)
         self.vectorized = vectorized
 
     def value_estimate(self, td, next_value=None, **kwargs):
@@ -174,7 +176,7 @@ class TD1Estimator(ValueEstimatorBase):
             _, next_value = self._call_value_nets(td)
         reward, done, terminated = self._get_done_terminated_reward(td)
         fn = F.vec_td1_return_estimate if self.vectorized else F.td1_return_estimate
-        return fn(float(self.gamma), next_value, reward, done, terminated)
+        return fn(self._gamma_float, next_value, reward, done, terminated)
 
     def forward(self, tensordict: TensorDictBase, **kwargs) -> TensorDictBase:
         value, _ = self._call_value_nets(tensordict)
@@ -190,7 +192,9 @@ class TDLambdaEstimator(ValueEstimatorBase):
     def __init__(self, *, gamma: float, lmbda: float = 0.95, value_network=None, shifted: bool = False, differentiable: bool = False, skip_existing=None, device=None, vectorized: bool = True, **kwargs):
         super().__init__(value_network, shifted, differentiable, skip_existing, device)
         self.register_buffer("gamma", torch.as_tensor(gamma))
+        self._gamma_float = float(gamma)
         self.register_buffer("lmbda", torch.as_tensor(lmbda))
+        self._lmbda_float = float(lmbda)
         self.vectorized = vectorized
 
     def value_estimate(self, td, next_value=None, **kwargs):
@@ -203,7 +207,7 @@ class TDLambdaEstimator(ValueEstimatorBase):
             else F.td_lambda_return_estimate
         )
         return fn(
-            float(self.gamma), float(self.lmbda), next_value, reward, done, terminated
+            self._gamma_float, self._lmbda_float, next_value, reward, done, terminated
         )
 
     def forward(self, tensordict: TensorDictBase, **kwargs) -> TensorDictBase:
@@ -236,7 +240,9 @@ class GAE(ValueEstimatorBase):
     ):
         super().__init__(value_network, shifted, differentiable, skip_existing, device)
         self.register_buffer("gamma", torch.as_tensor(gamma))
+        self._gamma_float = float(gamma)
         self.register_buffer("lmbda", torch.as_tensor(lmbda))
+        self._lmbda_float = float(lmbda)
         self.average_gae = average_gae
         self.vectorized = True if vectorized is None else vectorized
 
@@ -261,8 +267,8 @@ class GAE(ValueEstimatorBase):
             from ... import ops
 
             adv, value_target = ops.gae(
-                float(self.gamma),
-                float(self.lmbda),
+                self._gamma_float,
+                self._lmbda_float,
                 value,
                 next_value,
                 reward,
@@ -276,8 +282,8 @@ class GAE(ValueEstimatorBase):
                 else F.generalized_advantage_estimate
             )
             adv, value_target = fn(
-                float(self.gamma),
-                float(self.lmbda),
+                self._gamma_float,
+                self._lmbda_float,
                 value,
                 next_value,
                 reward,
@@ -312,6 +318,7 @@ class VTrace(ValueEstimatorBase):
     ):
         super().__init__(value_network, shifted, differentiable, None, device)
         self.register_buffer("gamma", torch.as_tensor(gamma))
+        self._gamma_float = float(gamma)
         self.rho_thresh = rho_thresh
         self.c_thresh = c_thresh
         self.actor_network = actor_network
@@ -337,7 +344,7 @@ class VTrace(ValueEstimatorBase):
             else F.vtrace_advantage_estimate
         )
         adv, value_target = fn(
-            float(self.gamma),
+            self._gamma_float,
             log_pi,
             log_mu,
             value,
