@@ -46,3 +46,4 @@ from .dreamer_v3 import (
 from .decision_transformer import OnlineDTActor
 from .llm_models import GPT2RewardModel, RewardModel
 from .gp import ExactGPRegressor, GPWorldModel, RBFController
+from .act import ACTModel

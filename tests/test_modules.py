@@ -185,3 +185,45 @@ class TestPILCOModels:
         assert u.abs().max() <= 0.7
         u.sum().backward()
         assert ctrl.weights.grad is not None
+
+
+class TestACTModel:
+    def test_training_and_inference_modes(self):
+        from rl_amd.modules import ACTModel
+
+        torch.manual_seed(0)
+        m = ACTModel(obs_dim=6, action_dim=3, chunk_size=5, hidden_dim=32,
+                     nheads=4, num_encoder_layers=1, num_decoder_layers=1,
+                     latent_dim=8, dim_feedforward=64)
+        obs = torch.randn(4, 6)
+        chunk = torch.randn(4, 5, 3)
+        pred, mu, logvar = m(obs, chunk)
+        assert pred.shape == (4, 5, 3) and mu.shape == (4, 8)
+        (pred.sum() + mu.sum() + logvar.sum()).backward()
+        # inference: no chunk → prior-mean latent, zero mu/logvar
+        pred2, mu2, lv2 = m(obs)
+        assert (mu2 == 0).all() and (lv2 == 0).all()
+        assert pred2.shape == (4, 5, 3)
+
+    def test_act_loss_integration(self):
+        from rl_amd.modules import ACTModel
+        from rl_amd.objectives import ACTLoss
+        from rl_amd.tensordict import TensorDict, TensorDictModule
+
+        torch.manual_seed(0)
+        m = ACTModel(obs_dim=4, action_dim=2, chunk_size=3, hidden_dim=16,
+                     nheads=2, num_encoder_layers=1, num_decoder_layers=1,
+                     latent_dim=4, dim_feedforward=32)
+        actor = TensorDictModule(
+            m,
+            in_keys=["observation", "action"],
+            out_keys=["action_pred", "latent_mu", "latent_logvar"],
+        )
+        loss = ACTLoss(actor, kl_weight=1.0)
+        td = TensorDict(
+            {"observation": torch.randn(4, 4), "action": torch.randn(4, 3, 2)},
+            batch_size=[4],
+        )
+        out = loss(td)
+        out.get("loss").backward()
+        assert torch.isfinite(out.get("loss"))
