@@ -105,11 +105,15 @@ def main():
     # overhead (hundreds of ~5us kernels); a single graph replay removes
     # it.  PER sample / priority write stay eager (buffer length grows).
     static_batch = None
+    static_keys = None
     graph = None
 
     def capture_update():
-        nonlocal static_batch, graph
+        nonlocal static_batch, static_keys, graph
         static_batch = rb.sample().clone()
+        # the loss writes td_error into the static buffer during capture;
+        # replays only refresh the keys a fresh sample actually carries
+        static_keys = list(static_batch.keys(True, True))
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -125,7 +129,7 @@ def main():
         for _ in range(args.utd):
             batch = rb.sample()
             if graph is not None:
-                for k in list(static_batch.keys(True, True)):
+                for k in static_keys:
                     static_batch.get(k).copy_(batch.get(k))
                 graph.replay()
                 rb.update_tensordict_priority(static_batch)
