@@ -16,8 +16,11 @@ from ..data import LazyTensorStorage, TensorDictPrioritizedReplayBuffer, TensorD
 from ..objectives import (
     A2CLoss,
     ClipPPOLoss,
+    CQLLoss,
     DDPGLoss,
     DQNLoss,
+    IQLLoss,
+    ReinforceLoss,
     SACLoss,
     SoftUpdate,
     TD3Loss,
@@ -252,3 +255,110 @@ class DQNTrainer(OffPolicyTrainer):
             logger=logger,
             **kwargs,
         )
+
+
+class ReinforceTrainer(OnPolicyTrainer):
+    """(reference algorithms/reinforce.py) — vanilla policy gradient with
+    a value baseline."""
+
+    def __init__(self, *, actor, critic, collector, total_frames, lr: float = 1e-3,
+                 gamma: float = 0.99, lmbda: float = 0.95, logger=None, **kwargs):
+        loss = ReinforceLoss(actor, critic)
+        gae = GAE(gamma=gamma, lmbda=lmbda, value_network=critic)
+        optim = torch.optim.Adam(loss.parameters(), lr=lr)
+        super().__init__(
+            collector=collector,
+            loss_module=loss,
+            optimizer=optim,
+            total_frames=total_frames,
+            value_estimator=gae,
+            num_epochs=1,
+            logger=logger,
+            **kwargs,
+        )
+
+
+class IQLTrainer(OffPolicyTrainer):
+    """(reference algorithms/iql.py) — implicit Q-learning."""
+
+    def __init__(self, *, actor, qvalue, value, collector, total_frames,
+                 lr: float = 3e-4, tau: float = 0.005, expectile: float = 0.7,
+                 temperature: float = 3.0, logger=None, **kwargs):
+        loss = IQLLoss(actor, qvalue, value, expectile=expectile, temperature=temperature)
+        loss.make_value_estimator()
+        optim = torch.optim.Adam(loss.parameters(), lr=lr)
+        updater = SoftUpdate(loss, tau=tau)
+        super().__init__(
+            collector=collector,
+            loss_module=loss,
+            optimizer=optim,
+            total_frames=total_frames,
+            target_updater=updater,
+            logger=logger,
+            **kwargs,
+        )
+
+
+class CQLTrainer(OffPolicyTrainer):
+    """(reference algorithms/cql.py) — conservative Q-learning."""
+
+    def __init__(self, *, actor, qvalue, collector, total_frames, lr: float = 3e-4,
+                 tau: float = 0.005, logger=None, **kwargs):
+        loss = CQLLoss(actor, qvalue)
+        loss.make_value_estimator()
+        optim = torch.optim.Adam(loss.parameters(), lr=lr)
+        updater = SoftUpdate(loss, tau=tau)
+        super().__init__(
+            collector=collector,
+            loss_module=loss,
+            optimizer=optim,
+            total_frames=total_frames,
+            target_updater=updater,
+            logger=logger,
+            **kwargs,
+        )
+
+
+class OfflineToOnlineTrainer(OffPolicyTrainer):
+    """(reference algorithms/offline_to_online.py) — pretrain on an
+    offline dataset, then continue online with the mixed
+    :class:`~rl_amd.data.offline_to_online.OfflineOnlineReplayBuffer`."""
+
+    def __init__(self, *, loss_module, collector, total_frames, offline_buffer,
+                 lr: float = 3e-4, offline_steps: int = 1000,
+                 offline_fraction: float = 0.5, batch_size: int = 256,
+                 buffer_size: int = 1_000_000,
+                 tau: float = 0.005, device=None, logger=None, **kwargs):
+        from ..data.offline_to_online import OfflineOnlineReplayBuffer
+
+        optim = torch.optim.Adam(loss_module.parameters(), lr=lr)
+        updater = SoftUpdate(loss_module, tau=tau)
+        online = TensorDictReplayBuffer(
+            storage=LazyTensorStorage(buffer_size, device=device),
+            batch_size=batch_size,
+        )
+        buf = OfflineOnlineReplayBuffer(
+            offline_buffer,
+            online,
+            offline_fraction=offline_fraction,
+            batch_size=batch_size,
+        )
+        super().__init__(
+            collector=collector,
+            loss_module=loss_module,
+            optimizer=optim,
+            total_frames=total_frames,
+            replay_buffer=buf,
+            batch_size=batch_size,
+            target_updater=updater,
+            logger=logger,
+            **kwargs,
+        )
+        self.offline_steps = offline_steps
+
+    def pretrain(self):
+        """Run ``offline_steps`` optimisation passes on offline data only
+        before any online collection."""
+        for _ in range(self.offline_steps):
+            batch = self.replay_buffer.offline.sample()
+            self.optim_steps(batch)
