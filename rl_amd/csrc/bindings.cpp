@@ -38,8 +38,8 @@ void launch_fused_actor(const float*, const float*, const float*, const float*,
                         float*, float*, float*, float*, int, int, int, int,
                         int, float, float, void*);
 int lstm_fused_lds_bytes(int);
-void launch_wgrad_splitk(const void*, const void*, float*, float*, long,
-                         int, int, void*);
+void launch_wgrad_splitk(const void*, const void*, float*, float*, float*,
+                         float*, long, int, int, void*);
 void launch_gru_fused(const float*, const float*, const float*, const bool*,
                       const float*, float*, float*, int, int, int, void*);
 void launch_lstm_fused(const float*, const float*, const bool*, const float*,
@@ -235,13 +235,18 @@ std::vector<torch::Tensor> wgrad_splitk(torch::Tensor dy, torch::Tensor x,
   long K = dy.size(0);
   int N = (int)dy.size(1), M = (int)x.size(1);
   auto opts = dy.options().dtype(torch::kFloat32);
-  auto dw = torch::zeros({N, M}, opts);
+  auto dw = torch::empty({N, M}, opts);
   torch::Tensor db;
-  if (want_bias) db = torch::zeros({N}, opts);
+  if (want_bias) db = torch::empty({N}, opts);
+  const long slabs = (K + 63) / 64;  // must match K_SLAB in wgrad.hip
+  const long tiles_n = (N + 63) / 64, tiles_m = (M + 63) / 64;
+  auto part = torch::empty({slabs, tiles_n * tiles_m * 64 * 64}, opts);
+  auto bias_part = torch::empty({slabs, tiles_n * 64}, opts);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_wgrad_splitk(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(),
-                      want_bias ? db.data_ptr<float>() : nullptr, K, N, M,
-                      (void*)stream);
+                      want_bias ? db.data_ptr<float>() : nullptr,
+                      part.data_ptr<float>(), bias_part.data_ptr<float>(), K,
+                      N, M, (void*)stream);
   if (want_bias) return {dw, db};
   return {dw};
 }

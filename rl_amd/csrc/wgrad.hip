@@ -14,9 +14,11 @@
 //  * each workgroup stages 64-row chunks of X and dY in LDS
 //    (coalesced ushort2 loads), then each of the 256 threads owns a
 //    4x4 output tile: 16 VALU FMAs per k-row from LDS broadcasts.
-//  * fp32 accumulation in registers; one atomicAdd per output element
-//    per workgroup into an fp32 workspace (contention: grid.x writers
-//    over 4096 addresses — negligible).
+//  * fp32 accumulation in registers; each workgroup writes its partial
+//    tile to a [slabs, N, M] buffer (NO atomics — measured: atomicAdd
+//    with 256 contending workgroups per address serialized at ~49 us;
+//    partials + a reduce kernel run in a fraction of that), then a
+//    second kernel reduces over the slab axis.
 //  * dBias[n] = sum_k dY[k][n] is fused (threads with tm==0).
 //
 // No MFMA: at these shapes the kernel is bound by the 2*(M+N)*B byte
@@ -99,18 +101,49 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_splitk_kernel(
     __syncthreads();
   }
 
-  // commit partials
+  // write this workgroup's partial tile (no atomics)
+  const long num_tiles_n = gridDim.y;
+  const long num_tiles_m = gridDim.z;
+  const long tile_elems = (long)64 * 64 * num_tiles_n * num_tiles_m;
+  float* part = dw + (long)blockIdx.x * tile_elems +
+                ((long)blockIdx.y * num_tiles_m + blockIdx.z) * 64 * 64;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
-    const int n = n_base + i;
-    if (n >= N) continue;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      const int m = m_base + j;
-      if (m < M) atomicAdd(&dw[(long)n * M + m], acc[i][j]);
+      part[(tn * 4 + i) * 64 + tm * 4 + j] = acc[i][j];
     }
     if (dbias != nullptr && tm == 0 && m0 == 0) {
-      atomicAdd(&dbias[n], bias_acc[i]);
+      dbias[(long)blockIdx.x * (num_tiles_n * 64) + n_base + i] = bias_acc[i];
+    }
+  }
+}
+
+// reduce the [slabs, tiles_n, tiles_m, 64, 64] partials into dW [N, M]
+// (and [slabs, tiles_n*64] bias partials into dBias [N])
+__global__ void wgrad_reduce_kernel(const float* __restrict__ part,
+                                    const float* __restrict__ bias_part,
+                                    float* __restrict__ dw,
+                                    float* __restrict__ dbias, int slabs,
+                                    int tiles_n, int tiles_m, int N, int M) {
+  const long tile_elems = (long)64 * 64 * tiles_n * tiles_m;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x;
+       idx < (long)N * M; idx += (long)gridDim.x * blockDim.x) {
+    const int n = (int)(idx / M), m = (int)(idx % M);
+    const int tn = n / 64, tm = m / 64;
+    const long off = ((long)tn * tiles_m + tm) * 64 * 64 +
+                     (long)(n % 64) * 64 + (m % 64);
+    float acc = 0.f;
+    for (int s = 0; s < slabs; ++s) acc += part[(long)s * tile_elems + off];
+    dw[idx] = acc;
+  }
+  if (dbias != nullptr) {
+    for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+         n += gridDim.x * blockDim.x) {
+      float acc = 0.f;
+      for (int s = 0; s < slabs; ++s)
+        acc += bias_part[(long)s * (tiles_n * 64) + n];
+      dbias[n] = acc;
     }
   }
 }
@@ -118,11 +151,18 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_splitk_kernel(
 }  // namespace
 
 extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
-                         float* dbias, long K, int N, int M, void* stream) {
+                         float* dbias, float* part, float* bias_part, long K,
+                         int N, int M, void* stream) {
   const int slabs = (int)((K + K_SLAB - 1) / K_SLAB);
-  dim3 grid(slabs, (N + 63) / 64, (M + 63) / 64);
+  const int tiles_n = (N + 63) / 64, tiles_m = (M + 63) / 64;
+  dim3 grid(slabs, tiles_n, tiles_m);
   hipLaunchKernelGGL(wgrad_splitk_kernel, grid, dim3(WG_THREADS), 0,
                      (hipStream_t)stream,
-                     (const __hip_bfloat16*)dy, (const __hip_bfloat16*)x, dw,
-                     dbias, K, N, M);
+                     (const __hip_bfloat16*)dy, (const __hip_bfloat16*)x, part,
+                     bias_part, K, N, M);
+  const long total = (long)N * M;
+  const int blocks = (int)min((total + 255) / 256, (long)1024);
+  hipLaunchKernelGGL(wgrad_reduce_kernel, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, part, bias_part, dw, dbias, slabs,
+                     tiles_n, tiles_m, N, M);
 }
