@@ -21,3 +21,4 @@ from .async_batched import AsyncBatchedCollector
 from .profiling import ProfileConfig, ProfilerHook, enable_profile
 from .graph import GraphedRollout
 from .rpc import RPCCollector
+from .ray import RayCollector, RayLLMCollector
