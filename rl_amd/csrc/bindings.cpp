@@ -38,6 +38,7 @@ void launch_fused_actor(const float*, const float*, const float*, const float*,
                         float*, float*, float*, float*, int, int, int, int,
                         int, float, float, void*);
 int lstm_fused_lds_bytes(int);
+int wgrad_slab_count(long);
 void launch_wgrad_splitk(const void*, const void*, float*, float*, float*,
                          float*, long, int, int, void*);
 void launch_gru_fused(const float*, const float*, const float*, const bool*,
@@ -238,7 +239,7 @@ std::vector<torch::Tensor> wgrad_splitk(torch::Tensor dy, torch::Tensor x,
   auto dw = torch::empty({N, M}, opts);
   torch::Tensor db;
   if (want_bias) db = torch::empty({N}, opts);
-  const long slabs = (K + 63) / 64;  // must match K_SLAB in wgrad.hip
+  const long slabs = wgrad_slab_count(K);
   const long tiles_n = (N + 63) / 64, tiles_m = (M + 63) / 64;
   auto part = torch::empty({slabs, tiles_n * tiles_m * 64 * 64}, opts);
   auto bias_part = torch::empty({slabs, tiles_n * 64}, opts);

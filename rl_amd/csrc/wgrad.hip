@@ -33,21 +33,22 @@
 
 #define WG_THREADS 256
 #define ROW_CHUNK 64   // k-rows staged in LDS per iteration
-#define K_SLAB 64      // k-rows owned by one workgroup (K=16384 -> 256 WGs; measured 38us at 128)
+// K_SLAB is chosen at launch: ~128 slabs regardless of K, so the reduce
+// stays cheap and the partials buffer bounded.
 
 namespace {
 
 __global__ void __launch_bounds__(WG_THREADS) wgrad_splitk_kernel(
     const __hip_bfloat16* __restrict__ dy,  // [K, N] row-major
     const __hip_bfloat16* __restrict__ x,   // [K, M] row-major
-    float* __restrict__ dw,                 // [N, M] fp32 accum workspace
-    float* __restrict__ dbias,              // [N] fp32 accum (or nullptr)
-    long K, int N, int M) {
+    float* __restrict__ dw,                 // partials, [tile, elem, slab]
+    float* __restrict__ dbias,              // bias partials [n, slab]
+    long K, int N, int M, int k_slab) {
   // output tile assigned to this workgroup
   const int n0 = blockIdx.y * 64;
   const int m0 = blockIdx.z * 64;
-  const long k_begin = (long)blockIdx.x * K_SLAB;
-  const long k_end = min(K, k_begin + (long)K_SLAB);
+  const long k_begin = (long)blockIdx.x * k_slab;
+  const long k_end = min(K, k_begin + (long)k_slab);
 
   // thread's 4x4 sub-tile
   const int tn = threadIdx.x / 16;  // 0..15
@@ -101,48 +102,50 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_splitk_kernel(
     __syncthreads();
   }
 
-  // write this workgroup's partial tile (no atomics)
-  const long num_tiles_n = gridDim.y;
+  // write this workgroup's partial tile, slab-axis INNERMOST so the
+  // reduce kernel streams contiguous runs of `slabs` floats per element
+  const long slabs = gridDim.x;
   const long num_tiles_m = gridDim.z;
-  const long tile_elems = (long)64 * 64 * num_tiles_n * num_tiles_m;
-  float* part = dw + (long)blockIdx.x * tile_elems +
-                ((long)blockIdx.y * num_tiles_m + blockIdx.z) * 64 * 64;
+  float* part = dw + ((long)blockIdx.y * num_tiles_m + blockIdx.z) *
+                         (64 * 64 * slabs);
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      part[(tn * 4 + i) * 64 + tm * 4 + j] = acc[i][j];
+      part[((long)(tn * 4 + i) * 64 + tm * 4 + j) * slabs + blockIdx.x] =
+          acc[i][j];
     }
     if (dbias != nullptr && tm == 0 && m0 == 0) {
-      dbias[(long)blockIdx.x * (num_tiles_n * 64) + n_base + i] = bias_acc[i];
+      dbias[(long)(n_base + i) * slabs + blockIdx.x] = bias_acc[i];
     }
   }
 }
 
-// reduce the [slabs, tiles_n, tiles_m, 64, 64] partials into dW [N, M]
-// (and [slabs, tiles_n*64] bias partials into dBias [N])
+// reduce the [tile, elem, slab] partials into dW [N, M] (and the
+// [n, slab] bias partials into dBias [N]); each element's slab run is
+// contiguous, so the inner loop streams.
 __global__ void wgrad_reduce_kernel(const float* __restrict__ part,
                                     const float* __restrict__ bias_part,
                                     float* __restrict__ dw,
                                     float* __restrict__ dbias, int slabs,
                                     int tiles_n, int tiles_m, int N, int M) {
-  const long tile_elems = (long)64 * 64 * tiles_n * tiles_m;
   for (long idx = blockIdx.x * blockDim.x + threadIdx.x;
        idx < (long)N * M; idx += (long)gridDim.x * blockDim.x) {
     const int n = (int)(idx / M), m = (int)(idx % M);
     const int tn = n / 64, tm = m / 64;
-    const long off = ((long)tn * tiles_m + tm) * 64 * 64 +
-                     (long)(n % 64) * 64 + (m % 64);
+    const float* p = part +
+                     ((long)tn * tiles_m + tm) * (64 * 64 * (long)slabs) +
+                     ((long)(n % 64) * 64 + (m % 64)) * slabs;
     float acc = 0.f;
-    for (int s = 0; s < slabs; ++s) acc += part[(long)s * tile_elems + off];
+    for (int s = 0; s < slabs; ++s) acc += p[s];
     dw[idx] = acc;
   }
   if (dbias != nullptr) {
     for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
          n += gridDim.x * blockDim.x) {
+      const float* p = bias_part + (long)n * slabs;
       float acc = 0.f;
-      for (int s = 0; s < slabs; ++s)
-        acc += bias_part[(long)s * (tiles_n * 64) + n];
+      for (int s = 0; s < slabs; ++s) acc += p[s];
       dbias[n] = acc;
     }
   }
@@ -150,16 +153,27 @@ __global__ void wgrad_reduce_kernel(const float* __restrict__ part,
 
 }  // namespace
 
+extern "C" int wgrad_slab_count(long K) {
+  // target ~128 slabs; slab size a multiple of ROW_CHUNK
+  long k_slab = (K + 127) / 128;
+  k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
+  if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
+  return (int)((K + k_slab - 1) / k_slab);
+}
+
 extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
                          float* dbias, float* part, float* bias_part, long K,
                          int N, int M, void* stream) {
-  const int slabs = (int)((K + K_SLAB - 1) / K_SLAB);
+  long k_slab = (K + 127) / 128;
+  k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
+  if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
+  const int slabs = (int)((K + k_slab - 1) / k_slab);
   const int tiles_n = (N + 63) / 64, tiles_m = (M + 63) / 64;
   dim3 grid(slabs, tiles_n, tiles_m);
   hipLaunchKernelGGL(wgrad_splitk_kernel, grid, dim3(WG_THREADS), 0,
                      (hipStream_t)stream,
                      (const __hip_bfloat16*)dy, (const __hip_bfloat16*)x, part,
-                     bias_part, K, N, M);
+                     bias_part, K, N, M, (int)k_slab);
   const long total = (long)N * M;
   const int blocks = (int)min((total + 255) / 256, (long)1024);
   hipLaunchKernelGGL(wgrad_reduce_kernel, dim3(blocks), dim3(256), 0,
