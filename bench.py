@@ -144,11 +144,23 @@ def main():
             args.splitk = False
 
     params = list(actor.parameters()) + list(critic.parameters())
-    optim = torch.optim.Adam(
-        params,
-        lr=3e-4,
-        capturable=bool(args.full_graph and cuda and world == 1),
-    )
+    # fused Adam: one multi-tensor kernel instead of ~10 elementwise
+    # launches per step (the r15 profile's long elementwise tail)
+    optim_kwargs = dict(lr=3e-4, capturable=bool(args.full_graph and cuda and world == 1))
+
+    def _fused_adam_ok():
+        try:
+            p = torch.zeros(4, device=device, requires_grad=True)
+            p.grad = torch.zeros_like(p)
+            torch.optim.Adam([p], fused=True, **optim_kwargs).step()
+            return True
+        except Exception:
+            return False
+
+    if cuda and _fused_adam_ok():
+        optim = torch.optim.Adam(params, fused=True, **optim_kwargs)
+    else:
+        optim = torch.optim.Adam(params, **optim_kwargs)
     loss_mod = ClipPPOLoss(actor, critic, clip_epsilon=0.2, entropy_coeff=0.01,
                            critic_coeff=0.5, normalize_advantage=True)
     gae = GAE(gamma=0.99, lmbda=0.95, value_network=critic, vectorized=True)
