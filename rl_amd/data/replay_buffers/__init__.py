@@ -32,6 +32,14 @@ from .writers import (
     TensorDictRoundRobinWriter,
     Writer,
 )
+from .checkpointers import (
+    FlatStorageCheckpointer,
+    H5StorageCheckpointer,
+    ListStorageCheckpointer,
+    NestedStorageCheckpointer,
+    StorageCheckpointerBase,
+    TensorStorageCheckpointer,
+)
 from .extras import (
     CompressedListStorage,
     ConsumingSampler,
@@ -40,4 +48,6 @@ from .extras import (
     PromptGroupSampler,
     StalenessAwareSampler,
     StepScheduler,
+    LambdaScheduler,
+    SchedulerList,
 )

@@ -25,6 +25,8 @@ __all__ = [
     "StalenessAwareSampler",
     "PromptGroupSampler",
     "CompressedListStorage",
+    "LambdaScheduler",
+    "SchedulerList",
 ]
 
 
@@ -200,3 +202,26 @@ class CompressedListStorage(ListStorage):
         if out and isinstance(out[0], TensorDictBase):
             return td_stack(out, 0)
         return out
+
+
+def LambdaScheduler(obj, param_name: str, fn: Callable[[int], float]) -> ParameterScheduler:
+    """Arbitrary-function schedule (reference scheduler.py LambdaScheduler)."""
+    return ParameterScheduler(obj, param_name, fn)
+
+
+class SchedulerList:
+    """Step several parameter schedulers together (reference
+    scheduler.py SchedulerList)."""
+
+    def __init__(self, schedulers):
+        self.schedulers = list(schedulers)
+
+    def step(self, n: int = 1):
+        return [s.step(n) for s in self.schedulers]
+
+    def state_dict(self):
+        return [s.state_dict() for s in self.schedulers]
+
+    def load_state_dict(self, sds):
+        for s, sd in zip(self.schedulers, sds):
+            s.load_state_dict(sd)

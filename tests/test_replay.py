@@ -241,3 +241,73 @@ class TestCheckpointing:
         assert len(rb2) == 10
         s = rb2.sample()
         assert s["x"].shape == (4, 2)
+
+
+class TestCheckpointers:
+    def _ted_buffer(self, n=20):
+        from rl_amd.data import LazyTensorStorage, TensorDictReplayBuffer
+        from rl_amd.tensordict import TensorDict
+
+        torch.manual_seed(0)
+        rb = TensorDictReplayBuffer(storage=LazyTensorStorage(50), batch_size=4)
+        obs = torch.randn(n, 3)
+        nxt_obs = torch.randn(n, 3)
+        done = torch.zeros(n, 1, dtype=torch.bool)
+        done[n // 2] = True
+        done[n - 1] = True
+        # TED consistency: next_obs[i] == obs[i+1] inside trajectories
+        nxt_obs[:-1] = obs[1:]
+        rb.extend(TensorDict(
+            {"obs": obs, "next": {"obs": nxt_obs, "done": done}}, batch_size=[n]
+        ))
+        return rb, obs, nxt_obs
+
+    def test_tensor_storage_checkpointer(self, tmp_path):
+        from rl_amd.data.replay_buffers import TensorStorageCheckpointer
+
+        rb, obs, nxt = self._ted_buffer()
+        cp = TensorStorageCheckpointer()
+        cp.dumps(rb._storage, str(tmp_path))
+        rb2, _, _ = self._ted_buffer(5)
+        cp.loads(rb2._storage, str(tmp_path))
+        assert len(rb2._storage) == 20
+        assert torch.equal(rb2._storage._storage["obs"][:20], obs)
+
+    def test_flat_checkpointer_roundtrip(self, tmp_path):
+        from rl_amd.data.replay_buffers import FlatStorageCheckpointer
+
+        rb, obs, nxt = self._ted_buffer()
+        cp = FlatStorageCheckpointer()
+        cp.dumps(rb._storage, str(tmp_path))
+        # the dedupe must shrink what's stored for duplicated keys
+        sd = torch.load(str(tmp_path) + "/flat_storage.pt", weights_only=False)
+        assert ("next", "obs") not in sd["compact"].keys(True, True)
+        rb2, _, _ = self._ted_buffer(5)
+        cp.loads(rb2._storage, str(tmp_path))
+        rebuilt = rb2._storage._storage
+        assert torch.equal(rebuilt["obs"][:20], obs)
+        assert torch.allclose(rebuilt["next", "obs"][:20], nxt)
+
+    def test_h5_gated(self):
+        import importlib.util
+
+        from rl_amd.data.replay_buffers import H5StorageCheckpointer
+
+        if importlib.util.find_spec("h5py") is None:
+            with pytest.raises(ImportError, match="h5py"):
+                H5StorageCheckpointer()
+
+    def test_scheduler_list(self):
+        from rl_amd.data.replay_buffers import LambdaScheduler, SchedulerList
+
+        class Obj:
+            a = 1.0
+            b = 2.0
+
+        o = Obj()
+        sl = SchedulerList([
+            LambdaScheduler(o, "a", lambda t: 1.0 / (1 + t)),
+            LambdaScheduler(o, "b", lambda t: 2.0 + t),
+        ])
+        sl.step()
+        assert o.a == 0.5 and o.b == 3.0
