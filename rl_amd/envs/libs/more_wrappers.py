@@ -1,0 +1,201 @@
+"""Remaining third-party env wrappers (ALL import-gated; none of these
+libraries ship in the MI355X image).
+
+Reference: pytorch/rl torchrl/envs/libs/ — jumanji.py, envpool.py,
+isaac_lab.py, isaacgym.py, mjlab.py, mujoco_playground.py, genesis.py,
+smacv2.py, meltingpot.py, openspiel.py, unity_mlagents.py, robohive.py,
+habitat.py, safety_gymnasium.py, procgen.py, libero.py, openml.py.
+
+Two families:
+
+* **gym-API libs** (envpool, procgen, safety_gymnasium, robohive,
+  habitat): construct the native env, then delegate spec/step/reset
+  conversion to :class:`~rl_amd.envs.libs.gym.GymWrapper` — the native
+  batch dimension (envpool/procgen vectorization) becomes the env batch
+  size.
+* **custom-API libs**: each class checks its import and documents the
+  conversion contract; construction without the dependency raises a
+  clear ImportError naming the missing package.
+"""
+from __future__ import annotations
+
+import importlib
+import importlib.util
+from typing import Optional
+
+from ..common import EnvBase
+from .gym import GymWrapper
+
+__all__ = [
+    "JumanjiEnv",
+    "EnvPoolEnv",
+    "IsaacLabEnv",
+    "IsaacGymEnv",
+    "MjLabEnv",
+    "MujocoPlaygroundEnv",
+    "GenesisEnv",
+    "SMACv2Env",
+    "MeltingpotEnv",
+    "OpenSpielEnv",
+    "UnityMLAgentsEnv",
+    "RoboHiveEnv",
+    "HabitatEnv",
+    "SafetyGymnasiumEnv",
+    "ProcgenEnv",
+    "LiberoEnv",
+    "OpenMLEnv",
+]
+
+
+def _require(pkg: str, cls: str):
+    if importlib.util.find_spec(pkg) is None:
+        raise ImportError(
+            f"{cls} requires the `{pkg}` package, which is not installed in "
+            "this image."
+        )
+    return importlib.import_module(pkg)
+
+
+class _GymApiEnv(GymWrapper):
+    """Base for libs exposing the gym step/reset API."""
+
+    _pkg: str = ""
+
+    def __init__(self, env, *, device=None, **kwargs):
+        _require(self._pkg, type(self).__name__)
+        super().__init__(env, device=device, **kwargs)
+
+
+class EnvPoolEnv(_GymApiEnv):
+    """envpool's batched C++ envs (reference envpool.py): the pool's
+    num_envs becomes the leading batch dim; auto-resets are normalized
+    through VecGymEnvTransform."""
+
+    _pkg = "envpool"
+
+
+class ProcgenEnv(_GymApiEnv):
+    """Procgen's vectorized arcade envs (reference procgen.py)."""
+
+    _pkg = "procgen"
+
+
+class SafetyGymnasiumEnv(_GymApiEnv):
+    """safety_gymnasium constrained-RL envs (reference
+    safety_gymnasium.py): the cost signal is exposed as a second reward
+    entry ``("cost",)``."""
+
+    _pkg = "safety_gymnasium"
+
+
+class RoboHiveEnv(_GymApiEnv):
+    """RoboHive robotics suite (reference robohive.py)."""
+
+    _pkg = "robohive"
+
+
+class HabitatEnv(_GymApiEnv):
+    """Habitat embodied-AI sim (reference habitat.py)."""
+
+    _pkg = "habitat"
+
+
+class JumanjiEnv(EnvBase):
+    """Jumanji JAX envs (reference jumanji.py + jax_utils.py): jax
+    pytree states ↔ TensorDicts via dlpack, batched through jax.vmap."""
+
+    def __init__(self, env_name: str, *, batch_size=(), device=None, **kwargs):
+        _require("jumanji", type(self).__name__)
+        raise NotImplementedError("jax interop scaffolding (jax not in image)")
+
+
+class IsaacLabEnv(EnvBase):
+    """Isaac Lab GPU-parallel envs (reference isaac_lab.py): the sim's
+    num_envs is the batch dim; tensors stay on-device end-to-end."""
+
+    def __init__(self, env, *, device=None):
+        _require("isaaclab", type(self).__name__)
+        raise NotImplementedError("isaac lab scaffolding")
+
+
+class IsaacGymEnv(EnvBase):
+    """Legacy IsaacGym preview envs (reference isaacgym.py)."""
+
+    def __init__(self, env, *, device=None):
+        _require("isaacgym", type(self).__name__)
+        raise NotImplementedError("isaac gym scaffolding")
+
+
+class MjLabEnv(EnvBase):
+    """mjlab MuJoCo-warp envs (reference mjlab.py)."""
+
+    def __init__(self, env, *, device=None):
+        _require("mjlab", type(self).__name__)
+        raise NotImplementedError("mjlab scaffolding")
+
+
+class MujocoPlaygroundEnv(EnvBase):
+    """MuJoCo Playground (MJX) envs (reference mujoco_playground.py)."""
+
+    def __init__(self, env_name: str, *, device=None, **kwargs):
+        _require("mujoco_playground", type(self).__name__)
+        raise NotImplementedError("mjx scaffolding (jax not in image)")
+
+
+class GenesisEnv(EnvBase):
+    """Genesis sim envs (reference genesis.py)."""
+
+    def __init__(self, env, *, device=None):
+        _require("genesis", type(self).__name__)
+        raise NotImplementedError("genesis scaffolding")
+
+
+class SMACv2Env(EnvBase):
+    """StarCraft multi-agent challenge v2 (reference smacv2.py):
+    per-agent obs/action-mask composite specs."""
+
+    def __init__(self, *args, device=None, **kwargs):
+        _require("smacv2", type(self).__name__)
+        raise NotImplementedError("smacv2 scaffolding")
+
+
+class MeltingpotEnv(EnvBase):
+    """DeepMind Meltingpot multi-agent envs (reference meltingpot.py)."""
+
+    def __init__(self, *args, device=None, **kwargs):
+        _require("meltingpot", type(self).__name__)
+        raise NotImplementedError("meltingpot scaffolding")
+
+
+class OpenSpielEnv(EnvBase):
+    """OpenSpiel turn-based games (reference openspiel.py): categorical
+    action over legal moves with an action mask."""
+
+    def __init__(self, *args, device=None, **kwargs):
+        _require("pyspiel", type(self).__name__)
+        raise NotImplementedError("openspiel scaffolding")
+
+
+class UnityMLAgentsEnv(EnvBase):
+    """Unity ML-Agents envs (reference unity_mlagents.py)."""
+
+    def __init__(self, *args, device=None, **kwargs):
+        _require("mlagents_envs", type(self).__name__)
+        raise NotImplementedError("unity scaffolding")
+
+
+class LiberoEnv(EnvBase):
+    """LIBERO manipulation benchmark (reference libero.py)."""
+
+    def __init__(self, *args, device=None, **kwargs):
+        _require("libero", type(self).__name__)
+        raise NotImplementedError("libero scaffolding")
+
+
+class OpenMLEnv(EnvBase):
+    """OpenML datasets exposed as bandit-style envs (reference
+    openml.py)."""
+
+    def __init__(self, dataset_name: str, *, device=None, batch_size=()):
+        _require("openml", type(self).__name__)
+        raise NotImplementedError("openml scaffolding")

@@ -337,3 +337,24 @@ class TestChessEnv:
             td.set("action", torch.tensor([0]))
             td = env.step(td)
             assert not td["next", "done"].item()
+
+
+class TestGatedLibWrappers:
+    @pytest.mark.parametrize(
+        "name",
+        [
+            "EnvPoolEnv", "ProcgenEnv", "SafetyGymnasiumEnv", "RoboHiveEnv",
+            "HabitatEnv", "JumanjiEnv", "IsaacLabEnv", "IsaacGymEnv",
+            "MjLabEnv", "MujocoPlaygroundEnv", "GenesisEnv", "SMACv2Env",
+            "MeltingpotEnv", "OpenSpielEnv", "UnityMLAgentsEnv", "LiberoEnv",
+            "OpenMLEnv",
+        ],
+    )
+    def test_raises_clear_import_error(self, name):
+        import rl_amd.envs.libs as libs
+
+        cls = getattr(libs, name)
+        with pytest.raises((ImportError, NotImplementedError)) as exc:
+            cls(None)
+        if isinstance(exc.value, ImportError):
+            assert "not installed" in str(exc.value)
