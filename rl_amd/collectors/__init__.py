@@ -22,3 +22,5 @@ from .profiling import ProfileConfig, ProfilerHook, enable_profile
 from .graph import GraphedRollout
 from .rpc import RPCCollector
 from .ray import RayCollector, RayLLMCollector
+from .utils import RandomPolicy, split_trajectories
+from .multi import _MultiCollectorBase as MultiCollector

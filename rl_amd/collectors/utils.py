@@ -10,7 +10,7 @@ import torch
 
 from ..tensordict import TensorDict, TensorDictBase, pad as td_pad, stack as td_stack
 
-__all__ = ["split_trajectories"]
+__all__ = ["split_trajectories", "RandomPolicy"]
 
 
 def split_trajectories(
@@ -61,3 +61,17 @@ def split_trajectories(
         piece.set("mask", mask)
         out.append(piece)
     return td_stack(out, 0)
+
+
+class RandomPolicy:
+    """Sample random actions from an action spec (reference
+    collectors/collectors.py RandomPolicy) — the default policy when a
+    collector is built without one."""
+
+    def __init__(self, action_spec, action_key="action"):
+        self.action_spec = action_spec
+        self.action_key = action_key
+
+    def __call__(self, td):
+        td.set(self.action_key, self.action_spec.rand())
+        return td

@@ -75,3 +75,4 @@ from .replay_buffers import (
     StepScheduler,
 )
 from .offline_to_online import OfflineOnlineReplayBuffer
+from .llm import AdaptiveKLController, ConstantKLController

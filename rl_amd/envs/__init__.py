@@ -15,7 +15,7 @@ from .gym_like import GymLikeEnv, default_info_dict_reader
 from .model_based import DreamerEnv, ModelBasedEnvBase
 from . import transforms
 from .llm import ChatEnv
-from .async_envs import AsyncEnvPool
+from .async_envs import AsyncEnvPool, ProcessorAsyncEnvPool, ThreadingAsyncEnvPool
 from .env_creator import EnvCreator, env_creator
 from .custom import ChessEnv, TicTacToeEnv, ToyVLAEnv
 from . import libs

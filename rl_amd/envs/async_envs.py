@@ -17,7 +17,7 @@ import torch
 from ..tensordict import TensorDict, TensorDictBase, stack as td_stack
 from .common import EnvBase
 
-__all__ = ["AsyncEnvPool"]
+__all__ = ["AsyncEnvPool", "ThreadingAsyncEnvPool", "ProcessorAsyncEnvPool"]
 
 
 class AsyncEnvPool:
@@ -128,3 +128,21 @@ class AsyncEnvPool:
     @property
     def observation_spec(self):
         return self._envs[0].observation_spec
+
+
+class ThreadingAsyncEnvPool(AsyncEnvPool):
+    """Thread-backed pool (reference async_envs.py) — right for
+    GIL-releasing simulators and GPU-resident envs."""
+
+    def __init__(self, env_makers, **kwargs):
+        kwargs.pop("backend", None)
+        super().__init__(env_makers, backend="threading", **kwargs)
+
+
+class ProcessorAsyncEnvPool(AsyncEnvPool):
+    """Process-backed pool (reference async_envs.py) — isolates
+    CPU-heavy or crash-prone simulators."""
+
+    def __init__(self, env_makers, **kwargs):
+        kwargs.pop("backend", None)
+        super().__init__(env_makers, backend="multiprocessing", **kwargs)

@@ -88,3 +88,4 @@ from .models import (
 from .models import GPT2RewardModel, OnlineDTActor, RewardModel
 from .models import ExactGPRegressor, GPWorldModel, RBFController
 from .models import ACTModel
+from .value_norm import PopArtValueNorm, RunningValueNorm, ValueNorm
