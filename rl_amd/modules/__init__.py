@@ -89,3 +89,4 @@ from .models import GPT2RewardModel, OnlineDTActor, RewardModel
 from .models import ExactGPRegressor, GPWorldModel, RBFController
 from .models import ACTModel
 from .value_norm import PopArtValueNorm, RunningValueNorm, ValueNorm
+from .vla import TinyVLA
