@@ -1,0 +1,232 @@
+"""IMPALA benchmark — Atari-Pong-shaped actor-learner on MI355X
+(BASELINE.json config 4: "IMPALA Atari Pong, MultiSyncDataCollector
+sharded across 8xMI355X (RCCL all-gather rollouts + V-trace HIP
+kernel)").
+
+One step = collect a T-step unroll from B synthetic Atari envs
+(uint8 [4, 84, 84] frames, 6 actions — Pong's shape) with the behavior
+policy, then one learner update: V-trace advantages (fused HIP kernel,
+csrc/value_scan.hip) + policy-gradient + value + entropy losses, Adam.
+
+Single-process benchmark; the driver's multi-GPU run shards ranks via
+torch.distributed (each rank collects its own unroll, gradients
+all-reduce over RCCL) when launched with torchrun — same contract as
+bench.py.  Metric: env frames/s (whole job).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.nn.functional as F
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from rl_amd import ops
+
+
+class SyntheticAtariEnv:
+    """GPU-resident Pong-shaped env: uint8 [4, 84, 84] frames evolve by a
+    cheap device-side rule; reward depends on the action matching a
+    hidden per-env target that drifts."""
+
+    def __init__(self, batch_size: int, device, frame_shape=(4, 84, 84), n_actions: int = 6):
+        self.B = batch_size
+        self.device = device
+        self.frame_shape = frame_shape
+        self.n_actions = n_actions
+        self.state = torch.randint(
+            0, 255, (batch_size, *frame_shape), dtype=torch.uint8, device=device
+        )
+        self.target = torch.randint(0, n_actions, (batch_size,), device=device)
+        self.t = torch.zeros(batch_size, device=device)
+
+    def reset(self):
+        self.state.random_(0, 255)
+        self.t.zero_()
+        return self.state
+
+    def step(self, action: torch.Tensor):
+        # scroll frames and inject noise — keeps the conv net honest
+        self.state = torch.roll(self.state, shifts=1, dims=1)
+        self.state[:, 0].random_(0, 255)
+        reward = (action == self.target).float().unsqueeze(-1) - 0.05
+        self.t += 1
+        done = (self.t >= 200).unsqueeze(-1)
+        self.t = torch.where(done.squeeze(-1), torch.zeros_like(self.t), self.t)
+        self.target = torch.where(
+            done.squeeze(-1), torch.randint_like(self.target, 0, self.n_actions), self.target
+        )
+        return self.state, reward, done
+
+
+class ImpalaNet(torch.nn.Module):
+    """Shallow IMPALA conv torso (Espeholt et al. 2018) + policy/value."""
+
+    def __init__(self, n_actions: int, device=None):
+        super().__init__()
+        self.conv = torch.nn.Sequential(
+            torch.nn.Conv2d(4, 32, 8, stride=4, device=device),
+            torch.nn.ReLU(),
+            torch.nn.Conv2d(32, 64, 4, stride=2, device=device),
+            torch.nn.ReLU(),
+            torch.nn.Conv2d(64, 64, 3, stride=1, device=device),
+            torch.nn.ReLU(),
+            torch.nn.Flatten(),
+        )
+        self.fc = torch.nn.Sequential(
+            torch.nn.Linear(64 * 7 * 7, 512, device=device), torch.nn.ReLU()
+        )
+        self.policy = torch.nn.Linear(512, n_actions, device=device)
+        self.value = torch.nn.Linear(512, 1, device=device)
+
+    def forward(self, frames: torch.Tensor):
+        x = frames.float() / 255.0
+        h = self.fc(self.conv(x))
+        return self.policy(h), self.value(h)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--envs", type=int, default=64)
+    p.add_argument("--unroll", type=int, default=20)
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if cuda else "cpu")
+    if cuda:
+        torch.cuda.set_device(device)
+    distributed = world > 1
+    if distributed:
+        torch.distributed.init_process_group(backend="nccl" if cuda else "gloo")
+    torch.manual_seed(7 + rank)
+
+    B, T, A = args.envs, args.unroll, 6
+    env = SyntheticAtariEnv(B, device)
+    net = ImpalaNet(A, device=device)
+    # behavior policy: slightly stale copy of the learner (IMPALA lag)
+    behavior = ImpalaNet(A, device=device)
+    behavior.load_state_dict(net.state_dict())
+    optim = torch.optim.Adam(net.parameters(), lr=6e-4)
+    gamma = 0.99
+
+    obs = env.reset()
+
+    def one_iteration():
+        nonlocal obs
+        frames = torch.empty(B, T, *env.frame_shape, dtype=torch.uint8, device=device)
+        actions = torch.empty(B, T, dtype=torch.long, device=device)
+        log_mu = torch.empty(B, T, device=device)
+        rewards = torch.empty(B, T, 1, device=device)
+        dones = torch.empty(B, T, 1, dtype=torch.bool, device=device)
+        with torch.no_grad():
+            for t in range(T):
+                logits, _ = behavior(obs)
+                dist = torch.distributions.Categorical(logits=logits, validate_args=False)
+                a = dist.sample()
+                frames[:, t] = obs
+                actions[:, t] = a
+                log_mu[:, t] = dist.log_prob(a)
+                obs, r, d = env.step(a)
+                rewards[:, t] = r
+                dones[:, t] = d
+        # learner: recompute pi under current weights over the unroll
+        flat = frames.reshape(B * T, *env.frame_shape)
+        logits, values = net(flat)
+        logits = logits.reshape(B, T, A)
+        values = values.reshape(B, T, 1)
+        dist = torch.distributions.Categorical(logits=logits, validate_args=False)
+        log_pi = dist.log_prob(actions)
+        with torch.no_grad():
+            _, next_v_last = net(obs)
+        next_values = torch.cat([values[:, 1:], next_v_last.reshape(B, 1, 1)], 1)
+        adv, vs = ops.vtrace(
+            gamma,
+            log_pi.detach().unsqueeze(-1),
+            log_mu.unsqueeze(-1),
+            values.detach(),
+            next_values.detach(),
+            rewards,
+            dones,
+            dones,
+        )
+        pg_loss = -(log_pi.unsqueeze(-1) * adv).mean()
+        v_loss = 0.5 * (values - vs).pow(2).mean()
+        ent_loss = -0.01 * dist.entropy().mean()
+        loss = pg_loss + v_loss + ent_loss
+        optim.zero_grad(set_to_none=True)
+        loss.backward()
+        if distributed:
+            grads = [p.grad for p in net.parameters() if p.grad is not None]
+            flat_g = torch.cat([g.reshape(-1) for g in grads])
+            torch.distributed.all_reduce(flat_g)
+            flat_g /= world
+            i = 0
+            for g in grads:
+                g.copy_(flat_g[i : i + g.numel()].reshape(g.shape))
+                i += g.numel()
+        optim.step()
+        behavior.load_state_dict(net.state_dict())
+
+    for _ in range(args.warmup):
+        one_iteration()
+    if distributed:
+        torch.distributed.barrier()
+    if cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_iteration()
+    if cuda:
+        torch.cuda.synchronize()
+    if distributed:
+        torch.distributed.barrier()
+    dt = time.perf_counter() - t0
+    tmax = torch.tensor([dt], device=device if cuda else "cpu")
+    if distributed:
+        torch.distributed.all_reduce(tmax, op=torch.distributed.ReduceOp.MAX)
+    dt = float(tmax.item())
+    frames_total = args.steps * B * T * world
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "impala_frames_per_sec",
+                    "value": frames_total / dt,
+                    "unit": "frames/s",
+                    "n_gpus": world,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": dt / args.steps * 1000,
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "fp32",
+                    "data": "synthetic",
+                    "config": {
+                        "model": "impala_shallow_pong",
+                        "global_batch": B * T * world,
+                        "unroll": T,
+                        "n_envs_per_gpu": B,
+                        "parallelism": f"dp{world}",
+                        "vtrace": "hip_kernel" if cuda and ops.HAS_HIP_EXT else "torch",
+                    },
+                }
+            )
+        )
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
