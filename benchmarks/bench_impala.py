@@ -51,16 +51,19 @@ class SyntheticAtariEnv:
         return self.state
 
     def step(self, action: torch.Tensor):
-        # scroll frames and inject noise — keeps the conv net honest
-        self.state = torch.roll(self.state, shifts=1, dims=1)
+        # scroll frames and inject noise — keeps the conv net honest.
+        # All state updates are IN-PLACE so the step loop is
+        # hipGraph-capturable (replays rewrite the same buffers).
+        self.state.copy_(torch.roll(self.state, shifts=1, dims=1))
         self.state[:, 0].random_(0, 255)
         reward = (action == self.target).float().unsqueeze(-1) - 0.05
         self.t += 1
         done = (self.t >= 200).unsqueeze(-1)
-        self.t = torch.where(done.squeeze(-1), torch.zeros_like(self.t), self.t)
-        self.target = torch.where(
-            done.squeeze(-1), torch.randint_like(self.target, 0, self.n_actions), self.target
-        )
+        dflat = done.squeeze(-1)
+        self.t.copy_(torch.where(dflat, torch.zeros_like(self.t), self.t))
+        self.target.copy_(torch.where(
+            dflat, torch.randint_like(self.target, 0, self.n_actions), self.target
+        ))
         return self.state, reward, done
 
 
@@ -97,6 +100,10 @@ def main():
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--envs", type=int, default=64)
     p.add_argument("--unroll", type=int, default=20)
+    p.add_argument("--graph", dest="graph", action="store_true",
+                   help="hipGraph-capture the T-step behavior rollout")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
+    p.set_defaults(graph=True)
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -121,25 +128,49 @@ def main():
     gamma = 0.99
 
     obs = env.reset()
+    # static rollout buffers: reused across iterations so the whole
+    # T-step behavior rollout can be hipGraph-captured and replayed
+    frames = torch.empty(B, T, *env.frame_shape, dtype=torch.uint8, device=device)
+    actions = torch.empty(B, T, dtype=torch.long, device=device)
+    log_mu = torch.empty(B, T, device=device)
+    rewards = torch.empty(B, T, 1, device=device)
+    dones = torch.empty(B, T, 1, dtype=torch.bool, device=device)
 
-    def one_iteration():
+    def rollout_body():
         nonlocal obs
-        frames = torch.empty(B, T, *env.frame_shape, dtype=torch.uint8, device=device)
-        actions = torch.empty(B, T, dtype=torch.long, device=device)
-        log_mu = torch.empty(B, T, device=device)
-        rewards = torch.empty(B, T, 1, device=device)
-        dones = torch.empty(B, T, 1, dtype=torch.bool, device=device)
         with torch.no_grad():
             for t in range(T):
                 logits, _ = behavior(obs)
                 dist = torch.distributions.Categorical(logits=logits, validate_args=False)
                 a = dist.sample()
-                frames[:, t] = obs
-                actions[:, t] = a
-                log_mu[:, t] = dist.log_prob(a)
-                obs, r, d = env.step(a)
-                rewards[:, t] = r
-                dones[:, t] = d
+                frames[:, t].copy_(obs)
+                actions[:, t].copy_(a)
+                log_mu[:, t].copy_(dist.log_prob(a))
+                _, r, d = env.step(a)  # obs aliases env.state (in-place)
+                rewards[:, t].copy_(r)
+                dones[:, t].copy_(d)
+
+    rollout = rollout_body
+    if args.graph and cuda:
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    rollout_body()
+            torch.cuda.current_stream().wait_stream(side)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                rollout_body()
+            rollout = g.replay
+        except Exception:
+            import traceback
+
+            traceback.print_exc()
+            args.graph = False
+
+    def one_iteration():
+        rollout()
         # learner: recompute pi under current weights over the unroll
         flat = frames.reshape(B * T, *env.frame_shape)
         logits, values = net(flat)
