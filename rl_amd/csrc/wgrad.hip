@@ -102,50 +102,52 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_splitk_kernel(
     __syncthreads();
   }
 
-  // write this workgroup's partial tile, slab-axis INNERMOST so the
-  // reduce kernel streams contiguous runs of `slabs` floats per element
+  // write this workgroup's partial tile, slab-axis OUTERMOST: the
+  // reduce's inner slab loop then reads CONSECUTIVE addresses across
+  // lanes (coalesced) each iteration.  (The slab-innermost variant gave
+  // each lane its own 512 B run — uncoalesced, measured 26.6 us.)
   const long slabs = gridDim.x;
   const long num_tiles_m = gridDim.z;
-  float* part = dw + ((long)blockIdx.y * num_tiles_m + blockIdx.z) *
-                         (64 * 64 * slabs);
+  const long tile_elems = (long)64 * 64 * gridDim.y * num_tiles_m;
+  float* part = dw + (long)blockIdx.x * tile_elems +
+                ((long)blockIdx.y * num_tiles_m + blockIdx.z) * (64 * 64);
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      part[((long)(tn * 4 + i) * 64 + tm * 4 + j) * slabs + blockIdx.x] =
-          acc[i][j];
+      part[(long)(tn * 4 + i) * 64 + tm * 4 + j] = acc[i][j];
     }
     if (dbias != nullptr && tm == 0 && m0 == 0) {
-      dbias[(long)(n_base + i) * slabs + blockIdx.x] = bias_acc[i];
+      dbias[(long)blockIdx.x * (gridDim.y * 64) + n_base + i] = bias_acc[i];
     }
   }
 }
 
-// reduce the [tile, elem, slab] partials into dW [N, M] (and the
-// [n, slab] bias partials into dBias [N]); each element's slab run is
-// contiguous, so the inner loop streams.
+// reduce the [slab, tile, elem] partials into dW [N, M] (and the
+// [slab, n] bias partials into dBias [N]); lanes read consecutive
+// elements of one slab per iteration — fully coalesced.
 __global__ void wgrad_reduce_kernel(const float* __restrict__ part,
                                     const float* __restrict__ bias_part,
                                     float* __restrict__ dw,
                                     float* __restrict__ dbias, int slabs,
                                     int tiles_n, int tiles_m, int N, int M) {
+  const long tile_elems = (long)64 * 64 * tiles_n * tiles_m;
   for (long idx = blockIdx.x * blockDim.x + threadIdx.x;
        idx < (long)N * M; idx += (long)gridDim.x * blockDim.x) {
     const int n = (int)(idx / M), m = (int)(idx % M);
     const int tn = n / 64, tm = m / 64;
-    const float* p = part +
-                     ((long)tn * tiles_m + tm) * (64 * 64 * (long)slabs) +
-                     ((long)(n % 64) * 64 + (m % 64)) * slabs;
+    const long off = ((long)tn * tiles_m + tm) * (64 * 64) +
+                     (long)(n % 64) * 64 + (m % 64);
     float acc = 0.f;
-    for (int s = 0; s < slabs; ++s) acc += p[s];
+    for (int s = 0; s < slabs; ++s) acc += part[(long)s * tile_elems + off];
     dw[idx] = acc;
   }
   if (dbias != nullptr) {
     for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
          n += gridDim.x * blockDim.x) {
-      const float* p = bias_part + (long)n * slabs;
       float acc = 0.f;
-      for (int s = 0; s < slabs; ++s) acc += p[s];
+      for (int s = 0; s < slabs; ++s)
+        acc += bias_part[(long)s * (tiles_n * 64) + n];
       dbias[n] = acc;
     }
   }
