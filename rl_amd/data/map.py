@@ -148,6 +148,34 @@ class PUCTScore:
         }
 
 
+class EXP3Score:
+    """EXP3 adversarial-bandit selection (reference mcts/scores.py:241):
+    exponential weights with a gamma-mixed uniform exploration floor."""
+
+    def __init__(self, gamma: float = 0.1):
+        if not 0 <= gamma <= 1:
+            raise ValueError("gamma must be in [0, 1]")
+        self.gamma = gamma
+
+    def __call__(self, node: "Tree") -> Dict[Any, float]:
+        import math
+
+        actions = list(node.children.keys())
+        k = len(actions)
+        if k == 0:
+            return {}
+        # weight = exp(gamma * mean_reward * visits / k), probability-mixed
+        ws = {
+            a: math.exp(min(50.0, self.gamma * ch.value * max(1, ch.visits) / k))
+            for a, ch in node.children.items()
+        }
+        total = sum(ws.values())
+        return {
+            a: (1 - self.gamma) * w / total + self.gamma / k
+            for a, w in ws.items()
+        }
+
+
 class UCBScore:
     """UCB1 (reference scores.py:150)."""
 
@@ -213,3 +241,49 @@ class MCTSForest:
 
     def __len__(self):
         return len(self._roots)
+
+
+class BinaryToDecimal(torch.nn.Module):
+    """Convert binary-encoded tensors to decimals (reference
+    map/hash.py:13): groups of ``num_bits`` bits fold to integers,
+    summed over the last dim when the input is wider."""
+
+    def __init__(self, num_bits: int, device=None, dtype=torch.int64, convert_to_binary: bool = True):
+        super().__init__()
+        self.num_bits = num_bits
+        self.convert_to_binary = convert_to_binary
+        self.register_buffer(
+            "bases", 2 ** torch.arange(num_bits - 1, -1, -1, device=device)
+        )
+        self.dtype = dtype
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.convert_to_binary:
+            x = torch.heaviside(x.float(), torch.zeros((), device=x.device))
+        L = x.shape[-1]
+        if L % self.num_bits:
+            raise ValueError("input length must divide num_bits")
+        g = x.reshape(*x.shape[:-1], L // self.num_bits, self.num_bits)
+        vals = (g.to(self.dtype) * self.bases.to(self.dtype)).sum(-1)
+        return vals.sum(-1) if vals.shape[-1] > 1 else vals.squeeze(-1)
+
+
+class HashToInt:
+    """Map arbitrary hash values to compact consecutive ints (reference
+    map/hash.py HashToInt) — stable indices for storage-backed maps."""
+
+    def __init__(self):
+        self._map: Dict[int, int] = {}
+
+    def __call__(self, key: torch.Tensor) -> torch.Tensor:
+        flat = key.reshape(-1)
+        out = torch.empty_like(flat)
+        for i, v in enumerate(flat.tolist()):
+            out[i] = self._map.setdefault(int(v), len(self._map))
+        return out.reshape(key.shape)
+
+    def state_dict(self):
+        return {"map": dict(self._map)}
+
+    def load_state_dict(self, sd):
+        self._map = {int(k): int(v) for k, v in sd["map"].items()}

@@ -90,3 +90,10 @@ from .models import ExactGPRegressor, GPWorldModel, RBFController
 from .models import ACTModel
 from .value_norm import PopArtValueNorm, RunningValueNorm, ValueNorm
 from .vla import TinyVLA
+from .value_transforms import (
+    ComposeValueTransform,
+    IdentityValueTransform,
+    SignedHyperbolicValueTransform,
+    SymLogValueTransform,
+    ValueTransform,
+)
