@@ -39,6 +39,10 @@ void launch_fused_actor(const float*, const float*, const float*, const float*,
                         int, float, float, void*);
 int lstm_fused_lds_bytes(int);
 int wgrad_slab_count(long);
+int synthetic_env_step_lds_bytes(int, int);
+void launch_synthetic_env_step(float*, const float*, const float*,
+                               const float*, float*, float*, float*, bool*,
+                               int, int, int, float, void*);
 void launch_wgrad_splitk(const void*, const void*, float*, float*, float*,
                          float*, long, int, int, void*);
 void launch_gru_fused(const float*, const float*, const float*, const bool*,
@@ -251,6 +255,33 @@ std::vector<torch::Tensor> wgrad_splitk(torch::Tensor dy, torch::Tensor x,
   if (want_bias) return {dw, db};
   return {dw};
 }
+// Fused synthetic-env transition: one launch replaces the ~12-kernel
+// eager step (GEMMs + tanh + reward + time/done bookkeeping).
+std::vector<torch::Tensor> synthetic_env_step(torch::Tensor state,
+                                              torch::Tensor action,
+                                              torch::Tensor A, torch::Tensor B,
+                                              torch::Tensor t,
+                                              double max_steps) {
+  TORCH_CHECK(state.is_cuda() && state.scalar_type() == torch::kFloat32,
+              "synthetic_env_step: fp32 cuda state");
+  TORCH_CHECK(state.is_contiguous() && action.is_contiguous() &&
+                  A.is_contiguous() && B.is_contiguous() && t.is_contiguous(),
+              "contiguous inputs");
+  const int Bn = (int)state.size(0), S = (int)state.size(1),
+            Aact = (int)action.size(1);
+  TORCH_CHECK(synthetic_env_step_lds_bytes(S, Aact) <= 160 * 1024,
+              "env dims exceed the LDS budget");
+  auto obs = torch::empty_like(state);
+  auto reward = torch::empty({Bn, 1}, state.options());
+  auto done = torch::empty({Bn, 1}, state.options().dtype(torch::kBool));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_synthetic_env_step(state.data_ptr<float>(), action.data_ptr<float>(),
+                            A.data_ptr<float>(), B.data_ptr<float>(),
+                            t.data_ptr<float>(), obs.data_ptr<float>(),
+                            reward.data_ptr<float>(), done.data_ptr<bool>(),
+                            Bn, S, Aact, (float)max_steps, (void*)stream);
+  return {obs, reward, done};
+}
 #endif  // RL_AMD_WITH_HIP
 
 // ---------------------------------------------------------------------------
@@ -280,6 +311,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
 #ifdef RL_AMD_WITH_HIP
   m.def("gae", &gae, "fused GAE scan (HIP)");
+  m.def("synthetic_env_step", &synthetic_env_step,
+        "fused synthetic-MuJoCo env transition (HIP)");
   m.def("wgrad_splitk", &wgrad_splitk,
         "split-K skinny weight gradient dW=dY^T X (HIP)");
   m.def("revscan", &revscan, "generic reverse linear-recurrence scan (HIP)");

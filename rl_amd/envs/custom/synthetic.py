@@ -73,6 +73,7 @@ class SyntheticMuJoCoEnv(EnvBase):
         self._t: Optional[torch.Tensor] = None
         self._gen = torch.Generator(device="cpu")
         self._capture_safe = False
+        self._zeros_done: Optional[torch.Tensor] = None
 
     def enable_capture_mode(self, mode: bool = True) -> "SyntheticMuJoCoEnv":
         """In-place state updates for hipGraph capture."""
@@ -114,6 +115,14 @@ class SyntheticMuJoCoEnv(EnvBase):
         )
 
     def _step(self, tensordict: TensorDictBase) -> TensorDictBase:
+        if (
+            self._capture_safe
+            and self.device.type == "cuda"
+            and self.dtype == torch.float32
+        ):
+            fused = self._fused_step(tensordict)
+            if fused is not None:
+                return fused
         action = tensordict.get("action").to(self.dtype).clamp(-1, 1)
         s = self._state
         drive = action @ self.B
@@ -136,6 +145,37 @@ class SyntheticMuJoCoEnv(EnvBase):
                 "done": truncated,
                 "terminated": torch.zeros_like(truncated),
                 "truncated": truncated,
+            },
+            batch_size=bs,
+            device=self.device,
+        )
+
+    def _fused_step(self, tensordict: TensorDictBase) -> Optional[TensorDictBase]:
+        """One-kernel transition (csrc/env_step.hip): the eager step is
+        ~12 launches of ~4 us inside the captured rollout."""
+        from ... import ops
+
+        if not ops.HAS_HIP_EXT:
+            return None
+        from ... import _C
+
+        bs = self.batch_size
+        action = tensordict.get("action")
+        if action.dtype != torch.float32:
+            action = action.float()
+        obs, reward, done = _C.synthetic_env_step(
+            self._state, action.contiguous(), self.A, self.B,
+            self._t.reshape(-1), float(self.max_steps),
+        )
+        if self._zeros_done is None or self._zeros_done.shape[0] != done.shape[0]:
+            self._zeros_done = torch.zeros_like(done)
+        return TensorDict(
+            {
+                "observation": obs,
+                "reward": reward,
+                "done": done,
+                "terminated": self._zeros_done,
+                "truncated": done,
             },
             batch_size=bs,
             device=self.device,
