@@ -224,6 +224,51 @@ def main():
         # close the loop: replays start from the final observation
         entry_obs.copy_(carrier.get("observation"))
 
+    # store-direct rollout: the fused actor and env kernels write action,
+    # log-prob, pre/post observations, reward and done STRAIGHT into the
+    # strided [B, T] store (zero copy kernels), and the env kernel
+    # auto-resets the carried state — 2 + 2T launches per iteration.
+    store_direct = bool(args.fused_actor and args.graph and cuda)
+    if store_direct:
+        try:
+            from rl_amd import _C
+            from rl_amd.ops import HAS_HIP_EXT
+
+            store_direct = HAS_HIP_EXT and hasattr(_C, "fused_actor_into")
+        except Exception:
+            store_direct = False
+    if store_direct:
+        fa = rollout_policy  # FusedTanhNormalActor
+        w1, w2, w3 = (l.weight for l in fa.linears)
+        b1, b2, b3 = (l.bias for l in fa.linears)
+        s_obs = store.get("observation")
+        s_act = store.get("action")
+        s_lp = store.get("sample_log_prob")
+        s_nobs = store.get(("next", "observation"))
+        s_rew = store.get(("next", "reward"))
+        s_done = store.get(("next", "done"))
+        store.get(("next", "terminated")).zero_()  # env never terminates
+        env.enable_capture_mode(True)
+
+        def rollout_body_direct():
+            with torch.no_grad():
+                eps_all = torch.randn(T, B, act_dim, device=device)
+                noise_all = torch.randn(T, B, obs_dim, device=device) * 0.1
+                for t in range(T):
+                    _C.fused_actor_into(
+                        env._state, w1, b1, w2, b2, w3, b3, eps_all[t],
+                        s_act[:, t], s_lp[:, t], fa.inv_softplus_bias,
+                        fa.scale_lb,
+                    )
+                    _C.synthetic_env_step_into(
+                        env._state, s_act[:, t], env.A, env.B,
+                        env._t.reshape(-1), s_nobs[:, t], s_obs[:, t],
+                        s_rew[:, t], s_done[:, t], noise_all[t],
+                        float(env.max_steps),
+                    )
+
+        rollout_body = rollout_body_direct
+
     rollout = rollout_body
     if args.graph and cuda:
         env.enable_capture_mode(True)
@@ -370,6 +415,7 @@ def main():
                 "hip_graph": bool(args.graph and cuda),
                 "fused_actor": bool(args.fused_actor and cuda),
                 "splitk_wgrad": bool(args.splitk and cuda),
+                "store_direct_rollout": bool(store_direct),
                 "full_step_graph": bool(args.full_graph and cuda and not distributed),
             },
         }

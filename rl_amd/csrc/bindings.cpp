@@ -35,14 +35,15 @@ int gru_fused_lds_bytes(int);
 int fused_actor_lds_bytes(int, int, int, int);
 void launch_fused_actor(const float*, const float*, const float*, const float*,
                         const float*, const float*, const float*, const float*,
-                        float*, float*, float*, float*, int, int, int, int,
-                        int, float, float, void*);
+                        float*, float*, float*, float*, long, long, int, int,
+                        int, int, int, float, float, void*);
 int lstm_fused_lds_bytes(int);
 int wgrad_slab_count(long);
 int synthetic_env_step_lds_bytes(int, int);
 void launch_synthetic_env_step(float*, const float*, const float*,
-                               const float*, float*, float*, float*, bool*,
-                               int, int, int, float, void*);
+                               const float*, float*, float*, float*, float*,
+                               bool*, const float*, long, long, long, int,
+                               int, int, float, void*);
 void launch_wgrad_splitk(const void*, const void*, float*, float*, float*,
                          float*, long, int, int, void*);
 void launch_gru_fused(const float*, const float*, const float*, const bool*,
@@ -223,8 +224,8 @@ std::vector<torch::Tensor> fused_actor(
       b3.data_ptr<float>(), eps.data_ptr<float>(), action.data_ptr<float>(),
       logp.data_ptr<float>(),
       want_loc_scale ? loc.data_ptr<float>() : nullptr,
-      want_loc_scale ? scale.data_ptr<float>() : nullptr, B, O, H1, H2, A,
-      (float)inv_softplus_bias, (float)scale_lb, (void*)stream);
+      want_loc_scale ? scale.data_ptr<float>() : nullptr, (long)A, 1L, B, O,
+      H1, H2, A, (float)inv_softplus_bias, (float)scale_lb, (void*)stream);
   if (want_loc_scale) return {action, logp, loc, scale};
   return {action, logp};
 }
@@ -278,9 +279,57 @@ std::vector<torch::Tensor> synthetic_env_step(torch::Tensor state,
   launch_synthetic_env_step(state.data_ptr<float>(), action.data_ptr<float>(),
                             A.data_ptr<float>(), B.data_ptr<float>(),
                             t.data_ptr<float>(), obs.data_ptr<float>(),
-                            reward.data_ptr<float>(), done.data_ptr<bool>(),
-                            Bn, S, Aact, (float)max_steps, (void*)stream);
+                            nullptr, reward.data_ptr<float>(),
+                            done.data_ptr<bool>(), nullptr, (long)S, 1,
+                            (long)Aact, Bn, S, Aact, (float)max_steps,
+                            (void*)stream);
   return {obs, reward, done};
+}
+
+// Store-direct variant: writes pre/post observations, reward and done
+// straight into strided [B, T] rollout-store views and auto-resets the
+// carried state from `reset_noise` — the 4-launch rollout step.
+void synthetic_env_step_into(torch::Tensor state, torch::Tensor action,
+                             torch::Tensor A, torch::Tensor B,
+                             torch::Tensor t, torch::Tensor next_obs,
+                             torch::Tensor prev_obs, torch::Tensor reward,
+                             torch::Tensor done, torch::Tensor reset_noise,
+                             double max_steps) {
+  TORCH_CHECK(state.is_cuda() && state.scalar_type() == torch::kFloat32,
+              "synthetic_env_step_into: fp32 cuda state");
+  TORCH_CHECK(next_obs.stride(-1) == 1 && prev_obs.stride(-1) == 1,
+              "obs views must be inner-contiguous");
+  const int Bn = (int)state.size(0), S = (int)state.size(1),
+            Aact = (int)action.size(-1);
+  TORCH_CHECK(action.stride(-1) == 1, "action inner-contiguous");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_synthetic_env_step(
+      state.data_ptr<float>(), action.data_ptr<float>(), A.data_ptr<float>(),
+      B.data_ptr<float>(), t.data_ptr<float>(), next_obs.data_ptr<float>(),
+      prev_obs.data_ptr<float>(), reward.data_ptr<float>(),
+      done.data_ptr<bool>(), reset_noise.data_ptr<float>(),
+      (long)next_obs.stride(0), (long)reward.stride(0),
+      (long)action.stride(0), Bn, S, Aact, (float)max_steps, (void*)stream);
+}
+// Store-direct fused actor: action/log-prob written straight into
+// strided [B, T] rollout-store views.
+void fused_actor_into(torch::Tensor obs, torch::Tensor w1, torch::Tensor b1,
+                      torch::Tensor w2, torch::Tensor b2, torch::Tensor w3,
+                      torch::Tensor b3, torch::Tensor eps,
+                      torch::Tensor action, torch::Tensor logp,
+                      double inv_softplus_bias, double scale_lb) {
+  TORCH_CHECK(obs.is_cuda() && obs.is_contiguous(), "obs device+contig");
+  TORCH_CHECK(action.stride(-1) == 1, "action inner-contiguous");
+  long B = obs.size(0), O = obs.size(1);
+  long H1 = w1.size(0), H2 = w2.size(0), A = w3.size(0) / 2;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_fused_actor(
+      obs.data_ptr<float>(), w1.data_ptr<float>(), b1.data_ptr<float>(),
+      w2.data_ptr<float>(), b2.data_ptr<float>(), w3.data_ptr<float>(),
+      b3.data_ptr<float>(), eps.data_ptr<float>(), action.data_ptr<float>(),
+      logp.data_ptr<float>(), nullptr, nullptr, (long)action.stride(0),
+      (long)logp.stride(0), B, O, H1, H2, A, (float)inv_softplus_bias,
+      (float)scale_lb, (void*)stream);
 }
 #endif  // RL_AMD_WITH_HIP
 
@@ -313,6 +362,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gae", &gae, "fused GAE scan (HIP)");
   m.def("synthetic_env_step", &synthetic_env_step,
         "fused synthetic-MuJoCo env transition (HIP)");
+  m.def("fused_actor_into", &fused_actor_into,
+        "store-direct fused actor (HIP)");
+  m.def("synthetic_env_step_into", &synthetic_env_step_into,
+        "store-direct fused env transition with auto-reset (HIP)");
   m.def("wgrad_splitk", &wgrad_splitk,
         "split-K skinny weight gradient dW=dY^T X (HIP)");
   m.def("revscan", &revscan, "generic reverse linear-recurrence scan (HIP)");

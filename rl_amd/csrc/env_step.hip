@@ -30,9 +30,12 @@ __global__ void __launch_bounds__(ENV_THREADS) synthetic_env_step_kernel(
     const float* __restrict__ Amat,   // [S, S] row-major (k, j)
     const float* __restrict__ Bmat,   // [Aact, S]
     float* __restrict__ t,            // [B, 1] in/out
-    float* __restrict__ obs_out,      // [B, S]
-    float* __restrict__ reward,       // [B, 1]
-    bool* __restrict__ done,          // [B, 1]
+    float* __restrict__ obs_out,      // next obs, row stride obs_rs
+    float* __restrict__ prev_out,     // pre-step obs (nullable), row stride obs_rs
+    float* __restrict__ reward,       // row stride sc_rs
+    bool* __restrict__ done,          // row stride sc_rs
+    const float* __restrict__ reset_noise,  // [B, S] (nullable): auto-reset
+    const long obs_rs, const long sc_rs, const long act_rs,
     const int Bn, const int S, const int Aact, const float max_steps) {
   extern __shared__ float smem[];
   const int apad = S + 1;
@@ -40,6 +43,7 @@ __global__ void __launch_bounds__(ENV_THREADS) synthetic_env_step_kernel(
   float* s_B = s_A + S * apad;       // [Aact, apad]
   float* s_state = s_B + Aact * apad;  // [ENV_ROWS, apad] (old state)
   float* s_act = s_state + ENV_ROWS * apad;  // [ENV_ROWS, Aact]
+  float* s_t = s_act + ENV_ROWS * Aact;      // [ENV_ROWS] (old t)
 
   const int tid = threadIdx.x;
   const int row0 = blockIdx.x * ENV_ROWS;
@@ -55,9 +59,10 @@ __global__ void __launch_bounds__(ENV_THREADS) synthetic_env_step_kernel(
   }
   for (int i = tid; i < rows * Aact; i += ENV_THREADS) {
     const int r = i / Aact;
-    float a = action[(size_t)(row0 + r) * Aact + i % Aact];
+    float a = action[(size_t)(row0 + r) * act_rs + i % Aact];
     s_act[r * Aact + i % Aact] = fminf(1.f, fmaxf(-1.f, a));
   }
+  for (int r = tid; r < rows; r += ENV_THREADS) s_t[r] = t[row0 + r];
   __syncthreads();
 
   for (int i = tid; i < rows * S; i += ENV_THREADS) {
@@ -71,19 +76,24 @@ __global__ void __launch_bounds__(ENV_THREADS) synthetic_env_step_kernel(
 #pragma unroll
     for (int k = 0; k < Aact; ++k) acc += ar[k] * s_B[k * apad + j];
     const float ns = tanhf(acc);
-    const size_t gi = (size_t)(row0 + r) * S + j;
-    state[gi] = ns;
-    obs_out[gi] = ns;
+    const int gr = row0 + r;
+    const bool trunc = (s_t[r] + 1.f) >= max_steps;
+    // step_and_maybe_reset semantics: the STORE sees the terminal obs,
+    // the carried state auto-resets (noise) so the next step continues
+    // a fresh trajectory with zero host involvement.
+    float carry = ns;
+    if (reset_noise != nullptr && trunc)
+      carry = reset_noise[(size_t)gr * S + j];
+    state[(size_t)gr * S + j] = carry;
+    obs_out[(size_t)gr * obs_rs + j] = ns;
+    if (prev_out != nullptr) prev_out[(size_t)gr * obs_rs + j] = sr[j];
     if (j == 0) {
       float ctrl = 0.f;
 #pragma unroll
       for (int k = 0; k < Aact; ++k) ctrl += ar[k] * ar[k];
-      const int gr = row0 + r;
-      const float tn = t[gr] + 1.f;
-      t[gr] = tn;
-      const bool trunc = tn >= max_steps;
-      reward[gr] = ns - 0.1f * ctrl;
-      done[gr] = trunc;
+      t[gr] = (reset_noise != nullptr && trunc) ? 0.f : s_t[r] + 1.f;
+      reward[(size_t)gr * sc_rs] = ns - 0.1f * ctrl;
+      done[(size_t)gr * sc_rs] = trunc;
     }
   }
 }
@@ -93,19 +103,23 @@ __global__ void __launch_bounds__(ENV_THREADS) synthetic_env_step_kernel(
 extern "C" int synthetic_env_step_lds_bytes(int S, int Aact) {
   const int apad = S + 1;
   return (int)sizeof(float) *
-         (S * apad + Aact * apad + ENV_ROWS * apad + ENV_ROWS * Aact);
+         (S * apad + Aact * apad + ENV_ROWS * apad + ENV_ROWS * Aact +
+          ENV_ROWS);
 }
 
 extern "C" void launch_synthetic_env_step(float* state, const float* action,
                                           const float* Amat, const float* Bmat,
                                           float* t, float* obs_out,
-                                          float* reward, bool* done, int Bn,
-                                          int S, int Aact, float max_steps,
-                                          void* stream) {
+                                          float* prev_out, float* reward,
+                                          bool* done, const float* reset_noise,
+                                          long obs_rs, long sc_rs, long act_rs,
+                                          int Bn, int S, int Aact,
+                                          float max_steps, void* stream) {
   const int blocks = (Bn + ENV_ROWS - 1) / ENV_ROWS;
   const int lds = synthetic_env_step_lds_bytes(S, Aact);
   hipLaunchKernelGGL(synthetic_env_step_kernel, dim3(blocks),
                      dim3(ENV_THREADS), lds, (hipStream_t)stream, state,
-                     action, Amat, Bmat, t, obs_out, reward, done, Bn, S, Aact,
+                     action, Amat, Bmat, t, obs_out, prev_out, reward, done,
+                     reset_noise, obs_rs, sc_rs, act_rs, Bn, S, Aact,
                      max_steps);
 }
