@@ -355,3 +355,54 @@ class TestFusedEnvStep:
                 a, b = out_e.get(k), out_f.get(k)
                 assert torch.allclose(a, b, rtol=1e-4, atol=1e-5), (i, k, (a - b).abs().max())
             assert torch.equal(out_e.get(("next", "done")), out_f.get(("next", "done")))
+
+
+@pytest.mark.gpu
+class TestStoreDirectEnvStep:
+    def test_auto_reset_semantics(self):
+        """synthetic_env_step_into must write the TERMINAL obs to the
+        store while carrying the RESET state (step_and_maybe_reset
+        contract), zeroing t at truncation."""
+        from rl_amd import _C
+        from rl_amd.envs.custom.synthetic import HalfCheetahVec
+
+        torch.manual_seed(0)
+        B, T = 32, 4
+        env = HalfCheetahVec(batch_size=[B], device="cuda", dtype=torch.float32)
+        env.set_seed(0)
+        env.reset()
+        env.enable_capture_mode(True)
+        S, A = env.obs_dim, env.act_dim
+        # force truncation on the next step
+        env._t.fill_(float(env.max_steps) - 1)
+        prev = torch.empty(B, T, S, device="cuda")
+        nxt = torch.empty(B, T, S, device="cuda")
+        rew = torch.empty(B, T, 1, device="cuda")
+        done = torch.empty(B, T, 1, dtype=torch.bool, device="cuda")
+        state_before = env._state.clone()
+        action = torch.rand(B, A, device="cuda") * 2 - 1
+        noise = torch.randn(B, S, device="cuda") * 0.1
+        t0 = 1
+        _C.synthetic_env_step_into(
+            env._state, action, env.A, env.B, env._t.reshape(-1),
+            nxt[:, t0], prev[:, t0], rew[:, t0], done[:, t0], noise,
+            float(env.max_steps),
+        )
+        assert done[:, t0].all()  # truncated
+        assert torch.equal(prev[:, t0], state_before)  # pre-step obs stored
+        # terminal obs stored ≠ carried state (which took the reset noise)
+        expected_terminal = torch.tanh(
+            state_before @ env.A + action.clamp(-1, 1) @ env.B
+        )
+        assert torch.allclose(nxt[:, t0], expected_terminal, atol=1e-5)
+        assert torch.allclose(env._state, noise, atol=1e-6)  # reset carried
+        assert (env._t == 0).all()  # t zeroed
+        # non-truncating step: state advances normally, t increments
+        _C.synthetic_env_step_into(
+            env._state, action, env.A, env.B, env._t.reshape(-1),
+            nxt[:, 2], prev[:, 2], rew[:, 2], done[:, 2], noise,
+            float(env.max_steps),
+        )
+        assert not done[:, 2].any()
+        assert (env._t == 1).all()
+        assert torch.allclose(env._state, nxt[:, 2], atol=1e-6)
