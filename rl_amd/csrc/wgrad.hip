@@ -21,15 +21,26 @@
 //    second kernel reduces over the slab axis.
 //  * dBias[n] = sum_k dY[k][n] is fused (threads with tm==0).
 //
-// No MFMA: at these shapes the kernel is bound by the 2*(M+N)*B byte
-// stream from HBM, which VALU FMAs already saturate; MFMA would add
-// fragment-layout complexity for zero wall-clock gain.
+// Two compute paths share the same LDS staging and partials layout:
+//  * wgrad_mfma_kernel — MFMA path (mfma_f32_16x16x32_bf16): each of
+//    the 4 waves owns 4 of the 16 16x16 output tiles; fragments are
+//    read straight from the padded LDS chunk.  Fragment mapping
+//    verified on hardware (benchmarks/mfma_probe.hip): A row/B col =
+//    lane&15, C/D row = (lane>>4)*4+reg, col = lane&15; the k
+//    enumeration only needs to be CONSISTENT between A and B (any
+//    k-permutation cancels in the contraction).
+//  * wgrad_splitk_kernel — VALU fallback for shapes that are not
+//    multiples of 16 in spirit (still correct for any N, M via the
+//    zero-padded staging; kept for comparison runs).
 //
 // Reference behavior: torch.nn.functional.linear backward (wgrad);
 // numerics validated vs fp32 torch.mm in tests/test_ops.py.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+
+using bf16_frag = __attribute__((ext_vector_type(8))) short;
+using f32_frag = __attribute__((ext_vector_type(4))) float;
 
 #define WG_THREADS 256
 #define ROW_CHUNK 64   // k-rows staged in LDS per iteration
@@ -123,6 +134,98 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_splitk_kernel(
   }
 }
 
+// MFMA split-K wgrad: same grid/staging/partials contract as the VALU
+// kernel; compute runs on the matrix cores.
+__global__ void __launch_bounds__(WG_THREADS) wgrad_mfma_kernel(
+    const __hip_bfloat16* __restrict__ dy,  // [K, N]
+    const __hip_bfloat16* __restrict__ x,   // [K, M]
+    float* __restrict__ dw,                 // partials [slab, tile, elem]
+    float* __restrict__ dbias,              // bias partials [slab, n]
+    long K, int N, int M, int k_slab) {
+  const int n0t = blockIdx.y * 64;
+  const int m0t = blockIdx.z * 64;
+  const long k_begin = (long)blockIdx.x * k_slab;
+  const long k_end = min(K, k_begin + (long)k_slab);
+
+  // bf16 LDS staging, padded rows (stride 66 shorts = 33 banks)
+  __shared__ __hip_bfloat16 s_dy[ROW_CHUNK][64 + 2];
+  __shared__ __hip_bfloat16 s_x[ROW_CHUNK][64 + 2];
+  __shared__ float s_bias[64];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;       // 0..3
+  const int lane = tid & 63;
+  // wave w owns output tiles (n_tile, m_tile) = (w, 0..3): n rows
+  // n0t + w*16 .. +15, all four 16-col m tiles.
+  f32_frag acc[4] = {};
+  float bias_acc = 0.f;
+
+  const int lr = tid / 64;  // staging: thread t loads column t%64
+  const int lc = tid % 64;
+  for (long kc = k_begin; kc < k_end; kc += ROW_CHUNK) {
+    const int rows = (int)min((long)ROW_CHUNK, k_end - kc);
+    for (int i = 0; i < ROW_CHUNK / 4; ++i) {
+      const int r = lr + 4 * i;
+      const long gk = kc + r;
+      __hip_bfloat16 dv = __hip_bfloat16(0.f), xv = __hip_bfloat16(0.f);
+      if (r < rows) {
+        if (lc < N - n0t) dv = dy[gk * N + n0t + lc];
+        if (lc < M - m0t) xv = x[gk * M + m0t + lc];
+      }
+      s_dy[r][lc] = dv;
+      s_x[r][lc] = xv;
+      bias_acc += __bfloat162float(dv);
+    }
+    __syncthreads();
+    // MFMA over the chunk: A[mf][k] = dy[k][n0w + mf], B[k][nf] = x[k][m0 + nf]
+    const int n_off = wave * 16 + (lane & 15);
+    for (int kk = 0; kk < ROW_CHUNK; kk += 32) {
+      bf16_frag a;
+#pragma unroll
+      for (int reg = 0; reg < 8; ++reg) {
+        const int k = kk + 8 * (lane >> 4) + reg;
+        a[reg] = *reinterpret_cast<const short*>(&s_dy[k][n_off]);
+      }
+#pragma unroll
+      for (int mt = 0; mt < 4; ++mt) {
+        bf16_frag b;
+        const int m_off = mt * 16 + (lane & 15);
+#pragma unroll
+        for (int reg = 0; reg < 8; ++reg) {
+          const int k = kk + 8 * (lane >> 4) + reg;
+          b[reg] = *reinterpret_cast<const short*>(&s_x[k][m_off]);
+        }
+        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // commit: D row = (lane>>4)*4 + reg (within the wave's 16-row band)
+  const long num_tiles_m = gridDim.z;
+  const long tile_elems = (long)64 * 64 * gridDim.y * num_tiles_m;
+  float* part = dw + (long)blockIdx.x * tile_elems +
+                ((long)blockIdx.y * num_tiles_m + blockIdx.z) * (64 * 64);
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int n_loc = wave * 16 + (lane >> 4) * 4 + reg;
+      const int m_loc = mt * 16 + (lane & 15);
+      part[(long)n_loc * 64 + m_loc] = acc[mt][reg];
+    }
+  }
+  // bias: 4 threads share each column; reduce through LDS
+  if (dbias != nullptr) {
+    if (lr == 0) s_bias[lc] = 0.f;
+    __syncthreads();
+    atomicAdd(&s_bias[lc], bias_acc);
+    __syncthreads();
+    if (lr == 0)
+      dbias[(long)blockIdx.x * (gridDim.y * 64) + blockIdx.y * 64 + lc] = s_bias[lc];
+  }
+}
+
 // reduce the [slab, tile, elem] partials into dW [N, M].  A thread per
 // ELEMENT is latency-bound (N*M = 4096 threads x 128 dependent loads
 // measured 26-56 us): instead RED_GROUPS threads share each element,
@@ -180,7 +283,7 @@ extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
   const int slabs = (int)((K + k_slab - 1) / k_slab);
   const int tiles_n = (N + 63) / 64, tiles_m = (M + 63) / 64;
   dim3 grid(slabs, tiles_n, tiles_m);
-  hipLaunchKernelGGL(wgrad_splitk_kernel, grid, dim3(WG_THREADS), 0,
+  hipLaunchKernelGGL(wgrad_mfma_kernel, grid, dim3(WG_THREADS), 0,
                      (hipStream_t)stream,
                      (const __hip_bfloat16*)dy, (const __hip_bfloat16*)x, part,
                      bias_part, K, N, M, (int)k_slab);
