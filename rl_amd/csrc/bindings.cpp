@@ -241,7 +241,7 @@ std::vector<torch::Tensor> wgrad_splitk(torch::Tensor dy, torch::Tensor x,
   long K = dy.size(0);
   int N = (int)dy.size(1), M = (int)x.size(1);
   auto opts = dy.options().dtype(torch::kFloat32);
-  auto dw = torch::zeros({N, M}, opts);  // reduce commits via atomicAdd
+  auto dw = torch::empty({N, M}, opts);  // wave-reduce writes every element
   torch::Tensor db;
   if (want_bias) db = torch::empty({N}, opts);
   const long slabs = wgrad_slab_count(K);

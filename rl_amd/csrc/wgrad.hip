@@ -226,12 +226,11 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_mfma_kernel(
   }
 }
 
-// reduce the [slab, tile, elem] partials into dW [N, M].  A thread per
-// ELEMENT is latency-bound (N*M = 4096 threads x 128 dependent loads
-// measured 26-56 us): instead RED_GROUPS threads share each element,
-// each summing slabs/RED_GROUPS values (8x the memory parallelism) and
-// committing with one fp32 atomicAdd (8 contenders per address).
-#define RED_GROUPS 8
+// reduce the [slab, tile, elem] partials into dW [N, M]: one WAVE per
+// output element — each lane sums slabs/64 strided values (all loads
+// independent and in flight), then a 6-step shuffle tree; lane 0 writes
+// without atomics.  (Thread-per-element was latency-bound at 26-56 us;
+// thread-group + atomicAdd left only 2 waves/CU in flight.)
 __global__ void wgrad_reduce_kernel(const float* __restrict__ part,
                                     const float* __restrict__ bias_part,
                                     float* __restrict__ dw,
@@ -239,27 +238,30 @@ __global__ void wgrad_reduce_kernel(const float* __restrict__ part,
                                     int tiles_n, int tiles_m, int N, int M) {
   const long tile_elems = (long)64 * 64 * tiles_n * tiles_m;
   const long NM = (long)N * M;
-  for (long idx = blockIdx.x * blockDim.x + threadIdx.x;
-       idx < NM * RED_GROUPS; idx += (long)gridDim.x * blockDim.x) {
-    // consecutive threads take consecutive elements (coalesced reads)
-    const int g = (int)(idx / NM);
-    const long e = idx % NM;
+  const long waves = ((long)gridDim.x * blockDim.x) >> 6;
+  const long wave_id = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  for (long e = wave_id; e < NM; e += waves) {
     const int n = (int)(e / M), m = (int)(e % M);
     const int tn = n / 64, tm = m / 64;
     const long off = ((long)tn * tiles_m + tm) * (64 * 64) +
                      (long)(n % 64) * 64 + (m % 64);
     float acc = 0.f;
-    for (int s = g; s < slabs; s += RED_GROUPS)
+    for (int s = lane; s < slabs; s += 64)
       acc += part[(long)s * tile_elems + off];
-    atomicAdd(&dw[e], acc);
+#pragma unroll
+    for (int d = 32; d > 0; d >>= 1) acc += __shfl_down(acc, d);
+    if (lane == 0) dw[e] = acc;
   }
   if (dbias != nullptr) {
-    for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
-         n += gridDim.x * blockDim.x) {
+    const long bias_waves = waves;
+    for (long n = wave_id; n < N; n += bias_waves) {
       float acc = 0.f;
-      for (int s = 0; s < slabs; ++s)
+      for (int s = lane; s < slabs; s += 64)
         acc += bias_part[(long)s * (tiles_n * 64) + n];
-      dbias[n] = acc;
+#pragma unroll
+      for (int d = 32; d > 0; d >>= 1) acc += __shfl_down(acc, d);
+      if (lane == 0) dbias[n] = acc;
     }
   }
 }
@@ -287,7 +289,7 @@ extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
                      (hipStream_t)stream,
                      (const __hip_bfloat16*)dy, (const __hip_bfloat16*)x, part,
                      bias_part, K, N, M, (int)k_slab);
-  const long total = (long)N * M * RED_GROUPS;
+  const long total = (long)N * M * 64;  // one wave per element
   const int blocks = (int)min((total + 255) / 256, (long)2048);
   hipLaunchKernelGGL(wgrad_reduce_kernel, dim3(blocks), dim3(256), 0,
                      (hipStream_t)stream, part, bias_part, dw, dbias, slabs,
