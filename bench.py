@@ -135,11 +135,21 @@ def main():
         # minibatch backward wgrad via the split-K kernel (csrc/wgrad.hip):
         # hipBLASLt runs these skinny [H,16k]x[16k,H] reductions on one
         # workgroup (measured 101us, 12.9% of the step in the r11 profile)
-        from rl_amd.ops import HAS_HIP_EXT, convert_linears_to_splitk
+        from rl_amd.ops import (
+            HAS_HIP_EXT,
+            convert_linears_to_splitk,
+            enable_splitk_bf16_cache,
+            refresh_splitk_caches,
+        )
 
         if HAS_HIP_EXT:
             convert_linears_to_splitk(actor)
             convert_linears_to_splitk(critic)
+            # one weight cast per layer per optimizer step instead of a
+            # re-cast on every minibatch call (autocast cache is off
+            # under graph capture; r29 profile: ~130 cast kernels/step)
+            enable_splitk_bf16_cache(actor)
+            enable_splitk_bf16_cache(critic)
         else:
             args.splitk = False
 
@@ -330,6 +340,8 @@ def main():
                                 off += k
                 torch.nn.utils.clip_grad_norm_(params, 1.0)
                 optim.step()
+                if args.splitk and cuda:
+                    refresh_splitk_caches(actor, critic)
 
     def one_step():
         rollout()

@@ -323,14 +323,82 @@ class _SplitKLinearFn(torch.autograd.Function):
         return dx, dw, db
 
 
+class _SplitKLinearCachedFn(torch.autograd.Function):
+    """Variant taking an explicitly-cached bf16 weight: under hipGraph
+    capture `autocast(cache_enabled=False)` re-casts every weight on
+    every call (~130 cast kernels per PPO step in the r29 profile);
+    with the cache the cast runs ONCE per step, and the wgrad kernel's
+    fp32 output feeds the fp32 master parameter directly (no bf16
+    round-trip on the gradient)."""
+
+    @staticmethod
+    def forward(ctx, x, w_bf, weight, b_bf, bias):
+        x = x.contiguous()
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        ctx.save_for_backward(x, w_bf)
+        ctx.has_bias = bias is not None
+        with torch.autocast("cuda", enabled=False):
+            return torch.nn.functional.linear(x, w_bf, b_bf)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w_bf = ctx.saved_tensors
+        dy = dy.contiguous()
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        dx = (dy @ w_bf).reshape(x.shape)
+        outs = _C.wgrad_splitk(dy2, x2, ctx.has_bias)
+        dw = outs[0]  # fp32 — matches the fp32 master weight
+        db = outs[1] if ctx.has_bias else None
+        return dx, None, dw, None, db
+
+
 class SplitKLinear(torch.nn.Linear):
     """nn.Linear whose backward uses the split-K wgrad kernel on
-    bf16/cuda inputs (eager semantics elsewhere)."""
+    bf16/cuda inputs (eager semantics elsewhere).  With
+    :meth:`enable_bf16_cache` the bf16 weight cast is amortized to once
+    per training step (call :func:`refresh_splitk_caches` inside the
+    step, after the optimizer update)."""
+
+    _bf16_cache: bool = False
+
+    def enable_bf16_cache(self):
+        self._bf16_cache = True
+        self.register_buffer("weight_bf16", self.weight.detach().to(torch.bfloat16))
+        if self.bias is not None:
+            self.register_buffer("bias_bf16", self.bias.detach().to(torch.bfloat16))
+        return self
+
+    def refresh_bf16_cache_(self):
+        self.weight_bf16.copy_(self.weight.detach())
+        if self.bias is not None:
+            self.bias_bf16.copy_(self.bias.detach())
 
     def forward(self, x):
         if x.is_cuda and HAS_HIP_EXT:
+            if self._bf16_cache and torch.is_autocast_enabled():
+                return _SplitKLinearCachedFn.apply(
+                    x, self.weight_bf16, self.weight,
+                    getattr(self, "bias_bf16", None), self.bias,
+                )
             return _SplitKLinearFn.apply(x, self.weight, self.bias)
         return super().forward(x)
+
+
+def enable_splitk_bf16_cache(module: torch.nn.Module) -> None:
+    for m in module.modules():
+        if isinstance(m, SplitKLinear):
+            m.enable_bf16_cache()
+
+
+def refresh_splitk_caches(*modules: torch.nn.Module) -> None:
+    """Re-cast all cached bf16 weights (one cast kernel per layer) —
+    call once per training step after the optimizer update."""
+    for module in modules:
+        for m in module.modules():
+            if isinstance(m, SplitKLinear) and m._bf16_cache:
+                m.refresh_bf16_cache_()
 
 
 def convert_linears_to_splitk(module: torch.nn.Module) -> torch.nn.Module:
