@@ -40,6 +40,11 @@ void launch_fused_actor(const float*, const float*, const float*, const float*,
 int lstm_fused_lds_bytes(int);
 int wgrad_slab_count(long);
 int synthetic_env_step_lds_bytes(int, int);
+void launch_tanh_normal_logprob_fwd(const float*, const float*, const float*,
+                                    float*, int, int, void*);
+void launch_tanh_normal_logprob_bwd(const float*, const float*, const float*,
+                                    const float*, float*, float*, int, int,
+                                    void*);
 void launch_synthetic_env_step(float*, const float*, const float*,
                                const float*, float*, float*, float*, float*,
                                bool*, const float*, long, long, long, int,
@@ -311,6 +316,39 @@ void synthetic_env_step_into(torch::Tensor state, torch::Tensor action,
       (long)next_obs.stride(0), (long)reward.stride(0),
       (long)action.stride(0), Bn, S, Aact, (float)max_steps, (void*)stream);
 }
+// Fused TanhNormal log-prob (PPO ratio; action is data).
+torch::Tensor tanh_normal_logprob(torch::Tensor loc, torch::Tensor scale,
+                                  torch::Tensor action) {
+  TORCH_CHECK(loc.is_cuda() && loc.scalar_type() == torch::kFloat32,
+              "tanh_normal_logprob: fp32 cuda");
+  TORCH_CHECK(loc.is_contiguous() && scale.is_contiguous() &&
+                  action.is_contiguous(),
+              "contiguous inputs");
+  const int N = (int)loc.size(0), A = (int)loc.size(1);
+  auto logp = torch::empty({N}, loc.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_tanh_normal_logprob_fwd(loc.data_ptr<float>(),
+                                 scale.data_ptr<float>(),
+                                 action.data_ptr<float>(),
+                                 logp.data_ptr<float>(), N, A, (void*)stream);
+  return logp;
+}
+
+std::vector<torch::Tensor> tanh_normal_logprob_bwd(torch::Tensor loc,
+                                                   torch::Tensor scale,
+                                                   torch::Tensor action,
+                                                   torch::Tensor gout) {
+  const int N = (int)loc.size(0), A = (int)loc.size(1);
+  auto dloc = torch::empty_like(loc);
+  auto dscale = torch::empty_like(scale);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_tanh_normal_logprob_bwd(
+      loc.data_ptr<float>(), scale.data_ptr<float>(), action.data_ptr<float>(),
+      gout.contiguous().data_ptr<float>(), dloc.data_ptr<float>(),
+      dscale.data_ptr<float>(), N, A, (void*)stream);
+  return {dloc, dscale};
+}
+
 // Store-direct fused actor: action/log-prob written straight into
 // strided [B, T] rollout-store views.
 void fused_actor_into(torch::Tensor obs, torch::Tensor w1, torch::Tensor b1,
@@ -362,6 +400,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gae", &gae, "fused GAE scan (HIP)");
   m.def("synthetic_env_step", &synthetic_env_step,
         "fused synthetic-MuJoCo env transition (HIP)");
+  m.def("tanh_normal_logprob", &tanh_normal_logprob,
+        "fused TanhNormal log-prob forward (HIP)");
+  m.def("tanh_normal_logprob_bwd", &tanh_normal_logprob_bwd,
+        "fused TanhNormal log-prob backward (HIP)");
   m.def("fused_actor_into", &fused_actor_into,
         "store-direct fused actor (HIP)");
   m.def("synthetic_env_step_into", &synthetic_env_step_into,

@@ -1,0 +1,101 @@
+// Fused TanhNormal log-prob for PPO ratio computation (CDNA4, gfx950).
+//
+// The eager chain (atanh, z-score, log-scale, tanh-jacobian, sum) is
+// ~15 elementwise launches forward + ~20 backward per minibatch.  For
+// the PPO ratio the ACTION IS DATA (sampled during the rollout), so
+// the gradient only flows to loc/scale — both analytic:
+//
+//   u  = atanh(clamp(a))                       (constant)
+//   z  = (u - loc) / scale
+//   lp = sum_a [ -0.5 z^2 - log(scale) - 0.5 log(2pi) - log1p(-a^2) ]
+//   dlp/dloc   =  z / scale
+//   dlp/dscale = (z^2 - 1) / scale
+//
+// log1p(-a^2) is torch's TanhTransform jacobian 2(log2 - u -
+// softplus(-2u)) in closed form — matched to the same clamping.
+// Trivial bounds (-1, 1) only (the PPO bench's TanhNormal); the eager
+// path remains for general bounds.  Validated vs the eager
+// distribution in tests/test_ops.py.
+
+#include <hip/hip_runtime.h>
+
+#define LP_THREADS 256
+
+namespace {
+
+__device__ __forceinline__ float softplusf(float x) {
+  if (x > 20.f) return x;
+  return log1pf(__expf(x));
+}
+
+__global__ void tanh_normal_logprob_fwd(
+    const float* __restrict__ loc,    // [N, A]
+    const float* __restrict__ scale,  // [N, A]
+    const float* __restrict__ action, // [N, A]
+    float* __restrict__ logp,         // [N]
+    const int N, const int A) {
+  const float LOG_SQRT_2PI = 0.9189385332046727f;
+  const float LOG2 = 0.6931471805599453f;
+  const float lim = 1.0f - 1.1920929e-7f;
+  for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+       n += gridDim.x * blockDim.x) {
+    float lp = 0.f;
+#pragma unroll 2
+    for (int a = 0; a < A; ++a) {
+      const size_t i = (size_t)n * A + a;
+      float y = fminf(fmaxf(action[i], -lim), lim);
+      const float u = atanhf(y);
+      const float s = scale[i];
+      const float z = (u - loc[i]) / s;
+      lp += -0.5f * z * z - __logf(s) - LOG_SQRT_2PI
+            - 2.0f * (LOG2 - u - softplusf(-2.0f * u));
+    }
+    logp[n] = lp;
+  }
+}
+
+__global__ void tanh_normal_logprob_bwd(
+    const float* __restrict__ loc,
+    const float* __restrict__ scale,
+    const float* __restrict__ action,
+    const float* __restrict__ gout,   // [N] upstream grad
+    float* __restrict__ dloc,         // [N, A]
+    float* __restrict__ dscale,       // [N, A]
+    const int N, const int A) {
+  const float lim = 1.0f - 1.1920929e-7f;
+  const long total = (long)N * A;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const int n = (int)(i / A);
+    float y = fminf(fmaxf(action[i], -lim), lim);
+    const float u = atanhf(y);
+    const float s = scale[i];
+    const float z = (u - loc[i]) / s;
+    const float g = gout[n];
+    dloc[i] = g * z / s;
+    dscale[i] = g * (z * z - 1.0f) / s;
+  }
+}
+
+}  // namespace
+
+extern "C" void launch_tanh_normal_logprob_fwd(const float* loc,
+                                               const float* scale,
+                                               const float* action,
+                                               float* logp, int N, int A,
+                                               void* stream) {
+  const int blocks = min((N + LP_THREADS - 1) / LP_THREADS, 2048);
+  hipLaunchKernelGGL(tanh_normal_logprob_fwd, dim3(blocks), dim3(LP_THREADS),
+                     0, (hipStream_t)stream, loc, scale, action, logp, N, A);
+}
+
+extern "C" void launch_tanh_normal_logprob_bwd(
+    const float* loc, const float* scale, const float* action,
+    const float* gout, float* dloc, float* dscale, int N, int A,
+    void* stream) {
+  const long total = (long)N * A;
+  const int blocks = (int)min((total + LP_THREADS - 1) / LP_THREADS, (long)2048);
+  hipLaunchKernelGGL(tanh_normal_logprob_bwd, dim3(blocks), dim3(LP_THREADS),
+                     0, (hipStream_t)stream, loc, scale, action, gout, dloc,
+                     dscale, N, A);
+}

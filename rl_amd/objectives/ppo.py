@@ -93,11 +93,29 @@ class PPOLoss(LossModule):
         prev_log_prob = td.get(self.tensor_keys.sample_log_prob)
         dist = self.actor_network.get_dist(td.clone(False))
         action = td.get(self.tensor_keys.action)
-        log_prob = dist.log_prob(action)
+        log_prob = self._fast_log_prob(dist, action)
         if prev_log_prob.dim() > log_prob.dim():
             prev_log_prob = prev_log_prob.squeeze(-1)
         log_weight = log_prob - prev_log_prob.detach()
         return log_weight, dist, log_prob
+
+    @staticmethod
+    def _fast_log_prob(dist, action):
+        """Route TanhNormal(-1,1) ratio log-probs through the fused
+        HIP kernel (csrc/loss_ops.hip) on GPU — the action carries no
+        gradient in the PPO objective."""
+        from ..modules.distributions.continuous import TanhNormal
+
+        if (
+            action.is_cuda
+            and isinstance(dist, TanhNormal)
+            and not dist.non_trivial_bounds
+        ):
+            from .. import ops
+
+            if ops.HAS_HIP_EXT:
+                return ops.tanh_normal_logprob(dist.loc, dist.scale, action)
+        return dist.log_prob(action)
 
     def _entropy(self, dist) -> torch.Tensor:
         try:

@@ -419,3 +419,38 @@ def convert_linears_to_splitk(module: torch.nn.Module) -> torch.nn.Module:
         else:
             convert_linears_to_splitk(child)
     return module
+
+
+class _TanhNormalLogProbFn(torch.autograd.Function):
+    """Fused TanhNormal log-prob for the PPO ratio (csrc/loss_ops.hip):
+    the action is DATA (sampled during the rollout), so the gradient
+    only flows to loc/scale — both analytic, one kernel each way
+    (the eager chain is ~15 elementwise launches per direction)."""
+
+    @staticmethod
+    def forward(ctx, loc, scale, action):
+        loc = loc.contiguous()
+        scale = scale.contiguous()
+        action = action.contiguous().detach()
+        ctx.save_for_backward(loc, scale, action)
+        return _C.tanh_normal_logprob(loc.float(), scale.float(), action.float())
+
+    @staticmethod
+    def backward(ctx, gout):
+        loc, scale, action = ctx.saved_tensors
+        dloc, dscale = _C.tanh_normal_logprob_bwd(
+            loc.float(), scale.float(), action.float(), gout
+        )
+        return dloc.to(loc.dtype), dscale.to(scale.dtype), None
+
+
+def tanh_normal_logprob(loc, scale, action):
+    """Fused log π(a | TanhNormal(loc, scale)) with bounds (-1, 1),
+    summed over the last dim.  GPU-only; used by the PPO ratio fast
+    path (no gradient through the action)."""
+    _require_ext()
+    flat_loc = loc.reshape(-1, loc.shape[-1])
+    flat_scale = scale.reshape(-1, scale.shape[-1])
+    flat_act = action.reshape(-1, action.shape[-1])
+    out = _TanhNormalLogProbFn.apply(flat_loc, flat_scale, flat_act)
+    return out.reshape(loc.shape[:-1])

@@ -31,6 +31,7 @@ if with_hip:
         os.path.join(CSRC, "fused_actor.hip"),
         os.path.join(CSRC, "wgrad.hip"),
         os.path.join(CSRC, "env_step.hip"),
+        os.path.join(CSRC, "loss_ops.hip"),
     ]
     define_macros.append(("RL_AMD_WITH_HIP", None))
     ext_cls = cpp_extension.CUDAExtension

@@ -406,3 +406,34 @@ class TestStoreDirectEnvStep:
         assert not done[:, 2].any()
         assert (env._t == 1).all()
         assert torch.allclose(env._state, nxt[:, 2], atol=1e-6)
+
+
+@pytest.mark.gpu
+class TestTanhNormalLogProb:
+    def test_matches_eager_with_grads(self):
+        from rl_amd import ops
+        from rl_amd.modules import TanhNormal
+
+        torch.manual_seed(0)
+        N, A = 4096, 6
+        loc = torch.randn(N, A, device="cuda", requires_grad=True)
+        scale = (torch.rand(N, A, device="cuda") * 0.9 + 0.1).requires_grad_()
+        with torch.no_grad():
+            action = TanhNormal(loc, scale).sample()
+        lp_f = ops.tanh_normal_logprob(loc, scale, action)
+        lp_e = TanhNormal(loc, scale).log_prob(action)
+        assert torch.allclose(lp_f, lp_e, rtol=1e-4, atol=1e-4), (
+            (lp_f - lp_e).abs().max().item()
+        )
+        g = torch.randn_like(lp_f)
+        dl_f, ds_f = torch.autograd.grad(lp_f, (loc, scale), g, retain_graph=True)
+        loc2 = loc.detach().clone().requires_grad_()
+        scale2 = scale.detach().clone().requires_grad_()
+        lp_e2 = TanhNormal(loc2, scale2).log_prob(action)
+        dl_e, ds_e = torch.autograd.grad(lp_e2, (loc2, scale2), g)
+        assert torch.allclose(dl_f, dl_e, rtol=1e-3, atol=1e-3), (
+            (dl_f - dl_e).abs().max().item()
+        )
+        assert torch.allclose(ds_f, ds_e, rtol=1e-3, atol=1e-3), (
+            (ds_f - ds_e).abs().max().item()
+        )
