@@ -45,6 +45,11 @@ void launch_tanh_normal_logprob_fwd(const float*, const float*, const float*,
 void launch_tanh_normal_logprob_bwd(const float*, const float*, const float*,
                                     const float*, float*, float*, int, int,
                                     void*);
+void launch_tanh_normal_entropy_fwd(const float*, const float*, const float*,
+                                    float*, int, int, void*);
+void launch_tanh_normal_entropy_bwd(const float*, const float*, const float*,
+                                    const float*, float*, float*, int, int,
+                                    void*);
 void launch_synthetic_env_step(float*, const float*, const float*,
                                const float*, float*, float*, float*, float*,
                                bool*, const float*, long, long, long, int,
@@ -349,6 +354,33 @@ std::vector<torch::Tensor> tanh_normal_logprob_bwd(torch::Tensor loc,
   return {dloc, dscale};
 }
 
+torch::Tensor tanh_normal_entropy(torch::Tensor loc, torch::Tensor scale,
+                                  torch::Tensor eps) {
+  const int N = (int)loc.size(0), A = (int)loc.size(1);
+  auto ent = torch::empty({N}, loc.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_tanh_normal_entropy_fwd(loc.data_ptr<float>(),
+                                 scale.data_ptr<float>(),
+                                 eps.data_ptr<float>(), ent.data_ptr<float>(),
+                                 N, A, (void*)stream);
+  return ent;
+}
+
+std::vector<torch::Tensor> tanh_normal_entropy_bwd(torch::Tensor loc,
+                                                   torch::Tensor scale,
+                                                   torch::Tensor eps,
+                                                   torch::Tensor gout) {
+  const int N = (int)loc.size(0), A = (int)loc.size(1);
+  auto dloc = torch::empty_like(loc);
+  auto dscale = torch::empty_like(scale);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_tanh_normal_entropy_bwd(
+      loc.data_ptr<float>(), scale.data_ptr<float>(), eps.data_ptr<float>(),
+      gout.contiguous().data_ptr<float>(), dloc.data_ptr<float>(),
+      dscale.data_ptr<float>(), N, A, (void*)stream);
+  return {dloc, dscale};
+}
+
 // Store-direct fused actor: action/log-prob written straight into
 // strided [B, T] rollout-store views.
 void fused_actor_into(torch::Tensor obs, torch::Tensor w1, torch::Tensor b1,
@@ -400,6 +432,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gae", &gae, "fused GAE scan (HIP)");
   m.def("synthetic_env_step", &synthetic_env_step,
         "fused synthetic-MuJoCo env transition (HIP)");
+  m.def("tanh_normal_entropy", &tanh_normal_entropy,
+        "fused reparameterized TanhNormal MC entropy (HIP)");
+  m.def("tanh_normal_entropy_bwd", &tanh_normal_entropy_bwd,
+        "fused entropy backward (HIP)");
   m.def("tanh_normal_logprob", &tanh_normal_logprob,
         "fused TanhNormal log-prob forward (HIP)");
   m.def("tanh_normal_logprob_bwd", &tanh_normal_logprob_bwd,

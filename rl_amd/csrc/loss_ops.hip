@@ -77,7 +77,69 @@ __global__ void tanh_normal_logprob_bwd(
   }
 }
 
+// MC entropy of TanhNormal with reparameterized gradients: for the
+// sample x = tanh(loc + scale*eps) (eps ~ N(0,1) passed in), the
+// estimate per element is  -lp(x) = 0.5 eps^2 + log(scale) +
+// 0.5 log(2pi) + log(1 - x^2),  and the TOTAL reparam gradients are
+// analytic:  d(-lp)/dloc = -2x,  d(-lp)/dscale = 1/scale - 2x*eps.
+__global__ void tanh_normal_entropy_fwd(
+    const float* __restrict__ loc, const float* __restrict__ scale,
+    const float* __restrict__ eps, float* __restrict__ ent, const int N,
+    const int A) {
+  const float LOG_SQRT_2PI = 0.9189385332046727f;
+  for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+       n += gridDim.x * blockDim.x) {
+    float acc = 0.f;
+#pragma unroll 2
+    for (int a = 0; a < A; ++a) {
+      const size_t i = (size_t)n * A + a;
+      const float e = eps[i];
+      const float s = scale[i];
+      const float y = tanhf(loc[i] + s * e);
+      acc += 0.5f * e * e + __logf(s) + LOG_SQRT_2PI + log1pf(-y * y);
+    }
+    ent[n] = acc;
+  }
+}
+
+__global__ void tanh_normal_entropy_bwd(
+    const float* __restrict__ loc, const float* __restrict__ scale,
+    const float* __restrict__ eps, const float* __restrict__ gout,
+    float* __restrict__ dloc, float* __restrict__ dscale, const int N,
+    const int A) {
+  const long total = (long)N * A;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const int n = (int)(i / A);
+    const float e = eps[i];
+    const float s = scale[i];
+    const float y = tanhf(loc[i] + s * e);
+    const float g = gout[n];
+    dloc[i] = g * (-2.0f * y);
+    dscale[i] = g * (1.0f / s - 2.0f * y * e);
+  }
+}
+
 }  // namespace
+
+extern "C" void launch_tanh_normal_entropy_fwd(const float* loc,
+                                               const float* scale,
+                                               const float* eps, float* ent,
+                                               int N, int A, void* stream) {
+  const int blocks = min((N + LP_THREADS - 1) / LP_THREADS, 2048);
+  hipLaunchKernelGGL(tanh_normal_entropy_fwd, dim3(blocks), dim3(LP_THREADS),
+                     0, (hipStream_t)stream, loc, scale, eps, ent, N, A);
+}
+
+extern "C" void launch_tanh_normal_entropy_bwd(
+    const float* loc, const float* scale, const float* eps, const float* gout,
+    float* dloc, float* dscale, int N, int A, void* stream) {
+  const long total = (long)N * A;
+  const int blocks = (int)min((total + LP_THREADS - 1) / LP_THREADS, (long)2048);
+  hipLaunchKernelGGL(tanh_normal_entropy_bwd, dim3(blocks), dim3(LP_THREADS),
+                     0, (hipStream_t)stream, loc, scale, eps, gout, dloc,
+                     dscale, N, A);
+}
 
 extern "C" void launch_tanh_normal_logprob_fwd(const float* loc,
                                                const float* scale,

@@ -118,6 +118,19 @@ class PPOLoss(LossModule):
         return dist.log_prob(action)
 
     def _entropy(self, dist) -> torch.Tensor:
+        from ..modules.distributions.continuous import TanhNormal
+
+        if (
+            isinstance(dist, TanhNormal)
+            and not dist.non_trivial_bounds
+            and dist.loc.is_cuda
+            and self.samples_mc_entropy == 1
+        ):
+            from .. import ops
+
+            if ops.HAS_HIP_EXT:
+                # fused reparameterized MC estimate (csrc/loss_ops.hip)
+                return ops.tanh_normal_entropy(dist.loc, dist.scale)
         try:
             ent = dist.entropy()
         except NotImplementedError:

@@ -454,3 +454,39 @@ def tanh_normal_logprob(loc, scale, action):
     flat_act = action.reshape(-1, action.shape[-1])
     out = _TanhNormalLogProbFn.apply(flat_loc, flat_scale, flat_act)
     return out.reshape(loc.shape[:-1])
+
+
+class _TanhNormalEntropyFn(torch.autograd.Function):
+    """Fused reparameterized MC entropy of TanhNormal (one sample):
+    analytic total gradients d(-lp)/dloc = -2x, d(-lp)/dscale =
+    1/scale - 2x*eps (csrc/loss_ops.hip)."""
+
+    @staticmethod
+    def forward(ctx, loc, scale, eps):
+        loc = loc.contiguous()
+        scale = scale.contiguous()
+        eps = eps.contiguous()
+        ctx.save_for_backward(loc, scale, eps)
+        return _C.tanh_normal_entropy(loc.float(), scale.float(), eps.float())
+
+    @staticmethod
+    def backward(ctx, gout):
+        loc, scale, eps = ctx.saved_tensors
+        dloc, dscale = _C.tanh_normal_entropy_bwd(
+            loc.float(), scale.float(), eps.float(), gout
+        )
+        return dloc.to(loc.dtype), dscale.to(scale.dtype), None
+
+
+def tanh_normal_entropy(loc, scale, eps=None):
+    """Single-sample reparameterized entropy estimate of
+    TanhNormal(loc, scale) with bounds (-1, 1), summed over the last
+    dim.  ``eps`` defaults to a fresh standard-normal draw."""
+    _require_ext()
+    if eps is None:
+        eps = torch.randn_like(loc)
+    flat_loc = loc.reshape(-1, loc.shape[-1])
+    flat_scale = scale.reshape(-1, scale.shape[-1])
+    flat_eps = eps.reshape(-1, eps.shape[-1])
+    out = _TanhNormalEntropyFn.apply(flat_loc, flat_scale, flat_eps)
+    return out.reshape(loc.shape[:-1])

@@ -437,3 +437,35 @@ class TestTanhNormalLogProb:
         assert torch.allclose(ds_f, ds_e, rtol=1e-3, atol=1e-3), (
             (ds_f - ds_e).abs().max().item()
         )
+
+
+@pytest.mark.gpu
+class TestTanhNormalEntropy:
+    def test_matches_eager_reparam_grads(self):
+        from rl_amd import ops
+        from rl_amd.modules import TanhNormal
+
+        torch.manual_seed(0)
+        N, A = 4096, 6
+        loc = torch.randn(N, A, device="cuda", requires_grad=True)
+        scale = (torch.rand(N, A, device="cuda") * 0.9 + 0.1).requires_grad_()
+        eps = torch.randn(N, A, device="cuda")
+        ent_f = ops.tanh_normal_entropy(loc, scale, eps)
+        # eager: same sample via explicit reparameterization
+        loc2 = loc.detach().clone().requires_grad_()
+        scale2 = scale.detach().clone().requires_grad_()
+        dist = TanhNormal(loc2, scale2)
+        x = torch.tanh(loc2 + scale2 * eps)
+        ent_e = -dist.log_prob(x)
+        assert torch.allclose(ent_f, ent_e, rtol=1e-3, atol=1e-3), (
+            (ent_f - ent_e).abs().max().item()
+        )
+        g = torch.randn_like(ent_f)
+        dl_f, ds_f = torch.autograd.grad(ent_f, (loc, scale), g)
+        dl_e, ds_e = torch.autograd.grad(ent_e, (loc2, scale2), g)
+        assert torch.allclose(dl_f, dl_e, rtol=1e-3, atol=2e-3), (
+            (dl_f - dl_e).abs().max().item()
+        )
+        assert torch.allclose(ds_f, ds_e, rtol=1e-3, atol=2e-3), (
+            (ds_f - ds_e).abs().max().item()
+        )
