@@ -12,6 +12,7 @@ from .distributed import DistributedCollector, DistributedSyncCollector
 from .evaluator import Evaluator
 from .llm import LLMCollector
 from .weight_update import (
+    RayWeightUpdater,
     MultiProcessedWeightUpdater,
     RemoteModuleWeightUpdater,
     VanillaWeightUpdater,

@@ -18,6 +18,7 @@ __all__ = [
     "VanillaWeightUpdater",
     "MultiProcessedWeightUpdater",
     "RemoteModuleWeightUpdater",
+    "RayWeightUpdater",
 ]
 
 
@@ -84,3 +85,19 @@ class MultiProcessedWeightUpdater(WeightUpdaterBase):
 
 class RemoteModuleWeightUpdater(MultiProcessedWeightUpdater):
     """Alias for parity: remote-module flavored pipe updater."""
+
+
+class RayWeightUpdater(WeightUpdaterBase):
+    """Ray-actor weight pushes (reference weight_update.py RayWeightUpdater)
+    — gated: ray is not installed in this image."""
+
+    def __init__(self, *args, **kwargs):
+        import importlib.util
+
+        if importlib.util.find_spec("ray") is None:
+            raise ImportError(
+                "RayWeightUpdater requires the `ray` package, which is not "
+                "installed in this image. Use MultiProcessedWeightUpdater or "
+                "the RCCL DistributedWeightSyncScheme instead."
+            )
+        raise NotImplementedError("ray backend scaffolding")
