@@ -119,6 +119,9 @@ def main():
     torch.manual_seed(7 + rank)
 
     B, T, A = args.envs, args.unroll, 6
+    # bf16 compute (no fp32 MFMA on CDNA4; see bench_sac r32 finding)
+    autocast = torch.autocast("cuda", dtype=torch.bfloat16, enabled=cuda,
+                              cache_enabled=False)
     env = SyntheticAtariEnv(B, device)
     net = ImpalaNet(A, device=device)
     # behavior policy: slightly stale copy of the learner (IMPALA lag)
@@ -138,14 +141,14 @@ def main():
 
     def rollout_body():
         nonlocal obs
-        with torch.no_grad():
+        with torch.no_grad(), autocast:
             for t in range(T):
                 logits, _ = behavior(obs)
                 dist = torch.distributions.Categorical(logits=logits, validate_args=False)
                 a = dist.sample()
                 frames[:, t].copy_(obs)
                 actions[:, t].copy_(a)
-                log_mu[:, t].copy_(dist.log_prob(a))
+                log_mu[:, t].copy_(dist.log_prob(a).float())
                 _, r, d = env.step(a)  # obs aliases env.state (in-place)
                 rewards[:, t].copy_(r)
                 dones[:, t].copy_(d)
@@ -173,13 +176,17 @@ def main():
         rollout()
         # learner: recompute pi under current weights over the unroll
         flat = frames.reshape(B * T, *env.frame_shape)
-        logits, values = net(flat)
+        with autocast:
+            logits, values = net(flat)
+        logits = logits.float()
+        values = values.float()
         logits = logits.reshape(B, T, A)
         values = values.reshape(B, T, 1)
         dist = torch.distributions.Categorical(logits=logits, validate_args=False)
         log_pi = dist.log_prob(actions)
-        with torch.no_grad():
+        with torch.no_grad(), autocast:
             _, next_v_last = net(obs)
+        next_v_last = next_v_last.float()
         next_values = torch.cat([values[:, 1:], next_v_last.reshape(B, 1, 1)], 1)
         adv, vs = ops.vtrace(
             gamma,
@@ -242,7 +249,7 @@ def main():
                     "higher_is_better": True,
                     "scaling": "weak",
                     "vs_baseline": None,
-                    "dtype": "fp32",
+                    "dtype": "bf16" if cuda else "fp32",
                     "data": "synthetic",
                     "config": {
                         "model": "impala_shallow_pong",
