@@ -69,6 +69,8 @@ def parse_args():
     p.add_argument("--splitk", dest="splitk", action="store_true",
                    help="split-K HIP wgrad kernel in the update backward")
     p.add_argument("--no-splitk", dest="splitk", action="store_false")
+    p.add_argument("--track-reward", action="store_true",
+                   help="print mean store reward every 20 steps (learning sanity)")
     p.set_defaults(graph=True, fused_actor=True, full_graph=True, splitk=True)
     return p.parse_args()
 
@@ -388,8 +390,11 @@ def main():
     if cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for it in range(args.steps):
         step_fn()
+        if args.track_reward and it % 20 == 0:
+            r = store.get(("next", "reward")).float().mean().item()
+            print(f"[reward] step {it}: {r:.4f}", file=sys.stderr)
     if cuda:
         torch.cuda.synchronize()
     if distributed:
