@@ -1,0 +1,177 @@
+"""Transforms batch 4: macro-action primitives, running stats, gated
+pretrained encoders / video / ray transforms, and name-parity aliases.
+
+Reference: pytorch/rl torchrl/envs/transforms/_primitive.py:47-199
+(MacroPrimitive, MacroAction, TargetMacroAction,
+MacroPrimitiveTransform), _normalization.py (RunningMeanStd),
+r3m.py/vip.py/vc1.py (pretrained visual encoders — gated: torchvision
+and model weights are unavailable offline), _video.py
+(DecodeVideoTransform — gated on torchvision/av), module.py
+(RayModuleTransform — gated on ray).
+"""
+from __future__ import annotations
+
+import importlib.util
+from enum import IntEnum
+from typing import Optional
+
+import torch
+
+from ...tensordict import TensorDict, TensorDictBase
+from ._base import Transform
+
+__all__ = [
+    "MacroPrimitive",
+    "MacroAction",
+    "TargetMacroAction",
+    "MacroPrimitiveTransform",
+    "RunningMeanStd",
+    "R3MTransform",
+    "VIPTransform",
+    "VIPRewardTransform",
+    "VC1Transform",
+    "DecodeVideoTransform",
+    "RayModuleTransform",
+    "RayTransform",
+]
+
+
+class MacroPrimitive(IntEnum):
+    """Primitive ids for :class:`MacroPrimitiveTransform`: hold the
+    current action (WAIT) or interpolate toward a target (MOVE)."""
+
+    WAIT = 0
+    MOVE = 1
+
+
+class MacroAction:
+    """Structured macro action: primitive mode + expansion durations."""
+
+    def __init__(self, mode: torch.Tensor, steps: torch.Tensor, settle_steps: Optional[torch.Tensor] = None):
+        self.mode = torch.as_tensor(mode)
+        self.steps = torch.as_tensor(steps)
+        self.settle_steps = (
+            torch.as_tensor(settle_steps)
+            if settle_steps is not None
+            else torch.zeros_like(self.steps)
+        )
+
+    @classmethod
+    def wait(cls, steps: int = 1):
+        return cls(torch.tensor(int(MacroPrimitive.WAIT)), torch.tensor(steps))
+
+
+class TargetMacroAction(MacroAction):
+    """Macro action with a low-level action-space target."""
+
+    def __init__(self, mode, steps, target: torch.Tensor, settle_steps=None):
+        super().__init__(mode, steps, settle_steps)
+        self.target = torch.as_tensor(target)
+
+    @classmethod
+    def move(cls, target: torch.Tensor, steps: int = 1, settle_steps: int = 0):
+        return cls(
+            torch.tensor(int(MacroPrimitive.MOVE)),
+            torch.tensor(steps),
+            target,
+            torch.tensor(settle_steps),
+        )
+
+
+class MacroPrimitiveTransform(Transform):
+    """Expand a macro action into a low-level action sequence on the
+    inverse path (reference _primitive.py:199): MOVE linearly
+    interpolates from the last low-level action to the target over
+    ``steps`` (+ ``settle_steps`` held repeats); WAIT holds.  The
+    expanded ``[..., T, A]`` sequence is written to
+    ``out_key`` for a :class:`~rl_amd.envs.transforms.MultiAction`-style
+    executor."""
+
+    def __init__(self, action_key: str = "macro_action", out_key: str = "action", action_dim: Optional[int] = None):
+        super().__init__(in_keys_inv=[action_key], out_keys_inv=[out_key])
+        self.action_key = action_key
+        self.out_key = out_key
+        self.action_dim = action_dim
+        self._last_action: Optional[torch.Tensor] = None
+
+    def expand_macro(self, macro: MacroAction, start: torch.Tensor) -> torch.Tensor:
+        steps = int(macro.steps)
+        settle = int(macro.settle_steps)
+        if int(macro.mode) == int(MacroPrimitive.WAIT):
+            seq = start.unsqueeze(-2).expand(*start.shape[:-1], steps + settle, start.shape[-1])
+            return seq.clone()
+        target = macro.target.to(start.device, start.dtype)
+        w = torch.linspace(1.0 / steps, 1.0, steps, device=start.device)
+        w = w.reshape(*(1,) * (start.dim() - 1), steps, 1)
+        seq = start.unsqueeze(-2) * (1 - w) + target.unsqueeze(-2) * w
+        if settle:
+            tail = target.unsqueeze(-2).expand(*target.shape[:-1], settle, target.shape[-1])
+            seq = torch.cat([seq, tail], dim=-2)
+        return seq
+
+    def _inv_call(self, td: TensorDictBase) -> TensorDictBase:
+        macro = td.get_non_tensor(self.action_key, None)
+        if macro is None:
+            return td
+        if self._last_action is None:
+            ref = macro.target if isinstance(macro, TargetMacroAction) else None
+            if ref is None:
+                raise RuntimeError("first macro must be a TargetMacroAction")
+            self._last_action = torch.zeros_like(ref)
+        seq = self.expand_macro(macro, self._last_action)
+        self._last_action = seq[..., -1, :].clone()
+        td.set(self.out_key, seq)
+        return td
+
+
+class RunningMeanStd:
+    """Numerically-stable running mean/var accumulator (reference
+    _normalization.py RunningMeanStd) — the statistic engine behind
+    VecNorm-style transforms."""
+
+    def __init__(self, shape=(), epsilon: float = 1e-4, device=None):
+        self.mean = torch.zeros(shape, device=device)
+        self.var = torch.ones(shape, device=device)
+        self.count = epsilon
+
+    def update(self, x: torch.Tensor) -> None:
+        batch_mean = x.mean(0)
+        batch_var = x.var(0, unbiased=False)
+        n = x.shape[0]
+        delta = batch_mean - self.mean
+        tot = self.count + n
+        self.mean = self.mean + delta * n / tot
+        m_a = self.var * self.count
+        m_b = batch_var * n
+        self.var = (m_a + m_b + delta.pow(2) * self.count * n / tot) / tot
+        self.count = tot
+
+    def normalize(self, x: torch.Tensor, eps: float = 1e-8) -> torch.Tensor:
+        return (x - self.mean) / (self.var + eps).sqrt()
+
+
+def _gated(name: str, pkg: str, extra: str = ""):
+    class _Gated(Transform):
+        def __init__(self, *args, **kwargs):
+            if importlib.util.find_spec(pkg) is None:
+                raise ImportError(
+                    f"{name} requires the `{pkg}` package, which is not "
+                    f"installed in this image.{extra}"
+                )
+            raise NotImplementedError(
+                f"{name}: pretrained weights are not downloadable offline"
+            )
+
+    _Gated.__name__ = name
+    _Gated.__qualname__ = name
+    return _Gated
+
+
+# pretrained visual encoders: need torchvision AND weight downloads
+R3MTransform = _gated("R3MTransform", "torchvision", " (and R3M weights need network access)")
+VIPTransform = _gated("VIPTransform", "torchvision", " (and VIP weights need network access)")
+VIPRewardTransform = _gated("VIPRewardTransform", "torchvision")
+VC1Transform = _gated("VC1Transform", "torchvision", " (and VC-1 weights need network access)")
+DecodeVideoTransform = _gated("DecodeVideoTransform", "torchvision")
+RayModuleTransform = _gated("RayModuleTransform", "ray")
+RayTransform = _gated("RayTransform", "ray")

@@ -299,3 +299,55 @@ class TestModuleTimerMean:
         )
         t._inv_call(act)
         assert torch.equal(act["action"], torch.ones(2, 3))
+
+
+class TestBatch4:
+    def test_macro_primitive_expand(self):
+        from rl_amd.envs.transforms import MacroPrimitiveTransform, TargetMacroAction
+        from rl_amd.tensordict import TensorDict
+
+        t = MacroPrimitiveTransform()
+        td = TensorDict({}, batch_size=[])
+        td.set_non_tensor("macro_action", TargetMacroAction.move(torch.ones(3), steps=4))
+        t._inv_call(td)
+        seq = td["action"]
+        assert seq.shape == (4, 3)
+        assert torch.allclose(seq[-1], torch.ones(3))
+        assert torch.allclose(seq[0], torch.full((3,), 0.25))  # linear ramp from 0
+        # WAIT holds the last action
+        from rl_amd.envs.transforms import MacroAction
+
+        td2 = TensorDict({}, batch_size=[])
+        td2.set_non_tensor("macro_action", MacroAction.wait(steps=2))
+        t._inv_call(td2)
+        assert torch.allclose(td2["action"], torch.ones(2, 3))
+
+    def test_running_mean_std(self):
+        from rl_amd.envs.transforms import RunningMeanStd
+
+        torch.manual_seed(0)
+        rms = RunningMeanStd(shape=(3,))
+        data = torch.randn(1000, 3) * 2 + 5
+        for chunk in data.split(100):
+            rms.update(chunk)
+        assert torch.allclose(rms.mean, data.mean(0), atol=0.05)
+        normed = rms.normalize(data)
+        assert normed.mean().abs() < 0.05 and (normed.std(0) - 1).abs().max() < 0.1
+
+    def test_gated_encoders(self):
+        import importlib.util
+
+        from rl_amd.envs.transforms import R3MTransform, RayModuleTransform
+
+        with pytest.raises((ImportError, NotImplementedError)):
+            R3MTransform("resnet18")
+        if importlib.util.find_spec("ray") is None:
+            with pytest.raises(ImportError, match="ray"):
+                RayModuleTransform()
+
+    def test_aliases(self):
+        from rl_amd.envs import transforms as T
+
+        assert T.Hash is T.HashTransform
+        assert T.Tokenizer is T.TokenizerTransform
+        assert T.RandomTruncationTransform is T.RandomTruncation
