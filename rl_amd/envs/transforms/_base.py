@@ -17,7 +17,7 @@ from ...data.tensor_specs import Composite, TensorSpec
 from ...tensordict import TensorDict, TensorDictBase, unravel_key
 from ..common import EnvBase
 
-__all__ = ["Transform", "Compose", "TransformedEnv"]
+__all__ = ["Transform", "Compose", "TransformedEnv", "AutoResetEnv"]
 
 
 class Transform(nn.Module):
@@ -376,3 +376,15 @@ class TransformedEnv(EnvBase):
 
     def __repr__(self):
         return f"TransformedEnv(env={self.base_env}, transform={self.transform})"
+
+
+class AutoResetEnv(TransformedEnv):
+    """TransformedEnv variant for natively auto-resetting base envs
+    (reference _base.py:2094): pairs with
+    :class:`~rl_amd.envs.transforms.AutoResetTransform`, which buffers
+    the post-done reset observation the lib delivers in-step."""
+
+    def _reset(self, tensordict=None, **kwargs):
+        if tensordict is not None:
+            tensordict = tensordict.select("_reset", strict=False)
+        return super().reset(tensordict, **kwargs)
