@@ -13,7 +13,7 @@ from .utils import (
     distance_loss,
     hold_out_net,
 )
-from .value import GAE, TD0Estimator, TD1Estimator, TDLambdaEstimator, ValueEstimatorBase, VTrace
+from .value import GAE, MultiAgentGAE, TD0Estimator, TD1Estimator, TDLambdaEstimator, ValueEstimatorBase, VTrace
 from .cql import CQLLoss, DiscreteCQLLoss
 from .iql import DiscreteIQLLoss, IQLLoss
 from .imitation import BCLoss, DTLoss, GAILLoss, OnlineDTLoss, RNDLoss

@@ -1,2 +1,2 @@
-from .advantages import GAE, TD0Estimator, TD1Estimator, TDLambdaEstimator, ValueEstimatorBase, VTrace
+from .advantages import GAE, MultiAgentGAE, TD0Estimator, TD1Estimator, TDLambdaEstimator, ValueEstimatorBase, VTrace
 from . import functional

@@ -299,6 +299,75 @@ class GAE(ValueEstimatorBase):
         return tensordict
 
 
+class MultiAgentGAE(GAE):
+    """GAE for per-agent values with team-level reward/done (reference
+    advantages.py:2378): when reward/done/terminated lack the agent
+    dimension of the ``[*B, T, n_agents, 1]`` value tensor, they are
+    broadcast along ``agent_dim`` before the standard recursion, so the
+    advantage/value_target match the per-agent value shape MAPPO/IPPO
+    losses expect.  Per-agent rewards pass through unchanged."""
+
+    def __init__(self, *, agent_dim: int = -2, **kwargs):
+        super().__init__(**kwargs)
+        self.agent_dim = agent_dim
+
+    def forward(self, tensordict: TensorDictBase, **kwargs) -> TensorDictBase:
+        keys = self.tensor_keys
+        value, next_value = self._call_value_nets(tensordict)
+        reward, done, terminated = self._get_done_terminated_reward(tensordict)
+        dim = self.agent_dim % value.ndim
+        n_agents = value.shape[dim]
+
+        def _bcast(x):
+            if x.ndim == value.ndim:
+                return x
+            xe = x.unsqueeze(dim)
+            return xe.expand(*xe.shape[:dim], n_agents, *xe.shape[dim + 1 :])
+
+        rewardb = _bcast(reward)
+        doneb = _bcast(done)
+        termb = _bcast(terminated)
+        # value layout [*B, T, A, 1] with dim = ndim-2: permute the agent
+        # dim out front, fold into the batch, run the (fused) recursion
+        vperm = value.movedim(dim, 0)
+        nvperm = next_value.movedim(dim, 0)
+        rperm = rewardb.movedim(dim, 0)
+        dperm = doneb.movedim(dim, 0)
+        tperm = termb.movedim(dim, 0)
+
+        shape = vperm.shape  # [A, *B, T, 1]
+        if vperm.is_cuda and self.vectorized and not self.differentiable:
+            from ... import ops
+
+            adv, vt = ops.gae(
+                self._gamma_float, self._lmbda_float,
+                vperm.reshape(-1, *shape[-2:]), nvperm.reshape(-1, *shape[-2:]),
+                rperm.reshape(-1, *shape[-2:]), dperm.reshape(-1, *shape[-2:]),
+                tperm.reshape(-1, *shape[-2:]),
+            )
+        else:
+            fn = (
+                F.vec_generalized_advantage_estimate
+                if self.vectorized
+                else F.generalized_advantage_estimate
+            )
+            adv, vt = fn(
+                self._gamma_float, self._lmbda_float,
+                vperm.reshape(-1, *shape[-2:]), nvperm.reshape(-1, *shape[-2:]),
+                rperm.reshape(-1, *shape[-2:]), dperm.reshape(-1, *shape[-2:]),
+                tperm.reshape(-1, *shape[-2:]),
+            )
+        adv = adv.reshape(shape).movedim(0, dim)
+        vt = vt.reshape(shape).movedim(0, dim)
+        if self.average_gae:
+            loc = adv.mean()
+            scale = adv.std().clamp_min(1e-6)
+            adv = (adv - loc) / scale
+        tensordict.set(keys.advantage, adv)
+        tensordict.set(keys.value_target, vt)
+        return tensordict
+
+
 class VTrace(ValueEstimatorBase):
     """V-trace off-policy correction (reference advantages.py:2484)."""
 

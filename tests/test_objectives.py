@@ -443,3 +443,64 @@ class TestPILCO:
         loss = ExponentialQuadraticCost()(td).get("loss_cost")
         loss.backward()
         assert m.grad is not None and torch.isfinite(m.grad).all()
+
+
+class TestMultiAgentGAE:
+    def test_team_reward_broadcasts(self):
+        from rl_amd.objectives import GAE, MultiAgentGAE
+        from rl_amd.tensordict import TensorDict
+
+        torch.manual_seed(0)
+        B, T, N = 3, 7, 4
+        td = TensorDict(
+            {
+                "state_value": torch.randn(B, T, N, 1),
+                "next": {
+                    "state_value": torch.randn(B, T, N, 1),
+                    "reward": torch.randn(B, T, 1),
+                    "done": torch.zeros(B, T, 1, dtype=torch.bool),
+                    "terminated": torch.zeros(B, T, 1, dtype=torch.bool),
+                },
+            },
+            batch_size=[B, T],
+        )
+        est = MultiAgentGAE(gamma=0.99, lmbda=0.95)
+        est(td)
+        adv = td.get("advantage")
+        assert adv.shape == (B, T, N, 1)
+        # agent a's advantage must equal single-agent GAE run on agent a
+        for a in range(N):
+            td_a = TensorDict(
+                {
+                    "state_value": td["state_value"][:, :, a],
+                    "next": {
+                        "state_value": td["next", "state_value"][:, :, a],
+                        "reward": td["next", "reward"],
+                        "done": td["next", "done"],
+                        "terminated": td["next", "terminated"],
+                    },
+                },
+                batch_size=[B, T],
+            )
+            GAE(gamma=0.99, lmbda=0.95)(td_a)
+            assert torch.allclose(adv[:, :, a], td_a["advantage"], atol=1e-5), a
+
+    def test_per_agent_reward_passthrough(self):
+        from rl_amd.objectives import MultiAgentGAE
+        from rl_amd.tensordict import TensorDict
+
+        B, T, N = 2, 5, 3
+        td = TensorDict(
+            {
+                "state_value": torch.randn(B, T, N, 1),
+                "next": {
+                    "state_value": torch.randn(B, T, N, 1),
+                    "reward": torch.randn(B, T, N, 1),
+                    "done": torch.zeros(B, T, N, 1, dtype=torch.bool),
+                    "terminated": torch.zeros(B, T, N, 1, dtype=torch.bool),
+                },
+            },
+            batch_size=[B, T],
+        )
+        MultiAgentGAE(gamma=0.9, lmbda=0.9)(td)
+        assert td["advantage"].shape == (B, T, N, 1)
