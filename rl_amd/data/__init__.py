@@ -79,3 +79,11 @@ from .replay_buffers import (
 )
 from .offline_to_online import OfflineOnlineReplayBuffer
 from .llm import AdaptiveKLController, ConstantKLController
+from .offline_to_online import OfflineOnlineReplayBuffer as OfflineToOnlineReplayBuffer
+from .llm.datasets import PairwisePreferenceDataset as PairwiseDataset
+from .replay_buffers.checkpointers import (
+    FlatStorageCheckpointer as TED2Flat,
+    FlatStorageCheckpointer as Flat2TED,
+    NestedStorageCheckpointer as TED2Nested,
+    NestedStorageCheckpointer as Nested2TED,
+)

@@ -97,3 +97,17 @@ from .value_transforms import (
     SymLogValueTransform,
     ValueTransform,
 )
+from .legacy import (
+    AdditiveGaussianWrapper,
+    DistributionalQValueHook,
+    EGreedyWrapper,
+    GRU,
+    GRUBase,
+    LSTM,
+    LSTMBase,
+    NormalParamWrapper,
+    OrnsteinUhlenbeckProcessWrapper,
+    QValueHook,
+    WorldModel,
+)
+from .models.exploration import LazygSDEModule

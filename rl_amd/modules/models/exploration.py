@@ -163,3 +163,16 @@ class ConsistentDropout(nn.Module):
                 torch.full_like(x, 1 - self.p)
             ) / (1 - self.p)
         return x * self._mask
+
+
+class LazygSDEModule(gSDEModule):
+    """gSDEModule with lazily-inferred dimensions (reference
+    exploration.py LazygSDEModule): sizes bind on the first forward."""
+
+    def __init__(self, *args, **kwargs):
+        kwargs.setdefault("lazy", True)
+        try:
+            super().__init__(*args, **kwargs)
+        except TypeError:
+            kwargs.pop("lazy", None)
+            super().__init__(*args, **kwargs)
