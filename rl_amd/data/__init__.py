@@ -87,3 +87,12 @@ from .replay_buffers.checkpointers import (
     NestedStorageCheckpointer as TED2Nested,
     NestedStorageCheckpointer as Nested2TED,
 )
+from .llm import (
+    PromptData,
+    PromptTensorDictTokenizer,
+    RewardData,
+    RolloutFromModel,
+    TensorDictTokenizer,
+    TokenizedDatasetLoader,
+    TopKRewardSelector,
+)

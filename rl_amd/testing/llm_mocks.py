@@ -23,11 +23,15 @@ class ByteTokenizer:
     pad_token_id = 256
     eos_token_id = 257
 
-    def __call__(self, texts: Union[str, List[str]], return_tensors=None, padding=False):
+    def __call__(self, texts: Union[str, List[str]], return_tensors=None,
+                 padding=False, max_length=None, truncation=False):
         if isinstance(texts, str):
             texts = [texts]
-        seqs = [list(t.encode("utf-8"))[:512] for t in texts]
+        cutoff = max_length if (truncation and max_length) else 512
+        seqs = [list(t.encode("utf-8"))[:cutoff] for t in texts]
         max_len = max(len(s) for s in seqs)
+        if padding == "max_length" and max_length:
+            max_len = max_length
         ids, mask = [], []
         for s in seqs:
             pad = max_len - len(s)

@@ -230,3 +230,50 @@ class TestDistillation:
             out = DistillationLoss(actor, kl_direction=d)(td.clone(False))
             assert torch.isfinite(out.get("loss_distill"))
             assert out.get("loss_distill") > 0
+
+
+class TestRLHFDataUtils:
+    def test_tokenizers_and_prompt_data(self, tok):
+        from rl_amd.data import PromptData, PromptTensorDictTokenizer, TensorDictTokenizer
+
+        t = TensorDictTokenizer(tok, max_length=16)
+        td = t(["hello world", "hi"])
+        assert td["input_ids"].shape == (2, 16)
+        pt = PromptTensorDictTokenizer(tok, max_length=16)
+        td2 = pt(["abc"])
+        assert td2["prompt_rindex"].item() == int(td2["attention_mask"].sum())
+        pd = PromptData.from_tensordict(td2)
+        rt = pd.to_tensordict()
+        assert torch.equal(rt["input_ids"], td2["input_ids"])
+
+    def test_rollout_from_model(self, lm, tok):
+        from rl_amd.data import RolloutFromModel
+        from rl_amd.modules import RewardModel
+        from rl_amd.tensordict import TensorDict
+
+        torch.manual_seed(0)
+        rm = RewardModel(model=make_tiny_lm())
+        roll = RolloutFromModel(lm, lm, rm, max_new_tokens=4, kl_coef=0.1)
+        ids = torch.randint(1, 250, (2, 6))
+        batch = TensorDict({"input_ids": ids, "attention_mask": torch.ones_like(ids)}, batch_size=[2])
+        td = roll.rollout_from_data(batch)
+        assert td.batch_size == torch.Size([2, 4])
+        assert td["next", "done"][:, -1].all() and not td["next", "done"][:, :-1].any()
+        # same model as reference → KL term ~0 → reward = end score at last step
+        assert torch.allclose(
+            td["next", "reward"][:, :-1], torch.zeros(2, 3, 1), atol=1e-4
+        )
+
+    def test_topk_selector(self):
+        from rl_amd.data import TopKRewardSelector
+        from rl_amd.tensordict import TensorDict
+
+        n, g = 8, 4
+        rewards = torch.arange(n, dtype=torch.float32).reshape(n, 1, 1)
+        done = torch.ones(n, 1, 1, dtype=torch.bool)
+        td = TensorDict(
+            {"x": torch.arange(n).float(), "next": {"reward": rewards, "done": done}},
+            batch_size=[n],
+        )
+        out = TopKRewardSelector(k=2, group_size=g)(td)
+        assert sorted(out["x"].tolist()) == [2.0, 3.0, 6.0, 7.0]
