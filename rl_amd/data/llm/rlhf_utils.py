@@ -170,9 +170,11 @@ class RolloutFromModel:
         B = gen.shape[0]
         done = torch.zeros(B, G, 1, dtype=torch.bool, device=gen.device)
         done[:, -1] = True
+        # per-step view of the full sequence (expanded stride, no copy)
+        ids_steps = gen.unsqueeze(1).expand(B, G, gen.shape[1])
         return TensorDict(
             {
-                "input_ids": gen,
+                "input_ids": ids_steps,
                 "sample_log_prob": lp,
                 "ref_log_prob": ref_lp,
                 "next": {
