@@ -12,6 +12,8 @@ from .distributed import DistributedCollector, DistributedSyncCollector
 from .evaluator import Evaluator
 from .llm import LLMCollector
 from .weight_update import (
+    DistributedWeightUpdater,
+    RPCWeightUpdater,
     RayWeightUpdater,
     MultiProcessedWeightUpdater,
     RemoteModuleWeightUpdater,

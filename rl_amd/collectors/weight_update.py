@@ -19,6 +19,8 @@ __all__ = [
     "MultiProcessedWeightUpdater",
     "RemoteModuleWeightUpdater",
     "RayWeightUpdater",
+    "DistributedWeightUpdater",
+    "RPCWeightUpdater",
 ]
 
 
@@ -101,3 +103,31 @@ class RayWeightUpdater(WeightUpdaterBase):
                 "the RCCL DistributedWeightSyncScheme instead."
             )
         raise NotImplementedError("ray backend scaffolding")
+
+
+class DistributedWeightUpdater(WeightUpdaterBase):
+    """Weight pushes to torch.distributed collector worker ranks over
+    RCCL/gloo stores (reference distributed/generic.py:1209; deprecated
+    there in favor of DistributedWeightSyncScheme — rl_amd keeps the
+    name as a thin adapter over the collector's own push)."""
+
+    def __init__(self, collector):
+        self.collector = collector
+
+    def push_weights(self, policy_or_weights=None, worker_ids=None):
+        self.collector.update_policy_weights_(policy_or_weights)
+
+    __call__ = push_weights
+
+
+class RPCWeightUpdater(WeightUpdaterBase):
+    """Weight pushes to RPC collector workers (reference
+    distributed/rpc.py:951) — adapter over RPCCollector's push."""
+
+    def __init__(self, collector):
+        self.collector = collector
+
+    def push_weights(self, policy_or_weights=None, worker_ids=None):
+        self.collector.update_policy_weights_(policy_or_weights)
+
+    __call__ = push_weights

@@ -30,6 +30,9 @@ from ..tensordict import TensorDict, TensorDictBase
 __all__ = [
     "Trainer",
     "TrainerHookBase",
+    "OptimizationStepper",
+    "DefaultOptimizationStepper",
+    "TD3OptimizationStepper",
     "ReplayBufferTrainer",
     "OptimizerHook",
     "ClearCudaCache",
@@ -649,3 +652,85 @@ class LogValidationReward(TrainerHookBase):
 
     def register(self, trainer: Trainer, name: str = "log_validation_reward"):
         trainer.register_op("post_steps_log", self, name)
+
+
+class OptimizationStepper(TrainerHookBase):
+    """Encapsulates one optimization step per sub-batch (reference
+    trainers.py:200) — override ``step`` for algorithms needing multiple
+    optimizers or delayed updates."""
+
+    def __init__(self):
+        self.trainer = None
+
+    def register(self, trainer, name: str = "optimization_stepper"):
+        self.trainer = trainer
+        trainer.register_op("optimizer", self, name)
+
+    def __call__(self, losses_td):
+        return self.step(losses_td)
+
+    def step(self, losses_td):
+        raise NotImplementedError
+
+    def state_dict(self):
+        return {}
+
+    def load_state_dict(self, sd):
+        pass
+
+
+class DefaultOptimizationStepper(OptimizationStepper):
+    """Single-optimizer step over the sum of ``loss_*`` entries
+    (reference trainers.py:249)."""
+
+    def __init__(self, optimizer, clip_grad_norm: Optional[float] = None):
+        super().__init__()
+        self.optimizer = optimizer
+        self.clip_grad_norm = clip_grad_norm
+
+    def step(self, losses_td):
+        loss = sum(
+            v for k, v in losses_td.items()
+            if isinstance(k, str) and k.startswith("loss")
+        )
+        self.optimizer.zero_grad(set_to_none=True)
+        loss.backward()
+        if self.clip_grad_norm is not None:
+            torch.nn.utils.clip_grad_norm_(
+                [p for g in self.optimizer.param_groups for p in g["params"]],
+                self.clip_grad_norm,
+            )
+        self.optimizer.step()
+        return losses_td
+
+
+class TD3OptimizationStepper(OptimizationStepper):
+    """Delayed-actor stepper (reference trainers.py TD3 variant): the
+    critic optimizer steps every call; the actor (and target-net
+    updater) only every ``policy_delay`` calls."""
+
+    def __init__(self, critic_optimizer, actor_optimizer, *, policy_delay: int = 2,
+                 target_updater=None):
+        super().__init__()
+        self.critic_optimizer = critic_optimizer
+        self.actor_optimizer = actor_optimizer
+        self.policy_delay = policy_delay
+        self.target_updater = target_updater
+        self._calls = 0
+
+    def step(self, losses_td):
+        critic_loss = losses_td.get("loss_qvalue", None)
+        if critic_loss is None:
+            critic_loss = losses_td.get("loss_value")
+        self.critic_optimizer.zero_grad(set_to_none=True)
+        critic_loss.backward(retain_graph=True)
+        self.critic_optimizer.step()
+        self._calls += 1
+        if self._calls % self.policy_delay == 0:
+            actor_loss = losses_td.get("loss_actor")
+            self.actor_optimizer.zero_grad(set_to_none=True)
+            actor_loss.backward()
+            self.actor_optimizer.step()
+            if self.target_updater is not None:
+                self.target_updater.step()
+        return losses_td
