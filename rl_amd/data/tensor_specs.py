@@ -799,3 +799,62 @@ OneHotDiscreteTensorSpec = OneHot
 MultiDiscreteTensorSpec = MultiCategorical
 BinaryDiscreteTensorSpec = Binary
 CompositeSpec = Composite
+
+
+# --------------------------------------------------------------------------- #
+# Reference name-parity aliases and box containers
+# (reference tensor_specs.py: BoundedContinuous/BoundedDiscrete are the
+# dtype-specialized views of Bounded; ContinuousBox/CategoricalBox/
+# BinaryBox/DiscreteBox describe the underlying value domain; legacy
+# long names map to the modern short ones.)
+# --------------------------------------------------------------------------- #
+BoundedContinuous = Bounded
+BoundedDiscrete = Bounded
+UnboundedContinuous = Unbounded
+UnboundedDiscrete = Unbounded
+DiscreteTensorSpec = Categorical
+OneHotDiscreteTensorSpec = OneHot
+MultiOneHotDiscreteTensorSpec = MultiOneHot
+MultiDiscreteTensorSpec = MultiCategorical
+BinaryDiscreteTensorSpec = Binary
+BoundedTensorSpec = Bounded
+UnboundedContinuousTensorSpec = Unbounded
+CompositeSpec = Composite
+NonTensorSpec = NonTensor
+
+
+class Box:
+    """Value-domain descriptor base (reference tensor_specs.py Box)."""
+
+
+class ContinuousBox(Box):
+    def __init__(self, low, high):
+        self.low = low
+        self.high = high
+
+    def __repr__(self):
+        return f"ContinuousBox(low={self.low}, high={self.high})"
+
+
+class CategoricalBox(Box):
+    def __init__(self, n: int):
+        self.n = n
+
+    def __repr__(self):
+        return f"CategoricalBox(n={self.n})"
+
+
+DiscreteBox = CategoricalBox
+
+
+class BinaryBox(Box):
+    def __init__(self, n: int):
+        self.n = n
+
+
+class BoxList(Box):
+    def __init__(self, boxes):
+        self.boxes = list(boxes)
+
+    def __len__(self):
+        return len(self.boxes)

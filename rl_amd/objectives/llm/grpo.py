@@ -242,3 +242,40 @@ class DistillationLoss(LossModule):
             },
             batch_size=[],
         )
+
+
+@dataclasses.dataclass
+class LLMLossOutput:
+    """Typed loss container base (reference llm losses *LossOutput)."""
+
+    loss_objective: "torch.Tensor" = None
+
+
+@dataclasses.dataclass
+class GRPOLossOutput(LLMLossOutput):
+    clip_fraction: "torch.Tensor" = None
+    ESS: "torch.Tensor" = None
+    loss_kl_to_ref: "torch.Tensor" = None
+    kl_to_ref: "torch.Tensor" = None
+    loss_entropy: "torch.Tensor" = None
+
+
+@dataclasses.dataclass
+class DAPOLossOutput(GRPOLossOutput):
+    pass
+
+
+@dataclasses.dataclass
+class CISPOLossOutput(GRPOLossOutput):
+    pass
+
+
+@dataclasses.dataclass
+class SFTLossOutput:
+    loss_sft: "torch.Tensor" = None
+
+
+@dataclasses.dataclass
+class DistillationLossOutput:
+    loss_distill: "torch.Tensor" = None
+    kl_to_teacher: "torch.Tensor" = None
