@@ -46,6 +46,10 @@ def main():
     p.add_argument("--batch", type=int, default=32, help="prompts per rank")
     p.add_argument("--prompt-len", type=int, default=64)
     p.add_argument("--gen-len", type=int, default=32)
+    p.add_argument("--graph-decode", dest="graph_decode", action="store_true",
+                   help="hipGraph-captured static-KV decode loop instead of HF generate")
+    p.add_argument("--no-graph-decode", dest="graph_decode", action="store_false")
+    p.set_defaults(graph_decode=True)
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -82,19 +86,90 @@ def main():
         resp = ids[:, P:]
         return lp.gather(-1, resp.unsqueeze(-1)).squeeze(-1), hidden
 
+    # ---- graph-captured static-KV decode -------------------------------- #
+    # HF generate runs ~6 ms of host-side logic per token; a captured
+    # decode step (single-token forward + multinomial + buffer writes)
+    # replays in a fraction of that.
+    decode_graph = None
+    if args.graph_decode and cuda:
+        from transformers import StaticCache
+
+        cache = StaticCache(
+            config=cfg, max_batch_size=B, max_cache_len=P + G, device=device,
+            dtype=torch.bfloat16,
+        )
+        cur = torch.zeros(B, 1, dtype=torch.long, device=device)
+        pos = torch.zeros(1, dtype=torch.long, device=device)
+        pos_out = torch.zeros(1, dtype=torch.long, device=device)
+        gen_buf = torch.zeros(B, G, dtype=torch.long, device=device)
+        prompt_buf = torch.zeros(B, P, dtype=torch.long, device=device)
+
+        def decode_body():
+            with autocast:
+                out = policy(
+                    input_ids=cur, past_key_values=cache, use_cache=True,
+                    cache_position=pos,
+                )
+            logits = out.logits[:, -1].float()
+            nxt = torch.multinomial(torch.softmax(logits, -1), 1)
+            cur.copy_(nxt)
+            gen_buf.index_copy_(1, pos_out, nxt)
+            pos.add_(1)
+            pos_out.add_(1)
+
+        def prefill(prompts):
+            prompt_buf.copy_(prompts)
+            with torch.no_grad(), autocast:
+                out = policy(
+                    input_ids=prompt_buf, past_key_values=cache, use_cache=True,
+                    cache_position=torch.arange(P, device=device),
+                )
+            logits = out.logits[:, -1].float()
+            first = torch.multinomial(torch.softmax(logits, -1), 1)
+            cur.copy_(first)
+            gen_buf[:, 0:1].copy_(first)
+            pos.fill_(P)
+            pos_out.fill_(1)
+
+        try:
+            with torch.no_grad():
+                prefill(torch.randint(0, V, (B, P), device=device))
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(3):
+                        decode_body()
+                torch.cuda.current_stream().wait_stream(side)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    decode_body()
+                decode_graph = g
+        except Exception:
+            import traceback
+
+            traceback.print_exc()
+            decode_graph = None
+
     def one_iteration():
         prompts = torch.randint(0, V, (B, P), device=device)
         mask = torch.ones_like(prompts)
-        with torch.no_grad(), autocast:
-            gen = policy.generate(
-                input_ids=prompts,
-                attention_mask=mask,
-                max_new_tokens=G,
-                min_new_tokens=G,
-                do_sample=True,
-                top_k=0,
-                pad_token_id=cfg.eos_token_id,
-            )
+        if decode_graph is not None:
+            with torch.no_grad():
+                prefill(prompts)
+                for _ in range(G - 1):
+                    decode_graph.replay()
+            gen = torch.cat([prompt_buf, gen_buf], 1)
+        else:
+            with torch.no_grad(), autocast:
+                gen = policy.generate(
+                    input_ids=prompts,
+                    attention_mask=mask,
+                    max_new_tokens=G,
+                    min_new_tokens=G,
+                    do_sample=True,
+                    top_k=0,
+                    pad_token_id=cfg.eos_token_id,
+                )
             full_mask = torch.ones_like(gen)
             old_lp, _ = token_log_probs(policy, gen, full_mask)
             ref_lp, _ = token_log_probs(ref, gen, full_mask)
@@ -170,6 +245,7 @@ def main():
                         "gen_len": G,
                         "parallelism": f"dp{world}",
                         "kl_to_ref": "k3",
+                        "graph_decode": bool(decode_graph is not None),
                     },
                 }
             )
