@@ -170,7 +170,8 @@ def main():
                     top_k=0,
                     pad_token_id=cfg.eos_token_id,
                 )
-            full_mask = torch.ones_like(gen)
+        full_mask = torch.ones_like(gen)
+        with torch.no_grad(), autocast:
             old_lp, _ = token_log_probs(policy, gen, full_mask)
             ref_lp, _ = token_log_probs(ref, gen, full_mask)
         # synthetic scalar reward per sequence (stand-in for a reward model)
