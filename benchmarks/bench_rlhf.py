@@ -46,10 +46,11 @@ def main():
     p.add_argument("--batch", type=int, default=32, help="prompts per rank")
     p.add_argument("--prompt-len", type=int, default=64)
     p.add_argument("--gen-len", type=int, default=32)
-    p.add_argument("--graph-decode", dest="graph_decode", action="store_true",
-                   help="hipGraph-captured static-KV decode loop instead of HF generate")
-    p.add_argument("--no-graph-decode", dest="graph_decode", action="store_false")
-    p.set_defaults(graph_decode=True)
+    p.add_argument("--decode-mode", choices=["generate", "static", "graph"],
+                   default="static",
+                   help="HF generate / eager static-KV loop / hipGraph-captured "
+                        "loop (graph: HF forward under capture segfaulted on "
+                        "ROCm 7.2 + transformers 5.15 — kept opt-in)")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -91,7 +92,8 @@ def main():
     # decode step (single-token forward + multinomial + buffer writes)
     # replays in a fraction of that.
     decode_graph = None
-    if args.graph_decode and cuda:
+    use_static = cuda and args.decode_mode in ("static", "graph")
+    if use_static:
         from transformers import StaticCache
 
         cache = StaticCache(
@@ -131,33 +133,38 @@ def main():
             pos.fill_(P)
             pos_out.fill_(1)
 
-        try:
-            with torch.no_grad():
-                prefill(torch.randint(0, V, (B, P), device=device))
-                side = torch.cuda.Stream()
-                side.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(side):
-                    for _ in range(3):
+        if args.decode_mode == "graph":
+            try:
+                with torch.no_grad():
+                    prefill(torch.randint(0, V, (B, P), device=device))
+                    side = torch.cuda.Stream()
+                    side.wait_stream(torch.cuda.current_stream())
+                    with torch.cuda.stream(side):
+                        for _ in range(3):
+                            decode_body()
+                    torch.cuda.current_stream().wait_stream(side)
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
                         decode_body()
-                torch.cuda.current_stream().wait_stream(side)
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
-                    decode_body()
-                decode_graph = g
-        except Exception:
-            import traceback
+                    decode_graph = g
+            except Exception:
+                import traceback
 
-            traceback.print_exc()
-            decode_graph = None
+                traceback.print_exc()
+                decode_graph = None
 
     def one_iteration():
         prompts = torch.randint(0, V, (B, P), device=device)
         mask = torch.ones_like(prompts)
-        if decode_graph is not None:
+        if use_static:
             with torch.no_grad():
                 prefill(prompts)
-                for _ in range(G - 1):
-                    decode_graph.replay()
+                if decode_graph is not None:
+                    for _ in range(G - 1):
+                        decode_graph.replay()
+                else:
+                    for _ in range(G - 1):
+                        decode_body()
             gen = torch.cat([prompt_buf, gen_buf], 1)
         else:
             with torch.no_grad(), autocast:
@@ -246,7 +253,8 @@ def main():
                         "gen_len": G,
                         "parallelism": f"dp{world}",
                         "kl_to_ref": "k3",
-                        "graph_decode": bool(decode_graph is not None),
+                        "decode": ("graph" if decode_graph is not None
+                                   else "static" if use_static else "generate"),
                     },
                 }
             )
