@@ -29,10 +29,13 @@ import torch.nn.functional as F
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
-def build_gpt2_small(device):
+def build_gpt2_small(device, attn: str = "sdpa"):
     from transformers import GPT2Config, GPT2LMHeadModel
 
     cfg = GPT2Config()  # gpt2-small: 12 layers, 768 hidden, 12 heads
+    # attn="eager": torch SDPA with a StaticCache segfaults on
+    # ROCm 7.2 + transformers 5.15 (single-token decode path)
+    cfg._attn_implementation = attn
     model = GPT2LMHeadModel(cfg).to(device)
     model.config.pad_token_id = cfg.eos_token_id
     return model, cfg
@@ -47,10 +50,12 @@ def main():
     p.add_argument("--prompt-len", type=int, default=64)
     p.add_argument("--gen-len", type=int, default=32)
     p.add_argument("--decode-mode", choices=["generate", "static", "graph"],
-                   default="static",
-                   help="HF generate / eager static-KV loop / hipGraph-captured "
-                        "loop (graph: HF forward under capture segfaulted on "
-                        "ROCm 7.2 + transformers 5.15 — kept opt-in)")
+                   default="generate",
+                   help="HF generate (default) / eager static-KV loop / "
+                        "hipGraph-captured loop.  static and graph SEGFAULT on "
+                        "this stack (ROCm 7.2 + transformers 5.15 GPT2 "
+                        "StaticCache decode, with both sdpa and eager "
+                        "attention) — kept opt-in for newer stacks")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -65,8 +70,9 @@ def main():
         torch.distributed.init_process_group(backend="nccl" if cuda else "gloo")
     torch.manual_seed(11 + rank)
 
-    policy, cfg = build_gpt2_small(device)
-    ref = build_gpt2_small(device)[0]
+    attn = "eager" if args.decode_mode in ("static", "graph") else "sdpa"
+    policy, cfg = build_gpt2_small(device, attn)
+    ref = build_gpt2_small(device, attn)[0]
     ref.eval()
     for prm in ref.parameters():
         prm.requires_grad_(False)
