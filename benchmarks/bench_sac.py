@@ -131,15 +131,28 @@ def main():
             update_body(static_batch)
         graph = g
 
+    def _sample_into_static():
+        # gather straight into the static graph buffers (index_select
+        # with out=) instead of materializing a fresh batch + copying
+        idx, info = rb._sampler.sample(rb._storage, args.batch)
+        idx = idx.to(device)
+        src = rb._storage._storage
+        for k in static_keys:
+            if k == "index":
+                static_batch.get(k).copy_(idx)
+            elif k == "_weight":
+                static_batch.get(k).copy_(info["_weight"].to(device))
+            else:
+                torch.index_select(src.get(k), 0, idx, out=static_batch.get(k))
+
     def train_step():
         for _ in range(args.utd):
-            batch = rb.sample()
             if graph is not None:
-                for k in static_keys:
-                    static_batch.get(k).copy_(batch.get(k))
+                _sample_into_static()
                 graph.replay()
                 rb.update_tensordict_priority(static_batch)
             else:
+                batch = rb.sample()
                 update_body(batch)
                 rb.update_tensordict_priority(batch)
 
