@@ -39,6 +39,12 @@ void launch_fused_actor(const float*, const float*, const float*, const float*,
                         int, int, int, float, float, void*);
 int lstm_fused_lds_bytes(int);
 int wgrad_slab_count(long);
+int mlp3_lds_bytes(int, int, int);
+void launch_mlp3_fwd(const void*, const void*, const void*, const void*,
+                     const void*, const void*, const void*, void*, void*,
+                     void*, int, int, int, int, void*);
+void launch_mlp3_bwd(const void*, const void*, const void*, const void*,
+                     const void*, void*, void*, int, int, int, void*);
 int synthetic_env_step_lds_bytes(int, int);
 void launch_tanh_normal_logprob_fwd(const float*, const float*, const float*,
                                     float*, int, int, void*);
@@ -321,6 +327,41 @@ void synthetic_env_step_into(torch::Tensor state, torch::Tensor action,
       (long)next_obs.stride(0), (long)reward.stride(0),
       (long)action.stride(0), Bn, S, Aact, (float)max_steps, (void*)stream);
 }
+// Fused 3-layer MLP forward/backward-dgrad (PPO update phase).
+std::vector<torch::Tensor> mlp3_fwd(torch::Tensor x, torch::Tensor w1,
+                                    torch::Tensor b1, torch::Tensor w2,
+                                    torch::Tensor b2, torch::Tensor w3,
+                                    torch::Tensor b3) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16,
+              "mlp3_fwd: bf16 cuda input");
+  TORCH_CHECK(x.is_contiguous(), "contiguous input");
+  const int N = (int)x.size(0), O = (int)x.size(1);
+  const int H = (int)w1.size(0), A2 = (int)w3.size(0);
+  TORCH_CHECK(mlp3_lds_bytes(O, H, A2) <= 160 * 1024, "MLP exceeds LDS budget");
+  auto out = torch::empty({N, A2}, x.options());
+  auto h1 = torch::empty({N, H}, x.options());
+  auto h2 = torch::empty({N, H}, x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_mlp3_fwd(x.data_ptr(), w1.data_ptr(), b1.data_ptr(), w2.data_ptr(),
+                  b2.data_ptr(), w3.data_ptr(), b3.data_ptr(), out.data_ptr(),
+                  h1.data_ptr(), h2.data_ptr(), N, O, H, A2, (void*)stream);
+  return {out, h1, h2};
+}
+
+std::vector<torch::Tensor> mlp3_bwd(torch::Tensor dout, torch::Tensor h1,
+                                    torch::Tensor h2, torch::Tensor w2,
+                                    torch::Tensor w3) {
+  const int N = (int)dout.size(0), A2 = (int)dout.size(1);
+  const int H = (int)h1.size(1);
+  auto dh1 = torch::empty_like(h1);
+  auto dh2 = torch::empty_like(h2);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_mlp3_bwd(dout.contiguous().data_ptr(), h1.data_ptr(), h2.data_ptr(),
+                  w2.data_ptr(), w3.data_ptr(), dh1.data_ptr(), dh2.data_ptr(),
+                  N, H, A2, (void*)stream);
+  return {dh1, dh2};
+}
+
 // Fused TanhNormal log-prob (PPO ratio; action is data).
 torch::Tensor tanh_normal_logprob(torch::Tensor loc, torch::Tensor scale,
                                   torch::Tensor action) {
@@ -432,6 +473,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gae", &gae, "fused GAE scan (HIP)");
   m.def("synthetic_env_step", &synthetic_env_step,
         "fused synthetic-MuJoCo env transition (HIP)");
+  m.def("mlp3_fwd", &mlp3_fwd, "fused 3-layer MLP forward (HIP)");
+  m.def("mlp3_bwd", &mlp3_bwd, "fused MLP backward dgrad chain (HIP)");
   m.def("tanh_normal_entropy", &tanh_normal_entropy,
         "fused reparameterized TanhNormal MC entropy (HIP)");
   m.def("tanh_normal_entropy_bwd", &tanh_normal_entropy_bwd,

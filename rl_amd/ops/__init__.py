@@ -490,3 +490,83 @@ def tanh_normal_entropy(loc, scale, eps=None):
     flat_eps = eps.reshape(-1, eps.shape[-1])
     out = _TanhNormalEntropyFn.apply(flat_loc, flat_scale, flat_eps)
     return out.reshape(loc.shape[:-1])
+
+
+class _FusedMLP3Fn(torch.autograd.Function):
+    """Whole 3-layer tanh MLP forward + backward on HIP
+    (csrc/fused_mlp.hip + the MFMA split-K wgrad): one forward launch
+    (saving h1/h2), one dgrad-chain launch, three wgrad launches —
+    replacing ~45 eager launches per network per minibatch.  The input
+    is the rollout observation (a leaf), so dX is skipped."""
+
+    @staticmethod
+    def forward(ctx, x, w1b, b1b, w2b, b2b, w3b, b3b, w1, b1, w2, b2, w3, b3):
+        x = x.contiguous()
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        out, h1, h2 = _C.mlp3_fwd(x, w1b, b1b, w2b, b2b, w3b, b3b)
+        ctx.save_for_backward(x, h1, h2, w2b, w3b)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        x, h1, h2, w2b, w3b = ctx.saved_tensors
+        dout = dout.contiguous()
+        if dout.dtype != torch.bfloat16:
+            dout = dout.to(torch.bfloat16)
+        dh1, dh2 = _C.mlp3_bwd(dout, h1, h2, w2b, w3b)
+        dw3, db3 = _C.wgrad_splitk(dout, h2, True)
+        dw2, db2 = _C.wgrad_splitk(dh2, h1, True)
+        dw1, db1 = _C.wgrad_splitk(dh1, x, True)
+        return (None, None, None, None, None, None, None,
+                dw1, db1, dw2, db2, dw3, db3)
+
+
+class FusedMLP3(torch.nn.Module):
+    """Routes a Linear-Tanh-Linear-Tanh-Linear stack through the fused
+    kernels on GPU (bf16 weight caches shared with SplitKLinear);
+    eager fallback elsewhere.  Built from three weight-sharing
+    :class:`SplitKLinear` layers with ``enable_bf16_cache`` on."""
+
+    def __init__(self, lin1, lin2, lin3, eager: torch.nn.Module):
+        super().__init__()
+        self.lin1, self.lin2, self.lin3 = lin1, lin2, lin3
+        self.eager = eager
+
+    def forward(self, x):
+        if (
+            x.is_cuda
+            and HAS_HIP_EXT
+            and getattr(self.lin1, "_bf16_cache", False)
+            and x.shape[-1] == self.lin1.in_features
+        ):
+            lead = x.shape[:-1]
+            flat = x.reshape(-1, x.shape[-1])
+            out = _FusedMLP3Fn.apply(
+                flat,
+                self.lin1.weight_bf16, self.lin1.bias_bf16,
+                self.lin2.weight_bf16, self.lin2.bias_bf16,
+                self.lin3.weight_bf16, self.lin3.bias_bf16,
+                self.lin1.weight, self.lin1.bias,
+                self.lin2.weight, self.lin2.bias,
+                self.lin3.weight, self.lin3.bias,
+            )
+            return out.reshape(*lead, out.shape[-1])
+        return self.eager(x)
+
+
+def fuse_mlp3(module: torch.nn.Module) -> torch.nn.Module:
+    """Wrap a Sequential [SplitKLinear, Tanh, SplitKLinear, Tanh,
+    SplitKLinear] (e.g. rl_amd MLP internals after
+    ``convert_linears_to_splitk`` + ``enable_splitk_bf16_cache``) in a
+    :class:`FusedMLP3`.  Returns the module unchanged if the pattern
+    does not match."""
+    import torch.nn as nn
+
+    linears = [m for m in module.modules() if isinstance(m, SplitKLinear)]
+    tanhs = [m for m in module.modules() if isinstance(m, nn.Tanh)]
+    if len(linears) == 3 and len(tanhs) == 2 and all(
+        getattr(l, "_bf16_cache", False) for l in linears
+    ):
+        return FusedMLP3(*linears, eager=module)
+    return module

@@ -469,3 +469,52 @@ class TestTanhNormalEntropy:
         assert torch.allclose(ds_f, ds_e, rtol=1e-3, atol=2e-3), (
             (ds_f - ds_e).abs().max().item()
         )
+
+
+@pytest.mark.gpu
+class TestFusedMLP3:
+    def test_matches_eager_fwd_bwd(self):
+        from rl_amd.ops import (
+            convert_linears_to_splitk,
+            enable_splitk_bf16_cache,
+            fuse_mlp3,
+            refresh_splitk_caches,
+        )
+
+        torch.manual_seed(0)
+        O, H, A2, N = 17, 64, 12, 8192
+        net = torch.nn.Sequential(
+            torch.nn.Linear(O, H), torch.nn.Tanh(),
+            torch.nn.Linear(H, H), torch.nn.Tanh(),
+            torch.nn.Linear(H, A2),
+        ).cuda()
+        ref = torch.nn.Sequential(
+            torch.nn.Linear(O, H), torch.nn.Tanh(),
+            torch.nn.Linear(H, H), torch.nn.Tanh(),
+            torch.nn.Linear(H, A2),
+        ).cuda()
+        ref.load_state_dict(net.state_dict())
+        convert_linears_to_splitk(net)
+        enable_splitk_bf16_cache(net)
+        fused = fuse_mlp3(net)
+        assert type(fused).__name__ == "FusedMLP3"
+        x = torch.randn(N, O, device="cuda")
+        with torch.autocast("cuda", dtype=torch.bfloat16, cache_enabled=False):
+            y_f = fused(x.to(torch.bfloat16))
+            y_e = ref(x.to(torch.bfloat16))
+        assert torch.allclose(y_f.float(), y_e.float(), rtol=2e-2, atol=2e-2), (
+            (y_f.float() - y_e.float()).abs().max().item()
+        )
+        # backward: same upstream grad, compare master-weight grads
+        g = torch.randn_like(y_f, dtype=torch.float32)
+        y_f.float().backward(g)
+        y_e.float().backward(g)
+        fused_linears = [fused.lin1, fused.lin2, fused.lin3]
+        ref_linears = [m for m in ref if isinstance(m, torch.nn.Linear)]
+        for lf, le in zip(fused_linears, ref_linears):
+            assert lf.weight.grad is not None
+            scale = le.weight.grad.abs().max().clamp_min(1e-3)
+            err = (lf.weight.grad - le.weight.grad).abs().max() / scale
+            assert err < 5e-2, err.item()
+            berr = (lf.bias.grad - le.bias.grad).abs().max() / le.bias.grad.abs().max().clamp_min(1e-3)
+            assert berr < 5e-2, berr.item()
