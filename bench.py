@@ -145,8 +145,6 @@ def main():
         )
 
         if HAS_HIP_EXT:
-            from rl_amd.ops import fuse_mlp3
-
             convert_linears_to_splitk(actor)
             convert_linears_to_splitk(critic)
             # one weight cast per layer per optimizer step instead of a
@@ -154,12 +152,11 @@ def main():
             # under graph capture; r29 profile: ~130 cast kernels/step)
             enable_splitk_bf16_cache(actor)
             enable_splitk_bf16_cache(critic)
-            # whole-MLP fused fwd/bwd for the update phase: one forward
-            # launch + one dgrad chain + three MFMA wgrads per network
-            # (csrc/fused_mlp.hip)
-            actor_seq = actor.module[0].module  # TDModule -> Sequential(MLP, extractor)
-            actor_seq[0] = fuse_mlp3(actor_seq[0])
-            critic.module = fuse_mlp3(critic.module)
+            # NOTE: ops.fuse_mlp3 (whole-MLP fused fwd/bwd) was measured
+            # SLOWER here: at 16k-64k-row minibatches the GEMMs belong
+            # on hipBLASLt's MFMA path, not the fused kernel's VALU
+            # dots (r37: T=64 7.49 ms vs 5.18).  It stays available for
+            # launch-bound small-batch updates.
         else:
             args.splitk = False
 

@@ -526,7 +526,13 @@ class FusedMLP3(torch.nn.Module):
     """Routes a Linear-Tanh-Linear-Tanh-Linear stack through the fused
     kernels on GPU (bf16 weight caches shared with SplitKLinear);
     eager fallback elsewhere.  Built from three weight-sharing
-    :class:`SplitKLinear` layers with ``enable_bf16_cache`` on."""
+    :class:`SplitKLinear` layers with ``enable_bf16_cache`` on.
+
+    .. note:: Wins only when the update is LAUNCH-bound (batches up to
+       a few thousand rows).  At the PPO bench's 16k-64k-row
+       minibatches hipBLASLt's MFMA GEMMs beat the kernel's VALU dots
+       (measured r37: T=64 7.49 ms vs 5.18) — the bench keeps the
+       eager GEMM path and uses only the MFMA wgrad."""
 
     def __init__(self, lin1, lin2, lin3, eager: torch.nn.Module):
         super().__init__()
