@@ -84,6 +84,10 @@ def main():
     )
 
     carrier = env.reset()
+    if cuda:
+        # in-place state + the fused one-launch env transition
+        # (csrc/env_step.hip) on the collect path
+        env.enable_capture_mode(True)
 
     def collect():
         nonlocal carrier
