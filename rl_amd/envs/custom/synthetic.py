@@ -152,12 +152,26 @@ class SyntheticMuJoCoEnv(EnvBase):
 
     def _fused_step(self, tensordict: TensorDictBase) -> Optional[TensorDictBase]:
         """One-kernel transition (csrc/env_step.hip): the eager step is
-        ~12 launches of ~4 us inside the captured rollout."""
+        ~12 launches of ~4 us inside the captured rollout.  Returns
+        ``None`` (eager fallback) when the extension is absent or the
+        state dims exceed the 160 KB LDS budget (e.g. Humanoid's
+        376-dim state: the A matrix alone is 565 KB)."""
+        if getattr(self, "_fused_step_ok", None) is False:
+            return None
         from ... import ops
 
         if not ops.HAS_HIP_EXT:
+            self._fused_step_ok = False
             return None
         from ... import _C
+
+        if getattr(self, "_fused_step_ok", None) is None:
+            apad = self.obs_dim + 1
+            lds = 4 * (self.obs_dim * apad + self.act_dim * apad + 16 * apad
+                       + 16 * self.act_dim + 16)
+            self._fused_step_ok = lds <= 160 * 1024
+            if not self._fused_step_ok:
+                return None
 
         bs = self.batch_size
         action = tensordict.get("action")
