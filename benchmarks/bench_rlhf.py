@@ -46,7 +46,7 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=8)
     p.add_argument("--warmup", type=int, default=2)
-    p.add_argument("--batch", type=int, default=64, help="prompts per rank")
+    p.add_argument("--batch", type=int, default=256, help="prompts per rank")
     p.add_argument("--prompt-len", type=int, default=64)
     p.add_argument("--gen-len", type=int, default=32)
     p.add_argument("--decode-mode", choices=["generate", "static", "graph"],
