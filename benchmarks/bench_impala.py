@@ -98,7 +98,7 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--envs", type=int, default=256)
+    p.add_argument("--envs", type=int, default=512)
     p.add_argument("--unroll", type=int, default=20)
     p.add_argument("--graph", dest="graph", action="store_true",
                    help="hipGraph-capture the T-step behavior rollout")
