@@ -89,18 +89,41 @@ def main():
         # (csrc/env_step.hip) on the collect path
         env.enable_capture_mode(True)
 
-    def collect():
-        nonlocal carrier
+    # collect: policy + env step captured as a hipGraph over a static
+    # entry-observation buffer (the env state updates in place); only
+    # rb.extend — whose writer index lives on the host — stays eager,
+    # reading the transition from the static snapshot.
+    entry_obs = carrier.get("observation").clone()
+    collect_snap = {}
+    collect_graph = None
+
+    def collect_body():
+        td = TensorDict({"observation": entry_obs}, batch_size=[args.envs], device=device)
         with torch.no_grad(), set_exploration_type(ExplorationType.RANDOM), autocast:
-            carrier = actor(carrier)
-            carrier.set("action", carrier.get("action").float())
-            carrier, nxt = env.step_and_maybe_reset(carrier)
-            rb.extend(carrier.select(
-                "observation", "action", "sample_log_prob",
-                ("next", "observation"), ("next", "reward"),
-                ("next", "done"), ("next", "terminated"),
-            ))
-            carrier = nxt
+            td = actor(td)
+            td.set("action", td.get("action").float())
+            td, nxt = env.step_and_maybe_reset(td)
+        snap = td.select(
+            "observation", "action", "sample_log_prob",
+            ("next", "observation"), ("next", "reward"),
+            ("next", "done"), ("next", "terminated"),
+        )
+        if not collect_snap:
+            for k in snap.keys(True, True):
+                collect_snap[k] = snap.get(k).clone()
+        else:
+            for k, buf in collect_snap.items():
+                buf.copy_(snap.get(k))
+        entry_obs.copy_(nxt.get("observation"))
+
+    def collect():
+        if collect_graph is not None:
+            collect_graph.replay()
+        else:
+            collect_body()
+        rb.extend(TensorDict(
+            dict(collect_snap), batch_size=[args.envs], device=device
+        ))
 
     def update_body(batch):
         with autocast:
@@ -166,6 +189,22 @@ def main():
 
     if use_graph:
         try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    collect_body()
+            torch.cuda.current_stream().wait_stream(side)
+            cg = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(cg):
+                collect_body()
+            collect_graph = cg
+        except Exception:
+            import traceback
+
+            traceback.print_exc()
+            collect_graph = None
+        try:
             capture_update()
         except Exception:
             import traceback
@@ -210,6 +249,7 @@ def main():
                     "parallelism": "dp1",
                     "buffer_device": str(device),
                     "update_graph": bool(graph is not None),
+                    "collect_graph": bool(collect_graph is not None),
                 },
             }
         )
