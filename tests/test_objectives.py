@@ -504,3 +504,64 @@ class TestMultiAgentGAE:
         )
         MultiAgentGAE(gamma=0.9, lmbda=0.9)(td)
         assert td["advantage"].shape == (B, T, N, 1)
+
+
+class TestBatchedEnsemble:
+    def test_bmm_path_matches_loop(self):
+        from rl_amd.objectives.common import _EnsembleModule
+        from rl_amd.modules import MLP, ValueOperator
+        from rl_amd.tensordict import TensorDict
+
+        torch.manual_seed(0)
+        q = ValueOperator(
+            MLP(in_features=7, out_features=1, num_cells=[16, 16]),
+            in_keys=["observation", "action"],
+        )
+        ens = _EnsembleModule(q, 4)
+        td = TensorDict(
+            {"observation": torch.randn(6, 4), "action": torch.randn(6, 3)},
+            batch_size=[6],
+        )
+        fast = ens(td.clone()).get("state_action_value")
+        ens._batched = False  # force the loop path on the same weights
+        loop = ens(td.clone()).get("state_action_value")
+        assert fast.shape == (4, 6, 1)
+        assert torch.allclose(fast, loop, atol=1e-5)
+        fast.sum().backward()
+        assert all(p.grad is not None for p in ens.parameters())
+
+    def test_sac_loss_still_correct(self):
+        from rl_amd.modules import MLP, NormalParamExtractor, ProbabilisticActor, TanhNormal, ValueOperator
+        from rl_amd.objectives import SACLoss
+        from rl_amd.tensordict import TensorDict, TensorDictModule
+
+        torch.manual_seed(0)
+        net = torch.nn.Sequential(
+            MLP(in_features=3, out_features=4, num_cells=[16]), NormalParamExtractor()
+        )
+        actor = ProbabilisticActor(
+            TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"]),
+            in_keys=["loc", "scale"], distribution_class=TanhNormal, return_log_prob=True,
+        )
+        q = ValueOperator(MLP(in_features=5, out_features=1, num_cells=[16]), in_keys=["observation", "action"])
+        loss = SACLoss(actor, q, num_qvalue_nets=2)
+        loss.make_value_estimator()
+        n = 8
+        td = TensorDict(
+            {
+                "observation": torch.randn(n, 3),
+                "action": torch.rand(n, 2) * 2 - 1,
+                "sample_log_prob": torch.randn(n),
+                "next": {
+                    "observation": torch.randn(n, 3),
+                    "reward": torch.randn(n, 1),
+                    "done": torch.zeros(n, 1, dtype=torch.bool),
+                    "terminated": torch.zeros(n, 1, dtype=torch.bool),
+                },
+            },
+            batch_size=[n],
+        )
+        out = loss(td)
+        total = out.get("loss_actor") + out.get("loss_qvalue") + out.get("loss_alpha")
+        total.backward()
+        assert torch.isfinite(total)
