@@ -6,3 +6,4 @@ from .mocking_classes import (
     MockSerialEnv,
     NestedCountingEnv,
 )
+from .dist_utils import assert_no_new_python_processes, snapshot_python_processes
