@@ -61,3 +61,56 @@ with torch.no_grad():
 
 print(json.dumps(results, indent=1))
 open("gpurun_out/kernel_microbench.json", "w").write(json.dumps(results, indent=1))
+
+# wgrad: MFMA split-K vs torch mm (bf16, fp32 accum)
+from rl_amd import _C
+for K, N, M in [(16384, 64, 64), (65536, 64, 64)]:
+    dy = torch.randn(K, N, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(K, M, device="cuda", dtype=torch.bfloat16)
+    results[f"wgrad_mfma_K{K}_ms"] = timeit_gpu(lambda: _C.wgrad_splitk(dy, x, True), n=50)
+    results[f"wgrad_torch_mm_K{K}_ms"] = timeit_gpu(
+        lambda: torch.mm(dy.t().float(), x.float()), n=50
+    )
+
+# fused actor vs eager policy chain (rollout shape)
+from rl_amd.modules import MLP, NormalParamExtractor, ProbabilisticActor, TanhNormal
+from rl_amd.ops import FusedTanhNormalActor
+from rl_amd.tensordict import TensorDict, TensorDictModule
+from rl_amd.envs.utils import ExplorationType, set_exploration_type
+net = torch.nn.Sequential(
+    MLP(in_features=17, out_features=12, num_cells=[64, 64],
+        activation_class=torch.nn.Tanh, device="cuda"),
+    NormalParamExtractor(),
+)
+actor = ProbabilisticActor(
+    TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"]),
+    in_keys=["loc", "scale"], distribution_class=TanhNormal, return_log_prob=True,
+)
+fa = FusedTanhNormalActor(actor)
+td = TensorDict({"observation": torch.randn(4096, 17, device="cuda")}, batch_size=[4096], device="cuda")
+with torch.no_grad(), set_exploration_type(ExplorationType.RANDOM):
+    results["actor_fused_B4096_ms"] = timeit_gpu(lambda: fa(td.clone(False)), n=50)
+    results["actor_eager_B4096_ms"] = timeit_gpu(lambda: actor(td.clone(False)), n=50)
+
+# fused env transition vs eager step
+from rl_amd.envs.custom.synthetic import HalfCheetahVec
+env = HalfCheetahVec(batch_size=[4096], device="cuda", dtype=torch.float32)
+env.reset()
+env.enable_capture_mode(True)
+act = torch.rand(4096, 6, device="cuda") * 2 - 1
+etd = TensorDict({"action": act}, batch_size=[4096], device="cuda")
+results["env_step_fused_B4096_ms"] = timeit_gpu(lambda: env._step(etd), n=50)
+env._fused_step_ok = False
+results["env_step_eager_B4096_ms"] = timeit_gpu(lambda: env._step(etd), n=50)
+
+# fused TanhNormal log-prob / entropy vs eager distribution
+from rl_amd import ops as _ops
+loc = torch.randn(16384, 6, device="cuda")
+scale = torch.rand(16384, 6, device="cuda") * 0.9 + 0.1
+with torch.no_grad():
+    a = TanhNormal(loc, scale).sample()
+results["logprob_fused_N16k_ms"] = timeit_gpu(lambda: _ops.tanh_normal_logprob(loc, scale, a), n=50)
+results["logprob_eager_N16k_ms"] = timeit_gpu(lambda: TanhNormal(loc, scale).log_prob(a), n=50)
+
+print(json.dumps(results, indent=1))
+open("gpurun_out/kernel_microbench.json", "w").write(json.dumps(results, indent=1))
