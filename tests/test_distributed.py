@@ -286,3 +286,40 @@ def test_learner_group_sync():
     # weights synced at init, grads averaged
     assert outs[0][0] == pytest.approx(outs[1][0])
     assert outs[0][1] == pytest.approx(outs[1][1])
+
+
+def _dist_scheme_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from rl_amd.weight_update import DistributedWeightSyncScheme
+
+    torch.manual_seed(10 + rank)  # DIFFERENT weights per rank
+    model = torch.nn.Linear(4, 2)
+    scheme = DistributedWeightSyncScheme(src=0)
+    scheme.connect(model) if hasattr(scheme, "connect") else setattr(scheme, "model", model)
+    if rank == 0:
+        scheme.send()
+    else:
+        scheme.receive(model)
+    q.put((rank, model.weight.sum().item()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_distributed_weight_sync_scheme():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_dist_scheme_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = {}
+    for _ in range(2):
+        rank, wsum = q.get(timeout=60)
+        outs[rank] = wsum
+    for p in procs:
+        p.join(30)
+    assert outs[0] == pytest.approx(outs[1])  # rank 1 received rank 0's weights
