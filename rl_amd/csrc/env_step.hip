@@ -38,7 +38,7 @@ __global__ void __launch_bounds__(ENV_THREADS) synthetic_env_step_kernel(
     const long obs_rs, const long sc_rs, const long act_rs,
     const int Bn, const int S, const int Aact, const float max_steps) {
   extern __shared__ float smem[];
-  const int apad = S + 1;
+  const int apad = S | 1;  // odd stride: no LDS bank collisions
   float* s_A = smem;                 // [S, apad]
   float* s_B = s_A + S * apad;       // [Aact, apad]
   float* s_state = s_B + Aact * apad;  // [ENV_ROWS, apad] (old state)
@@ -101,7 +101,7 @@ __global__ void __launch_bounds__(ENV_THREADS) synthetic_env_step_kernel(
 }  // namespace
 
 extern "C" int synthetic_env_step_lds_bytes(int S, int Aact) {
-  const int apad = S + 1;
+  const int apad = S | 1;
   return (int)sizeof(float) *
          (S * apad + Aact * apad + ENV_ROWS * apad + ENV_ROWS * Aact +
           ENV_ROWS);
