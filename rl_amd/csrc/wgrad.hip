@@ -269,8 +269,9 @@ __global__ void wgrad_reduce_kernel(const float* __restrict__ part,
 }  // namespace
 
 extern "C" int wgrad_slab_count(long K) {
-  // target ~128 slabs; slab size a multiple of ROW_CHUNK
-  long k_slab = (K + 127) / 128;
+  // target ~256 slabs (one per CU): slab size a multiple of ROW_CHUNK.
+  // (128 slabs left half the chip idle at K=65536 — measured 46 us.)
+  long k_slab = (K + 255) / 256;
   k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
   if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
   return (int)((K + k_slab - 1) / k_slab);
@@ -279,7 +280,7 @@ extern "C" int wgrad_slab_count(long K) {
 extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
                          float* dbias, float* part, float* bias_part, long K,
                          int N, int M, void* stream) {
-  long k_slab = (K + 127) / 128;
+  long k_slab = (K + 255) / 256;
   k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
   if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
   const int slabs = (int)((K + k_slab - 1) / k_slab);
