@@ -267,10 +267,23 @@ def main():
         store.get(("next", "terminated")).zero_()  # env never terminates
         env.enable_capture_mode(True)
 
+        mega = hasattr(_C, "fused_rollout")
+
         def rollout_body_direct():
             with torch.no_grad():
                 eps_all = torch.randn(T, B, act_dim, device=device)
                 noise_all = torch.randn(T, B, obs_dim, device=device) * 0.1
+                if mega:
+                    # env rows are independent: the whole T-step rollout
+                    # runs as ONE kernel (csrc/rollout_fused.hip) — no
+                    # per-step launch latency at all
+                    _C.fused_rollout(
+                        env._state, env._t.reshape(-1), w1, b1, w2, b2, w3,
+                        b3, env.A, env.B, eps_all, noise_all, s_obs, s_act,
+                        s_lp, s_nobs, s_rew, s_done, float(env.max_steps),
+                        fa.inv_softplus_bias, fa.scale_lb,
+                    )
+                    return
                 for t in range(T):
                     _C.fused_actor_into(
                         env._state, w1, b1, w2, b2, w3, b3, eps_all[t],

@@ -40,6 +40,13 @@ void launch_fused_actor(const float*, const float*, const float*, const float*,
 int lstm_fused_lds_bytes(int);
 int wgrad_slab_count(long);
 int mlp3_lds_bytes(int, int, int);
+int fused_rollout_lds_bytes(int, int, int, int);
+void launch_fused_rollout(float*, float*, const float*, const float*,
+                          const float*, const float*, const float*,
+                          const float*, const float*, const float*,
+                          const float*, const float*, float*, float*, float*,
+                          float*, float*, bool*, int, int, int, int, int, int,
+                          float, float, float, void*);
 void launch_mlp3_fwd(const void*, const void*, const void*, const void*,
                      const void*, const void*, const void*, void*, void*,
                      void*, int, int, int, int, void*);
@@ -327,6 +334,41 @@ void synthetic_env_step_into(torch::Tensor state, torch::Tensor action,
       (long)next_obs.stride(0), (long)reward.stride(0),
       (long)action.stride(0), Bn, S, Aact, (float)max_steps, (void*)stream);
 }
+// Whole-rollout mega-kernel: the entire T-step rollout in one launch.
+void fused_rollout(torch::Tensor state, torch::Tensor step_ct,
+                   torch::Tensor w1, torch::Tensor b1, torch::Tensor w2,
+                   torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
+                   torch::Tensor A, torch::Tensor B, torch::Tensor eps,
+                   torch::Tensor noise, torch::Tensor st_obs,
+                   torch::Tensor st_act, torch::Tensor st_logp,
+                   torch::Tensor st_nobs, torch::Tensor st_rew,
+                   torch::Tensor st_done, double max_steps,
+                   double inv_softplus_bias, double scale_lb) {
+  TORCH_CHECK(state.is_cuda() && state.scalar_type() == torch::kFloat32,
+              "fused_rollout: fp32 cuda state");
+  const int Bn = (int)state.size(0), S = (int)state.size(1);
+  const int H1 = (int)w1.size(0), H2 = (int)w2.size(0);
+  const int Aact = (int)w3.size(0) / 2;
+  const int T = (int)eps.size(0);
+  TORCH_CHECK(fused_rollout_lds_bytes(S, H1, H2, Aact) <= 160 * 1024,
+              "fused_rollout exceeds the LDS budget");
+  TORCH_CHECK(st_obs.is_contiguous() && st_act.is_contiguous() &&
+                  st_logp.is_contiguous() && st_nobs.is_contiguous() &&
+                  st_rew.is_contiguous() && st_done.is_contiguous(),
+              "store tensors must be contiguous [B, T, ...]");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_fused_rollout(
+      state.data_ptr<float>(), step_ct.data_ptr<float>(),
+      w1.data_ptr<float>(), b1.data_ptr<float>(), w2.data_ptr<float>(),
+      b2.data_ptr<float>(), w3.data_ptr<float>(), b3.data_ptr<float>(),
+      A.data_ptr<float>(), B.data_ptr<float>(), eps.data_ptr<float>(),
+      noise.data_ptr<float>(), st_obs.data_ptr<float>(),
+      st_act.data_ptr<float>(), st_logp.data_ptr<float>(),
+      st_nobs.data_ptr<float>(), st_rew.data_ptr<float>(),
+      st_done.data_ptr<bool>(), Bn, S, H1, H2, Aact, T, (float)max_steps,
+      (float)inv_softplus_bias, (float)scale_lb, (void*)stream);
+}
+
 // Fused 3-layer MLP forward/backward-dgrad (PPO update phase).
 std::vector<torch::Tensor> mlp3_fwd(torch::Tensor x, torch::Tensor w1,
                                     torch::Tensor b1, torch::Tensor w2,
@@ -473,6 +515,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gae", &gae, "fused GAE scan (HIP)");
   m.def("synthetic_env_step", &synthetic_env_step,
         "fused synthetic-MuJoCo env transition (HIP)");
+  m.def("fused_rollout", &fused_rollout,
+        "entire T-step rollout in one launch (HIP)");
   m.def("mlp3_fwd", &mlp3_fwd, "fused 3-layer MLP forward (HIP)");
   m.def("mlp3_bwd", &mlp3_bwd, "fused MLP backward dgrad chain (HIP)");
   m.def("tanh_normal_entropy", &tanh_normal_entropy,

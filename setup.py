@@ -33,6 +33,7 @@ if with_hip:
         os.path.join(CSRC, "env_step.hip"),
         os.path.join(CSRC, "loss_ops.hip"),
         os.path.join(CSRC, "fused_mlp.hip"),
+        os.path.join(CSRC, "rollout_fused.hip"),
     ]
     define_macros.append(("RL_AMD_WITH_HIP", None))
     ext_cls = cpp_extension.CUDAExtension

@@ -518,3 +518,72 @@ class TestFusedMLP3:
             assert err < 5e-2, err.item()
             berr = (lf.bias.grad - le.bias.grad).abs().max() / le.bias.grad.abs().max().clamp_min(1e-3)
             assert berr < 5e-2, berr.item()
+
+
+@pytest.mark.gpu
+class TestFusedRollout:
+    def test_matches_per_step_kernels(self):
+        """The whole-rollout mega-kernel must reproduce the per-step
+        fused_actor_into + synthetic_env_step_into pair bit-for-bit
+        (same eps/noise streams)."""
+        from rl_amd import _C
+        from rl_amd.envs.custom.synthetic import HalfCheetahVec
+        from rl_amd.modules import MLP, NormalParamExtractor
+
+        torch.manual_seed(0)
+        B, T = 64, 12
+        env1 = HalfCheetahVec(batch_size=[B], device="cuda", dtype=torch.float32)
+        env2 = HalfCheetahVec(batch_size=[B], device="cuda", dtype=torch.float32)
+        env1.set_seed(0); env2.set_seed(0)
+        env1.reset(); env2.reset()
+        env2._state.copy_(env1._state); env2._t.copy_(env1._t)
+        env1.enable_capture_mode(True); env2.enable_capture_mode(True)
+        S, A = env1.obs_dim, env1.act_dim
+        net = torch.nn.Sequential(
+            MLP(in_features=S, out_features=2 * A, num_cells=[64, 64],
+                activation_class=torch.nn.Tanh, device="cuda"),
+            NormalParamExtractor(),
+        )
+        lins = [m for m in net.modules() if isinstance(m, torch.nn.Linear)]
+        w1, b1 = lins[0].weight, lins[0].bias
+        w2, b2 = lins[1].weight, lins[1].bias
+        w3, b3 = lins[2].weight, lins[2].bias
+        isb = 0.5413248546129181
+        eps = torch.randn(T, B, A, device="cuda")
+        noise = torch.randn(T, B, S, device="cuda") * 0.1
+
+        def alloc():
+            return dict(
+                obs=torch.zeros(B, T, S, device="cuda"),
+                act=torch.zeros(B, T, A, device="cuda"),
+                lp=torch.zeros(B, T, device="cuda"),
+                nobs=torch.zeros(B, T, S, device="cuda"),
+                rew=torch.zeros(B, T, 1, device="cuda"),
+                done=torch.zeros(B, T, 1, dtype=torch.bool, device="cuda"),
+            )
+
+        st1, st2 = alloc(), alloc()
+        # force a mid-rollout reset
+        env1._t.fill_(float(env1.max_steps) - 5)
+        env2._t.fill_(float(env2.max_steps) - 5)
+        _C.fused_rollout(env1._state, env1._t.reshape(-1), w1, b1, w2, b2,
+                         w3, b3, env1.A, env1.B, eps, noise,
+                         st1["obs"], st1["act"], st1["lp"], st1["nobs"],
+                         st1["rew"], st1["done"], float(env1.max_steps), isb, 1e-4)
+        for t in range(T):
+            _C.fused_actor_into(env2._state, w1, b1, w2, b2, w3, b3, eps[t],
+                                st2["act"][:, t], st2["lp"][:, t], isb, 1e-4)
+            _C.synthetic_env_step_into(
+                env2._state, st2["act"][:, t], env2.A, env2.B,
+                env2._t.reshape(-1), st2["nobs"][:, t], st2["obs"][:, t],
+                st2["rew"][:, t], st2["done"][:, t], noise[t],
+                float(env2.max_steps))
+        for k in st1:
+            a, b = st1[k], st2[k]
+            if a.dtype == torch.bool:
+                assert torch.equal(a, b), k
+            else:
+                assert torch.allclose(a, b, atol=1e-5), (k, (a - b).abs().max())
+        assert torch.allclose(env1._state, env2._state, atol=1e-5)
+        assert torch.equal(env1._t, env2._t)
+        assert st1["done"][:, 4].all()  # the forced reset happened
