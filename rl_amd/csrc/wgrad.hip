@@ -147,9 +147,11 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_mfma_kernel(
   const long k_begin = (long)blockIdx.x * k_slab;
   const long k_end = min(K, k_begin + (long)k_slab);
 
-  // bf16 LDS staging, padded rows (stride 66 shorts = 33 banks)
-  __shared__ __hip_bfloat16 s_dy[ROW_CHUNK][64 + 2];
-  __shared__ __hip_bfloat16 s_x[ROW_CHUNK][64 + 2];
+  // bf16 LDS staging, DOUBLE-BUFFERED (pad: stride 66 shorts = 33 banks):
+  // the next chunk's global loads land in the idle buffer while MFMA
+  // consumes the current one, hiding the HBM latency.
+  __shared__ __hip_bfloat16 s_dy[2][ROW_CHUNK][64 + 2];
+  __shared__ __hip_bfloat16 s_x[2][ROW_CHUNK][64 + 2];
   __shared__ float s_bias[64];
 
   const int tid = threadIdx.x;
@@ -162,7 +164,8 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_mfma_kernel(
 
   const int lr = tid / 64;  // staging: thread t loads column t%64
   const int lc = tid % 64;
-  for (long kc = k_begin; kc < k_end; kc += ROW_CHUNK) {
+
+  auto stage = [&](long kc, int buf) {
     const int rows = (int)min((long)ROW_CHUNK, k_end - kc);
     for (int i = 0; i < ROW_CHUNK / 4; ++i) {
       const int r = lr + 4 * i;
@@ -172,11 +175,17 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_mfma_kernel(
         if (lc < N - n0t) dv = dy[gk * N + n0t + lc];
         if (lc < M - m0t) xv = x[gk * M + m0t + lc];
       }
-      s_dy[r][lc] = dv;
-      s_x[r][lc] = xv;
+      s_dy[buf][r][lc] = dv;
+      s_x[buf][r][lc] = xv;
       bias_acc += __bfloat162float(dv);
     }
-    __syncthreads();
+  };
+
+  int cur = 0;
+  stage(k_begin, 0);
+  __syncthreads();
+  for (long kc = k_begin; kc < k_end; kc += ROW_CHUNK) {
+    if (kc + ROW_CHUNK < k_end) stage(kc + ROW_CHUNK, 1 - cur);
     // MFMA over the chunk: A[mf][k] = dy[k][n0w + mf], B[k][nf] = x[k][m0 + nf]
     const int n_off = wave * 16 + (lane & 15);
     for (int kk = 0; kk < ROW_CHUNK; kk += 32) {
@@ -184,7 +193,7 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_mfma_kernel(
 #pragma unroll
       for (int reg = 0; reg < 8; ++reg) {
         const int k = kk + 8 * (lane >> 4) + reg;
-        a[reg] = *reinterpret_cast<const short*>(&s_dy[k][n_off]);
+        a[reg] = *reinterpret_cast<const short*>(&s_dy[cur][k][n_off]);
       }
 #pragma unroll
       for (int mt = 0; mt < 4; ++mt) {
@@ -193,12 +202,13 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_mfma_kernel(
 #pragma unroll
         for (int reg = 0; reg < 8; ++reg) {
           const int k = kk + 8 * (lane >> 4) + reg;
-          b[reg] = *reinterpret_cast<const short*>(&s_x[k][m_off]);
+          b[reg] = *reinterpret_cast<const short*>(&s_x[cur][k][m_off]);
         }
         acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt], 0, 0, 0);
       }
     }
     __syncthreads();
+    cur = 1 - cur;
   }
 
   // commit: D row = (lane>>4)*4 + reg (within the wave's 16-row band)
