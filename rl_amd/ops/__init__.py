@@ -312,7 +312,8 @@ class _SplitKLinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         dy2 = dy.reshape(-1, dy.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
-        dx = (dy @ weight).reshape(x.shape)
+        # first-layer inputs are leaf observations: skip the dx GEMM
+        dx = (dy @ weight).reshape(x.shape) if ctx.needs_input_grad[0] else None
         if dy2.is_cuda and dy2.dtype == torch.bfloat16 and HAS_HIP_EXT:
             outs = _C.wgrad_splitk(dy2, x2, ctx.has_bias)
             dw = outs[0].to(weight.dtype)
@@ -347,7 +348,7 @@ class _SplitKLinearCachedFn(torch.autograd.Function):
         dy = dy.contiguous()
         dy2 = dy.reshape(-1, dy.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
-        dx = (dy @ w_bf).reshape(x.shape)
+        dx = (dy @ w_bf).reshape(x.shape) if ctx.needs_input_grad[0] else None
         outs = _C.wgrad_splitk(dy2, x2, ctx.has_bias)
         dw = outs[0]  # fp32 — matches the fp32 master weight
         db = outs[1] if ctx.has_bias else None
