@@ -96,3 +96,4 @@ from .llm import (
     TokenizedDatasetLoader,
     TopKRewardSelector,
 )
+from .video import VideoClipRef

@@ -19,3 +19,4 @@ from .async_envs import AsyncEnvPool, ProcessorAsyncEnvPool, ThreadingAsyncEnvPo
 from .env_creator import EnvCreator, env_creator
 from .custom import ChessEnv, TicTacToeEnv, ToyVLAEnv
 from . import libs
+from .transforms import HumanoidMacroAction

@@ -24,6 +24,7 @@ __all__ = [
     "MacroPrimitive",
     "MacroAction",
     "TargetMacroAction",
+    "HumanoidMacroAction",
     "MacroPrimitiveTransform",
     "RunningMeanStd",
     "R3MTransform",
@@ -76,6 +77,12 @@ class TargetMacroAction(MacroAction):
             target,
             torch.tensor(settle_steps),
         )
+
+
+class HumanoidMacroAction(TargetMacroAction):
+    """Humanoid actuator-control macro (reference
+    mujoco/_humanoid_primitives.py:15): a TargetMacroAction whose
+    target lives directly in actuator-control coordinates."""
 
 
 class MacroPrimitiveTransform(Transform):

@@ -89,7 +89,7 @@ from .models import GPT2RewardModel, OnlineDTActor, RewardModel
 from .models import ExactGPRegressor, GPWorldModel, RBFController
 from .models import ACTModel
 from .value_norm import PopArtValueNorm, RunningValueNorm, ValueNorm
-from .vla import TinyVLA
+from .vla import LeRobotPolicyWrapper, TinyVLA
 from .value_transforms import (
     ComposeValueTransform,
     IdentityValueTransform,

@@ -17,7 +17,7 @@ from torch import nn
 
 from ..tensordict import TensorDictBase
 
-__all__ = ["TinyVLA"]
+__all__ = ["TinyVLA", "LeRobotPolicyWrapper"]
 
 
 def _hash_instruction(text: str, vocab: int) -> int:
@@ -117,3 +117,20 @@ class TinyVLA(nn.Module):
             td.set("action_tokens", tokens)
             td.set("action", tokens[:, 0])
         return td
+
+
+class LeRobotPolicyWrapper(torch.nn.Module):
+    """Adapt a LeRobot-style pretrained chunk policy to the rl_amd VLA
+    schema (reference modules/vla/wrappers.py:24) — gated: `lerobot` is
+    not installed in this image; TinyVLA covers the schema offline."""
+
+    def __init__(self, *args, **kwargs):
+        import importlib.util
+
+        super().__init__()
+        if importlib.util.find_spec("lerobot") is None:
+            raise ImportError(
+                "LeRobotPolicyWrapper requires the `lerobot` package, which "
+                "is not installed in this image. Use TinyVLA for offline work."
+            )
+        raise NotImplementedError("lerobot adapter scaffolding")

@@ -30,3 +30,4 @@ from .dreamer_v3 import (
 )
 from .tqc import TQCLoss
 from .act import ACTLoss, DiffusionActor, DiffusionBCLoss
+from .common import AUTO_LOSS_MASK_KEYS

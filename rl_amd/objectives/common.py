@@ -36,6 +36,10 @@ def params_to_buffers(module: nn.Module) -> nn.Module:
     return module
 
 
+# keys a loss will auto-mask on when present (reference common.py:41)
+AUTO_LOSS_MASK_KEYS = (("collector", "mask"), "shifted_valid")
+
+
 class _EnsembleModule(nn.Module):
     """N copies of a module evaluated as one vmapped call
     (reference vmap-over-params, common.py:939-969)."""
