@@ -340,6 +340,14 @@ class _SplitKLinearCachedFn(torch.autograd.Function):
         ctx.save_for_backward(x, w_bf)
         ctx.has_bias = bias is not None
         with torch.autocast("cuda", enabled=False):
+            if w_bf.shape[0] == 1:
+                # single-output head: hipBLASLt runs the [N,H]x[H,1]
+                # gemv at ~29 us (MT1x4x256, one-ish WG); an
+                # elementwise-multiply + row reduce is ~6 us
+                out = (x * w_bf[0]).sum(-1, keepdim=True)
+                if b_bf is not None:
+                    out = out + b_bf
+                return out
             return torch.nn.functional.linear(x, w_bf, b_bf)
 
     @staticmethod
@@ -348,7 +356,13 @@ class _SplitKLinearCachedFn(torch.autograd.Function):
         dy = dy.contiguous()
         dy2 = dy.reshape(-1, dy.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
-        dx = (dy @ w_bf).reshape(x.shape) if ctx.needs_input_grad[0] else None
+        if not ctx.needs_input_grad[0]:
+            dx = None
+        elif w_bf.shape[0] == 1:
+            # rank-1 dgrad: dy [N,1] @ w [1,H] is a broadcast multiply
+            dx = (dy2 * w_bf).reshape(x.shape)
+        else:
+            dx = (dy @ w_bf).reshape(x.shape)
         outs = _C.wgrad_splitk(dy2, x2, ctx.has_bias)
         dw = outs[0]  # fp32 — matches the fp32 master weight
         db = outs[1] if ctx.has_bias else None
