@@ -104,6 +104,40 @@ __global__ void verify_cand_kernel(int cand, float* max_err) {
   atomicMax((int*)max_err, __float_as_int(err));
 }
 
+// ---- 32x32x16 bf16 verification (same method) -----------------------------
+// Candidate mappings: A row = lane&31, B col = lane&31,
+// k = 8*(lane>>5) + reg (8 regs).  C/D (guide, hardware-verified m74/m101):
+// col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5), reg in [0,16).
+using frag_cd32 = __attribute__((ext_vector_type(16))) float;
+
+__global__ void verify32_kernel(float* max_err) {
+  const int lane = threadIdx.x;
+  frag_ab a, b;
+  const int am = lane & 31;
+  const int bn = lane & 31;
+  for (int reg = 0; reg < 8; ++reg) {
+    const int k = 8 * (lane >> 5) + reg;
+    a[reg] = f2bf((float)am + 0.01f * (float)k);
+    b[reg] = f2bf((k == bn % 16) ? (1.0f + (float)k / 100.0f) : 0.0f);
+  }
+  frag_cd32 acc = {};
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+  auto bf = [](float f) { union { float f; unsigned u; } v{f}; v.u &= 0xFFFF0000u; return v.f; };
+  float err = 0.f;
+  for (int reg = 0; reg < 16; ++reg) {
+    const int m = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+    const int n = lane & 31;
+    // expected: D[m][n] = A[m][n%16] * (1 + (n%16)/100)
+    const float a1 = bf((float)m + 0.01f * (float)(n % 16));
+    const float w1 = bf(1.0f + (float)(n % 16) / 100.0f);
+    const float expect = a1 * w1;
+    err = fmaxf(err, fabsf(acc[reg] - expect));
+  }
+  atomicMax((int*)max_err, __float_as_int(err));
+}
+
+
+
 int main() {
   float* d_err;
   hipMalloc(&d_err, sizeof(float));
@@ -117,5 +151,16 @@ int main() {
     printf("candidate %d: max |err| = %f  %s\n", cand, err,
            err < 0.15f ? "<-- MATCHES" : "");
   }
+  {
+    float zero = 0.f;
+    hipMemcpy(d_err, &zero, sizeof(float), hipMemcpyHostToDevice);
+    hipLaunchKernelGGL(verify32_kernel, dim3(1), dim3(64), 0, 0, d_err);
+    hipDeviceSynchronize();
+    float err;
+    hipMemcpy(&err, d_err, sizeof(float), hipMemcpyDeviceToHost);
+    printf("32x32x16 candidate: max |err| = %f  %s\n", err,
+           err < 0.15f ? "<-- MATCHES" : "");
+  }
   return 0;
 }
+

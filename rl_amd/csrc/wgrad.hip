@@ -41,6 +41,7 @@
 
 using bf16_frag = __attribute__((ext_vector_type(8))) short;
 using f32_frag = __attribute__((ext_vector_type(4))) float;
+using f32_frag16 = __attribute__((ext_vector_type(16))) float;
 
 #define WG_THREADS 256
 #define ROW_CHUNK 64   // k-rows staged in LDS per iteration
@@ -236,6 +237,95 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad_mfma_kernel(
   }
 }
 
+// 32x32-tile MFMA variant: each of the 4 waves owns ONE 32x32 output
+// tile (mfma_f32_32x32x16_bf16; layout hardware-verified in
+// benchmarks/mfma_probe.hip: A row / B col = lane&31,
+// k = 8*(lane>>5)+reg, C/D col = lane&31,
+// row = (reg&3)+8*(reg>>2)+4*(lane>>5)).  2.5x fewer LDS fragment
+// reads per MFMA than the 16x16 tiling.
+__global__ void __launch_bounds__(WG_THREADS) wgrad_mfma32_kernel(
+    const __hip_bfloat16* __restrict__ dy,  // [K, N]
+    const __hip_bfloat16* __restrict__ x,   // [K, M]
+    float* __restrict__ dw,                 // partials [slab, tile, elem]
+    float* __restrict__ dbias,              // bias partials [slab, n]
+    long K, int N, int M, int k_slab) {
+  const int n0t = blockIdx.y * 64;
+  const int m0t = blockIdx.z * 64;
+  const long k_begin = (long)blockIdx.x * k_slab;
+  const long k_end = min(K, k_begin + (long)k_slab);
+
+  __shared__ __hip_bfloat16 s_dy[2][ROW_CHUNK][64 + 2];
+  __shared__ __hip_bfloat16 s_x[2][ROW_CHUNK][64 + 2];
+  __shared__ float s_bias[64];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int tile_n = (wave >> 1) * 32;
+  const int tile_m = (wave & 1) * 32;
+  f32_frag16 acc = {};
+  float bias_acc = 0.f;
+
+  const int lr = tid / 64;
+  const int lc = tid % 64;
+
+  auto stage = [&](long kc, int buf) {
+    const int rows = (int)min((long)ROW_CHUNK, k_end - kc);
+    for (int i = 0; i < ROW_CHUNK / 4; ++i) {
+      const int r = lr + 4 * i;
+      const long gk = kc + r;
+      __hip_bfloat16 dv = __hip_bfloat16(0.f), xv = __hip_bfloat16(0.f);
+      if (r < rows) {
+        if (lc < N - n0t) dv = dy[gk * N + n0t + lc];
+        if (lc < M - m0t) xv = x[gk * M + m0t + lc];
+      }
+      s_dy[buf][r][lc] = dv;
+      s_x[buf][r][lc] = xv;
+      bias_acc += __bfloat162float(dv);
+    }
+  };
+
+  int cur = 0;
+  stage(k_begin, 0);
+  __syncthreads();
+  const int n_off = tile_n + (lane & 31);
+  const int m_off = tile_m + (lane & 31);
+  for (long kc = k_begin; kc < k_end; kc += ROW_CHUNK) {
+    if (kc + ROW_CHUNK < k_end) stage(kc + ROW_CHUNK, 1 - cur);
+    for (int kk = 0; kk < ROW_CHUNK; kk += 16) {
+      bf16_frag a, b;
+#pragma unroll
+      for (int reg = 0; reg < 8; ++reg) {
+        const int k = kk + 8 * (lane >> 5) + reg;
+        a[reg] = *reinterpret_cast<const short*>(&s_dy[cur][k][n_off]);
+        b[reg] = *reinterpret_cast<const short*>(&s_x[cur][k][m_off]);
+      }
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+    }
+    __syncthreads();
+    cur = 1 - cur;
+  }
+
+  const long num_tiles_m = gridDim.z;
+  const long tile_elems = (long)64 * 64 * gridDim.y * num_tiles_m;
+  float* part = dw + (long)blockIdx.x * tile_elems +
+                ((long)blockIdx.y * num_tiles_m + blockIdx.z) * (64 * 64);
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int n_loc = tile_n + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+    const int m_loc = tile_m + (lane & 31);
+    part[(long)n_loc * 64 + m_loc] = acc[reg];
+  }
+  if (dbias != nullptr) {
+    if (lr == 0) s_bias[lc] = 0.f;
+    __syncthreads();
+    atomicAdd(&s_bias[lc], bias_acc);
+    __syncthreads();
+    if (lr == 0)
+      dbias[(long)blockIdx.x * (gridDim.y * 64) + blockIdx.y * 64 + lc] = s_bias[lc];
+  }
+}
+
 // reduce the [slab, tile, elem] partials into dW [N, M]: one WAVE per
 // output element — each lane sums slabs/64 strided values (all loads
 // independent and in flight), then a 6-step shuffle tree; lane 0 writes
@@ -296,7 +386,7 @@ extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
   const int slabs = (int)((K + k_slab - 1) / k_slab);
   const int tiles_n = (N + 63) / 64, tiles_m = (M + 63) / 64;
   dim3 grid(slabs, tiles_n, tiles_m);
-  hipLaunchKernelGGL(wgrad_mfma_kernel, grid, dim3(WG_THREADS), 0,
+  hipLaunchKernelGGL(wgrad_mfma32_kernel, grid, dim3(WG_THREADS), 0,
                      (hipStream_t)stream,
                      (const __hip_bfloat16*)dy, (const __hip_bfloat16*)x, part,
                      bias_part, K, N, M, (int)k_slab);
