@@ -324,6 +324,47 @@ def main():
     # full-step graph capture needs stable grad buffers
     zero_set_to_none = not (args.full_graph and cuda and not distributed)
 
+    # Distributed mode cannot capture the whole step (the RCCL
+    # all-reduce sits between backward and optimizer), but the
+    # minibatch fwd+bwd — the launch-heavy part — CAN be captured and
+    # replayed with the comm/step/refresh left eager, keeping the
+    # per-GPU step close to the single-GPU captured one.
+    mb_graph = {"graph": None, "sub": None, "total": None}
+
+    def _mb_fwd_bwd(sub):
+        with autocast:
+            out = loss_mod(sub)
+            total = (
+                out.get("loss_objective")
+                + out.get("loss_critic")
+                + out.get("loss_entropy")
+            )
+        optim.zero_grad(set_to_none=False)
+        total.backward()
+        return total
+
+    def _capture_mb_graph(example_sub):
+        static_sub = example_sub.clone(False)
+        for k in list(static_sub.keys(True, True)):
+            static_sub.set(k, static_sub.get(k).clone())
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                _mb_fwd_bwd(static_sub)
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            _mb_fwd_bwd(static_sub)
+        mb_graph["graph"] = g
+        mb_graph["sub"] = static_sub
+
+    use_mb_graph = bool(
+        (distributed or os.environ.get("RL_AMD_BENCH_FORCE_MB_GRAPH") == "1")
+        and cuda
+        and args.full_graph
+    )
+
     def update_phase():
         batch = store
         with torch.no_grad(), autocast:
@@ -339,15 +380,40 @@ def main():
             shuffled = flat[perm]
             for i in range(args.minibatches):
                 sub = shuffled[i * mb : (i + 1) * mb]
-                with autocast:
-                    out = loss_mod(sub)
-                    total = (
-                        out.get("loss_objective")
-                        + out.get("loss_critic")
-                        + out.get("loss_entropy")
-                    )
-                optim.zero_grad(set_to_none=zero_set_to_none)
-                total.backward()
+                if use_mb_graph:
+                    if mb_graph["graph"] is None:
+                        try:
+                            _capture_mb_graph(sub)
+                        except Exception:
+                            import traceback
+
+                            traceback.print_exc(file=sys.stderr)
+                            mb_graph["graph"] = False
+                    if mb_graph["graph"] not in (None, False):
+                        static_sub = mb_graph["sub"]
+                        for k in list(static_sub.keys(True, True)):
+                            static_sub.get(k).copy_(sub.get(k))
+                        mb_graph["graph"].replay()
+                    else:
+                        with autocast:
+                            out = loss_mod(sub)
+                            total = (
+                                out.get("loss_objective")
+                                + out.get("loss_critic")
+                                + out.get("loss_entropy")
+                            )
+                        optim.zero_grad(set_to_none=zero_set_to_none)
+                        total.backward()
+                else:
+                    with autocast:
+                        out = loss_mod(sub)
+                        total = (
+                            out.get("loss_objective")
+                            + out.get("loss_critic")
+                            + out.get("loss_entropy")
+                        )
+                    optim.zero_grad(set_to_none=zero_set_to_none)
+                    total.backward()
                 if distributed:
                     with torch.no_grad():
                         flat_grads = torch.cat(
@@ -377,7 +443,7 @@ def main():
         update_phase()
 
     step_fn = one_step
-    if args.full_graph and cuda and not distributed:
+    if args.full_graph and cuda and not distributed and not use_mb_graph:
         # capture EVERYTHING (rollout + GAE + fwd/bwd/Adam): grads must be
         # pre-allocated and kept (set_to_none=False) so buffers are static
         try:
