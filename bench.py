@@ -520,7 +520,8 @@ def main():
                 "fused_actor": bool(args.fused_actor and cuda),
                 "splitk_wgrad": bool(args.splitk and cuda),
                 "store_direct_rollout": bool(store_direct),
-                "full_step_graph": bool(args.full_graph and cuda and not distributed),
+                "full_step_graph": bool(args.full_graph and cuda and not distributed and not use_mb_graph),
+                "minibatch_graph": bool(use_mb_graph and mb_graph["graph"] not in (None, False)),
             },
         }
         print(json.dumps(result))
