@@ -8,6 +8,7 @@ xGMI for every collective.
 """
 __version__ = "0.1.0"
 
-from ._utils import logger, timeit, seed_everything, set_profiling_enabled
+from ._utils import logger, timeit, seed_everything, set_profiling_enabled, warn
+from .services import service_backend
 
-__all__ = ["logger", "timeit", "seed_everything", "set_profiling_enabled"]
+__all__ = ["logger", "timeit", "seed_everything", "set_profiling_enabled", "warn", "service_backend"]

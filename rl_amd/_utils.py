@@ -292,3 +292,12 @@ def implement_for(module_name: str, from_version: str = None, to_version: str = 
         return fn
 
     return deco
+
+
+def warn(msg: str, category=UserWarning, stacklevel: int = 2) -> None:
+    """One-stop warning emitter (reference torchrl.warn): routes through
+    the package logger AND python warnings so both sinks see it."""
+    import warnings as _warnings
+
+    logger.warning(msg)
+    _warnings.warn(msg, category=category, stacklevel=stacklevel)

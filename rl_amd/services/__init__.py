@@ -11,7 +11,7 @@ from __future__ import annotations
 import threading
 from typing import Any, Dict, Optional
 
-__all__ = ["ServiceBase", "LocalService", "get_services", "reset_services"]
+__all__ = ["ServiceBase", "LocalService", "get_services", "reset_services", "service_backend"]
 
 
 class ServiceBase:
@@ -84,3 +84,16 @@ def get_services(backend: str = "local", **kwargs) -> ServiceBase:
 def reset_services() -> None:
     global _GLOBAL
     _GLOBAL = None
+
+
+def service_backend(name: str = "local", **kwargs):
+    """Select a service backend by name (reference
+    torchrl/_comm/backends.py:191): ``"local"`` returns the in-process
+    registry; ``"ray"`` is gated (ray is not installed here)."""
+    if name == "local":
+        return get_services()
+    if name == "ray":
+        raise RuntimeError(
+            "service backend 'ray' not available in this build (no ray)"
+        )
+    raise ValueError(f"unknown service backend {name!r}")
