@@ -1,2 +1,2 @@
 from .transformers_wrapper import LLMWrapperBase, TransformersWrapper
-from .vllm_wrapper import vLLMWrapper
+from .vllm_wrapper import SGLangWrapper, vLLMWrapper

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import importlib.util
 
-__all__ = ["vLLMWrapper"]
+__all__ = ["vLLMWrapper", "SGLangWrapper"]
 
 _has_vllm = importlib.util.find_spec("vllm") is not None
 
@@ -31,3 +31,19 @@ class vLLMWrapper:
         raise NotImplementedError(
             "vLLM backend scaffolding: install vllm and implement engine glue"
         )
+
+
+class SGLangWrapper:
+    """SGLang-backed LLM policy wrapper (reference
+    policies/sglang_wrapper.py:53) — gated: sglang is not installed in
+    this image; TransformersWrapper covers the TensorDict interface."""
+
+    def __init__(self, *args, **kwargs):
+        import importlib.util
+
+        if importlib.util.find_spec("sglang") is None:
+            raise ImportError(
+                "SGLangWrapper requires the `sglang` package, which is not "
+                "installed in this image. Use TransformersWrapper instead."
+            )
+        raise NotImplementedError("sglang backend scaffolding")
