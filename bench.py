@@ -443,6 +443,13 @@ def main():
         update_phase()
 
     step_fn = one_step
+    if args.full_graph and B * T > 2_000_000:
+        # full-step capture at multi-million-frame stores exhausted the
+        # capture pool (core dump at 65536 envs x T=64); the rollout
+        # mega-kernel path carries those sizes without the whole-step
+        # graph (measured 103M frames/s at 65536 x T=16)
+        print("[bench] store > 2M frames: full-step graph disabled", file=sys.stderr)
+        args.full_graph = False
     if args.full_graph and cuda and not distributed and not use_mb_graph:
         # capture EVERYTHING (rollout + GAE + fwd/bwd/Adam): grads must be
         # pre-allocated and kept (set_to_none=False) so buffers are static
