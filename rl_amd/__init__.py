@@ -12,3 +12,9 @@ from ._utils import logger, timeit, seed_everything, set_profiling_enabled, warn
 from .services import service_backend
 
 __all__ = ["logger", "timeit", "seed_everything", "set_profiling_enabled", "warn", "service_backend"]
+
+# reference-parity re-exports (torchrl subpackage-level __all__)
+from ._utils import (  # noqa: F401
+    implement_for,
+)
+__all__ = sorted(set(list(globals().get('__all__', [])) + ['implement_for']))

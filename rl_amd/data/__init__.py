@@ -97,3 +97,23 @@ from .llm import (
     TopKRewardSelector,
 )
 from .video import VideoClipRef
+
+# reference-parity re-exports (torchrl subpackage-level __all__)
+from .replay_buffers import (  # noqa: F401
+    FlatStorageCheckpointer,
+    H5StorageCheckpointer,
+    ListStorageCheckpointer,
+    NestedStorageCheckpointer,
+    RemoteTensorDictReplayBuffer,
+    StorageCheckpointerBase,
+    TensorStorageCheckpointer,
+)
+from .tensor_specs import (  # noqa: F401
+    BoundedContinuous,
+    DEVICE_TYPING,
+)
+from .vla import (  # noqa: F401
+    ActionTokenizerBase,
+    UniformActionTokenizer,
+)
+__all__ = sorted(set(list(globals().get('__all__', [])) + ['ActionTokenizerBase', 'BoundedContinuous', 'DEVICE_TYPING', 'FlatStorageCheckpointer', 'H5StorageCheckpointer', 'ListStorageCheckpointer', 'NestedStorageCheckpointer', 'RemoteTensorDictReplayBuffer', 'StorageCheckpointerBase', 'TensorStorageCheckpointer', 'UniformActionTokenizer']))

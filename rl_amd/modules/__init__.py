@@ -111,3 +111,25 @@ from .legacy import (
     WorldModel,
 )
 from .models.exploration import LazygSDEModule
+
+# reference-parity re-exports (torchrl subpackage-level __all__)
+from .functional import (  # noqa: F401
+    symexp,
+    symlog,
+)
+from .value_transforms import (  # noqa: F401
+    signed_hyperbolic,
+    signed_parabolic,
+)
+from rl_amd.collectors import (  # noqa: F401
+    RandomPolicy,
+)
+from rl_amd.data import (  # noqa: F401
+    EXP3Score,
+    PUCTScore,
+    UCBScore,
+)
+from rl_amd.objectives import (  # noqa: F401
+    DiffusionActor,
+)
+__all__ = sorted(set(list(globals().get('__all__', [])) + ['DiffusionActor', 'EXP3Score', 'PUCTScore', 'RandomPolicy', 'UCBScore', 'signed_hyperbolic', 'signed_parabolic', 'symexp', 'symlog']))

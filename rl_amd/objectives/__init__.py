@@ -31,3 +31,18 @@ from .dreamer_v3 import (
 from .tqc import TQCLoss
 from .act import ACTLoss, DiffusionActor, DiffusionBCLoss
 from .common import AUTO_LOSS_MASK_KEYS
+
+# reference-parity re-exports (torchrl subpackage-level __all__)
+from .utils import (  # noqa: F401
+    default_value_kwargs,
+    hold_out_params,
+    next_state_value,
+)
+from rl_amd.modules.functional import (  # noqa: F401
+    symexp,
+    symlog,
+    two_hot_cross_entropy,
+    two_hot_decode,
+    two_hot_encode,
+)
+__all__ = sorted(set(list(globals().get('__all__', [])) + ['default_value_kwargs', 'hold_out_params', 'next_state_value', 'symexp', 'symlog', 'two_hot_cross_entropy', 'two_hot_decode', 'two_hot_encode']))
