@@ -124,12 +124,10 @@ from .value_transforms import (  # noqa: F401
 from rl_amd.collectors import (  # noqa: F401
     RandomPolicy,
 )
-from rl_amd.data import (  # noqa: F401
+from rl_amd.data.map import (  # noqa: F401
     EXP3Score,
     PUCTScore,
     UCBScore,
 )
-from rl_amd.objectives import (  # noqa: F401
-    DiffusionActor,
-)
+from rl_amd.objectives.act import DiffusionActor  # noqa: F401
 __all__ = sorted(set(list(globals().get('__all__', [])) + ['DiffusionActor', 'EXP3Score', 'PUCTScore', 'RandomPolicy', 'UCBScore', 'signed_hyperbolic', 'signed_parabolic', 'symexp', 'symlog']))
