@@ -61,3 +61,22 @@ class EnvCreator:
 
 def env_creator(fn: Callable) -> EnvCreator:
     return EnvCreator(fn)
+
+
+def get_env_metadata(env_or_creator, kwargs=None):
+    """Extract picklable metadata (specs, batch size, device) from an
+    env or an env-making callable (reference env_creator.py): batched /
+    remote workers rebuild specs from this without instantiating the
+    env again."""
+    from .common import EnvMetaData
+
+    from .common import EnvBase
+
+    if isinstance(env_or_creator, EnvBase):
+        env = env_or_creator
+    else:
+        env = env_or_creator(**(kwargs or {}))
+    return EnvMetaData.build(env)
+
+
+__all__.append("get_env_metadata")

@@ -199,3 +199,137 @@ class OpenMLEnv(EnvBase):
     def __init__(self, dataset_name: str, *, device=None, batch_size=()):
         _require("openml", type(self).__name__)
         raise NotImplementedError("openml scaffolding")
+
+
+# ---------------------------------------------------------------------------
+# reference-parity wrapper aliases and remaining gated libs
+# (torchrl exports both XxxEnv — build-from-name — and XxxWrapper —
+#  wrap-an-instance — for most libraries; reference envs/__init__.py)
+# ---------------------------------------------------------------------------
+
+
+class BraxEnv(EnvBase):
+    """Brax JAX physics envs (reference libs/brax.py) — gated: brax/jax
+    are not in this image."""
+
+    def __init__(self, env_name: str, *, batch_size=(), device=None, **kwargs):
+        _require("brax", type(self).__name__)
+        raise NotImplementedError("jax interop scaffolding (jax not in image)")
+
+
+class BraxWrapper(BraxEnv):
+    """Wrap an existing brax env instance (reference libs/brax.py)."""
+
+
+class VmasEnv(EnvBase):
+    """VMAS vectorized multi-agent envs (reference libs/vmas.py) —
+    gated on `vmas`; grouping follows MarlGroupMapType."""
+
+    def __init__(self, scenario, *, num_envs: int = 1, device=None, **kwargs):
+        _require("vmas", type(self).__name__)
+        raise NotImplementedError("vmas scaffolding")
+
+
+class VmasWrapper(VmasEnv):
+    """Wrap an existing vmas env instance (reference libs/vmas.py)."""
+
+
+class PettingZooEnv(EnvBase):
+    """PettingZoo multi-agent envs (reference libs/pettingzoo.py) —
+    gated on `pettingzoo`; grouping follows MarlGroupMapType."""
+
+    def __init__(self, task: str = None, *, parallel: bool = True, device=None, **kwargs):
+        _require("pettingzoo", type(self).__name__)
+        raise NotImplementedError("pettingzoo scaffolding")
+
+
+class PettingZooWrapper(PettingZooEnv):
+    """Wrap an existing PettingZoo env instance."""
+
+
+class MOGymEnv(EnvBase):
+    """mo-gymnasium multi-objective envs (reference libs/meltingpot.py
+    neighborhood: mo_gym.py) — the reward is a vector spec."""
+
+    def __init__(self, env_name: str, *, device=None, **kwargs):
+        _require("mo_gymnasium", type(self).__name__)
+        raise NotImplementedError("mo_gymnasium scaffolding")
+
+
+class MOGymWrapper(MOGymEnv):
+    """Wrap an existing mo-gymnasium env instance."""
+
+
+class MultiThreadedEnvWrapper(EnvPoolEnv):
+    """envpool thread-pool batched envs (reference libs/envpool.py:
+    MultiThreadedEnvWrapper) — alias of the envpool wrapper."""
+
+
+class MultiThreadedEnv(EnvPoolEnv):
+    """Build an envpool batched env by task name (reference envpool.py)."""
+
+
+# wrap-an-instance aliases for existing gated by-name classes
+JumanjiWrapper = JumanjiEnv
+IsaacLabWrapper = IsaacLabEnv
+IsaacGymWrapper = IsaacGymEnv
+MJLabWrapper = MjLabEnv
+MJLabEnv = MjLabEnv
+MujocoPlaygroundWrapper = MujocoPlaygroundEnv
+GenesisWrapper = GenesisEnv
+SMACv2Wrapper = SMACv2Env
+MeltingpotWrapper = MeltingpotEnv
+OpenSpielWrapper = OpenSpielEnv
+UnityMLAgentsWrapper = UnityMLAgentsEnv
+LiberoWrapper = LiberoEnv
+ProcgenWrapper = ProcgenEnv
+
+
+class MujocoPlaygroundAgentSpec:
+    """Per-agent description for MuJoCo-Playground multi-agent tasks
+    (reference libs/mujoco_playground.py): name + obs/action slices."""
+
+    def __init__(self, name: str, obs_slice=None, action_slice=None):
+        self.name = name
+        self.obs_slice = obs_slice
+        self.action_slice = action_slice
+
+
+class MujocoPlaygroundAgentMapping:
+    """Maps playground agents to rl_amd MARL groups (reference
+    libs/mujoco_playground.py)."""
+
+    def __init__(self, agent_specs):
+        self.agent_specs = list(agent_specs)
+
+    def group_map(self):
+        return {"agents": [a.name for a in self.agent_specs]}
+
+
+__all__ += [
+    "BraxEnv",
+    "BraxWrapper",
+    "VmasEnv",
+    "VmasWrapper",
+    "PettingZooEnv",
+    "PettingZooWrapper",
+    "MOGymEnv",
+    "MOGymWrapper",
+    "MultiThreadedEnvWrapper",
+    "MultiThreadedEnv",
+    "JumanjiWrapper",
+    "IsaacLabWrapper",
+    "IsaacGymWrapper",
+    "MJLabWrapper",
+    "MJLabEnv",
+    "MujocoPlaygroundWrapper",
+    "GenesisWrapper",
+    "SMACv2Wrapper",
+    "MeltingpotWrapper",
+    "OpenSpielWrapper",
+    "UnityMLAgentsWrapper",
+    "LiberoWrapper",
+    "ProcgenWrapper",
+    "MujocoPlaygroundAgentSpec",
+    "MujocoPlaygroundAgentMapping",
+]
