@@ -131,3 +131,59 @@ from rl_amd.data.map import (  # noqa: F401
 )
 from rl_amd.objectives.act import DiffusionActor  # noqa: F401
 __all__ = sorted(set(list(globals().get('__all__', [])) + ['DiffusionActor', 'EXP3Score', 'PUCTScore', 'RandomPolicy', 'UCBScore', 'signed_hyperbolic', 'signed_parabolic', 'symexp', 'symlog']))
+
+# reference-parity: MCTS scores, multi-agent/cross-group critics, RNN +
+# primer utilities, discrete-distribution extras
+from .extras import (  # noqa: F401
+    CrossCriticGroupSpec,
+    CrossGroupCritic,
+    MCTSScore,
+    MCTSScores,
+    MultiAgentNetBase,
+    RecurrentMatmulPrecision,
+    RecurrentMatmulPrecisionUserMode,
+    UCB1TunedScore,
+    VLAWrapperBase,
+    VmapModule,
+    canonicalize_rnn_subset,
+    get_env_transforms_from_module,
+    get_primers_from_module,
+    get_recurrent_matmul_precision,
+    set_exploration_modules_spec_from_env,
+    set_recurrent_matmul_precision,
+)
+from .distributions.discrete import OneHotOrdinal, ReparamGradientStrategy  # noqa: F401
+from ..tensordict.nn import (  # noqa: F401
+    ProbabilisticTensorDictSequential as SafeProbabilisticTensorDictSequential,
+)
+
+def distributions_maps(distribution_class: str):
+    """Map a distribution name to its class (reference
+    distributions/__init__.py distributions_maps)."""
+    from . import distributions as _d
+
+    maps = {
+        "delta": getattr(_d, "Delta", None),
+        "tanhnormal": getattr(_d, "TanhNormal", None),
+        "truncnormal": getattr(_d, "TruncatedNormal", None),
+        "tanhdelta": getattr(_d, "TanhDelta", None),
+        "onehotcategorical": getattr(_d, "OneHotCategorical", None),
+        "categorical": getattr(_d, "MaskedCategorical", None),
+        "ordinal": getattr(_d, "Ordinal", None),
+        "onehotordinal": OneHotOrdinal,
+    }
+    key = distribution_class.lower().replace("_", "").replace("-", "")
+    if key not in maps or maps[key] is None:
+        raise NotImplementedError(f"unknown distribution {distribution_class!r}")
+    return maps[key]
+
+__all__ = sorted(set(__all__) | {
+    "CrossCriticGroupSpec", "CrossGroupCritic", "MCTSScore", "MCTSScores",
+    "MultiAgentNetBase", "RecurrentMatmulPrecision",
+    "RecurrentMatmulPrecisionUserMode", "UCB1TunedScore", "VLAWrapperBase",
+    "VmapModule", "canonicalize_rnn_subset", "get_env_transforms_from_module",
+    "get_primers_from_module", "get_recurrent_matmul_precision",
+    "set_exploration_modules_spec_from_env", "set_recurrent_matmul_precision",
+    "OneHotOrdinal", "ReparamGradientStrategy",
+    "SafeProbabilisticTensorDictSequential", "distributions_maps",
+})

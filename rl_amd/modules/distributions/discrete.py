@@ -6,6 +6,7 @@ Reference: pytorch/rl torchrl/modules/distributions/discrete.py
 """
 from __future__ import annotations
 
+from enum import Enum
 from typing import Optional
 
 import torch
@@ -139,3 +140,23 @@ class LLMMaskedCategorical(MaskedCategorical):
         lp = super().log_prob(tokens)
         valid = self._token_mask.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
         return (lp * valid.to(lp.dtype)).sum(-1)
+
+
+class ReparamGradientStrategy(Enum):
+    """How discrete samples get surrogate gradients (reference
+    discrete.py:60): straight-through pass-through, or relaxed one-hot
+    (Gumbel-softmax style)."""
+
+    PassThrough = 1
+    RelaxedOneHot = 2
+
+
+class OneHotOrdinal(OneHotCategorical):
+    """One-hot version of :class:`Ordinal` (reference discrete.py:677):
+    ordinal-structured logits, one-hot samples/log-probs."""
+
+    def __init__(self, scores: torch.Tensor):
+        super().__init__(logits=_ordinal_logits(scores))
+
+
+__all__ += ["ReparamGradientStrategy", "OneHotOrdinal"]

@@ -46,3 +46,7 @@ from rl_amd.modules.functional import (  # noqa: F401
     two_hot_encode,
 )
 __all__ = sorted(set(list(globals().get('__all__', [])) + ['default_value_kwargs', 'hold_out_params', 'next_state_value', 'symexp', 'symlog', 'two_hot_cross_entropy', 'two_hot_decode', 'two_hot_encode']))
+
+from .dreamer_v3 import categorical_kl_balanced  # noqa: F401
+from .utils import add_random_module, group_optimizers  # noqa: F401
+__all__ = sorted(set(__all__) | {"categorical_kl_balanced", "add_random_module", "group_optimizers"})

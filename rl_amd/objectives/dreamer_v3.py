@@ -290,3 +290,22 @@ class DreamerV3ValueLoss(LossModule):
             value = td.get(self.tensor_keys.value)
             loss = (symlog(target) - value).pow(2).mean()
         return TensorDict({"loss_value": loss}, batch_size=[])
+
+
+def categorical_kl_balanced(
+    posterior_logits: torch.Tensor,
+    prior_logits: torch.Tensor,
+    alpha: float = 0.8,
+    free_bits: float = 1.0,
+) -> torch.Tensor:
+    """Balanced categorical KL (reference dreamer_v3.py:114):
+    ``alpha·KL(sg(post)||prior) + (1−alpha)·KL(post||sg(prior))`` with
+    free-bits clamping — the scalar-weighted form of
+    :func:`categorical_kl_terms`."""
+    dyn, rep = categorical_kl_terms(
+        posterior_logits, prior_logits, free_nats=free_bits
+    )
+    return alpha * dyn + (1.0 - alpha) * rep
+
+
+__all__.append("categorical_kl_balanced")

@@ -180,3 +180,42 @@ def next_state_value(
         nxt = tensordict.get("next").clone(False)
         pred_next_val = operator(nxt).get(next_val_key)
     return reward + gamma * (~done).to(reward.dtype) * pred_next_val
+
+
+RANDOM_MODULE_LIST: tuple = ()
+
+
+def add_random_module(module) -> None:
+    """Register a module class as stochastic so loss vmap-randomness
+    detection treats it as random (reference common.py:1012)."""
+    global RANDOM_MODULE_LIST
+    RANDOM_MODULE_LIST = RANDOM_MODULE_LIST + (module,)
+
+
+def group_optimizers(*optimizers) -> torch.optim.Optimizer:
+    """Merge several same-type optimizers into one (reference
+    utils.py:997): their param groups are concatenated so a single
+    ``step()`` drives them all — one fused multi-tensor pass instead
+    of several small ones."""
+    cls = None
+    groups = []
+    for opt in optimizers:
+        if opt is None:
+            continue
+        if cls is None:
+            cls = type(opt)
+        elif type(opt) is not cls:
+            raise ValueError(
+                f"all optimizers must share a type; got {cls.__name__} and {type(opt).__name__}"
+            )
+        groups.extend(opt.param_groups)
+    if cls is None:
+        raise ValueError("no optimizers given")
+    merged = cls([{"params": []}])
+    merged.param_groups = []
+    for g in groups:
+        merged.add_param_group(g)
+    return merged
+
+
+__all__ += ["add_random_module", "group_optimizers", "RANDOM_MODULE_LIST"]
