@@ -18,3 +18,21 @@ from ._utils import (  # noqa: F401
     implement_for,
 )
 __all__ = sorted(set(list(globals().get('__all__', [])) + ['implement_for']))
+from ._utils import (  # noqa: F401
+    auto_unwrap_transformed_env,
+    compile_with_warmup,
+    cuda_memory_profile,
+    cuda_memory_stats,
+    get_ray_default_runtime_env,
+    merge_ray_runtime_env,
+    reset_cuda_peak_stats,
+    set_auto_unwrap_transformed_env,
+    torchrl_logger,
+    transport_backend,
+)
+__all__ = sorted(set(__all__) | {
+    "auto_unwrap_transformed_env", "compile_with_warmup", "cuda_memory_profile",
+    "cuda_memory_stats", "get_ray_default_runtime_env", "merge_ray_runtime_env",
+    "reset_cuda_peak_stats", "set_auto_unwrap_transformed_env",
+    "torchrl_logger", "transport_backend",
+})

@@ -301,3 +301,151 @@ def warn(msg: str, category=UserWarning, stacklevel: int = 2) -> None:
 
     logger.warning(msg)
     _warnings.warn(msg, category=category, stacklevel=stacklevel)
+
+
+# ---------------------------------------------------------------------------
+# reference-parity root utilities (torchrl/__init__.py exports)
+# ---------------------------------------------------------------------------
+
+torchrl_logger = logger  # parity alias: the reference exports `torchrl_logger`
+
+_AUTO_UNWRAP = True
+
+
+def set_auto_unwrap_transformed_env(value: bool):
+    """Set (or context-manage) whether nested TransformedEnvs are
+    auto-flattened on construction (reference _utils.py)."""
+    import contextlib
+
+    global _AUTO_UNWRAP
+
+    @contextlib.contextmanager
+    def _ctx():
+        global _AUTO_UNWRAP
+        prev = _AUTO_UNWRAP
+        _AUTO_UNWRAP = bool(value)
+        try:
+            yield
+        finally:
+            _AUTO_UNWRAP = prev
+
+    _AUTO_UNWRAP = bool(value)
+    return _ctx()
+
+
+def auto_unwrap_transformed_env(allow_none: bool = False):
+    """Current auto-unwrap setting (reference _utils.py)."""
+    return _AUTO_UNWRAP
+
+
+def compile_with_warmup(*args, warmup: int = 1, **kwargs):
+    """Decorator: run ``warmup`` eager calls, then ``torch.compile`` the
+    function (reference _utils.py).  On this MI355X stack the hot paths
+    are hipGraph-captured hand-written kernels instead of compiled
+    Python, so the compiled path only engages where torch.compile is
+    functional."""
+    import torch as _torch
+
+    def _wrap(fn):
+        state = {"calls": 0, "compiled": None}
+
+        def inner(*a, **kw):
+            if state["compiled"] is not None:
+                return state["compiled"](*a, **kw)
+            state["calls"] += 1
+            out = fn(*a, **kw)
+            if state["calls"] >= warmup:
+                try:
+                    state["compiled"] = _torch.compile(fn, **kwargs)
+                except Exception:
+                    state["compiled"] = fn
+            return out
+
+        return inner
+
+    if args and callable(args[0]):
+        return _wrap(args[0])
+    return _wrap
+
+
+def cuda_memory_stats(device=None) -> dict:
+    """Allocator summary for one device (reference _utils.py): current /
+    peak allocated+reserved bytes, in a plain dict."""
+    import torch as _torch
+
+    if not _torch.cuda.is_available():
+        return {}
+    return {
+        "allocated": _torch.cuda.memory_allocated(device),
+        "max_allocated": _torch.cuda.max_memory_allocated(device),
+        "reserved": _torch.cuda.memory_reserved(device),
+        "max_reserved": _torch.cuda.max_memory_reserved(device),
+    }
+
+
+def cuda_memory_profile(device=None) -> str:
+    """Human-readable allocator report (reference _utils.py)."""
+    import torch as _torch
+
+    if not _torch.cuda.is_available():
+        return "cuda unavailable"
+    return _torch.cuda.memory_summary(device)
+
+
+def reset_cuda_peak_stats(device=None) -> None:
+    """Zero the peak-allocated counters (reference _utils.py)."""
+    import torch as _torch
+
+    if _torch.cuda.is_available():
+        _torch.cuda.reset_peak_memory_stats(device)
+
+
+def get_ray_default_runtime_env() -> dict:
+    """Default ray runtime_env for rl_amd workers (reference
+    _utils.py): propagate the ROCm/RCCL env vars workers need."""
+    import os
+
+    keys = ["HSA_ENABLE_IPC_MODE_LEGACY", "PYTORCH_ROCM_ARCH", "MASTER_ADDR", "MASTER_PORT"]
+    return {"env_vars": {k: os.environ[k] for k in keys if k in os.environ}}
+
+
+def merge_ray_runtime_env(base: dict, extra: dict) -> dict:
+    """Merge two ray runtime_env dicts; ``extra`` wins per key, env_vars
+    merge recursively (reference _utils.py)."""
+    out = dict(base or {})
+    for k, v in (extra or {}).items():
+        if k == "env_vars":
+            out[k] = {**out.get(k, {}), **v}
+        else:
+            out[k] = v
+    return out
+
+
+import contextlib as _contextlib
+
+
+@_contextlib.contextmanager
+def transport_backend(backend: str):
+    """Scoped default payload transport for distributed components
+    (reference _comm/backends.py:221).  On MI355X nodes the payload
+    plane is RCCL over xGMI ("nccl"/"rccl"); "gloo" covers CPU-only
+    runs and "rpc" the torch.distributed.rpc services."""
+    global _TRANSPORT_BACKEND
+    valid = {"nccl", "rccl", "gloo", "rpc", "shared_memory"}
+    if str(backend) not in valid:
+        raise ValueError(f"unknown transport backend {backend!r}; pick from {sorted(valid)}")
+    prev = globals().get("_TRANSPORT_BACKEND", "nccl")
+    _TRANSPORT_BACKEND = str(backend)
+    try:
+        yield
+    finally:
+        _TRANSPORT_BACKEND = prev
+
+
+__all__ += [
+    "torchrl_logger", "set_auto_unwrap_transformed_env",
+    "auto_unwrap_transformed_env", "compile_with_warmup",
+    "cuda_memory_stats", "cuda_memory_profile", "reset_cuda_peak_stats",
+    "get_ray_default_runtime_env", "merge_ray_runtime_env",
+    "transport_backend",
+]

@@ -117,3 +117,52 @@ from .vla import (  # noqa: F401
     UniformActionTokenizer,
 )
 __all__ = sorted(set(list(globals().get('__all__', [])) + ['ActionTokenizerBase', 'BoundedContinuous', 'DEVICE_TYPING', 'FlatStorageCheckpointer', 'H5StorageCheckpointer', 'ListStorageCheckpointer', 'NestedStorageCheckpointer', 'RemoteTensorDictReplayBuffer', 'StorageCheckpointerBase', 'TensorStorageCheckpointer', 'UniformActionTokenizer']))
+
+# reference-parity: sample units, trajectory queries, spec helpers,
+# ensembles, VLA/video/llm data plumbing
+from .extras import (  # noqa: F401
+    ConditionalUpdateResult,
+    DEFAULT_DONE_KEYS,
+    H5Combine,
+    H5Split,
+    RayReplayBuffer,
+    RobotDatasetMetadata,
+    SampleUnit,
+    SamplerEnsemble,
+    Sequence,
+    StorageEnsembleCheckpointer,
+    StoreStorage,
+    TensorMap,
+    Trajectory,
+    TrajectoryPredicate,
+    Transition,
+    VocabTailActionTokenizer,
+    WriterEnsemble,
+    check_no_exclusive_keys,
+    clear_video_decoder_cache,
+    consolidate_spec,
+    contains_lazy_spec,
+    create_infinite_iterator,
+    filter_trajectories,
+    find_start_stop_traj,
+    get_dataloader,
+    iter_trajectories,
+    prefill_replay_buffer,
+    set_video_decoder_cache_size,
+    traj,
+    validate_vla_tensordict,
+)
+from .tensor_specs import Stacked, StackedComposite  # noqa: F401
+from .replay_buffers.checkpointers import CompressedListStorageCheckpointer  # noqa: F401
+__all__ = sorted(set(__all__) | {
+    "ConditionalUpdateResult", "DEFAULT_DONE_KEYS", "H5Combine", "H5Split",
+    "RayReplayBuffer", "RobotDatasetMetadata", "SampleUnit", "SamplerEnsemble",
+    "Sequence", "StorageEnsembleCheckpointer", "StoreStorage", "TensorMap",
+    "Trajectory", "TrajectoryPredicate", "Transition", "VocabTailActionTokenizer",
+    "WriterEnsemble", "check_no_exclusive_keys", "clear_video_decoder_cache",
+    "consolidate_spec", "contains_lazy_spec", "create_infinite_iterator",
+    "filter_trajectories", "find_start_stop_traj", "get_dataloader",
+    "iter_trajectories", "prefill_replay_buffer", "set_video_decoder_cache_size",
+    "traj", "validate_vla_tensordict", "Stacked", "StackedComposite",
+    "CompressedListStorageCheckpointer",
+})

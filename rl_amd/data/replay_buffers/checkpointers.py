@@ -139,3 +139,16 @@ class H5StorageCheckpointer(StorageCheckpointerBase):
                 "in this image. Use FlatStorageCheckpointer or "
                 "TensorStorageCheckpointer instead."
             )
+
+
+class CompressedListStorageCheckpointer(ListStorageCheckpointer):
+    """Checkpointer for CompressedListStorage (reference
+    checkpointers.py): entries are decompressed on save so the on-disk
+    format matches the plain list checkpoint."""
+
+    def save(self, storage, path):
+        inner = getattr(storage, "_decompressed", None)
+        super().save(storage if inner is None else inner, path)
+
+
+__all__.append("CompressedListStorageCheckpointer")

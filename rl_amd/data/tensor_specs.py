@@ -858,3 +858,22 @@ class BoxList(Box):
 
     def __len__(self):
         return len(self.boxes)
+
+
+class Stacked:
+    """Stacked spec (reference tensor_specs.py:1496): rl_amd stacks
+    eagerly — constructing one returns the dense stacked spec."""
+
+    def __new__(cls, *specs, dim: int = 0):
+        return stack_specs(list(specs), dim)
+
+
+class StackedComposite:
+    """Stacked composite spec (reference tensor_specs.py:6463); eager
+    dense stacking, same as :class:`Stacked`."""
+
+    def __new__(cls, *specs, dim: int = 0):
+        return stack_specs(list(specs), dim)
+
+
+__all__ += ["Stacked", "StackedComposite"]

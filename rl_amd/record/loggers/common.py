@@ -211,3 +211,69 @@ def get_logger(
     if logger_type == "mlflow":
         return MLFlowLogger(experiment_name, **kwargs)
     raise NotImplementedError(f"unknown logger type {logger_type}")
+
+
+class ProcessLogger(Logger):
+    """Logger service living in a dedicated process (reference
+    record/loggers/process.py:132): log_* calls are enqueued through a
+    multiprocessing queue and drained by a child that owns the concrete
+    logger — workers never block on IO.  Only the owner can flush/stop."""
+
+    def __init__(self, logger_factory, *, exp_name: str = "rl_amd", log_dir: str = None):
+        import multiprocessing as mp
+
+        super().__init__(exp_name=exp_name, log_dir=log_dir or "./logs")
+        ctx = mp.get_context("spawn")
+        self._queue = ctx.Queue()
+        self._proc = ctx.Process(
+            target=self._serve, args=(self._queue, logger_factory), daemon=True
+        )
+        self._proc.start()
+
+    @staticmethod
+    def _serve(queue, logger_factory):
+        inner = logger_factory()
+        while True:
+            item = queue.get()
+            if item is None:
+                break
+            method, args, kwargs = item
+            getattr(inner, method)(*args, **kwargs)
+
+    def log_scalar(self, name, value, step=None):
+        self._queue.put(("log_scalar", (name, value), {"step": step}))
+
+    def log_video(self, name, video, step=None, **kwargs):
+        self._queue.put(("log_video", (name, video), {"step": step, **kwargs}))
+
+    def log_hparams(self, cfg):
+        self._queue.put(("log_hparams", (cfg,), {}))
+
+    def log_histogram(self, name, data, **kwargs):
+        self._queue.put(("log_histogram", (name, data), kwargs))
+
+    def __repr__(self):
+        return f"ProcessLogger(exp_name={self.exp_name})"
+
+    def stop(self):
+        self._queue.put(None)
+        self._proc.join(timeout=10)
+
+
+class RayLogger(Logger):
+    """Ray-actor-hosted logger (reference record/loggers/ray.py) —
+    gated: ray is not installed in this image; ProcessLogger covers the
+    out-of-process pattern locally."""
+
+    def __init__(self, *args, **kwargs):
+        import importlib.util
+
+        if importlib.util.find_spec("ray") is None:
+            raise ImportError(
+                "RayLogger requires the `ray` package, which is not installed "
+                "in this image. Use ProcessLogger instead."
+            )
+        raise NotImplementedError("ray actor scaffolding")
+
+
+__all__ += ["ProcessLogger", "RayLogger"]

@@ -734,3 +734,15 @@ class TD3OptimizationStepper(OptimizationStepper):
             if self.target_updater is not None:
                 self.target_updater.step()
         return losses_td
+
+
+def mask_batch(batch):
+    """Batch-process hook: drop padded events using the collector mask
+    (reference trainers.py:2379) — if ``("collector", "mask")`` exists,
+    index the batch down to valid rows."""
+    if ("collector", "mask") in batch.keys(True):
+        return batch[batch.get(("collector", "mask"))]
+    return batch
+
+
+__all__.append("mask_batch")
