@@ -113,6 +113,14 @@ class GraphedRollout:
             probe = self.policy(self._entry.clone(False))
             probe, _ = self.env.step_and_maybe_reset(probe)
         self._alloc_store(probe.get("next"), probe)
+        # the probe advanced the env's internal state one step past the
+        # snapshotted entry observation; re-reset so they agree at the
+        # first collect (envs whose state is not fully in the tensordict)
+        carrier0 = self.env.reset()
+        for k in list(self._entry.keys(True, True)):
+            nv = carrier0.get(k, None)
+            if nv is not None:
+                self._entry.get(k).copy_(nv)
         if self._use_graph:
             if hasattr(self.env, "enable_capture_mode"):
                 self.env.enable_capture_mode(True)

@@ -147,6 +147,20 @@ class ReplayBuffer:
                     pass
                 data.set("_weight", w.reshape(data.batch_size[:1]) if data.batch_size else w)
             data.set("index", index.to(data.device) if data.device else index)
+            trunc = info.get("truncated", None)
+            if trunc is not None:
+                # mark slice boundaries truncated (SliceSampler contract)
+                tkey = info.get("truncated_key", ("next", "truncated"))
+                cur = data.get(tkey, None)
+                trunc = trunc.to(data.device) if data.device else trunc
+                if cur is not None:
+                    data.set(tkey, cur | trunc.reshape(cur.shape))
+                else:
+                    data.set(tkey, trunc.reshape(*data.batch_size, 1))
+                dkey = ("next", "done") if not isinstance(tkey, str) else "done"
+                done = data.get(dkey, None)
+                if done is not None:
+                    data.set(dkey, done | trunc.reshape(done.shape))
         if self._pin_memory and isinstance(data, TensorDictBase):
             data = data.pin_memory()
         data = self._apply_transforms(data)

@@ -398,10 +398,17 @@ class SliceSampler(Sampler):
             slice_starts.unsqueeze(1)
             + torch.arange(slice_len, device=v_starts.device).unsqueeze(0)
         ).reshape(-1)
-        info = {
-            ("next", "truncated") if not isinstance(self.truncated_key, str) else self.truncated_key: None
+        # the last step of every slice is a truncation boundary (reference
+        # SliceSampler sets ('next','truncated') there); the buffer applies
+        # this mask to the gathered batch
+        truncated = torch.zeros(index.numel(), dtype=torch.bool, device=index.device)
+        truncated[slice_len - 1 :: slice_len] = True
+        return index, {
+            "slice_len": slice_len,
+            "num_slices": num_slices,
+            "truncated": truncated,
+            "truncated_key": self.truncated_key,
         }
-        return index, {"slice_len": slice_len, "num_slices": num_slices}
 
 
 class SliceSamplerWithoutReplacement(SliceSampler):
@@ -444,7 +451,14 @@ class SliceSamplerWithoutReplacement(SliceSampler):
             slice_starts.unsqueeze(1)
             + torch.arange(slice_len, device=starts.device).unsqueeze(0)
         ).reshape(-1)
-        return index, {"slice_len": slice_len, "num_slices": num_slices}
+        truncated = torch.zeros(index.numel(), dtype=torch.bool, device=index.device)
+        truncated[slice_len - 1 :: slice_len] = True
+        return index, {
+            "slice_len": slice_len,
+            "num_slices": num_slices,
+            "truncated": truncated,
+            "truncated_key": self.truncated_key,
+        }
 
     def _empty(self):
         self._used = None

@@ -69,11 +69,23 @@ class _EnsembleModule(nn.Module):
 
         if self._batched is not None:
             return self._batched
+        _FAST = (nn.Linear, nn.Tanh, nn.ReLU, nn.ELU, nn.SiLU)
+        _INERT = (nn.Identity,)
         try:
             per_copy = []
             for m in self.modules_list:
-                layers = [sub for sub in m.modules()
-                          if isinstance(sub, (nn.Linear, nn.Tanh, nn.ReLU, nn.ELU, nn.SiLU))]
+                # the fast path must account for EVERY compute submodule:
+                # any leaf outside the whitelist (LayerNorm, Dropout, GELU,
+                # BatchNorm, ...) forces the exact per-copy loop
+                leaves = [
+                    sub
+                    for sub in m.modules()
+                    if next(iter(sub.children()), None) is None
+                ]
+                if not all(isinstance(l, _FAST + _INERT) for l in leaves):
+                    self._batched = False
+                    return False
+                layers = [sub for sub in m.modules() if isinstance(sub, _FAST)]
                 per_copy.append(layers)
             sig = [
                 (type(l).__name__,

@@ -95,7 +95,14 @@ class SACLoss(LossModule):
                 if spec is not None and hasattr(spec, "shape") and len(spec.shape):
                     te = -float(np.prod(spec.shape[-1:]))
                 else:
-                    te = -1.0
+                    # matching the reference (torchrl/objectives/sac.py):
+                    # silently guessing -1.0 mis-tunes alpha for
+                    # multi-dim actions, so refuse instead
+                    raise RuntimeError(
+                        "target_entropy='auto' requires the actor to expose an "
+                        "action spec (actor_network.spec with a non-empty shape). "
+                        "Pass a numeric target_entropy instead."
+                    )
             self._target_entropy_val = float(te)
         return self._target_entropy_val
 

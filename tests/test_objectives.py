@@ -74,11 +74,14 @@ def make_actor():
         MLP(in_features=OBS, out_features=2 * ACT, num_cells=[32]),
         NormalParamExtractor(),
     )
+    from rl_amd.data import Bounded
+
     return ProbabilisticActor(
         TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"]),
         in_keys=["loc", "scale"],
         distribution_class=TanhNormal,
         return_log_prob=True,
+        spec=Bounded(-1.0, 1.0, shape=(ACT,)),
     )
 
 
@@ -539,9 +542,12 @@ class TestBatchedEnsemble:
         net = torch.nn.Sequential(
             MLP(in_features=3, out_features=4, num_cells=[16]), NormalParamExtractor()
         )
+        from rl_amd.data import Bounded
+
         actor = ProbabilisticActor(
             TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"]),
             in_keys=["loc", "scale"], distribution_class=TanhNormal, return_log_prob=True,
+            spec=Bounded(-1.0, 1.0, shape=(2,)),
         )
         q = ValueOperator(MLP(in_features=5, out_features=1, num_cells=[16]), in_keys=["observation", "action"])
         loss = SACLoss(actor, q, num_qvalue_nets=2)

@@ -33,11 +33,14 @@ def make_cont_actor_critic(obs_dim=7, act_dim=5):
         MLP(in_features=obs_dim, out_features=2 * act_dim, num_cells=[32]),
         NormalParamExtractor(),
     )
+    from rl_amd.data import Bounded
+
     actor = ProbabilisticActor(
         TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"]),
         in_keys=["loc", "scale"],
         distribution_class=TanhNormal,
         return_log_prob=True,
+        spec=Bounded(-1.0, 1.0, shape=(act_dim,)),
     )
     critic = ValueOperator(
         MLP(in_features=obs_dim, out_features=1, num_cells=[32]),
@@ -255,11 +258,14 @@ class TestMoreAlgorithmTrainers:
             MLP(in_features=7, out_features=2 * 5, num_cells=[16]),
             NormalParamExtractor(),
         )
+        from rl_amd.data import Bounded
+
         actor = ProbabilisticActor(
             TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"]),
             in_keys=["loc", "scale"],
             distribution_class=TanhNormal,
             return_log_prob=True,
+            spec=Bounded(-1.0, 1.0, shape=(5,)),
         )
         qvalue = ValueOperator(
             MLP(in_features=12, out_features=1, num_cells=[16]),
