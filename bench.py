@@ -4,18 +4,22 @@ Metric (BASELINE.json): frames/sec (whole job) for PPO with 4096 vec-envs,
 HalfCheetah-v4 shapes (obs 17 / act 6), synthetic dynamics + random-init
 weights (no network for datasets), bf16 compute, GAE on the fused HIP scan.
 
-One full PPO iteration per step:
-  rollout T env steps x 4096 envs (policy sample + env step, on-device,
-  written into a pre-allocated [B, T] HBM store; with --graph the whole
-  T-step rollout is captured ONCE as a hipGraph and replayed per iter)
-  -> GAE (fused HIP scan) -> 1 epoch of 4 minibatch ClipPPO updates
-  -> optimizer step (+ flat RCCL all-reduce when world_size > 1).
+The loop is constructed ONLY from public library classes:
+  HalfCheetahVec env → Collector (GPU fast path: mega-kernel / hipGraph
+  rollout) → GAE (fused HIP scan) → ClipPPOLoss → Adam, orchestrated by
+  trainers.GraphedPPO which captures the ENTIRE iteration as one hipGraph
+  at world=1 and, when distributed, captures the minibatch fwd+bwd and
+  overlaps the bucketed RCCL gradient all-reduce (parallel.GradAllReducer)
+  on a second HIP stream.
 
 Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
       --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
 Weak scaling: each rank owns its own 4096 envs; gradients all-reduce.
+
+Secondary configs (same JSON contract): --config sac|impala|rlhf|dqn
+dispatch to benchmarks/bench_<config>.py.
 """
 from __future__ import annotations
 
@@ -29,26 +33,19 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+from rl_amd.collectors import Collector
 from rl_amd.envs.custom.synthetic import HalfCheetahVec
-from rl_amd.envs.utils import ExplorationType, set_exploration_type
 from rl_amd.modules import MLP, NormalParamExtractor, ProbabilisticActor, TanhNormal, ValueOperator
 from rl_amd.objectives import ClipPPOLoss
 from rl_amd.objectives.value.advantages import GAE
 from rl_amd.tensordict import TensorDict, TensorDictModule
-
-STORE_KEYS = [
-    "observation",
-    "action",
-    "sample_log_prob",
-    ("next", "observation"),
-    ("next", "reward"),
-    ("next", "done"),
-    ("next", "terminated"),
-]
+from rl_amd.trainers import GraphedPPO
 
 
-def parse_args():
+def parse_args(argv=None):
     p = argparse.ArgumentParser()
+    p.add_argument("--config", choices=["ppo", "sac", "impala", "rlhf", "dqn"],
+                   default="ppo")
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
@@ -58,13 +55,18 @@ def parse_args():
     p.add_argument("--epochs", type=int, default=1)
     p.add_argument("--hidden", type=int, default=64)
     p.add_argument("--device", type=str, default=None)
-    p.add_argument("--graph", action="store_true", help="hipGraph-capture the rollout")
+    p.add_argument("--seed", type=int, default=1234)
+    p.add_argument("--same-seed", action="store_true",
+                   help="identical seed on every rank (numerics tests: "
+                        "world=N must then reproduce world=1 bit-for-bit)")
+    p.add_argument("--dump-params", type=str, default=None,
+                   help="save actor+critic state_dict after the timed run")
+    p.add_argument("--graph", action="store_true", help="hipGraph capture")
     p.add_argument("--no-graph", dest="graph", action="store_false")
     p.add_argument("--fused-actor", action="store_true",
-                   help="single-kernel MLP+TanhNormal rollout policy")
+                   help="(kept for compat; the Collector fast path fuses automatically)")
     p.add_argument("--no-fused-actor", dest="fused_actor", action="store_false")
-    p.add_argument("--full-graph", dest="full_graph", action="store_true",
-                   help="capture rollout+GAE+PPO update as one hipGraph")
+    p.add_argument("--full-graph", dest="full_graph", action="store_true")
     p.add_argument("--no-full-graph", dest="full_graph", action="store_false")
     p.add_argument("--splitk", dest="splitk", action="store_true",
                    help="split-K HIP wgrad kernel in the update backward")
@@ -72,7 +74,7 @@ def parse_args():
     p.add_argument("--track-reward", action="store_true",
                    help="print mean store reward every 20 steps (learning sanity)")
     p.set_defaults(graph=True, fused_actor=True, full_graph=True, splitk=True)
-    return p.parse_args()
+    return p.parse_args(argv)
 
 
 def build(args, device):
@@ -108,8 +110,7 @@ def build(args, device):
     return env, actor, critic
 
 
-def main():
-    args = parse_args()
+def run_ppo(args):
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
@@ -123,16 +124,17 @@ def main():
         device = torch.device("cpu")
     if cuda:
         torch.cuda.set_device(device)
-    dtype = torch.bfloat16
 
     distributed = world > 1
-    if distributed:
+    if distributed and not torch.distributed.is_initialized():
         torch.distributed.init_process_group(backend="nccl" if cuda else "gloo")
 
-    torch.manual_seed(1234 + rank)
+    seed = args.seed if args.same_seed else args.seed + rank
+    torch.manual_seed(seed)
     env, actor, critic = build(args, device)
-    env.set_seed(1234 + rank)
+    env.set_seed(seed)
 
+    refresh_hook = None
     if args.splitk and cuda:
         # minibatch backward wgrad via the split-K kernel (csrc/wgrad.hip):
         # hipBLASLt runs these skinny [H,16k]x[16k,H] reductions on one
@@ -152,17 +154,14 @@ def main():
             # under graph capture; r29 profile: ~130 cast kernels/step)
             enable_splitk_bf16_cache(actor)
             enable_splitk_bf16_cache(critic)
-            # NOTE: ops.fuse_mlp3 (whole-MLP fused fwd/bwd) was measured
-            # SLOWER here: at 16k-64k-row minibatches the GEMMs belong
-            # on hipBLASLt's MFMA path, not the fused kernel's VALU
-            # dots (r37: T=64 7.49 ms vs 5.18).  It stays available for
-            # launch-bound small-batch updates.
+
+            def refresh_hook():
+                refresh_splitk_caches(actor, critic)
         else:
             args.splitk = False
 
     params = list(actor.parameters()) + list(critic.parameters())
-    # fused Adam: one multi-tensor kernel instead of ~10 elementwise
-    # launches per step (the r15 profile's long elementwise tail)
+    # fused multi-tensor Adam, capturable for full-step graph capture
     optim_kwargs = dict(lr=3e-4, capturable=bool(args.full_graph and cuda and world == 1))
 
     def _fused_adam_ok():
@@ -178,303 +177,36 @@ def main():
         optim = torch.optim.Adam(params, fused=True, **optim_kwargs)
     else:
         optim = torch.optim.Adam(params, **optim_kwargs)
+
     loss_mod = ClipPPOLoss(actor, critic, clip_epsilon=0.2, entropy_coeff=0.01,
                            critic_coeff=0.5, normalize_advantage=True)
     gae = GAE(gamma=0.99, lmbda=0.95, value_network=critic, vectorized=True)
 
     T, B = args.horizon, args.envs
     frames_per_step = T * B
-    # cache_enabled=False: the autocast weight cache allocates during
-    # hipGraph capture, which is forbidden mid-capture
-    autocast = torch.autocast(
-        device_type="cuda", dtype=dtype, enabled=cuda, cache_enabled=False
-    )
-    obs_dim, act_dim = env.obs_dim, env.act_dim
 
-    # pre-allocated [B, T] rollout store resident in HBM
-    store = TensorDict(
-        {
-            "observation": torch.zeros(B, T, obs_dim, device=device),
-            "action": torch.zeros(B, T, act_dim, device=device),
-            "sample_log_prob": torch.zeros(B, T, device=device),
-            "next": {
-                "observation": torch.zeros(B, T, obs_dim, device=device),
-                "reward": torch.zeros(B, T, 1, device=device),
-                "done": torch.zeros(B, T, 1, dtype=torch.bool, device=device),
-                "terminated": torch.zeros(B, T, 1, dtype=torch.bool, device=device),
-            },
-        },
-        batch_size=[B, T],
-        device=device,
+    collector = Collector(
+        env,
+        actor,
+        frames_per_batch=frames_per_step,
+        total_frames=-1,
+        use_graph="auto" if (args.graph and cuda) else False,
     )
 
-    rollout_policy = actor
-    if args.fused_actor and cuda:
-        try:
-            from rl_amd.ops import FusedTanhNormalActor
-
-            rollout_policy = FusedTanhNormalActor(actor)
-        except Exception as e:
-            print(f"[bench] fused actor unavailable ({e!r}); eager policy", file=sys.stderr)
-            args.fused_actor = False
-
-    carrier0 = env.reset()
-    entry_obs = carrier0.get("observation")  # static entry buffer
-
-    def rollout_body():
-        carrier = TensorDict(
-            {"observation": entry_obs}, batch_size=[B], device=device
-        )
-        with torch.no_grad(), set_exploration_type(ExplorationType.RANDOM):
-            for t in range(T):
-                if args.fused_actor and cuda:
-                    carrier = rollout_policy(carrier)
-                else:
-                    with autocast:
-                        carrier = rollout_policy(carrier)
-                carrier.set("action", carrier.get("action").float())
-                carrier, next_root = env.step_and_maybe_reset(carrier)
-                for k in STORE_KEYS:
-                    col = store.get(k)[:, t]
-                    col.copy_(carrier.get(k).reshape(col.shape))
-                carrier = next_root
-        # close the loop: replays start from the final observation
-        entry_obs.copy_(carrier.get("observation"))
-
-    # store-direct rollout: the fused actor and env kernels write action,
-    # log-prob, pre/post observations, reward and done STRAIGHT into the
-    # strided [B, T] store (zero copy kernels), and the env kernel
-    # auto-resets the carried state — 2 + 2T launches per iteration.
-    store_direct = bool(args.fused_actor and args.graph and cuda)
-    if store_direct:
-        try:
-            from rl_amd import _C
-            from rl_amd.ops import HAS_HIP_EXT
-
-            store_direct = HAS_HIP_EXT and hasattr(_C, "fused_actor_into")
-        except Exception:
-            store_direct = False
-    if store_direct:
-        fa = rollout_policy  # FusedTanhNormalActor
-        w1, w2, w3 = (l.weight for l in fa.linears)
-        b1, b2, b3 = (l.bias for l in fa.linears)
-        s_obs = store.get("observation")
-        s_act = store.get("action")
-        s_lp = store.get("sample_log_prob")
-        s_nobs = store.get(("next", "observation"))
-        s_rew = store.get(("next", "reward"))
-        s_done = store.get(("next", "done"))
-        store.get(("next", "terminated")).zero_()  # env never terminates
-        env.enable_capture_mode(True)
-
-        mega = hasattr(_C, "fused_rollout")
-
-        def rollout_body_direct():
-            with torch.no_grad():
-                eps_all = torch.randn(T, B, act_dim, device=device)
-                noise_all = torch.randn(T, B, obs_dim, device=device) * 0.1
-                if mega:
-                    # env rows are independent: the whole T-step rollout
-                    # runs as ONE kernel (csrc/rollout_fused.hip) — no
-                    # per-step launch latency at all
-                    _C.fused_rollout(
-                        env._state, env._t.reshape(-1), w1, b1, w2, b2, w3,
-                        b3, env.A, env.B, eps_all, noise_all, s_obs, s_act,
-                        s_lp, s_nobs, s_rew, s_done, float(env.max_steps),
-                        fa.inv_softplus_bias, fa.scale_lb,
-                    )
-                    return
-                for t in range(T):
-                    _C.fused_actor_into(
-                        env._state, w1, b1, w2, b2, w3, b3, eps_all[t],
-                        s_act[:, t], s_lp[:, t], fa.inv_softplus_bias,
-                        fa.scale_lb,
-                    )
-                    _C.synthetic_env_step_into(
-                        env._state, s_act[:, t], env.A, env.B,
-                        env._t.reshape(-1), s_nobs[:, t], s_obs[:, t],
-                        s_rew[:, t], s_done[:, t], noise_all[t],
-                        float(env.max_steps),
-                    )
-
-        rollout_body = rollout_body_direct
-
-    rollout = rollout_body
-    if args.graph and cuda:
-        env.enable_capture_mode(True)
-        try:
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                for _ in range(3):
-                    rollout_body()
-            torch.cuda.current_stream().wait_stream(side)
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                rollout_body()
-            rollout = g.replay
-        except Exception as e:  # capture unsupported → eager fallback
-            import traceback
-
-            traceback.print_exc(file=sys.stderr)
-            print(f"[bench] hipGraph capture failed ({e!r}); running eager", file=sys.stderr)
-            rollout = rollout_body
-            args.graph = False
-
-    # full-step graph capture needs stable grad buffers
-    zero_set_to_none = not (args.full_graph and cuda and not distributed)
-
-    # Distributed mode cannot capture the whole step (the RCCL
-    # all-reduce sits between backward and optimizer), but the
-    # minibatch fwd+bwd — the launch-heavy part — CAN be captured and
-    # replayed with the comm/step/refresh left eager, keeping the
-    # per-GPU step close to the single-GPU captured one.
-    mb_graph = {"graph": None, "sub": None, "total": None}
-
-    def _mb_fwd_bwd(sub):
-        with autocast:
-            out = loss_mod(sub)
-            total = (
-                out.get("loss_objective")
-                + out.get("loss_critic")
-                + out.get("loss_entropy")
-            )
-        optim.zero_grad(set_to_none=False)
-        total.backward()
-        return total
-
-    def _capture_mb_graph(example_sub):
-        static_sub = example_sub.clone(False)
-        for k in list(static_sub.keys(True, True)):
-            static_sub.set(k, static_sub.get(k).clone())
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            for _ in range(3):
-                _mb_fwd_bwd(static_sub)
-        torch.cuda.current_stream().wait_stream(side)
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            _mb_fwd_bwd(static_sub)
-        mb_graph["graph"] = g
-        mb_graph["sub"] = static_sub
-
-    use_mb_graph = bool(
-        (distributed or os.environ.get("RL_AMD_BENCH_FORCE_MB_GRAPH") == "1")
-        and cuda
-        and args.full_graph
+    runner = GraphedPPO(
+        collector,
+        gae,
+        loss_mod,
+        optim,
+        minibatches=args.minibatches,
+        epochs=args.epochs,
+        capture="auto" if (args.full_graph and cuda) else False,
+        post_optim_hook=refresh_hook,
     )
-
-    def update_phase():
-        batch = store
-        with torch.no_grad(), autocast:
-            gae(batch)
-        flat = batch.reshape(-1)
-        n = flat.batch_size[0]
-        mb = n // args.minibatches
-        for _ in range(args.epochs):
-            perm = torch.randperm(n, device=device)
-            # one gather of the whole flat store (7 kernels), then the
-            # minibatches are contiguous zero-copy slices — instead of
-            # 7 gathers per minibatch
-            shuffled = flat[perm]
-            for i in range(args.minibatches):
-                sub = shuffled[i * mb : (i + 1) * mb]
-                if use_mb_graph:
-                    if mb_graph["graph"] is None:
-                        try:
-                            _capture_mb_graph(sub)
-                        except Exception:
-                            import traceback
-
-                            traceback.print_exc(file=sys.stderr)
-                            mb_graph["graph"] = False
-                    if mb_graph["graph"] not in (None, False):
-                        static_sub = mb_graph["sub"]
-                        for k in list(static_sub.keys(True, True)):
-                            static_sub.get(k).copy_(sub.get(k))
-                        mb_graph["graph"].replay()
-                    else:
-                        with autocast:
-                            out = loss_mod(sub)
-                            total = (
-                                out.get("loss_objective")
-                                + out.get("loss_critic")
-                                + out.get("loss_entropy")
-                            )
-                        optim.zero_grad(set_to_none=zero_set_to_none)
-                        total.backward()
-                else:
-                    with autocast:
-                        out = loss_mod(sub)
-                        total = (
-                            out.get("loss_objective")
-                            + out.get("loss_critic")
-                            + out.get("loss_entropy")
-                        )
-                    optim.zero_grad(set_to_none=zero_set_to_none)
-                    total.backward()
-                if distributed:
-                    with torch.no_grad():
-                        flat_grads = torch.cat(
-                            [p.grad.reshape(-1) for p in params if p.grad is not None]
-                        )
-                        torch.distributed.all_reduce(flat_grads)
-                        flat_grads /= world
-                        off = 0
-                        for p in params:
-                            if p.grad is not None:
-                                k = p.grad.numel()
-                                p.grad.copy_(flat_grads[off : off + k].view_as(p.grad))
-                                off += k
-                torch.nn.utils.clip_grad_norm_(params, 1.0)
-                optim.step()
-                if args.splitk and cuda:
-                    refresh_splitk_caches(actor, critic)
-
-    def one_step():
-        rollout()
-        update_phase()
-
-    def one_step_eager_rollout():
-        # used inside full-step capture: a graph cannot replay another
-        # graph, so the rollout body runs inline
-        rollout_body()
-        update_phase()
-
-    step_fn = one_step
-    if args.full_graph and B * T > 2_000_000:
-        # full-step capture at multi-million-frame stores exhausted the
-        # capture pool (core dump at 65536 envs x T=64); the rollout
-        # mega-kernel path carries those sizes without the whole-step
-        # graph (measured 103M frames/s at 65536 x T=16)
-        print("[bench] store > 2M frames: full-step graph disabled", file=sys.stderr)
-        args.full_graph = False
-    if args.full_graph and cuda and not distributed and not use_mb_graph:
-        # capture EVERYTHING (rollout + GAE + fwd/bwd/Adam): grads must be
-        # pre-allocated and kept (set_to_none=False) so buffers are static
-        try:
-            env.enable_capture_mode(True)
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                for _ in range(3):
-                    one_step_eager_rollout()
-            torch.cuda.current_stream().wait_stream(side)
-            g_full = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g_full):
-                one_step_eager_rollout()
-            step_fn = g_full.replay
-        except Exception as e:
-            import traceback
-
-            traceback.print_exc(file=sys.stderr)
-            print(f"[bench] full-step capture failed ({e!r}); rollout-graph only", file=sys.stderr)
-            args.full_graph = False
-            step_fn = one_step
+    runner.initialize()
 
     for _ in range(args.warmup):
-        step_fn()
+        runner.step()
 
     if distributed:
         torch.distributed.barrier()
@@ -482,10 +214,12 @@ def main():
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for it in range(args.steps):
-        step_fn()
+        runner.step()
         if args.track_reward and it % 20 == 0:
-            r = store.get(("next", "reward")).float().mean().item()
-            print(f"[reward] step {it}: {r:.4f}", file=sys.stderr)
+            store = collector._graphed.store if collector._graphed is not None else None
+            if store is not None:
+                r = store.get(("next", "reward")).float().mean().item()
+                print(f"[reward] step {it}: {r:.4f}", file=sys.stderr)
     if cuda:
         torch.cuda.synchronize()
     if distributed:
@@ -501,6 +235,14 @@ def main():
     fps = world * frames_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    if args.dump_params:
+        sd = {
+            "actor": {k: v.detach().cpu() for k, v in actor.state_dict().items()},
+            "critic": {k: v.detach().cpu() for k, v in critic.state_dict().items()},
+        }
+        torch.save(sd, args.dump_params + (f".rank{rank}" if world > 1 else ""))
+
+    graphed = collector._graphed
     if rank == 0:
         result = {
             "metric": "frames_per_sec_ppo_4096envs",
@@ -523,17 +265,47 @@ def main():
                 "n_envs_per_gpu": B,
                 "ppo_epochs": args.epochs,
                 "minibatches": args.minibatches,
-                "hip_graph": bool(args.graph and cuda),
-                "fused_actor": bool(args.fused_actor and cuda),
+                "library_api": True,
+                "hip_graph": bool(graphed is not None and (graphed.captured or graphed.mega)),
+                "store_direct_rollout": bool(graphed is not None and graphed.mega),
                 "splitk_wgrad": bool(args.splitk and cuda),
-                "store_direct_rollout": bool(store_direct),
-                "full_step_graph": bool(args.full_graph and cuda and not distributed and not use_mb_graph),
-                "minibatch_graph": bool(use_mb_graph and mb_graph["graph"] not in (None, False)),
+                "full_step_graph": runner.full_graph,
+                "minibatch_graph": runner.minibatch_graph,
+                "overlapped_comm": bool(runner.reducer is not None),
             },
         }
         print(json.dumps(result))
     if distributed:
         torch.distributed.destroy_process_group()
+
+
+def _dispatch_secondary(config: str, argv):
+    """Run benchmarks/bench_<config>.py's main() with the remaining args."""
+    import importlib.util
+
+    name = {"sac": "bench_sac", "impala": "bench_impala",
+            "rlhf": "bench_rlhf", "dqn": "bench_dqn_cpu"}[config]
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "benchmarks", f"{name}.py")
+    spec = importlib.util.spec_from_file_location(name, path)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    sys.argv = [path] + argv
+    mod.main()
+
+
+def main():
+    argv = sys.argv[1:]
+    if "--config" in argv:
+        i = argv.index("--config")
+        config = argv[i + 1]
+        rest = argv[:i] + argv[i + 2 :]
+        if config != "ppo":
+            _dispatch_secondary(config, rest)
+            return
+        argv = rest
+    args = parse_args(argv)
+    run_ppo(args)
 
 
 if __name__ == "__main__":

@@ -28,6 +28,7 @@ from rl_amd.testing import DiscreteActionVecMockEnv
 
 def main():
     p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--envs", type=int, default=16)

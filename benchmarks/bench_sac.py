@@ -32,6 +32,7 @@ from rl_amd.tensordict import TensorDict, TensorDictModule, stack as td_stack
 
 def main():
     p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=50)
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch", type=int, default=256)

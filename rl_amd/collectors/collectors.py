@@ -105,6 +105,7 @@ class Collector(BaseCollector):
         cudagraph_policy: bool = False,
         no_cuda_sync: bool = False,
         weight_updater=None,
+        use_graph: Union[bool, str] = "auto",
         **kwargs,
     ):
         if isinstance(create_env_fn, EnvBase):
@@ -161,6 +162,11 @@ class Collector(BaseCollector):
 
         self._frames = 0
         self._iter = 0
+        # GPU fast path: hipGraph-captured (or mega-kernel) rollout into a
+        # static [B, T] HBM store — engaged lazily on the first rollout
+        self.use_graph = use_graph
+        self._graphed = None
+        self._graphed_tried = False
         # persistent carrier (reference _make_carrier:1388)
         self._carrier: Optional[TensorDictBase] = None
         # trajectory ids (reference _update_traj_ids:1985)
@@ -191,8 +197,84 @@ class Collector(BaseCollector):
             ids[done_flat.reshape(-1)] = new_ids
             self._traj_ids = ids.reshape(self._traj_ids.shape)
 
+    def _try_graphed(self):
+        """Build the GraphedRollout fast path when eligible: GPU-resident
+        vectorized env with masked auto-reset + GPU policy.  The whole
+        T-step loop then runs as one hipGraph replay (or, when the env
+        advertises ``supports_fused_rollout`` for the policy, as a single
+        mega-kernel launch) writing straight into a static [B, T] HBM
+        store.  The returned batch is that store, overwritten per
+        iteration (``return_same_td`` semantics); ('collector','traj_ids')
+        is not stamped on this path."""
+        if not self.use_graph:
+            return None
+        if not (
+            torch.cuda.is_available()
+            and self.env.device is not None
+            and self.env.device.type == "cuda"
+            and len(self.env_batch) == 1
+            and self.steps_per_batch >= 1
+            and self.interruptor is None
+            and self.init_random_frames == 0
+            and not self.reset_at_each_iter
+            and (self.storing_device is None or self.storing_device == self.env.device)
+            and not self._cast_policy
+            and getattr(self.env, "_supports_masked_reset", False)
+        ):
+            return None
+        policy = self.policy
+        try:
+            from ..ops import HAS_HIP_EXT, FusedTanhNormalActor
+
+            if HAS_HIP_EXT and not isinstance(policy, FusedTanhNormalActor):
+                try:
+                    policy = FusedTanhNormalActor(policy)
+                except Exception:
+                    policy = self.policy
+        except Exception:
+            pass
+        try:
+            from .graph import GraphedRollout
+
+            gr = GraphedRollout(
+                self.env,
+                policy,
+                horizon=self.steps_per_batch,
+                exploration_type=self.exploration_type,
+            ).initialize()
+        except Exception:
+            if self.use_graph is True:
+                raise
+            return None
+        if not (gr.mega or gr.captured) and self.use_graph == "auto":
+            # neither capture nor the mega-kernel engaged: the graphed
+            # body would just be the eager loop without collector
+            # metadata — keep the standard path
+            return None
+        return gr
+
+    def rollout_inline(self) -> TensorDictBase:
+        """``rollout`` variant safe to run inside an OUTER hipGraph
+        capture: the fast path runs its uncaptured body (a graph cannot
+        replay another graph).  Used by trainers.GraphedPPO."""
+        if not self._graphed_tried:
+            self._graphed_tried = True
+            self._graphed = self._try_graphed()
+        if self._graphed is not None:
+            self._graphed.body()
+            self._frames += self.frames_per_batch
+            return self._graphed.store
+        return self.rollout()
+
     def rollout(self) -> TensorDictBase:
         """The per-step hot loop (reference _single.py:2014)."""
+        if not self._graphed_tried:
+            self._graphed_tried = True
+            self._graphed = self._try_graphed()
+        if self._graphed is not None:
+            batch = self._graphed.collect()
+            self._frames += self.frames_per_batch
+            return batch
         if self._carrier is None or self.reset_at_each_iter:
             self._carrier = self.env.reset()
         carrier = self._carrier

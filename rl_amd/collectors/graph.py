@@ -70,6 +70,14 @@ class GraphedRollout:
         self.store: Optional[TensorDictBase] = None
         self._entry: Optional[TensorDictBase] = None
         self._warmup_iters = warmup_iters
+        # whole-rollout mega-kernel mode: the env itself runs the entire
+        # [B, T] rollout as ONE HIP launch (csrc/rollout_fused.hip) when
+        # it advertises support for this (policy, env) pair
+        self._mega = bool(
+            hasattr(env, "supports_fused_rollout")
+            and hasattr(env, "fused_rollout_into")
+            and env.supports_fused_rollout(policy)
+        )
 
     def _alloc_store(self, example_next: TensorDictBase, example_root: TensorDictBase):
         store = TensorDict({}, batch_size=[self.B, self.T], device=self.device)
@@ -85,6 +93,16 @@ class GraphedRollout:
                 ),
             )
         self.store = store
+
+    def body(self):
+        """One uncaptured T-step rollout into the store — exposed so an
+        outer full-step hipGraph (trainers.GraphedPPO) can inline it
+        (a graph cannot replay another graph)."""
+        if self._mega:
+            with torch.no_grad():
+                self.env.fused_rollout_into(self.policy, self.store)
+            return
+        self._body()
 
     def _body(self):
         carrier = self._entry.clone(False)
@@ -105,7 +123,34 @@ class GraphedRollout:
             if nv is not None:
                 self._entry.get(k).copy_(nv)
 
+    def _initialize_mega(self) -> "GraphedRollout":
+        env = self.env
+        B, T = self.B, self.T
+        dev = self.device
+        self.store = TensorDict(
+            {
+                "observation": torch.zeros(B, T, env.obs_dim, device=dev),
+                "action": torch.zeros(B, T, env.act_dim, device=dev),
+                "sample_log_prob": torch.zeros(B, T, device=dev),
+                "next": {
+                    "observation": torch.zeros(B, T, env.obs_dim, device=dev),
+                    "reward": torch.zeros(B, T, 1, device=dev),
+                    "done": torch.zeros(B, T, 1, dtype=torch.bool, device=dev),
+                    "terminated": torch.zeros(B, T, 1, dtype=torch.bool, device=dev),
+                },
+            },
+            batch_size=[B, T],
+            device=dev,
+        )
+        if hasattr(env, "enable_capture_mode"):
+            env.enable_capture_mode(True)
+        env.reset()
+        self._entry = TensorDict({}, batch_size=[B], device=dev)  # unused in mega mode
+        return self
+
     def initialize(self) -> "GraphedRollout":
+        if self._mega:
+            return self._initialize_mega()
         carrier0 = self.env.reset()
         self._entry = carrier0.clone()
         # probe one policy step for store allocation
@@ -146,12 +191,19 @@ class GraphedRollout:
     def captured(self) -> bool:
         return self._graph is not None
 
+    @property
+    def mega(self) -> bool:
+        """True when the rollout runs as the single-launch mega-kernel."""
+        return self._mega
+
     def collect(self) -> TensorDictBase:
         """Run one T-step rollout; returns the [B, T] store (overwritten
         per call — clone if you need to keep it)."""
         if self._entry is None:
             self.initialize()
-        if self._graph is not None:
+        if self._mega:
+            self.body()
+        elif self._graph is not None:
             self._graph.replay()
         else:
             self._body()

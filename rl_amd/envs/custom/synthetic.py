@@ -200,6 +200,81 @@ class SyntheticMuJoCoEnv(EnvBase):
             self._gen.manual_seed(seed)
         return seed
 
+    # ------------------------------------------------------------------ #
+    # Whole-rollout mega-kernel protocol (consumed by GraphedRollout /
+    # Collector fast path): env rows are independent, so the entire
+    # T-step rollout of (fused actor + this env) runs as ONE kernel
+    # (csrc/rollout_fused.hip).
+    # ------------------------------------------------------------------ #
+    def supports_fused_rollout(self, policy) -> bool:
+        """True when (policy, self) can run as the single-launch rollout
+        mega-kernel: GPU-resident fp32 env, 3-Linear fused TanhNormal
+        actor, and the weight set fits the 160 KB LDS budget."""
+        if self.device is None or self.device.type != "cuda" or self.dtype != torch.float32:
+            return False
+        from ... import ops
+
+        if not ops.HAS_HIP_EXT:
+            return False
+        from ... import _C
+
+        if not hasattr(_C, "fused_rollout"):
+            return False
+        if not isinstance(policy, ops.FusedTanhNormalActor):
+            return False
+        linears = policy.linears
+        S, H1, H2 = self.obs_dim, linears[0].out_features, linears[1].out_features
+        if linears[0].in_features != S or linears[2].out_features != 2 * self.act_dim:
+            return False
+        return _C.fused_rollout_lds_ok(S, H1, H2, self.act_dim) if hasattr(
+            _C, "fused_rollout_lds_ok"
+        ) else self._fused_rollout_lds_ok(S, H1, H2, self.act_dim)
+
+    @staticmethod
+    def _fused_rollout_lds_ok(S, H1, H2, A):
+        # mirror of csrc/rollout_fused.hip fused_rollout_lds_bytes
+        RO_ROWS = 8
+        w1s, w2s, w3s, as_ = S | 1, H1 | 1, H2 | 1, S | 1
+        bufw = max(S, H1, H2, 2 * A)
+        floats = (
+            H1 * w1s + H1 + H2 * w2s + H2 + 2 * A * w3s + 2 * A
+            + S * as_ + A * as_ + RO_ROWS * as_ + RO_ROWS * A
+            + 2 * RO_ROWS * bufw + RO_ROWS
+        )
+        return 4 * floats <= 160 * 1024
+
+    def fused_rollout_into(self, policy, store) -> None:
+        """Run one [B, T] rollout with ``policy`` (a FusedTanhNormalActor)
+        into the pre-allocated store (keys: observation, action,
+        sample_log_prob, next/observation, next/reward, next/done).
+        Carried state auto-resets; the caller owns loop closure."""
+        from ... import _C
+
+        B, T = store.batch_size
+        if self._state is None:
+            self.reset()
+        w1, w2, w3 = policy.linears
+        eps_all = torch.randn(T, B, self.act_dim, device=self.device)
+        noise_all = torch.randn(T, B, self.obs_dim, device=self.device) * 0.1
+        _C.fused_rollout(
+            self._state,
+            self._t.reshape(-1),
+            w1.weight, w1.bias,
+            w2.weight, w2.bias,
+            w3.weight, w3.bias,
+            self.A, self.B,
+            eps_all, noise_all,
+            store.get("observation"),
+            store.get("action"),
+            store.get("sample_log_prob"),
+            store.get(("next", "observation")),
+            store.get(("next", "reward")),
+            store.get(("next", "done")),
+            float(self.max_steps),
+            policy.inv_softplus_bias,
+            policy.scale_lb,
+        )
+
 
 class HalfCheetahVec(SyntheticMuJoCoEnv):
     """HalfCheetah-v4 shapes: obs 17, act 6."""

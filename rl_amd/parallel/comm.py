@@ -168,7 +168,14 @@ class GradAllReducer:
         world_size: Optional[int] = None,
         bucket_bytes: int = DEFAULT_BUCKET_BYTES,
         group=None,
+        register_hooks: bool = True,
     ):
+        """``register_hooks=True`` overlaps bucket all-reduces with an
+        EAGER backward (hooks fire as grads accumulate).  With
+        ``register_hooks=False`` the backward may be a hipGraph replay
+        (hooks cannot fire inside a replay): call :meth:`reduce` after
+        the replay — buckets still go out as async RCCL calls on the
+        comm stream, then :meth:`finalize` applies them."""
         self.params = [p for p in params if p.requires_grad]
         self.world_size = world_size or dist.get_world_size(group)
         self.group = group
@@ -194,18 +201,29 @@ class GradAllReducer:
         self._pending = [0] * len(self.buckets)
         self._works: List = []
         self._flat: List[Optional[torch.Tensor]] = [None] * len(self.buckets)
+        self.hooks_enabled = True
         self._hooks = []
-        for p in self.params:
-            self._hooks.append(
-                p.register_post_accumulate_grad_hook(self._on_grad)
-            )
+        if register_hooks:
+            for p in self.params:
+                self._hooks.append(
+                    p.register_post_accumulate_grad_hook(self._on_grad)
+                )
         self._reset_counts()
+
+    def reduce(self):
+        """Explicit launch of every bucket's all-reduce (for use after a
+        graph-replayed backward, where hooks cannot fire).  Follow with
+        :meth:`finalize` before the optimizer step."""
+        for bi in range(len(self.buckets)):
+            self._launch(bi)
 
     def _reset_counts(self):
         self._pending = [len(b) for b in self.buckets]
         self._works = []
 
     def _on_grad(self, p: torch.nn.Parameter):
+        if not self.hooks_enabled:
+            return
         bi = self._param_bucket[p]
         self._pending[bi] -= 1
         if self._pending[bi] == 0:

@@ -323,3 +323,40 @@ def test_distributed_weight_sync_scheme():
     for p in procs:
         p.join(30)
     assert outs[0] == pytest.approx(outs[1])  # rank 1 received rank 0's weights
+
+
+# ---------------------------------------------------------------------- #
+# bench.py distributed entry: the EXACT `torchrun bench.py --gpus N`
+# command the driver uses, world=2 on CPU gloo, must reproduce world=1
+# numerics bit-for-bit when every rank sees the same data (--same-seed:
+# all-reduced mean of identical gradients == the single-rank gradient).
+# ---------------------------------------------------------------------- #
+@pytest.mark.timeout(300)
+def test_bench_world2_matches_world1_bitwise(tmp_path):
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    bench = os.path.join(repo, "bench.py")
+    common = ["--envs", "8", "--horizon", "4", "--steps", "3", "--warmup", "1",
+              "--same-seed"]
+    p1 = str(tmp_path / "w1.pt")
+    subprocess.run(
+        [sys.executable, bench, *common, "--dump-params", p1],
+        check=True, cwd=repo, capture_output=True, timeout=240,
+    )
+    p2 = str(tmp_path / "w2.pt")
+    port = _free_port()
+    subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), bench, "--gpus", "2", *common,
+         "--dump-params", p2],
+        check=True, cwd=repo, capture_output=True, timeout=240,
+    )
+    a = torch.load(p1, weights_only=False)
+    for rank in range(2):
+        b = torch.load(p2 + f".rank{rank}", weights_only=False)
+        for part in ("actor", "critic"):
+            for k in a[part]:
+                assert torch.equal(a[part][k], b[part][k]), (rank, part, k)
