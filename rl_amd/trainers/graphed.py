@@ -108,6 +108,13 @@ class GraphedPPO:
 
             self.reducer = GradAllReducer(self._params, world_size=self._world)
         self._distributed = self.reducer is not None and self._world > 1
+        # test hook: exercise the distributed-mode step shape (minibatch
+        # graph + eager comm/step) at world=1 — used to argue multi-GPU
+        # per-rank throughput from single-GPU measurements
+        import os
+
+        if os.environ.get("RL_AMD_FORCE_MB_GRAPH") == "1":
+            self._distributed = True
 
         self._initialized = False
         self._step_fn: Optional[Callable[[], None]] = None
