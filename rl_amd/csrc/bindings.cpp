@@ -74,6 +74,18 @@ void launch_gru_fused(const float*, const float*, const float*, const bool*,
 void launch_lstm_fused(const float*, const float*, const bool*, const float*,
                        const float*, float*, float*, float*, int, int, int,
                        void*);
+void launch_gru_train_fwd(const float*, const void*, const float*,
+                          const bool*, const float*, float*, int, int, int,
+                          void*);
+void launch_gru_bwd(const float*, const void*, const float*, const bool*,
+                    const float*, const float*, const float*, float*, float*,
+                    float*, float*, int, int, int, void*);
+void launch_lstm_train_fwd(const float*, const void*, const bool*,
+                           const float*, const float*, float*, float*, int,
+                           int, int, void*);
+void launch_lstm_bwd(const float*, const void*, const bool*, const float*,
+                     const float*, const float*, const float*, const float*,
+                     float*, float*, float*, float*, int, int, int, void*);
 }
 
 static void check_gae_args(const torch::Tensor& t, const char* name) {
@@ -223,6 +235,99 @@ std::vector<torch::Tensor> lstm_fused(torch::Tensor gates_x,
                     c_out.data_ptr<float>(), B, T, H, (void*)stream);
   return {ys, h_out, c_out};
 }
+
+// Training-path scans (reverse-time backward with gate recompute).
+// ``wt`` is the pre-transposed bf16 W_hh^T [H, G*H]: staged to LDS when
+// it fits the 160 KB budget, streamed from L2 otherwise (large H).
+static const void* bf16_ptr(const torch::Tensor& t) {
+  return (const void*)t.data_ptr<at::BFloat16>();
+}
+
+torch::Tensor gru_train_fwd(torch::Tensor gates_x, torch::Tensor wt,
+                            torch::Tensor bias_hh, torch::Tensor is_init,
+                            torch::Tensor h0) {
+  TORCH_CHECK(gates_x.is_cuda() && gates_x.is_contiguous());
+  TORCH_CHECK(wt.is_contiguous() && wt.scalar_type() == torch::kBFloat16,
+              "wt must be contiguous bf16 [H, 3H]");
+  long B = gates_x.size(0), T = gates_x.size(1);
+  long H = gates_x.size(2) / 3;
+  auto ys = torch::empty({B, T, H}, gates_x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_gru_train_fwd(gates_x.data_ptr<float>(), bf16_ptr(wt),
+                       bias_hh.data_ptr<float>(), is_init.data_ptr<bool>(),
+                       h0.numel() ? h0.data_ptr<float>() : nullptr,
+                       ys.data_ptr<float>(), B, T, H, (void*)stream);
+  return ys;
+}
+
+std::vector<torch::Tensor> gru_bwd(torch::Tensor gates_x, torch::Tensor wt,
+                                   torch::Tensor bias_hh,
+                                   torch::Tensor is_init, torch::Tensor h0,
+                                   torch::Tensor ys, torch::Tensor dys) {
+  TORCH_CHECK(gates_x.is_cuda() && gates_x.is_contiguous());
+  TORCH_CHECK(dys.is_contiguous(), "dys must be contiguous");
+  long B = gates_x.size(0), T = gates_x.size(1);
+  long H = gates_x.size(2) / 3;
+  auto dgx = torch::empty_like(gates_x);
+  auto dgh = torch::empty_like(gates_x);
+  auto hprev = torch::empty({B, T, H}, gates_x.options());
+  auto dh0 = torch::empty({B, H}, gates_x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_gru_bwd(gates_x.data_ptr<float>(), bf16_ptr(wt),
+                 bias_hh.data_ptr<float>(), is_init.data_ptr<bool>(),
+                 h0.numel() ? h0.data_ptr<float>() : nullptr,
+                 ys.data_ptr<float>(), dys.data_ptr<float>(),
+                 dgx.data_ptr<float>(), dgh.data_ptr<float>(),
+                 hprev.data_ptr<float>(), dh0.data_ptr<float>(), B, T, H,
+                 (void*)stream);
+  return {dgx, dgh, hprev, dh0};
+}
+
+std::vector<torch::Tensor> lstm_train_fwd(torch::Tensor gates_x,
+                                          torch::Tensor wt,
+                                          torch::Tensor is_init,
+                                          torch::Tensor h0, torch::Tensor c0) {
+  TORCH_CHECK(gates_x.is_cuda() && gates_x.is_contiguous());
+  TORCH_CHECK(wt.is_contiguous() && wt.scalar_type() == torch::kBFloat16,
+              "wt must be contiguous bf16 [H, 4H]");
+  long B = gates_x.size(0), T = gates_x.size(1);
+  long H = gates_x.size(2) / 4;
+  auto ys = torch::empty({B, T, H}, gates_x.options());
+  auto cs = torch::empty({B, T, H}, gates_x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_lstm_train_fwd(gates_x.data_ptr<float>(), bf16_ptr(wt),
+                        is_init.data_ptr<bool>(),
+                        h0.numel() ? h0.data_ptr<float>() : nullptr,
+                        c0.numel() ? c0.data_ptr<float>() : nullptr,
+                        ys.data_ptr<float>(), cs.data_ptr<float>(), B, T, H,
+                        (void*)stream);
+  return {ys, cs};
+}
+
+std::vector<torch::Tensor> lstm_bwd(torch::Tensor gates_x, torch::Tensor wt,
+                                    torch::Tensor is_init, torch::Tensor h0,
+                                    torch::Tensor c0, torch::Tensor ys,
+                                    torch::Tensor cs, torch::Tensor dys) {
+  TORCH_CHECK(gates_x.is_cuda() && gates_x.is_contiguous());
+  TORCH_CHECK(dys.is_contiguous(), "dys must be contiguous");
+  long B = gates_x.size(0), T = gates_x.size(1);
+  long H = gates_x.size(2) / 4;
+  auto dg = torch::empty_like(gates_x);
+  auto hprev = torch::empty({B, T, H}, gates_x.options());
+  auto dh0 = torch::empty({B, H}, gates_x.options());
+  auto dc0 = torch::empty({B, H}, gates_x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_lstm_bwd(gates_x.data_ptr<float>(), bf16_ptr(wt),
+                  is_init.data_ptr<bool>(),
+                  h0.numel() ? h0.data_ptr<float>() : nullptr,
+                  c0.numel() ? c0.data_ptr<float>() : nullptr,
+                  ys.data_ptr<float>(), cs.data_ptr<float>(),
+                  dys.data_ptr<float>(), dg.data_ptr<float>(),
+                  hprev.data_ptr<float>(), dh0.data_ptr<float>(),
+                  dc0.data_ptr<float>(), B, T, H, (void*)stream);
+  return {dg, hprev, dh0, dc0};
+}
+
 // Fused actor: MLP(tanh)x2 + heads + TanhNormal sample + log-prob.
 std::vector<torch::Tensor> fused_actor(
     torch::Tensor obs, torch::Tensor w1, torch::Tensor b1, torch::Tensor w2,
@@ -540,6 +645,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tree_update", &tree_update, "segment-tree leaf update+recompute (HIP)");
   m.def("gru_fused", &gru_fused, "fused GRU forward scan with resets (HIP)");
   m.def("lstm_fused", &lstm_fused, "fused LSTM forward scan with resets (HIP)");
+  m.def("gru_train_fwd", &gru_train_fwd,
+        "differentiable GRU forward scan, any H (HIP)");
+  m.def("gru_bwd", &gru_bwd,
+        "GRU reverse-time backward scan with gate recompute (HIP)");
+  m.def("lstm_train_fwd", &lstm_train_fwd,
+        "LSTM forward scan emitting per-step cell states (HIP)");
+  m.def("lstm_bwd", &lstm_bwd,
+        "LSTM reverse-time backward scan with gate recompute (HIP)");
   m.def("fused_actor", &fused_actor,
         "fused MLP+TanhNormal actor forward (HIP)");
 #endif

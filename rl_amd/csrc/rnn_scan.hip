@@ -203,7 +203,526 @@ __global__ void lstm_fused_kernel(
 }
 
 }  // namespace
+// ------------------------------------------------------------------- //
+// Training path: differentiable fused scans (reference capability:
+// _rnn_triton.py:329 _gru_bwd_kernel, :822 _lstm_bwd_kernel).
+//
+// MI355X design: one workgroup scans all T steps for its batch tile;
+// the backward runs REVERSE-time and RECOMPUTES the gates from gates_x
+// + h_{t-1} (read back from ys) — no per-step gate storage.  W_hh is a
+// pre-transposed bf16 [H, G*H] array (k-major); for H where it fits the
+// 160 KB LDS it is staged once (template LDSW=true, padded stride so
+// the backward's per-k reads land on distinct banks), otherwise it is
+// read straight from global memory — W stays L2-resident (bf16 W at
+// H=256 GRU is 384 KB against 4 MB L2/XCD), so large-H pays L2 latency
+// but the scan is still ONE launch instead of T·(GEMM + ~10 elementwise)
+// launches.  Parameter gradients leave the kernel as per-step gate
+// grads + the h_{t-1} sequence; the [G*H, H] weight gradient is ONE
+// GEMM on the host.
+// ------------------------------------------------------------------- //
 
+#define BW_B_TILE 8
+#define BW_B_TILE_LSTM 4
+
+namespace {
+
+template <bool LDSW>
+__global__ void __launch_bounds__(THREADS) gru_train_fwd_kernel(
+    const float* __restrict__ gates_x,       // [B, T, 3H]
+    const __hip_bfloat16* __restrict__ wt_g, // [H, 3H] (W_hh^T, k-major)
+    const float* __restrict__ bias_hh,       // [3H]
+    const bool* __restrict__ is_init,        // [B, T]
+    const float* __restrict__ h0,            // [B, H] or nullptr
+    float* __restrict__ ys,                  // [B, T, H]
+    const int B, const int T, const int H) {
+  extern __shared__ unsigned char smem[];
+  const int GH = 3 * H;
+  const int WS = LDSW ? GH + 2 : GH;
+  __hip_bfloat16* w_t = reinterpret_cast<__hip_bfloat16*>(smem);
+  float* h_s = LDSW
+      ? reinterpret_cast<float*>(smem + (size_t)H * WS * sizeof(__hip_bfloat16))
+      : reinterpret_cast<float*>(smem);
+  const __hip_bfloat16* W = LDSW ? w_t : wt_g;
+
+  const int tid = threadIdx.x;
+  const int row0 = blockIdx.x * BW_B_TILE;
+  const int rows = min(BW_B_TILE, B - row0);
+
+  if (LDSW) {
+    for (int i = tid; i < GH * H; i += THREADS) {
+      const int k = i / GH;
+      const int j = i % GH;
+      w_t[(size_t)k * WS + j] = wt_g[(size_t)k * GH + j];
+    }
+  }
+  for (int i = tid; i < rows * H; i += THREADS) {
+    const int b = i / H;
+    const int k = i % H;
+    h_s[b * H + k] = h0 ? h0[(size_t)(row0 + b) * H + k] : 0.0f;
+  }
+  __syncthreads();
+
+  const int pairs = rows * H;
+  for (int t = 0; t < T; ++t) {
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      if (is_init[(size_t)(row0 + b) * T + t]) h_s[i] = 0.0f;
+    }
+    __syncthreads();
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      float acc_r = bias_hh[jh];
+      float acc_z = bias_hh[H + jh];
+      float acc_n = bias_hh[2 * H + jh];
+      const float* hrow = &h_s[b * H];
+      for (int k = 0; k < H; ++k) {
+        const float hv = hrow[k];
+        const size_t base = (size_t)k * WS;
+        acc_r += hv * __bfloat162float(W[base + jh]);
+        acc_z += hv * __bfloat162float(W[base + H + jh]);
+        acc_n += hv * __bfloat162float(W[base + 2 * H + jh]);
+      }
+      const size_t gbase = ((size_t)(row0 + b) * T + t) * GH;
+      const float r = sigmoidf_(gates_x[gbase + jh] + acc_r);
+      const float z = sigmoidf_(gates_x[gbase + H + jh] + acc_z);
+      const float n = tanhf(gates_x[gbase + 2 * H + jh] + r * acc_n);
+      ys[((size_t)(row0 + b) * T + t) * H + jh] = (1.0f - z) * n + z * hrow[jh];
+    }
+    __syncthreads();
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      h_s[i] = ys[((size_t)(row0 + b) * T + t) * H + jh];
+    }
+    __syncthreads();
+  }
+}
+
+template <bool LDSW>
+__global__ void __launch_bounds__(THREADS) gru_bwd_kernel(
+    const float* __restrict__ gates_x,       // [B, T, 3H]
+    const __hip_bfloat16* __restrict__ wt_g, // [H, 3H]
+    const float* __restrict__ bias_hh,       // [3H]
+    const bool* __restrict__ is_init,        // [B, T]
+    const float* __restrict__ h0,            // [B, H] or nullptr
+    const float* __restrict__ ys,            // [B, T, H] (forward outputs)
+    const float* __restrict__ dys,           // [B, T, H] (grad wrt ys)
+    float* __restrict__ dgx,                 // [B, T, 3H] grad wrt gates_x
+    float* __restrict__ dgh,   // [B, T, 3H] grad wrt (W_hh h + b_hh)
+    float* __restrict__ hprev, // [B, T, H] effective h_{t-1}
+    float* __restrict__ dh0,   // [B, H]
+    const int B, const int T, const int H) {
+  extern __shared__ unsigned char smem[];
+  const int GH = 3 * H;
+  const int WS = LDSW ? GH + 2 : GH;
+  __hip_bfloat16* w_t = reinterpret_cast<__hip_bfloat16*>(smem);
+  float* f32base = LDSW
+      ? reinterpret_cast<float*>(smem + (size_t)H * WS * sizeof(__hip_bfloat16))
+      : reinterpret_cast<float*>(smem);
+  float* hp = f32base;             // [B_TILE][H]
+  float* dh = hp + BW_B_TILE * H;  // [B_TILE][H]
+  float* zdh = dh + BW_B_TILE * H; // [B_TILE][H]
+  float* g_s = zdh + BW_B_TILE * H;// [B_TILE][3H]
+  const __hip_bfloat16* W = LDSW ? w_t : wt_g;
+
+  const int tid = threadIdx.x;
+  const int row0 = blockIdx.x * BW_B_TILE;
+  const int rows = min(BW_B_TILE, B - row0);
+
+  if (LDSW) {
+    for (int i = tid; i < GH * H; i += THREADS) {
+      const int k = i / GH;
+      const int j = i % GH;
+      w_t[(size_t)k * WS + j] = wt_g[(size_t)k * GH + j];
+    }
+  }
+  const int pairs = rows * H;
+  for (int i = tid; i < pairs; i += THREADS) dh[i] = 0.0f;
+  __syncthreads();
+
+  for (int t = T - 1; t >= 0; --t) {
+    // stage the effective h_{t-1} (zeroed at trajectory starts)
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      const bool init = is_init[(size_t)(row0 + b) * T + t];
+      float v = 0.0f;
+      if (!init) {
+        if (t > 0)
+          v = ys[((size_t)(row0 + b) * T + t - 1) * H + k];
+        else if (h0)
+          v = h0[(size_t)(row0 + b) * H + k];
+      }
+      hp[b * H + k] = v;
+    }
+    __syncthreads();
+    // recompute gates, form gate grads
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      float accr = bias_hh[jh];
+      float accz = bias_hh[H + jh];
+      float accn = bias_hh[2 * H + jh];
+      const float* hrow = &hp[b * H];
+      for (int k = 0; k < H; ++k) {
+        const float hv = hrow[k];
+        const size_t base = (size_t)k * WS;
+        accr += hv * __bfloat162float(W[base + jh]);
+        accz += hv * __bfloat162float(W[base + H + jh]);
+        accn += hv * __bfloat162float(W[base + 2 * H + jh]);
+      }
+      const size_t gbase = ((size_t)(row0 + b) * T + t) * GH;
+      const float r = sigmoidf_(gates_x[gbase + jh] + accr);
+      const float z = sigmoidf_(gates_x[gbase + H + jh] + accz);
+      const float n = tanhf(gates_x[gbase + 2 * H + jh] + r * accn);
+      const size_t ybase = ((size_t)(row0 + b) * T + t) * H + jh;
+      const float dh_t = dys[ybase] + dh[b * H + jh];
+      const float dz = dh_t * (hrow[jh] - n);
+      const float dgz = dz * z * (1.0f - z);
+      const float dn = dh_t * (1.0f - z) * (1.0f - n * n);
+      const float dan = dn * r;
+      const float dgr = dn * accn * r * (1.0f - r);
+      dgx[gbase + jh] = dgr;
+      dgx[gbase + H + jh] = dgz;
+      dgx[gbase + 2 * H + jh] = dn;
+      dgh[gbase + jh] = dgr;
+      dgh[gbase + H + jh] = dgz;
+      dgh[gbase + 2 * H + jh] = dan;
+      hprev[ybase] = hrow[jh];
+      g_s[b * GH + jh] = dgr;
+      g_s[b * GH + H + jh] = dgz;
+      g_s[b * GH + 2 * H + jh] = dan;
+      zdh[b * H + jh] = dh_t * z;
+    }
+    __syncthreads();
+    // carry: dh_{t-1} = z·dh + W^T dgates  (cut at trajectory starts)
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      float acc = 0.0f;
+      if (!is_init[(size_t)(row0 + b) * T + t]) {
+        acc = zdh[b * H + k];
+        const float* grow = &g_s[b * GH];
+        const size_t base = (size_t)k * WS;
+        for (int j = 0; j < GH; ++j)
+          acc += grow[j] * __bfloat162float(W[base + j]);
+      }
+      dh[b * H + k] = acc;
+    }
+    __syncthreads();
+  }
+  for (int i = tid; i < pairs; i += THREADS) {
+    const int b = i / H;
+    const int k = i % H;
+    dh0[(size_t)(row0 + b) * H + k] = dh[b * H + k];
+  }
+}
+
+// LSTM training forward: also emits per-step cell states (cs), which
+// the backward needs (c is a recurrence — unlike the gates it cannot
+// be recomputed in reverse).
+template <bool LDSW>
+__global__ void __launch_bounds__(THREADS) lstm_train_fwd_kernel(
+    const float* __restrict__ gates_x,       // [B, T, 4H] (+b_ih+b_hh folded)
+    const __hip_bfloat16* __restrict__ wt_g, // [H, 4H]
+    const bool* __restrict__ is_init,        // [B, T]
+    const float* __restrict__ h0, const float* __restrict__ c0,
+    float* __restrict__ ys,  // [B, T, H]
+    float* __restrict__ cs,  // [B, T, H]
+    const int B, const int T, const int H) {
+  extern __shared__ unsigned char smem[];
+  const int GH = 4 * H;
+  const int WS = LDSW ? GH + 2 : GH;
+  __hip_bfloat16* w_t = reinterpret_cast<__hip_bfloat16*>(smem);
+  float* h_s = LDSW
+      ? reinterpret_cast<float*>(smem + (size_t)H * WS * sizeof(__hip_bfloat16))
+      : reinterpret_cast<float*>(smem);
+  float* c_s = h_s + BW_B_TILE_LSTM * H;
+  const __hip_bfloat16* W = LDSW ? w_t : wt_g;
+
+  const int tid = threadIdx.x;
+  const int row0 = blockIdx.x * BW_B_TILE_LSTM;
+  const int rows = min(BW_B_TILE_LSTM, B - row0);
+
+  if (LDSW) {
+    for (int i = tid; i < GH * H; i += THREADS) {
+      const int k = i / GH;
+      const int j = i % GH;
+      w_t[(size_t)k * WS + j] = wt_g[(size_t)k * GH + j];
+    }
+  }
+  for (int i = tid; i < rows * H; i += THREADS) {
+    const int b = i / H;
+    const int k = i % H;
+    h_s[b * H + k] = h0 ? h0[(size_t)(row0 + b) * H + k] : 0.0f;
+    c_s[b * H + k] = c0 ? c0[(size_t)(row0 + b) * H + k] : 0.0f;
+  }
+  __syncthreads();
+
+  const int pairs = rows * H;
+  for (int t = 0; t < T; ++t) {
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      if (is_init[(size_t)(row0 + b) * T + t]) {
+        h_s[i] = 0.0f;
+        c_s[i] = 0.0f;
+      }
+    }
+    __syncthreads();
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      float acc_i = 0.f, acc_f = 0.f, acc_g = 0.f, acc_o = 0.f;
+      const float* hrow = &h_s[b * H];
+      for (int k = 0; k < H; ++k) {
+        const float hv = hrow[k];
+        const size_t base = (size_t)k * WS;
+        acc_i += hv * __bfloat162float(W[base + jh]);
+        acc_f += hv * __bfloat162float(W[base + H + jh]);
+        acc_g += hv * __bfloat162float(W[base + 2 * H + jh]);
+        acc_o += hv * __bfloat162float(W[base + 3 * H + jh]);
+      }
+      const size_t gbase = ((size_t)(row0 + b) * T + t) * GH;
+      const float ig = sigmoidf_(gates_x[gbase + jh] + acc_i);
+      const float fg = sigmoidf_(gates_x[gbase + H + jh] + acc_f);
+      const float gg = tanhf(gates_x[gbase + 2 * H + jh] + acc_g);
+      const float og = sigmoidf_(gates_x[gbase + 3 * H + jh] + acc_o);
+      const size_t ybase = ((size_t)(row0 + b) * T + t) * H + jh;
+      const float cnew = fg * c_s[b * H + jh] + ig * gg;
+      ys[ybase] = og * tanhf(cnew);
+      cs[ybase] = cnew;
+    }
+    __syncthreads();
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      const size_t ybase = ((size_t)(row0 + b) * T + t) * H + jh;
+      h_s[i] = ys[ybase];
+      c_s[i] = cs[ybase];
+    }
+    __syncthreads();
+  }
+}
+
+template <bool LDSW>
+__global__ void __launch_bounds__(THREADS) lstm_bwd_kernel(
+    const float* __restrict__ gates_x,       // [B, T, 4H]
+    const __hip_bfloat16* __restrict__ wt_g, // [H, 4H]
+    const bool* __restrict__ is_init,        // [B, T]
+    const float* __restrict__ h0, const float* __restrict__ c0,
+    const float* __restrict__ ys,   // [B, T, H]
+    const float* __restrict__ cs,   // [B, T, H]
+    const float* __restrict__ dys,  // [B, T, H]
+    float* __restrict__ dg,    // [B, T, 4H] grad wrt gates (x and h side)
+    float* __restrict__ hprev, // [B, T, H]
+    float* __restrict__ dh0, float* __restrict__ dc0,  // [B, H]
+    const int B, const int T, const int H) {
+  extern __shared__ unsigned char smem[];
+  const int GH = 4 * H;
+  const int WS = LDSW ? GH + 2 : GH;
+  __hip_bfloat16* w_t = reinterpret_cast<__hip_bfloat16*>(smem);
+  float* f32base = LDSW
+      ? reinterpret_cast<float*>(smem + (size_t)H * WS * sizeof(__hip_bfloat16))
+      : reinterpret_cast<float*>(smem);
+  float* hp = f32base;                  // [B_TILE][H]
+  float* cp = hp + BW_B_TILE_LSTM * H;  // [B_TILE][H]
+  float* dh = cp + BW_B_TILE_LSTM * H;  // [B_TILE][H]
+  float* dc = dh + BW_B_TILE_LSTM * H;  // [B_TILE][H]
+  float* g_s = dc + BW_B_TILE_LSTM * H; // [B_TILE][4H]
+  const __hip_bfloat16* W = LDSW ? w_t : wt_g;
+
+  const int tid = threadIdx.x;
+  const int row0 = blockIdx.x * BW_B_TILE_LSTM;
+  const int rows = min(BW_B_TILE_LSTM, B - row0);
+
+  if (LDSW) {
+    for (int i = tid; i < GH * H; i += THREADS) {
+      const int k = i / GH;
+      const int j = i % GH;
+      w_t[(size_t)k * WS + j] = wt_g[(size_t)k * GH + j];
+    }
+  }
+  const int pairs = rows * H;
+  for (int i = tid; i < pairs; i += THREADS) {
+    dh[i] = 0.0f;
+    dc[i] = 0.0f;
+  }
+  __syncthreads();
+
+  for (int t = T - 1; t >= 0; --t) {
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      const bool init = is_init[(size_t)(row0 + b) * T + t];
+      float hv = 0.0f, cv = 0.0f;
+      if (!init) {
+        if (t > 0) {
+          const size_t pb = ((size_t)(row0 + b) * T + t - 1) * H + k;
+          hv = ys[pb];
+          cv = cs[pb];
+        } else {
+          if (h0) hv = h0[(size_t)(row0 + b) * H + k];
+          if (c0) cv = c0[(size_t)(row0 + b) * H + k];
+        }
+      }
+      hp[b * H + k] = hv;
+      cp[b * H + k] = cv;
+    }
+    __syncthreads();
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      float acc_i = 0.f, acc_f = 0.f, acc_g = 0.f, acc_o = 0.f;
+      const float* hrow = &hp[b * H];
+      for (int k = 0; k < H; ++k) {
+        const float hv = hrow[k];
+        const size_t base = (size_t)k * WS;
+        acc_i += hv * __bfloat162float(W[base + jh]);
+        acc_f += hv * __bfloat162float(W[base + H + jh]);
+        acc_g += hv * __bfloat162float(W[base + 2 * H + jh]);
+        acc_o += hv * __bfloat162float(W[base + 3 * H + jh]);
+      }
+      const size_t gbase = ((size_t)(row0 + b) * T + t) * GH;
+      const float ig = sigmoidf_(gates_x[gbase + jh] + acc_i);
+      const float fg = sigmoidf_(gates_x[gbase + H + jh] + acc_f);
+      const float gg = tanhf(gates_x[gbase + 2 * H + jh] + acc_g);
+      const float og = sigmoidf_(gates_x[gbase + 3 * H + jh] + acc_o);
+      const size_t ybase = ((size_t)(row0 + b) * T + t) * H + jh;
+      const float ct = cs[ybase];
+      const float tc = tanhf(ct);
+      const float dh_t = dys[ybase] + dh[b * H + jh];
+      const float dgo = dh_t * tc * og * (1.0f - og);
+      const float dct = dc[b * H + jh] + dh_t * og * (1.0f - tc * tc);
+      const float dgf = dct * cp[b * H + jh] * fg * (1.0f - fg);
+      const float dgi = dct * gg * ig * (1.0f - ig);
+      const float dgg = dct * ig * (1.0f - gg * gg);
+      dg[gbase + jh] = dgi;
+      dg[gbase + H + jh] = dgf;
+      dg[gbase + 2 * H + jh] = dgg;
+      dg[gbase + 3 * H + jh] = dgo;
+      hprev[ybase] = hrow[jh];
+      g_s[b * GH + jh] = dgi;
+      g_s[b * GH + H + jh] = dgf;
+      g_s[b * GH + 2 * H + jh] = dgg;
+      g_s[b * GH + 3 * H + jh] = dgo;
+      dc[b * H + jh] =
+          is_init[(size_t)(row0 + b) * T + t] ? 0.0f : dct * fg;
+    }
+    __syncthreads();
+    for (int i = tid; i < pairs; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      float acc = 0.0f;
+      if (!is_init[(size_t)(row0 + b) * T + t]) {
+        const float* grow = &g_s[b * GH];
+        const size_t base = (size_t)k * WS;
+        for (int j = 0; j < GH; ++j)
+          acc += grow[j] * __bfloat162float(W[base + j]);
+      }
+      dh[b * H + k] = acc;
+    }
+    __syncthreads();
+  }
+  for (int i = tid; i < pairs; i += THREADS) {
+    const int b = i / H;
+    const int k = i % H;
+    dh0[(size_t)(row0 + b) * H + k] = dh[b * H + k];
+    dc0[(size_t)(row0 + b) * H + k] = dc[b * H + k];
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+// LDS bytes when W is resident (LDSW=true); the f32 scratch alone when
+// it is not (large H streams W from L2).
+int gru_train_lds_bytes(int H, int ldsw) {
+  const size_t w = ldsw ? (size_t)H * (3 * H + 2) * sizeof(__hip_bfloat16) : 0;
+  return (int)(w + (size_t)BW_B_TILE * (3 * H + 3 * H) * sizeof(float));
+}
+
+int lstm_train_lds_bytes(int H, int ldsw) {
+  const size_t w = ldsw ? (size_t)H * (4 * H + 2) * sizeof(__hip_bfloat16) : 0;
+  return (int)(w + (size_t)BW_B_TILE_LSTM * (4 * H + 4 * H) * sizeof(float));
+}
+
+void launch_gru_train_fwd(const float* gates_x, const void* wt_g,
+                          const float* bias_hh, const bool* is_init,
+                          const float* h0, float* ys, int B, int T, int H,
+                          void* stream) {
+  const int blocks = (B + BW_B_TILE - 1) / BW_B_TILE;
+  const bool ldsw = gru_train_lds_bytes(H, 1) <= 160 * 1024;
+  const int lds = gru_train_lds_bytes(H, ldsw);
+  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
+  if (ldsw)
+    hipLaunchKernelGGL(gru_train_fwd_kernel<true>, dim3(blocks), dim3(THREADS),
+                       lds, (hipStream_t)stream, gates_x, wt, bias_hh, is_init,
+                       h0, ys, B, T, H);
+  else
+    hipLaunchKernelGGL(gru_train_fwd_kernel<false>, dim3(blocks),
+                       dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
+                       bias_hh, is_init, h0, ys, B, T, H);
+}
+
+void launch_gru_bwd(const float* gates_x, const void* wt_g,
+                    const float* bias_hh, const bool* is_init,
+                    const float* h0, const float* ys, const float* dys,
+                    float* dgx, float* dgh, float* hprev, float* dh0, int B,
+                    int T, int H, void* stream) {
+  const int blocks = (B + BW_B_TILE - 1) / BW_B_TILE;
+  const bool ldsw = gru_train_lds_bytes(H, 1) <= 160 * 1024;
+  const int lds = gru_train_lds_bytes(H, ldsw);
+  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
+  if (ldsw)
+    hipLaunchKernelGGL(gru_bwd_kernel<true>, dim3(blocks), dim3(THREADS), lds,
+                       (hipStream_t)stream, gates_x, wt, bias_hh, is_init, h0,
+                       ys, dys, dgx, dgh, hprev, dh0, B, T, H);
+  else
+    hipLaunchKernelGGL(gru_bwd_kernel<false>, dim3(blocks), dim3(THREADS), lds,
+                       (hipStream_t)stream, gates_x, wt, bias_hh, is_init, h0,
+                       ys, dys, dgx, dgh, hprev, dh0, B, T, H);
+}
+
+void launch_lstm_train_fwd(const float* gates_x, const void* wt_g,
+                           const bool* is_init, const float* h0,
+                           const float* c0, float* ys, float* cs, int B,
+                           int T, int H, void* stream) {
+  const int blocks = (B + BW_B_TILE_LSTM - 1) / BW_B_TILE_LSTM;
+  const bool ldsw = lstm_train_lds_bytes(H, 1) <= 160 * 1024;
+  const int lds = lstm_train_lds_bytes(H, ldsw);
+  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
+  if (ldsw)
+    hipLaunchKernelGGL(lstm_train_fwd_kernel<true>, dim3(blocks),
+                       dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
+                       is_init, h0, c0, ys, cs, B, T, H);
+  else
+    hipLaunchKernelGGL(lstm_train_fwd_kernel<false>, dim3(blocks),
+                       dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
+                       is_init, h0, c0, ys, cs, B, T, H);
+}
+
+void launch_lstm_bwd(const float* gates_x, const void* wt_g,
+                     const bool* is_init, const float* h0, const float* c0,
+                     const float* ys, const float* cs, const float* dys,
+                     float* dg, float* hprev, float* dh0, float* dc0, int B,
+                     int T, int H, void* stream) {
+  const int blocks = (B + BW_B_TILE_LSTM - 1) / BW_B_TILE_LSTM;
+  const bool ldsw = lstm_train_lds_bytes(H, 1) <= 160 * 1024;
+  const int lds = lstm_train_lds_bytes(H, ldsw);
+  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
+  if (ldsw)
+    hipLaunchKernelGGL(lstm_bwd_kernel<true>, dim3(blocks), dim3(THREADS),
+                       lds, (hipStream_t)stream, gates_x, wt, is_init, h0, c0,
+                       ys, cs, dys, dg, hprev, dh0, dc0, B, T, H);
+  else
+    hipLaunchKernelGGL(lstm_bwd_kernel<false>, dim3(blocks), dim3(THREADS),
+                       lds, (hipStream_t)stream, gates_x, wt, is_init, h0, c0,
+                       ys, cs, dys, dg, hprev, dh0, dc0, B, T, H);
+}
+
+}  // extern "C"
 extern "C" {
 
 int gru_fused_lds_bytes(int H) {

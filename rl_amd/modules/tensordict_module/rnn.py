@@ -205,6 +205,20 @@ def _fused_available(x: torch.Tensor, hidden: int) -> bool:
         return False
 
 
+def _fused_train_available(x: torch.Tensor, hidden: int, kind: str) -> bool:
+    """Differentiable fused scan (HIP fwd + reverse-time recompute bwd)."""
+    if not x.is_cuda:
+        return False
+    try:
+        from ... import ops
+
+        # any H is supported: W_hh^T is LDS-resident when it fits the
+        # 160 KB budget, L2-streamed otherwise (csrc/rnn_scan.hip)
+        return ops.HAS_HIP_EXT and hasattr(ops.ext_module(), f"{kind}_bwd")
+    except Exception:
+        return False
+
+
 class _RNNModuleBase(TensorDictModuleBase):
     cell_cls = None
 
@@ -276,7 +290,16 @@ class LSTMModule(_RNNModuleBase):
             if h0 is not None and h0.dim() == 3:
                 h0 = h0[:, 0]
                 c0 = c0[:, 0]
-            ys, h, cs = lstm_scan(self.cell, x, is_init, h0, c0)
+            if self.backend == "fused" and _fused_train_available(
+                x, self.hidden_size, "lstm"
+            ):
+                # one HIP launch fwd (+ the reverse-time recompute scan
+                # in backward when grads are on)
+                from ... import ops
+
+                ys, h, cs = ops.lstm_train(self.cell, x, is_init, h0, c0)
+            else:
+                ys, h, cs = lstm_scan(self.cell, x, is_init, h0, c0)
             td.set(self.out_key, ys)
             td.set(("next", "recurrent_state_h"), ys)
             td.set(("next", "recurrent_state_c"), cs)
@@ -324,7 +347,13 @@ class GRUModule(_RNNModuleBase):
             h0 = td.get("recurrent_state", None)
             if h0 is not None and h0.dim() == 3:
                 h0 = h0[:, 0]
-            if self.backend == "fused" and _fused_available(x, self.hidden_size):
+            if self.backend == "fused" and torch.is_grad_enabled() and _fused_train_available(
+                x, self.hidden_size, "gru"
+            ):
+                from ... import ops
+
+                ys, h = ops.gru_train(self.cell, x, is_init, h0)
+            elif self.backend == "fused" and _fused_available(x, self.hidden_size):
                 from ... import ops
 
                 ys, h = ops.gru_fused(self.cell, x, is_init, h0)

@@ -221,6 +221,111 @@ def lstm_fused(cell, x: torch.Tensor, is_init: torch.Tensor, h0=None, c0=None):
     return ys.to(x.dtype), h.to(x.dtype), c.to(x.dtype)
 
 
+def gru_train_lds_ok(H: int) -> bool:
+    """True when W_hh^T fits LDS (else the scan streams it from L2 —
+    still a single launch, just slower weight reads).  Informational."""
+    return H * (3 * H + 2) * 2 + 8 * 6 * H * 4 <= 160 * 1024
+
+
+def lstm_train_lds_ok(H: int) -> bool:
+    return H * (4 * H + 2) * 2 + 4 * 8 * H * 4 <= 160 * 1024
+
+
+class _GRUFusedTrainFn(torch.autograd.Function):
+    """Differentiable fused GRU scan: forward = the single-launch HIP
+    scan (csrc/rnn_scan.hip gru_train_fwd_kernel), backward = the
+    reverse-time gate-RECOMPUTE scan (gru_bwd_kernel) + ONE GEMM for
+    the [3H, H] weight gradient (dgates_h^T @ h_prev).
+    Reference capability: _rnn_triton.py:329 (_gru_bwd_kernel)."""
+
+    @staticmethod
+    def forward(ctx, gates_x, w_hh, bias_hh, is_init, h0):
+        # pre-transposed bf16 weight (k-major) shared by fwd and bwd
+        wt = w_hh.detach().t().contiguous().to(torch.bfloat16)
+        bias = bias_hh.detach().contiguous()
+        ys = _C.gru_train_fwd(gates_x, wt, bias, is_init, h0)
+        ctx.save_for_backward(gates_x, wt, bias, is_init, h0, ys)
+        return ys, ys[:, -1]
+
+    @staticmethod
+    def backward(ctx, dys, dh_last):
+        gates_x, wt, bias, is_init, h0, ys = ctx.saved_tensors
+        dys = dys.contiguous()
+        if dh_last is not None and dh_last.numel():
+            dys = dys.clone()
+            dys[:, -1] += dh_last
+        dgx, dgh, hprev, dh0 = _C.gru_bwd(
+            gates_x, wt, bias, is_init, h0, ys, dys
+        )
+        H = hprev.shape[-1]
+        dW_hh = dgh.reshape(-1, 3 * H).T @ hprev.reshape(-1, H)
+        db_hh = dgh.sum((0, 1))
+        return dgx, dW_hh, db_hh, None, (dh0 if h0.numel() else None)
+
+
+class _LSTMFusedTrainFn(torch.autograd.Function):
+    """Differentiable fused LSTM scan (forward saves per-step cell
+    states; backward recomputes the gates reverse-time).  Reference
+    capability: _rnn_triton.py:822 (_lstm_bwd_kernel)."""
+
+    @staticmethod
+    def forward(ctx, gates_x, w_hh, is_init, h0, c0):
+        wt = w_hh.detach().t().contiguous().to(torch.bfloat16)
+        ys, cs = _C.lstm_train_fwd(gates_x, wt, is_init, h0, c0)
+        ctx.save_for_backward(gates_x, wt, is_init, h0, c0, ys, cs)
+        # grads flow through ys only; the cell-state sequence is exposed
+        # as data (the backward kernel has no dcs input)
+        ctx.mark_non_differentiable(cs)
+        return ys, cs
+
+    @staticmethod
+    def backward(ctx, dys, dcs):
+        gates_x, wt, is_init, h0, c0, ys, cs = ctx.saved_tensors
+        dys = dys.contiguous()
+        dg, hprev, dh0, dc0 = _C.lstm_bwd(
+            gates_x, wt, is_init, h0, c0, ys, cs, dys
+        )
+        H = hprev.shape[-1]
+        dW_hh = dg.reshape(-1, 4 * H).T @ hprev.reshape(-1, H)
+        return (
+            dg,
+            dW_hh,
+            None,
+            (dh0 if h0.numel() else None),
+            (dc0 if c0.numel() else None),
+        )
+
+
+def gru_train(cell, x: torch.Tensor, is_init: torch.Tensor, h0: Optional[torch.Tensor] = None):
+    """Differentiable fused GRU over [B, T, F] with per-step resets.
+    The x@W_ih GEMM stays a torch op (dW_ih/db_ih/dx flow through torch
+    autograd); the recurrent scan runs fwd+bwd on HIP."""
+    _require_ext()
+    gates_x = (x.float() @ cell.weight_ih.T.float() + cell.bias_ih.float()).contiguous()
+    ii = is_init.squeeze(-1) if is_init.dim() == 3 else is_init
+    h0c = h0.float().contiguous() if h0 is not None else torch.Tensor()
+    ys, h = _GRUFusedTrainFn.apply(
+        gates_x, cell.weight_hh, cell.bias_hh, ii.contiguous(), h0c
+    )
+    return ys.to(x.dtype), h.to(x.dtype)
+
+
+def lstm_train(cell, x: torch.Tensor, is_init: torch.Tensor, h0=None, c0=None):
+    """Differentiable fused LSTM over [B, T, F] with per-step resets.
+    Returns (ys, h_last, cs)."""
+    _require_ext()
+    gates_x = (
+        x.float() @ cell.weight_ih.T.float() + cell.bias_ih.float() + cell.bias_hh.float()
+    ).contiguous()
+    ii = is_init.squeeze(-1) if is_init.dim() == 3 else is_init
+    h0c = h0.float().contiguous() if h0 is not None else torch.Tensor()
+    c0c = c0.float().contiguous() if c0 is not None else torch.Tensor()
+    ys, cs = _LSTMFusedTrainFn.apply(
+        gates_x, cell.weight_hh, ii.contiguous(), h0c, c0c
+    )
+    return ys.to(x.dtype), ys[:, -1].to(x.dtype), cs.to(x.dtype)
+
+
 class FusedTanhNormalActor(torch.nn.Module):
     """Drop-in rollout policy: MLP(tanh)x2 + heads + TanhNormal sample +
     log-prob as ONE kernel (rl_amd/csrc/fused_actor.hip).
