@@ -75,17 +75,19 @@ void launch_lstm_fused(const float*, const float*, const bool*, const float*,
                        const float*, float*, float*, float*, int, int, int,
                        void*);
 void launch_gru_train_fwd(const float*, const void*, const float*,
-                          const bool*, const float*, float*, int, int, int,
-                          void*);
-void launch_gru_bwd(const float*, const void*, const float*, const bool*,
-                    const float*, const float*, const float*, float*, float*,
-                    float*, float*, int, int, int, void*);
+                          const bool*, const float*, float*, float*, int,
+                          int, int, void*);
+void launch_gru_bwd(const float*, const void*, const void*, const float*,
+                    const bool*, const float*, const float*, const float*,
+                    float*, float*, float*, float*, float*, int, int, int,
+                    void*);
 void launch_lstm_train_fwd(const float*, const void*, const bool*,
-                           const float*, const float*, float*, float*, int,
-                           int, int, void*);
-void launch_lstm_bwd(const float*, const void*, const bool*, const float*,
+                           const float*, const float*, float*, float*,
+                           float*, int, int, int, void*);
+void launch_lstm_bwd(const float*, const void*, const void*, const bool*,
                      const float*, const float*, const float*, const float*,
-                     float*, float*, float*, float*, int, int, int, void*);
+                     const float*, float*, float*, float*, float*, float*,
+                     int, int, int, void*);
 }
 
 static void check_gae_args(const torch::Tensor& t, const char* name) {
@@ -252,15 +254,19 @@ torch::Tensor gru_train_fwd(torch::Tensor gates_x, torch::Tensor wt,
   long B = gates_x.size(0), T = gates_x.size(1);
   long H = gates_x.size(2) / 3;
   auto ys = torch::empty({B, T, H}, gates_x.options());
+  long B_pad = (B + 15) / 16 * 16;
+  auto gscr = torch::empty({B_pad * 3 * H}, gates_x.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_gru_train_fwd(gates_x.data_ptr<float>(), bf16_ptr(wt),
                        bias_hh.data_ptr<float>(), is_init.data_ptr<bool>(),
                        h0.numel() ? h0.data_ptr<float>() : nullptr,
-                       ys.data_ptr<float>(), B, T, H, (void*)stream);
+                       ys.data_ptr<float>(), gscr.data_ptr<float>(), B, T, H,
+                       (void*)stream);
   return ys;
 }
 
 std::vector<torch::Tensor> gru_bwd(torch::Tensor gates_x, torch::Tensor wt,
+                                   torch::Tensor w_row,
                                    torch::Tensor bias_hh,
                                    torch::Tensor is_init, torch::Tensor h0,
                                    torch::Tensor ys, torch::Tensor dys) {
@@ -272,14 +278,16 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gates_x, torch::Tensor wt,
   auto dgh = torch::empty_like(gates_x);
   auto hprev = torch::empty({B, T, H}, gates_x.options());
   auto dh0 = torch::empty({B, H}, gates_x.options());
+  long B_pad = (B + 15) / 16 * 16;
+  auto gscr = torch::empty({B_pad * 3 * H}, gates_x.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  launch_gru_bwd(gates_x.data_ptr<float>(), bf16_ptr(wt),
+  launch_gru_bwd(gates_x.data_ptr<float>(), bf16_ptr(wt), bf16_ptr(w_row),
                  bias_hh.data_ptr<float>(), is_init.data_ptr<bool>(),
                  h0.numel() ? h0.data_ptr<float>() : nullptr,
                  ys.data_ptr<float>(), dys.data_ptr<float>(),
                  dgx.data_ptr<float>(), dgh.data_ptr<float>(),
-                 hprev.data_ptr<float>(), dh0.data_ptr<float>(), B, T, H,
-                 (void*)stream);
+                 hprev.data_ptr<float>(), dh0.data_ptr<float>(),
+                 gscr.data_ptr<float>(), B, T, H, (void*)stream);
   return {dgx, dgh, hprev, dh0};
 }
 
@@ -294,17 +302,20 @@ std::vector<torch::Tensor> lstm_train_fwd(torch::Tensor gates_x,
   long H = gates_x.size(2) / 4;
   auto ys = torch::empty({B, T, H}, gates_x.options());
   auto cs = torch::empty({B, T, H}, gates_x.options());
+  long B_pad = (B + 15) / 16 * 16;
+  auto gscr = torch::empty({B_pad * 4 * H}, gates_x.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_lstm_train_fwd(gates_x.data_ptr<float>(), bf16_ptr(wt),
                         is_init.data_ptr<bool>(),
                         h0.numel() ? h0.data_ptr<float>() : nullptr,
                         c0.numel() ? c0.data_ptr<float>() : nullptr,
-                        ys.data_ptr<float>(), cs.data_ptr<float>(), B, T, H,
-                        (void*)stream);
+                        ys.data_ptr<float>(), cs.data_ptr<float>(),
+                        gscr.data_ptr<float>(), B, T, H, (void*)stream);
   return {ys, cs};
 }
 
 std::vector<torch::Tensor> lstm_bwd(torch::Tensor gates_x, torch::Tensor wt,
+                                    torch::Tensor w_row,
                                     torch::Tensor is_init, torch::Tensor h0,
                                     torch::Tensor c0, torch::Tensor ys,
                                     torch::Tensor cs, torch::Tensor dys) {
@@ -316,15 +327,18 @@ std::vector<torch::Tensor> lstm_bwd(torch::Tensor gates_x, torch::Tensor wt,
   auto hprev = torch::empty({B, T, H}, gates_x.options());
   auto dh0 = torch::empty({B, H}, gates_x.options());
   auto dc0 = torch::empty({B, H}, gates_x.options());
+  long B_pad = (B + 15) / 16 * 16;
+  auto gscr = torch::empty({B_pad * 4 * H}, gates_x.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  launch_lstm_bwd(gates_x.data_ptr<float>(), bf16_ptr(wt),
+  launch_lstm_bwd(gates_x.data_ptr<float>(), bf16_ptr(wt), bf16_ptr(w_row),
                   is_init.data_ptr<bool>(),
                   h0.numel() ? h0.data_ptr<float>() : nullptr,
                   c0.numel() ? c0.data_ptr<float>() : nullptr,
                   ys.data_ptr<float>(), cs.data_ptr<float>(),
                   dys.data_ptr<float>(), dg.data_ptr<float>(),
                   hprev.data_ptr<float>(), dh0.data_ptr<float>(),
-                  dc0.data_ptr<float>(), B, T, H, (void*)stream);
+                  dc0.data_ptr<float>(), gscr.data_ptr<float>(), B, T, H,
+                  (void*)stream);
   return {dg, hprev, dh0, dc0};
 }
 

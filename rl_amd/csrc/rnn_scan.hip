@@ -634,6 +634,445 @@ __global__ void __launch_bounds__(THREADS) lstm_bwd_kernel(
 
 }  // namespace
 
+// ------------------------------------------------------------------- //
+// MFMA training scans (the h@W_hh product on the matrix cores).
+//
+// At production batch sizes the recurrent product is GEMM-shaped
+// ([B_TILE, H] x [H, G*H] per step) — VALU dot loops lose to the MFMA
+// pipes by ~12x there, so these variants compute every gate block with
+// v_mfma_f32_16x16x32_bf16 (fragment mapping hardware-verified in
+// benchmarks/mfma_probe.hip: A row / B col = lane&15, k = 8*(lane>>4)
+// + reg, D row = (lane>>4)*4 + reg).
+//
+//  * one workgroup = 16 batch rows (one MFMA A-tile) x 4 waves; waves
+//    split the gate-column tiles; accumulators persist across the
+//    K loop in AGPRs.
+//  * W is staged through LDS in 32-row K-chunks per step (bf16,
+//    padded stride) — no residency requirement, so ANY H%32==0 works
+//    with the same code path and W stays L2-resident across steps.
+//  * per-step gate pre-activations land in a global fp32 scratch
+//    [B, G*H] (written/read once per step, coalesced) instead of LDS —
+//    this keeps the LDS budget at ~75-140 KB for H=256.
+//  * backward: two MFMA GEMMs per step — gate recompute (hp @ W^T) and
+//    the carry (dgates @ W, dgates rounded to bf16 for the A fragment;
+//    parameter-gradient outputs stay fp32).
+// ------------------------------------------------------------------- //
+
+namespace {
+
+using bf16_frag = __attribute__((ext_vector_type(8))) short;
+using f32_frag = __attribute__((ext_vector_type(4))) float;
+
+#define MF_ROWS 16
+#define MF_MAX_ACC 16  // max col tiles per wave (LSTM H=256: 4H/16/4)
+
+// One 16x16 D tile at column j0, contracting a 32-row K-chunk:
+// A(row, k) = a16[row * lda + (kc + k)], k-major chunk B(k, col) =
+// bchunk[k * ldb + j0 + col].
+__device__ __forceinline__ f32_frag mfma_chunk_tile(
+    const __hip_bfloat16* a16, int lda, int kc,
+    const __hip_bfloat16* bchunk, int ldb, int j0, int lane, f32_frag acc) {
+  bf16_frag a, b;
+  const int row = lane & 15;
+  const int koff = 8 * (lane >> 4);
+#pragma unroll
+  for (int r = 0; r < 8; ++r) {
+    const int k = koff + r;
+    a[r] = *reinterpret_cast<const short*>(&a16[(size_t)row * lda + kc + k]);
+    b[r] = *reinterpret_cast<const short*>(&bchunk[(size_t)k * ldb + j0 + row]);
+  }
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+}
+
+// out[16][N] (global fp32, stride N) = A16 [16][K] @ B [K][N]; B comes
+// from global bsrc (k-major, stride N) staged through the LDS chunk
+// buffer (stride N+2).  Every wave owns col tiles wave, wave+4, ...
+__device__ void mfma_gemm_store(const __hip_bfloat16* a16, int lda,
+                                const __hip_bfloat16* bsrc, int N, int K,
+                                __hip_bfloat16* chunk, float* out, int ldo,
+                                int wave, int lane, int tid) {
+  f32_frag acc[MF_MAX_ACC] = {};
+  const int ntiles = N / 16;
+  const int ldc = N + 2;
+  for (int kc = 0; kc < K; kc += 32) {
+    __syncthreads();  // previous consumers done before restage
+    for (int i = tid; i < 32 * N; i += THREADS) {
+      const int k = i / N;
+      const int j = i % N;
+      chunk[(size_t)k * ldc + j] = bsrc[(size_t)(kc + k) * N + j];
+    }
+    __syncthreads();
+    int ai = 0;
+    for (int ct = wave; ct < ntiles; ct += 4, ++ai)
+      acc[ai] = mfma_chunk_tile(a16, lda, kc, chunk, ldc, ct * 16, lane, acc[ai]);
+  }
+  int ai = 0;
+  for (int ct = wave; ct < ntiles; ct += 4, ++ai) {
+    const int col = ct * 16 + (lane & 15);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = (lane >> 4) * 4 + r;
+      out[(size_t)row * ldo + col] = acc[ai][r];
+    }
+  }
+  __threadfence_block();  // out is global scratch read back after the
+                          // caller's __syncthreads
+}
+
+// GRU MFMA forward: gates scratch gscr is [B, 3H] fp32 (per-step reuse).
+__global__ void __launch_bounds__(THREADS) gru_train_fwd_mfma_kernel(
+    const float* __restrict__ gates_x,       // [B, T, 3H]
+    const __hip_bfloat16* __restrict__ wt_g, // [H, 3H]
+    const float* __restrict__ bias_hh,       // [3H]
+    const bool* __restrict__ is_init,        // [B, T]
+    const float* __restrict__ h0,            // [B, H] or nullptr
+    float* __restrict__ ys,                  // [B, T, H]
+    float* __restrict__ gscr,                // [B, 3H] scratch
+    const int B, const int T, const int H) {
+  extern __shared__ unsigned char smem[];
+  const int GH = 3 * H;
+  const int lda = H + 2;
+  __hip_bfloat16* h_bf = reinterpret_cast<__hip_bfloat16*>(smem);           // [16][H+2]
+  float* h_f = reinterpret_cast<float*>(smem + (size_t)MF_ROWS * lda * 2);  // [16][H]
+  __hip_bfloat16* chunk =
+      reinterpret_cast<__hip_bfloat16*>(h_f + MF_ROWS * H);                 // [32][3H+2]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int row0 = blockIdx.x * MF_ROWS;
+  const int rows = min(MF_ROWS, B - row0);
+  float* gs = gscr + (size_t)row0 * GH;
+
+  for (int i = tid; i < MF_ROWS * H; i += THREADS) {
+    const int b = i / H;
+    const int k = i % H;
+    const float v =
+        (b < rows && h0) ? h0[(size_t)(row0 + b) * H + k] : 0.0f;
+    h_f[b * H + k] = v;
+    h_bf[b * lda + k] = __float2bfloat16(v);
+  }
+  __syncthreads();
+
+  for (int t = 0; t < T; ++t) {
+    for (int i = tid; i < rows * H; i += THREADS) {
+      const int b = i / H;
+      if (is_init[(size_t)(row0 + b) * T + t]) {
+        h_f[b * H + i % H] = 0.0f;
+        h_bf[b * lda + i % H] = __float2bfloat16(0.0f);
+      }
+    }
+    // gates_h = h @ W^T  (A = h_bf, B = wt chunks)
+    mfma_gemm_store(h_bf, lda, wt_g, GH, H, chunk, gs, GH, wave, lane, tid);
+    __syncthreads();
+    for (int i = tid; i < rows * H; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      const size_t gbase = ((size_t)(row0 + b) * T + t) * GH;
+      const float* grow = &gs[(size_t)b * GH];
+      const float r = sigmoidf_(gates_x[gbase + jh] + grow[jh] + bias_hh[jh]);
+      const float z =
+          sigmoidf_(gates_x[gbase + H + jh] + grow[H + jh] + bias_hh[H + jh]);
+      const float n = tanhf(gates_x[gbase + 2 * H + jh] +
+                            r * (grow[2 * H + jh] + bias_hh[2 * H + jh]));
+      const float hnew = (1.0f - z) * n + z * h_f[b * H + jh];
+      ys[((size_t)(row0 + b) * T + t) * H + jh] = hnew;
+      h_f[b * H + jh] = hnew;
+      h_bf[b * lda + jh] = __float2bfloat16(hnew);
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void __launch_bounds__(THREADS) gru_bwd_mfma_kernel(
+    const float* __restrict__ gates_x,        // [B, T, 3H]
+    const __hip_bfloat16* __restrict__ wt_g,  // [H, 3H] (W^T, k-major)
+    const __hip_bfloat16* __restrict__ w_row, // [3H, H] (W, row-major)
+    const float* __restrict__ bias_hh,        // [3H]
+    const bool* __restrict__ is_init,         // [B, T]
+    const float* __restrict__ h0,             // [B, H] or nullptr
+    const float* __restrict__ ys,             // [B, T, H]
+    const float* __restrict__ dys,            // [B, T, H]
+    float* __restrict__ dgx,                  // [B, T, 3H]
+    float* __restrict__ dgh,                  // [B, T, 3H]
+    float* __restrict__ hprev,                // [B, T, H]
+    float* __restrict__ dh0,                  // [B, H]
+    float* __restrict__ gscr,                 // [B, 3H] scratch
+    const int B, const int T, const int H) {
+  extern __shared__ unsigned char smem[];
+  const int GH = 3 * H;
+  const int lda = H + 2;
+  const int ldg = GH + 2;
+  __hip_bfloat16* hp_bf = reinterpret_cast<__hip_bfloat16*>(smem);  // [16][H+2]
+  __hip_bfloat16* g_bf = hp_bf + MF_ROWS * lda;                     // [16][3H+2]
+  float* dh = reinterpret_cast<float*>(g_bf + MF_ROWS * ldg);       // [16][H]
+  __hip_bfloat16* chunk = reinterpret_cast<__hip_bfloat16*>(dh + MF_ROWS * H);
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int row0 = blockIdx.x * MF_ROWS;
+  const int rows = min(MF_ROWS, B - row0);
+  float* gs = gscr + (size_t)row0 * GH;
+
+  for (int i = tid; i < MF_ROWS * H; i += THREADS) dh[i] = 0.0f;
+  for (int i = tid; i < MF_ROWS * ldg; i += THREADS)
+    g_bf[i] = __float2bfloat16(0.0f);
+  __syncthreads();
+
+  for (int t = T - 1; t >= 0; --t) {
+    // stage effective h_{t-1} (bf16 for the recompute GEMM)
+    for (int i = tid; i < MF_ROWS * H; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      float v = 0.0f;
+      if (b < rows && !is_init[(size_t)(row0 + b) * T + t]) {
+        if (t > 0)
+          v = ys[((size_t)(row0 + b) * T + t - 1) * H + k];
+        else if (h0)
+          v = h0[(size_t)(row0 + b) * H + k];
+      }
+      hp_bf[b * lda + k] = __float2bfloat16(v);
+    }
+    // recompute gates_h = hp @ W^T into the global scratch
+    mfma_gemm_store(hp_bf, lda, wt_g, GH, H, chunk, gs, GH, wave, lane, tid);
+    __syncthreads();
+    // pointwise grads; stash bf16 dgates for the carry GEMM, fold
+    // z·dh into the dh buffer in place (its old value is consumed here)
+    for (int i = tid; i < rows * H; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      const float* grow = &gs[(size_t)b * GH];
+      const size_t gbase = ((size_t)(row0 + b) * T + t) * GH;
+      const float accn = grow[2 * H + jh] + bias_hh[2 * H + jh];
+      const float r = sigmoidf_(gates_x[gbase + jh] + grow[jh] + bias_hh[jh]);
+      const float z =
+          sigmoidf_(gates_x[gbase + H + jh] + grow[H + jh] + bias_hh[H + jh]);
+      const float n = tanhf(gates_x[gbase + 2 * H + jh] + r * accn);
+      const size_t ybase = ((size_t)(row0 + b) * T + t) * H + jh;
+      const float hpv = __bfloat162float(hp_bf[b * lda + jh]);
+      const float dh_t = dys[ybase] + dh[b * H + jh];
+      const float dz = dh_t * (hpv - n);
+      const float dgz = dz * z * (1.0f - z);
+      const float dn = dh_t * (1.0f - z) * (1.0f - n * n);
+      const float dan = dn * r;
+      const float dgr = dn * accn * r * (1.0f - r);
+      dgx[gbase + jh] = dgr;
+      dgx[gbase + H + jh] = dgz;
+      dgx[gbase + 2 * H + jh] = dn;
+      dgh[gbase + jh] = dgr;
+      dgh[gbase + H + jh] = dgz;
+      dgh[gbase + 2 * H + jh] = dan;
+      hprev[ybase] = hpv;
+      g_bf[b * ldg + jh] = __float2bfloat16(dgr);
+      g_bf[b * ldg + H + jh] = __float2bfloat16(dgz);
+      g_bf[b * ldg + 2 * H + jh] = __float2bfloat16(dan);
+      dh[b * H + jh] = dh_t * z;  // carry contribution through the z path
+    }
+    __syncthreads();
+    // carry GEMM: dcontrib[16, H] = dgates [16, 3H] @ W [3H, H],
+    // written to the FRONT of the scratch row block (fp32 [16, H])
+    mfma_gemm_store(g_bf, ldg, w_row, H, GH, chunk, gs, H, wave, lane, tid);
+    __syncthreads();
+    for (int i = tid; i < rows * H; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      dh[b * H + k] = is_init[(size_t)(row0 + b) * T + t]
+                          ? 0.0f
+                          : dh[b * H + k] + gs[(size_t)b * H + k];
+    }
+    __syncthreads();
+  }
+  for (int i = tid; i < rows * H; i += THREADS) {
+    const int b = i / H;
+    const int k = i % H;
+    dh0[(size_t)(row0 + b) * H + k] = dh[b * H + k];
+  }
+}
+
+__global__ void __launch_bounds__(THREADS) lstm_train_fwd_mfma_kernel(
+    const float* __restrict__ gates_x,       // [B, T, 4H]
+    const __hip_bfloat16* __restrict__ wt_g, // [H, 4H]
+    const bool* __restrict__ is_init,        // [B, T]
+    const float* __restrict__ h0, const float* __restrict__ c0,
+    float* __restrict__ ys,   // [B, T, H]
+    float* __restrict__ cs,   // [B, T, H]
+    float* __restrict__ gscr, // [B, 4H]
+    const int B, const int T, const int H) {
+  extern __shared__ unsigned char smem[];
+  const int GH = 4 * H;
+  const int lda = H + 2;
+  __hip_bfloat16* h_bf = reinterpret_cast<__hip_bfloat16*>(smem);
+  float* h_f = reinterpret_cast<float*>(smem + (size_t)MF_ROWS * lda * 2);
+  float* c_f = h_f + MF_ROWS * H;
+  __hip_bfloat16* chunk = reinterpret_cast<__hip_bfloat16*>(c_f + MF_ROWS * H);
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int row0 = blockIdx.x * MF_ROWS;
+  const int rows = min(MF_ROWS, B - row0);
+  float* gs = gscr + (size_t)row0 * GH;
+
+  for (int i = tid; i < MF_ROWS * H; i += THREADS) {
+    const int b = i / H;
+    const int k = i % H;
+    const float hv = (b < rows && h0) ? h0[(size_t)(row0 + b) * H + k] : 0.0f;
+    const float cv = (b < rows && c0) ? c0[(size_t)(row0 + b) * H + k] : 0.0f;
+    h_f[b * H + k] = hv;
+    c_f[b * H + k] = cv;
+    h_bf[b * lda + k] = __float2bfloat16(hv);
+  }
+  __syncthreads();
+
+  for (int t = 0; t < T; ++t) {
+    for (int i = tid; i < rows * H; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      if (is_init[(size_t)(row0 + b) * T + t]) {
+        h_f[b * H + k] = 0.0f;
+        c_f[b * H + k] = 0.0f;
+        h_bf[b * lda + k] = __float2bfloat16(0.0f);
+      }
+    }
+    mfma_gemm_store(h_bf, lda, wt_g, GH, H, chunk, gs, GH, wave, lane, tid);
+    __syncthreads();
+    for (int i = tid; i < rows * H; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      const size_t gbase = ((size_t)(row0 + b) * T + t) * GH;
+      const float* grow = &gs[(size_t)b * GH];
+      const float ig = sigmoidf_(gates_x[gbase + jh] + grow[jh]);
+      const float fg = sigmoidf_(gates_x[gbase + H + jh] + grow[H + jh]);
+      const float gg = tanhf(gates_x[gbase + 2 * H + jh] + grow[2 * H + jh]);
+      const float og = sigmoidf_(gates_x[gbase + 3 * H + jh] + grow[3 * H + jh]);
+      const size_t ybase = ((size_t)(row0 + b) * T + t) * H + jh;
+      const float cnew = fg * c_f[b * H + jh] + ig * gg;
+      const float hnew = og * tanhf(cnew);
+      ys[ybase] = hnew;
+      cs[ybase] = cnew;
+      h_f[b * H + jh] = hnew;
+      c_f[b * H + jh] = cnew;
+      h_bf[b * lda + jh] = __float2bfloat16(hnew);
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void __launch_bounds__(THREADS) lstm_bwd_mfma_kernel(
+    const float* __restrict__ gates_x,        // [B, T, 4H]
+    const __hip_bfloat16* __restrict__ wt_g,  // [H, 4H]
+    const __hip_bfloat16* __restrict__ w_row, // [4H, H]
+    const bool* __restrict__ is_init,         // [B, T]
+    const float* __restrict__ h0, const float* __restrict__ c0,
+    const float* __restrict__ ys,   // [B, T, H]
+    const float* __restrict__ cs,   // [B, T, H]
+    const float* __restrict__ dys,  // [B, T, H]
+    float* __restrict__ dg,         // [B, T, 4H]
+    float* __restrict__ hprev,      // [B, T, H]
+    float* __restrict__ dh0, float* __restrict__ dc0,
+    float* __restrict__ gscr,       // [B, 4H]
+    const int B, const int T, const int H) {
+  extern __shared__ unsigned char smem[];
+  const int GH = 4 * H;
+  const int lda = H + 2;
+  const int ldg = GH + 2;
+  __hip_bfloat16* hp_bf = reinterpret_cast<__hip_bfloat16*>(smem);  // [16][H+2]
+  __hip_bfloat16* g_bf = hp_bf + MF_ROWS * lda;                     // [16][4H+2]
+  float* dh = reinterpret_cast<float*>(g_bf + MF_ROWS * ldg);       // [16][H]
+  float* dc = dh + MF_ROWS * H;                                     // [16][H]
+  __hip_bfloat16* chunk = reinterpret_cast<__hip_bfloat16*>(dc + MF_ROWS * H);
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int row0 = blockIdx.x * MF_ROWS;
+  const int rows = min(MF_ROWS, B - row0);
+  float* gs = gscr + (size_t)row0 * GH;
+
+  for (int i = tid; i < MF_ROWS * H; i += THREADS) {
+    dh[i] = 0.0f;
+    dc[i] = 0.0f;
+  }
+  for (int i = tid; i < MF_ROWS * ldg; i += THREADS)
+    g_bf[i] = __float2bfloat16(0.0f);
+  __syncthreads();
+
+  for (int t = T - 1; t >= 0; --t) {
+    for (int i = tid; i < MF_ROWS * H; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      float hv = 0.0f;
+      if (b < rows && !is_init[(size_t)(row0 + b) * T + t]) {
+        if (t > 0)
+          hv = ys[((size_t)(row0 + b) * T + t - 1) * H + k];
+        else if (h0)
+          hv = h0[(size_t)(row0 + b) * H + k];
+      }
+      hp_bf[b * lda + k] = __float2bfloat16(hv);
+    }
+    mfma_gemm_store(hp_bf, lda, wt_g, GH, H, chunk, gs, GH, wave, lane, tid);
+    __syncthreads();
+    for (int i = tid; i < rows * H; i += THREADS) {
+      const int b = i / H;
+      const int jh = i % H;
+      const bool init = is_init[(size_t)(row0 + b) * T + t];
+      const float* grow = &gs[(size_t)b * GH];
+      const size_t gbase = ((size_t)(row0 + b) * T + t) * GH;
+      const float ig = sigmoidf_(gates_x[gbase + jh] + grow[jh]);
+      const float fg = sigmoidf_(gates_x[gbase + H + jh] + grow[H + jh]);
+      const float gg = tanhf(gates_x[gbase + 2 * H + jh] + grow[2 * H + jh]);
+      const float og = sigmoidf_(gates_x[gbase + 3 * H + jh] + grow[3 * H + jh]);
+      const size_t ybase = ((size_t)(row0 + b) * T + t) * H + jh;
+      // effective c_{t-1}
+      float cpv = 0.0f;
+      if (!init) {
+        if (t > 0)
+          cpv = cs[((size_t)(row0 + b) * T + t - 1) * H + jh];
+        else if (c0)
+          cpv = c0[(size_t)(row0 + b) * H + jh];
+      }
+      const float ct = cs[ybase];
+      const float tc = tanhf(ct);
+      const float dh_t = dys[ybase] + dh[b * H + jh];
+      const float dgo = dh_t * tc * og * (1.0f - og);
+      const float dct = dc[b * H + jh] + dh_t * og * (1.0f - tc * tc);
+      const float dgf = dct * cpv * fg * (1.0f - fg);
+      const float dgi = dct * gg * ig * (1.0f - ig);
+      const float dgg = dct * ig * (1.0f - gg * gg);
+      dg[gbase + jh] = dgi;
+      dg[gbase + H + jh] = dgf;
+      dg[gbase + 2 * H + jh] = dgg;
+      dg[gbase + 3 * H + jh] = dgo;
+      hprev[ybase] = __bfloat162float(hp_bf[b * lda + jh]);
+      g_bf[b * ldg + jh] = __float2bfloat16(dgi);
+      g_bf[b * ldg + H + jh] = __float2bfloat16(dgf);
+      g_bf[b * ldg + 2 * H + jh] = __float2bfloat16(dgg);
+      g_bf[b * ldg + 3 * H + jh] = __float2bfloat16(dgo);
+      dc[b * H + jh] = init ? 0.0f : dct * fg;
+    }
+    __syncthreads();
+    mfma_gemm_store(g_bf, ldg, w_row, H, GH, chunk, gs, H, wave, lane, tid);
+    __syncthreads();
+    for (int i = tid; i < rows * H; i += THREADS) {
+      const int b = i / H;
+      const int k = i % H;
+      dh[b * H + k] = is_init[(size_t)(row0 + b) * T + t]
+                          ? 0.0f
+                          : gs[(size_t)b * H + k];
+    }
+    __syncthreads();
+  }
+  for (int i = tid; i < rows * H; i += THREADS) {
+    const int b = i / H;
+    const int k = i % H;
+    dh0[(size_t)(row0 + b) * H + k] = dh[b * H + k];
+    dc0[(size_t)(row0 + b) * H + k] = dc[b * H + k];
+  }
+}
+
+}  // namespace
+
 extern "C" {
 
 // LDS bytes when W is resident (LDSW=true); the f32 scratch alone when
@@ -650,12 +1089,21 @@ int lstm_train_lds_bytes(int H, int ldsw) {
 
 void launch_gru_train_fwd(const float* gates_x, const void* wt_g,
                           const float* bias_hh, const bool* is_init,
-                          const float* h0, float* ys, int B, int T, int H,
-                          void* stream) {
+                          const float* h0, float* ys, float* gscr, int B,
+                          int T, int H, void* stream) {
+  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
+  if (H % 32 == 0 && 3 * H <= 1024) {
+    const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
+    const int lds = (int)(MF_ROWS * (H + 2) * 2 + MF_ROWS * H * 4 +
+                          32 * (3 * H + 2) * 2);
+    hipLaunchKernelGGL(gru_train_fwd_mfma_kernel, dim3(blocks), dim3(THREADS),
+                       lds, (hipStream_t)stream, gates_x, wt, bias_hh,
+                       is_init, h0, ys, gscr, B, T, H);
+    return;
+  }
   const int blocks = (B + BW_B_TILE - 1) / BW_B_TILE;
   const bool ldsw = gru_train_lds_bytes(H, 1) <= 160 * 1024;
   const int lds = gru_train_lds_bytes(H, ldsw);
-  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
   if (ldsw)
     hipLaunchKernelGGL(gru_train_fwd_kernel<true>, dim3(blocks), dim3(THREADS),
                        lds, (hipStream_t)stream, gates_x, wt, bias_hh, is_init,
@@ -667,14 +1115,26 @@ void launch_gru_train_fwd(const float* gates_x, const void* wt_g,
 }
 
 void launch_gru_bwd(const float* gates_x, const void* wt_g,
-                    const float* bias_hh, const bool* is_init,
-                    const float* h0, const float* ys, const float* dys,
-                    float* dgx, float* dgh, float* hprev, float* dh0, int B,
-                    int T, int H, void* stream) {
+                    const void* w_row_g, const float* bias_hh,
+                    const bool* is_init, const float* h0, const float* ys,
+                    const float* dys, float* dgx, float* dgh, float* hprev,
+                    float* dh0, float* gscr, int B, int T, int H,
+                    void* stream) {
+  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
+  const __hip_bfloat16* w_row = (const __hip_bfloat16*)w_row_g;
+  if (H % 32 == 0 && 3 * H <= 1024) {
+    const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
+    const int lds = (int)(MF_ROWS * (H + 2) * 2 + MF_ROWS * (3 * H + 2) * 2 +
+                          MF_ROWS * H * 4 + 32 * (3 * H + 2) * 2);
+    hipLaunchKernelGGL(gru_bwd_mfma_kernel, dim3(blocks), dim3(THREADS), lds,
+                       (hipStream_t)stream, gates_x, wt, w_row, bias_hh,
+                       is_init, h0, ys, dys, dgx, dgh, hprev, dh0, gscr, B, T,
+                       H);
+    return;
+  }
   const int blocks = (B + BW_B_TILE - 1) / BW_B_TILE;
   const bool ldsw = gru_train_lds_bytes(H, 1) <= 160 * 1024;
   const int lds = gru_train_lds_bytes(H, ldsw);
-  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
   if (ldsw)
     hipLaunchKernelGGL(gru_bwd_kernel<true>, dim3(blocks), dim3(THREADS), lds,
                        (hipStream_t)stream, gates_x, wt, bias_hh, is_init, h0,
@@ -687,12 +1147,21 @@ void launch_gru_bwd(const float* gates_x, const void* wt_g,
 
 void launch_lstm_train_fwd(const float* gates_x, const void* wt_g,
                            const bool* is_init, const float* h0,
-                           const float* c0, float* ys, float* cs, int B,
-                           int T, int H, void* stream) {
+                           const float* c0, float* ys, float* cs, float* gscr,
+                           int B, int T, int H, void* stream) {
+  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
+  if (H % 32 == 0 && 4 * H <= 1024) {
+    const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
+    const int lds = (int)(MF_ROWS * (H + 2) * 2 + 2 * MF_ROWS * H * 4 +
+                          32 * (4 * H + 2) * 2);
+    hipLaunchKernelGGL(lstm_train_fwd_mfma_kernel, dim3(blocks),
+                       dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
+                       is_init, h0, c0, ys, cs, gscr, B, T, H);
+    return;
+  }
   const int blocks = (B + BW_B_TILE_LSTM - 1) / BW_B_TILE_LSTM;
   const bool ldsw = lstm_train_lds_bytes(H, 1) <= 160 * 1024;
   const int lds = lstm_train_lds_bytes(H, ldsw);
-  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
   if (ldsw)
     hipLaunchKernelGGL(lstm_train_fwd_kernel<true>, dim3(blocks),
                        dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
@@ -704,14 +1173,25 @@ void launch_lstm_train_fwd(const float* gates_x, const void* wt_g,
 }
 
 void launch_lstm_bwd(const float* gates_x, const void* wt_g,
-                     const bool* is_init, const float* h0, const float* c0,
-                     const float* ys, const float* cs, const float* dys,
-                     float* dg, float* hprev, float* dh0, float* dc0, int B,
-                     int T, int H, void* stream) {
+                     const void* w_row_g, const bool* is_init,
+                     const float* h0, const float* c0, const float* ys,
+                     const float* cs, const float* dys, float* dg,
+                     float* hprev, float* dh0, float* dc0, float* gscr,
+                     int B, int T, int H, void* stream) {
+  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
+  const __hip_bfloat16* w_row = (const __hip_bfloat16*)w_row_g;
+  if (H % 32 == 0 && 4 * H <= 1024) {
+    const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
+    const int lds = (int)(MF_ROWS * (H + 2) * 2 + MF_ROWS * (4 * H + 2) * 2 +
+                          2 * MF_ROWS * H * 4 + 32 * (4 * H + 2) * 2);
+    hipLaunchKernelGGL(lstm_bwd_mfma_kernel, dim3(blocks), dim3(THREADS), lds,
+                       (hipStream_t)stream, gates_x, wt, w_row, is_init, h0,
+                       c0, ys, cs, dys, dg, hprev, dh0, dc0, gscr, B, T, H);
+    return;
+  }
   const int blocks = (B + BW_B_TILE_LSTM - 1) / BW_B_TILE_LSTM;
   const bool ldsw = lstm_train_lds_bytes(H, 1) <= 160 * 1024;
   const int lds = lstm_train_lds_bytes(H, ldsw);
-  const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
   if (ldsw)
     hipLaunchKernelGGL(lstm_bwd_kernel<true>, dim3(blocks), dim3(THREADS),
                        lds, (hipStream_t)stream, gates_x, wt, is_init, h0, c0,

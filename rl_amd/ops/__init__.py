@@ -240,22 +240,24 @@ class _GRUFusedTrainFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, gates_x, w_hh, bias_hh, is_init, h0):
-        # pre-transposed bf16 weight (k-major) shared by fwd and bwd
+        # pre-transposed bf16 weight (k-major) shared by fwd and bwd;
+        # row-major bf16 copy feeds the backward's carry GEMM
         wt = w_hh.detach().t().contiguous().to(torch.bfloat16)
+        w_row = w_hh.detach().contiguous().to(torch.bfloat16)
         bias = bias_hh.detach().contiguous()
         ys = _C.gru_train_fwd(gates_x, wt, bias, is_init, h0)
-        ctx.save_for_backward(gates_x, wt, bias, is_init, h0, ys)
+        ctx.save_for_backward(gates_x, wt, w_row, bias, is_init, h0, ys)
         return ys, ys[:, -1]
 
     @staticmethod
     def backward(ctx, dys, dh_last):
-        gates_x, wt, bias, is_init, h0, ys = ctx.saved_tensors
+        gates_x, wt, w_row, bias, is_init, h0, ys = ctx.saved_tensors
         dys = dys.contiguous()
         if dh_last is not None and dh_last.numel():
             dys = dys.clone()
             dys[:, -1] += dh_last
         dgx, dgh, hprev, dh0 = _C.gru_bwd(
-            gates_x, wt, bias, is_init, h0, ys, dys
+            gates_x, wt, w_row, bias, is_init, h0, ys, dys
         )
         H = hprev.shape[-1]
         dW_hh = dgh.reshape(-1, 3 * H).T @ hprev.reshape(-1, H)
@@ -271,8 +273,9 @@ class _LSTMFusedTrainFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, gates_x, w_hh, is_init, h0, c0):
         wt = w_hh.detach().t().contiguous().to(torch.bfloat16)
+        w_row = w_hh.detach().contiguous().to(torch.bfloat16)
         ys, cs = _C.lstm_train_fwd(gates_x, wt, is_init, h0, c0)
-        ctx.save_for_backward(gates_x, wt, is_init, h0, c0, ys, cs)
+        ctx.save_for_backward(gates_x, wt, w_row, is_init, h0, c0, ys, cs)
         # grads flow through ys only; the cell-state sequence is exposed
         # as data (the backward kernel has no dcs input)
         ctx.mark_non_differentiable(cs)
@@ -280,10 +283,10 @@ class _LSTMFusedTrainFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dys, dcs):
-        gates_x, wt, is_init, h0, c0, ys, cs = ctx.saved_tensors
+        gates_x, wt, w_row, is_init, h0, c0, ys, cs = ctx.saved_tensors
         dys = dys.contiguous()
         dg, hprev, dh0, dc0 = _C.lstm_bwd(
-            gates_x, wt, is_init, h0, c0, ys, cs, dys
+            gates_x, wt, w_row, is_init, h0, c0, ys, cs, dys
         )
         H = hprev.shape[-1]
         dW_hh = dg.reshape(-1, 4 * H).T @ hprev.reshape(-1, H)

@@ -6,3 +6,8 @@ from .weight_sync_schemes import (
     WeightStrategy,
     WeightSyncScheme,
 )
+from .llm import (
+    LLMCollectiveWeightSyncScheme,
+    LLMDoubleBufferWeightSyncScheme,
+    get_model_metadata,
+)

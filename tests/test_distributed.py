@@ -360,3 +360,87 @@ def test_bench_world2_matches_world1_bitwise(tmp_path):
         for part in ("actor", "critic"):
             for k in a[part]:
                 assert torch.equal(a[part][k], b[part][k]), (rank, part, k)
+
+
+# ---------------------------------------------------------------------- #
+# LLM weight publication: trainer → generation worker
+# ---------------------------------------------------------------------- #
+def _llm_nccl_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from transformers import GPT2Config, GPT2LMHeadModel
+
+    from rl_amd.weight_update import LLMCollectiveWeightSyncScheme
+
+    cfg = GPT2Config(n_layer=2, n_head=2, n_embd=64, vocab_size=128,
+                     bos_token_id=0, eos_token_id=0)
+    torch.manual_seed(100 + rank)  # DIFFERENT init per rank
+    model = GPT2LMHeadModel(cfg)
+    scheme = LLMCollectiveWeightSyncScheme(src=0)
+    scheme.connect(model)
+    if rank == 0:
+        # "mid-training": take an optimizer step before publishing
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        ids = torch.randint(0, 128, (2, 8))
+        loss = model(input_ids=ids, labels=ids).loss
+        loss.backward()
+        opt.step()
+        scheme.send()
+    else:
+        scheme.receive(model)
+    sig = sum(float(v.float().sum()) for v in model.state_dict().values())
+    q.put((rank, sig))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_llm_collective_broadcast_gpt2_mid_training():
+    pytest.importorskip("transformers")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_llm_nccl_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = {}
+    for _ in range(2):
+        rank, sig = q.get(timeout=240)
+        outs[rank] = sig
+    for p in procs:
+        p.join(60)
+    # the generation worker (rank 1) now holds the trainer's weights
+    assert outs[0] == pytest.approx(outs[1], rel=1e-6)
+
+
+def test_llm_double_buffer_roundtrip(tmp_path):
+    pytest.importorskip("transformers")
+    from transformers import GPT2Config, GPT2LMHeadModel
+
+    from rl_amd.weight_update import LLMDoubleBufferWeightSyncScheme
+
+    cfg = GPT2Config(n_layer=2, n_head=2, n_embd=64, vocab_size=128,
+                     bos_token_id=0, eos_token_id=0)
+    torch.manual_seed(0)
+    trainer = GPT2LMHeadModel(cfg)
+    torch.manual_seed(1)
+    worker = GPT2LMHeadModel(cfg)
+    scheme = LLMDoubleBufferWeightSyncScheme(str(tmp_path), model=trainer)
+
+    assert scheme.receive(worker) is False  # nothing published yet
+    scheme.send()
+    assert scheme.receive(worker) is True
+    for k, v in trainer.state_dict().items():
+        assert torch.equal(v, worker.state_dict()[k]), k
+    assert scheme.receive(worker) is False  # same version: no-op
+
+    # A/B alternation: mutate, publish again, worker picks up buffer B
+    with torch.no_grad():
+        for p in trainer.parameters():
+            p.add_(0.25)
+    scheme.send()
+    assert scheme.receive(worker) is True
+    for k, v in trainer.state_dict().items():
+        assert torch.equal(v, worker.state_dict()[k]), k
