@@ -666,45 +666,52 @@ using f32_frag = __attribute__((ext_vector_type(4))) float;
 #define MF_ROWS 16
 #define MF_MAX_ACC 16  // max col tiles per wave (LSTM H=256: 4H/16/4)
 
-// One 16x16 D tile at column j0, contracting a 32-row K-chunk:
-// A(row, k) = a16[row * lda + (kc + k)], k-major chunk B(k, col) =
-// bchunk[k * ldb + j0 + col].
+#define MF_KPAD 40  // chunk row stride in shorts: 32 k + 8 pad → 80 B rows,
+                    // 16 B-aligned fragment reads, odd-dword bank spread
+
+// One 16x16 D tile at column j0, contracting a 32-row K-chunk staged
+// TRANSPOSED (chunkT[j][k], stride MF_KPAD): both fragments load as one
+// 16-byte ds_read_b128.  A(row, k) = a16[row * lda + (kc + k)].
 __device__ __forceinline__ f32_frag mfma_chunk_tile(
     const __hip_bfloat16* a16, int lda, int kc,
-    const __hip_bfloat16* bchunk, int ldb, int j0, int lane, f32_frag acc) {
-  bf16_frag a, b;
+    const __hip_bfloat16* chunkT, int j0, int lane, f32_frag acc) {
   const int row = lane & 15;
   const int koff = 8 * (lane >> 4);
-#pragma unroll
-  for (int r = 0; r < 8; ++r) {
-    const int k = koff + r;
-    a[r] = *reinterpret_cast<const short*>(&a16[(size_t)row * lda + kc + k]);
-    b[r] = *reinterpret_cast<const short*>(&bchunk[(size_t)k * ldb + j0 + row]);
-  }
+  const bf16_frag a =
+      *reinterpret_cast<const bf16_frag*>(&a16[(size_t)row * lda + kc + koff]);
+  const bf16_frag b = *reinterpret_cast<const bf16_frag*>(
+      &chunkT[(size_t)(j0 + row) * MF_KPAD + koff]);
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
 }
 
-// out[16][N] (global fp32, stride N) = A16 [16][K] @ B [K][N]; B comes
-// from global bsrc (k-major, stride N) staged through the LDS chunk
-// buffer (stride N+2).  Every wave owns col tiles wave, wave+4, ...
+// out[16][N] (global fp32, stride ldo) = A16 [16][K] @ B [K][N]; B comes
+// from global bsrc (k-major, stride N) staged transposed through the LDS
+// chunk buffer.  Every wave owns col tiles wave, wave+4, ...
 __device__ void mfma_gemm_store(const __hip_bfloat16* a16, int lda,
                                 const __hip_bfloat16* bsrc, int N, int K,
                                 __hip_bfloat16* chunk, float* out, int ldo,
                                 int wave, int lane, int tid) {
   f32_frag acc[MF_MAX_ACC] = {};
   const int ntiles = N / 16;
-  const int ldc = N + 2;
+  const int n8 = N / 8;
   for (int kc = 0; kc < K; kc += 32) {
     __syncthreads();  // previous consumers done before restage
-    for (int i = tid; i < 32 * N; i += THREADS) {
-      const int k = i / N;
-      const int j = i % N;
-      chunk[(size_t)k * ldc + j] = bsrc[(size_t)(kc + k) * N + j];
+    // 16-byte coalesced global reads, 2-byte transposed LDS scatter
+    for (int i = tid; i < 32 * n8; i += THREADS) {
+      const int k = i / n8;
+      const int j8 = i % n8;
+      const uint4 v4 = *reinterpret_cast<const uint4*>(
+          &bsrc[(size_t)(kc + k) * N + j8 * 8]);
+      const unsigned short* vals = reinterpret_cast<const unsigned short*>(&v4);
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+        chunk[(size_t)(j8 * 8 + m) * MF_KPAD + k] =
+            *reinterpret_cast<const __hip_bfloat16*>(&vals[m]);
     }
     __syncthreads();
     int ai = 0;
     for (int ct = wave; ct < ntiles; ct += 4, ++ai)
-      acc[ai] = mfma_chunk_tile(a16, lda, kc, chunk, ldc, ct * 16, lane, acc[ai]);
+      acc[ai] = mfma_chunk_tile(a16, lda, kc, chunk, ct * 16, lane, acc[ai]);
   }
   int ai = 0;
   for (int ct = wave; ct < ntiles; ct += 4, ++ai) {
@@ -731,7 +738,7 @@ __global__ void __launch_bounds__(THREADS) gru_train_fwd_mfma_kernel(
     const int B, const int T, const int H) {
   extern __shared__ unsigned char smem[];
   const int GH = 3 * H;
-  const int lda = H + 2;
+  const int lda = H + 8;
   __hip_bfloat16* h_bf = reinterpret_cast<__hip_bfloat16*>(smem);           // [16][H+2]
   float* h_f = reinterpret_cast<float*>(smem + (size_t)MF_ROWS * lda * 2);  // [16][H]
   __hip_bfloat16* chunk =
@@ -801,8 +808,8 @@ __global__ void __launch_bounds__(THREADS) gru_bwd_mfma_kernel(
     const int B, const int T, const int H) {
   extern __shared__ unsigned char smem[];
   const int GH = 3 * H;
-  const int lda = H + 2;
-  const int ldg = GH + 2;
+  const int lda = H + 8;
+  const int ldg = GH + 8;
   __hip_bfloat16* hp_bf = reinterpret_cast<__hip_bfloat16*>(smem);  // [16][H+2]
   __hip_bfloat16* g_bf = hp_bf + MF_ROWS * lda;                     // [16][3H+2]
   float* dh = reinterpret_cast<float*>(g_bf + MF_ROWS * ldg);       // [16][H]
@@ -901,7 +908,7 @@ __global__ void __launch_bounds__(THREADS) lstm_train_fwd_mfma_kernel(
     const int B, const int T, const int H) {
   extern __shared__ unsigned char smem[];
   const int GH = 4 * H;
-  const int lda = H + 2;
+  const int lda = H + 8;
   __hip_bfloat16* h_bf = reinterpret_cast<__hip_bfloat16*>(smem);
   float* h_f = reinterpret_cast<float*>(smem + (size_t)MF_ROWS * lda * 2);
   float* c_f = h_f + MF_ROWS * H;
@@ -975,8 +982,8 @@ __global__ void __launch_bounds__(THREADS) lstm_bwd_mfma_kernel(
     const int B, const int T, const int H) {
   extern __shared__ unsigned char smem[];
   const int GH = 4 * H;
-  const int lda = H + 2;
-  const int ldg = GH + 2;
+  const int lda = H + 8;
+  const int ldg = GH + 8;
   __hip_bfloat16* hp_bf = reinterpret_cast<__hip_bfloat16*>(smem);  // [16][H+2]
   __hip_bfloat16* g_bf = hp_bf + MF_ROWS * lda;                     // [16][4H+2]
   float* dh = reinterpret_cast<float*>(g_bf + MF_ROWS * ldg);       // [16][H]
@@ -1094,8 +1101,8 @@ void launch_gru_train_fwd(const float* gates_x, const void* wt_g,
   const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
   if (H % 32 == 0 && 3 * H <= 1024) {
     const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
-    const int lds = (int)(MF_ROWS * (H + 2) * 2 + MF_ROWS * H * 4 +
-                          32 * (3 * H + 2) * 2);
+    const int lds = (int)(MF_ROWS * (H + 8) * 2 + MF_ROWS * H * 4 +
+                          3 * H * MF_KPAD * 2);
     hipLaunchKernelGGL(gru_train_fwd_mfma_kernel, dim3(blocks), dim3(THREADS),
                        lds, (hipStream_t)stream, gates_x, wt, bias_hh,
                        is_init, h0, ys, gscr, B, T, H);
@@ -1124,8 +1131,8 @@ void launch_gru_bwd(const float* gates_x, const void* wt_g,
   const __hip_bfloat16* w_row = (const __hip_bfloat16*)w_row_g;
   if (H % 32 == 0 && 3 * H <= 1024) {
     const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
-    const int lds = (int)(MF_ROWS * (H + 2) * 2 + MF_ROWS * (3 * H + 2) * 2 +
-                          MF_ROWS * H * 4 + 32 * (3 * H + 2) * 2);
+    const int lds = (int)(MF_ROWS * (H + 8) * 2 + MF_ROWS * (3 * H + 8) * 2 +
+                          MF_ROWS * H * 4 + 3 * H * MF_KPAD * 2);
     hipLaunchKernelGGL(gru_bwd_mfma_kernel, dim3(blocks), dim3(THREADS), lds,
                        (hipStream_t)stream, gates_x, wt, w_row, bias_hh,
                        is_init, h0, ys, dys, dgx, dgh, hprev, dh0, gscr, B, T,
@@ -1152,8 +1159,8 @@ void launch_lstm_train_fwd(const float* gates_x, const void* wt_g,
   const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
   if (H % 32 == 0 && 4 * H <= 1024) {
     const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
-    const int lds = (int)(MF_ROWS * (H + 2) * 2 + 2 * MF_ROWS * H * 4 +
-                          32 * (4 * H + 2) * 2);
+    const int lds = (int)(MF_ROWS * (H + 8) * 2 + 2 * MF_ROWS * H * 4 +
+                          4 * H * MF_KPAD * 2);
     hipLaunchKernelGGL(lstm_train_fwd_mfma_kernel, dim3(blocks),
                        dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
                        is_init, h0, c0, ys, cs, gscr, B, T, H);
@@ -1182,8 +1189,8 @@ void launch_lstm_bwd(const float* gates_x, const void* wt_g,
   const __hip_bfloat16* w_row = (const __hip_bfloat16*)w_row_g;
   if (H % 32 == 0 && 4 * H <= 1024) {
     const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
-    const int lds = (int)(MF_ROWS * (H + 2) * 2 + MF_ROWS * (4 * H + 2) * 2 +
-                          2 * MF_ROWS * H * 4 + 32 * (4 * H + 2) * 2);
+    const int lds = (int)(MF_ROWS * (H + 8) * 2 + MF_ROWS * (4 * H + 8) * 2 +
+                          2 * MF_ROWS * H * 4 + 4 * H * MF_KPAD * 2);
     hipLaunchKernelGGL(lstm_bwd_mfma_kernel, dim3(blocks), dim3(THREADS), lds,
                        (hipStream_t)stream, gates_x, wt, w_row, is_init, h0,
                        c0, ys, cs, dys, dg, hprev, dh0, dc0, gscr, B, T, H);

@@ -143,25 +143,39 @@ def _parallel_worker(
     env_kwargs: dict,
     shared_td: Optional[TensorDictBase],
 ):
-    """Worker loop (reference _run_worker_pipe_shared_mem:3322)."""
+    """Worker loop (reference _run_worker_pipe_shared_mem:3322).
+
+    Buffered commands complete via a shared-memory done-flag write
+    (reference _signal_done :3390-3397) — the parent spin-waits one
+    cache line per worker instead of draining N pipe acks; pipes carry
+    only the command tuples and the unbuffered fallback payloads."""
     torch.set_num_threads(1)
     env = env_fn(**env_kwargs)
     my_slice = shared_td[idx] if shared_td is not None else None
+    flags: Optional[torch.Tensor] = None
+    seq = 0
     pipe.send(("meta", EnvMetaData.build(env)))
     root_td: Optional[TensorDictBase] = None
     action_keys = env.full_action_spec.keys(True, True) or ["action"]
+
+    def _signal_done():
+        nonlocal seq
+        seq += 1
+        flags[idx] = seq
+
     try:
         while True:
             cmd, data = pipe.recv()
             if cmd == "set_buffer":
                 # shared-memory payload buffer: this worker owns row idx
-                my_slice = data[idx]
+                my_slice = data[0][idx]
+                flags = data[1]
                 pipe.send(("buffer_set", None))
             elif cmd == "reset":
                 root_td = env.reset(data)
                 if my_slice is not None:
                     _copy_into(my_slice.get("root_next"), root_td)
-                    pipe.send(("done", None))
+                    _signal_done()
                 else:
                     pipe.send(("done", root_td))
             elif cmd == "step":
@@ -175,7 +189,7 @@ def _parallel_worker(
                 next_td = td.get("next")
                 if my_slice is not None:
                     _copy_into(my_slice.get("next"), next_td)
-                    pipe.send(("done", None))
+                    _signal_done()
                 else:
                     pipe.send(("done", next_td))
                 root_td = step_mdp(td)
@@ -190,7 +204,7 @@ def _parallel_worker(
                 if my_slice is not None:
                     _copy_into(my_slice.get("next"), td.get("next"))
                     _copy_into(my_slice.get("root_next"), next_root)
-                    pipe.send(("done", None))
+                    _signal_done()
                 else:
                     pipe.send(("done", (td.get("next"), next_root)))
                 root_td = next_root
@@ -276,10 +290,52 @@ class ParallelEnv(BatchedEnvBase):
         buf.set("root_next", root)
         buf.share_memory_()
         self._shared = buf
-        self._send_all("set_buffer", [buf] * n)
+        # completion flags: one shared int64 per worker; workers bump
+        # their slot, the parent spin-waits the vector (reference
+        # shm done-flags, batched_envs.py:3385-3397)
+        self._flags = torch.zeros(n, dtype=torch.int64).share_memory_()
+        self._seq = 0
+        self._send_all("set_buffer", [(buf, self._flags)] * n)
         for pipe in self.parent_pipes:
             msg, _ = pipe.recv()
             assert msg == "buffer_set"
+        if self.device is not None and self.device.type == "cuda":
+            self._pin_shared_buffer()
+
+    def _pin_shared_buffer(self):
+        """Host-register the shared-memory pages (hipHostRegister via
+        torch's cudart shim) so parent-side ``.to(device)`` copies are
+        true async DMA from pinned memory (reference keeps pinned
+        shared buffers for GPU parents)."""
+        try:
+            cudart = torch.cuda.cudart()
+            for _k, v in self._shared.items(True, True):
+                storage = v.untyped_storage()
+                r = cudart.cudaHostRegister(storage.data_ptr(), storage.nbytes(), 0)
+                if int(r) != 0:
+                    raise RuntimeError(f"cudaHostRegister -> {r}")
+            self._pinned = True
+        except Exception as e:  # best-effort: unpinned copies still work
+            logger.debug(f"ParallelEnv: host-register failed ({e!r})")
+            self._pinned = False
+
+    def _wait_flags(self):
+        """Spin-wait until every worker's flag reaches the current
+        sequence number (short pure spin, then yielding sleep)."""
+        self._seq += 1
+        seq = self._seq
+        flags = self._flags
+        deadline = time.monotonic() + BATCHED_PIPE_TIMEOUT
+        spins = 0
+        while True:
+            if bool((flags >= seq).all()):
+                return
+            spins += 1
+            if spins > 2000:
+                time.sleep(5e-5)
+            if spins % 512 == 0 and time.monotonic() > deadline:
+                self._check()
+                raise TimeoutError("ParallelEnv worker timed out (flags)")
 
     def _set_specs_from_meta(self, meta: EnvMetaData):
         n = self.num_workers
@@ -316,13 +372,13 @@ class ParallelEnv(BatchedEnvBase):
             sub = tensordict[i].clone(False) if tensordict is not None else None
             datas.append(sub)
         self._send_all("reset", datas)
-        outs = self._recv_all()
         if self._shared is not None:
+            self._wait_flags()
             out = self._shared.get("root_next").clone()
         else:
-            out = td_stack(outs, 0)
+            out = td_stack(self._recv_all(), 0)
         if self.device is not None and out.device != self.device:
-            out = out.to(self.device)
+            out = out.to(self.device, non_blocking=getattr(self, "_pinned", False))
         return out
 
     def _write_actions(self, tensordict: TensorDictBase) -> None:
@@ -336,7 +392,7 @@ class ParallelEnv(BatchedEnvBase):
         if self._shared is not None:
             self._write_actions(tensordict)
             self._send_all("step", [None] * self.num_workers)
-            self._recv_all()
+            self._wait_flags()
             out = self._shared.get("next").clone()
         else:
             datas = []
@@ -355,7 +411,7 @@ class ParallelEnv(BatchedEnvBase):
         if self._shared is not None:
             self._write_actions(tensordict)
             self._send_all("step_and_maybe_reset", [None] * self.num_workers)
-            self._recv_all()
+            self._wait_flags()
             next_tds = self._shared.get("next").clone()
             next_roots = self._shared.get("root_next").clone()
         else:
@@ -367,8 +423,9 @@ class ParallelEnv(BatchedEnvBase):
             next_tds = td_stack([o[0] for o in outs], 0)
             next_roots = td_stack([o[1] for o in outs], 0)
         if self.device is not None:
-            next_tds = next_tds.to(self.device)
-            next_roots = next_roots.to(self.device)
+            nb = getattr(self, "_pinned", False)
+            next_tds = next_tds.to(self.device, non_blocking=nb)
+            next_roots = next_roots.to(self.device, non_blocking=nb)
         self._complete_done(next_tds)
         tensordict.set("next", next_tds)
         return tensordict, next_roots
