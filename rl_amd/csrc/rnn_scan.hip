@@ -647,14 +647,17 @@ __global__ void __launch_bounds__(THREADS) lstm_bwd_kernel(
 //  * one workgroup = 16 batch rows (one MFMA A-tile) x 4 waves; waves
 //    split the gate-column tiles; accumulators persist across the
 //    K loop in AGPRs.
-//  * W is staged through LDS in 32-row K-chunks per step (bf16,
-//    padded stride) — no residency requirement, so ANY H%32==0 works
-//    with the same code path and W stays L2-resident across steps.
+//  * W placement adapts to H (template RESW): when the TRANSPOSED
+//    bf16 W fits LDS next to the state tiles (H<=128-ish) it is staged
+//    ONCE and every per-step B fragment is a single 16-byte
+//    ds_read_b128 — no staging work or barriers inside the scan.
+//    Otherwise W is staged through LDS in 32-row K-chunks per step
+//    (k-major, coalesced) and stays L2-resident across steps.
 //  * per-step gate pre-activations land in a global fp32 scratch
-//    [B, G*H] (written/read once per step, coalesced) instead of LDS —
-//    this keeps the LDS budget at ~75-140 KB for H=256.
-//  * backward: two MFMA GEMMs per step — gate recompute (hp @ W^T) and
-//    the carry (dgates @ W, dgates rounded to bf16 for the A fragment;
+//    [B, G*H] (written/read once per step, coalesced) instead of LDS.
+//  * backward: two MFMA GEMMs per step — gate recompute (hp @ W^T,
+//    RESW-eligible) and the carry (dgates @ W, always chunked from the
+//    row-major copy; dgates rounded to bf16 for the A fragment,
 //    parameter-gradient outputs stay fp32).
 // ------------------------------------------------------------------- //
 
@@ -666,52 +669,53 @@ using f32_frag = __attribute__((ext_vector_type(4))) float;
 #define MF_ROWS 16
 #define MF_MAX_ACC 16  // max col tiles per wave (LSTM H=256: 4H/16/4)
 
-#define MF_KPAD 40  // chunk row stride in shorts: 32 k + 8 pad → 80 B rows,
-                    // 16 B-aligned fragment reads, odd-dword bank spread
-
-// One 16x16 D tile at column j0, contracting a 32-row K-chunk staged
-// TRANSPOSED (chunkT[j][k], stride MF_KPAD): both fragments load as one
-// 16-byte ds_read_b128.  A(row, k) = a16[row * lda + (kc + k)].
+// ---- chunked path: B staged k-major [32][N+2] per K-chunk ---------- //
 __device__ __forceinline__ f32_frag mfma_chunk_tile(
-    const __hip_bfloat16* a16, int lda, int kc,
-    const __hip_bfloat16* chunkT, int j0, int lane, f32_frag acc) {
+    const __hip_bfloat16* a16, int lda, int ka,
+    const __hip_bfloat16* bchunk, int ldb, int j0, int lane, f32_frag acc) {
   const int row = lane & 15;
   const int koff = 8 * (lane >> 4);
+  // A rows are 16-byte aligned (lda % 8 == 0): one ds_read_b128
   const bf16_frag a =
-      *reinterpret_cast<const bf16_frag*>(&a16[(size_t)row * lda + kc + koff]);
-  const bf16_frag b = *reinterpret_cast<const bf16_frag*>(
-      &chunkT[(size_t)(j0 + row) * MF_KPAD + koff]);
+      *reinterpret_cast<const bf16_frag*>(&a16[(size_t)row * lda + ka + koff]);
+  bf16_frag b;
+#pragma unroll
+  for (int r = 0; r < 8; ++r) {
+    const int k = koff + r;
+    b[r] = *reinterpret_cast<const short*>(&bchunk[(size_t)k * ldb + j0 + row]);
+  }
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
 }
 
-// out[16][N] (global fp32, stride ldo) = A16 [16][K] @ B [K][N]; B comes
-// from global bsrc (k-major, stride N) staged transposed through the LDS
-// chunk buffer.  Every wave owns col tiles wave, wave+4, ...
-__device__ void mfma_gemm_store(const __hip_bfloat16* a16, int lda,
-                                const __hip_bfloat16* bsrc, int N, int K,
-                                __hip_bfloat16* chunk, float* out, int ldo,
-                                int wave, int lane, int tid) {
+// out[16][N] (global fp32, stride ldo) = A16 [16][K] @ B [K][N]; B
+// staged from global bsrc (k-major, stride N) in 32-row chunks.
+__device__ void mfma_gemm_chunked(const __hip_bfloat16* a16, int lda,
+                                  const __hip_bfloat16* bsrc, int N, int K,
+                                  __hip_bfloat16* chunk, int chunk_rows,
+                                  float* out, int ldo, int wave, int lane,
+                                  int tid) {
   f32_frag acc[MF_MAX_ACC] = {};
   const int ntiles = N / 16;
-  const int n8 = N / 8;
-  for (int kc = 0; kc < K; kc += 32) {
+  const int ldc = N + 8;  // 16-byte-aligned rows for vector staging
+  const int n8 = N / 8;   // N % 8 == 0 always (H % 32 == 0)
+  for (int kc = 0; kc < K; kc += chunk_rows) {
+    const int rows_now = min(chunk_rows, K - kc);
     __syncthreads();  // previous consumers done before restage
-    // 16-byte coalesced global reads, 2-byte transposed LDS scatter
-    for (int i = tid; i < 32 * n8; i += THREADS) {
+    for (int i = tid; i < rows_now * n8; i += THREADS) {
       const int k = i / n8;
       const int j8 = i % n8;
-      const uint4 v4 = *reinterpret_cast<const uint4*>(
-          &bsrc[(size_t)(kc + k) * N + j8 * 8]);
-      const unsigned short* vals = reinterpret_cast<const unsigned short*>(&v4);
-#pragma unroll
-      for (int m = 0; m < 8; ++m)
-        chunk[(size_t)(j8 * 8 + m) * MF_KPAD + k] =
-            *reinterpret_cast<const __hip_bfloat16*>(&vals[m]);
+      *reinterpret_cast<uint4*>(&chunk[(size_t)k * ldc + j8 * 8]) =
+          *reinterpret_cast<const uint4*>(
+              &bsrc[(size_t)(kc + k) * N + j8 * 8]);
     }
     __syncthreads();
-    int ai = 0;
-    for (int ct = wave; ct < ntiles; ct += 4, ++ai)
-      acc[ai] = mfma_chunk_tile(a16, lda, kc, chunk, ct * 16, lane, acc[ai]);
+    for (int kk = 0; kk < rows_now; kk += 32) {
+      int ai = 0;
+      for (int ct = wave; ct < ntiles; ct += 4, ++ai)
+        acc[ai] =
+            mfma_chunk_tile(a16, lda, kc + kk, chunk + (size_t)kk * ldc, ldc,
+                            ct * 16, lane, acc[ai]);
+    }
   }
   int ai = 0;
   for (int ct = wave; ct < ntiles; ct += 4, ++ai) {
@@ -726,7 +730,49 @@ __device__ void mfma_gemm_store(const __hip_bfloat16* a16, int lda,
                           // caller's __syncthreads
 }
 
+// ---- resident path: W pre-staged TRANSPOSED [N][K+8] --------------- //
+// B(k, col) = wres[(j0+col) * (K+8) + k]: both fragments are single
+// 16-byte LDS reads; no staging inside the scan.
+__device__ void stage_w_resident(const __hip_bfloat16* bsrc, int N, int K,
+                                 __hip_bfloat16* wres, int tid) {
+  const int ldw = K + 8;
+  for (int i = tid; i < K * N; i += THREADS) {
+    const int k = i / N;  // coalesced global reads (consecutive j)
+    const int j = i % N;
+    wres[(size_t)j * ldw + k] = bsrc[(size_t)k * N + j];
+  }
+}
+
+__device__ void mfma_gemm_resident(const __hip_bfloat16* a16, int lda,
+                                   const __hip_bfloat16* wres, int N, int K,
+                                   float* out, int ldo, int wave, int lane) {
+  f32_frag acc[MF_MAX_ACC] = {};
+  const int ntiles = N / 16;
+  const int ldw = K + 8;
+  const int row = lane & 15;
+  const int koff = 8 * (lane >> 4);
+  for (int kc = 0; kc < K; kc += 32) {
+    const bf16_frag a = *reinterpret_cast<const bf16_frag*>(
+        &a16[(size_t)row * lda + kc + koff]);
+    int ai = 0;
+    for (int ct = wave; ct < ntiles; ct += 4, ++ai) {
+      const bf16_frag b = *reinterpret_cast<const bf16_frag*>(
+          &wres[(size_t)(ct * 16 + row) * ldw + kc + koff]);
+      acc[ai] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[ai], 0, 0, 0);
+    }
+  }
+  int ai = 0;
+  for (int ct = wave; ct < ntiles; ct += 4, ++ai) {
+    const int col = ct * 16 + (lane & 15);
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      out[(size_t)((lane >> 4) * 4 + r) * ldo + col] = acc[ai][r];
+  }
+  __threadfence_block();
+}
+
 // GRU MFMA forward: gates scratch gscr is [B, 3H] fp32 (per-step reuse).
+template <bool RESW>
 __global__ void __launch_bounds__(THREADS) gru_train_fwd_mfma_kernel(
     const float* __restrict__ gates_x,       // [B, T, 3H]
     const __hip_bfloat16* __restrict__ wt_g, // [H, 3H]
@@ -735,14 +781,14 @@ __global__ void __launch_bounds__(THREADS) gru_train_fwd_mfma_kernel(
     const float* __restrict__ h0,            // [B, H] or nullptr
     float* __restrict__ ys,                  // [B, T, H]
     float* __restrict__ gscr,                // [B, 3H] scratch
-    const int B, const int T, const int H) {
+    const int B, const int T, const int H, const int crows) {
   extern __shared__ unsigned char smem[];
   const int GH = 3 * H;
   const int lda = H + 8;
-  __hip_bfloat16* h_bf = reinterpret_cast<__hip_bfloat16*>(smem);           // [16][H+2]
+  __hip_bfloat16* h_bf = reinterpret_cast<__hip_bfloat16*>(smem);           // [16][H+8]
   float* h_f = reinterpret_cast<float*>(smem + (size_t)MF_ROWS * lda * 2);  // [16][H]
-  __hip_bfloat16* chunk =
-      reinterpret_cast<__hip_bfloat16*>(h_f + MF_ROWS * H);                 // [32][3H+2]
+  __hip_bfloat16* wbuf =
+      reinterpret_cast<__hip_bfloat16*>(h_f + MF_ROWS * H);  // Wres or chunk
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -751,11 +797,11 @@ __global__ void __launch_bounds__(THREADS) gru_train_fwd_mfma_kernel(
   const int rows = min(MF_ROWS, B - row0);
   float* gs = gscr + (size_t)row0 * GH;
 
+  if (RESW) stage_w_resident(wt_g, GH, H, wbuf, tid);
   for (int i = tid; i < MF_ROWS * H; i += THREADS) {
     const int b = i / H;
     const int k = i % H;
-    const float v =
-        (b < rows && h0) ? h0[(size_t)(row0 + b) * H + k] : 0.0f;
+    const float v = (b < rows && h0) ? h0[(size_t)(row0 + b) * H + k] : 0.0f;
     h_f[b * H + k] = v;
     h_bf[b * lda + k] = __float2bfloat16(v);
   }
@@ -769,8 +815,12 @@ __global__ void __launch_bounds__(THREADS) gru_train_fwd_mfma_kernel(
         h_bf[b * lda + i % H] = __float2bfloat16(0.0f);
       }
     }
-    // gates_h = h @ W^T  (A = h_bf, B = wt chunks)
-    mfma_gemm_store(h_bf, lda, wt_g, GH, H, chunk, gs, GH, wave, lane, tid);
+    __syncthreads();
+    if (RESW)
+      mfma_gemm_resident(h_bf, lda, wbuf, GH, H, gs, GH, wave, lane);
+    else
+      mfma_gemm_chunked(h_bf, lda, wt_g, GH, H, wbuf, crows, gs, GH, wave,
+                        lane, tid);
     __syncthreads();
     for (int i = tid; i < rows * H; i += THREADS) {
       const int b = i / H;
@@ -791,6 +841,7 @@ __global__ void __launch_bounds__(THREADS) gru_train_fwd_mfma_kernel(
   }
 }
 
+template <bool RESW>
 __global__ void __launch_bounds__(THREADS) gru_bwd_mfma_kernel(
     const float* __restrict__ gates_x,        // [B, T, 3H]
     const __hip_bfloat16* __restrict__ wt_g,  // [H, 3H] (W^T, k-major)
@@ -805,15 +856,18 @@ __global__ void __launch_bounds__(THREADS) gru_bwd_mfma_kernel(
     float* __restrict__ hprev,                // [B, T, H]
     float* __restrict__ dh0,                  // [B, H]
     float* __restrict__ gscr,                 // [B, 3H] scratch
-    const int B, const int T, const int H) {
+    const int B, const int T, const int H, const int crows_rec,
+    const int crows_car) {
   extern __shared__ unsigned char smem[];
   const int GH = 3 * H;
   const int lda = H + 8;
   const int ldg = GH + 8;
-  __hip_bfloat16* hp_bf = reinterpret_cast<__hip_bfloat16*>(smem);  // [16][H+2]
-  __hip_bfloat16* g_bf = hp_bf + MF_ROWS * lda;                     // [16][3H+2]
+  __hip_bfloat16* hp_bf = reinterpret_cast<__hip_bfloat16*>(smem);  // [16][H+8]
+  __hip_bfloat16* g_bf = hp_bf + MF_ROWS * lda;                     // [16][3H+8]
   float* dh = reinterpret_cast<float*>(g_bf + MF_ROWS * ldg);       // [16][H]
+  // chunk buffer for the carry GEMM (always chunked), then Wres
   __hip_bfloat16* chunk = reinterpret_cast<__hip_bfloat16*>(dh + MF_ROWS * H);
+  __hip_bfloat16* wres = chunk + (size_t)crows_car * (H + 8);
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -822,13 +876,13 @@ __global__ void __launch_bounds__(THREADS) gru_bwd_mfma_kernel(
   const int rows = min(MF_ROWS, B - row0);
   float* gs = gscr + (size_t)row0 * GH;
 
+  if (RESW) stage_w_resident(wt_g, GH, H, wres, tid);
   for (int i = tid; i < MF_ROWS * H; i += THREADS) dh[i] = 0.0f;
   for (int i = tid; i < MF_ROWS * ldg; i += THREADS)
     g_bf[i] = __float2bfloat16(0.0f);
   __syncthreads();
 
   for (int t = T - 1; t >= 0; --t) {
-    // stage effective h_{t-1} (bf16 for the recompute GEMM)
     for (int i = tid; i < MF_ROWS * H; i += THREADS) {
       const int b = i / H;
       const int k = i % H;
@@ -841,11 +895,13 @@ __global__ void __launch_bounds__(THREADS) gru_bwd_mfma_kernel(
       }
       hp_bf[b * lda + k] = __float2bfloat16(v);
     }
-    // recompute gates_h = hp @ W^T into the global scratch
-    mfma_gemm_store(hp_bf, lda, wt_g, GH, H, chunk, gs, GH, wave, lane, tid);
     __syncthreads();
-    // pointwise grads; stash bf16 dgates for the carry GEMM, fold
-    // z·dh into the dh buffer in place (its old value is consumed here)
+    if (RESW)
+      mfma_gemm_resident(hp_bf, lda, wres, GH, H, gs, GH, wave, lane);
+    else
+      mfma_gemm_chunked(hp_bf, lda, wt_g, GH, H, chunk, crows_rec, gs, GH,
+                        wave, lane, tid);
+    __syncthreads();
     for (int i = tid; i < rows * H; i += THREADS) {
       const int b = i / H;
       const int jh = i % H;
@@ -877,9 +933,9 @@ __global__ void __launch_bounds__(THREADS) gru_bwd_mfma_kernel(
       dh[b * H + jh] = dh_t * z;  // carry contribution through the z path
     }
     __syncthreads();
-    // carry GEMM: dcontrib[16, H] = dgates [16, 3H] @ W [3H, H],
-    // written to the FRONT of the scratch row block (fp32 [16, H])
-    mfma_gemm_store(g_bf, ldg, w_row, H, GH, chunk, gs, H, wave, lane, tid);
+    // carry GEMM: dcontrib[16, H] = dgates [16, 3H] @ W [3H, H]
+    mfma_gemm_chunked(g_bf, ldg, w_row, H, GH, chunk, crows_car, gs, H, wave,
+                      lane, tid);
     __syncthreads();
     for (int i = tid; i < rows * H; i += THREADS) {
       const int b = i / H;
@@ -897,6 +953,7 @@ __global__ void __launch_bounds__(THREADS) gru_bwd_mfma_kernel(
   }
 }
 
+template <bool RESW>
 __global__ void __launch_bounds__(THREADS) lstm_train_fwd_mfma_kernel(
     const float* __restrict__ gates_x,       // [B, T, 4H]
     const __hip_bfloat16* __restrict__ wt_g, // [H, 4H]
@@ -905,14 +962,14 @@ __global__ void __launch_bounds__(THREADS) lstm_train_fwd_mfma_kernel(
     float* __restrict__ ys,   // [B, T, H]
     float* __restrict__ cs,   // [B, T, H]
     float* __restrict__ gscr, // [B, 4H]
-    const int B, const int T, const int H) {
+    const int B, const int T, const int H, const int crows) {
   extern __shared__ unsigned char smem[];
   const int GH = 4 * H;
   const int lda = H + 8;
   __hip_bfloat16* h_bf = reinterpret_cast<__hip_bfloat16*>(smem);
   float* h_f = reinterpret_cast<float*>(smem + (size_t)MF_ROWS * lda * 2);
   float* c_f = h_f + MF_ROWS * H;
-  __hip_bfloat16* chunk = reinterpret_cast<__hip_bfloat16*>(c_f + MF_ROWS * H);
+  __hip_bfloat16* wbuf = reinterpret_cast<__hip_bfloat16*>(c_f + MF_ROWS * H);
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -921,6 +978,7 @@ __global__ void __launch_bounds__(THREADS) lstm_train_fwd_mfma_kernel(
   const int rows = min(MF_ROWS, B - row0);
   float* gs = gscr + (size_t)row0 * GH;
 
+  if (RESW) stage_w_resident(wt_g, GH, H, wbuf, tid);
   for (int i = tid; i < MF_ROWS * H; i += THREADS) {
     const int b = i / H;
     const int k = i % H;
@@ -942,7 +1000,12 @@ __global__ void __launch_bounds__(THREADS) lstm_train_fwd_mfma_kernel(
         h_bf[b * lda + k] = __float2bfloat16(0.0f);
       }
     }
-    mfma_gemm_store(h_bf, lda, wt_g, GH, H, chunk, gs, GH, wave, lane, tid);
+    __syncthreads();
+    if (RESW)
+      mfma_gemm_resident(h_bf, lda, wbuf, GH, H, gs, GH, wave, lane);
+    else
+      mfma_gemm_chunked(h_bf, lda, wt_g, GH, H, wbuf, crows, gs, GH, wave,
+                        lane, tid);
     __syncthreads();
     for (int i = tid; i < rows * H; i += THREADS) {
       const int b = i / H;
@@ -966,6 +1029,7 @@ __global__ void __launch_bounds__(THREADS) lstm_train_fwd_mfma_kernel(
   }
 }
 
+template <bool RESW>
 __global__ void __launch_bounds__(THREADS) lstm_bwd_mfma_kernel(
     const float* __restrict__ gates_x,        // [B, T, 4H]
     const __hip_bfloat16* __restrict__ wt_g,  // [H, 4H]
@@ -979,16 +1043,18 @@ __global__ void __launch_bounds__(THREADS) lstm_bwd_mfma_kernel(
     float* __restrict__ hprev,      // [B, T, H]
     float* __restrict__ dh0, float* __restrict__ dc0,
     float* __restrict__ gscr,       // [B, 4H]
-    const int B, const int T, const int H) {
+    const int B, const int T, const int H, const int crows_rec,
+    const int crows_car) {
   extern __shared__ unsigned char smem[];
   const int GH = 4 * H;
   const int lda = H + 8;
   const int ldg = GH + 8;
-  __hip_bfloat16* hp_bf = reinterpret_cast<__hip_bfloat16*>(smem);  // [16][H+2]
-  __hip_bfloat16* g_bf = hp_bf + MF_ROWS * lda;                     // [16][4H+2]
+  __hip_bfloat16* hp_bf = reinterpret_cast<__hip_bfloat16*>(smem);  // [16][H+8]
+  __hip_bfloat16* g_bf = hp_bf + MF_ROWS * lda;                     // [16][4H+8]
   float* dh = reinterpret_cast<float*>(g_bf + MF_ROWS * ldg);       // [16][H]
   float* dc = dh + MF_ROWS * H;                                     // [16][H]
   __hip_bfloat16* chunk = reinterpret_cast<__hip_bfloat16*>(dc + MF_ROWS * H);
+  __hip_bfloat16* wres = chunk + (size_t)crows_car * (H + 8);
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -997,6 +1063,7 @@ __global__ void __launch_bounds__(THREADS) lstm_bwd_mfma_kernel(
   const int rows = min(MF_ROWS, B - row0);
   float* gs = gscr + (size_t)row0 * GH;
 
+  if (RESW) stage_w_resident(wt_g, GH, H, wres, tid);
   for (int i = tid; i < MF_ROWS * H; i += THREADS) {
     dh[i] = 0.0f;
     dc[i] = 0.0f;
@@ -1018,7 +1085,12 @@ __global__ void __launch_bounds__(THREADS) lstm_bwd_mfma_kernel(
       }
       hp_bf[b * lda + k] = __float2bfloat16(hv);
     }
-    mfma_gemm_store(hp_bf, lda, wt_g, GH, H, chunk, gs, GH, wave, lane, tid);
+    __syncthreads();
+    if (RESW)
+      mfma_gemm_resident(hp_bf, lda, wres, GH, H, gs, GH, wave, lane);
+    else
+      mfma_gemm_chunked(hp_bf, lda, wt_g, GH, H, chunk, crows_rec, gs, GH,
+                        wave, lane, tid);
     __syncthreads();
     for (int i = tid; i < rows * H; i += THREADS) {
       const int b = i / H;
@@ -1059,7 +1131,8 @@ __global__ void __launch_bounds__(THREADS) lstm_bwd_mfma_kernel(
       dc[b * H + jh] = init ? 0.0f : dct * fg;
     }
     __syncthreads();
-    mfma_gemm_store(g_bf, ldg, w_row, H, GH, chunk, gs, H, wave, lane, tid);
+    mfma_gemm_chunked(g_bf, ldg, w_row, H, GH, chunk, crows_car, gs, H, wave,
+                      lane, tid);
     __syncthreads();
     for (int i = tid; i < rows * H; i += THREADS) {
       const int b = i / H;
@@ -1079,6 +1152,15 @@ __global__ void __launch_bounds__(THREADS) lstm_bwd_mfma_kernel(
 }
 
 }  // namespace
+
+static int _floor32(int x) { return (x / 32) * 32; }
+static int _chunk_rows(int avail_bytes, int row_shorts, int K) {
+  int r = _floor32(avail_bytes / 2 / row_shorts);
+  if (r < 32) r = 32;
+  if (r > 32) r = 32;  // measured: larger K-chunks do not pay (less
+                       // barrier-amortization win than L2-reuse loss)
+  return r;
+}
 
 extern "C" {
 
@@ -1101,11 +1183,19 @@ void launch_gru_train_fwd(const float* gates_x, const void* wt_g,
   const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
   if (H % 32 == 0 && 3 * H <= 1024) {
     const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
-    const int lds = (int)(MF_ROWS * (H + 8) * 2 + MF_ROWS * H * 4 +
-                          3 * H * MF_KPAD * 2);
-    hipLaunchKernelGGL(gru_train_fwd_mfma_kernel, dim3(blocks), dim3(THREADS),
-                       lds, (hipStream_t)stream, gates_x, wt, bias_hh,
-                       is_init, h0, ys, gscr, B, T, H);
+    const int base = MF_ROWS * (H + 8) * 2 + MF_ROWS * H * 4;
+    const int lds_res = base + 3 * H * (H + 8) * 2;
+    if (lds_res <= 160 * 1024) {
+      hipLaunchKernelGGL(gru_train_fwd_mfma_kernel<true>, dim3(blocks),
+                         dim3(THREADS), lds_res, (hipStream_t)stream, gates_x,
+                         wt, bias_hh, is_init, h0, ys, gscr, B, T, H, 0);
+    } else {
+      const int cr = _chunk_rows(160 * 1024 - base, 3 * H + 8, H);
+      const int lds = base + cr * (3 * H + 8) * 2;
+      hipLaunchKernelGGL(gru_train_fwd_mfma_kernel<false>, dim3(blocks),
+                         dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
+                         bias_hh, is_init, h0, ys, gscr, B, T, H, cr);
+    }
     return;
   }
   const int blocks = (B + BW_B_TILE - 1) / BW_B_TILE;
@@ -1131,12 +1221,32 @@ void launch_gru_bwd(const float* gates_x, const void* wt_g,
   const __hip_bfloat16* w_row = (const __hip_bfloat16*)w_row_g;
   if (H % 32 == 0 && 3 * H <= 1024) {
     const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
-    const int lds = (int)(MF_ROWS * (H + 8) * 2 + MF_ROWS * (3 * H + 8) * 2 +
-                          MF_ROWS * H * 4 + 3 * H * MF_KPAD * 2);
-    hipLaunchKernelGGL(gru_bwd_mfma_kernel, dim3(blocks), dim3(THREADS), lds,
-                       (hipStream_t)stream, gates_x, wt, w_row, bias_hh,
-                       is_init, h0, ys, dys, dgx, dgh, hprev, dh0, gscr, B, T,
-                       H);
+    const int base = MF_ROWS * (H + 8) * 2 + MF_ROWS * (3 * H + 8) * 2 +
+                     MF_ROWS * H * 4;
+    const int wres_b = 3 * H * (H + 8) * 2;
+    if (base + 32 * (H + 8) * 2 + wres_b <= 160 * 1024) {
+      // resident recompute W; size the carry chunk from what is left
+      const int cr_car =
+          _chunk_rows(160 * 1024 - base - wres_b, H + 8, 3 * H);
+      const int lds_res = base + cr_car * (H + 8) * 2 + wres_b;
+      hipLaunchKernelGGL(gru_bwd_mfma_kernel<true>, dim3(blocks),
+                         dim3(THREADS), lds_res, (hipStream_t)stream, gates_x,
+                         wt, w_row, bias_hh, is_init, h0, ys, dys, dgx, dgh,
+                         hprev, dh0, gscr, B, T, H, 0, cr_car);
+    } else {
+      // one chunk buffer shared by both GEMMs
+      const int avail = 160 * 1024 - base;
+      const int cr_rec = _chunk_rows(avail, 3 * H + 8, H);
+      int cr_car = _chunk_rows(avail, H + 8, 3 * H);
+      const int bufshorts = cr_rec * (3 * H + 8) > cr_car * (H + 8)
+                                ? cr_rec * (3 * H + 8)
+                                : cr_car * (H + 8);
+      const int lds = base + bufshorts * 2;
+      hipLaunchKernelGGL(gru_bwd_mfma_kernel<false>, dim3(blocks),
+                         dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
+                         w_row, bias_hh, is_init, h0, ys, dys, dgx, dgh,
+                         hprev, dh0, gscr, B, T, H, cr_rec, cr_car);
+    }
     return;
   }
   const int blocks = (B + BW_B_TILE - 1) / BW_B_TILE;
@@ -1159,11 +1269,19 @@ void launch_lstm_train_fwd(const float* gates_x, const void* wt_g,
   const __hip_bfloat16* wt = (const __hip_bfloat16*)wt_g;
   if (H % 32 == 0 && 4 * H <= 1024) {
     const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
-    const int lds = (int)(MF_ROWS * (H + 8) * 2 + 2 * MF_ROWS * H * 4 +
-                          4 * H * MF_KPAD * 2);
-    hipLaunchKernelGGL(lstm_train_fwd_mfma_kernel, dim3(blocks),
-                       dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
-                       is_init, h0, c0, ys, cs, gscr, B, T, H);
+    const int base = MF_ROWS * (H + 8) * 2 + 2 * MF_ROWS * H * 4;
+    const int lds_res = base + 4 * H * (H + 8) * 2;
+    if (lds_res <= 160 * 1024) {
+      hipLaunchKernelGGL(lstm_train_fwd_mfma_kernel<true>, dim3(blocks),
+                         dim3(THREADS), lds_res, (hipStream_t)stream, gates_x,
+                         wt, is_init, h0, c0, ys, cs, gscr, B, T, H, 0);
+    } else {
+      const int cr = _chunk_rows(160 * 1024 - base, 4 * H + 8, H);
+      const int lds = base + cr * (4 * H + 8) * 2;
+      hipLaunchKernelGGL(lstm_train_fwd_mfma_kernel<false>, dim3(blocks),
+                         dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
+                         is_init, h0, c0, ys, cs, gscr, B, T, H, cr);
+    }
     return;
   }
   const int blocks = (B + BW_B_TILE_LSTM - 1) / BW_B_TILE_LSTM;
@@ -1189,11 +1307,30 @@ void launch_lstm_bwd(const float* gates_x, const void* wt_g,
   const __hip_bfloat16* w_row = (const __hip_bfloat16*)w_row_g;
   if (H % 32 == 0 && 4 * H <= 1024) {
     const int blocks = (B + MF_ROWS - 1) / MF_ROWS;
-    const int lds = (int)(MF_ROWS * (H + 8) * 2 + MF_ROWS * (4 * H + 8) * 2 +
-                          2 * MF_ROWS * H * 4 + 4 * H * MF_KPAD * 2);
-    hipLaunchKernelGGL(lstm_bwd_mfma_kernel, dim3(blocks), dim3(THREADS), lds,
-                       (hipStream_t)stream, gates_x, wt, w_row, is_init, h0,
-                       c0, ys, cs, dys, dg, hprev, dh0, dc0, gscr, B, T, H);
+    const int base = MF_ROWS * (H + 8) * 2 + MF_ROWS * (4 * H + 8) * 2 +
+                     2 * MF_ROWS * H * 4;
+    const int wres_b = 4 * H * (H + 8) * 2;
+    if (base + 32 * (H + 8) * 2 + wres_b <= 160 * 1024) {
+      const int cr_car =
+          _chunk_rows(160 * 1024 - base - wres_b, H + 8, 4 * H);
+      const int lds_res = base + cr_car * (H + 8) * 2 + wres_b;
+      hipLaunchKernelGGL(lstm_bwd_mfma_kernel<true>, dim3(blocks),
+                         dim3(THREADS), lds_res, (hipStream_t)stream, gates_x,
+                         wt, w_row, is_init, h0, c0, ys, cs, dys, dg, hprev,
+                         dh0, dc0, gscr, B, T, H, 0, cr_car);
+    } else {
+      const int avail = 160 * 1024 - base;
+      const int cr_rec = _chunk_rows(avail, 4 * H + 8, H);
+      int cr_car = _chunk_rows(avail, H + 8, 4 * H);
+      const int bufshorts = cr_rec * (4 * H + 8) > cr_car * (H + 8)
+                                ? cr_rec * (4 * H + 8)
+                                : cr_car * (H + 8);
+      const int lds = base + bufshorts * 2;
+      hipLaunchKernelGGL(lstm_bwd_mfma_kernel<false>, dim3(blocks),
+                         dim3(THREADS), lds, (hipStream_t)stream, gates_x, wt,
+                         w_row, is_init, h0, c0, ys, cs, dys, dg, hprev, dh0,
+                         dc0, gscr, B, T, H, cr_rec, cr_car);
+    }
     return;
   }
   const int blocks = (B + BW_B_TILE_LSTM - 1) / BW_B_TILE_LSTM;

@@ -860,20 +860,167 @@ class BoxList(Box):
         return len(self.boxes)
 
 
+def _specs_homogeneous(specs) -> bool:
+    first = specs[0]
+    if isinstance(first, Composite):
+        k0 = set(first.keys(True, True))
+        return all(
+            isinstance(s, Composite)
+            and set(s.keys(True, True)) == k0
+            and all(s[k].shape == first[k].shape and s[k].dtype == first[k].dtype
+                    for k in k0)
+            for s in specs[1:]
+        )
+    return all(
+        type(s) is type(first) and s.shape == first.shape and s.dtype == first.dtype
+        for s in specs[1:]
+    )
+
+
+def _merged_shape(shapes, dim, n):
+    """Per-dim merge with -1 where components disagree (the reference's
+    heterogeneous Stacked shape convention)."""
+    nd = max(len(s) for s in shapes)
+    dims = []
+    for i in range(nd):
+        vals = {s[i] if i < len(s) else None for s in shapes}
+        dims.append(vals.pop() if len(vals) == 1 and None not in vals else -1)
+    return torch.Size([*dims[:dim], n, *dims[dim:]])
+
+
+class _LazyStackedSpecBase:
+    """Shared plumbing for lazily stacked (heterogeneous) specs."""
+
+    def __init__(self, *specs, dim: int = 0):
+        if len(specs) == 1 and isinstance(specs[0], (list, tuple)):
+            specs = tuple(specs[0])
+        self._specs_list = list(specs)
+        self.dim = dim
+
+    def __len__(self):
+        return len(self._specs_list)
+
+    def __getitem__(self, idx):
+        if isinstance(idx, int):
+            return self._specs_list[idx]
+        raise KeyError(idx)
+
+    @property
+    def device(self):
+        return self._specs_list[0].device
+
+    @property
+    def shape(self):
+        return _merged_shape(
+            [tuple(s.shape) for s in self._specs_list], self.dim,
+            len(self._specs_list),
+        )
+
+    def clone(self):
+        return type(self)(*[s.clone() for s in self._specs_list], dim=self.dim)
+
+    def to(self, device):
+        return type(self)(
+            *[s.to(device) for s in self._specs_list], dim=self.dim
+        )
+
+
+class LazyStackedSpec(_LazyStackedSpecBase):
+    """Lazily stacked LEAF specs with heterogeneous shapes (reference
+    Stacked, tensor_specs.py:1496).  ``shape`` reports -1 at
+    disagreeing dims; ``rand``/``zero`` return per-component lists."""
+
+    @property
+    def dtype(self):
+        return self._specs_list[0].dtype
+
+    def rand(self, shape=None):
+        return [s.rand() for s in self._specs_list]
+
+    def zero(self, shape=None):
+        return [s.zero() for s in self._specs_list]
+
+    def is_in(self, val) -> bool:
+        vals = val if isinstance(val, (list, tuple)) else val.unbind(self.dim)
+        return all(s.is_in(v) for s, v in zip(self._specs_list, vals))
+
+
+class LazyStackedComposite(_LazyStackedSpecBase):
+    """Lazily stacked Composite specs (reference StackedComposite,
+    tensor_specs.py:6463) — the spec of a LazyStackedTensorDict, e.g.
+    heterogeneous multi-agent groups."""
+
+    def keys(self, include_nested: bool = False, leaves_only: bool = False):
+        common = None
+        for s in self._specs_list:
+            ks = set(s.keys(include_nested, leaves_only))
+            common = ks if common is None else common & ks
+        return [
+            k
+            for k in self._specs_list[0].keys(include_nested, leaves_only)
+            if k in common
+        ]
+
+    def items(self, include_nested: bool = False, leaves_only: bool = False):
+        for k in self.keys(include_nested, leaves_only):
+            yield k, self[k]
+
+    def __getitem__(self, key):
+        if isinstance(key, int):
+            return self._specs_list[key]
+        children = [s[key] for s in self._specs_list]
+        return Stacked(*children, dim=self.dim)
+
+    def __contains__(self, key):
+        try:
+            self[key]
+            return True
+        except KeyError:
+            return False
+
+    def rand(self, shape=None):
+        from ..tensordict import lazy_stack
+
+        return lazy_stack([s.rand() for s in self._specs_list], self.dim)
+
+    def zero(self, shape=None):
+        from ..tensordict import lazy_stack
+
+        return lazy_stack([s.zero() for s in self._specs_list], self.dim)
+
+    def is_in(self, val) -> bool:
+        from ..tensordict import LazyStackedTensorDict
+
+        if isinstance(val, LazyStackedTensorDict):
+            parts = val.tensordicts
+        else:
+            parts = [val[i] for i in range(len(self._specs_list))]
+        return all(s.is_in(v) for s, v in zip(self._specs_list, parts))
+
+
 class Stacked:
-    """Stacked spec (reference tensor_specs.py:1496): rl_amd stacks
-    eagerly — constructing one returns the dense stacked spec."""
+    """Stacked spec (reference tensor_specs.py:1496): homogeneous specs
+    stack EAGERLY into a dense spec; heterogeneous specs return a
+    :class:`LazyStackedSpec` / :class:`LazyStackedComposite` that keeps
+    the components (shape reports -1 at disagreeing dims)."""
 
     def __new__(cls, *specs, dim: int = 0):
-        return stack_specs(list(specs), dim)
+        if len(specs) == 1 and isinstance(specs[0], (list, tuple)):
+            specs = tuple(specs[0])
+        if _specs_homogeneous(list(specs)):
+            return stack_specs(list(specs), dim)
+        if isinstance(specs[0], Composite):
+            return LazyStackedComposite(*specs, dim=dim)
+        return LazyStackedSpec(*specs, dim=dim)
 
 
 class StackedComposite:
-    """Stacked composite spec (reference tensor_specs.py:6463); eager
-    dense stacking, same as :class:`Stacked`."""
+    """Stacked composite spec (reference tensor_specs.py:6463)."""
 
     def __new__(cls, *specs, dim: int = 0):
-        return stack_specs(list(specs), dim)
+        if _specs_homogeneous(list(specs)):
+            return stack_specs(list(specs), dim)
+        return LazyStackedComposite(*specs, dim=dim)
 
 
-__all__ += ["Stacked", "StackedComposite"]
+__all__ += ["Stacked", "StackedComposite", "LazyStackedSpec", "LazyStackedComposite"]

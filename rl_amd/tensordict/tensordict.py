@@ -153,6 +153,11 @@ class _TensorDictKeysView:
         self._leaves_only = leaves_only
 
     def _iter(self, td, prefix):
+        if not hasattr(td, "_data"):
+            # lazy-stacked child: delegate to its own keys view
+            for sub in td.keys(self._include_nested, self._leaves_only):
+                yield prefix + (sub if isinstance(sub, tuple) else (sub,)) if prefix else sub
+            return
         for k, v in td._data.items():
             is_td = isinstance(v, TensorDictBase)
             key = prefix + (k,) if prefix else k
@@ -325,6 +330,16 @@ class TensorDict(TensorDictBase):
     # get / set
     # ------------------------------------------------------------------ #
     def _validate_value(self, value, check_shape: bool = True):
+        if isinstance(value, TensorDictBase) and not hasattr(value, "_batch_size"):
+            # lazy-stacked child: batch_size is computed, not stored
+            if check_shape:
+                bs = self._batch_size
+                if value.batch_size[: len(bs)] != bs:
+                    raise RuntimeError(
+                        f"nested lazy stack batch_size {value.batch_size} "
+                        f"incompatible with parent batch_size {bs}"
+                    )
+            return value
         if isinstance(value, TensorDictBase):
             if check_shape:
                 bs = self._batch_size
@@ -413,8 +428,11 @@ class TensorDict(TensorDictBase):
             if isinstance(key, str):
                 return self._data[key]
             obj = self
-            for k in key:
-                obj = obj._data[k]
+            for i, k in enumerate(key):
+                if hasattr(obj, "_data"):
+                    obj = obj._data[k]
+                else:  # lazy-stacked child: delegate the remaining path
+                    return obj.get(key[i:], default)
             return obj
         except (KeyError, AttributeError):
             if default is _NO_DEFAULT:
@@ -1271,11 +1289,19 @@ def _common_keys(tds: Sequence[TensorDict]) -> List[NestedKey]:
 
 
 def stack(tds: Sequence[TensorDict], dim: int = 0) -> TensorDict:
-    """Stack TensorDicts along a new batch dim (eager; the reference's
-    LazyStackedTensorDict becomes an eager stack on MI355X — HBM is big)."""
+    """Stack TensorDicts along a new batch dim.
+
+    Homogeneous inputs stack EAGERLY (dense layouts feed kernels; HBM3E
+    is big).  Heterogeneous inputs — per-element shape or key-set
+    mismatches — return a :class:`~rl_amd.tensordict.LazyStackedTensorDict`
+    instead (the reference's lazy-stack behavior)."""
     tds = list(tds)
     if not tds:
         raise ValueError("cannot stack an empty sequence of TensorDicts")
+    if _is_heterogeneous(tds):
+        from .lazy import LazyStackedTensorDict
+
+        return LazyStackedTensorDict(*tds, stack_dim=dim)
     bs = tds[0].batch_size
     if dim < 0:
         dim = len(bs) + 1 + dim
@@ -1313,7 +1339,38 @@ def _merge_non_tensor(vals: Sequence[NonTensorData]) -> NonTensorData:
     return NonTensorData(datas)
 
 
-lazy_stack = stack
+def _is_heterogeneous(tds: Sequence[TensorDict]) -> bool:
+    """True when elements disagree on batch size, leaf key sets or leaf
+    shapes (then only a lazy stack can represent them)."""
+    first = tds[0]
+    bs = first.batch_size
+    keys0 = set(first.keys(True, True))
+    shapes0 = None
+    for td in tds[1:]:
+        if td.batch_size != bs:
+            return True
+        if set(td.keys(True, True)) != keys0:
+            return True
+        if shapes0 is None:
+            try:
+                shapes0 = {k: tuple(first.get(k).shape)
+                           for k in keys0
+                           if isinstance(first.get(k), torch.Tensor)}
+            except RuntimeError:
+                # a component holds a heterogeneous lazy stack itself
+                return True
+        for k, shp in shapes0.items():
+            v = td.get(k, None)
+            if isinstance(v, torch.Tensor) and tuple(v.shape) != shp:
+                return True
+    return False
+
+
+def lazy_stack(tds: Sequence[TensorDict], dim: int = 0):
+    """Always-lazy stack (reference LazyStackedTensorDict constructor)."""
+    from .lazy import LazyStackedTensorDict
+
+    return LazyStackedTensorDict(*list(tds), stack_dim=dim)
 
 
 def cat(tds: Sequence[TensorDict], dim: int = 0) -> TensorDict:

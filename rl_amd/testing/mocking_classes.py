@@ -300,3 +300,243 @@ class EnvThatErrors(CountingEnv):
         if self._n >= self.error_at:
             raise RuntimeError("EnvThatErrors: deliberate failure")
         return super()._step(tensordict)
+
+
+class MultiKeyCountingEnv(EnvBase):
+    """Counting env with MULTIPLE observation/action/reward keys, some
+    nested under an agent group (reference mocking_classes.py:1992
+    MultiKeyCountingEnv): exercises key-selection machinery in
+    collectors, step_mdp and losses."""
+
+    _supports_masked_reset = True
+
+    def __init__(self, max_steps: int = 5, batch_size=(), device=None):
+        super().__init__(device=device, batch_size=batch_size)
+        self.max_steps = max_steps
+        bs = self.batch_size
+        self.observation_spec = Composite(
+            {
+                "observation": Unbounded(shape=(*bs, 3), device=self.device),
+                "observation_orig": Unbounded(shape=(*bs, 3), device=self.device),
+                "nested_1": Composite(
+                    {"observation": Unbounded(shape=(*bs, 2), device=self.device)},
+                    shape=bs,
+                    device=self.device,
+                ),
+                "nested_2": Composite(
+                    {"observation": Unbounded(shape=(*bs, 4), device=self.device)},
+                    shape=bs,
+                    device=self.device,
+                ),
+            },
+            shape=bs,
+            device=self.device,
+        )
+        self.full_action_spec = Composite(
+            {
+                "action": Binary(shape=(*bs, 1), device=self.device, dtype=torch.bool),
+                "nested_1": Composite(
+                    {
+                        "action": Categorical(
+                            n=3, shape=(*bs,), device=self.device, dtype=torch.int64
+                        )
+                    },
+                    shape=bs,
+                    device=self.device,
+                ),
+                "nested_2": Composite(
+                    {
+                        "azione": Binary(
+                            shape=(*bs, 1), device=self.device, dtype=torch.bool
+                        )
+                    },
+                    shape=bs,
+                    device=self.device,
+                ),
+            },
+            shape=bs,
+            device=self.device,
+        )
+        self.full_reward_spec = Composite(
+            {
+                "reward": Unbounded(shape=(*bs, 1), device=self.device),
+                "nested_1": Composite(
+                    {"gift": Unbounded(shape=(*bs, 1), device=self.device)},
+                    shape=bs,
+                    device=self.device,
+                ),
+            },
+            shape=bs,
+            device=self.device,
+        )
+        self.count = torch.zeros((*bs, 1), dtype=torch.float32, device=self.device)
+
+    def _obs_td(self):
+        bs = self.batch_size
+        c = self.count
+        return TensorDict(
+            {
+                "observation": c.expand(*bs, 3).clone(),
+                "observation_orig": c.expand(*bs, 3).clone(),
+                ("nested_1", "observation"): c.expand(*bs, 2).clone(),
+                ("nested_2", "observation"): c.expand(*bs, 4).clone(),
+            },
+            batch_size=bs,
+            device=self.device,
+        )
+
+    def _reset(self, tensordict=None, **kwargs):
+        if tensordict is not None and "_reset" in tensordict:
+            mask = tensordict.get("_reset").reshape(self.count.shape)
+            self.count = torch.where(mask, torch.zeros_like(self.count), self.count)
+        else:
+            self.count.fill_(0.0)
+        td = self._obs_td()
+        td.set("done", torch.zeros_like(self.count, dtype=torch.bool))
+        td.set("terminated", torch.zeros_like(self.count, dtype=torch.bool))
+        return td
+
+    def _step(self, tensordict):
+        self.count = self.count + 1.0
+        done = self.count >= self.max_steps
+        td = self._obs_td()
+        td.set("reward", torch.ones_like(self.count))
+        td.set(("nested_1", "gift"), torch.full_like(self.count, 2.0))
+        td.set("done", done)
+        td.set("terminated", done)
+        return td
+
+    def _set_seed(self, seed):
+        return seed
+
+
+class HeterogeneousCountingEnv(EnvBase):
+    """Multi-agent counting env whose agents have DIFFERENT observation
+    widths — per-step data is a LazyStackedTensorDict under "agents"
+    and the spec a LazyStackedComposite (reference
+    mocking_classes.py:1787 HeterogeneousCountingEnv)."""
+
+    def __init__(self, n_agents: int = 3, max_steps: int = 5, device=None):
+        super().__init__(device=device, batch_size=())
+        from ..data.tensor_specs import Stacked
+
+        self.n_agents = n_agents
+        self.max_steps = max_steps
+        agent_specs = [
+            Composite(
+                {
+                    "observation": Unbounded(shape=(i + 1,), device=self.device),
+                },
+                shape=(),
+                device=self.device,
+            )
+            for i in range(n_agents)
+        ]
+        self.observation_spec = Composite(
+            {},
+            shape=(),
+            device=self.device,
+        )
+        self.observation_spec.set("agents", Stacked(*agent_specs, dim=0))
+        self.action_spec = Bounded(
+            low=-1.0, high=1.0, shape=(n_agents,), device=self.device
+        )
+        self.reward_spec = Unbounded(shape=(1,), device=self.device)
+        self.count = 0.0
+
+    def _agents_td(self):
+        from ..tensordict import lazy_stack
+
+        return lazy_stack(
+            [
+                TensorDict(
+                    {
+                        "observation": torch.full(
+                            (i + 1,), self.count, device=self.device
+                        ),
+                    },
+                    batch_size=(),
+                    device=self.device,
+                )
+                for i in range(self.n_agents)
+            ],
+            0,
+        )
+
+    def _reset(self, tensordict=None, **kwargs):
+        self.count = 0.0
+        td = TensorDict({}, batch_size=(), device=self.device)
+        td.set("agents", self._agents_td())
+        td.set("done", torch.zeros(1, dtype=torch.bool, device=self.device))
+        td.set("terminated", torch.zeros(1, dtype=torch.bool, device=self.device))
+        return td
+
+    def _step(self, tensordict):
+        self.count += 1.0
+        done = torch.full(
+            (1,), self.count >= self.max_steps, dtype=torch.bool, device=self.device
+        )
+        td = TensorDict({}, batch_size=(), device=self.device)
+        td.set("agents", self._agents_td())
+        td.set("reward", torch.ones(1, device=self.device))
+        td.set("done", done)
+        td.set("terminated", done)
+        return td
+
+    def _set_seed(self, seed):
+        return seed
+
+
+class EnvWithDynamicSpec(EnvBase):
+    """Observation shape GROWS each step (reference
+    mocking_classes.py:2307): rollouts cannot densify and must come back
+    as LazyStackedTensorDicts (``return_contiguous=False``)."""
+
+    def __init__(self, max_steps: int = 4, device=None):
+        super().__init__(device=device, batch_size=())
+        self.max_steps = max_steps
+        self.observation_spec = Composite(
+            {"observation": Unbounded(shape=(1, 2), device=self.device)},
+            shape=(),
+            device=self.device,
+        )
+        self.action_spec = Bounded(low=-1.0, high=1.0, shape=(2,), device=self.device)
+        self.reward_spec = Unbounded(shape=(1,), device=self.device)
+        self._t = 0
+
+    def _obs(self):
+        return torch.ones((self._t + 1, 2), device=self.device) * self._t
+
+    def _reset(self, tensordict=None, **kwargs):
+        self._t = 0
+        return TensorDict(
+            {
+                "observation": self._obs(),
+                "done": torch.zeros(1, dtype=torch.bool, device=self.device),
+                "terminated": torch.zeros(1, dtype=torch.bool, device=self.device),
+            },
+            batch_size=(),
+            device=self.device,
+        )
+
+    def _step(self, tensordict):
+        self._t += 1
+        done = torch.full(
+            (1,), self._t >= self.max_steps, dtype=torch.bool, device=self.device
+        )
+        return TensorDict(
+            {
+                "observation": self._obs(),
+                "reward": torch.ones(1, device=self.device),
+                "done": done,
+                "terminated": done,
+            },
+            batch_size=(),
+            device=self.device,
+        )
+
+    def _set_seed(self, seed):
+        return seed
+
+
+__all__ += ["MultiKeyCountingEnv", "HeterogeneousCountingEnv", "EnvWithDynamicSpec"]
