@@ -67,6 +67,11 @@ from .datasets import (
     LocalMemmapExperienceReplay,
     MinariExperienceReplay,
     OpenXExperienceReplay,
+    StreamingEpisodeStorage,
+    convert_atari_shards,
+    convert_d4rl_hdf5,
+    convert_minari_hdf5,
+    download_file,
 )
 from .replay_buffers import (
     CompressedListStorage,
