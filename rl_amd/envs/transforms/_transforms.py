@@ -396,6 +396,15 @@ class ObservationNorm(Transform):
             return obs * scale + loc
         return (obs - loc) / scale
 
+    def transform_observation_spec(self, spec):
+        from ...data.tensor_specs import Unbounded as _U
+
+        for in_key, out_key in zip(self.in_keys, self.out_keys):
+            if in_key in spec:
+                s = spec[in_key]
+                spec[out_key] = _U(shape=s.shape, dtype=s.dtype, device=s.device)
+        return spec
+
 
 class CatFrames(Transform):
     """Stack the last N frames along ``dim`` (reference _observation.py:867).
@@ -660,6 +669,19 @@ class RewardClipping(Transform):
     def _apply_transform(self, r):
         return r.clamp(self.clamp_min, self.clamp_max)
 
+    def transform_reward_spec(self, spec):
+        from ...data.tensor_specs import Bounded as _B
+
+        if self.clamp_min is not None and self.clamp_max is not None:
+            for in_key, out_key in zip(self.in_keys, self.out_keys):
+                if in_key in spec:
+                    s = spec[in_key]
+                    spec[out_key] = _B(
+                        low=self.clamp_min, high=self.clamp_max,
+                        shape=s.shape, dtype=s.dtype, device=s.device,
+                    )
+        return spec
+
 
 class RewardScaling(Transform):
     """reward ← reward * scale + loc (reference _normalization.py:366)."""
@@ -674,6 +696,15 @@ class RewardScaling(Transform):
         if self.standard_normal:
             return (r - self.loc) / self.scale
         return r * self.scale + self.loc
+
+    def transform_reward_spec(self, spec):
+        from ...data.tensor_specs import Unbounded as _U
+
+        for in_key, out_key in zip(self.in_keys, self.out_keys):
+            if in_key in spec:
+                s = spec[in_key]
+                spec[out_key] = _U(shape=s.shape, dtype=s.dtype, device=s.device)
+        return spec
 
 
 class RewardSum(Transform):
@@ -719,6 +750,16 @@ class BinarizeReward(Transform):
     def _apply_transform(self, r):
         return (r > 0).to(r.dtype)
 
+    def transform_reward_spec(self, spec):
+        from ...data.tensor_specs import Bounded as _B
+
+        for in_key, out_key in zip(self.in_keys, self.out_keys):
+            if in_key in spec:
+                s = spec[in_key]
+                spec[out_key] = _B(low=0.0, high=1.0, shape=s.shape,
+                                   dtype=s.dtype, device=s.device)
+        return spec
+
 
 class SignTransform(Transform):
     def __init__(self, in_keys=("reward",), out_keys=None):
@@ -726,6 +767,16 @@ class SignTransform(Transform):
 
     def _apply_transform(self, r):
         return r.sign()
+
+    def transform_reward_spec(self, spec):
+        from ...data.tensor_specs import Bounded as _B
+
+        for in_key, out_key in zip(self.in_keys, self.out_keys):
+            if in_key in spec:
+                s = spec[in_key]
+                spec[out_key] = _B(low=-1.0, high=1.0, shape=s.shape,
+                                   dtype=s.dtype, device=s.device)
+        return spec
 
 
 class TargetReturn(Transform):
@@ -954,9 +1005,20 @@ class DTypeCastTransform(Transform):
     def _inv_call(self, td):
         if self.in_keys_inv:
             return super()._inv_call(td)
+        # auto mode: with a parent env, only cast back the inputs the
+        # BASE env actually declares as dtype_in (e.g. float64 MuJoCo
+        # actions) — a float32-native env must keep float32 actions
+        base_in = None
+        parent = self.parent
+        if parent is not None and hasattr(parent, "base_env"):
+            base_in = parent.base_env.full_action_spec
         for k in list(td.keys(True, True)):
             v = td.get(k)
             if isinstance(v, torch.Tensor) and v.dtype == self.dtype_out:
+                if base_in is not None:
+                    sp = base_in.get(k, None) if hasattr(base_in, "get") else None
+                    if sp is None or sp.dtype != self.dtype_in:
+                        continue
                 td.set(k, v.to(self.dtype_in))
         return td
 
@@ -1050,6 +1112,24 @@ class ClipTransform(Transform):
 
     def _apply_transform(self, x):
         return x.clamp(self.low, self.high)
+
+    def _clip_spec(self, spec):
+        from ...data.tensor_specs import Bounded as _B
+
+        if self.low is None or self.high is None:
+            return spec
+        for in_key, out_key in zip(self.in_keys, self.out_keys):
+            if in_key in spec:
+                s = spec[in_key]
+                spec[out_key] = _B(low=self.low, high=self.high,
+                                   shape=s.shape, dtype=s.dtype, device=s.device)
+        return spec
+
+    def transform_observation_spec(self, spec):
+        return self._clip_spec(spec)
+
+    def transform_reward_spec(self, spec):
+        return self._clip_spec(spec)
 
 
 class ActionMask(Transform):

@@ -105,5 +105,15 @@ class VecNorm(Transform):
         var = (ssq / count.clamp_min(1.0) - mean * mean).clamp_min(0.0)
         return mean, var.sqrt() + self.eps
 
+    def transform_observation_spec(self, spec):
+        from ..._utils import logger  # noqa: F401  (import check)
+        from ...data.tensor_specs import Unbounded as _U
+
+        for in_key, out_key in zip(self.in_keys, self.out_keys):
+            if in_key in spec:
+                sp = spec[in_key]
+                spec[out_key] = _U(shape=sp.shape, dtype=sp.dtype, device=sp.device)
+        return spec
+
 
 VecNormV2 = VecNorm

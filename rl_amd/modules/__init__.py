@@ -52,7 +52,12 @@ from .tensordict_module import (
 from .models import Mixer, MultiAgentConvNet, MultiAgentMLP, QMixer, VDNMixer
 from .models import ConsistentDropout, NoisyLazyLinear, NoisyLinear, gSDEModule, reset_noise
 from .planners import CEMPlanner, MPCPlannerBase, MPPIPlanner
-from .inference_server import InferenceServer, PolicyClient
+from .inference_server import (
+    InferenceServer,
+    PolicyClient,
+    ProcessInferenceServer,
+    SlotPolicyClient,
+)
 from .llm import LLMWrapperBase, TransformersWrapper
 from .models import (
     DTActor,

@@ -253,3 +253,79 @@ class TestReferenceAllParity:
             if gone:
                 missing[sp or "root"] = gone
         assert not missing, missing
+
+
+# ---------------------------------------------------------------------- #
+# ProcessInferenceServer: shared-memory slots, own server process
+# (reference inference_server/_server.py:961)
+# ---------------------------------------------------------------------- #
+def _linear_policy_factory():
+    import torch
+
+    from rl_amd.tensordict import TensorDictModule
+
+    torch.manual_seed(7)
+    return TensorDictModule(
+        torch.nn.Linear(4, 2), in_keys=["observation"], out_keys=["action"]
+    )
+
+
+@pytest.mark.timeout(120)
+def test_process_inference_server_slots():
+    import torch
+
+    from rl_amd.modules import ProcessInferenceServer
+    from rl_amd.tensordict import TensorDict
+
+    req = TensorDict({"observation": torch.zeros(4)}, batch_size=[])
+    resp = TensorDict({"action": torch.zeros(2)}, batch_size=[])
+    server = ProcessInferenceServer(
+        _linear_policy_factory, req, resp, n_slots=4
+    )
+    with server:
+        clients = [server.make_client(timeout=60.0) for _ in range(3)]
+        torch.manual_seed(7)
+        ref_policy = torch.nn.Linear(4, 2)
+        outs = []
+        for i, c in enumerate(clients):
+            td = TensorDict({"observation": torch.full((4,), float(i))}, batch_size=[])
+            outs.append(c(td).get("action"))
+        with torch.no_grad():
+            for i, a in enumerate(outs):
+                expect = ref_policy(torch.full((1, 4), float(i)))[0]
+                assert torch.allclose(a, expect, atol=1e-5), i
+        stats = server.stats
+    assert stats["requests"] == 3
+
+
+@pytest.mark.timeout(120)
+def test_process_inference_server_batches_concurrent():
+    import threading
+
+    import torch
+
+    from rl_amd.modules import ProcessInferenceServer
+    from rl_amd.tensordict import TensorDict
+
+    req = TensorDict({"observation": torch.zeros(4)}, batch_size=[])
+    resp = TensorDict({"action": torch.zeros(2)}, batch_size=[])
+    server = ProcessInferenceServer(_linear_policy_factory, req, resp, n_slots=8)
+    with server:
+        clients = [server.make_client(timeout=60.0) for _ in range(6)]
+        results = [None] * 6
+
+        def run(i):
+            for _ in range(5):
+                td = TensorDict(
+                    {"observation": torch.full((4,), float(i))}, batch_size=[]
+                )
+                results[i] = clients[i](td).get("action").clone()
+
+        threads = [threading.Thread(target=run, args=(i,)) for i in range(6)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        stats = server.stats
+    assert stats["requests"] == 30
+    assert all(r is not None for r in results)
