@@ -90,10 +90,11 @@ class RemoteModuleWeightUpdater(MultiProcessedWeightUpdater):
 
 
 class RayWeightUpdater(WeightUpdaterBase):
-    """Ray-actor weight pushes (reference weight_update.py RayWeightUpdater)
-    — gated: ray is not installed in this image."""
+    """Ray-actor weight pushes (reference weight_update.py
+    RayWeightUpdater): broadcasts the learner state through the object
+    store to every remote collector actor — gated on `ray`."""
 
-    def __init__(self, *args, **kwargs):
+    def __init__(self, policy, remote_collectors, max_interval: int = 0):
         import importlib.util
 
         if importlib.util.find_spec("ray") is None:
@@ -102,7 +103,23 @@ class RayWeightUpdater(WeightUpdaterBase):
                 "installed in this image. Use MultiProcessedWeightUpdater or "
                 "the RCCL DistributedWeightSyncScheme instead."
             )
-        raise NotImplementedError("ray backend scaffolding")
+        self.policy = policy
+        self.remote_collectors = list(remote_collectors)
+        self.max_interval = max_interval
+        self._updates = 0
+
+    def update_weights(self, weights=None) -> None:
+        import ray
+
+        if weights is None:
+            weights = {
+                k: v.detach().cpu() for k, v in self.policy.state_dict().items()
+            }
+        ref = ray.put(weights)
+        ray.get(
+            [w.update_policy_weights_.remote(ref) for w in self.remote_collectors]
+        )
+        self._updates += 1
 
 
 class DistributedWeightUpdater(WeightUpdaterBase):

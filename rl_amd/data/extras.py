@@ -582,15 +582,83 @@ def clear_video_decoder_cache() -> None:
         cache.clear()
 
 
-class RayReplayBuffer:
-    """Ray-actor-hosted remote replay buffer (reference ray_buffer.py)
-    — gated: ray is not installed in this image."""
+class _RayBufferActor:
+    """Actor body hosting the real buffer (decorated at runtime)."""
 
-    def __init__(self, *args, **kwargs):
+    def __init__(self, buffer_factory):
+        self.rb = buffer_factory()
+
+    def add(self, data):
+        return self.rb.add(data)
+
+    def extend(self, data):
+        return self.rb.extend(data)
+
+    def sample(self, batch_size=None):
+        return self.rb.sample(batch_size)
+
+    def update_priority(self, index, priority):
+        if hasattr(self.rb, "update_priority"):
+            self.rb.update_priority(index, priority)
+
+    def __len__(self):
+        return len(self.rb)
+
+    def length(self):
+        return len(self.rb)
+
+
+class RayReplayBuffer:
+    """Ray-actor-hosted remote replay buffer (reference ray_buffer.py):
+    collectors on any node ``extend`` through the object store while the
+    learner ``sample``s — gated on `ray` (not in this image; use
+    ReplayBufferService / RemoteTensorDictReplayBuffer offline)."""
+
+    def __init__(self, buffer_factory=None, *, remote_configs=None, **kwargs):
         if importlib.util.find_spec("ray") is None:
             raise ImportError(
                 "RayReplayBuffer requires the `ray` package, which is not "
                 "installed in this image. Use RemoteTensorDictReplayBuffer "
                 "(torch.distributed.rpc) instead."
             )
-        raise NotImplementedError("ray actor scaffolding")
+        import ray
+
+        if not ray.is_initialized():
+            ray.init(ignore_reinit_error=True)
+        if buffer_factory is None:
+            from .replay_buffers.buffers import TensorDictReplayBuffer
+            from .replay_buffers.storages import LazyTensorStorage
+
+            def buffer_factory():
+                return TensorDictReplayBuffer(
+                    storage=LazyTensorStorage(kwargs.pop("capacity", 100_000)),
+                    **kwargs,
+                )
+
+        Actor = ray.remote(**(remote_configs or {"num_cpus": 1}))(_RayBufferActor)
+        self._actor = Actor.remote(buffer_factory)
+
+    def add(self, data):
+        import ray
+
+        return ray.get(self._actor.add.remote(data))
+
+    def extend(self, data):
+        import ray
+
+        return ray.get(self._actor.extend.remote(data))
+
+    def sample(self, batch_size=None):
+        import ray
+
+        return ray.get(self._actor.sample.remote(batch_size))
+
+    def update_priority(self, index, priority):
+        import ray
+
+        ray.get(self._actor.update_priority.remote(index, priority))
+
+    def __len__(self):
+        import ray
+
+        return ray.get(self._actor.length.remote())

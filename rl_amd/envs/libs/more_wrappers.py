@@ -221,30 +221,33 @@ class BraxWrapper(BraxEnv):
     """Wrap an existing brax env instance (reference libs/brax.py)."""
 
 
-class VmasEnv(EnvBase):
-    """VMAS vectorized multi-agent envs (reference libs/vmas.py) —
-    gated on `vmas`; grouping follows MarlGroupMapType."""
-
-    def __init__(self, scenario, *, num_envs: int = 1, device=None, **kwargs):
-        _require("vmas", type(self).__name__)
-        raise NotImplementedError("vmas scaffolding")
+from .misc_wrappers import PettingZooWrapper, VmasWrapper  # noqa: F401 (re-export)
 
 
-class VmasWrapper(VmasEnv):
-    """Wrap an existing vmas env instance (reference libs/vmas.py)."""
+def VmasEnv(scenario, *, num_envs: int = 1, device=None, **kwargs):
+    """Build a VMAS env by scenario name and wrap it (reference
+    libs/vmas.py VmasEnv; the wrap-an-instance class is
+    :class:`~rl_amd.envs.libs.misc_wrappers.VmasWrapper`)."""
+    vmas = _require("vmas", "VmasEnv")
+    from .misc_wrappers import VmasWrapper as _W
+
+    env = vmas.make_env(
+        scenario=scenario, num_envs=num_envs, device=device or "cpu", **kwargs
+    )
+    return _W(env, device=device)
 
 
-class PettingZooEnv(EnvBase):
-    """PettingZoo multi-agent envs (reference libs/pettingzoo.py) —
-    gated on `pettingzoo`; grouping follows MarlGroupMapType."""
+def PettingZooEnv(task: str = None, *, parallel: bool = True, device=None, **kwargs):
+    """Build a PettingZoo env by dotted task name (e.g.
+    ``"mpe.simple_spread_v3"``) and wrap it (reference
+    libs/pettingzoo.py PettingZooEnv; the wrap-an-instance class is
+    :class:`~rl_amd.envs.libs.misc_wrappers.PettingZooWrapper`)."""
+    _require("pettingzoo", "PettingZooEnv")
+    from .misc_wrappers import PettingZooWrapper as _W
 
-    def __init__(self, task: str = None, *, parallel: bool = True, device=None, **kwargs):
-        _require("pettingzoo", type(self).__name__)
-        raise NotImplementedError("pettingzoo scaffolding")
-
-
-class PettingZooWrapper(PettingZooEnv):
-    """Wrap an existing PettingZoo env instance."""
+    mod = importlib.import_module(f"pettingzoo.{task}")
+    env = mod.parallel_env(**kwargs) if parallel else mod.env(**kwargs)
+    return _W(env, device=device)
 
 
 class MOGymEnv(EnvBase):

@@ -180,5 +180,48 @@ VIPTransform = _gated("VIPTransform", "torchvision", " (and VIP weights need net
 VIPRewardTransform = _gated("VIPRewardTransform", "torchvision")
 VC1Transform = _gated("VC1Transform", "torchvision", " (and VC-1 weights need network access)")
 DecodeVideoTransform = _gated("DecodeVideoTransform", "torchvision")
-RayModuleTransform = _gated("RayModuleTransform", "ray")
-RayTransform = _gated("RayTransform", "ray")
+
+
+class _RayModuleActor:
+    def __init__(self, module_factory):
+        self.module = module_factory()
+
+    def __call__(self, td):
+        import torch
+
+        with torch.no_grad():
+            return self.module(td)
+
+
+class RayTransform(Transform):
+    """Run a module remotely in a Ray actor on the forward path
+    (reference transforms/module.py:123 / ray_service.py:130): heavy
+    encoders live once per cluster instead of once per worker — gated
+    on `ray`."""
+
+    def __init__(self, module_factory, *, in_keys=None, out_keys=None,
+                 remote_configs=None):
+        if importlib.util.find_spec("ray") is None:
+            raise ImportError(
+                "RayTransform requires the `ray` package, which is not "
+                "installed in this image."
+            )
+        import ray
+
+        super().__init__(in_keys=in_keys, out_keys=out_keys)
+        if not ray.is_initialized():
+            ray.init(ignore_reinit_error=True)
+        Actor = ray.remote(**(remote_configs or {"num_cpus": 1}))(_RayModuleActor)
+        self._actor = Actor.remote(module_factory)
+
+    def _call(self, td):
+        import ray
+
+        out = ray.get(self._actor.__call__.remote(td))
+        td.update(out)
+        return td
+
+
+class RayModuleTransform(RayTransform):
+    """Alias flavor of :class:`RayTransform` (reference
+    transforms/module.py:26)."""

@@ -260,12 +260,29 @@ class ProcessLogger(Logger):
         self._proc.join(timeout=10)
 
 
-class RayLogger(Logger):
-    """Ray-actor-hosted logger (reference record/loggers/ray.py) —
-    gated: ray is not installed in this image; ProcessLogger covers the
-    out-of-process pattern locally."""
+class _RayLoggerActor:
+    """Actor body hosting the real logger (decorated at runtime)."""
 
-    def __init__(self, *args, **kwargs):
+    def __init__(self, logger_factory):
+        self.logger = logger_factory()
+
+    def log_scalar(self, name, value, step=None):
+        self.logger.log_scalar(name, value, step=step)
+
+    def log_hparams(self, cfg):
+        self.logger.log_hparams(cfg)
+
+    def log_video(self, name, video, step=None, **kw):
+        self.logger.log_video(name, video, step=step, **kw)
+
+
+class RayLogger(Logger):
+    """Ray-actor-hosted logger (reference record/loggers/ray.py): every
+    worker logs through one actor so multi-node runs share a single
+    experiment sink — gated on `ray` (ProcessLogger covers the
+    out-of-process pattern locally)."""
+
+    def __init__(self, logger_factory, *, remote_configs=None):
         import importlib.util
 
         if importlib.util.find_spec("ray") is None:
@@ -273,7 +290,22 @@ class RayLogger(Logger):
                 "RayLogger requires the `ray` package, which is not installed "
                 "in this image. Use ProcessLogger instead."
             )
-        raise NotImplementedError("ray actor scaffolding")
+        import ray
+
+        if not ray.is_initialized():
+            ray.init(ignore_reinit_error=True)
+        Actor = ray.remote(**(remote_configs or {"num_cpus": 0.1}))(_RayLoggerActor)
+        self._actor = Actor.remote(logger_factory)
+        self.exp_name = "ray"
+
+    def log_scalar(self, name, value, step=None):
+        self._actor.log_scalar.remote(name, value, step)
+
+    def log_hparams(self, cfg):
+        self._actor.log_hparams.remote(cfg)
+
+    def log_video(self, name, video, step=None, **kw):
+        self._actor.log_video.remote(name, video, step, **kw)
 
 
 __all__ += ["ProcessLogger", "RayLogger"]

@@ -343,7 +343,7 @@ class TestBatch4:
             R3MTransform("resnet18")
         if importlib.util.find_spec("ray") is None:
             with pytest.raises(ImportError, match="ray"):
-                RayModuleTransform()
+                RayModuleTransform(lambda: None)
 
     def test_aliases(self):
         from rl_amd.envs import transforms as T
