@@ -36,10 +36,28 @@ class BaseCollector:
     def iterator(self):
         raise NotImplementedError
 
-    def update_policy_weights_(self, policy_or_weights=None, **kwargs) -> None:
+    def register_weight_sync_scheme(self, model_id: str, scheme, model=None) -> None:
+        """Register a WeightSyncScheme for ``model_id`` (reference
+        per-model_id scheme registry, weight_sync_schemes.py:346); its
+        sender is driven by ``update_policy_weights_(model_id=...)``."""
+        if not hasattr(self, "_weight_senders"):
+            self._weight_senders = {}
+        target = model if model is not None else getattr(self, "policy", None)
+        self._weight_senders[model_id] = scheme.create_sender(model_id, target)
+
+    def update_policy_weights_(self, policy_or_weights=None, *, model_id=None, **kwargs) -> None:
         """Push new policy weights into the collector (reference
-        _base.py:924).  Single-process: copy state-dict into the local
-        policy."""
+        _base.py:924).  With registered schemes, route through the
+        model_id's sender; single-process default copies the state-dict
+        into the local policy."""
+        senders = getattr(self, "_weight_senders", None)
+        if senders:
+            if model_id is not None:
+                senders[model_id].send(policy_or_weights)
+            else:
+                for sender in senders.values():
+                    sender.send(policy_or_weights)
+            return
         if policy_or_weights is None:
             return
         if isinstance(policy_or_weights, dict):

@@ -54,8 +54,42 @@ class WeightStrategy:
             raise TypeError(f"cannot apply weights of type {type(weights)}")
 
 
+class WeightSender:
+    """Learner-side handle bound to one (scheme, model_id) pair
+    (reference weight_sync_schemes.py sender half)."""
+
+    def __init__(self, scheme: "WeightSyncScheme", model_id: str, model=None):
+        self.scheme = scheme
+        self.model_id = model_id
+        if model is not None:
+            scheme.connect(model)
+
+    def send(self, weights=None) -> None:
+        self.scheme.send(weights)
+
+    def __repr__(self):
+        return f"WeightSender({type(self.scheme).__name__}, {self.model_id!r})"
+
+
+class WeightReceiver:
+    """Worker-side handle bound to one (scheme, model_id) pair."""
+
+    def __init__(self, scheme: "WeightSyncScheme", model_id: str, model):
+        self.scheme = scheme
+        self.model_id = model_id
+        self.model = model
+
+    def receive(self) -> bool:
+        return self.scheme.receive(self.model)
+
+    def __repr__(self):
+        return f"WeightReceiver({type(self.scheme).__name__}, {self.model_id!r})"
+
+
 class WeightSyncScheme:
-    """ABC (reference :346): create sender/receiver pairs per model_id."""
+    """ABC (reference :346): a scheme pairs a transport with a weight
+    format; collectors register one per model_id and talk to it through
+    :class:`WeightSender` / :class:`WeightReceiver` handles."""
 
     def __init__(self, strategy: str = "tensordict"):
         self.strategy = WeightStrategy(strategy)
@@ -63,6 +97,13 @@ class WeightSyncScheme:
     def connect(self, model) -> None:
         """Learner-side registration."""
         self.model = model
+
+    # -- sender/receiver factory (reference per-model_id matrix) -------- #
+    def create_sender(self, model_id: str = "policy", model=None) -> WeightSender:
+        return WeightSender(self, model_id, model)
+
+    def create_receiver(self, model, model_id: str = "policy") -> WeightReceiver:
+        return WeightReceiver(self, model_id, model)
 
     def send(self, weights=None) -> None:
         raise NotImplementedError
@@ -198,3 +239,56 @@ class DistributedWeightSyncScheme(WeightSyncScheme):
             # extract_as == tensordict (from_module shares storage)
             pass
         return True
+
+
+# ---------------------------------------------------------------------- #
+# RPC transport (reference weight_update/_rpc.py:19): weights pushed by
+# torch.distributed.rpc calls into a per-worker module registry.
+# ---------------------------------------------------------------------- #
+_RPC_MODEL_REGISTRY: Dict[str, Any] = {}
+
+
+def rpc_register_model(model_id: str, model) -> None:
+    """Worker-side: register the module RPC weight pushes should hit."""
+    _RPC_MODEL_REGISTRY[model_id] = model
+
+
+def _rpc_apply_weights(model_id: str, state_dict: dict) -> bool:
+    model = _RPC_MODEL_REGISTRY.get(model_id)
+    if model is None:
+        return False
+    model.load_state_dict(state_dict)
+    return True
+
+
+class RPCWeightSyncScheme(WeightSyncScheme):
+    """torch.distributed.rpc (TensorPipe) transport: the sender issues
+    one ``rpc_sync`` per worker applying the state dict into that
+    worker's registered model (reference _rpc.py:19,143)."""
+
+    def __init__(self, worker_names: Sequence[str], model_id: str = "policy"):
+        super().__init__("state_dict")
+        self.worker_names = list(worker_names)
+        self.model_id = model_id
+
+    def send(self, weights=None) -> None:
+        from torch.distributed import rpc
+
+        sd = weights if isinstance(weights, dict) else {
+            k: v.detach().cpu()
+            for k, v in (weights or self.model).state_dict().items()
+        }
+        futs = [
+            rpc.rpc_async(name, _rpc_apply_weights, args=(self.model_id, sd))
+            for name in self.worker_names
+        ]
+        for f in futs:
+            f.wait()
+
+    def receive(self, model) -> bool:
+        # worker side is push-based: register once, RPC calls do the rest
+        rpc_register_model(self.model_id, model)
+        return False
+
+
+__all__ += ["WeightSender", "WeightReceiver", "RPCWeightSyncScheme", "rpc_register_model"]
