@@ -55,6 +55,9 @@ def parse_args(argv=None):
     p.add_argument("--epochs", type=int, default=1)
     p.add_argument("--hidden", type=int, default=64)
     p.add_argument("--device", type=str, default=None)
+    p.add_argument("--backend", type=str, default=None,
+                   help="process-group backend override (e.g. gloo for a "
+                        "2-rank smoke sharing one GPU)")
     p.add_argument("--seed", type=int, default=1234)
     p.add_argument("--same-seed", action="store_true",
                    help="identical seed on every rank (numerics tests: "
@@ -127,7 +130,8 @@ def run_ppo(args):
 
     distributed = world > 1
     if distributed and not torch.distributed.is_initialized():
-        torch.distributed.init_process_group(backend="nccl" if cuda else "gloo")
+        backend = args.backend or ("nccl" if cuda else "gloo")
+        torch.distributed.init_process_group(backend=backend)
 
     seed = args.seed if args.same_seed else args.seed + rank
     torch.manual_seed(seed)
