@@ -80,11 +80,19 @@ class CSVLogger(Logger):
         with open(path, "a", newline="") as f:
             csv.writer(f).writerow([step if step is not None else "", value])
 
-    def log_video(self, name: str, video, step: Optional[int] = None, **kwargs) -> None:
+    def log_video(self, name: str, video, step: Optional[int] = None, video_format: str = "gif", **kwargs) -> None:
         safe = name.replace("/", ".")
         suffix = f"_{step}" if step is not None else ""
-        path = os.path.join(self.experiment_dir, "videos", f"{safe}{suffix}.pt")
-        torch.save(video, path)
+        base = os.path.join(self.experiment_dir, "videos", f"{safe}{suffix}")
+        if video_format == "gif":
+            from ...render.video import write_gif
+
+            v = video
+            if hasattr(v, "dim") and v.dim() == 5:  # [B, T, C, H, W]
+                v = v[0]
+            write_gif(v, base + ".gif", fps=int(kwargs.get("fps", 30)))
+        else:
+            torch.save(video, base + ".pt")
 
     def log_hparams(self, cfg: Dict[str, Any]) -> None:
         path = os.path.join(self.experiment_dir, "texts", "hparams.txt")

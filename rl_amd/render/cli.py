@@ -42,6 +42,8 @@ def render_rollout(
     steps: int = 200,
     deterministic: bool = True,
     out_path: Optional[str] = None,
+    video_path: Optional[str] = None,
+    pixel_key: str = "pixels",
 ) -> TensorDictBase:
     etype = (
         ExplorationType.DETERMINISTIC if deterministic else ExplorationType.RANDOM
@@ -50,6 +52,18 @@ def render_rollout(
         rollout = env.rollout(steps, policy=policy, break_when_any_done=True)
     if out_path is not None:
         torch.save(rollout, out_path)
+    if video_path is not None:
+        frames = rollout.get(pixel_key, None)
+        if frames is None:
+            frames = rollout.get(("next", pixel_key), None)
+        if frames is None:
+            raise KeyError(
+                f"rollout has no {pixel_key!r} frames — add a pixel transform "
+                "or PixelRenderTransform to the env"
+            )
+        from .video import write_gif
+
+        write_gif(frames.reshape(-1, *frames.shape[-3:]), video_path)
     return rollout
 
 
@@ -61,6 +75,8 @@ def main(argv=None) -> int:
     parser.add_argument("--steps", type=int, default=200)
     parser.add_argument("--episodes", type=int, default=1)
     parser.add_argument("--out", type=str, default=None)
+    parser.add_argument("--video", type=str, default=None,
+                        help="write an animated GIF of the pixel frames")
     parser.add_argument("--stochastic", action="store_true")
     args = parser.parse_args(argv)
 
@@ -84,6 +100,9 @@ def main(argv=None) -> int:
             deterministic=not args.stochastic,
             out_path=(
                 f"{args.out}_ep{ep}.pt" if args.out and args.episodes > 1 else args.out
+            ),
+            video_path=(
+                f"{args.video}_ep{ep}.gif" if args.video and args.episodes > 1 else args.video
             ),
         )
         r = float(rollout.get(("next", "reward")).sum())

@@ -329,3 +329,25 @@ def test_process_inference_server_batches_concurrent():
         stats = server.stats
     assert stats["requests"] == 30
     assert all(r is not None for r in results)
+
+
+def test_gif_video_pipeline(tmp_path):
+    """VideoRecorder → CSVLogger → animated GIF artifact (render/video.py
+    GIF89a encoder, no video deps)."""
+    from rl_amd.record import CSVLogger, VideoRecorder
+    from rl_amd.render import write_gif
+
+    logger = CSVLogger("vid", log_dir=str(tmp_path))
+    rec = VideoRecorder(logger, in_keys=["pixels"], skip=1)
+    for t in range(5):
+        td = TensorDict({"pixels": torch.rand(3, 16, 16)}, batch_size=[])
+        rec._call(td)
+    rec.dump()
+    vids = list((tmp_path / "vid").glob("**/videos/*.gif"))
+    assert len(vids) == 1
+    data = vids[0].read_bytes()
+    assert data[:6] == b"GIF89a" and data.endswith(b"\x3b")
+    # direct writer: HWC input, float range
+    p = str(tmp_path / "direct.gif")
+    write_gif(torch.rand(4, 8, 8, 3), p, fps=5)
+    assert open(p, "rb").read()[:6] == b"GIF89a"
