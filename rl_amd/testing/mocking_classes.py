@@ -540,3 +540,271 @@ class EnvWithDynamicSpec(EnvBase):
 
 
 __all__ += ["MultiKeyCountingEnv", "HeterogeneousCountingEnv", "EnvWithDynamicSpec"]
+
+
+class StatelessCountingEnv(EnvBase):
+    """Counter carried IN the tensordict (no hidden env state) —
+    exercises state-in-data plumbing (reference stateless mocks;
+    also the case GraphedRollout's entry snapshot covers trivially)."""
+
+    def __init__(self, max_steps: int = 5, batch_size=(), device=None):
+        super().__init__(device=device, batch_size=batch_size)
+        self.max_steps = max_steps
+        bs = self.batch_size
+        self.observation_spec = Composite(
+            {
+                "count": Unbounded(shape=(*bs, 1), device=self.device),
+            },
+            shape=bs,
+            device=self.device,
+        )
+        self.action_spec = Binary(shape=(*bs, 1), device=self.device, dtype=torch.bool)
+        self.reward_spec = Unbounded(shape=(*bs, 1), device=self.device)
+
+    def _reset(self, tensordict=None, **kwargs):
+        bs = self.batch_size
+        return TensorDict(
+            {
+                "count": torch.zeros(*bs, 1, device=self.device),
+                "done": torch.zeros(*bs, 1, dtype=torch.bool, device=self.device),
+                "terminated": torch.zeros(*bs, 1, dtype=torch.bool, device=self.device),
+            },
+            batch_size=bs,
+            device=self.device,
+        )
+
+    def _step(self, tensordict):
+        count = tensordict.get("count") + tensordict.get("action").float()
+        done = count >= self.max_steps
+        return TensorDict(
+            {
+                "count": count,
+                "reward": torch.ones_like(count),
+                "done": done,
+                "terminated": done,
+            },
+            batch_size=self.batch_size,
+            device=self.device,
+        )
+
+    def _set_seed(self, seed):
+        return seed
+
+
+class MockBatchedLockedEnv(CountingEnv):
+    """Env whose batch size is FIXED at construction — rejects inputs of
+    any other batch shape (reference mocking_classes MockBatchedLockedEnv)."""
+
+    batch_locked = True
+
+    def _step(self, tensordict):
+        if tuple(tensordict.batch_size) != tuple(self.batch_size):
+            raise RuntimeError(
+                f"batch-locked env: expected {tuple(self.batch_size)}, got "
+                f"{tuple(tensordict.batch_size)}"
+            )
+        return super()._step(tensordict)
+
+
+class MockBatchedUnLockedEnv(CountingEnv):
+    """Env accepting arbitrary leading batch shapes (reference
+    MockBatchedUnLockedEnv)."""
+
+    batch_locked = False
+
+    def _step(self, tensordict):
+        bs = tensordict.batch_size
+        action = tensordict.get("action")
+        count = tensordict.get("observation", torch.zeros(*bs, 1)) + action.float()
+        done = count >= self.max_steps
+        return TensorDict(
+            {
+                "observation": count,
+                "reward": torch.ones_like(count),
+                "done": done,
+                "terminated": done,
+            },
+            batch_size=bs,
+            device=self.device,
+        )
+
+
+class DiscreteActionConvMockEnv(EnvBase):
+    """Pixel observation + categorical action (the DQN/Atari shape;
+    reference DiscreteActionConvMockEnv)."""
+
+    _supports_masked_reset = True
+
+    def __init__(self, batch_size=(), device=None, pixel_shape=(1, 7, 7), n_actions: int = 7, max_steps: int = 10):
+        super().__init__(device=device, batch_size=batch_size)
+        self.max_steps = max_steps
+        self.pixel_shape = tuple(pixel_shape)
+        bs = self.batch_size
+        self.observation_spec = Composite(
+            {
+                "pixels": Bounded(
+                    low=0.0, high=1.0, shape=(*bs, *pixel_shape), device=self.device
+                )
+            },
+            shape=bs,
+            device=self.device,
+        )
+        self.action_spec = Categorical(n_actions, shape=bs, device=self.device, dtype=torch.int64)
+        self.reward_spec = Unbounded(shape=(*bs, 1), device=self.device)
+        self._t = torch.zeros(*bs, 1, device=self.device)
+
+    def _pix(self):
+        return torch.rand(*self.batch_size, *self.pixel_shape, device=self.device)
+
+    def _reset(self, tensordict=None, **kwargs):
+        self._t = torch.zeros(*self.batch_size, 1, device=self.device)
+        bs = self.batch_size
+        return TensorDict(
+            {
+                "pixels": self._pix(),
+                "done": torch.zeros(*bs, 1, dtype=torch.bool, device=self.device),
+                "terminated": torch.zeros(*bs, 1, dtype=torch.bool, device=self.device),
+            },
+            batch_size=bs,
+            device=self.device,
+        )
+
+    def _step(self, tensordict):
+        self._t = self._t + 1
+        done = self._t >= self.max_steps
+        act = tensordict.get("action").reshape(*self.batch_size, 1).float()
+        return TensorDict(
+            {
+                "pixels": self._pix(),
+                "reward": act / 10.0 + 1.0,
+                "done": done,
+                "terminated": done,
+            },
+            batch_size=self.batch_size,
+            device=self.device,
+        )
+
+    def _set_seed(self, seed):
+        return seed
+
+
+class ContinuousActionConvMockEnv(DiscreteActionConvMockEnv):
+    """Pixel observation + continuous action (reference
+    ContinuousActionConvMockEnv)."""
+
+    def __init__(self, batch_size=(), device=None, pixel_shape=(1, 7, 7), act_dim: int = 4, max_steps: int = 10):
+        super().__init__(batch_size=batch_size, device=device, pixel_shape=pixel_shape, max_steps=max_steps)
+        bs = self.batch_size
+        self.action_spec = Bounded(low=-1.0, high=1.0, shape=(*bs, act_dim), device=self.device)
+
+    def _step(self, tensordict):
+        self._t = self._t + 1
+        done = self._t >= self.max_steps
+        act = tensordict.get("action")
+        return TensorDict(
+            {
+                "pixels": self._pix(),
+                "reward": act.pow(2).sum(-1, keepdim=True),
+                "done": done,
+                "terminated": done,
+            },
+            batch_size=self.batch_size,
+            device=self.device,
+        )
+
+
+class MultiAgentCountingEnv(EnvBase):
+    """Homogeneous n-agent counting env with ("agents", ...) grouping
+    (reference MultiAgentCountingEnv — the dense-MARL complement of
+    HeterogeneousCountingEnv)."""
+
+    def __init__(self, n_agents: int = 3, max_steps: int = 5, batch_size=(), device=None):
+        super().__init__(device=device, batch_size=batch_size)
+        self.n_agents = n_agents
+        self.max_steps = max_steps
+        bs = self.batch_size
+        self.observation_spec = Composite(
+            {("agents", "observation"): Unbounded(shape=(*bs, n_agents, 3), device=self.device)},
+            shape=bs,
+            device=self.device,
+        )
+        self.full_action_spec = Composite(
+            {("agents", "action"): Binary(shape=(*bs, n_agents, 1), device=self.device, dtype=torch.bool)},
+            shape=bs,
+            device=self.device,
+        )
+        self.full_reward_spec = Composite(
+            {("agents", "reward"): Unbounded(shape=(*bs, n_agents, 1), device=self.device)},
+            shape=bs,
+            device=self.device,
+        )
+        self.count = torch.zeros(*bs, n_agents, 1, device=self.device)
+
+    def _reset(self, tensordict=None, **kwargs):
+        bs = self.batch_size
+        self.count = torch.zeros(*bs, self.n_agents, 1, device=self.device)
+        td = TensorDict({}, batch_size=bs, device=self.device)
+        td.set(("agents", "observation"), self.count.expand(*bs, self.n_agents, 3).clone())
+        td.set("done", torch.zeros(*bs, 1, dtype=torch.bool, device=self.device))
+        td.set("terminated", torch.zeros(*bs, 1, dtype=torch.bool, device=self.device))
+        return td
+
+    def _step(self, tensordict):
+        act = tensordict.get(("agents", "action")).float()
+        self.count = self.count + act
+        done = (self.count >= self.max_steps).any(-2)
+        td = TensorDict({}, batch_size=self.batch_size, device=self.device)
+        td.set(("agents", "observation"), self.count.expand(*self.batch_size, self.n_agents, 3).clone())
+        td.set(("agents", "reward"), torch.ones_like(self.count))
+        td.set("done", done)
+        td.set("terminated", done.clone())
+        return td
+
+    def _set_seed(self, seed):
+        return seed
+
+
+class EnvWithMetadata(CountingEnv):
+    """Counting env carrying a NON-TENSOR leaf through reset/step
+    (reference EnvWithMetadata): exercises NonTensorData plumbing."""
+
+    def _reset(self, tensordict=None, **kwargs):
+        td = super()._reset(tensordict, **kwargs)
+        td.set_non_tensor("info_str", "reset")
+        return td
+
+    def _step(self, tensordict):
+        td = super()._step(tensordict)
+        td.set_non_tensor("info_str", f"step{int(self.count.reshape(-1)[0])}")
+        return td
+
+
+class CountingPolicy:
+    """Deterministic mock policy: always emits action=1 for counting
+    envs (reference CountingEnvCountPolicy)."""
+
+    def __init__(self, action_spec=None, action_key="action"):
+        self.action_spec = action_spec
+        self.action_key = action_key
+
+    def __call__(self, td):
+        shape = (
+            self.action_spec.shape
+            if self.action_spec is not None
+            else (*td.batch_size, 1)
+        )
+        dtype = self.action_spec.dtype if self.action_spec is not None else torch.bool
+        td.set(self.action_key, torch.ones(shape, dtype=dtype, device=td.device))
+        return td
+
+
+__all__ += [
+    "StatelessCountingEnv",
+    "MockBatchedLockedEnv",
+    "MockBatchedUnLockedEnv",
+    "DiscreteActionConvMockEnv",
+    "ContinuousActionConvMockEnv",
+    "MultiAgentCountingEnv",
+    "EnvWithMetadata",
+    "CountingPolicy",
+]

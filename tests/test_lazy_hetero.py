@@ -157,3 +157,60 @@ class TestNewMockEnvs:
         ]
         with pytest.raises(RuntimeError, match="heterogeneous"):
             r.get("observation")
+
+
+class TestMoreMockEnvs:
+    def test_stateless_counting(self):
+        from rl_amd.testing import StatelessCountingEnv
+
+        env = StatelessCountingEnv(batch_size=[3])
+        check_env_specs(env)
+
+    def test_batch_locked(self):
+        from rl_amd.testing import MockBatchedLockedEnv
+
+        env = MockBatchedLockedEnv(batch_size=[2])
+        check_env_specs(env)
+        bad = env.reset()[0:1]
+        with pytest.raises(RuntimeError, match="batch-locked"):
+            env._step(bad)
+
+    def test_conv_mocks(self):
+        from rl_amd.testing import (
+            ContinuousActionConvMockEnv,
+            DiscreteActionConvMockEnv,
+        )
+
+        check_env_specs(DiscreteActionConvMockEnv(batch_size=[2]))
+        check_env_specs(ContinuousActionConvMockEnv(batch_size=[2]))
+
+    def test_multiagent_counting_rollout(self):
+        from rl_amd.testing import MultiAgentCountingEnv
+
+        env = MultiAgentCountingEnv(n_agents=3, batch_size=[2])
+        r = env.rollout(3)
+        assert r.get(("agents", "observation")).shape == (2, 3, 3, 3)
+        assert r.get(("next", "agents", "reward")).shape == (2, 3, 3, 1)
+
+    def test_env_with_metadata(self):
+        from rl_amd.testing import EnvWithMetadata
+
+        env = EnvWithMetadata(batch_size=[2])
+        td = env.reset()
+        assert td.get_non_tensor("info_str") == "reset"
+        r = env.rollout(2)
+        assert r is not None
+
+    def test_counting_policy_drives_counting_env(self):
+        from rl_amd.collectors import Collector
+        from rl_amd.testing import CountingEnv, CountingPolicy
+
+        env = CountingEnv(batch_size=[2], max_steps=100)
+        pol = CountingPolicy(env.action_spec)
+        col = Collector(env, pol, frames_per_batch=10, total_frames=10)
+        batch = next(iter(col))
+        # counting semantics assertable exactly
+        assert torch.equal(
+            batch.get("observation")[0].reshape(-1),
+            torch.arange(0, 5, dtype=torch.float32),
+        )
