@@ -87,47 +87,26 @@ def main():
         beta=0.5,
     )
 
-    carrier = env.reset()
-    if cuda:
-        # in-place state + the fused one-launch env transition
-        # (csrc/env_step.hip) on the collect path
-        env.enable_capture_mode(True)
+    # collect through the PUBLIC Collector API: its GPU fast path
+    # captures the policy+env step as a hipGraph (GraphedRollout) and the
+    # iterator extends the replay buffer (only the writer index stays on
+    # the host).  AutocastPolicy puts the 256x256 actor GEMMs on the
+    # matrix cores with float32 outputs into the buffer.
+    from rl_amd.collectors import Collector
+    from rl_amd.modules import AutocastPolicy
 
-    # collect: policy + env step captured as a hipGraph over a static
-    # entry-observation buffer (the env state updates in place); only
-    # rb.extend — whose writer index lives on the host — stays eager,
-    # reading the transition from the static snapshot.
-    entry_obs = carrier.get("observation").clone()
-    collect_snap = {}
-    collect_graph = None
-
-    def collect_body():
-        td = TensorDict({"observation": entry_obs}, batch_size=[args.envs], device=device)
-        with torch.no_grad(), set_exploration_type(ExplorationType.RANDOM), autocast:
-            td = actor(td)
-            td.set("action", td.get("action").float())
-            td, nxt = env.step_and_maybe_reset(td)
-        snap = td.select(
-            "observation", "action", "sample_log_prob",
-            ("next", "observation"), ("next", "reward"),
-            ("next", "done"), ("next", "terminated"),
-        )
-        if not collect_snap:
-            for k in snap.keys(True, True):
-                collect_snap[k] = snap.get(k).clone()
-        else:
-            for k, buf in collect_snap.items():
-                buf.copy_(snap.get(k))
-        entry_obs.copy_(nxt.get("observation"))
+    collector = Collector(
+        env,
+        AutocastPolicy(actor),
+        frames_per_batch=args.envs,
+        total_frames=-1,
+        replay_buffer=rb,
+        use_graph="auto" if (args.graph and cuda) else False,
+    )
+    col_iter = iter(collector)
 
     def collect():
-        if collect_graph is not None:
-            collect_graph.replay()
-        else:
-            collect_body()
-        rb.extend(TensorDict(
-            dict(collect_snap), batch_size=[args.envs], device=device
-        ))
+        next(col_iter)
 
     def update_body(batch):
         with autocast:
@@ -193,22 +172,6 @@ def main():
 
     if use_graph:
         try:
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                for _ in range(3):
-                    collect_body()
-            torch.cuda.current_stream().wait_stream(side)
-            cg = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(cg):
-                collect_body()
-            collect_graph = cg
-        except Exception:
-            import traceback
-
-            traceback.print_exc()
-            collect_graph = None
-        try:
             capture_update()
         except Exception:
             import traceback
@@ -253,7 +216,11 @@ def main():
                     "parallelism": "dp1",
                     "buffer_device": str(device),
                     "update_graph": bool(graph is not None),
-                    "collect_graph": bool(collect_graph is not None),
+                    "collect_graph": bool(
+                        collector._graphed is not None
+                        and (collector._graphed.captured or collector._graphed.mega)
+                    ),
+                    "library_api_collection": True,
                 },
             }
         )

@@ -140,6 +140,7 @@ __all__ = sorted(set(list(globals().get('__all__', [])) + ['DiffusionActor', 'EX
 # reference-parity: MCTS scores, multi-agent/cross-group critics, RNN +
 # primer utilities, discrete-distribution extras
 from .extras import (  # noqa: F401
+    AutocastPolicy,
     CrossCriticGroupSpec,
     CrossGroupCritic,
     MCTSScore,

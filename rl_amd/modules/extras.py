@@ -24,6 +24,7 @@ from torch import nn
 from ..tensordict import TensorDict, TensorDictBase, TensorDictModuleBase
 
 __all__ = [
+    "AutocastPolicy",
     "MCTSScore",
     "MCTSScores",
     "UCB1TunedScore",
@@ -387,3 +388,35 @@ class VLAWrapperBase(nn.Module):
         td.set(("vla_action", "chunk"), chunk)
         td.set("action", chunk[..., 0, :])
         return td
+
+
+class AutocastPolicy(torch.nn.Module):
+    """Run a policy under bf16 autocast with float32 outputs — the
+    rollout-side compute-dtype wrapper for collector fast paths
+    (CDNA4 has no fp32 MFMA, so fp32 policy GEMMs run on the vector
+    ALU; autocast moves them to the matrix cores).  ``cache_enabled``
+    stays False so the forward is hipGraph-capture-safe."""
+
+    def __init__(self, policy, dtype=torch.bfloat16, cast_outputs_to_float: bool = True):
+        super().__init__()
+        self.policy = policy
+        self.dtype = dtype
+        self.cast_outputs_to_float = cast_outputs_to_float
+        self.in_keys = getattr(policy, "in_keys", [])
+        self.out_keys = getattr(policy, "out_keys", [])
+
+    def forward(self, td):
+        enabled = torch.cuda.is_available() and (
+            td.device is not None and td.device.type == "cuda"
+        )
+        with torch.autocast("cuda", dtype=self.dtype, enabled=enabled, cache_enabled=False):
+            out = self.policy(td)
+        if self.cast_outputs_to_float:
+            for k in self.out_keys:
+                v = out.get(k, None)
+                if v is not None and v.dtype == self.dtype:
+                    out.set(k, v.float())
+        return out
+
+    def get_dist(self, td):
+        return self.policy.get_dist(td)

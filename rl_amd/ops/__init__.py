@@ -360,14 +360,36 @@ class FusedTanhNormalActor(torch.nn.Module):
         )
         self.in_keys = [in_key]
         self.out_keys = ["action", "sample_log_prob", "loc", "scale"]
+        self._kernel_ok = None  # probed on first GPU forward
 
     def forward(self, td):
         obs = td.get(self.in_key)
-        if not (obs.is_cuda and not torch.is_grad_enabled() and HAS_HIP_EXT):
+        if not (
+            obs.is_cuda
+            and not torch.is_grad_enabled()
+            and HAS_HIP_EXT
+            and self._kernel_ok is not False
+        ):
             return self.eager_actor(td)
         w1, w2, w3 = self.linears
         A = w3.out_features // 2
         eps = torch.randn(obs.shape[0], A, device=obs.device)
+        if self._kernel_ok is None:
+            # dims may exceed the kernel's LDS budget (e.g. Humanoid's
+            # 376-wide obs): probe once, fall back to the eager actor
+            try:
+                _C.fused_actor(
+                    obs.contiguous().float(),
+                    w1.weight.detach(), w1.bias.detach(),
+                    w2.weight.detach(), w2.bias.detach(),
+                    w3.weight.detach(), w3.bias.detach(),
+                    eps, float(self.inv_softplus_bias), float(self.scale_lb),
+                    True,
+                )
+                self._kernel_ok = True
+            except RuntimeError:
+                self._kernel_ok = False
+                return self.eager_actor(td)
         action, logp, loc, scale = _C.fused_actor(
             obs.contiguous().float(),
             w1.weight.detach(), w1.bias.detach(),
