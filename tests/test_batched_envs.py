@@ -152,3 +152,41 @@ class TestParallelEnvSharedMemory:
         finally:
             env_sh.close()
             env_pk.close()
+
+
+# ---------------------------------------------------------------------- #
+# HIP-IPC tensor sharing + event sync (reference batched_envs.py
+# 3349-3361: CUDA tensors shared across processes with event handshake;
+# on this pool HSA_ENABLE_IPC_MODE_LEGACY=0 routes through dmabuf IPC)
+# ---------------------------------------------------------------------- #
+def _hip_ipc_child(q_in, q_out):
+    import torch
+
+    t, evt = q_in.get(timeout=60)
+    evt.synchronize()  # producer's writes visible
+    assert float(t[0]) == 1.0
+    t.add_(41.0)  # write back through the SAME device memory
+    done = torch.cuda.Event(interprocess=False)
+    done.record()
+    done.synchronize()
+    q_out.put(True)
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(180)
+def test_hip_ipc_tensor_sharing_with_event_sync():
+    import torch.multiprocessing as tmp
+
+    ctx = tmp.get_context("spawn")
+    q_in, q_out = ctx.Queue(), ctx.Queue()
+    proc = ctx.Process(target=_hip_ipc_child, args=(q_in, q_out))
+    proc.start()
+    t = torch.zeros(4, device="cuda")
+    t += 1.0
+    evt = torch.cuda.Event(interprocess=True)
+    evt.record()
+    q_in.put((t, evt))
+    assert q_out.get(timeout=120) is True
+    proc.join(30)
+    torch.cuda.synchronize()
+    assert float(t[0]) == 42.0  # child's write visible through IPC
