@@ -318,3 +318,78 @@ class TestPreemption:
             assert n == 2
         finally:
             col.shutdown()
+
+
+class TestGraphedPPONumerics:
+    """GraphedPPO's orchestration must be EXACTLY the textbook loop —
+    this is the math under bench.py's measured number."""
+
+    def _build(self, seed):
+        from rl_amd.collectors import Collector
+        from rl_amd.modules import MLP, NormalParamExtractor, ProbabilisticActor, TanhNormal, ValueOperator
+        from rl_amd.objectives import ClipPPOLoss
+        from rl_amd.objectives.value.advantages import GAE
+        from rl_amd.tensordict import TensorDictModule
+        from rl_amd.testing import ContinuousActionVecMockEnv
+
+        torch.manual_seed(seed)
+        env = ContinuousActionVecMockEnv(batch_size=[4], max_steps=100)
+        env.set_seed(seed)
+        net = torch.nn.Sequential(
+            MLP(in_features=7, out_features=2 * 5, num_cells=[16]),
+            NormalParamExtractor(),
+        )
+        actor = ProbabilisticActor(
+            TensorDictModule(net, in_keys=["observation"], out_keys=["loc", "scale"]),
+            in_keys=["loc", "scale"],
+            distribution_class=TanhNormal,
+            return_log_prob=True,
+        )
+        critic = ValueOperator(
+            MLP(in_features=7, out_features=1, num_cells=[16]), in_keys=["observation"]
+        )
+        col = Collector(env, actor, frames_per_batch=16, total_frames=-1)
+        loss = ClipPPOLoss(actor, critic, normalize_advantage=True)
+        gae = GAE(gamma=0.99, lmbda=0.95, value_network=critic, vectorized=True)
+        optim = torch.optim.Adam(
+            list(actor.parameters()) + list(critic.parameters()), lr=1e-3
+        )
+        return col, actor, critic, loss, gae, optim
+
+    def test_step_matches_manual_loop(self):
+        from rl_amd.trainers import GraphedPPO
+
+        col1, actor1, critic1, loss1, gae1, optim1 = self._build(7)
+        runner = GraphedPPO(col1, gae1, loss1, optim1, minibatches=2, epochs=1,
+                            capture=False)
+        runner.step()
+        runner.step()
+
+        col2, actor2, critic2, loss2, gae2, optim2 = self._build(7)
+        for _ in range(2):
+            batch = col2.rollout()
+            with torch.no_grad():
+                gae2(batch)
+            flat = batch.reshape(-1)
+            n = flat.batch_size[0]
+            mb = n // 2
+            perm = torch.randperm(n)
+            shuffled = flat[perm]
+            for i in range(2):
+                sub = shuffled[i * mb : (i + 1) * mb]
+                out = loss2(sub)
+                total = sum(
+                    v for k, v in out.items()
+                    if isinstance(k, str) and k.startswith("loss_")
+                )
+                optim2.zero_grad(set_to_none=False)
+                total.backward()
+                torch.nn.utils.clip_grad_norm_(
+                    list(actor2.parameters()) + list(critic2.parameters()), 1.0
+                )
+                optim2.step()
+
+        for p1, p2 in zip(actor1.parameters(), actor2.parameters()):
+            assert torch.equal(p1, p2)
+        for p1, p2 in zip(critic1.parameters(), critic2.parameters()):
+            assert torch.equal(p1, p2)
