@@ -147,7 +147,12 @@ class Trainer:
     def _run_hooks(self, stage: str, arg=None):
         out = arg
         for op, _name in self._hooks[stage]:
-            res = op(out) if out is not None else op()
+            if out is not None or stage == "process_optim_batch":
+                # batch-processing hooks accept an (optionally None)
+                # batch — async collection samples with no batch at all
+                res = op(out)
+            else:
+                res = op()
             if res is not None:
                 out = res
         return out
@@ -211,6 +216,8 @@ class Trainer:
 
     def train(self) -> None:
         """(reference trainers.py:1400)"""
+        if self.async_collection:
+            return self._train_async()
         pbar = None
         if self.progress_bar:
             try:
@@ -245,6 +252,39 @@ class Trainer:
                 break
         if pbar is not None:
             pbar.close()
+        self.collector.shutdown()
+
+    def _train_async(self) -> None:
+        """Collect/train overlap (reference trainers.py:1409-1415): the
+        collector streams into its replay buffer from a background
+        thread while the learner samples and steps concurrently."""
+        import time as _time
+
+        rb = getattr(self.collector, "replay_buffer", None)
+        if rb is None:
+            # adopt the trainer's own buffer (OffPolicyTrainer builds one)
+            rb = getattr(self, "replay_buffer", None)
+            if rb is not None:
+                self.collector.replay_buffer = rb
+                self.collector.extend_buffer = True
+        if rb is None:
+            raise RuntimeError(
+                "async_collection requires a replay buffer (on the collector "
+                "or the trainer)"
+            )
+        self.collector.start()
+        min_fill = getattr(rb, "_batch_size", None) or 1
+        deadline = _time.monotonic() + 300
+        while len(rb) < min_fill:
+            if _time.monotonic() > deadline:
+                raise TimeoutError("async collection never filled the buffer")
+            _time.sleep(0.01)
+        while self.collector.frames < self.total_frames:
+            self.optim_steps(None)
+            self._run_hooks("post_steps")
+            self.collected_frames = self.collector.frames
+        self.collected_frames = self.collector.frames
+        self.collector.async_shutdown()
         self.collector.shutdown()
 
     # ------------------------------------------------------------------ #

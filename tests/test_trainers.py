@@ -367,3 +367,34 @@ class TestGraphedPPOTrainer:
         assert tr.collected_frames == 64
         assert any(p.grad is not None for p in actor.parameters())
         tr.shutdown()
+
+
+class TestAsyncCollection:
+    @pytest.mark.timeout(120)
+    def test_sac_trainer_async_collection(self):
+        """Collect/train overlap (reference trainers.py:1409): the
+        collector thread streams into the buffer while the learner
+        samples concurrently."""
+        from rl_amd.trainers import SACTrainer
+
+        env = ContinuousActionVecMockEnv(batch_size=[2], max_steps=50)
+        actor, _ = make_cont_actor_critic()
+        qnet = ValueOperator(
+            MLP(in_features=7 + 5, out_features=1, num_cells=[32]),
+            in_keys=["observation", "action"],
+        )
+        col = Collector(env, actor, frames_per_batch=16, total_frames=96)
+        tr = SACTrainer(
+            actor=actor,
+            qvalue=qnet,
+            collector=col,
+            total_frames=96,
+            batch_size=8,
+            buffer_size=500,
+            optim_steps_per_batch=1,
+            progress_bar=False,
+            async_collection=True,
+        )
+        tr.train()
+        assert tr.collected_frames >= 96
+        assert tr._optim_count > 0
