@@ -51,13 +51,21 @@ class Evaluator:
         self._last_eval = 0
         self._thread: Optional[threading.Thread] = None
         self.last_result: Optional[float] = None
+        # async eval runs on a SNAPSHOT of the weights (reference
+        # _ThreadEvalBackend): the learner may update mid-rollout
+        self._eval_policy = None
+        if async_eval and hasattr(policy, "state_dict"):
+            import copy
 
-    def evaluate(self, step: Optional[int] = None) -> float:
+            self._eval_policy = copy.deepcopy(policy)
+
+    def evaluate(self, step: Optional[int] = None, policy=None) -> float:
+        policy = policy if policy is not None else self.policy
         rewards = []
         with set_exploration_type(self.exploration_type), torch.no_grad():
             for _ in range(self.num_episodes):
                 rollout = self.env.rollout(
-                    self.max_steps, policy=self.policy, break_when_any_done=True
+                    self.max_steps, policy=policy, break_when_any_done=True
                 )
                 rewards.append(rollout.get(("next", "reward")).sum().item())
         mean_r = sum(rewards) / len(rewards)
@@ -73,8 +81,13 @@ class Evaluator:
         if self.async_eval:
             if self._thread is not None and self._thread.is_alive():
                 return self.last_result
+            eval_policy = self.policy
+            if self._eval_policy is not None:
+                self._eval_policy.load_state_dict(self.policy.state_dict())
+                eval_policy = self._eval_policy
             self._thread = threading.Thread(
-                target=self.evaluate, args=(collected_frames,), daemon=True
+                target=self.evaluate, args=(collected_frames, eval_policy),
+                daemon=True,
             )
             self._thread.start()
             return self.last_result
