@@ -538,12 +538,22 @@ def enable_splitk_bf16_cache(module: torch.nn.Module) -> None:
 
 
 def refresh_splitk_caches(*modules: torch.nn.Module) -> None:
-    """Re-cast all cached bf16 weights (one cast kernel per layer) —
-    call once per training step after the optimizer update."""
+    """Re-cast all cached bf16 weights — batched into a single foreach
+    copy (the per-layer casts were 78 launches/step ≈ 10% of the PPO
+    step in the T=64 profile); call once per training step after the
+    optimizer update."""
+    dsts: list = []
+    srcs: list = []
     for module in modules:
         for m in module.modules():
             if isinstance(m, SplitKLinear) and m._bf16_cache:
-                m.refresh_bf16_cache_()
+                dsts.append(m.weight_bf16)
+                srcs.append(m.weight.detach())
+                if m.bias is not None:
+                    dsts.append(m.bias_bf16)
+                    srcs.append(m.bias.detach())
+    if dsts:
+        torch._foreach_copy_(dsts, srcs)
 
 
 def convert_linears_to_splitk(module: torch.nn.Module) -> torch.nn.Module:
