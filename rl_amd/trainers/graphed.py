@@ -142,7 +142,11 @@ class GraphedPPO:
         with self._autocast:
             out = self.loss_module(sub)
             total = self._total_loss(out)
-        self.optimizer.zero_grad(set_to_none=False)
+        # set_to_none=True even under capture: grads allocated inside the
+        # capture live in the graph's private pool (stable across
+        # replays), and it removes a zero-fill + accumulate-add per
+        # parameter per minibatch (~100 kernels/step in the T=16 profile)
+        self.optimizer.zero_grad(set_to_none=True)
         total.backward()
 
     def _capture_mb_graph(self, example_sub: TensorDictBase) -> None:
