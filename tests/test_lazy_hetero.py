@@ -214,3 +214,52 @@ class TestMoreMockEnvs:
             batch.get("observation")[0].reshape(-1),
             torch.arange(0, 5, dtype=torch.float32),
         )
+
+
+class TestLazyDenseParity:
+    """On HOMOGENEOUS data a forced lazy stack must behave exactly like
+    the dense stack for the common operations."""
+
+    def _pair(self):
+        torch.manual_seed(0)
+        tds = [
+            TensorDict(
+                {"a": torch.randn(4, 3), "n": {"b": torch.randn(4, 2)}},
+                batch_size=[4],
+            )
+            for _ in range(5)
+        ]
+        return lazy_stack(tds, 0), stack([td.clone() for td in tds], 0)
+
+    def test_get_and_keys(self):
+        lz, dn = self._pair()
+        assert lz.batch_size == dn.batch_size
+        assert set(map(str, lz.keys(True, True))) == set(map(str, dn.keys(True, True)))
+        assert torch.equal(lz.get("a"), dn.get("a"))
+        assert torch.equal(lz.get(("n", "b")), dn.get(("n", "b")))
+
+    def test_indexing(self):
+        lz, dn = self._pair()
+        assert torch.equal(lz[2].get("a"), dn[2].get("a"))
+        assert torch.equal(lz[1:4].get("a"), dn[1:4].get("a"))
+        assert torch.equal(lz[2, 1:3].get("a"), dn[2, 1:3].get("a"))
+
+    def test_set_roundtrip(self):
+        lz, dn = self._pair()
+        v = torch.randn(5, 4, 7)
+        lz.set("c", v)
+        dn.set("c", v)
+        assert torch.equal(lz.get("c"), dn.get("c"))
+
+    def test_to_tensordict_apply_select(self):
+        lz, dn = self._pair()
+        assert torch.equal(lz.to_tensordict().get("a"), dn.get("a"))
+        assert torch.equal(lz.apply(lambda t: t * 2).get("a"), dn.get("a") * 2)
+        assert torch.equal(lz.select("a").get("a"), dn.get("a"))
+
+    def test_update(self):
+        lz, dn = self._pair()
+        upd = TensorDict({"a": torch.ones(5, 4, 3)}, batch_size=[5, 4])
+        lz.update(upd)
+        dn.update(upd)
+        assert torch.equal(lz.get("a"), dn.get("a"))
