@@ -365,7 +365,12 @@ class EnvBase(nn.Module):
         """step + partial auto-reset; returns ``(td_with_next, next_root)``
         (reference torchrl/envs/common.py:4090)."""
         tensordict = self.step(tensordict)
-        next_root = step_mdp(tensordict)
+        next_root = step_mdp(
+            tensordict,
+            reward_keys=self.reward_keys,
+            done_keys=self.done_keys,
+            action_keys=self.action_keys,
+        )
         if self._supports_masked_reset:
             done = next_root.get("done", None)
             if done is None:
@@ -437,11 +442,21 @@ class EnvBase(nn.Module):
             if break_when_all_done and bool(done.all()):
                 break
             if break_when_any_done or break_when_all_done:
-                td = step_mdp(td)
+                td = step_mdp(
+                    td,
+                    reward_keys=self.reward_keys,
+                    done_keys=self.done_keys,
+                    action_keys=self.action_keys,
+                )
             else:
                 td, _next_root = None, None
                 last = tds[-1]
-                next_root = step_mdp(last)
+                next_root = step_mdp(
+                    last,
+                    reward_keys=self.reward_keys,
+                    done_keys=self.done_keys,
+                    action_keys=self.action_keys,
+                )
                 if terminated_or_truncated(next_root, key="_reset"):
                     next_root = self.reset(next_root)
                 td = next_root

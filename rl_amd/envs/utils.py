@@ -59,23 +59,37 @@ def step_mdp(
     out = nxt.clone(False)
     if exclude_reward:
         for k in reward_keys:
-            out.pop(k, None)
+            _pop_nested(out, k)
     if exclude_done:
         for k in done_keys:
-            out.pop(k, None)
+            _pop_nested(out, k)
+    nested_actions = [
+        unravel_key(k) for k in action_keys if isinstance(unravel_key(k), tuple)
+    ]
     if keep_other:
         skip = {"next"}
         if exclude_action:
-            skip |= {unravel_key(k) for k in action_keys}
+            skip |= {k for k in map(unravel_key, action_keys) if isinstance(k, str)}
         for k, v in tensordict._data.items():
             if k in skip or k in out._data:
                 continue
             out._data[k] = v
+        if exclude_action:
+            for k in nested_actions:
+                _pop_nested(out, k)
     elif not exclude_action:
         for k in action_keys:
             if k in tensordict and k not in out._data:
                 out.set(k, tensordict.get(k))
     return out
+
+
+def _pop_nested(td: TensorDictBase, key) -> None:
+    key = unravel_key(key)
+    try:
+        td.del_(key)
+    except KeyError:
+        pass
 
 
 def terminated_or_truncated(

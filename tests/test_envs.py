@@ -358,3 +358,26 @@ class TestGatedLibWrappers:
             cls(None)
         if isinstance(exc.value, ImportError):
             assert "not installed" in str(exc.value)
+
+
+def test_step_mdp_excludes_nested_reward_and_action():
+    """Nested (agent-grouped) reward/action keys must not leak into the
+    next root (regression: leaked ('agents','reward') made collector
+    batches heterogeneous → silent lazy-stack degradation)."""
+    from rl_amd.testing import MultiAgentCountingEnv
+
+    env = MultiAgentCountingEnv(n_agents=3, batch_size=[2])
+    td = env.reset()
+    td = env.rand_action(td)
+    td, next_root = env.step_and_maybe_reset(td)
+    assert ("agents", "reward") not in next_root
+    assert ("agents", "action") not in next_root
+    assert ("agents", "observation") in next_root
+    # collector batches stay DENSE
+    from rl_amd.collectors import Collector
+    from rl_amd.tensordict import LazyStackedTensorDict
+
+    col = Collector(env, None, frames_per_batch=8, total_frames=8)
+    batch = next(iter(col))
+    assert not isinstance(batch, LazyStackedTensorDict)
+    assert not isinstance(batch.get("agents"), LazyStackedTensorDict)
