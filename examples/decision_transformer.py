@@ -53,7 +53,7 @@ def build_offline_buffer(n_traj=50, T=20, device="cpu"):
         sampler=SliceSampler(slice_len=CTX),
         batch_size=4 * CTX,
     )
-    rb.extend(data)
+    rb.extend(data.to(device))
     return rb
 
 
