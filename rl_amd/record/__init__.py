@@ -8,5 +8,5 @@ from .loggers import (
     get_logger,
 )
 from .recorder import Every, LoggerMonitor, PixelRenderTransform, TensorDictRecorder, VideoRecorder
-from .loggers.common import ProcessLogger, RayLogger  # noqa: F401
+from .loggers.common import ProcessLogger, RayLogger, TrackioLogger  # noqa: F401
 __all__ = sorted(set(list(globals().get("__all__", [])) + ["ProcessLogger", "RayLogger"]))

@@ -221,6 +221,37 @@ def get_logger(
     raise NotImplementedError(f"unknown logger type {logger_type}")
 
 
+class TrackioLogger(Logger):
+    """trackio experiment logger (reference record/loggers/trackio.py:21)
+    — gated on the `trackio` package."""
+
+    def __init__(self, exp_name: str, project: str = "rl_amd", **kwargs):
+        import importlib.util
+
+        if importlib.util.find_spec("trackio") is None:
+            raise ImportError(
+                "TrackioLogger requires the `trackio` package, which is not "
+                "installed in this image. Use CSVLogger instead."
+            )
+        import trackio
+
+        super().__init__(exp_name)
+        self.run = trackio.init(project=project, name=exp_name, **kwargs)
+
+    def log_scalar(self, name, value, step=None):
+        import trackio
+
+        trackio.log({name: value}, step=step)
+
+    def log_hparams(self, cfg):
+        self.run.config.update(dict(cfg))
+
+    def log_video(self, name, video, step=None, **kw):
+        import trackio
+
+        trackio.log({name: video}, step=step)
+
+
 class ProcessLogger(Logger):
     """Logger service living in a dedicated process (reference
     record/loggers/process.py:132): log_* calls are enqueued through a
@@ -316,4 +347,4 @@ class RayLogger(Logger):
         self._actor.log_video.remote(name, video, step, **kw)
 
 
-__all__ += ["ProcessLogger", "RayLogger"]
+__all__ += ["ProcessLogger", "RayLogger", "TrackioLogger"]
