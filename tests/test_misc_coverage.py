@@ -351,3 +351,29 @@ def test_gif_video_pipeline(tmp_path):
     p = str(tmp_path / "direct.gif")
     write_gif(torch.rand(4, 8, 8, 3), p, fps=5)
     assert open(p, "rb").read()[:6] == b"GIF89a"
+
+
+def test_autocast_policy_float_outputs():
+    from rl_amd.modules import AutocastPolicy
+    from rl_amd.tensordict import TensorDictModule
+
+    pol = TensorDictModule(torch.nn.Linear(4, 2), in_keys=["observation"], out_keys=["action"])
+    wrapped = AutocastPolicy(pol)
+    td = TensorDict({"observation": torch.randn(3, 4)}, batch_size=[3])
+    out = wrapped(td)
+    assert out.get("action").dtype == torch.float32
+    assert wrapped.in_keys == ["observation"] and wrapped.out_keys == ["action"]
+
+
+def test_splitk_refresh_foreach_matches_per_layer():
+    from rl_amd.ops import SplitKLinear, refresh_splitk_caches
+
+    lin1 = SplitKLinear(4, 4).enable_bf16_cache()
+    lin2 = SplitKLinear(4, 2).enable_bf16_cache()
+    mod = torch.nn.Sequential(lin1, lin2)
+    with torch.no_grad():
+        lin1.weight.add_(1.0)
+        lin2.bias.add_(2.0)
+    refresh_splitk_caches(mod)
+    assert torch.equal(lin1.weight_bf16, lin1.weight.detach().to(torch.bfloat16))
+    assert torch.equal(lin2.bias_bf16, lin2.bias.detach().to(torch.bfloat16))
