@@ -168,12 +168,109 @@ class MeltingpotEnv(EnvBase):
 
 
 class OpenSpielEnv(EnvBase):
-    """OpenSpiel turn-based games (reference openspiel.py): categorical
-    action over legal moves with an action mask."""
+    """OpenSpiel turn-based games (reference openspiel.py): observation
+    tensor + legal-action mask + ``current_player`` per step; chance
+    nodes resolve by their declared distribution; rewards are the acting
+    player's returns delta."""
 
-    def __init__(self, *args, device=None, **kwargs):
-        _require("pyspiel", type(self).__name__)
-        raise NotImplementedError("openspiel scaffolding")
+    def __init__(self, game_string: str = "tic_tac_toe", *, device=None, **kwargs):
+        pyspiel = _require("pyspiel", type(self).__name__)
+        super().__init__(device=device, batch_size=())
+        from ...data.tensor_specs import (
+            Binary,
+            Categorical,
+            Composite,
+            Unbounded,
+        )
+        import torch
+
+        self._game = pyspiel.load_game(game_string, kwargs or None)
+        n_act = self._game.num_distinct_actions()
+        obs_len = int(
+            torch.tensor(self._game.observation_tensor_shape()).prod()
+        )
+        self.observation_spec = Composite(
+            {
+                "observation": Unbounded(shape=(obs_len,), device=self.device),
+                "action_mask": Binary(
+                    shape=(n_act,), device=self.device, dtype=torch.bool
+                ),
+                "current_player": Unbounded(
+                    shape=(1,), device=self.device, dtype=torch.int64
+                ),
+            },
+            shape=(),
+            device=self.device,
+        )
+        self.action_spec = Categorical(n_act, shape=(), device=self.device, dtype=torch.int64)
+        self.reward_spec = Unbounded(shape=(1,), device=self.device)
+        self._state = None
+        self._rng = None
+
+    def _resolve_chance(self):
+        import random
+
+        while self._state.is_chance_node():
+            actions, probs = zip(*self._state.chance_outcomes())
+            self._state.apply_action(
+                (self._rng or random).choices(actions, weights=probs)[0]
+            )
+
+    def _obs_td(self, reward=0.0, done=False):
+        import torch
+
+        from ...tensordict import TensorDict
+
+        player = max(self._state.current_player(), 0)
+        if done:
+            obs = torch.zeros(self.observation_spec["observation"].shape, device=self.device)
+            mask = torch.zeros(
+                self.observation_spec["action_mask"].shape,
+                dtype=torch.bool,
+                device=self.device,
+            )
+        else:
+            obs = torch.tensor(
+                self._state.observation_tensor(player),
+                dtype=torch.float32,
+                device=self.device,
+            ).reshape(-1)
+            mask = torch.zeros(
+                self.observation_spec["action_mask"].shape,
+                dtype=torch.bool,
+                device=self.device,
+            )
+            mask[self._state.legal_actions(player)] = True
+        td = TensorDict({}, batch_size=(), device=self.device)
+        td.set("observation", obs)
+        td.set("action_mask", mask)
+        td.set("current_player", torch.tensor([player], device=self.device))
+        td.set("done", torch.tensor([done], device=self.device))
+        td.set("terminated", torch.tensor([done], device=self.device))
+        return td
+
+    def _reset(self, tensordict=None, **kwargs):
+        self._state = self._game.new_initial_state()
+        self._resolve_chance()
+        return self._obs_td()
+
+    def _step(self, tensordict):
+        import torch
+
+        player = self._state.current_player()
+        self._state.apply_action(int(tensordict.get("action")))
+        self._resolve_chance()
+        done = self._state.is_terminal()
+        r = self._state.returns()[player] if done else self._state.rewards()[player]
+        td = self._obs_td(done=done)
+        td.set("reward", torch.tensor([float(r)], device=self.device))
+        return td
+
+    def _set_seed(self, seed):
+        import random
+
+        self._rng = random.Random(seed)
+        return seed
 
 
 class UnityMLAgentsEnv(EnvBase):
@@ -250,17 +347,42 @@ def PettingZooEnv(task: str = None, *, parallel: bool = True, device=None, **kwa
     return _W(env, device=device)
 
 
-class MOGymEnv(EnvBase):
-    """mo-gymnasium multi-objective envs (reference libs/meltingpot.py
-    neighborhood: mo_gym.py) — the reward is a vector spec."""
+class MOGymWrapper(GymWrapper):
+    """Wrap a mo-gymnasium env instance (reference libs/mo_gym):
+    identical gym step/reset conventions with a VECTOR reward — the
+    reward spec takes the env's ``reward_space``/``reward_dim`` shape."""
 
-    def __init__(self, env_name: str, *, device=None, **kwargs):
+    def __init__(self, env, *, device=None, **kwargs):
         _require("mo_gymnasium", type(self).__name__)
-        raise NotImplementedError("mo_gymnasium scaffolding")
+        super().__init__(env, device=device, **kwargs)
+
+    def _make_specs(self):
+        super()._make_specs()
+        from ...data.tensor_specs import Unbounded
+
+        env = self._env
+        d = None
+        space = getattr(env, "reward_space", None)
+        if space is not None and getattr(space, "shape", None):
+            d = int(space.shape[0])
+        elif getattr(env, "reward_dim", None):
+            d = int(env.reward_dim)
+        if d:
+            self.reward_spec = Unbounded(shape=(d,), device=self.device)
+
+    def read_reward(self, reward):
+        import numpy as _np
+        import torch as _torch
+
+        return _torch.as_tensor(
+            _np.asarray(reward), dtype=_torch.float32, device=self.device
+        ).reshape(self.reward_spec.shape)
 
 
-class MOGymWrapper(MOGymEnv):
-    """Wrap an existing mo-gymnasium env instance."""
+def MOGymEnv(env_name: str, *, device=None, **kwargs):
+    """Build a mo-gymnasium env by name and wrap it."""
+    mo_gym = _require("mo_gymnasium", "MOGymEnv")
+    return MOGymWrapper(mo_gym.make(env_name, **kwargs), device=device)
 
 
 class MultiThreadedEnvWrapper(EnvPoolEnv):

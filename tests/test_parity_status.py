@@ -19,11 +19,9 @@ DOCUMENTED_SCAFFOLDING = {
     "MjLabEnv",
     "SMACv2Env",
     "MeltingpotEnv",
-    "OpenSpielEnv",
     "UnityMLAgentsEnv",
     "LiberoEnv",
     "OpenMLEnv",
-    "MOGymEnv",
     "BraxWrapper",  # misc_wrappers' jax-gated instance wrapper
     "LeRobotPolicyWrapper",
     "MujocoEnv",
