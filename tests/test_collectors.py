@@ -393,3 +393,23 @@ class TestGraphedPPONumerics:
             assert torch.equal(p1, p2)
         for p1, p2 in zip(critic1.parameters(), critic2.parameters()):
             assert torch.equal(p1, p2)
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_multisync_collector_gpu_envs():
+    """MultiSync workers with GPU-resident envs on one device (HIP IPC
+    shares the result buffers across processes)."""
+    from rl_amd.collectors import MultiSyncCollector
+    from rl_amd.envs.custom.synthetic import HalfCheetahVec
+
+    def make_env():
+        return HalfCheetahVec(batch_size=[8], device="cuda")
+
+    col = MultiSyncCollector(
+        [make_env] * 2, frames_per_batch=64, total_frames=128
+    )
+    batches = list(col)
+    assert sum(b.numel() for b in batches) == 128
+    assert batches[0].get("observation").shape[-1] == 17
+    col.shutdown()
