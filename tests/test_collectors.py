@@ -395,19 +395,21 @@ class TestGraphedPPONumerics:
             assert torch.equal(p1, p2)
 
 
+def _gpu_cheetah_env():
+    from rl_amd.envs.custom.synthetic import HalfCheetahVec
+
+    return HalfCheetahVec(batch_size=[8], device="cuda")
+
+
 @pytest.mark.gpu
 @pytest.mark.timeout(300)
 def test_multisync_collector_gpu_envs():
     """MultiSync workers with GPU-resident envs on one device (HIP IPC
     shares the result buffers across processes)."""
     from rl_amd.collectors import MultiSyncCollector
-    from rl_amd.envs.custom.synthetic import HalfCheetahVec
-
-    def make_env():
-        return HalfCheetahVec(batch_size=[8], device="cuda")
 
     col = MultiSyncCollector(
-        [make_env] * 2, frames_per_batch=64, total_frames=128
+        [_gpu_cheetah_env] * 2, frames_per_batch=64, total_frames=128
     )
     batches = list(col)
     assert sum(b.numel() for b in batches) == 128
