@@ -63,6 +63,15 @@ void launch_tanh_normal_entropy_fwd(const float*, const float*, const float*,
 void launch_tanh_normal_entropy_bwd(const float*, const float*, const float*,
                                     const float*, float*, float*, int, int,
                                     void*);
+void launch_adv_stats(const float*, float*, float*, long, void*);
+void launch_ppo_clip_fwd(const float*, const float*, const float*, float*,
+                         float*, float, float, long, void*);
+void launch_ppo_clip_bwd(const float*, const float*, const float*,
+                         const float*, float*, float, float, long, void*);
+void launch_smooth_l1_fwd(const void*, const float*, float*, float*, long,
+                          int, void*);
+void launch_smooth_l1_bwd(const void*, const float*, const float*, void*,
+                          long, int, void*);
 void launch_synthetic_env_step(float*, const float*, const float*,
                                const float*, float*, float*, float*, float*,
                                bool*, const float*, long, long, long, int,
@@ -603,6 +612,80 @@ void fused_actor_into(torch::Tensor obs, torch::Tensor w1, torch::Tensor b1,
       (long)logp.stride(0), B, O, H1, H2, A, (float)inv_softplus_bias,
       (float)scale_lb, (void*)stream);
 }
+// Fused ClipPPO objective (reference torchrl/objectives/ppo.py:1082
+// ClipPPOLoss.forward elementwise/reduction chain).  Returns
+// (out[3] = {loss_objective, ESS/N, clip_fraction}, stats[2]) where
+// stats = (mu, 1/sigma) of the optional advantage normalization —
+// saved for backward so the normalized advantage is never
+// materialized.
+std::vector<torch::Tensor> ppo_clip_fwd(torch::Tensor lw, torch::Tensor adv,
+                                        double lo, double hi,
+                                        bool normalize) {
+  TORCH_CHECK(lw.is_cuda() && lw.scalar_type() == torch::kFloat32 &&
+                  adv.scalar_type() == torch::kFloat32,
+              "ppo_clip_fwd: fp32 cuda");
+  TORCH_CHECK(lw.is_contiguous() && adv.is_contiguous(), "contiguous");
+  const long N = lw.numel();
+  TORCH_CHECK(adv.numel() == N, "ppo_clip_fwd: numel mismatch");
+  auto opt = lw.options();
+  auto part = torch::empty({256 * 4}, opt);
+  auto out = torch::empty({3}, opt);
+  auto stats = normalize ? torch::empty({2}, opt) : torch::Tensor();
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (normalize)
+    launch_adv_stats(adv.data_ptr<float>(), part.data_ptr<float>(),
+                     stats.data_ptr<float>(), N, (void*)stream);
+  launch_ppo_clip_fwd(lw.data_ptr<float>(), adv.data_ptr<float>(),
+                      normalize ? stats.data_ptr<float>() : nullptr,
+                      part.data_ptr<float>(), out.data_ptr<float>(),
+                      (float)lo, (float)hi, N, (void*)stream);
+  if (!normalize) stats = torch::empty({0}, opt);
+  return {out, stats};
+}
+
+torch::Tensor ppo_clip_bwd(torch::Tensor lw, torch::Tensor adv,
+                           torch::Tensor stats, torch::Tensor gout,
+                           double lo, double hi) {
+  const long N = lw.numel();
+  auto dlw = torch::empty_like(lw);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_ppo_clip_bwd(
+      lw.data_ptr<float>(), adv.data_ptr<float>(),
+      stats.numel() ? stats.data_ptr<float>() : nullptr,
+      gout.data_ptr<float>(), dlw.data_ptr<float>(), (float)lo, (float)hi, N,
+      (void*)stream);
+  return dlw;
+}
+
+// Fused mean smooth-L1 (beta=1) critic loss; value may be bf16 (autocast).
+torch::Tensor smooth_l1_fwd(torch::Tensor v, torch::Tensor t) {
+  TORCH_CHECK(v.is_cuda() && v.is_contiguous() && t.is_contiguous(),
+              "smooth_l1: cuda contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, "target fp32");
+  const bool bf16 = v.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || v.scalar_type() == torch::kFloat32, "value fp32/bf16");
+  const long N = v.numel();
+  TORCH_CHECK(t.numel() == N, "smooth_l1: numel mismatch");
+  auto part = torch::empty({256}, t.options());
+  auto out = torch::empty({1}, t.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_smooth_l1_fwd(v.data_ptr(), t.data_ptr<float>(),
+                       part.data_ptr<float>(), out.data_ptr<float>(), N,
+                       bf16 ? 1 : 0, (void*)stream);
+  return out;
+}
+
+torch::Tensor smooth_l1_bwd(torch::Tensor v, torch::Tensor t,
+                            torch::Tensor gout) {
+  const bool bf16 = v.scalar_type() == torch::kBFloat16;
+  const long N = v.numel();
+  auto dv = torch::empty_like(v);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_smooth_l1_bwd(v.data_ptr(), t.data_ptr<float>(),
+                       gout.data_ptr<float>(), dv.data_ptr(), N, bf16 ? 1 : 0,
+                       (void*)stream);
+  return dv;
+}
 #endif  // RL_AMD_WITH_HIP
 
 // ---------------------------------------------------------------------------
@@ -648,6 +731,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused TanhNormal log-prob backward (HIP)");
   m.def("fused_actor_into", &fused_actor_into,
         "store-direct fused actor (HIP)");
+  m.def("ppo_clip_fwd", &ppo_clip_fwd,
+        "fused ClipPPO objective + diagnostics forward (HIP)");
+  m.def("ppo_clip_bwd", &ppo_clip_bwd, "fused ClipPPO backward (HIP)");
+  m.def("smooth_l1_fwd", &smooth_l1_fwd,
+        "fused mean smooth-L1 critic loss forward (HIP)");
+  m.def("smooth_l1_bwd", &smooth_l1_bwd,
+        "fused smooth-L1 backward (HIP)");
   m.def("synthetic_env_step_into", &synthetic_env_step_into,
         "store-direct fused env transition with auto-reset (HIP)");
   m.def("wgrad_splitk", &wgrad_splitk,

@@ -160,6 +160,32 @@ class PPOLoss(LossModule):
         scale = adv.std().clamp_min(1e-6)
         return (adv - loc) / scale
 
+    def _loss_critic_reduced(self, td: TensorDictBase) -> torch.Tensor:
+        """``self._reduce(self.loss_critic(td))`` with a fused HIP path:
+        smooth-L1 + mean in two launches, one-launch analytic backward
+        (csrc/loss_ops.hip) when no value clipping is in play."""
+        if (
+            self.clip_value is None
+            and self.loss_critic_type in ("smooth_l1", "huber")
+            and self.reduction == "mean"
+        ):
+            target = td.get(self.tensor_keys.value_target)
+            value = self.critic_network(td.clone(False)).get(self.tensor_keys.value)
+            if (
+                value.is_cuda
+                and value.dtype in (torch.float32, torch.bfloat16)
+                and target.dtype == torch.float32
+                and value.numel() == target.numel()
+            ):
+                from .. import ops
+
+                if ops.HAS_HIP_EXT:
+                    return ops.smooth_l1_mean(value, target)
+            return self._reduce(
+                distance_loss(value, target, self.loss_critic_type)
+            )
+        return self._reduce(self.loss_critic(td))
+
     def _reduce(self, x: torch.Tensor) -> torch.Tensor:
         if self.reduction == "mean":
             return x.mean()
@@ -191,8 +217,7 @@ class PPOLoss(LossModule):
             out.set("entropy", entropy.detach().mean())
             out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
         if self.critic_coeff is not None and self.critic_coeff > 0:
-            loss_critic = self.loss_critic(td)
-            out.set("loss_critic", self.critic_coeff * self._reduce(loss_critic))
+            out.set("loss_critic", self.critic_coeff * self._loss_critic_reduced(td))
         out.set("ESS", ess / log_weight.numel())
         return out
 
@@ -219,25 +244,48 @@ class ClipPPOLoss(PPOLoss):
                 self.make_value_estimator()
             self.value_estimator(td)
             advantage = td.get(self.tensor_keys.advantage)
-        if self.normalize_advantage and advantage.numel() > 1:
-            advantage = self._normalize_adv(advantage)
+        normalize = self.normalize_advantage and advantage.numel() > 1
         log_weight, dist, _ = self._log_weight(td)
         if log_weight.dim() < advantage.dim():
             log_weight = log_weight.unsqueeze(-1)
-        ratio = log_weight.exp()
-        gain1 = ratio * advantage
-        ratio_clamped = log_weight.clamp(*self._clip_bounds).exp()
-        gain2 = ratio_clamped * advantage
-        gain = torch.minimum(gain1, gain2)
-        with torch.no_grad():
-            lw_flat = log_weight.reshape(-1)
-            ess = lw_flat.exp().sum().pow(2) / lw_flat.mul(2).exp().sum().clamp_min(1e-12)
-            clip_fraction = (ratio_clamped != ratio).float().mean()
+        fused = None
+        if (
+            self.reduction == "mean"
+            and log_weight.is_cuda
+            and log_weight.dtype == torch.float32
+            and advantage.dtype == torch.float32
+            and not advantage.requires_grad
+            and log_weight.numel() == advantage.numel()
+        ):
+            from .. import ops
+
+            if ops.HAS_HIP_EXT:
+                # fused objective + diagnostics + advantage
+                # normalization (csrc/loss_ops.hip)
+                fused = ops.ppo_clip_objective(
+                    log_weight, advantage, *self._clip_bounds, normalize
+                )
+        if fused is not None:
+            loss_objective, ess_per_sample, clip_fraction = fused
+        else:
+            if normalize:
+                advantage = self._normalize_adv(advantage)
+            ratio = log_weight.exp()
+            gain1 = ratio * advantage
+            ratio_clamped = log_weight.clamp(*self._clip_bounds).exp()
+            gain2 = ratio_clamped * advantage
+            gain = torch.minimum(gain1, gain2)
+            with torch.no_grad():
+                lw_flat = log_weight.reshape(-1)
+                ess = lw_flat.exp().sum().pow(2) / lw_flat.mul(2).exp().sum().clamp_min(1e-12)
+                clip_fraction = (ratio_clamped != ratio).float().mean()
+                ess_per_sample = ess / log_weight.numel()
+            loss_objective = -self._reduce(gain)
         out = TensorDict(
             {
-                "loss_objective": -self._reduce(gain),
+                "loss_objective": loss_objective,
                 "clip_fraction": clip_fraction,
-                "ESS": ess / log_weight.numel(),
+                "ESS": ess_per_sample,
             },
             batch_size=[],
         )
@@ -246,8 +294,7 @@ class ClipPPOLoss(PPOLoss):
             out.set("entropy", entropy.detach().mean())
             out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
         if self.critic_coeff is not None and self.critic_coeff > 0:
-            loss_critic = self.loss_critic(td)
-            out.set("loss_critic", self.critic_coeff * self._reduce(loss_critic))
+            out.set("loss_critic", self.critic_coeff * self._loss_critic_reduced(td))
         return out
 
 
@@ -308,5 +355,5 @@ class KLPENPPOLoss(PPOLoss):
             out.set("entropy", entropy.detach().mean())
             out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
         if self.critic_coeff is not None and self.critic_coeff > 0:
-            out.set("loss_critic", self.critic_coeff * self._reduce(self.loss_critic(td)))
+            out.set("loss_critic", self.critic_coeff * self._loss_critic_reduced(td))
         return out
