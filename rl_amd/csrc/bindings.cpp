@@ -72,6 +72,14 @@ void launch_smooth_l1_fwd(const void*, const float*, float*, float*, long,
                           int, void*);
 void launch_smooth_l1_bwd(const void*, const float*, const float*, void*,
                           long, int, void*);
+void launch_ppo_head_fwd(const void*, const float*, const float*,
+                         const float*, const float*, const float*, float*,
+                         float*, float, float, float, float, float, long, int,
+                         int, void*);
+void launch_ppo_head_bwd(const void*, const float*, const float*,
+                         const float*, const float*, const float*,
+                         const float*, const float*, void*, float, float,
+                         float, float, float, long, int, int, void*);
 void launch_synthetic_env_step(float*, const float*, const float*,
                                const float*, float*, float*, float*, float*,
                                bool*, const float*, long, long, long, int,
@@ -686,6 +694,72 @@ torch::Tensor smooth_l1_bwd(torch::Tensor v, torch::Tensor t,
                        (void*)stream);
   return dv;
 }
+// Mega-fused TanhNormal head loss: raw actor-head output [N, 2A] ->
+// (out[5] = {loss_objective, ESS/N, clip_fraction, entropy_mean,
+// loss_entropy}, stats).  See csrc/loss_ops.hip.
+std::vector<torch::Tensor> ppo_head_fwd(torch::Tensor head,
+                                        torch::Tensor action,
+                                        torch::Tensor eps, torch::Tensor prev,
+                                        torch::Tensor adv, double sp_bias,
+                                        double lb, double lo, double hi,
+                                        double ent_coeff, bool normalize) {
+  TORCH_CHECK(head.is_cuda() && head.is_contiguous(), "head cuda contiguous");
+  const bool bf16 = head.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || head.scalar_type() == torch::kFloat32,
+              "head fp32/bf16");
+  TORCH_CHECK(action.scalar_type() == torch::kFloat32 &&
+                  eps.scalar_type() == torch::kFloat32 &&
+                  prev.scalar_type() == torch::kFloat32 &&
+                  adv.scalar_type() == torch::kFloat32,
+              "fp32 data inputs");
+  TORCH_CHECK(action.is_contiguous() && eps.is_contiguous() &&
+                  prev.is_contiguous() && adv.is_contiguous(),
+              "contiguous");
+  const long N = head.size(0);
+  const int A = (int)(head.size(1) / 2);
+  TORCH_CHECK(head.size(1) == 2 * A && action.numel() == N * A &&
+                  eps.numel() == N * A && prev.numel() == N &&
+                  adv.numel() == N,
+              "shape mismatch");
+  auto opt = action.options();
+  auto part = torch::empty({256 * 5}, opt);
+  auto out = torch::empty({5}, opt);
+  auto stats = normalize ? torch::empty({2}, opt) : torch::empty({0}, opt);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (normalize)
+    launch_adv_stats(adv.data_ptr<float>(), part.data_ptr<float>(),
+                     stats.data_ptr<float>(), N, (void*)stream);
+  launch_ppo_head_fwd(head.data_ptr(), action.data_ptr<float>(),
+                      eps.data_ptr<float>(), prev.data_ptr<float>(),
+                      adv.data_ptr<float>(),
+                      normalize ? stats.data_ptr<float>() : nullptr,
+                      part.data_ptr<float>(), out.data_ptr<float>(),
+                      (float)sp_bias, (float)lb, (float)lo, (float)hi,
+                      (float)ent_coeff, N, A, bf16 ? 1 : 0, (void*)stream);
+  return {out, stats};
+}
+
+torch::Tensor ppo_head_bwd(torch::Tensor head, torch::Tensor action,
+                           torch::Tensor eps, torch::Tensor prev,
+                           torch::Tensor adv, torch::Tensor stats,
+                           torch::Tensor gobj, torch::Tensor gent,
+                           double sp_bias, double lb, double lo, double hi,
+                           double ent_coeff) {
+  const bool bf16 = head.scalar_type() == torch::kBFloat16;
+  const long N = head.size(0);
+  const int A = (int)(head.size(1) / 2);
+  auto dhead = torch::empty_like(head);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_ppo_head_bwd(
+      head.data_ptr(), action.data_ptr<float>(), eps.data_ptr<float>(),
+      prev.data_ptr<float>(), adv.data_ptr<float>(),
+      stats.numel() ? stats.data_ptr<float>() : nullptr,
+      gobj.numel() ? gobj.data_ptr<float>() : nullptr,
+      gent.numel() ? gent.data_ptr<float>() : nullptr, dhead.data_ptr(),
+      (float)sp_bias, (float)lb, (float)lo, (float)hi, (float)ent_coeff, N, A,
+      bf16 ? 1 : 0, (void*)stream);
+  return dhead;
+}
 #endif  // RL_AMD_WITH_HIP
 
 // ---------------------------------------------------------------------------
@@ -738,6 +812,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused mean smooth-L1 critic loss forward (HIP)");
   m.def("smooth_l1_bwd", &smooth_l1_bwd,
         "fused smooth-L1 backward (HIP)");
+  m.def("ppo_head_fwd", &ppo_head_fwd,
+        "mega-fused TanhNormal head -> PPO losses forward (HIP)");
+  m.def("ppo_head_bwd", &ppo_head_bwd,
+        "mega-fused head-loss backward (HIP)");
   m.def("synthetic_env_step_into", &synthetic_env_step_into,
         "store-direct fused env transition with auto-reset (HIP)");
   m.def("wgrad_splitk", &wgrad_splitk,

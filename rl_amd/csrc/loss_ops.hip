@@ -426,3 +426,226 @@ extern "C" void launch_smooth_l1_bwd(const void* v, const float* t,
                        0, (hipStream_t)stream, (const float*)v, t, gout,
                        (float*)dv, N);
 }
+
+// ---------------------------------------------------------------------------
+// Mega-fused TanhNormal head loss: raw actor-head output -> PPO losses.
+//
+// Fuses, per minibatch, everything between the actor MLP's last GEMM
+// and the scalar losses (reference torchrl/objectives/ppo.py:1082
+// ClipPPOLoss.forward + the NormalParamExtractor in actor
+// construction):
+//   NormalParamExtractor (chunk + biased softplus + clamp_min)
+//   + TanhNormal log-prob of the stored action (ratio numerator)
+//   + log-weight, clipped surrogate, ESS, clip-fraction
+//   + single-sample reparameterized MC entropy + its mean/loss
+//   + optional advantage (x-mean)/std normalization
+// into one grid-stride pass + a 1-WG finalize (plus the 2-launch
+// advantage-stats pre-pass), and the ENTIRE backward — d(raw head) for
+// both loss_objective and loss_entropy — into ONE analytic kernel that
+// recomputes the forward per row (cheaper than saving activations at
+// these sizes).  Eagerly this chain is ~25 forward + ~20 backward
+// launches per minibatch.  clamp_min backward passes gradient where
+// s >= lb (boundary included), matching torch.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+#define PH_LOG_SQRT_2PI 0.9189385332046727f
+#define PH_LOG2 0.6931471805599453f
+#define PH_ATANH_LIM (1.0f - 1.1920929e-7f)
+
+template <typename TV>
+__global__ void ppo_head_fwd_partials_k(
+    const TV* __restrict__ head,      // [N, 2A]: loc | raw scale
+    const float* __restrict__ action, // [N, A]
+    const float* __restrict__ eps,    // [N, A] entropy sample
+    const float* __restrict__ prev,   // [N] behavior log-prob
+    const float* __restrict__ adv,    // [N]
+    const float* __restrict__ stats,  // nullable (mu, 1/sigma)
+    float* __restrict__ part,         // [nWG, 5]
+    const float sp_bias, const float lb, const float lo, const float hi,
+    const long N, const int A) {
+  const float mu = stats ? stats[0] : 0.f;
+  const float isd = stats ? stats[1] : 1.f;
+  float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f;
+  for (long n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+       n += (long)gridDim.x * blockDim.x) {
+    float lp = 0.f, ent = 0.f;
+    for (int a = 0; a < A; ++a) {
+      const float loc = (float)head[n * 2 * A + a];
+      const float spre = (float)head[n * 2 * A + A + a] + sp_bias;
+      const float s = fmaxf(softplusf(spre), lb);
+      const float y = fminf(fmaxf(action[n * A + a], -PH_ATANH_LIM),
+                            PH_ATANH_LIM);
+      const float u = atanhf(y);
+      const float z = (u - loc) / s;
+      lp += -0.5f * z * z - __logf(s) - PH_LOG_SQRT_2PI
+            - 2.0f * (PH_LOG2 - u - softplusf(-2.0f * u));
+      const float e = eps[n * A + a];
+      const float x = tanhf(loc + s * e);
+      ent += 0.5f * e * e + __logf(s) + PH_LOG_SQRT_2PI + log1pf(-x * x);
+    }
+    const float w = lp - prev[n];
+    const float an = (adv[n] - mu) * isd;
+    const float r = expf(w);
+    const float rc = expf(fminf(fmaxf(w, lo), hi));
+    sg += fminf(r * an, rc * an);
+    sr += r;
+    sr2 += expf(2.0f * w);
+    sc += (rc != r) ? 1.f : 0.f;
+    se += ent;
+  }
+  __shared__ float smem[8];
+  float t;
+  t = block_sum(sg, smem);
+  if (threadIdx.x == 0) part[blockIdx.x * 5] = t;
+  __syncthreads();
+  t = block_sum(sr, smem);
+  if (threadIdx.x == 0) part[blockIdx.x * 5 + 1] = t;
+  __syncthreads();
+  t = block_sum(sr2, smem);
+  if (threadIdx.x == 0) part[blockIdx.x * 5 + 2] = t;
+  __syncthreads();
+  t = block_sum(sc, smem);
+  if (threadIdx.x == 0) part[blockIdx.x * 5 + 3] = t;
+  __syncthreads();
+  t = block_sum(se, smem);
+  if (threadIdx.x == 0) part[blockIdx.x * 5 + 4] = t;
+}
+
+__global__ void ppo_head_finalize_k(const float* __restrict__ part,
+                                    const int nwg, const long N,
+                                    const float ent_coeff,
+                                    float* __restrict__ out) {
+  float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f;
+  for (int i = threadIdx.x; i < nwg; i += blockDim.x) {
+    sg += part[5 * i];
+    sr += part[5 * i + 1];
+    sr2 += part[5 * i + 2];
+    sc += part[5 * i + 3];
+    se += part[5 * i + 4];
+  }
+  __shared__ float smem[8];
+  float t;
+  t = block_sum(sg, smem);
+  if (threadIdx.x == 0) out[0] = -t / (float)N;  // loss_objective
+  __syncthreads();
+  t = block_sum(sr, smem);
+  if (threadIdx.x == 0) smem[4] = t;
+  __syncthreads();
+  const float tsr = smem[4];
+  t = block_sum(sr2, smem);
+  if (threadIdx.x == 0)
+    out[1] = tsr * tsr / fmaxf(t, 1e-12f) / (float)N;  // ESS / N
+  __syncthreads();
+  t = block_sum(sc, smem);
+  if (threadIdx.x == 0) out[2] = t / (float)N;  // clip_fraction
+  __syncthreads();
+  t = block_sum(se, smem);
+  if (threadIdx.x == 0) {
+    out[3] = t / (float)N;                   // entropy (mean)
+    out[4] = -ent_coeff * t / (float)N;      // loss_entropy
+  }
+}
+
+template <typename TV>
+__global__ void ppo_head_bwd_k(
+    const TV* __restrict__ head, const float* __restrict__ action,
+    const float* __restrict__ eps, const float* __restrict__ prev,
+    const float* __restrict__ adv, const float* __restrict__ stats,
+    const float* __restrict__ gobj,  // 0-d upstream grad of loss_objective
+    const float* __restrict__ gent,  // nullable: grad of loss_entropy
+    TV* __restrict__ dhead, const float sp_bias, const float lb,
+    const float lo, const float hi, const float ent_coeff, const long N,
+    const int A) {
+  const float mu = stats ? stats[0] : 0.f;
+  const float isd = stats ? stats[1] : 1.f;
+  const float g1s = gobj ? -gobj[0] / (float)N : 0.f;
+  const float ges = gent ? -ent_coeff * gent[0] / (float)N : 0.f;
+  for (long n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+       n += (long)gridDim.x * blockDim.x) {
+    // pass 1: recompute lp -> per-row dlw
+    float lp = 0.f;
+    for (int a = 0; a < A; ++a) {
+      const float loc = (float)head[n * 2 * A + a];
+      const float s = fmaxf(
+          softplusf((float)head[n * 2 * A + A + a] + sp_bias), lb);
+      const float y = fminf(fmaxf(action[n * A + a], -PH_ATANH_LIM),
+                            PH_ATANH_LIM);
+      const float u = atanhf(y);
+      const float z = (u - loc) / s;
+      lp += -0.5f * z * z - __logf(s) - PH_LOG_SQRT_2PI
+            - 2.0f * (PH_LOG2 - u - softplusf(-2.0f * u));
+    }
+    const float w = lp - prev[n];
+    const float an = (adv[n] - mu) * isd;
+    const float r = expf(w);
+    const float rc = expf(fminf(fmaxf(w, lo), hi));
+    const float dlw = (r * an <= rc * an) ? g1s * an * r : 0.f;
+    // pass 2: analytic d(head)
+    for (int a = 0; a < A; ++a) {
+      const float loc = (float)head[n * 2 * A + a];
+      const float spre = (float)head[n * 2 * A + A + a] + sp_bias;
+      const float s0 = softplusf(spre);
+      const float s = fmaxf(s0, lb);
+      const float y = fminf(fmaxf(action[n * A + a], -PH_ATANH_LIM),
+                            PH_ATANH_LIM);
+      const float u = atanhf(y);
+      const float z = (u - loc) / s;
+      const float e = eps[n * A + a];
+      const float x = tanhf(loc + s * e);
+      const float dloc = dlw * z / s + ges * (-2.0f * x);
+      float ds = dlw * (z * z - 1.0f) / s + ges * (1.0f / s - 2.0f * x * e);
+      ds = (s0 >= lb) ? ds : 0.f;  // clamp_min backward
+      const float sig = 1.0f / (1.0f + __expf(-spre));  // softplus'
+      dhead[n * 2 * A + a] = (TV)dloc;
+      dhead[n * 2 * A + A + a] = (TV)(ds * sig);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" void launch_ppo_head_fwd(const void* head, const float* action,
+                                    const float* eps, const float* prev,
+                                    const float* adv, const float* stats,
+                                    float* part, float* out, float sp_bias,
+                                    float lb, float lo, float hi,
+                                    float ent_coeff, long N, int A,
+                                    int head_is_bf16, void* stream) {
+  const int blocks = red_blocks(N);
+  if (head_is_bf16)
+    hipLaunchKernelGGL(ppo_head_fwd_partials_k<__hip_bfloat16>, dim3(blocks),
+                       dim3(LP_THREADS), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)head, action, eps, prev, adv,
+                       stats, part, sp_bias, lb, lo, hi, N, A);
+  else
+    hipLaunchKernelGGL(ppo_head_fwd_partials_k<float>, dim3(blocks),
+                       dim3(LP_THREADS), 0, (hipStream_t)stream,
+                       (const float*)head, action, eps, prev, adv, stats,
+                       part, sp_bias, lb, lo, hi, N, A);
+  hipLaunchKernelGGL(ppo_head_finalize_k, dim3(1), dim3(LP_THREADS), 0,
+                     (hipStream_t)stream, part, blocks, N, ent_coeff, out);
+}
+
+extern "C" void launch_ppo_head_bwd(const void* head, const float* action,
+                                    const float* eps, const float* prev,
+                                    const float* adv, const float* stats,
+                                    const float* gobj, const float* gent,
+                                    void* dhead, float sp_bias, float lb,
+                                    float lo, float hi, float ent_coeff,
+                                    long N, int A, int head_is_bf16,
+                                    void* stream) {
+  const int blocks = red_blocks(N);
+  if (head_is_bf16)
+    hipLaunchKernelGGL(ppo_head_bwd_k<__hip_bfloat16>, dim3(blocks),
+                       dim3(LP_THREADS), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)head, action, eps, prev, adv,
+                       stats, gobj, gent, (__hip_bfloat16*)dhead, sp_bias,
+                       lb, lo, hi, ent_coeff, N, A);
+  else
+    hipLaunchKernelGGL(ppo_head_bwd_k<float>, dim3(blocks), dim3(LP_THREADS),
+                       0, (hipStream_t)stream, (const float*)head, action,
+                       eps, prev, adv, stats, gobj, gent, (float*)dhead,
+                       sp_bias, lb, lo, hi, ent_coeff, N, A);
+}

@@ -236,6 +236,110 @@ class ClipPPOLoss(PPOLoss):
             math.log1p(float(self.clip_epsilon)),
         )
 
+    def _mega_probe(self):
+        """One-time introspection of the actor for the mega-fused head
+        path (csrc/loss_ops.hip ppo_head kernels): ProbabilisticActor
+        over Sequential(body…, NormalParamExtractor(biased_softplus))
+        with a TanhNormal(-1, 1) distribution.  Returns a dict of
+        pieces, or False when the structure does not match."""
+        cached = self.__dict__.get("_mega_info", None)
+        if cached is not None:
+            return cached
+        info = False
+        try:
+            from ..modules.models.models import NormalParamExtractor
+            from ..modules.distributions.continuous import TanhNormal
+            from ..tensordict.nn import (
+                ProbabilisticTensorDictModule,
+                TensorDictModule,
+            )
+
+            actor = self.actor_network
+            mods = list(actor.module)
+            if len(mods) == 2:
+                tdm, prob = mods
+                dk = getattr(prob, "distribution_kwargs", None) or {}
+                low, high = dk.get("low", -1.0), dk.get("high", 1.0)
+                if (
+                    isinstance(prob, ProbabilisticTensorDictModule)
+                    and prob.distribution_class is TanhNormal
+                    and float(low) == -1.0
+                    and float(high) == 1.0
+                    and list(prob.in_keys) == ["loc", "scale"]
+                    and isinstance(tdm, TensorDictModule)
+                    and list(tdm.out_keys) == ["loc", "scale"]
+                    and len(tdm.in_keys) == 1
+                    and isinstance(tdm.module, torch.nn.Sequential)
+                    and len(tdm.module) >= 2
+                    and isinstance(tdm.module[-1], NormalParamExtractor)
+                    and tdm.module[-1]._inv_softplus_bias is not None
+                ):
+                    ext = tdm.module[-1]
+                    info = {
+                        "body": list(tdm.module)[:-1],
+                        "obs_key": tdm.in_keys[0],
+                        "sp_bias": float(ext._inv_softplus_bias),
+                        "scale_lb": float(ext.scale_lb),
+                    }
+        except Exception:
+            info = False
+        self.__dict__["_mega_info"] = info
+        return info
+
+    def _mega_head_loss(self, td, advantage, normalize):
+        """Fused actor-loss path: raw head -> (loss_objective,
+        loss_entropy, entropy, ESS, clip_fraction) in 3-5 launches +
+        1-kernel backward.  Returns None when ineligible."""
+        if not (
+            self.entropy_bonus
+            and self.samples_mc_entropy == 1
+            and self.reduction == "mean"
+            and not advantage.requires_grad
+            and advantage.dtype == torch.float32
+        ):
+            return None
+        from .. import ops
+
+        if not ops.HAS_HIP_EXT:
+            return None
+        info = self._mega_probe()
+        if not info:
+            return None
+        obs = td.get(info["obs_key"], None)
+        action = td.get(self.tensor_keys.action, None)
+        prev_lp = td.get(self.tensor_keys.sample_log_prob, None)
+        if obs is None or action is None or prev_lp is None or not obs.is_cuda:
+            return None
+        if action.dtype != torch.float32 or prev_lp.dtype != torch.float32:
+            return None
+        A = action.shape[-1]
+        N = action.numel() // A
+        if advantage.numel() != N or prev_lp.numel() != N:
+            return None
+        head = obs
+        for m in info["body"]:
+            head = m(head)
+        if head.dtype not in (torch.float32, torch.bfloat16):
+            return None
+        if head.shape[-1] != 2 * A:
+            return None
+        eps = torch.randn(N, A, device=action.device, dtype=torch.float32)
+        lo, hi = self._clip_bounds
+        loss_obj, loss_ent, ent_mean, ess, clip_frac = ops.ppo_head_loss(
+            head.reshape(N, 2 * A),
+            action.reshape(N, A).float(),
+            prev_lp.reshape(N),
+            advantage.reshape(N),
+            eps,
+            sp_bias=info["sp_bias"],
+            scale_lb=info["scale_lb"],
+            lo=lo,
+            hi=hi,
+            entropy_coeff=float(self.entropy_coeff),
+            normalize=normalize,
+        )
+        return loss_obj, loss_ent, ent_mean, ess, clip_frac
+
     def forward(self, tensordict: TensorDictBase) -> TensorDictBase:
         td = tensordict.clone(False)
         advantage = td.get(self.tensor_keys.advantage, None)
@@ -245,6 +349,25 @@ class ClipPPOLoss(PPOLoss):
             self.value_estimator(td)
             advantage = td.get(self.tensor_keys.advantage)
         normalize = self.normalize_advantage and advantage.numel() > 1
+        mega = self._mega_head_loss(td, advantage, normalize)
+        if mega is not None:
+            loss_obj, loss_ent, ent_mean, ess, clip_frac = mega
+            out = TensorDict(
+                {
+                    "loss_objective": loss_obj,
+                    "clip_fraction": clip_frac,
+                    "ESS": ess,
+                    "entropy": ent_mean,
+                    "loss_entropy": loss_ent,
+                },
+                batch_size=[],
+            )
+            if self.critic_coeff is not None and self.critic_coeff > 0:
+                out.set(
+                    "loss_critic",
+                    self.critic_coeff * self._loss_critic_reduced(td),
+                )
+            return out
         log_weight, dist, _ = self._log_weight(td)
         if log_weight.dim() < advantage.dim():
             log_weight = log_weight.unsqueeze(-1)

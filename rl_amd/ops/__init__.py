@@ -687,6 +687,63 @@ def ppo_clip_objective(log_weight, advantage, lo, hi, normalize):
     return _PPOClipFn.apply(log_weight, advantage, float(lo), float(hi), bool(normalize))
 
 
+class _PPOHeadLossFn(torch.autograd.Function):
+    """Mega-fused TanhNormal head loss (csrc/loss_ops.hip): raw actor
+    head [N, 2A] -> (loss_objective, loss_entropy, entropy_mean, ESS/N,
+    clip_fraction) in one grid pass + finalize; the entire backward —
+    NormalParamExtractor + log-prob + clipped surrogate + entropy — is
+    ONE analytic kernel emitting d(head) in the head's dtype (bf16
+    under autocast, so the MLP backward needs no casts)."""
+
+    @staticmethod
+    def forward(ctx, head, action, prev_lp, adv, eps, sp_bias, lb, lo, hi,
+                ent_coeff, normalize):
+        head = head.contiguous()
+        action = action.contiguous().detach()
+        prev_lp = prev_lp.contiguous().detach()
+        adv = adv.contiguous().detach()
+        eps = eps.contiguous()
+        out, stats = _C.ppo_head_fwd(
+            head, action, eps, prev_lp, adv, sp_bias, lb, lo, hi, ent_coeff,
+            normalize,
+        )
+        ctx.save_for_backward(head, action, eps, prev_lp, adv, stats)
+        ctx.cfg = (sp_bias, lb, lo, hi, ent_coeff)
+        loss_obj, ess, clip_frac, ent_mean, loss_ent = (
+            out[0], out[1], out[2], out[3], out[4]
+        )
+        ctx.mark_non_differentiable(ess, clip_frac, ent_mean)
+        return loss_obj, loss_ent, ent_mean, ess, clip_frac
+
+    @staticmethod
+    def backward(ctx, g_obj, g_ent, g_em, g_ess, g_cf):
+        head, action, eps, prev_lp, adv, stats = ctx.saved_tensors
+        sp_bias, lb, lo, hi, ent_coeff = ctx.cfg
+        empty = head.new_empty(0, dtype=torch.float32)
+        gobj = g_obj.contiguous() if g_obj is not None else empty
+        gent = g_ent.contiguous() if g_ent is not None else empty
+        dhead = _C.ppo_head_bwd(
+            head, action, eps, prev_lp, adv, stats, gobj, gent,
+            sp_bias, lb, lo, hi, ent_coeff,
+        )
+        return (dhead, None, None, None, None, None, None, None, None, None,
+                None)
+
+
+def ppo_head_loss(head, action, prev_log_prob, advantage, eps, *, sp_bias,
+                  scale_lb, lo, hi, entropy_coeff, normalize):
+    """Fused ClipPPO actor losses straight from the raw policy-head
+    output (``[N, 2A]`` = loc | pre-softplus scale).  Returns
+    ``(loss_objective, loss_entropy, entropy_mean, ESS_per_sample,
+    clip_fraction)``; the first two carry gradient to ``head``."""
+    _require_ext()
+    return _PPOHeadLossFn.apply(
+        head, action, prev_log_prob, advantage, eps, float(sp_bias),
+        float(scale_lb), float(lo), float(hi), float(entropy_coeff),
+        bool(normalize),
+    )
+
+
 class _SmoothL1MeanFn(torch.autograd.Function):
     """Fused mean smooth-L1 (beta=1) critic loss (csrc/loss_ops.hip):
     one partials + one finalize launch forward, one analytic backward

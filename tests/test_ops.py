@@ -733,3 +733,86 @@ class TestSmoothL1Fused:
         e.backward()
         atol = 1e-6 if dtype == torch.float32 else 1e-2
         assert torch.allclose(v.grad.float(), v2.grad.float(), atol=atol)
+
+
+@pytest.mark.gpu
+class TestPPOHeadMega:
+    """csrc/loss_ops.hip mega-fused head loss vs the eager chain."""
+
+    def _eager(self, head, action, prev, adv, eps, bias, lb, lo, hi, coeff,
+               normalize):
+        A = action.shape[-1]
+        loc, sraw = head.float().chunk(2, -1)
+        scale = torch.nn.functional.softplus(sraw + bias).clamp_min(lb)
+        if normalize:
+            adv = (adv - adv.mean()) / adv.std().clamp_min(1e-6)
+        lim = 1.0 - 1.1920929e-7
+        y = action.clamp(-lim, lim)
+        u = torch.atanh(y)
+        z = (u - loc) / scale
+        jac = 2.0 * (math.log(2.0) - u - torch.nn.functional.softplus(-2 * u))
+        lp = (-0.5 * z * z - scale.log() - 0.5 * math.log(2 * math.pi) - jac).sum(-1)
+        lw = lp - prev
+        ratio = lw.exp()
+        rc = lw.clamp(lo, hi).exp()
+        gain = torch.minimum(ratio * adv, rc * adv)
+        x = torch.tanh(loc + scale * eps)
+        ent = (0.5 * eps * eps + scale.log() + 0.5 * math.log(2 * math.pi)
+               + torch.log1p(-x * x)).sum(-1)
+        ess = lw.exp().sum().pow(2) / lw.mul(2).exp().sum().clamp_min(1e-12)
+        cf = (rc != ratio).float().mean()
+        return (-gain.mean(), -coeff * ent.mean(), ent.mean().detach(),
+                ess / lw.numel(), cf)
+
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    @pytest.mark.parametrize("normalize", [False, True])
+    def test_matches_eager(self, dtype, normalize):
+        from rl_amd import ops
+
+        torch.manual_seed(0)
+        N, A = 2049, 6
+        head = (0.4 * torch.randn(N, 2 * A, device="cuda")).to(dtype).requires_grad_()
+        action = (torch.rand(N, A, device="cuda") * 1.8 - 0.9)
+        prev = torch.randn(N, device="cuda") * 0.1
+        adv = torch.randn(N, device="cuda")
+        eps = torch.randn(N, A, device="cuda")
+        bias, lb = 0.5413248546129181, 1e-4
+        lo, hi = math.log1p(-0.2), math.log1p(0.2)
+        o_f = ops.ppo_head_loss(head, action, prev, adv, eps, sp_bias=bias,
+                                scale_lb=lb, lo=lo, hi=hi, entropy_coeff=0.01,
+                                normalize=normalize)
+        head2 = head.detach().clone().requires_grad_()
+        o_e = self._eager(head2, action, prev, adv, eps, bias, lb, lo, hi,
+                          0.01, normalize)
+        tol = 1e-5 if dtype == torch.float32 else 5e-3
+        for i, (a, b) in enumerate(zip(o_f, o_e)):
+            assert torch.allclose(a, b, atol=tol, rtol=1e-3), (i, a, b)
+        (o_f[0] + o_f[1]).backward()
+        (o_e[0] + o_e[1]).backward()
+        gtol = 1e-5 if dtype == torch.float32 else 1e-2
+        assert torch.allclose(head.grad.float(), head2.grad.float(),
+                              atol=gtol), (head.grad - head2.grad).abs().max()
+
+    def test_objective_only_grad(self):
+        """g_ent=None path (autograd.grad on loss_objective alone)."""
+        from rl_amd import ops
+
+        torch.manual_seed(1)
+        N, A = 513, 3
+        head = (0.3 * torch.randn(N, 2 * A, device="cuda")).requires_grad_()
+        action = torch.rand(N, A, device="cuda") * 1.6 - 0.8
+        prev = torch.randn(N, device="cuda") * 0.1
+        adv = torch.randn(N, device="cuda")
+        eps = torch.randn(N, A, device="cuda")
+        lo, hi = math.log1p(-0.2), math.log1p(0.2)
+        o = ops.ppo_head_loss(head, action, prev, adv, eps,
+                              sp_bias=0.5413248546129181, scale_lb=1e-4,
+                              lo=lo, hi=hi, entropy_coeff=0.01,
+                              normalize=True)
+        (g,) = torch.autograd.grad(o[0], head)
+        head2 = head.detach().clone().requires_grad_()
+        o_e = TestPPOHeadMega._eager(self, head2, action, prev, adv, eps,
+                                     0.5413248546129181, 1e-4, lo, hi, 0.01,
+                                     True)
+        (g_e,) = torch.autograd.grad(o_e[0], head2)
+        assert torch.allclose(g, g_e, atol=1e-5), (g - g_e).abs().max()
