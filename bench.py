@@ -158,6 +158,12 @@ def run_ppo(args):
             # under graph capture; r29 profile: ~130 cast kernels/step)
             enable_splitk_bf16_cache(actor)
             enable_splitk_bf16_cache(critic)
+            # whole-MLP MFMA kernels for the update phase: 3 GEMMs +
+            # bias + tanh in one launch each way (ops.FusedMLP3)
+            from rl_amd.ops import fuse_mlp3
+
+            actor.module[0].module[0] = fuse_mlp3(actor.module[0].module[0])
+            critic.module = fuse_mlp3(critic.module)
 
             def refresh_hook():
                 refresh_splitk_caches(actor, critic)

@@ -80,6 +80,14 @@ void launch_ppo_head_bwd(const void*, const float*, const float*,
                          const float*, const float*, const float*,
                          const float*, const float*, void*, float, float,
                          float, float, float, long, int, int, void*);
+int mlp3_mfma_lds_bytes(int, int, int);
+void launch_mlp3_mfma_fwd(const void*, int, const void*, const void*,
+                          const void*, const void*, const void*, const void*,
+                          void*, void*, void*, void*, long, int, int, int,
+                          void*);
+void launch_mlp3_mfma_bwd(const void*, int, const void*, const void*,
+                          const void*, const void*, void*, void*, long, int,
+                          int, void*);
 void launch_synthetic_env_step(float*, const float*, const float*,
                                const float*, float*, float*, float*, float*,
                                bool*, const float*, long, long, long, int,
@@ -694,6 +702,63 @@ torch::Tensor smooth_l1_bwd(torch::Tensor v, torch::Tensor t,
                        (void*)stream);
   return dv;
 }
+// MFMA whole-MLP forward/backward (csrc/fused_mlp.hip v2): three
+// matrix-core GEMMs + biases + tanh in one launch; dgrad chain + tanh'
+// in one launch.  Replaces the eager hipBLASLt GEMM + tanh + cast
+// chains of the PPO update phase.
+bool mlp3_mfma_ok(long O, long H, long A2) {
+  return H % 16 == 0 && H >= 16 && H <= 512 && O >= 1 && A2 >= 1 &&
+         mlp3_mfma_lds_bytes((int)O, (int)H, (int)A2) <= 160 * 1024;
+}
+
+std::vector<torch::Tensor> mlp3_mfma_fwd(torch::Tensor x, torch::Tensor w1,
+                                         torch::Tensor b1, torch::Tensor w2,
+                                         torch::Tensor b2, torch::Tensor w3,
+                                         torch::Tensor b3) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x cuda contiguous");
+  const bool xf32 = x.scalar_type() == torch::kFloat32;
+  TORCH_CHECK(xf32 || x.scalar_type() == torch::kBFloat16, "x fp32/bf16");
+  TORCH_CHECK(w1.scalar_type() == torch::kBFloat16 &&
+                  w2.scalar_type() == torch::kBFloat16 &&
+                  w3.scalar_type() == torch::kBFloat16,
+              "bf16 weights");
+  const long N = x.size(0), O = x.size(1);
+  const long H = w1.size(0), A2 = w3.size(0);
+  TORCH_CHECK(w1.size(1) == O && w2.size(0) == H && w2.size(1) == H &&
+                  w3.size(1) == H,
+              "weight shape mismatch");
+  TORCH_CHECK(mlp3_mfma_ok(O, H, A2), "mlp3_mfma: unsupported dims");
+  auto bopt = w1.options();
+  auto out = torch::empty({N, A2}, bopt);
+  auto h1 = torch::empty({N, H}, bopt);
+  auto h2 = torch::empty({N, H}, bopt);
+  auto xb = xf32 ? torch::empty({N, O}, bopt) : x;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_mlp3_mfma_fwd(x.data_ptr(), xf32 ? 1 : 0, w1.data_ptr(),
+                       b1.data_ptr(), w2.data_ptr(), b2.data_ptr(),
+                       w3.data_ptr(), b3.data_ptr(), out.data_ptr(),
+                       h1.data_ptr(), h2.data_ptr(), xb.data_ptr(), N, (int)O,
+                       (int)H, (int)A2, (void*)stream);
+  return {out, h1, h2, xb};
+}
+
+std::vector<torch::Tensor> mlp3_mfma_bwd(torch::Tensor dout, torch::Tensor h1,
+                                         torch::Tensor h2, torch::Tensor w2,
+                                         torch::Tensor w3) {
+  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous(), "dout cuda contiguous");
+  const bool df32 = dout.scalar_type() == torch::kFloat32;
+  const long N = dout.size(0), A2 = dout.size(1), H = h1.size(1);
+  auto bopt = h1.options();
+  auto dh1 = torch::empty({N, H}, bopt);
+  auto dh2 = torch::empty({N, H}, bopt);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_mlp3_mfma_bwd(dout.data_ptr(), df32 ? 1 : 0, h1.data_ptr(),
+                       h2.data_ptr(), w2.data_ptr(), w3.data_ptr(),
+                       dh1.data_ptr(), dh2.data_ptr(), N, (int)H, (int)A2,
+                       (void*)stream);
+  return {dh1, dh2};
+}
+
 // Mega-fused TanhNormal head loss: raw actor-head output [N, 2A] ->
 // (out[5] = {loss_objective, ESS/N, clip_fraction, entropy_mean,
 // loss_entropy}, stats).  See csrc/loss_ops.hip.
@@ -812,6 +877,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused mean smooth-L1 critic loss forward (HIP)");
   m.def("smooth_l1_bwd", &smooth_l1_bwd,
         "fused smooth-L1 backward (HIP)");
+  m.def("mlp3_mfma_ok", &mlp3_mfma_ok, "MFMA MLP3 shape eligibility");
+  m.def("mlp3_mfma_fwd", &mlp3_mfma_fwd,
+        "MFMA whole-MLP forward: 3 GEMMs + tanh in one launch (HIP)");
+  m.def("mlp3_mfma_bwd", &mlp3_mfma_bwd,
+        "MFMA dgrad chain + tanh' in one launch (HIP)");
   m.def("ppo_head_fwd", &ppo_head_fwd,
         "mega-fused TanhNormal head -> PPO losses forward (HIP)");
   m.def("ppo_head_bwd", &ppo_head_bwd,

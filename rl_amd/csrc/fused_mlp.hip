@@ -221,3 +221,310 @@ extern "C" void launch_mlp3_bwd(const void* dout, const void* h1,
                      (const __hip_bfloat16*)w2, (const __hip_bfloat16*)w3,
                      (__hip_bfloat16*)dh1, (__hip_bfloat16*)dh2, N, H, A2);
 }
+
+// ---------------------------------------------------------------------------
+// MFMA whole-MLP kernels (v2).  The VALU kernels above win only for
+// launch-bound small batches; at the PPO update's 16K-row minibatches
+// their dot loops lose to hipBLASLt (measured r37: T=64 7.49 ms vs
+// 5.18).  These variants put the three GEMMs on the matrix cores
+// (v_mfma_f32_16x16x32_bf16, fragment mapping hardware-verified in
+// benchmarks/mfma_probe.hip) with the whole layer chain fused:
+//
+//  * mlp3_mfma_fwd_kernel — 64 rows per workgroup (grid = N/64: 256+
+//    workgroups at bench sizes — fills the 8 XCDs), weights staged in
+//    LDS once.  W1/W2/W3 are [out,in] row-major, which IS the B^T
+//    fragment layout: b[reg] = W[j][k0+reg] is a single 16-byte LDS
+//    read per fragment (rows padded to 16-byte alignment).  tanh +
+//    bias run in registers; h1/h2 (bf16) go to LDS for the next layer
+//    and to global for the backward; fp32 input is converted during
+//    staging (kills the eager obs->bf16 cast kernel).
+//  * mlp3_mfma_bwd_kernel — the dgrad chain in one launch; W2/W3 are
+//    staged TRANSPOSED so dX = dY @ W also reads single-b128 B
+//    fragments; tanh' multiplies against the saved activations.
+//
+// Eager semantics matched: bf16 inputs/weights, fp32 accumulation,
+// fp32 tanh, bf16 activations (= autocast's eager rounding).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+using mfrag_b = __attribute__((ext_vector_type(8))) short;
+using mfrag_f = __attribute__((ext_vector_type(4))) float;
+
+#define M3_ROWS 64       // rows per workgroup (4 waves x 16)
+#define M3_MAXCT 8       // max col tiles per layer (H <= 128)
+
+__device__ __forceinline__ int m3_pad32(int k) { return (k + 31) & ~31; }
+
+// one 16x16 output tile over the full K: A rows from s_a (stride lda,
+// 16B-aligned), B rows = weight rows j (stride ldb) — both b128 reads.
+__device__ __forceinline__ void m3_gemm_tile(
+    const __hip_bfloat16* s_a, int lda, const __hip_bfloat16* s_w, int ldb,
+    int j0, int Kp, int lane, mfrag_f* acc) {
+  const int row = lane & 15;
+  const int koff = 8 * (lane >> 4);
+  for (int kc = 0; kc < Kp; kc += 32) {
+    const mfrag_b a = *reinterpret_cast<const mfrag_b*>(
+        &s_a[(size_t)row * lda + kc + koff]);
+    const mfrag_b b = *reinterpret_cast<const mfrag_b*>(
+        &s_w[(size_t)(j0 + row) * ldb + kc + koff]);
+    *acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, *acc, 0, 0, 0);
+  }
+}
+
+template <typename TX>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
+    const TX* __restrict__ x,               // [N, O]
+    const __hip_bfloat16* __restrict__ w1,  // [H, O]
+    const __hip_bfloat16* __restrict__ b1,  // [H]
+    const __hip_bfloat16* __restrict__ w2,  // [H, H]
+    const __hip_bfloat16* __restrict__ b2,  // [H]
+    const __hip_bfloat16* __restrict__ w3,  // [A2, H]
+    const __hip_bfloat16* __restrict__ b3,  // [A2]
+    __hip_bfloat16* __restrict__ out,       // [N, A2]
+    __hip_bfloat16* __restrict__ h1_out,    // [N, H]
+    __hip_bfloat16* __restrict__ h2_out,    // [N, H]
+    __hip_bfloat16* __restrict__ xb_out,    // [N, O] bf16 copy (wgrad)
+    const int N, const int O, const int H, const int A2) {
+  extern __shared__ __hip_bfloat16 smem[];
+  const int Op = m3_pad32(O), Hp = m3_pad32(H);
+  const int lx = Op + 8, lh = Hp + 8;
+  const int A2p = (A2 + 15) & ~15;
+  __hip_bfloat16* s_x = smem;                  // [64][lx]
+  __hip_bfloat16* s_w1 = s_x + M3_ROWS * lx;   // [H][lx]
+  __hip_bfloat16* s_h1 = s_w1 + H * lx;        // [64][lh]
+  __hip_bfloat16* s_w2 = s_h1 + M3_ROWS * lh;  // [H][lh]
+  __hip_bfloat16* s_h2 = s_w2 + H * lh;        // [64][lh]
+  __hip_bfloat16* s_w3 = s_h2 + M3_ROWS * lh;  // [A2p][lh]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const long row0 = (long)blockIdx.x * M3_ROWS;
+  const int rows = (int)min((long)M3_ROWS, (long)N - row0);
+
+  for (int i = tid; i < M3_ROWS * lx; i += MLP_THREADS) {
+    const int r = i / lx, k = i % lx;
+    __hip_bfloat16 v = __hip_bfloat16(0.f);
+    if (r < rows && k < O) v = __hip_bfloat16((float)x[(row0 + r) * O + k]);
+    s_x[i] = v;
+    if (r < rows && k < O) xb_out[(row0 + r) * O + k] = v;
+  }
+  for (int i = tid; i < H * lx; i += MLP_THREADS) {
+    const int j = i / lx, k = i % lx;
+    s_w1[i] = (k < O) ? w1[(size_t)j * O + k] : __hip_bfloat16(0.f);
+  }
+  for (int i = tid; i < H * lh; i += MLP_THREADS) {
+    const int j = i / lh, k = i % lh;
+    s_w2[i] = (k < H) ? w2[(size_t)j * H + k] : __hip_bfloat16(0.f);
+  }
+  for (int i = tid; i < A2p * lh; i += MLP_THREADS) {
+    const int j = i / lh, k = i % lh;
+    s_w3[i] = (j < A2 && k < H) ? w3[(size_t)j * H + k] : __hip_bfloat16(0.f);
+  }
+  __syncthreads();
+
+  const int erow = (lane >> 4) * 4;  // epilogue row base (+reg)
+  const int ecol = lane & 15;
+  // layer 1 + tanh
+  for (int ct = 0; ct < H / 16; ++ct) {
+    mfrag_f acc = {};
+    m3_gemm_tile(&s_x[(size_t)wave * 16 * lx], lx, s_w1, lx, ct * 16, Op,
+                 lane, &acc);
+    const int col = ct * 16 + ecol;
+    const float bias = __bfloat162float(b1[col]);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = wave * 16 + erow + r;
+      const __hip_bfloat16 h = __hip_bfloat16(tanhf(acc[r] + bias));
+      s_h1[(size_t)row * lh + col] = h;
+      if (row < rows) h1_out[(row0 + row) * H + col] = h;
+    }
+  }
+  // zero the k-pad of h tiles once (Hp > H only when H % 32 == 16)
+  for (int i = tid; i < M3_ROWS; i += MLP_THREADS)
+    for (int k = H; k < Hp; ++k) {
+      s_h1[(size_t)i * lh + k] = __hip_bfloat16(0.f);
+      s_h2[(size_t)i * lh + k] = __hip_bfloat16(0.f);
+    }
+  __syncthreads();
+  // layer 2 + tanh
+  for (int ct = 0; ct < H / 16; ++ct) {
+    mfrag_f acc = {};
+    m3_gemm_tile(&s_h1[(size_t)wave * 16 * lh], lh, s_w2, lh, ct * 16, Hp,
+                 lane, &acc);
+    const int col = ct * 16 + ecol;
+    const float bias = __bfloat162float(b2[col]);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = wave * 16 + erow + r;
+      const __hip_bfloat16 h = __hip_bfloat16(tanhf(acc[r] + bias));
+      s_h2[(size_t)row * lh + col] = h;
+      if (row < rows) h2_out[(row0 + row) * H + col] = h;
+    }
+  }
+  __syncthreads();
+  // layer 3 (head)
+  for (int ct = 0; ct < A2p / 16; ++ct) {
+    mfrag_f acc = {};
+    m3_gemm_tile(&s_h2[(size_t)wave * 16 * lh], lh, s_w3, lh, ct * 16, Hp,
+                 lane, &acc);
+    const int col = ct * 16 + ecol;
+    if (col < A2) {
+      const float bias = __bfloat162float(b3[col]);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = wave * 16 + erow + r;
+        if (row < rows)
+          out[(row0 + row) * A2 + col] = __hip_bfloat16(acc[r] + bias);
+      }
+    }
+  }
+}
+
+template <typename TD>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
+    const TD* __restrict__ dout,            // [N, A2]
+    const __hip_bfloat16* __restrict__ h1,  // [N, H]
+    const __hip_bfloat16* __restrict__ h2,  // [N, H]
+    const __hip_bfloat16* __restrict__ w2,  // [H, H]
+    const __hip_bfloat16* __restrict__ w3,  // [A2, H]
+    __hip_bfloat16* __restrict__ dh1_out,   // [N, H]
+    __hip_bfloat16* __restrict__ dh2_out,   // [N, H]
+    const int N, const int H, const int A2) {
+  extern __shared__ __hip_bfloat16 smem[];
+  const int A2p32 = m3_pad32(A2), Hp = m3_pad32(H);
+  const int ld = A2p32 + 8, lh = Hp + 8;
+  __hip_bfloat16* s_dy = smem;                   // [64][ld]
+  __hip_bfloat16* s_w3t = s_dy + M3_ROWS * ld;   // [H][ld]  w3t[j][k]=w3[k][j]
+  __hip_bfloat16* s_dh2 = s_w3t + H * ld;        // [64][lh]
+  __hip_bfloat16* s_w2t = s_dh2 + M3_ROWS * lh;  // [H][lh]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const long row0 = (long)blockIdx.x * M3_ROWS;
+  const int rows = (int)min((long)M3_ROWS, (long)N - row0);
+
+  for (int i = tid; i < M3_ROWS * ld; i += MLP_THREADS) {
+    const int r = i / ld, k = i % ld;
+    s_dy[i] = (r < rows && k < A2)
+                  ? __hip_bfloat16((float)dout[(row0 + r) * A2 + k])
+                  : __hip_bfloat16(0.f);
+  }
+  for (int i = tid; i < H * ld; i += MLP_THREADS) {
+    const int j = i / ld, k = i % ld;
+    s_w3t[i] = (k < A2) ? w3[(size_t)k * H + j] : __hip_bfloat16(0.f);
+  }
+  for (int i = tid; i < H * lh; i += MLP_THREADS) {
+    const int j = i / lh, k = i % lh;
+    s_w2t[i] = (k < H) ? w2[(size_t)k * H + j] : __hip_bfloat16(0.f);
+  }
+  __syncthreads();
+
+  const int erow = (lane >> 4) * 4;
+  const int ecol = lane & 15;
+  // dh2 = (dY @ W3) * (1 - h2^2)
+  for (int ct = 0; ct < H / 16; ++ct) {
+    mfrag_f acc = {};
+    m3_gemm_tile(&s_dy[(size_t)wave * 16 * ld], ld, s_w3t, ld, ct * 16,
+                 A2p32, lane, &acc);
+    const int col = ct * 16 + ecol;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = wave * 16 + erow + r;
+      float g = 0.f;
+      if (row < rows) {
+        const float hv = __bfloat162float(h2[(row0 + row) * H + col]);
+        g = acc[r] * (1.f - hv * hv);
+        dh2_out[(row0 + row) * H + col] = __hip_bfloat16(g);
+      }
+      s_dh2[(size_t)row * lh + col] = __hip_bfloat16(g);
+    }
+  }
+  for (int i = tid; i < M3_ROWS; i += MLP_THREADS)
+    for (int k = H; k < Hp; ++k) s_dh2[(size_t)i * lh + k] = __hip_bfloat16(0.f);
+  __syncthreads();
+  // dh1 = (dh2 @ W2) * (1 - h1^2)
+  for (int ct = 0; ct < H / 16; ++ct) {
+    mfrag_f acc = {};
+    m3_gemm_tile(&s_dh2[(size_t)wave * 16 * lh], lh, s_w2t, lh, ct * 16, Hp,
+                 lane, &acc);
+    const int col = ct * 16 + ecol;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = wave * 16 + erow + r;
+      if (row < rows) {
+        const float hv = __bfloat162float(h1[(row0 + row) * H + col]);
+        dh1_out[(row0 + row) * H + col] =
+            __hip_bfloat16(acc[r] * (1.f - hv * hv));
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" int mlp3_mfma_lds_bytes(int O, int H, int A2) {
+  const int Op = (O + 31) & ~31, Hp = (H + 31) & ~31;
+  const int lx = Op + 8, lh = Hp + 8, A2p = (A2 + 15) & ~15;
+  const int ld = ((A2 + 31) & ~31) + 8;
+  const int fwd = 2 * (M3_ROWS * lx + H * lx + 2 * M3_ROWS * lh + H * lh +
+                       A2p * lh);
+  const int bwd = 2 * (M3_ROWS * ld + H * ld + M3_ROWS * lh + H * lh);
+  return fwd > bwd ? fwd : bwd;
+}
+
+extern "C" void launch_mlp3_mfma_fwd(const void* x, int x_is_f32,
+                                     const void* w1, const void* b1,
+                                     const void* w2, const void* b2,
+                                     const void* w3, const void* b3,
+                                     void* out, void* h1, void* h2, void* xb,
+                                     long N, int O, int H, int A2,
+                                     void* stream) {
+  const int blocks = (int)((N + M3_ROWS - 1) / M3_ROWS);
+  const int Op = (O + 31) & ~31, Hp = (H + 31) & ~31;
+  const int lx = Op + 8, lh = Hp + 8, A2p = (A2 + 15) & ~15;
+  const int lds =
+      2 * (M3_ROWS * lx + H * lx + 2 * M3_ROWS * lh + H * lh + A2p * lh);
+  if (x_is_f32)
+    hipLaunchKernelGGL(mlp3_mfma_fwd_kernel<float>, dim3(blocks),
+                       dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                       (const float*)x, (const __hip_bfloat16*)w1,
+                       (const __hip_bfloat16*)b1, (const __hip_bfloat16*)w2,
+                       (const __hip_bfloat16*)b2, (const __hip_bfloat16*)w3,
+                       (const __hip_bfloat16*)b3, (__hip_bfloat16*)out,
+                       (__hip_bfloat16*)h1, (__hip_bfloat16*)h2,
+                       (__hip_bfloat16*)xb, (int)N, O, H, A2);
+  else
+    hipLaunchKernelGGL(mlp3_mfma_fwd_kernel<__hip_bfloat16>, dim3(blocks),
+                       dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)w1,
+                       (const __hip_bfloat16*)b1, (const __hip_bfloat16*)w2,
+                       (const __hip_bfloat16*)b2, (const __hip_bfloat16*)w3,
+                       (const __hip_bfloat16*)b3, (__hip_bfloat16*)out,
+                       (__hip_bfloat16*)h1, (__hip_bfloat16*)h2,
+                       (__hip_bfloat16*)xb, (int)N, O, H, A2);
+}
+
+extern "C" void launch_mlp3_mfma_bwd(const void* dout, int d_is_f32,
+                                     const void* h1, const void* h2,
+                                     const void* w2, const void* w3,
+                                     void* dh1, void* dh2, long N, int H,
+                                     int A2, void* stream) {
+  const int blocks = (int)((N + M3_ROWS - 1) / M3_ROWS);
+  const int ld = ((A2 + 31) & ~31) + 8, lh = ((H + 31) & ~31) + 8;
+  const int lds = 2 * (M3_ROWS * ld + H * ld + M3_ROWS * lh + H * lh);
+  if (d_is_f32)
+    hipLaunchKernelGGL(mlp3_mfma_bwd_kernel<float>, dim3(blocks),
+                       dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                       (const float*)dout, (const __hip_bfloat16*)h1,
+                       (const __hip_bfloat16*)h2, (const __hip_bfloat16*)w2,
+                       (const __hip_bfloat16*)w3, (__hip_bfloat16*)dh1,
+                       (__hip_bfloat16*)dh2, (int)N, H, A2);
+  else
+    hipLaunchKernelGGL(mlp3_mfma_bwd_kernel<__hip_bfloat16>, dim3(blocks),
+                       dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                       (const __hip_bfloat16*)dout, (const __hip_bfloat16*)h1,
+                       (const __hip_bfloat16*)h2, (const __hip_bfloat16*)w2,
+                       (const __hip_bfloat16*)w3, (__hip_bfloat16*)dh1,
+                       (__hip_bfloat16*)dh2, (int)N, H, A2);
+}

@@ -782,6 +782,15 @@ class _FusedMLP3Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w1b, b1b, w2b, b2b, w3b, b3b, w1, b1, w2, b2, w3, b3):
         x = x.contiguous()
+        O, H, A2 = x.shape[-1], w1b.shape[0], w3b.shape[0]
+        if _C.mlp3_mfma_ok(O, H, A2):
+            # MFMA path: fp32 input converted during staging (the bf16
+            # copy for the wgrad comes back as xb)
+            out, h1, h2, xb = _C.mlp3_mfma_fwd(x, w1b, b1b, w2b, b2b, w3b, b3b)
+            ctx.mfma = True
+            ctx.save_for_backward(xb, h1, h2, w2b, w3b)
+            return out
+        ctx.mfma = False
         if x.dtype != torch.bfloat16:
             x = x.to(torch.bfloat16)
         out, h1, h2 = _C.mlp3_fwd(x, w1b, b1b, w2b, b2b, w3b, b3b)
@@ -792,9 +801,14 @@ class _FusedMLP3Fn(torch.autograd.Function):
     def backward(ctx, dout):
         x, h1, h2, w2b, w3b = ctx.saved_tensors
         dout = dout.contiguous()
-        if dout.dtype != torch.bfloat16:
-            dout = dout.to(torch.bfloat16)
-        dh1, dh2 = _C.mlp3_bwd(dout, h1, h2, w2b, w3b)
+        if ctx.mfma:
+            dh1, dh2 = _C.mlp3_mfma_bwd(dout, h1, h2, w2b, w3b)
+            if dout.dtype != torch.bfloat16:
+                dout = dout.to(torch.bfloat16)
+        else:
+            if dout.dtype != torch.bfloat16:
+                dout = dout.to(torch.bfloat16)
+            dh1, dh2 = _C.mlp3_bwd(dout, h1, h2, w2b, w3b)
         dw3, db3 = _C.wgrad_splitk(dout, h2, True)
         dw2, db2 = _C.wgrad_splitk(dh2, h1, True)
         dw1, db1 = _C.wgrad_splitk(dh1, x, True)
@@ -808,11 +822,13 @@ class FusedMLP3(torch.nn.Module):
     eager fallback elsewhere.  Built from three weight-sharing
     :class:`SplitKLinear` layers with ``enable_bf16_cache`` on.
 
-    .. note:: Wins only when the update is LAUNCH-bound (batches up to
-       a few thousand rows).  At the PPO bench's 16k-64k-row
-       minibatches hipBLASLt's MFMA GEMMs beat the kernel's VALU dots
-       (measured r37: T=64 7.49 ms vs 5.18) — the bench keeps the
-       eager GEMM path and uses only the MFMA wgrad."""
+    .. note:: Two kernel generations: the original VALU dot kernels
+       (small/odd shapes) and the MFMA v2 kernels (H % 16 == 0) that
+       run the three GEMMs on the matrix cores — the v2 path beats the
+       eager hipBLASLt chain at the PPO bench's 16k-row minibatches
+       because it folds the tanh/bias/cast launches into the GEMM
+       launch (the earlier VALU version lost there, measured r37:
+       T=64 7.49 ms vs 5.18)."""
 
     def __init__(self, lin1, lin2, lin3, eager: torch.nn.Module):
         super().__init__()
