@@ -81,6 +81,10 @@ void launch_ppo_head_bwd(const void*, const float*, const float*,
                          const float*, const float*, void*, float, float,
                          float, float, float, long, int, int, void*);
 int mlp3_mfma_lds_bytes(int, int, int);
+int wgrad3_slab_count(long);
+void launch_wgrad3(const void* const*, const void* const*, float* const*,
+                   float* const*, float* const*, float* const*, const int*,
+                   const int*, long, void*);
 void launch_mlp3_mfma_fwd(const void*, int, const void*, const void*,
                           const void*, const void*, const void*, const void*,
                           void*, void*, void*, void*, long, int, int, int,
@@ -702,6 +706,54 @@ torch::Tensor smooth_l1_bwd(torch::Tensor v, torch::Tensor t,
                        (void*)stream);
   return dv;
 }
+// Batched 3-layer wgrad (csrc/wgrad.hip): the three (dY, X) pairs of a
+// fused-MLP backward in one mfma + one reduce launch.  All dims <= 64
+// and a shared K are required (the PPO/critic MLP shape).
+std::vector<torch::Tensor> wgrad_splitk3(torch::Tensor dy0, torch::Tensor x0,
+                                         torch::Tensor dy1, torch::Tensor x1,
+                                         torch::Tensor dy2,
+                                         torch::Tensor x2) {
+  torch::Tensor dys[3] = {dy0, dy1, dy2};
+  torch::Tensor xs[3] = {x0, x1, x2};
+  const long K = dy0.size(0);
+  const int slabs = wgrad3_slab_count(K);
+  const void* dyp[3];
+  const void* xp[3];
+  float *partp[3], *biasp[3], *dwp[3], *dbp[3];
+  int N[3], M[3];
+  std::vector<torch::Tensor> out;
+  std::vector<torch::Tensor> keep;
+  auto fopt = dy0.options().dtype(torch::kFloat32);
+  for (int l = 0; l < 3; ++l) {
+    TORCH_CHECK(dys[l].scalar_type() == torch::kBFloat16 &&
+                    xs[l].scalar_type() == torch::kBFloat16,
+                "wgrad3: bf16 inputs");
+    TORCH_CHECK(dys[l].is_contiguous() && xs[l].is_contiguous(),
+                "wgrad3: contiguous");
+    TORCH_CHECK(dys[l].size(0) == K && xs[l].size(0) == K, "wgrad3: K");
+    N[l] = (int)dys[l].size(1);
+    M[l] = (int)xs[l].size(1);
+    TORCH_CHECK(N[l] <= 64 && M[l] <= 64, "wgrad3: dims <= 64");
+    auto dw = torch::empty({N[l], M[l]}, fopt);
+    auto db = torch::empty({N[l]}, fopt);
+    auto part = torch::empty({slabs, 64 * 64}, fopt);
+    auto bias_part = torch::empty({slabs, 64}, fopt);
+    dyp[l] = dys[l].data_ptr();
+    xp[l] = xs[l].data_ptr();
+    partp[l] = part.data_ptr<float>();
+    biasp[l] = bias_part.data_ptr<float>();
+    dwp[l] = dw.data_ptr<float>();
+    dbp[l] = db.data_ptr<float>();
+    out.push_back(dw);
+    out.push_back(db);
+    keep.push_back(part);
+    keep.push_back(bias_part);
+  }
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_wgrad3(dyp, xp, partp, biasp, dwp, dbp, N, M, K, (void*)stream);
+  return out;
+}
+
 // MFMA whole-MLP forward/backward (csrc/fused_mlp.hip v2): three
 // matrix-core GEMMs + biases + tanh in one launch; dgrad chain + tanh'
 // in one launch.  Replaces the eager hipBLASLt GEMM + tanh + cast
@@ -878,6 +930,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("smooth_l1_bwd", &smooth_l1_bwd,
         "fused smooth-L1 backward (HIP)");
   m.def("mlp3_mfma_ok", &mlp3_mfma_ok, "MFMA MLP3 shape eligibility");
+  m.def("wgrad_splitk3", &wgrad_splitk3,
+        "batched 3-layer split-K wgrad (HIP)");
   m.def("mlp3_mfma_fwd", &mlp3_mfma_fwd,
         "MFMA whole-MLP forward: 3 GEMMs + tanh in one launch (HIP)");
   m.def("mlp3_mfma_bwd", &mlp3_mfma_bwd,

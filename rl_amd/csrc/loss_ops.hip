@@ -360,7 +360,11 @@ __global__ void smooth_l1_bwd_k(const TV* __restrict__ v,
 }
 
 inline int red_blocks(long N) {
-  int b = (int)((N + LP_THREADS - 1) / LP_THREADS);
+  // one row per thread, but spread over >= 4x more workgroups than a
+  // dense packing so the 256-CU chip is filled even at 16K rows (a
+  // 64-WG launch left 3/4 of the XCDs idle — each row's transcendental
+  // chain is latency-bound, not ALU-bound)
+  int b = (int)((N + 63) / 64);
   return b < 1 ? 1 : (b > 256 ? 256 : b);
 }
 

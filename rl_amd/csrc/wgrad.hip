@@ -396,3 +396,168 @@ extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
                      (hipStream_t)stream, part, bias_part, dw, dbias, slabs,
                      tiles_n, tiles_m, N, M);
 }
+
+// ---------------------------------------------------------------------------
+// Batched 3-layer wgrad: the three (dY, X) weight-gradient pairs of one
+// fused-MLP backward (layers emitted together by mlp3_mfma_bwd) in ONE
+// mfma launch + ONE reduce launch instead of six.  All three layers
+// share K (the minibatch rows); dims are <= 64 (one output tile each),
+// which is exactly the PPO/critic MLP shape.  blockIdx.y selects the
+// layer; the slab target halves vs the single-layer path (grid is 3x
+// wider, so ~128 slabs x 3 layers still covers the 256 CUs) — the
+// partials buffer and its reduce traffic halve with it.
+// ---------------------------------------------------------------------------
+
+struct Wg3Args {
+  const __hip_bfloat16* dy[3];
+  const __hip_bfloat16* x[3];
+  float* part[3];       // [slabs, 64*64] each
+  float* bias_part[3];  // [slabs, 64] each
+  float* dw[3];
+  float* db[3];
+  int N[3];
+  int M[3];
+};
+
+namespace {
+
+__global__ void __launch_bounds__(WG_THREADS) wgrad3_mfma_kernel(
+    const Wg3Args args, long K, int k_slab) {
+  const int l = blockIdx.y;
+  const __hip_bfloat16* __restrict__ dy = args.dy[l];
+  const __hip_bfloat16* __restrict__ x = args.x[l];
+  const int N = args.N[l], M = args.M[l];
+  const long k_begin = (long)blockIdx.x * k_slab;
+  const long k_end = min(K, k_begin + (long)k_slab);
+
+  __shared__ __hip_bfloat16 s_dy[2][ROW_CHUNK][64 + 2];
+  __shared__ __hip_bfloat16 s_x[2][ROW_CHUNK][64 + 2];
+  __shared__ float s_bias[64];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int tile_n = (wave >> 1) * 32;
+  const int tile_m = (wave & 1) * 32;
+  f32_frag16 acc = {};
+  float bias_acc = 0.f;
+
+  const int lr = tid / 64;
+  const int lc = tid % 64;
+
+  auto stage = [&](long kc, int buf) {
+    const int rows = (int)min((long)ROW_CHUNK, k_end - kc);
+    for (int i = 0; i < ROW_CHUNK / 4; ++i) {
+      const int r = lr + 4 * i;
+      const long gk = kc + r;
+      __hip_bfloat16 dv = __hip_bfloat16(0.f), xv = __hip_bfloat16(0.f);
+      if (r < rows) {
+        if (lc < N) dv = dy[gk * N + lc];
+        if (lc < M) xv = x[gk * M + lc];
+      }
+      s_dy[buf][r][lc] = dv;
+      s_x[buf][r][lc] = xv;
+      bias_acc += __bfloat162float(dv);
+    }
+  };
+
+  int cur = 0;
+  stage(k_begin, 0);
+  __syncthreads();
+  const int n_off = tile_n + (lane & 31);
+  const int m_off = tile_m + (lane & 31);
+  for (long kc = k_begin; kc < k_end; kc += ROW_CHUNK) {
+    if (kc + ROW_CHUNK < k_end) stage(kc + ROW_CHUNK, 1 - cur);
+    for (int kk = 0; kk < ROW_CHUNK; kk += 16) {
+      bf16_frag a, b;
+#pragma unroll
+      for (int reg = 0; reg < 8; ++reg) {
+        const int k = kk + 8 * (lane >> 5) + reg;
+        a[reg] = *reinterpret_cast<const short*>(&s_dy[cur][k][n_off]);
+        b[reg] = *reinterpret_cast<const short*>(&s_x[cur][k][m_off]);
+      }
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+    }
+    __syncthreads();
+    cur = 1 - cur;
+  }
+
+  float* part = args.part[l] + (long)blockIdx.x * (64 * 64);
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int n_loc = tile_n + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+    const int m_loc = tile_m + (lane & 31);
+    part[(long)n_loc * 64 + m_loc] = acc[reg];
+  }
+  if (lr == 0) s_bias[lc] = 0.f;
+  __syncthreads();
+  atomicAdd(&s_bias[lc], bias_acc);
+  __syncthreads();
+  if (lr == 0) args.bias_part[l][(long)blockIdx.x * 64 + lc] = s_bias[lc];
+}
+
+__global__ void wgrad3_reduce_kernel(const Wg3Args args, int slabs) {
+  const int l = blockIdx.y;
+  const int N = args.N[l], M = args.M[l];
+  const float* __restrict__ part = args.part[l];
+  const long NM = (long)N * M;
+  const long waves = ((long)gridDim.x * blockDim.x) >> 6;
+  const long wave_id = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  for (long e = wave_id; e < NM; e += waves) {
+    const int n = (int)(e / M), m = (int)(e % M);
+    const long off = (long)n * 64 + m;
+    float acc = 0.f;
+    for (int s = lane; s < slabs; s += 64)
+      acc += part[(long)s * (64 * 64) + off];
+#pragma unroll
+    for (int d = 32; d > 0; d >>= 1) acc += __shfl_down(acc, d);
+    if (lane == 0) args.dw[l][e] = acc;
+  }
+  for (long n = wave_id; n < N; n += waves) {
+    float acc = 0.f;
+    for (int s = lane; s < slabs; s += 64)
+      acc += args.bias_part[l][(long)s * 64 + n];
+#pragma unroll
+    for (int d = 32; d > 0; d >>= 1) acc += __shfl_down(acc, d);
+    if (lane == 0) args.db[l][n] = acc;
+  }
+}
+
+}  // namespace
+
+extern "C" int wgrad3_slab_count(long K) {
+  // ~128 slabs: x3 layers in grid.y still fills the 256 CUs
+  long k_slab = (K + 127) / 128;
+  k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
+  if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
+  return (int)((K + k_slab - 1) / k_slab);
+}
+
+extern "C" void launch_wgrad3(const void* const* dy, const void* const* x,
+                              float* const* part, float* const* bias_part,
+                              float* const* dw, float* const* db,
+                              const int* N, const int* M, long K,
+                              void* stream) {
+  Wg3Args a;
+  for (int l = 0; l < 3; ++l) {
+    a.dy[l] = (const __hip_bfloat16*)dy[l];
+    a.x[l] = (const __hip_bfloat16*)x[l];
+    a.part[l] = part[l];
+    a.bias_part[l] = bias_part[l];
+    a.dw[l] = dw[l];
+    a.db[l] = db[l];
+    a.N[l] = N[l];
+    a.M[l] = M[l];
+  }
+  const Wg3Args* args = &a;
+  long k_slab = (K + 127) / 128;
+  k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
+  if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
+  const int slabs = (int)((K + k_slab - 1) / k_slab);
+  hipLaunchKernelGGL(wgrad3_mfma_kernel, dim3(slabs, 3), dim3(WG_THREADS), 0,
+                     (hipStream_t)stream, *args, K, (int)k_slab);
+  const int blocks = 512;  // 3 * 512 WGs cover the chip for the reduce
+  hipLaunchKernelGGL(wgrad3_reduce_kernel, dim3(blocks, 3), dim3(256), 0,
+                     (hipStream_t)stream, *args, slabs);
+}

@@ -658,6 +658,7 @@ class _PPOClipFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, lw, adv, lo, hi, normalize):
+        ctx.set_materialize_grads(False)
         lwf = lw.contiguous().reshape(-1)
         advf = adv.contiguous().reshape(-1)
         out, stats = _C.ppo_clip_fwd(lwf, advf, lo, hi, normalize)
@@ -672,6 +673,8 @@ class _PPOClipFn(torch.autograd.Function):
     def backward(ctx, g_loss, g_ess, g_cf):
         lwf, advf, stats = ctx.saved_tensors
         lo, hi = ctx.bounds
+        if g_loss is None:
+            return None, None, None, None, None
         dlw = _C.ppo_clip_bwd(lwf, advf, stats, g_loss.contiguous(), lo, hi)
         return dlw.reshape(ctx.lw_shape), None, None, None, None
 
@@ -752,6 +755,7 @@ class _SmoothL1MeanFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, value, target):
+        ctx.set_materialize_grads(False)
         vf = value.contiguous().reshape(-1)
         tf = target.contiguous().reshape(-1)
         ctx.save_for_backward(vf, tf)
@@ -761,6 +765,8 @@ class _SmoothL1MeanFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, gout):
         vf, tf = ctx.saved_tensors
+        if gout is None:
+            return None, None
         dv = _C.smooth_l1_bwd(vf, tf, gout.contiguous())
         return dv.reshape(ctx.v_shape), None
 
@@ -809,9 +815,15 @@ class _FusedMLP3Fn(torch.autograd.Function):
             if dout.dtype != torch.bfloat16:
                 dout = dout.to(torch.bfloat16)
             dh1, dh2 = _C.mlp3_bwd(dout, h1, h2, w2b, w3b)
-        dw3, db3 = _C.wgrad_splitk(dout, h2, True)
-        dw2, db2 = _C.wgrad_splitk(dh2, h1, True)
-        dw1, db1 = _C.wgrad_splitk(dh1, x, True)
+        if max(dout.shape[1], h2.shape[1], x.shape[1]) <= 64:
+            # all three layer wgrads in one mfma + one reduce launch
+            dw3, db3, dw2, db2, dw1, db1 = _C.wgrad_splitk3(
+                dout, h2, dh2, h1, dh1, x
+            )
+        else:
+            dw3, db3 = _C.wgrad_splitk(dout, h2, True)
+            dw2, db2 = _C.wgrad_splitk(dh2, h1, True)
+            dw1, db1 = _C.wgrad_splitk(dh1, x, True)
         return (None, None, None, None, None, None, None,
                 dw1, db1, dw2, db2, dw3, db3)
 
