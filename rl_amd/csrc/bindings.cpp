@@ -82,6 +82,8 @@ void launch_ppo_head_bwd(const void*, const float*, const float*,
                          float, float, float, long, int, int, void*);
 int mlp3_mfma_lds_bytes(int, int, int);
 int wgrad3_slab_count(long);
+void launch_grad_clip_coef(const void*, int, float, float*, void*);
+void launch_multi_gather(const void*, int, const long*, long, void*);
 void launch_wgrad3(const void* const*, const void* const*, float* const*,
                    float* const*, float* const*, float* const*, const int*,
                    const int*, long, void*);
@@ -706,6 +708,65 @@ torch::Tensor smooth_l1_bwd(torch::Tensor v, torch::Tensor t,
                        (void*)stream);
   return dv;
 }
+// Fused gradient clipping (csrc/loss_ops.hip): one single-workgroup
+// kernel computes the global 2-norm over up to 32 fp32 gradients and
+// the clamped scale coefficient (torch's clip_grad_norm_ runs ~10
+// launches for the same).  Apply with torch._foreach_mul_(grads, coef).
+torch::Tensor fused_grad_clip_coef(std::vector<torch::Tensor> grads,
+                                   double max_norm) {
+  struct {
+    const float* g[32];
+    int len[32];
+  } args;
+  TORCH_CHECK(!grads.empty() && grads.size() <= 32, "1..32 gradients");
+  for (size_t i = 0; i < grads.size(); ++i) {
+    TORCH_CHECK(grads[i].is_cuda() &&
+                    grads[i].scalar_type() == torch::kFloat32 &&
+                    grads[i].is_contiguous(),
+                "fused_grad_clip: fp32 cuda contiguous");
+    args.g[i] = grads[i].data_ptr<float>();
+    args.len[i] = (int)grads[i].numel();
+  }
+  auto coef = torch::empty({}, grads[0].options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_grad_clip_coef(&args, (int)grads.size(), (float)max_norm,
+                        coef.data_ptr<float>(), (void*)stream);
+  return coef;
+}
+
+// Batched shuffle gather: dst[t][i] = src[t][perm[i]] for up to 8 fp32
+// row-major tensors sharing the leading dim (csrc/loss_ops.hip) — one
+// launch instead of one index kernel per tensordict key.
+std::vector<torch::Tensor> multi_gather(torch::Tensor perm,
+                                        std::vector<torch::Tensor> srcs) {
+  struct {
+    const float* src[8];
+    float* dst[8];
+    int w[8];
+  } args;
+  TORCH_CHECK(perm.is_cuda() && perm.scalar_type() == torch::kLong &&
+                  perm.is_contiguous(),
+              "perm int64 cuda");
+  const long n = perm.numel();
+  TORCH_CHECK(!srcs.empty() && srcs.size() <= 8, "1..8 tensors");
+  std::vector<torch::Tensor> out;
+  for (size_t i = 0; i < srcs.size(); ++i) {
+    auto& s = srcs[i];
+    TORCH_CHECK(s.is_cuda() && s.scalar_type() == torch::kFloat32 &&
+                    s.is_contiguous() && s.size(0) == n,
+                "multi_gather: fp32 cuda contiguous with shared dim 0");
+    auto d = torch::empty_like(s);
+    args.src[i] = s.data_ptr<float>();
+    args.dst[i] = d.data_ptr<float>();
+    args.w[i] = (int)(s.numel() / n);
+    out.push_back(d);
+  }
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_multi_gather(&args, (int)srcs.size(), perm.data_ptr<long>(), n,
+                      (void*)stream);
+  return out;
+}
+
 // Batched 3-layer wgrad (csrc/wgrad.hip): the three (dY, X) pairs of a
 // fused-MLP backward in one mfma + one reduce launch.  All dims <= 64
 // and a shared K are required (the PPO/critic MLP shape).
@@ -932,6 +993,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlp3_mfma_ok", &mlp3_mfma_ok, "MFMA MLP3 shape eligibility");
   m.def("wgrad_splitk3", &wgrad_splitk3,
         "batched 3-layer split-K wgrad (HIP)");
+  m.def("fused_grad_clip_coef", &fused_grad_clip_coef,
+        "single-kernel global grad-norm clip coefficient (HIP)");
+  m.def("multi_gather", &multi_gather,
+        "batched row gather over up to 8 tensors (HIP)");
   m.def("mlp3_mfma_fwd", &mlp3_mfma_fwd,
         "MFMA whole-MLP forward: 3 GEMMs + tanh in one launch (HIP)");
   m.def("mlp3_mfma_bwd", &mlp3_mfma_bwd,

@@ -251,8 +251,13 @@ namespace {
 using mfrag_b = __attribute__((ext_vector_type(8))) short;
 using mfrag_f = __attribute__((ext_vector_type(4))) float;
 
-#define M3_ROWS 64       // rows per workgroup (4 waves x 16)
-#define M3_MAXCT 8       // max col tiles per layer (H <= 128)
+// Rows per workgroup is a template parameter: 32 (2 row blocks x 16,
+// waves split row-block x col-parity) for launches that would not fill
+// the chip at 64 — a 64-row tile at 16K rows is 256 WGs = 1 WG/CU with
+// no latency hiding (measured 20 us/fwd); 64 (4 waves x 16 rows) once
+// N/64 >= 512 WGs, where the halved per-WG weight staging wins
+// (measured: 32-row tile cost ~5% at 65K rows).
+#define M3_ROWS_SWITCH 32768
 
 __device__ __forceinline__ int m3_pad32(int k) { return (k + 31) & ~31; }
 
@@ -272,7 +277,7 @@ __device__ __forceinline__ void m3_gemm_tile(
   }
 }
 
-template <typename TX>
+template <typename TX, int R>
 __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
     const TX* __restrict__ x,               // [N, O]
     const __hip_bfloat16* __restrict__ w1,  // [H, O]
@@ -291,18 +296,18 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
   const int lx = Op + 8, lh = Hp + 8;
   const int A2p = (A2 + 15) & ~15;
   __hip_bfloat16* s_x = smem;                  // [64][lx]
-  __hip_bfloat16* s_w1 = s_x + M3_ROWS * lx;   // [H][lx]
+  __hip_bfloat16* s_w1 = s_x + R * lx;   // [H][lx]
   __hip_bfloat16* s_h1 = s_w1 + H * lx;        // [64][lh]
-  __hip_bfloat16* s_w2 = s_h1 + M3_ROWS * lh;  // [H][lh]
+  __hip_bfloat16* s_w2 = s_h1 + R * lh;  // [H][lh]
   __hip_bfloat16* s_h2 = s_w2 + H * lh;        // [64][lh]
-  __hip_bfloat16* s_w3 = s_h2 + M3_ROWS * lh;  // [A2p][lh]
+  __hip_bfloat16* s_w3 = s_h2 + R * lh;  // [A2p][lh]
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  const long row0 = (long)blockIdx.x * M3_ROWS;
-  const int rows = (int)min((long)M3_ROWS, (long)N - row0);
+  const long row0 = (long)blockIdx.x * R;
+  const int rows = (int)min((long)R, (long)N - row0);
 
-  for (int i = tid; i < M3_ROWS * lx; i += MLP_THREADS) {
+  for (int i = tid; i < R * lx; i += MLP_THREADS) {
     const int r = i / lx, k = i % lx;
     __hip_bfloat16 v = __hip_bfloat16(0.f);
     if (r < rows && k < O) v = __hip_bfloat16((float)x[(row0 + r) * O + k]);
@@ -325,38 +330,41 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
 
   const int erow = (lane >> 4) * 4;  // epilogue row base (+reg)
   const int ecol = lane & 15;
+  const int rblk = (R == 64 ? wave : (wave & 1)) * 16;
+  const int ct0 = (R == 64 ? 0 : (wave >> 1));
+  const int cts = (R == 64 ? 1 : 2);
   // layer 1 + tanh
-  for (int ct = 0; ct < H / 16; ++ct) {
+  for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};
-    m3_gemm_tile(&s_x[(size_t)wave * 16 * lx], lx, s_w1, lx, ct * 16, Op,
+    m3_gemm_tile(&s_x[(size_t)rblk * lx], lx, s_w1, lx, ct * 16, Op,
                  lane, &acc);
     const int col = ct * 16 + ecol;
     const float bias = __bfloat162float(b1[col]);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wave * 16 + erow + r;
+      const int row = rblk + erow + r;
       const __hip_bfloat16 h = __hip_bfloat16(tanhf(acc[r] + bias));
       s_h1[(size_t)row * lh + col] = h;
       if (row < rows) h1_out[(row0 + row) * H + col] = h;
     }
   }
   // zero the k-pad of h tiles once (Hp > H only when H % 32 == 16)
-  for (int i = tid; i < M3_ROWS; i += MLP_THREADS)
+  for (int i = tid; i < R; i += MLP_THREADS)
     for (int k = H; k < Hp; ++k) {
       s_h1[(size_t)i * lh + k] = __hip_bfloat16(0.f);
       s_h2[(size_t)i * lh + k] = __hip_bfloat16(0.f);
     }
   __syncthreads();
   // layer 2 + tanh
-  for (int ct = 0; ct < H / 16; ++ct) {
+  for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};
-    m3_gemm_tile(&s_h1[(size_t)wave * 16 * lh], lh, s_w2, lh, ct * 16, Hp,
+    m3_gemm_tile(&s_h1[(size_t)rblk * lh], lh, s_w2, lh, ct * 16, Hp,
                  lane, &acc);
     const int col = ct * 16 + ecol;
     const float bias = __bfloat162float(b2[col]);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wave * 16 + erow + r;
+      const int row = rblk + erow + r;
       const __hip_bfloat16 h = __hip_bfloat16(tanhf(acc[r] + bias));
       s_h2[(size_t)row * lh + col] = h;
       if (row < rows) h2_out[(row0 + row) * H + col] = h;
@@ -364,16 +372,16 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
   }
   __syncthreads();
   // layer 3 (head)
-  for (int ct = 0; ct < A2p / 16; ++ct) {
+  for (int ct = ct0; ct < A2p / 16; ct += cts) {
     mfrag_f acc = {};
-    m3_gemm_tile(&s_h2[(size_t)wave * 16 * lh], lh, s_w3, lh, ct * 16, Hp,
+    m3_gemm_tile(&s_h2[(size_t)rblk * lh], lh, s_w3, lh, ct * 16, Hp,
                  lane, &acc);
     const int col = ct * 16 + ecol;
     if (col < A2) {
       const float bias = __bfloat162float(b3[col]);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = wave * 16 + erow + r;
+        const int row = rblk + erow + r;
         if (row < rows)
           out[(row0 + row) * A2 + col] = __hip_bfloat16(acc[r] + bias);
       }
@@ -381,7 +389,7 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
   }
 }
 
-template <typename TD>
+template <typename TD, int R>
 __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
     const TD* __restrict__ dout,            // [N, A2]
     const __hip_bfloat16* __restrict__ h1,  // [N, H]
@@ -395,16 +403,16 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
   const int A2p32 = m3_pad32(A2), Hp = m3_pad32(H);
   const int ld = A2p32 + 8, lh = Hp + 8;
   __hip_bfloat16* s_dy = smem;                   // [64][ld]
-  __hip_bfloat16* s_w3t = s_dy + M3_ROWS * ld;   // [H][ld]  w3t[j][k]=w3[k][j]
+  __hip_bfloat16* s_w3t = s_dy + R * ld;   // [H][ld]  w3t[j][k]=w3[k][j]
   __hip_bfloat16* s_dh2 = s_w3t + H * ld;        // [64][lh]
-  __hip_bfloat16* s_w2t = s_dh2 + M3_ROWS * lh;  // [H][lh]
+  __hip_bfloat16* s_w2t = s_dh2 + R * lh;  // [H][lh]
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  const long row0 = (long)blockIdx.x * M3_ROWS;
-  const int rows = (int)min((long)M3_ROWS, (long)N - row0);
+  const long row0 = (long)blockIdx.x * R;
+  const int rows = (int)min((long)R, (long)N - row0);
 
-  for (int i = tid; i < M3_ROWS * ld; i += MLP_THREADS) {
+  for (int i = tid; i < R * ld; i += MLP_THREADS) {
     const int r = i / ld, k = i % ld;
     s_dy[i] = (r < rows && k < A2)
                   ? __hip_bfloat16((float)dout[(row0 + r) * A2 + k])
@@ -422,15 +430,18 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
 
   const int erow = (lane >> 4) * 4;
   const int ecol = lane & 15;
+  const int rblk = (R == 64 ? wave : (wave & 1)) * 16;
+  const int ct0 = (R == 64 ? 0 : (wave >> 1));
+  const int cts = (R == 64 ? 1 : 2);
   // dh2 = (dY @ W3) * (1 - h2^2)
-  for (int ct = 0; ct < H / 16; ++ct) {
+  for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};
-    m3_gemm_tile(&s_dy[(size_t)wave * 16 * ld], ld, s_w3t, ld, ct * 16,
+    m3_gemm_tile(&s_dy[(size_t)rblk * ld], ld, s_w3t, ld, ct * 16,
                  A2p32, lane, &acc);
     const int col = ct * 16 + ecol;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wave * 16 + erow + r;
+      const int row = rblk + erow + r;
       float g = 0.f;
       if (row < rows) {
         const float hv = __bfloat162float(h2[(row0 + row) * H + col]);
@@ -440,18 +451,18 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
       s_dh2[(size_t)row * lh + col] = __hip_bfloat16(g);
     }
   }
-  for (int i = tid; i < M3_ROWS; i += MLP_THREADS)
+  for (int i = tid; i < R; i += MLP_THREADS)
     for (int k = H; k < Hp; ++k) s_dh2[(size_t)i * lh + k] = __hip_bfloat16(0.f);
   __syncthreads();
   // dh1 = (dh2 @ W2) * (1 - h1^2)
-  for (int ct = 0; ct < H / 16; ++ct) {
+  for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};
-    m3_gemm_tile(&s_dh2[(size_t)wave * 16 * lh], lh, s_w2t, lh, ct * 16, Hp,
+    m3_gemm_tile(&s_dh2[(size_t)rblk * lh], lh, s_w2t, lh, ct * 16, Hp,
                  lane, &acc);
     const int col = ct * 16 + ecol;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wave * 16 + erow + r;
+      const int row = rblk + erow + r;
       if (row < rows) {
         const float hv = __bfloat162float(h1[(row0 + row) * H + col]);
         dh1_out[(row0 + row) * H + col] =
@@ -464,12 +475,12 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
 }  // namespace
 
 extern "C" int mlp3_mfma_lds_bytes(int O, int H, int A2) {
+  const int R = 64;  // LDS bound checked for the larger tile
   const int Op = (O + 31) & ~31, Hp = (H + 31) & ~31;
   const int lx = Op + 8, lh = Hp + 8, A2p = (A2 + 15) & ~15;
   const int ld = ((A2 + 31) & ~31) + 8;
-  const int fwd = 2 * (M3_ROWS * lx + H * lx + 2 * M3_ROWS * lh + H * lh +
-                       A2p * lh);
-  const int bwd = 2 * (M3_ROWS * ld + H * ld + M3_ROWS * lh + H * lh);
+  const int fwd = 2 * (R * lx + H * lx + 2 * R * lh + H * lh + A2p * lh);
+  const int bwd = 2 * (R * ld + H * ld + R * lh + H * lh);
   return fwd > bwd ? fwd : bwd;
 }
 
@@ -480,29 +491,28 @@ extern "C" void launch_mlp3_mfma_fwd(const void* x, int x_is_f32,
                                      void* out, void* h1, void* h2, void* xb,
                                      long N, int O, int H, int A2,
                                      void* stream) {
-  const int blocks = (int)((N + M3_ROWS - 1) / M3_ROWS);
+  const int R = N >= M3_ROWS_SWITCH ? 64 : 32;
+  const int blocks = (int)((N + R - 1) / R);
   const int Op = (O + 31) & ~31, Hp = (H + 31) & ~31;
   const int lx = Op + 8, lh = Hp + 8, A2p = (A2 + 15) & ~15;
-  const int lds =
-      2 * (M3_ROWS * lx + H * lx + 2 * M3_ROWS * lh + H * lh + A2p * lh);
-  if (x_is_f32)
-    hipLaunchKernelGGL(mlp3_mfma_fwd_kernel<float>, dim3(blocks),
-                       dim3(MLP_THREADS), lds, (hipStream_t)stream,
-                       (const float*)x, (const __hip_bfloat16*)w1,
-                       (const __hip_bfloat16*)b1, (const __hip_bfloat16*)w2,
-                       (const __hip_bfloat16*)b2, (const __hip_bfloat16*)w3,
-                       (const __hip_bfloat16*)b3, (__hip_bfloat16*)out,
-                       (__hip_bfloat16*)h1, (__hip_bfloat16*)h2,
-                       (__hip_bfloat16*)xb, (int)N, O, H, A2);
-  else
-    hipLaunchKernelGGL(mlp3_mfma_fwd_kernel<__hip_bfloat16>, dim3(blocks),
-                       dim3(MLP_THREADS), lds, (hipStream_t)stream,
-                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)w1,
-                       (const __hip_bfloat16*)b1, (const __hip_bfloat16*)w2,
-                       (const __hip_bfloat16*)b2, (const __hip_bfloat16*)w3,
-                       (const __hip_bfloat16*)b3, (__hip_bfloat16*)out,
-                       (__hip_bfloat16*)h1, (__hip_bfloat16*)h2,
-                       (__hip_bfloat16*)xb, (int)N, O, H, A2);
+  const int lds = 2 * (R * lx + H * lx + 2 * R * lh + H * lh + A2p * lh);
+#define M3_LAUNCH_FWD(TX, RR)                                               \
+  hipLaunchKernelGGL((mlp3_mfma_fwd_kernel<TX, RR>), dim3(blocks),          \
+                     dim3(MLP_THREADS), lds, (hipStream_t)stream,           \
+                     (const TX*)x, (const __hip_bfloat16*)w1,               \
+                     (const __hip_bfloat16*)b1, (const __hip_bfloat16*)w2,  \
+                     (const __hip_bfloat16*)b2, (const __hip_bfloat16*)w3,  \
+                     (const __hip_bfloat16*)b3, (__hip_bfloat16*)out,       \
+                     (__hip_bfloat16*)h1, (__hip_bfloat16*)h2,              \
+                     (__hip_bfloat16*)xb, (int)N, O, H, A2)
+  if (x_is_f32) {
+    if (R == 64) M3_LAUNCH_FWD(float, 64);
+    else M3_LAUNCH_FWD(float, 32);
+  } else {
+    if (R == 64) M3_LAUNCH_FWD(__hip_bfloat16, 64);
+    else M3_LAUNCH_FWD(__hip_bfloat16, 32);
+  }
+#undef M3_LAUNCH_FWD
 }
 
 extern "C" void launch_mlp3_mfma_bwd(const void* dout, int d_is_f32,
@@ -510,21 +520,23 @@ extern "C" void launch_mlp3_mfma_bwd(const void* dout, int d_is_f32,
                                      const void* w2, const void* w3,
                                      void* dh1, void* dh2, long N, int H,
                                      int A2, void* stream) {
-  const int blocks = (int)((N + M3_ROWS - 1) / M3_ROWS);
+  const int R = N >= M3_ROWS_SWITCH ? 64 : 32;
+  const int blocks = (int)((N + R - 1) / R);
   const int ld = ((A2 + 31) & ~31) + 8, lh = ((H + 31) & ~31) + 8;
-  const int lds = 2 * (M3_ROWS * ld + H * ld + M3_ROWS * lh + H * lh);
-  if (d_is_f32)
-    hipLaunchKernelGGL(mlp3_mfma_bwd_kernel<float>, dim3(blocks),
-                       dim3(MLP_THREADS), lds, (hipStream_t)stream,
-                       (const float*)dout, (const __hip_bfloat16*)h1,
-                       (const __hip_bfloat16*)h2, (const __hip_bfloat16*)w2,
-                       (const __hip_bfloat16*)w3, (__hip_bfloat16*)dh1,
-                       (__hip_bfloat16*)dh2, (int)N, H, A2);
-  else
-    hipLaunchKernelGGL(mlp3_mfma_bwd_kernel<__hip_bfloat16>, dim3(blocks),
-                       dim3(MLP_THREADS), lds, (hipStream_t)stream,
-                       (const __hip_bfloat16*)dout, (const __hip_bfloat16*)h1,
-                       (const __hip_bfloat16*)h2, (const __hip_bfloat16*)w2,
-                       (const __hip_bfloat16*)w3, (__hip_bfloat16*)dh1,
-                       (__hip_bfloat16*)dh2, (int)N, H, A2);
+  const int lds = 2 * (R * ld + H * ld + R * lh + H * lh);
+#define M3_LAUNCH_BWD(TD, RR)                                               \
+  hipLaunchKernelGGL((mlp3_mfma_bwd_kernel<TD, RR>), dim3(blocks),          \
+                     dim3(MLP_THREADS), lds, (hipStream_t)stream,           \
+                     (const TD*)dout, (const __hip_bfloat16*)h1,            \
+                     (const __hip_bfloat16*)h2, (const __hip_bfloat16*)w2,  \
+                     (const __hip_bfloat16*)w3, (__hip_bfloat16*)dh1,       \
+                     (__hip_bfloat16*)dh2, (int)N, H, A2)
+  if (d_is_f32) {
+    if (R == 64) M3_LAUNCH_BWD(float, 64);
+    else M3_LAUNCH_BWD(float, 32);
+  } else {
+    if (R == 64) M3_LAUNCH_BWD(__hip_bfloat16, 64);
+    else M3_LAUNCH_BWD(__hip_bfloat16, 32);
+  }
+#undef M3_LAUNCH_BWD
 }

@@ -653,3 +653,78 @@ extern "C" void launch_ppo_head_bwd(const void* head, const float* action,
                        eps, prev, adv, stats, gobj, gent, (float*)dhead,
                        sp_bias, lb, lo, hi, ent_coeff, N, A);
 }
+
+// ---------------------------------------------------------------------------
+// Step-glue kernels: fused gradient clipping and multi-tensor gather.
+//
+// clip_grad_norm_ over the PPO nets' ~17K gradient elements is 10
+// launches in torch (foreach norm + stack + reduce + scalar chain +
+// foreach mul); here ONE single-workgroup kernel computes the global
+// norm and the clamped coefficient (the pointer table rides in the
+// kernel-argument struct — gradient addresses are stable across graph
+// replays), and torch._foreach_mul_ applies it.  The minibatch shuffle
+// gather (one index kernel per tensordict key) is one batched kernel.
+// ---------------------------------------------------------------------------
+
+struct ClipArgs {
+  const float* g[32];
+  int len[32];
+};
+
+struct GatherArgs {
+  const float* src[8];
+  float* dst[8];
+  int w[8];
+};
+
+namespace {
+
+__global__ void grad_clip_coef_k(const ClipArgs args, const int nt,
+                                 const float max_norm,
+                                 float* __restrict__ coef) {
+  float ss = 0.f;
+  for (int t = 0; t < nt; ++t) {
+    const float* __restrict__ p = args.g[t];
+    const int L = args.len[t];
+    for (int i = threadIdx.x; i < L; i += blockDim.x) {
+      const float v = p[i];
+      ss += v * v;
+    }
+  }
+  __shared__ float smem[8];
+  const float tot = block_sum(ss, smem);
+  if (threadIdx.x == 0) {
+    const float c = max_norm / (sqrtf(tot) + 1e-6f);
+    coef[0] = c < 1.f ? c : 1.f;
+  }
+}
+
+__global__ void multi_gather_k(const GatherArgs args,
+                               const long* __restrict__ perm, const long n) {
+  const int t = blockIdx.y;
+  const float* __restrict__ src = args.src[t];
+  float* __restrict__ dst = args.dst[t];
+  const int w = args.w[t];
+  const long total = n * w;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long r = i / w;
+    dst[i] = src[perm[r] * w + (i - r * w)];
+  }
+}
+
+}  // namespace
+
+extern "C" void launch_grad_clip_coef(const void* args, int nt,
+                                      float max_norm, float* coef,
+                                      void* stream) {
+  hipLaunchKernelGGL(grad_clip_coef_k, dim3(1), dim3(LP_THREADS), 0,
+                     (hipStream_t)stream, *(const ClipArgs*)args, nt,
+                     max_norm, coef);
+}
+
+extern "C" void launch_multi_gather(const void* args, int nt,
+                                    const long* perm, long n, void* stream) {
+  hipLaunchKernelGGL(multi_gather_k, dim3(256, nt), dim3(LP_THREADS), 0,
+                     (hipStream_t)stream, *(const GatherArgs*)args, perm, n);
+}

@@ -733,6 +733,58 @@ class _PPOHeadLossFn(torch.autograd.Function):
                 None)
 
 
+def fused_grad_clip_(parameters, max_norm: float) -> bool:
+    """clip_grad_norm_ in two launches (csrc/loss_ops.hip): one
+    single-WG kernel for the global norm + clamped coefficient, one
+    foreach multiply.  Returns False (caller should use torch's
+    clip_grad_norm_) when the gradients don't fit the fused kernel."""
+    if not HAS_HIP_EXT:
+        return False
+    grads = [p.grad for p in parameters if p.grad is not None]
+    if not grads or len(grads) > 32:
+        return False
+    for g in grads:
+        if not (g.is_cuda and g.dtype == torch.float32 and g.is_contiguous()):
+            return False
+    coef = _C.fused_grad_clip_coef(grads, float(max_norm))
+    torch._foreach_mul_(grads, coef)
+    return True
+
+
+def multi_gather_td(flat_td, perm: torch.Tensor):
+    """Shuffle-gather a flat TensorDict of fp32 cuda leaves in ONE
+    kernel (csrc/loss_ops.hip) instead of one index kernel per key.
+    Returns None when ineligible (caller falls back to flat_td[perm])."""
+    if not HAS_HIP_EXT:
+        return None
+    try:
+        items = list(flat_td.items())
+    except Exception:
+        return None
+    if not (1 <= len(items) <= 8):
+        return None
+    n = perm.numel()
+    srcs = []
+    for _, v in items:
+        if not (
+            isinstance(v, torch.Tensor)
+            and v.is_cuda
+            and v.dtype == torch.float32
+            and v.is_contiguous()
+            and v.dim() >= 1
+            and v.shape[0] == n
+        ):
+            return None
+        srcs.append(v)
+    outs = _C.multi_gather(perm.contiguous(), srcs)
+    from ..tensordict import TensorDict
+
+    return TensorDict(
+        {k: o for (k, _), o in zip(items, outs)}, batch_size=[n],
+        device=perm.device,
+    )
+
+
 def ppo_head_loss(head, action, prev_log_prob, advantage, eps, *, sp_bias,
                   scale_lb, lo, hi, entropy_coeff, normalize):
     """Fused ClipPPO actor losses straight from the raw policy-head

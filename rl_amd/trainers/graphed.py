@@ -196,23 +196,68 @@ class GraphedPPO:
         if self.reducer is not None:
             self.reducer.finalize()
         if self.max_grad_norm:
-            torch.nn.utils.clip_grad_norm_(self._params, self.max_grad_norm)
+            # single-kernel global-norm clip (csrc/loss_ops.hip);
+            # torch's foreach chain is ~10 launches for the same
+            from .. import ops
+
+            if not ops.fused_grad_clip_(self._params, self.max_grad_norm):
+                torch.nn.utils.clip_grad_norm_(self._params, self.max_grad_norm)
         self.optimizer.step()
         if self.post_optim_hook is not None:
             self.post_optim_hook()
+
+    def _minibatch_keys(self, flat: TensorDictBase):
+        """Keys the update phase actually reads (loss tensor_keys +
+        network in_keys); shuffling anything else is wasted gather
+        bandwidth.  None = shuffle everything (unknown loss shape)."""
+        if "_mb_keys" in self.__dict__:
+            return self.__dict__["_mb_keys"]
+        keys = None
+        try:
+            tk = self.loss_module.tensor_keys
+            keys = set()
+            for attr in ("advantage", "value_target", "action",
+                         "sample_log_prob", "value"):
+                k = getattr(tk, attr, None)
+                if k is not None:
+                    keys.add(k)
+            for net_attr in ("actor_network", "critic_network"):
+                net = getattr(self.loss_module, net_attr, None)
+                if net is not None:
+                    keys.update(net.in_keys)
+            present = set(flat.keys(include_nested=True, leaves_only=True))
+            keys = [k for k in keys if k in present]
+            needed_min = {"advantage", "action"}
+            if not needed_min.issubset({k if isinstance(k, str) else k[-1]
+                                        for k in keys}):
+                keys = None
+        except Exception:
+            keys = None
+        self.__dict__["_mb_keys"] = keys
+        return keys
 
     def _update_phase(self, batch: TensorDictBase) -> None:
         with torch.no_grad(), self._autocast:
             self.advantage(batch)
         flat = batch.reshape(-1)
+        keys = self._minibatch_keys(flat)
+        if keys is not None:
+            flat = flat.select(*keys)
         n = flat.batch_size[0]
         mb = n // self.minibatches
         device = flat.device
+        from .. import ops
+
         for _ in range(self.epochs):
             perm = torch.randperm(n, device=device)
             # one gather of the whole flat store, then minibatches are
-            # contiguous zero-copy slices
-            shuffled = flat[perm]
+            # contiguous zero-copy slices; the gather itself is one
+            # batched kernel when the leaves allow it
+            shuffled = None
+            if flat.device is not None and flat.device.type == "cuda":
+                shuffled = ops.multi_gather_td(flat, perm)
+            if shuffled is None:
+                shuffled = flat[perm]
             for i in range(self.minibatches):
                 self._run_minibatch(shuffled[i * mb : (i + 1) * mb])
 
