@@ -65,7 +65,7 @@ void launch_tanh_normal_entropy_bwd(const float*, const float*, const float*,
                                     void*);
 void launch_adv_stats(const float*, float*, float*, long, void*);
 void launch_ppo_clip_fwd(const float*, const float*, const float*, float*,
-                         float*, float, float, long, void*);
+                         float*, float*, float*, float, float, long, void*);
 void launch_ppo_clip_bwd(const float*, const float*, const float*,
                          const float*, float*, float, float, long, void*);
 void launch_smooth_l1_fwd(const void*, const float*, float*, float*, long,
@@ -74,16 +74,17 @@ void launch_smooth_l1_bwd(const void*, const float*, const float*, void*,
                           long, int, void*);
 void launch_ppo_head_fwd(const void*, const float*, const float*,
                          const float*, const float*, const float*, float*,
-                         float*, float, float, float, float, float, long, int,
-                         int, void*);
+                         float* const*, float, float, float, float, float,
+                         long, int, int, void*);
 void launch_ppo_head_bwd(const void*, const float*, const float*,
                          const float*, const float*, const float*,
                          const float*, const float*, void*, float, float,
                          float, float, float, long, int, int, void*);
 int mlp3_mfma_lds_bytes(int, int, int);
 int wgrad3_slab_count(long);
-void launch_grad_clip_coef(const void*, int, float, float*, void*);
+void launch_grad_clip_coef(const void*, int, float, float*, float*, void*);
 void launch_multi_gather(const void*, int, const long*, long, void*);
+void launch_multi_shuffle(const void*, int, const int*, long, void*);
 void launch_wgrad3(const void* const*, const void* const*, float* const*,
                    float* const*, float* const*, float* const*, const int*,
                    const int*, long, void*);
@@ -651,18 +652,20 @@ std::vector<torch::Tensor> ppo_clip_fwd(torch::Tensor lw, torch::Tensor adv,
   TORCH_CHECK(adv.numel() == N, "ppo_clip_fwd: numel mismatch");
   auto opt = lw.options();
   auto part = torch::empty({256 * 4}, opt);
-  auto out = torch::empty({3}, opt);
-  auto stats = normalize ? torch::empty({2}, opt) : torch::Tensor();
+  auto o_loss = torch::empty({}, opt);
+  auto o_ess = torch::empty({}, opt);
+  auto o_cf = torch::empty({}, opt);
+  auto stats = normalize ? torch::empty({2}, opt) : torch::empty({0}, opt);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   if (normalize)
     launch_adv_stats(adv.data_ptr<float>(), part.data_ptr<float>(),
                      stats.data_ptr<float>(), N, (void*)stream);
   launch_ppo_clip_fwd(lw.data_ptr<float>(), adv.data_ptr<float>(),
                       normalize ? stats.data_ptr<float>() : nullptr,
-                      part.data_ptr<float>(), out.data_ptr<float>(),
+                      part.data_ptr<float>(), o_loss.data_ptr<float>(),
+                      o_ess.data_ptr<float>(), o_cf.data_ptr<float>(),
                       (float)lo, (float)hi, N, (void*)stream);
-  if (!normalize) stats = torch::empty({0}, opt);
-  return {out, stats};
+  return {o_loss, o_ess, o_cf, stats};
 }
 
 torch::Tensor ppo_clip_bwd(torch::Tensor lw, torch::Tensor adv,
@@ -689,7 +692,7 @@ torch::Tensor smooth_l1_fwd(torch::Tensor v, torch::Tensor t) {
   const long N = v.numel();
   TORCH_CHECK(t.numel() == N, "smooth_l1: numel mismatch");
   auto part = torch::empty({256}, t.options());
-  auto out = torch::empty({1}, t.options());
+  auto out = torch::empty({}, t.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_smooth_l1_fwd(v.data_ptr(), t.data_ptr<float>(),
                        part.data_ptr<float>(), out.data_ptr<float>(), N,
@@ -728,9 +731,11 @@ torch::Tensor fused_grad_clip_coef(std::vector<torch::Tensor> grads,
     args.len[i] = (int)grads[i].numel();
   }
   auto coef = torch::empty({}, grads[0].options());
+  auto part = torch::empty({32}, grads[0].options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_grad_clip_coef(&args, (int)grads.size(), (float)max_norm,
-                        coef.data_ptr<float>(), (void*)stream);
+                        part.data_ptr<float>(), coef.data_ptr<float>(),
+                        (void*)stream);
   return coef;
 }
 
@@ -764,6 +769,39 @@ std::vector<torch::Tensor> multi_gather(torch::Tensor perm,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_multi_gather(&args, (int)srcs.size(), perm.data_ptr<long>(), n,
                       (void*)stream);
+  return out;
+}
+
+// Feistel-shuffle variant: dst[t][i] = src[t][perm(i)] where perm is a
+// keyed 4-round Feistel permutation computed inline — replaces
+// randperm's radix sort + per-key index kernels with ONE launch.
+std::vector<torch::Tensor> multi_shuffle(torch::Tensor keys,
+                                         std::vector<torch::Tensor> srcs) {
+  struct {
+    const float* src[8];
+    float* dst[8];
+    int w[8];
+  } args;
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt32 &&
+                  keys.numel() >= 4 && keys.is_contiguous(),
+              "keys int32[4] cuda");
+  TORCH_CHECK(!srcs.empty() && srcs.size() <= 8, "1..8 tensors");
+  const long n = srcs[0].size(0);
+  std::vector<torch::Tensor> out;
+  for (size_t i = 0; i < srcs.size(); ++i) {
+    auto& s = srcs[i];
+    TORCH_CHECK(s.is_cuda() && s.scalar_type() == torch::kFloat32 &&
+                    s.is_contiguous() && s.size(0) == n,
+                "multi_shuffle: fp32 cuda contiguous with shared dim 0");
+    auto d = torch::empty_like(s);
+    args.src[i] = s.data_ptr<float>();
+    args.dst[i] = d.data_ptr<float>();
+    args.w[i] = (int)(s.numel() / n);
+    out.push_back(d);
+  }
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_multi_shuffle(&args, (int)srcs.size(), keys.data_ptr<int>(), n,
+                       (void*)stream);
   return out;
 }
 
@@ -901,7 +939,12 @@ std::vector<torch::Tensor> ppo_head_fwd(torch::Tensor head,
               "shape mismatch");
   auto opt = action.options();
   auto part = torch::empty({256 * 5}, opt);
-  auto out = torch::empty({5}, opt);
+  std::vector<torch::Tensor> outs;
+  float* outp[5];
+  for (int i = 0; i < 5; ++i) {
+    outs.push_back(torch::empty({}, opt));
+    outp[i] = outs.back().data_ptr<float>();
+  }
   auto stats = normalize ? torch::empty({2}, opt) : torch::empty({0}, opt);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   if (normalize)
@@ -911,10 +954,11 @@ std::vector<torch::Tensor> ppo_head_fwd(torch::Tensor head,
                       eps.data_ptr<float>(), prev.data_ptr<float>(),
                       adv.data_ptr<float>(),
                       normalize ? stats.data_ptr<float>() : nullptr,
-                      part.data_ptr<float>(), out.data_ptr<float>(),
-                      (float)sp_bias, (float)lb, (float)lo, (float)hi,
-                      (float)ent_coeff, N, A, bf16 ? 1 : 0, (void*)stream);
-  return {out, stats};
+                      part.data_ptr<float>(), outp, (float)sp_bias,
+                      (float)lb, (float)lo, (float)hi, (float)ent_coeff, N,
+                      A, bf16 ? 1 : 0, (void*)stream);
+  outs.push_back(stats);
+  return outs;
 }
 
 torch::Tensor ppo_head_bwd(torch::Tensor head, torch::Tensor action,
@@ -997,6 +1041,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "single-kernel global grad-norm clip coefficient (HIP)");
   m.def("multi_gather", &multi_gather,
         "batched row gather over up to 8 tensors (HIP)");
+  m.def("multi_shuffle", &multi_shuffle,
+        "batched Feistel-permutation shuffle (HIP)");
   m.def("mlp3_mfma_fwd", &mlp3_mfma_fwd,
         "MFMA whole-MLP forward: 3 GEMMs + tanh in one launch (HIP)");
   m.def("mlp3_mfma_bwd", &mlp3_mfma_bwd,

@@ -274,7 +274,9 @@ __global__ void ppo_clip_fwd_partials_k(const float* __restrict__ lw,
 
 __global__ void ppo_clip_finalize_k(const float* __restrict__ part,
                                     const int nwg, const long N,
-                                    float* __restrict__ out) {
+                                    float* __restrict__ o_loss,
+                                    float* __restrict__ o_ess,
+                                    float* __restrict__ o_cf) {
   float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f;
   for (int i = threadIdx.x; i < nwg; i += blockDim.x) {
     sg += part[4 * i];
@@ -285,7 +287,7 @@ __global__ void ppo_clip_finalize_k(const float* __restrict__ part,
   __shared__ float smem[8];
   float t;
   t = block_sum(sg, smem);
-  if (threadIdx.x == 0) out[0] = -t / (float)N;  // loss_objective (mean)
+  if (threadIdx.x == 0) *o_loss = -t / (float)N;  // loss_objective (mean)
   __syncthreads();
   t = block_sum(sr, smem);
   if (threadIdx.x == 0) smem[4] = t;
@@ -293,10 +295,10 @@ __global__ void ppo_clip_finalize_k(const float* __restrict__ part,
   const float tsr = smem[4];
   t = block_sum(sr2, smem);
   if (threadIdx.x == 0)
-    out[1] = tsr * tsr / fmaxf(t, 1e-12f) / (float)N;  // ESS / N
+    *o_ess = tsr * tsr / fmaxf(t, 1e-12f) / (float)N;  // ESS / N
   __syncthreads();
   t = block_sum(sc, smem);
-  if (threadIdx.x == 0) out[2] = t / (float)N;  // clip_fraction
+  if (threadIdx.x == 0) *o_cf = t / (float)N;  // clip_fraction
 }
 
 __global__ void ppo_clip_bwd_k(const float* __restrict__ lw,
@@ -381,13 +383,15 @@ extern "C" void launch_adv_stats(const float* adv, float* part, float* stats,
 
 extern "C" void launch_ppo_clip_fwd(const float* lw, const float* adv,
                                     const float* stats, float* part,
-                                    float* out, float lo, float hi, long N,
+                                    float* o_loss, float* o_ess, float* o_cf,
+                                    float lo, float hi, long N,
                                     void* stream) {
   const int blocks = red_blocks(N);
   hipLaunchKernelGGL(ppo_clip_fwd_partials_k, dim3(blocks), dim3(LP_THREADS),
                      0, (hipStream_t)stream, lw, adv, stats, part, lo, hi, N);
   hipLaunchKernelGGL(ppo_clip_finalize_k, dim3(1), dim3(LP_THREADS), 0,
-                     (hipStream_t)stream, part, blocks, N, out);
+                     (hipStream_t)stream, part, blocks, N, o_loss, o_ess,
+                     o_cf);
 }
 
 extern "C" void launch_ppo_clip_bwd(const float* lw, const float* adv,
@@ -520,7 +524,11 @@ __global__ void ppo_head_fwd_partials_k(
 __global__ void ppo_head_finalize_k(const float* __restrict__ part,
                                     const int nwg, const long N,
                                     const float ent_coeff,
-                                    float* __restrict__ out) {
+                                    float* __restrict__ o_loss,
+                                    float* __restrict__ o_ess,
+                                    float* __restrict__ o_cf,
+                                    float* __restrict__ o_ent,
+                                    float* __restrict__ o_lent) {
   float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f;
   for (int i = threadIdx.x; i < nwg; i += blockDim.x) {
     sg += part[5 * i];
@@ -532,7 +540,7 @@ __global__ void ppo_head_finalize_k(const float* __restrict__ part,
   __shared__ float smem[8];
   float t;
   t = block_sum(sg, smem);
-  if (threadIdx.x == 0) out[0] = -t / (float)N;  // loss_objective
+  if (threadIdx.x == 0) *o_loss = -t / (float)N;  // loss_objective
   __syncthreads();
   t = block_sum(sr, smem);
   if (threadIdx.x == 0) smem[4] = t;
@@ -540,15 +548,15 @@ __global__ void ppo_head_finalize_k(const float* __restrict__ part,
   const float tsr = smem[4];
   t = block_sum(sr2, smem);
   if (threadIdx.x == 0)
-    out[1] = tsr * tsr / fmaxf(t, 1e-12f) / (float)N;  // ESS / N
+    *o_ess = tsr * tsr / fmaxf(t, 1e-12f) / (float)N;  // ESS / N
   __syncthreads();
   t = block_sum(sc, smem);
-  if (threadIdx.x == 0) out[2] = t / (float)N;  // clip_fraction
+  if (threadIdx.x == 0) *o_cf = t / (float)N;  // clip_fraction
   __syncthreads();
   t = block_sum(se, smem);
   if (threadIdx.x == 0) {
-    out[3] = t / (float)N;                   // entropy (mean)
-    out[4] = -ent_coeff * t / (float)N;      // loss_entropy
+    *o_ent = t / (float)N;                   // entropy (mean)
+    *o_lent = -ent_coeff * t / (float)N;     // loss_entropy
   }
 }
 
@@ -613,9 +621,9 @@ __global__ void ppo_head_bwd_k(
 extern "C" void launch_ppo_head_fwd(const void* head, const float* action,
                                     const float* eps, const float* prev,
                                     const float* adv, const float* stats,
-                                    float* part, float* out, float sp_bias,
-                                    float lb, float lo, float hi,
-                                    float ent_coeff, long N, int A,
+                                    float* part, float* const* outs,
+                                    float sp_bias, float lb, float lo,
+                                    float hi, float ent_coeff, long N, int A,
                                     int head_is_bf16, void* stream) {
   const int blocks = red_blocks(N);
   if (head_is_bf16)
@@ -629,7 +637,8 @@ extern "C" void launch_ppo_head_fwd(const void* head, const float* action,
                        (const float*)head, action, eps, prev, adv, stats,
                        part, sp_bias, lb, lo, hi, N, A);
   hipLaunchKernelGGL(ppo_head_finalize_k, dim3(1), dim3(LP_THREADS), 0,
-                     (hipStream_t)stream, part, blocks, N, ent_coeff, out);
+                     (hipStream_t)stream, part, blocks, N, ent_coeff,
+                     outs[0], outs[1], outs[2], outs[3], outs[4]);
 }
 
 extern "C" void launch_ppo_head_bwd(const void* head, const float* action,
@@ -679,24 +688,59 @@ struct GatherArgs {
 
 namespace {
 
-__global__ void grad_clip_coef_k(const ClipArgs args, const int nt,
-                                 const float max_norm,
-                                 float* __restrict__ coef) {
+#define CLIP_BLOCKS 32
+
+__global__ void grad_clip_partials_k(const ClipArgs args, const int nt,
+                                     float* __restrict__ part) {
   float ss = 0.f;
+  const int stride = gridDim.x * blockDim.x;
+  const int base = blockIdx.x * blockDim.x + threadIdx.x;
   for (int t = 0; t < nt; ++t) {
     const float* __restrict__ p = args.g[t];
     const int L = args.len[t];
-    for (int i = threadIdx.x; i < L; i += blockDim.x) {
+    for (int i = base; i < L; i += stride) {
       const float v = p[i];
       ss += v * v;
     }
   }
   __shared__ float smem[8];
   const float tot = block_sum(ss, smem);
+  if (threadIdx.x == 0) part[blockIdx.x] = tot;
+}
+
+__global__ void grad_clip_finalize_k(const float* __restrict__ part,
+                                     const float max_norm,
+                                     float* __restrict__ coef) {
+  float ss = (threadIdx.x < CLIP_BLOCKS) ? part[threadIdx.x] : 0.f;
+  __shared__ float smem[8];
+  const float tot = block_sum(ss, smem);
   if (threadIdx.x == 0) {
     const float c = max_norm / (sqrtf(tot) + 1e-6f);
     coef[0] = c < 1.f ? c : 1.f;
   }
+}
+
+// 4-round Feistel network over ceil(log2 n) bits with cycle-walking:
+// a keyed pseudorandom PERMUTATION of [0, n) computed inline — replaces
+// torch.randperm's ~8-launch radix sort for the epoch shuffle.  The
+// round keys come from a device tensor (philox-fresh per replay).
+__device__ __forceinline__ long feistel_perm(long i, const int* keys,
+                                             int half_bits, long n) {
+  const unsigned mask = (1u << half_bits) - 1u;
+  unsigned v = (unsigned)i;
+  do {
+    unsigned L = v >> half_bits, R = v & mask;
+    for (int r = 0; r < 4; ++r) {
+      unsigned f = (R + (unsigned)keys[r]) * 2654435761u;
+      f ^= f >> 13;
+      f *= 0x5bd1e995u;
+      unsigned nl = R;
+      R = (L ^ (f & mask));
+      L = nl;
+    }
+    v = (L << half_bits) | R;
+  } while ((long)v >= n);
+  return (long)v;
 }
 
 __global__ void multi_gather_k(const GatherArgs args,
@@ -713,18 +757,47 @@ __global__ void multi_gather_k(const GatherArgs args,
   }
 }
 
+__global__ void multi_shuffle_k(const GatherArgs args,
+                                const int* __restrict__ keys, const long n,
+                                const int half_bits) {
+  const int t = blockIdx.y;
+  const float* __restrict__ src = args.src[t];
+  float* __restrict__ dst = args.dst[t];
+  const int w = args.w[t];
+  const long total = n * w;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long r = i / w;
+    const long pr = feistel_perm(r, keys, half_bits, n);
+    dst[i] = src[pr * w + (i - r * w)];
+  }
+}
+
 }  // namespace
 
 extern "C" void launch_grad_clip_coef(const void* args, int nt,
-                                      float max_norm, float* coef,
-                                      void* stream) {
-  hipLaunchKernelGGL(grad_clip_coef_k, dim3(1), dim3(LP_THREADS), 0,
-                     (hipStream_t)stream, *(const ClipArgs*)args, nt,
-                     max_norm, coef);
+                                      float max_norm, float* part,
+                                      float* coef, void* stream) {
+  hipLaunchKernelGGL(grad_clip_partials_k, dim3(CLIP_BLOCKS),
+                     dim3(LP_THREADS), 0, (hipStream_t)stream,
+                     *(const ClipArgs*)args, nt, part);
+  hipLaunchKernelGGL(grad_clip_finalize_k, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, part, max_norm, coef);
 }
 
 extern "C" void launch_multi_gather(const void* args, int nt,
                                     const long* perm, long n, void* stream) {
   hipLaunchKernelGGL(multi_gather_k, dim3(256, nt), dim3(LP_THREADS), 0,
                      (hipStream_t)stream, *(const GatherArgs*)args, perm, n);
+}
+
+extern "C" void launch_multi_shuffle(const void* args, int nt,
+                                     const int* keys, long n,
+                                     void* stream) {
+  int bits = 1;
+  while ((1L << bits) < n) ++bits;
+  const int half = (bits + 1) / 2;
+  hipLaunchKernelGGL(multi_shuffle_k, dim3(256, nt), dim3(LP_THREADS), 0,
+                     (hipStream_t)stream, *(const GatherArgs*)args, keys, n,
+                     half);
 }

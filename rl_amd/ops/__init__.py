@@ -661,11 +661,10 @@ class _PPOClipFn(torch.autograd.Function):
         ctx.set_materialize_grads(False)
         lwf = lw.contiguous().reshape(-1)
         advf = adv.contiguous().reshape(-1)
-        out, stats = _C.ppo_clip_fwd(lwf, advf, lo, hi, normalize)
+        loss, ess, clip_frac, stats = _C.ppo_clip_fwd(lwf, advf, lo, hi, normalize)
         ctx.save_for_backward(lwf, advf, stats)
         ctx.bounds = (lo, hi)
         ctx.lw_shape = lw.shape
-        loss, ess, clip_frac = out[0], out[1], out[2]
         ctx.mark_non_differentiable(ess, clip_frac)
         return loss, ess, clip_frac
 
@@ -706,15 +705,12 @@ class _PPOHeadLossFn(torch.autograd.Function):
         prev_lp = prev_lp.contiguous().detach()
         adv = adv.contiguous().detach()
         eps = eps.contiguous()
-        out, stats = _C.ppo_head_fwd(
+        loss_obj, ess, clip_frac, ent_mean, loss_ent, stats = _C.ppo_head_fwd(
             head, action, eps, prev_lp, adv, sp_bias, lb, lo, hi, ent_coeff,
             normalize,
         )
         ctx.save_for_backward(head, action, eps, prev_lp, adv, stats)
         ctx.cfg = (sp_bias, lb, lo, hi, ent_coeff)
-        loss_obj, ess, clip_frac, ent_mean, loss_ent = (
-            out[0], out[1], out[2], out[3], out[4]
-        )
         ctx.mark_non_differentiable(ess, clip_frac, ent_mean)
         return loss_obj, loss_ent, ent_mean, ess, clip_frac
 
@@ -785,6 +781,42 @@ def multi_gather_td(flat_td, perm: torch.Tensor):
     )
 
 
+def multi_shuffle_td(flat_td, keys: torch.Tensor):
+    """Epoch shuffle via a keyed Feistel permutation computed inline in
+    ONE kernel (csrc/loss_ops.hip), replacing randperm radix sort plus
+    per-key gathers.  keys is an int32[4] cuda tensor of round keys
+    (draw fresh per epoch so graph replays get a new permutation).
+    Returns None when the leaves do not fit."""
+    if not HAS_HIP_EXT:
+        return None
+    try:
+        items = list(flat_td.items())
+    except Exception:
+        return None
+    if not (1 <= len(items) <= 8):
+        return None
+    n = flat_td.batch_size[0]
+    srcs = []
+    for _, v in items:
+        if not (
+            isinstance(v, torch.Tensor)
+            and v.is_cuda
+            and v.dtype == torch.float32
+            and v.is_contiguous()
+            and v.dim() >= 1
+            and v.shape[0] == n
+        ):
+            return None
+        srcs.append(v)
+    outs = _C.multi_shuffle(keys, srcs)
+    from ..tensordict import TensorDict
+
+    return TensorDict(
+        {k: o for (k, _), o in zip(items, outs)}, batch_size=[n],
+        device=srcs[0].device,
+    )
+
+
 def ppo_head_loss(head, action, prev_log_prob, advantage, eps, *, sp_bias,
                   scale_lb, lo, hi, entropy_coeff, normalize):
     """Fused ClipPPO actor losses straight from the raw policy-head
@@ -812,7 +844,7 @@ class _SmoothL1MeanFn(torch.autograd.Function):
         tf = target.contiguous().reshape(-1)
         ctx.save_for_backward(vf, tf)
         ctx.v_shape = value.shape
-        return _C.smooth_l1_fwd(vf, tf)[0]
+        return _C.smooth_l1_fwd(vf, tf)
 
     @staticmethod
     def backward(ctx, gout):

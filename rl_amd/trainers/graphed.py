@@ -248,16 +248,27 @@ class GraphedPPO:
         device = flat.device
         from .. import ops
 
+        on_gpu = any(
+            isinstance(v, torch.Tensor) and v.is_cuda for v in flat.values()
+        )
         for _ in range(self.epochs):
-            perm = torch.randperm(n, device=device)
-            # one gather of the whole flat store, then minibatches are
-            # contiguous zero-copy slices; the gather itself is one
-            # batched kernel when the leaves allow it
+            # one shuffle-gather of the whole flat store, then
+            # minibatches are contiguous zero-copy slices.  On GPU the
+            # shuffle is ONE kernel applying a keyed Feistel
+            # permutation inline (fresh philox keys per epoch);
+            # randperm's radix sort + per-key index kernels otherwise.
             shuffled = None
-            if flat.device is not None and flat.device.type == "cuda":
-                shuffled = ops.multi_gather_td(flat, perm)
+            if on_gpu:
+                keys = torch.randint(
+                    -(2 ** 31), 2 ** 31 - 1, (4,), device=device,
+                    dtype=torch.int32,
+                )
+                shuffled = ops.multi_shuffle_td(flat, keys)
             if shuffled is None:
-                shuffled = flat[perm]
+                perm = torch.randperm(n, device=device)
+                shuffled = ops.multi_gather_td(flat, perm) if on_gpu else None
+                if shuffled is None:
+                    shuffled = flat[perm]
             for i in range(self.minibatches):
                 self._run_minibatch(shuffled[i * mb : (i + 1) * mb])
 
