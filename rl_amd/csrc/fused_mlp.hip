@@ -333,7 +333,7 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
   const int rblk = (R == 64 ? wave : (wave & 1)) * 16;
   const int ct0 = (R == 64 ? 0 : (wave >> 1));
   const int cts = (R == 64 ? 1 : 2);
-  // layer 1 + tanh
+  // layer 1 + tanh (global h1 goes out later via a coalesced LDS copy)
   for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};
     m3_gemm_tile(&s_x[(size_t)rblk * lx], lx, s_w1, lx, ct * 16, Op,
@@ -343,9 +343,7 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = rblk + erow + r;
-      const __hip_bfloat16 h = __hip_bfloat16(tanhf(acc[r] + bias));
-      s_h1[(size_t)row * lh + col] = h;
-      if (row < rows) h1_out[(row0 + row) * H + col] = h;
+      s_h1[(size_t)row * lh + col] = __hip_bfloat16(tanhf(acc[r] + bias));
     }
   }
   // zero the k-pad of h tiles once (Hp > H only when H % 32 == 16)
@@ -355,7 +353,13 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
       s_h2[(size_t)i * lh + k] = __hip_bfloat16(0.f);
     }
   __syncthreads();
-  // layer 2 + tanh
+  // coalesced vectorized h1 store (8 bf16 per thread) + layer 2
+  const int h8 = H / 8;  // H % 16 == 0
+  for (int i = tid; i < rows * h8; i += MLP_THREADS) {
+    const int r = i / h8, c8 = i - r * h8;
+    *reinterpret_cast<uint4*>(&h1_out[(row0 + r) * H + 8 * c8]) =
+        *reinterpret_cast<const uint4*>(&s_h1[(size_t)r * lh + 8 * c8]);
+  }
   for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};
     m3_gemm_tile(&s_h1[(size_t)rblk * lh], lh, s_w2, lh, ct * 16, Hp,
@@ -365,27 +369,32 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = rblk + erow + r;
-      const __hip_bfloat16 h = __hip_bfloat16(tanhf(acc[r] + bias));
-      s_h2[(size_t)row * lh + col] = h;
-      if (row < rows) h2_out[(row0 + row) * H + col] = h;
+      s_h2[(size_t)row * lh + col] = __hip_bfloat16(tanhf(acc[r] + bias));
     }
   }
   __syncthreads();
-  // layer 3 (head)
+  // coalesced h2 store + layer 3 (head result staged over s_h1)
+  for (int i = tid; i < rows * h8; i += MLP_THREADS) {
+    const int r = i / h8, c8 = i - r * h8;
+    *reinterpret_cast<uint4*>(&h2_out[(row0 + r) * H + 8 * c8]) =
+        *reinterpret_cast<const uint4*>(&s_h2[(size_t)r * lh + 8 * c8]);
+  }
   for (int ct = ct0; ct < A2p / 16; ct += cts) {
     mfrag_f acc = {};
     m3_gemm_tile(&s_h2[(size_t)rblk * lh], lh, s_w3, lh, ct * 16, Hp,
                  lane, &acc);
     const int col = ct * 16 + ecol;
-    if (col < A2) {
-      const float bias = __bfloat162float(b3[col]);
+    const float bias = (col < A2) ? __bfloat162float(b3[col]) : 0.f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = rblk + erow + r;
-        if (row < rows)
-          out[(row0 + row) * A2 + col] = __hip_bfloat16(acc[r] + bias);
-      }
+    for (int r = 0; r < 4; ++r) {
+      const int row = rblk + erow + r;
+      s_h1[(size_t)row * lh + col] = __hip_bfloat16(acc[r] + bias);
     }
+  }
+  __syncthreads();
+  for (int i = tid; i < rows * A2; i += MLP_THREADS) {
+    const int r = i / A2, c = i - r * A2;
+    out[(row0 + r) * A2 + c] = s_h1[(size_t)r * lh + c];
   }
 }
 
@@ -402,10 +411,11 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
   extern __shared__ __hip_bfloat16 smem[];
   const int A2p32 = m3_pad32(A2), Hp = m3_pad32(H);
   const int ld = A2p32 + 8, lh = Hp + 8;
-  __hip_bfloat16* s_dy = smem;                   // [64][ld]
+  __hip_bfloat16* s_dy = smem;                   // [R][ld]
   __hip_bfloat16* s_w3t = s_dy + R * ld;   // [H][ld]  w3t[j][k]=w3[k][j]
-  __hip_bfloat16* s_dh2 = s_w3t + H * ld;        // [64][lh]
+  __hip_bfloat16* s_dh2 = s_w3t + H * ld;        // [R][lh]
   __hip_bfloat16* s_w2t = s_dh2 + R * lh;  // [H][lh]
+  __hip_bfloat16* s_dh1 = s_w2t + H * lh;        // [R][lh]
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
@@ -446,7 +456,6 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
       if (row < rows) {
         const float hv = __bfloat162float(h2[(row0 + row) * H + col]);
         g = acc[r] * (1.f - hv * hv);
-        dh2_out[(row0 + row) * H + col] = __hip_bfloat16(g);
       }
       s_dh2[(size_t)row * lh + col] = __hip_bfloat16(g);
     }
@@ -454,7 +463,13 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
   for (int i = tid; i < R; i += MLP_THREADS)
     for (int k = H; k < Hp; ++k) s_dh2[(size_t)i * lh + k] = __hip_bfloat16(0.f);
   __syncthreads();
-  // dh1 = (dh2 @ W2) * (1 - h1^2)
+  // coalesced dh2 store + dh1 = (dh2 @ W2) * (1 - h1^2)
+  const int h8 = H / 8;
+  for (int i = tid; i < rows * h8; i += MLP_THREADS) {
+    const int r = i / h8, c8 = i - r * h8;
+    *reinterpret_cast<uint4*>(&dh2_out[(row0 + r) * H + 8 * c8]) =
+        *reinterpret_cast<const uint4*>(&s_dh2[(size_t)r * lh + 8 * c8]);
+  }
   for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};
     m3_gemm_tile(&s_dh2[(size_t)rblk * lh], lh, s_w2t, lh, ct * 16, Hp,
@@ -463,12 +478,19 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = rblk + erow + r;
+      float g = 0.f;
       if (row < rows) {
         const float hv = __bfloat162float(h1[(row0 + row) * H + col]);
-        dh1_out[(row0 + row) * H + col] =
-            __hip_bfloat16(acc[r] * (1.f - hv * hv));
+        g = acc[r] * (1.f - hv * hv);
       }
+      s_dh1[(size_t)row * lh + col] = __hip_bfloat16(g);
     }
+  }
+  __syncthreads();
+  for (int i = tid; i < rows * h8; i += MLP_THREADS) {
+    const int r = i / h8, c8 = i - r * h8;
+    *reinterpret_cast<uint4*>(&dh1_out[(row0 + r) * H + 8 * c8]) =
+        *reinterpret_cast<const uint4*>(&s_dh1[(size_t)r * lh + 8 * c8]);
   }
 }
 
@@ -480,7 +502,7 @@ extern "C" int mlp3_mfma_lds_bytes(int O, int H, int A2) {
   const int lx = Op + 8, lh = Hp + 8, A2p = (A2 + 15) & ~15;
   const int ld = ((A2 + 31) & ~31) + 8;
   const int fwd = 2 * (R * lx + H * lx + 2 * R * lh + H * lh + A2p * lh);
-  const int bwd = 2 * (R * ld + H * ld + R * lh + H * lh);
+  const int bwd = 2 * (R * ld + H * ld + 2 * R * lh + H * lh);
   return fwd > bwd ? fwd : bwd;
 }
 
@@ -523,7 +545,7 @@ extern "C" void launch_mlp3_mfma_bwd(const void* dout, int d_is_f32,
   const int R = N >= M3_ROWS_SWITCH ? 64 : 32;
   const int blocks = (int)((N + R - 1) / R);
   const int ld = ((A2 + 31) & ~31) + 8, lh = ((H + 31) & ~31) + 8;
-  const int lds = 2 * (R * ld + H * ld + R * lh + H * lh);
+  const int lds = 2 * (R * ld + H * ld + 2 * R * lh + H * lh);
 #define M3_LAUNCH_BWD(TD, RR)                                               \
   hipLaunchKernelGGL((mlp3_mfma_bwd_kernel<TD, RR>), dim3(blocks),          \
                      dim3(MLP_THREADS), lds, (hipStream_t)stream,           \

@@ -430,8 +430,10 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad3_mfma_kernel(
   const long k_begin = (long)blockIdx.x * k_slab;
   const long k_end = min(K, k_begin + (long)k_slab);
 
-  __shared__ __hip_bfloat16 s_dy[2][ROW_CHUNK][64 + 2];
-  __shared__ __hip_bfloat16 s_x[2][ROW_CHUNK][64 + 2];
+  // row stride 72 shorts = 144 B: 16-byte aligned rows so full chunks
+  // stage with uint4 (8 bf16) loads/stores when the layer is 64 wide
+  __shared__ __hip_bfloat16 s_dy[2][ROW_CHUNK][64 + 8];
+  __shared__ __hip_bfloat16 s_x[2][ROW_CHUNK][64 + 8];
   __shared__ float s_bias[64];
 
   const int tid = threadIdx.x;
@@ -440,13 +442,31 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad3_mfma_kernel(
   const int tile_n = (wave >> 1) * 32;
   const int tile_m = (wave & 1) * 32;
   f32_frag16 acc = {};
-  float bias_acc = 0.f;
+  float bias_acc = 0.f;     // scalar path: this thread's column lc
+  float bias_v[8] = {};     // vector path: columns c8*8 .. c8*8+7
+  const int c8 = tid & 7;   // fixed per thread (stride 256 keeps i&7)
 
   const int lr = tid / 64;
   const int lc = tid % 64;
 
   auto stage = [&](long kc, int buf) {
     const int rows = (int)min((long)ROW_CHUNK, k_end - kc);
+    if (N == 64 && M == 64 && rows == ROW_CHUNK) {
+      for (int i = tid; i < ROW_CHUNK * 8; i += WG_THREADS) {
+        const int r = i >> 3;
+        const uint4 dv =
+            *reinterpret_cast<const uint4*>(&dy[(kc + r) * 64 + c8 * 8]);
+        const uint4 xv =
+            *reinterpret_cast<const uint4*>(&x[(kc + r) * 64 + c8 * 8]);
+        *reinterpret_cast<uint4*>(&s_dy[buf][r][c8 * 8]) = dv;
+        *reinterpret_cast<uint4*>(&s_x[buf][r][c8 * 8]) = xv;
+        const __hip_bfloat16* hv =
+            reinterpret_cast<const __hip_bfloat16*>(&dv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) bias_v[j] += __bfloat162float(hv[j]);
+      }
+      return;
+    }
     for (int i = 0; i < ROW_CHUNK / 4; ++i) {
       const int r = lr + 4 * i;
       const long gk = kc + r;
@@ -492,6 +512,9 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad3_mfma_kernel(
   if (lr == 0) s_bias[lc] = 0.f;
   __syncthreads();
   atomicAdd(&s_bias[lc], bias_acc);
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    if (bias_v[j] != 0.f) atomicAdd(&s_bias[c8 * 8 + j], bias_v[j]);
   __syncthreads();
   if (lr == 0) args.bias_part[l][(long)blockIdx.x * 64 + lc] = s_bias[lc];
 }
