@@ -787,17 +787,21 @@ def multi_shuffle_td(flat_td, keys: torch.Tensor):
     per-key gathers.  keys is an int32[4] cuda tensor of round keys
     (draw fresh per epoch so graph replays get a new permutation).
     Returns None when the leaves do not fit."""
+    dbg = os.environ.get("RL_AMD_DEBUG_SHUFFLE") == "1"
     if not HAS_HIP_EXT:
+        if dbg: print("shuffle: no ext", file=__import__("sys").stderr)
         return None
     try:
         items = list(flat_td.items())
-    except Exception:
+    except Exception as e:
+        if dbg: print("shuffle: items()", e, file=__import__("sys").stderr)
         return None
     if not (1 <= len(items) <= 8):
+        if dbg: print("shuffle: n items", len(items), file=__import__("sys").stderr)
         return None
     n = flat_td.batch_size[0]
     srcs = []
-    for _, v in items:
+    for k, v in items:
         if not (
             isinstance(v, torch.Tensor)
             and v.is_cuda
@@ -806,6 +810,7 @@ def multi_shuffle_td(flat_td, keys: torch.Tensor):
             and v.dim() >= 1
             and v.shape[0] == n
         ):
+            if dbg: print("shuffle: leaf", k, type(v), getattr(v, "dtype", None), getattr(v, "shape", None), v.is_contiguous() if isinstance(v, torch.Tensor) else None, file=__import__("sys").stderr)
             return None
         srcs.append(v)
     outs = _C.multi_shuffle(keys, srcs)

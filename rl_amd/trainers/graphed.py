@@ -216,8 +216,10 @@ class GraphedPPO:
         try:
             tk = self.loss_module.tensor_keys
             keys = set()
-            for attr in ("advantage", "value_target", "action",
-                         "sample_log_prob", "value"):
+            attrs = ["advantage", "value_target", "action", "sample_log_prob"]
+            if getattr(self.loss_module, "clip_value", None) is not None:
+                attrs.append("value")  # old values only needed for clipping
+            for attr in attrs:
                 k = getattr(tk, attr, None)
                 if k is not None:
                     keys.add(k)
