@@ -68,18 +68,19 @@ void launch_ppo_clip_fwd(const float*, const float*, const float*, float*,
                          float*, float*, float*, float, float, long, void*);
 void launch_ppo_clip_bwd(const float*, const float*, const float*,
                          const float*, float*, float, float, long, void*);
-void launch_smooth_l1_fwd(const void*, const float*, float*, float*, long,
-                          int, void*);
-void launch_smooth_l1_bwd(const void*, const float*, const float*, void*,
+void launch_smooth_l1_fwd(const void*, const float*, float*, float*, float,
                           long, int, void*);
+void launch_smooth_l1_bwd(const void*, const float*, const float*, void*,
+                          float, long, int, void*);
 void launch_ppo_head_fwd(const void*, const float*, const float*,
                          const float*, const float*, const float*, float*,
                          float* const*, float, float, float, float, float,
                          long, int, int, void*);
 void launch_ppo_head_bwd(const void*, const float*, const float*,
                          const float*, const float*, const float*,
-                         const float*, const float*, void*, float, float,
-                         float, float, float, long, int, int, void*);
+                         const float*, const float*, const float*, void*,
+                         float, float, float, float, float, long, int, int,
+                         void*);
 int mlp3_mfma_lds_bytes(int, int, int);
 int wgrad3_slab_count(long);
 void launch_grad_clip_coef(const void*, int, float, float*, float*, void*);
@@ -683,7 +684,8 @@ torch::Tensor ppo_clip_bwd(torch::Tensor lw, torch::Tensor adv,
 }
 
 // Fused mean smooth-L1 (beta=1) critic loss; value may be bf16 (autocast).
-torch::Tensor smooth_l1_fwd(torch::Tensor v, torch::Tensor t) {
+torch::Tensor smooth_l1_fwd(torch::Tensor v, torch::Tensor t,
+                            double scale) {
   TORCH_CHECK(v.is_cuda() && v.is_contiguous() && t.is_contiguous(),
               "smooth_l1: cuda contiguous");
   TORCH_CHECK(t.scalar_type() == torch::kFloat32, "target fp32");
@@ -695,20 +697,20 @@ torch::Tensor smooth_l1_fwd(torch::Tensor v, torch::Tensor t) {
   auto out = torch::empty({}, t.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_smooth_l1_fwd(v.data_ptr(), t.data_ptr<float>(),
-                       part.data_ptr<float>(), out.data_ptr<float>(), N,
-                       bf16 ? 1 : 0, (void*)stream);
+                       part.data_ptr<float>(), out.data_ptr<float>(),
+                       (float)scale, N, bf16 ? 1 : 0, (void*)stream);
   return out;
 }
 
 torch::Tensor smooth_l1_bwd(torch::Tensor v, torch::Tensor t,
-                            torch::Tensor gout) {
+                            torch::Tensor gout, double scale) {
   const bool bf16 = v.scalar_type() == torch::kBFloat16;
   const long N = v.numel();
   auto dv = torch::empty_like(v);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_smooth_l1_bwd(v.data_ptr(), t.data_ptr<float>(),
-                       gout.data_ptr<float>(), dv.data_ptr(), N, bf16 ? 1 : 0,
-                       (void*)stream);
+                       gout.data_ptr<float>(), dv.data_ptr(), (float)scale,
+                       N, bf16 ? 1 : 0, (void*)stream);
   return dv;
 }
 // Fused gradient clipping (csrc/loss_ops.hip): one single-workgroup
@@ -940,8 +942,8 @@ std::vector<torch::Tensor> ppo_head_fwd(torch::Tensor head,
   auto opt = action.options();
   auto part = torch::empty({256 * 5}, opt);
   std::vector<torch::Tensor> outs;
-  float* outp[5];
-  for (int i = 0; i < 5; ++i) {
+  float* outp[6];
+  for (int i = 0; i < 6; ++i) {
     outs.push_back(torch::empty({}, opt));
     outp[i] = outs.back().data_ptr<float>();
   }
@@ -965,8 +967,8 @@ torch::Tensor ppo_head_bwd(torch::Tensor head, torch::Tensor action,
                            torch::Tensor eps, torch::Tensor prev,
                            torch::Tensor adv, torch::Tensor stats,
                            torch::Tensor gobj, torch::Tensor gent,
-                           double sp_bias, double lb, double lo, double hi,
-                           double ent_coeff) {
+                           torch::Tensor gact, double sp_bias, double lb,
+                           double lo, double hi, double ent_coeff) {
   const bool bf16 = head.scalar_type() == torch::kBFloat16;
   const long N = head.size(0);
   const int A = (int)(head.size(1) / 2);
@@ -977,7 +979,8 @@ torch::Tensor ppo_head_bwd(torch::Tensor head, torch::Tensor action,
       prev.data_ptr<float>(), adv.data_ptr<float>(),
       stats.numel() ? stats.data_ptr<float>() : nullptr,
       gobj.numel() ? gobj.data_ptr<float>() : nullptr,
-      gent.numel() ? gent.data_ptr<float>() : nullptr, dhead.data_ptr(),
+      gent.numel() ? gent.data_ptr<float>() : nullptr,
+      gact.numel() ? gact.data_ptr<float>() : nullptr, dhead.data_ptr(),
       (float)sp_bias, (float)lb, (float)lo, (float)hi, (float)ent_coeff, N, A,
       bf16 ? 1 : 0, (void*)stream);
   return dhead;

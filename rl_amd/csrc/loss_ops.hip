@@ -340,20 +340,22 @@ __global__ void smooth_l1_partials_k(const TV* __restrict__ v,
 
 __global__ void sum_finalize_mean_k(const float* __restrict__ part,
                                     const int nwg, const long N,
+                                    const float scale,
                                     float* __restrict__ out) {
   float s = 0.f;
   for (int i = threadIdx.x; i < nwg; i += blockDim.x) s += part[i];
   __shared__ float smem[8];
   const float ts = block_sum(s, smem);
-  if (threadIdx.x == 0) out[0] = ts / (float)N;
+  if (threadIdx.x == 0) out[0] = scale * ts / (float)N;
 }
 
 template <typename TV>
 __global__ void smooth_l1_bwd_k(const TV* __restrict__ v,
                                 const float* __restrict__ t,
                                 const float* __restrict__ gout,
-                                TV* __restrict__ dv, const long N) {
-  const float g = gout[0] / (float)N;
+                                TV* __restrict__ dv, const float scale,
+                                const long N) {
+  const float g = scale * gout[0] / (float)N;
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < N;
        i += (long)gridDim.x * blockDim.x) {
     const float z = (float)v[i] - t[i];
@@ -405,8 +407,8 @@ extern "C" void launch_ppo_clip_bwd(const float* lw, const float* adv,
 }
 
 extern "C" void launch_smooth_l1_fwd(const void* v, const float* t,
-                                     float* part, float* out, long N,
-                                     int v_is_bf16, void* stream) {
+                                     float* part, float* out, float scale,
+                                     long N, int v_is_bf16, void* stream) {
   const int blocks = red_blocks(N);
   if (v_is_bf16)
     hipLaunchKernelGGL(smooth_l1_partials_k<__hip_bfloat16>, dim3(blocks),
@@ -417,22 +419,23 @@ extern "C" void launch_smooth_l1_fwd(const void* v, const float* t,
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const float*)v, t, part, N);
   hipLaunchKernelGGL(sum_finalize_mean_k, dim3(1), dim3(LP_THREADS), 0,
-                     (hipStream_t)stream, part, blocks, N, out);
+                     (hipStream_t)stream, part, blocks, N, scale, out);
 }
 
 extern "C" void launch_smooth_l1_bwd(const void* v, const float* t,
-                                     const float* gout, void* dv, long N,
-                                     int v_is_bf16, void* stream) {
+                                     const float* gout, void* dv,
+                                     float scale, long N, int v_is_bf16,
+                                     void* stream) {
   const int blocks = red_blocks(N);
   if (v_is_bf16)
     hipLaunchKernelGGL(smooth_l1_bwd_k<__hip_bfloat16>, dim3(blocks),
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)v, t, gout, (__hip_bfloat16*)dv,
-                       N);
+                       scale, N);
   else
     hipLaunchKernelGGL(smooth_l1_bwd_k<float>, dim3(blocks), dim3(LP_THREADS),
                        0, (hipStream_t)stream, (const float*)v, t, gout,
-                       (float*)dv, N);
+                       (float*)dv, scale, N);
 }
 
 // ---------------------------------------------------------------------------
@@ -528,7 +531,8 @@ __global__ void ppo_head_finalize_k(const float* __restrict__ part,
                                     float* __restrict__ o_ess,
                                     float* __restrict__ o_cf,
                                     float* __restrict__ o_ent,
-                                    float* __restrict__ o_lent) {
+                                    float* __restrict__ o_lent,
+                                    float* __restrict__ o_act) {
   float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f;
   for (int i = threadIdx.x; i < nwg; i += blockDim.x) {
     sg += part[5 * i];
@@ -556,7 +560,10 @@ __global__ void ppo_head_finalize_k(const float* __restrict__ part,
   t = block_sum(se, smem);
   if (threadIdx.x == 0) {
     *o_ent = t / (float)N;                   // entropy (mean)
-    *o_lent = -ent_coeff * t / (float)N;     // loss_entropy
+    const float lent = -ent_coeff * t / (float)N;
+    *o_lent = lent;                          // loss_entropy
+    *o_act = *o_loss + lent;  // pre-summed actor loss (saves the eager
+                              // add + its backward per minibatch)
   }
 }
 
@@ -567,13 +574,16 @@ __global__ void ppo_head_bwd_k(
     const float* __restrict__ adv, const float* __restrict__ stats,
     const float* __restrict__ gobj,  // 0-d upstream grad of loss_objective
     const float* __restrict__ gent,  // nullable: grad of loss_entropy
+    const float* __restrict__ gact,  // nullable: grad of the pre-summed
+                                     // actor loss (adds to both)
     TV* __restrict__ dhead, const float sp_bias, const float lb,
     const float lo, const float hi, const float ent_coeff, const long N,
     const int A) {
   const float mu = stats ? stats[0] : 0.f;
   const float isd = stats ? stats[1] : 1.f;
-  const float g1s = gobj ? -gobj[0] / (float)N : 0.f;
-  const float ges = gent ? -ent_coeff * gent[0] / (float)N : 0.f;
+  const float ga = gact ? gact[0] : 0.f;
+  const float g1s = -(ga + (gobj ? gobj[0] : 0.f)) / (float)N;
+  const float ges = -ent_coeff * (ga + (gent ? gent[0] : 0.f)) / (float)N;
   for (long n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
        n += (long)gridDim.x * blockDim.x) {
     // pass 1: recompute lp -> per-row dlw
@@ -638,29 +648,30 @@ extern "C" void launch_ppo_head_fwd(const void* head, const float* action,
                        part, sp_bias, lb, lo, hi, N, A);
   hipLaunchKernelGGL(ppo_head_finalize_k, dim3(1), dim3(LP_THREADS), 0,
                      (hipStream_t)stream, part, blocks, N, ent_coeff,
-                     outs[0], outs[1], outs[2], outs[3], outs[4]);
+                     outs[0], outs[1], outs[2], outs[3], outs[4],
+                     outs[5]);
 }
 
 extern "C" void launch_ppo_head_bwd(const void* head, const float* action,
                                     const float* eps, const float* prev,
                                     const float* adv, const float* stats,
                                     const float* gobj, const float* gent,
-                                    void* dhead, float sp_bias, float lb,
-                                    float lo, float hi, float ent_coeff,
-                                    long N, int A, int head_is_bf16,
-                                    void* stream) {
+                                    const float* gact, void* dhead,
+                                    float sp_bias, float lb, float lo,
+                                    float hi, float ent_coeff, long N, int A,
+                                    int head_is_bf16, void* stream) {
   const int blocks = red_blocks(N);
   if (head_is_bf16)
     hipLaunchKernelGGL(ppo_head_bwd_k<__hip_bfloat16>, dim3(blocks),
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)head, action, eps, prev, adv,
-                       stats, gobj, gent, (__hip_bfloat16*)dhead, sp_bias,
-                       lb, lo, hi, ent_coeff, N, A);
+                       stats, gobj, gent, gact, (__hip_bfloat16*)dhead,
+                       sp_bias, lb, lo, hi, ent_coeff, N, A);
   else
     hipLaunchKernelGGL(ppo_head_bwd_k<float>, dim3(blocks), dim3(LP_THREADS),
                        0, (hipStream_t)stream, (const float*)head, action,
-                       eps, prev, adv, stats, gobj, gent, (float*)dhead,
-                       sp_bias, lb, lo, hi, ent_coeff, N, A);
+                       eps, prev, adv, stats, gobj, gent, gact,
+                       (float*)dhead, sp_bias, lb, lo, hi, ent_coeff, N, A);
 }
 
 // ---------------------------------------------------------------------------

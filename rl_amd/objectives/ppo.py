@@ -160,10 +160,11 @@ class PPOLoss(LossModule):
         scale = adv.std().clamp_min(1e-6)
         return (adv - loc) / scale
 
-    def _loss_critic_reduced(self, td: TensorDictBase) -> torch.Tensor:
-        """``self._reduce(self.loss_critic(td))`` with a fused HIP path:
-        smooth-L1 + mean in two launches, one-launch analytic backward
-        (csrc/loss_ops.hip) when no value clipping is in play."""
+    def _loss_critic_reduced(self, td: TensorDictBase, scale: float = 1.0) -> torch.Tensor:
+        """``scale * self._reduce(self.loss_critic(td))`` with a fused
+        HIP path: smooth-L1 + mean in two launches, one-launch analytic
+        backward (csrc/loss_ops.hip) when no value clipping is in play;
+        the coefficient is folded into the kernel."""
         if (
             self.clip_value is None
             and self.loss_critic_type in ("smooth_l1", "huber")
@@ -180,11 +181,11 @@ class PPOLoss(LossModule):
                 from .. import ops
 
                 if ops.HAS_HIP_EXT:
-                    return ops.smooth_l1_mean(value, target)
-            return self._reduce(
+                    return ops.smooth_l1_mean(value, target, scale)
+            return scale * self._reduce(
                 distance_loss(value, target, self.loss_critic_type)
             )
-        return self._reduce(self.loss_critic(td))
+        return scale * self._reduce(self.loss_critic(td))
 
     def _reduce(self, x: torch.Tensor) -> torch.Tensor:
         if self.reduction == "mean":
@@ -217,7 +218,10 @@ class PPOLoss(LossModule):
             out.set("entropy", entropy.detach().mean())
             out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
         if self.critic_coeff is not None and self.critic_coeff > 0:
-            out.set("loss_critic", self.critic_coeff * self._loss_critic_reduced(td))
+            out.set(
+                "loss_critic",
+                self._loss_critic_reduced(td, float(self.critic_coeff)),
+            )
         out.set("ESS", ess / log_weight.numel())
         return out
 
@@ -325,7 +329,7 @@ class ClipPPOLoss(PPOLoss):
             return None
         eps = torch.randn(N, A, device=action.device, dtype=torch.float32)
         lo, hi = self._clip_bounds
-        loss_obj, loss_ent, ent_mean, ess, clip_frac = ops.ppo_head_loss(
+        loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act = ops.ppo_head_loss(
             head.reshape(N, 2 * A),
             action.reshape(N, A).float(),
             prev_lp.reshape(N),
@@ -338,7 +342,7 @@ class ClipPPOLoss(PPOLoss):
             entropy_coeff=float(self.entropy_coeff),
             normalize=normalize,
         )
-        return loss_obj, loss_ent, ent_mean, ess, clip_frac
+        return loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act
 
     def forward(self, tensordict: TensorDictBase) -> TensorDictBase:
         td = tensordict.clone(False)
@@ -351,7 +355,7 @@ class ClipPPOLoss(PPOLoss):
         normalize = self.normalize_advantage and advantage.numel() > 1
         mega = self._mega_head_loss(td, advantage, normalize)
         if mega is not None:
-            loss_obj, loss_ent, ent_mean, ess, clip_frac = mega
+            loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act = mega
             out = TensorDict(
                 {
                     "loss_objective": loss_obj,
@@ -359,13 +363,16 @@ class ClipPPOLoss(PPOLoss):
                     "ESS": ess,
                     "entropy": ent_mean,
                     "loss_entropy": loss_ent,
+                    # pre-summed loss_objective + loss_entropy: trainers
+                    # that recognize it (GraphedPPO) skip the eager adds
+                    "_loss_actor": loss_act,
                 },
                 batch_size=[],
             )
             if self.critic_coeff is not None and self.critic_coeff > 0:
                 out.set(
                     "loss_critic",
-                    self.critic_coeff * self._loss_critic_reduced(td),
+                    self._loss_critic_reduced(td, float(self.critic_coeff)),
                 )
             return out
         log_weight, dist, _ = self._log_weight(td)
@@ -417,7 +424,10 @@ class ClipPPOLoss(PPOLoss):
             out.set("entropy", entropy.detach().mean())
             out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
         if self.critic_coeff is not None and self.critic_coeff > 0:
-            out.set("loss_critic", self.critic_coeff * self._loss_critic_reduced(td))
+            out.set(
+                "loss_critic",
+                self._loss_critic_reduced(td, float(self.critic_coeff)),
+            )
         return out
 
 
@@ -478,5 +488,8 @@ class KLPENPPOLoss(PPOLoss):
             out.set("entropy", entropy.detach().mean())
             out.set("loss_entropy", -self.entropy_coeff * self._reduce(entropy))
         if self.critic_coeff is not None and self.critic_coeff > 0:
-            out.set("loss_critic", self.critic_coeff * self._loss_critic_reduced(td))
+            out.set(
+                "loss_critic",
+                self._loss_critic_reduced(td, float(self.critic_coeff)),
+            )
         return out

@@ -700,29 +700,32 @@ class _PPOHeadLossFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, head, action, prev_lp, adv, eps, sp_bias, lb, lo, hi,
                 ent_coeff, normalize):
+        ctx.set_materialize_grads(False)
         head = head.contiguous()
         action = action.contiguous().detach()
         prev_lp = prev_lp.contiguous().detach()
         adv = adv.contiguous().detach()
         eps = eps.contiguous()
-        loss_obj, ess, clip_frac, ent_mean, loss_ent, stats = _C.ppo_head_fwd(
+        (loss_obj, ess, clip_frac, ent_mean, loss_ent, loss_act,
+         stats) = _C.ppo_head_fwd(
             head, action, eps, prev_lp, adv, sp_bias, lb, lo, hi, ent_coeff,
             normalize,
         )
         ctx.save_for_backward(head, action, eps, prev_lp, adv, stats)
         ctx.cfg = (sp_bias, lb, lo, hi, ent_coeff)
         ctx.mark_non_differentiable(ess, clip_frac, ent_mean)
-        return loss_obj, loss_ent, ent_mean, ess, clip_frac
+        return loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act
 
     @staticmethod
-    def backward(ctx, g_obj, g_ent, g_em, g_ess, g_cf):
+    def backward(ctx, g_obj, g_ent, g_em, g_ess, g_cf, g_act):
         head, action, eps, prev_lp, adv, stats = ctx.saved_tensors
         sp_bias, lb, lo, hi, ent_coeff = ctx.cfg
         empty = head.new_empty(0, dtype=torch.float32)
         gobj = g_obj.contiguous() if g_obj is not None else empty
         gent = g_ent.contiguous() if g_ent is not None else empty
+        gact = g_act.contiguous() if g_act is not None else empty
         dhead = _C.ppo_head_bwd(
-            head, action, eps, prev_lp, adv, stats, gobj, gent,
+            head, action, eps, prev_lp, adv, stats, gobj, gent, gact,
             sp_bias, lb, lo, hi, ent_coeff,
         )
         return (dhead, None, None, None, None, None, None, None, None, None,
@@ -827,7 +830,9 @@ def ppo_head_loss(head, action, prev_log_prob, advantage, eps, *, sp_bias,
     """Fused ClipPPO actor losses straight from the raw policy-head
     output (``[N, 2A]`` = loc | pre-softplus scale).  Returns
     ``(loss_objective, loss_entropy, entropy_mean, ESS_per_sample,
-    clip_fraction)``; the first two carry gradient to ``head``."""
+    clip_fraction, loss_actor)`` where ``loss_actor`` is the pre-summed
+    ``loss_objective + loss_entropy``; the two losses and the sum carry
+    gradient to ``head``."""
     _require_ext()
     return _PPOHeadLossFn.apply(
         head, action, prev_log_prob, advantage, eps, float(sp_bias),
@@ -843,28 +848,30 @@ class _SmoothL1MeanFn(torch.autograd.Function):
     autocast); the target is fp32 data."""
 
     @staticmethod
-    def forward(ctx, value, target):
+    def forward(ctx, value, target, scale):
         ctx.set_materialize_grads(False)
         vf = value.contiguous().reshape(-1)
         tf = target.contiguous().reshape(-1)
         ctx.save_for_backward(vf, tf)
         ctx.v_shape = value.shape
-        return _C.smooth_l1_fwd(vf, tf)
+        ctx.scale = scale
+        return _C.smooth_l1_fwd(vf, tf, scale)
 
     @staticmethod
     def backward(ctx, gout):
         vf, tf = ctx.saved_tensors
         if gout is None:
-            return None, None
-        dv = _C.smooth_l1_bwd(vf, tf, gout.contiguous())
-        return dv.reshape(ctx.v_shape), None
+            return None, None, None
+        dv = _C.smooth_l1_bwd(vf, tf, gout.contiguous(), ctx.scale)
+        return dv.reshape(ctx.v_shape), None, None
 
 
-def smooth_l1_mean(value, target):
-    """Fused ``F.smooth_l1_loss(value, target, reduction="mean")`` on
-    GPU; gradient flows to ``value`` only (the target is data)."""
+def smooth_l1_mean(value, target, scale: float = 1.0):
+    """Fused ``scale * F.smooth_l1_loss(value, target, reduction=
+    "mean")`` on GPU; gradient flows to ``value`` only (the target is
+    data).  ``scale`` folds a loss coefficient into the kernel."""
     _require_ext()
-    return _SmoothL1MeanFn.apply(value, target.detach())
+    return _SmoothL1MeanFn.apply(value, target.detach(), float(scale))
 
 
 class _FusedMLP3Fn(torch.autograd.Function):

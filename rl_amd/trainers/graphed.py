@@ -116,6 +116,7 @@ class GraphedPPO:
         if os.environ.get("RL_AMD_FORCE_MB_GRAPH") == "1":
             self._distributed = True
 
+        self._seed_one = None
         self._initialized = False
         self._step_fn: Optional[Callable[[], None]] = None
         self._full_graph = False
@@ -131,9 +132,13 @@ class GraphedPPO:
 
     # ------------------------------------------------------------------ #
     def _total_loss(self, out: TensorDictBase) -> torch.Tensor:
-        total = None
+        # "_loss_actor" is the kernel-side pre-summed
+        # loss_objective + loss_entropy (ClipPPOLoss mega path)
+        presummed = out.get("_loss_actor", None) if hasattr(out, "get") else None
+        skip = {"loss_objective", "loss_entropy"} if presummed is not None else set()
+        total = presummed
         for k in out.keys():
-            if isinstance(k, str) and k.startswith("loss_"):
+            if isinstance(k, str) and k.startswith("loss_") and k not in skip:
                 v = out.get(k)
                 total = v if total is None else total + v
         return total
@@ -147,7 +152,14 @@ class GraphedPPO:
         # replays), and it removes a zero-fill + accumulate-add per
         # parameter per minibatch (~100 kernels/step in the T=16 profile)
         self.optimizer.zero_grad(set_to_none=True)
-        total.backward()
+        if total.is_cuda:
+            if self._seed_one is None or self._seed_one.device != total.device:
+                self._seed_one = torch.ones((), device=total.device)
+            # preallocated seed: backward() otherwise fills a fresh ones
+            # scalar every minibatch
+            total.backward(gradient=self._seed_one)
+        else:
+            total.backward()
 
     def _capture_mb_graph(self, example_sub: TensorDictBase) -> None:
         static_sub = example_sub.clone(False)

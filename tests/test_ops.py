@@ -787,6 +787,8 @@ class TestPPOHeadMega:
         tol = 1e-5 if dtype == torch.float32 else 5e-3
         for i, (a, b) in enumerate(zip(o_f, o_e)):
             assert torch.allclose(a, b, atol=tol, rtol=1e-3), (i, a, b)
+        # 6th output: kernel-side pre-summed actor loss
+        assert torch.allclose(o_f[5], o_f[0] + o_f[1], atol=1e-6)
         (o_f[0] + o_f[1]).backward()
         (o_e[0] + o_e[1]).backward()
         gtol = 1e-5 if dtype == torch.float32 else 1e-2
