@@ -344,6 +344,38 @@ class ClipPPOLoss(PPOLoss):
         )
         return loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act
 
+    def _loss_critic_side_stream(self, td: TensorDictBase) -> torch.Tensor:
+        """Critic loss forward on a side HIP stream: the critic chain
+        (MLP forward, smooth-L1, and — because autograd replays each
+        node on its forward stream — the whole critic backward and its
+        wgrads) runs CONCURRENTLY with the actor chain on the main
+        stream.  Inside a hipGraph capture this records as parallel
+        graph branches.  Trainers must join ``_side_streams`` after
+        backward before touching gradients (GraphedPPO does)."""
+        coeff = float(self.critic_coeff)
+        n = td.batch_size[0] if td.batch_dims >= 1 else 0
+        if not torch.cuda.is_available() or n < 32768:
+            # below ~32K rows the two branches' kernels just contend
+            # for CUs (measured: -4% at 16K rows, +3% at 64K)
+            return self._loss_critic_reduced(td, coeff)
+        cs = self.__dict__.get("_crit_stream")
+        if cs is None:
+            cs = torch.cuda.Stream()
+            self.__dict__["_crit_stream"] = cs
+            self.__dict__["_side_streams"] = [cs]
+        cur = torch.cuda.current_stream()
+        cs.wait_stream(cur)
+        with torch.cuda.stream(cs):
+            loss_critic = self._loss_critic_reduced(td, coeff)
+        cur.wait_stream(cs)
+        if not torch.cuda.is_current_stream_capturing():
+            loss_critic.record_stream(cur)
+        return loss_critic
+
+    @property
+    def _side_streams(self):
+        return self.__dict__.get("_side_streams_list", self.__dict__.get("_side_streams", []))
+
     def forward(self, tensordict: TensorDictBase) -> TensorDictBase:
         td = tensordict.clone(False)
         advantage = td.get(self.tensor_keys.advantage, None)
@@ -370,10 +402,7 @@ class ClipPPOLoss(PPOLoss):
                 batch_size=[],
             )
             if self.critic_coeff is not None and self.critic_coeff > 0:
-                out.set(
-                    "loss_critic",
-                    self._loss_critic_reduced(td, float(self.critic_coeff)),
-                )
+                out.set("loss_critic", self._loss_critic_side_stream(td))
             return out
         log_weight, dist, _ = self._log_weight(td)
         if log_weight.dim() < advantage.dim():

@@ -158,6 +158,10 @@ class GraphedPPO:
             # preallocated seed: backward() otherwise fills a fresh ones
             # scalar every minibatch
             total.backward(gradient=self._seed_one)
+            # join any loss-module side streams (e.g. the ClipPPOLoss
+            # critic branch) before gradients are read
+            for s in getattr(self.loss_module, "_side_streams", []):
+                torch.cuda.current_stream().wait_stream(s)
         else:
             total.backward()
 
