@@ -26,7 +26,7 @@
 #include <hip/hip_runtime.h>
 
 #define RO_THREADS 256
-#define RO_ROWS 8
+#define RO_ROWS 4
 
 namespace {
 
