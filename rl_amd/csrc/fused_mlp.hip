@@ -330,9 +330,9 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
 
   const int erow = (lane >> 4) * 4;  // epilogue row base (+reg)
   const int ecol = lane & 15;
-  const int rblk = (R == 64 ? wave : (wave & 1)) * 16;
-  const int ct0 = (R == 64 ? 0 : (wave >> 1));
-  const int cts = (R == 64 ? 1 : 2);
+  const int rblk = (R == 64 ? wave : (R == 32 ? (wave & 1) : 0)) * 16;
+  const int ct0 = (R == 64 ? 0 : (R == 32 ? (wave >> 1) : wave));
+  const int cts = (R == 64 ? 1 : (R == 32 ? 2 : 4));
   // layer 1 + tanh (global h1 goes out later via a coalesced LDS copy)
   for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};
@@ -440,9 +440,9 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
 
   const int erow = (lane >> 4) * 4;
   const int ecol = lane & 15;
-  const int rblk = (R == 64 ? wave : (wave & 1)) * 16;
-  const int ct0 = (R == 64 ? 0 : (wave >> 1));
-  const int cts = (R == 64 ? 1 : 2);
+  const int rblk = (R == 64 ? wave : (R == 32 ? (wave & 1) : 0)) * 16;
+  const int ct0 = (R == 64 ? 0 : (R == 32 ? (wave >> 1) : wave));
+  const int cts = (R == 64 ? 1 : (R == 32 ? 2 : 4));
   // dh2 = (dY @ W3) * (1 - h2^2)
   for (int ct = ct0; ct < H / 16; ct += cts) {
     mfrag_f acc = {};

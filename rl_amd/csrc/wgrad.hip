@@ -550,8 +550,8 @@ __global__ void wgrad3_reduce_kernel(const Wg3Args args, int slabs) {
 }  // namespace
 
 extern "C" int wgrad3_slab_count(long K) {
-  // ~128 slabs: x3 layers in grid.y still fills the 256 CUs
-  long k_slab = (K + 127) / 128;
+  // ~256 slabs: x3 layers in grid.y
+  long k_slab = (K + 255) / 256;
   k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
   if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
   return (int)((K + k_slab - 1) / k_slab);
@@ -574,7 +574,7 @@ extern "C" void launch_wgrad3(const void* const* dy, const void* const* x,
     a.M[l] = M[l];
   }
   const Wg3Args* args = &a;
-  long k_slab = (K + 127) / 128;
+  long k_slab = (K + 255) / 256;
   k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
   if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
   const int slabs = (int)((K + k_slab - 1) / k_slab);
