@@ -88,7 +88,12 @@ void launch_multi_gather(const void*, int, const long*, long, void*);
 void launch_multi_shuffle(const void*, int, const int*, long, void*);
 void launch_wgrad3(const void* const*, const void* const*, float* const*,
                    float* const*, float* const*, float* const*, const int*,
-                   const int*, long, void*);
+                   const int*, int, long, void*);
+void launch_mlp3_mfma_fwd2(const void*, int, const void* const*,
+                           void* const*, void*, long, int, const int*,
+                           const int*, void*);
+void launch_mlp3_mfma_bwd2(const void* const*, void* const*, long,
+                           const int*, const int*, void*);
 void launch_mlp3_mfma_fwd(const void*, int, const void*, const void*,
                           const void*, const void*, const void*, const void*,
                           void*, void*, void*, void*, long, int, int, int,
@@ -810,22 +815,26 @@ std::vector<torch::Tensor> multi_shuffle(torch::Tensor keys,
 // Batched 3-layer wgrad (csrc/wgrad.hip): the three (dY, X) pairs of a
 // fused-MLP backward in one mfma + one reduce launch.  All dims <= 64
 // and a shared K are required (the PPO/critic MLP shape).
-std::vector<torch::Tensor> wgrad_splitk3(torch::Tensor dy0, torch::Tensor x0,
-                                         torch::Tensor dy1, torch::Tensor x1,
-                                         torch::Tensor dy2,
-                                         torch::Tensor x2) {
-  torch::Tensor dys[3] = {dy0, dy1, dy2};
-  torch::Tensor xs[3] = {x0, x1, x2};
-  const long K = dy0.size(0);
+bool mlp3_mfma_ok(long O, long H, long A2);
+
+std::vector<torch::Tensor> wgrad_splitk_batch(
+    std::vector<torch::Tensor> dys_v, std::vector<torch::Tensor> xs_v) {
+  const int n_layers = (int)dys_v.size();
+  TORCH_CHECK(n_layers >= 1 && n_layers <= 6 &&
+                  xs_v.size() == dys_v.size(),
+              "wgrad batch: 1..6 layer pairs");
+  torch::Tensor* dys = dys_v.data();
+  torch::Tensor* xs = xs_v.data();
+  const long K = dys[0].size(0);
   const int slabs = wgrad3_slab_count(K);
-  const void* dyp[3];
-  const void* xp[3];
-  float *partp[3], *biasp[3], *dwp[3], *dbp[3];
-  int N[3], M[3];
+  const void* dyp[6];
+  const void* xp[6];
+  float *partp[6], *biasp[6], *dwp[6], *dbp[6];
+  int N[6], M[6];
   std::vector<torch::Tensor> out;
   std::vector<torch::Tensor> keep;
-  auto fopt = dy0.options().dtype(torch::kFloat32);
-  for (int l = 0; l < 3; ++l) {
+  auto fopt = dys[0].options().dtype(torch::kFloat32);
+  for (int l = 0; l < n_layers; ++l) {
     TORCH_CHECK(dys[l].scalar_type() == torch::kBFloat16 &&
                     xs[l].scalar_type() == torch::kBFloat16,
                 "wgrad3: bf16 inputs");
@@ -851,8 +860,87 @@ std::vector<torch::Tensor> wgrad_splitk3(torch::Tensor dy0, torch::Tensor x0,
     keep.push_back(bias_part);
   }
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  launch_wgrad3(dyp, xp, partp, biasp, dwp, dbp, N, M, K, (void*)stream);
+  launch_wgrad3(dyp, xp, partp, biasp, dwp, dbp, N, M, n_layers, K,
+                (void*)stream);
   return out;
+}
+
+std::vector<torch::Tensor> wgrad_splitk3(torch::Tensor dy0, torch::Tensor x0,
+                                         torch::Tensor dy1, torch::Tensor x1,
+                                         torch::Tensor dy2,
+                                         torch::Tensor x2) {
+  return wgrad_splitk_batch({dy0, dy1, dy2}, {x0, x1, x2});
+}
+
+// Dual-network (actor+critic) whole-MLP forward/backward: both nets
+// over the SAME input rows in one launch each way (grid.y = net).
+std::vector<torch::Tensor> mlp3_mfma_fwd2(torch::Tensor x,
+                                          std::vector<torch::Tensor> aw,
+                                          std::vector<torch::Tensor> cw) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x cuda contiguous");
+  TORCH_CHECK(aw.size() == 6 && cw.size() == 6, "6 weight tensors per net");
+  const bool xf32 = x.scalar_type() == torch::kFloat32;
+  const long N = x.size(0), O = x.size(1);
+  int H[2] = {(int)aw[0].size(0), (int)cw[0].size(0)};
+  int A2[2] = {(int)aw[4].size(0), (int)cw[4].size(0)};
+  TORCH_CHECK(mlp3_mfma_ok(O, H[0], A2[0]) && mlp3_mfma_ok(O, H[1], A2[1]),
+              "mlp3_mfma_fwd2: unsupported dims");
+  auto bopt = aw[0].options();
+  std::vector<torch::Tensor> res;
+  const void* wp[12];
+  void* op[6];
+  for (int n = 0; n < 2; ++n) {
+    auto& w = n == 0 ? aw : cw;
+    for (int i = 0; i < 6; ++i) {
+      TORCH_CHECK(w[i].scalar_type() == torch::kBFloat16 &&
+                      w[i].is_contiguous(),
+                  "bf16 contiguous weights");
+      wp[n * 6 + i] = w[i].data_ptr();
+    }
+    auto outn = torch::empty({N, A2[n]}, bopt);
+    auto h1n = torch::empty({N, H[n]}, bopt);
+    auto h2n = torch::empty({N, H[n]}, bopt);
+    op[n * 3 + 0] = outn.data_ptr();
+    op[n * 3 + 1] = h1n.data_ptr();
+    op[n * 3 + 2] = h2n.data_ptr();
+    res.push_back(outn);
+    res.push_back(h1n);
+    res.push_back(h2n);
+  }
+  auto xb = xf32 ? torch::empty({N, O}, bopt) : x;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_mlp3_mfma_fwd2(x.data_ptr(), xf32 ? 1 : 0, wp, op, xb.data_ptr(), N,
+                        (int)O, H, A2, (void*)stream);
+  res.push_back(xb);
+  return res;  // a_out, a_h1, a_h2, c_out, c_h1, c_h2, xb
+}
+
+std::vector<torch::Tensor> mlp3_mfma_bwd2(
+    torch::Tensor a_dout, torch::Tensor a_h1, torch::Tensor a_h2,
+    torch::Tensor a_w2, torch::Tensor a_w3, torch::Tensor c_dout,
+    torch::Tensor c_h1, torch::Tensor c_h2, torch::Tensor c_w2,
+    torch::Tensor c_w3) {
+  const long N = a_dout.size(0);
+  int H[2] = {(int)a_h1.size(1), (int)c_h1.size(1)};
+  int A2[2] = {(int)a_dout.size(1), (int)c_dout.size(1)};
+  TORCH_CHECK(a_dout.scalar_type() == torch::kBFloat16 &&
+                  c_dout.scalar_type() == torch::kBFloat16,
+              "bf16 douts");
+  TORCH_CHECK(a_dout.is_contiguous() && c_dout.is_contiguous(), "contiguous");
+  auto bopt = a_h1.options();
+  const void* dp[10] = {a_dout.data_ptr(), a_h1.data_ptr(), a_h2.data_ptr(),
+                        a_w2.data_ptr(),  a_w3.data_ptr(), c_dout.data_ptr(),
+                        c_h1.data_ptr(),  c_h2.data_ptr(), c_w2.data_ptr(),
+                        c_w3.data_ptr()};
+  auto a_dh1 = torch::empty({N, H[0]}, bopt);
+  auto a_dh2 = torch::empty({N, H[0]}, bopt);
+  auto c_dh1 = torch::empty({N, H[1]}, bopt);
+  auto c_dh2 = torch::empty({N, H[1]}, bopt);
+  void* dh[4] = {a_dh1.data_ptr(), a_dh2.data_ptr(), c_dh1.data_ptr(),
+                 c_dh2.data_ptr()};
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_mlp3_mfma_bwd2(dp, dh, N, H, A2, (void*)stream);
+  return {a_dh1, a_dh2, c_dh1, c_dh2};
 }
 
 // MFMA whole-MLP forward/backward (csrc/fused_mlp.hip v2): three
@@ -1040,6 +1128,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlp3_mfma_ok", &mlp3_mfma_ok, "MFMA MLP3 shape eligibility");
   m.def("wgrad_splitk3", &wgrad_splitk3,
         "batched 3-layer split-K wgrad (HIP)");
+  m.def("wgrad_splitk_batch", &wgrad_splitk_batch,
+        "batched 1..6-layer split-K wgrad (HIP)");
+  m.def("mlp3_mfma_fwd2", &mlp3_mfma_fwd2,
+        "dual-network MFMA MLP forward (HIP)");
+  m.def("mlp3_mfma_bwd2", &mlp3_mfma_bwd2,
+        "dual-network MFMA dgrad chain (HIP)");
   m.def("fused_grad_clip_coef", &fused_grad_clip_coef,
         "single-kernel global grad-norm clip coefficient (HIP)");
   m.def("multi_gather", &multi_gather,

@@ -278,7 +278,7 @@ __device__ __forceinline__ void m3_gemm_tile(
 }
 
 template <typename TX, int R>
-__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
+__device__ void mlp3_mfma_fwd_impl(
     const TX* __restrict__ x,               // [N, O]
     const __hip_bfloat16* __restrict__ w1,  // [H, O]
     const __hip_bfloat16* __restrict__ b1,  // [H]
@@ -312,7 +312,8 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
     __hip_bfloat16 v = __hip_bfloat16(0.f);
     if (r < rows && k < O) v = __hip_bfloat16((float)x[(row0 + r) * O + k]);
     s_x[i] = v;
-    if (r < rows && k < O) xb_out[(row0 + r) * O + k] = v;
+    if (xb_out != nullptr && r < rows && k < O)
+      xb_out[(row0 + r) * O + k] = v;
   }
   for (int i = tid; i < H * lx; i += MLP_THREADS) {
     const int j = i / lx, k = i % lx;
@@ -398,8 +399,52 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
   }
 }
 
+template <typename TX, int R>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd_kernel(
+    const TX* __restrict__ x, const __hip_bfloat16* __restrict__ w1,
+    const __hip_bfloat16* __restrict__ b1,
+    const __hip_bfloat16* __restrict__ w2,
+    const __hip_bfloat16* __restrict__ b2,
+    const __hip_bfloat16* __restrict__ w3,
+    const __hip_bfloat16* __restrict__ b3, __hip_bfloat16* __restrict__ out,
+    __hip_bfloat16* __restrict__ h1_out, __hip_bfloat16* __restrict__ h2_out,
+    __hip_bfloat16* __restrict__ xb_out, const int N, const int O,
+    const int H, const int A2) {
+  mlp3_mfma_fwd_impl<TX, R>(x, w1, b1, w2, b2, w3, b3, out, h1_out, h2_out,
+                            xb_out, N, O, H, A2);
+}
+
+// Dual-network forward: blockIdx.y selects the net (0 = actor, 1 =
+// critic).  Both read the SAME input rows; the bf16 input copy for the
+// wgrad is written once (net 0).
+struct AC2Fwd {
+  const __hip_bfloat16* w1[2];
+  const __hip_bfloat16* b1[2];
+  const __hip_bfloat16* w2[2];
+  const __hip_bfloat16* b2[2];
+  const __hip_bfloat16* w3[2];
+  const __hip_bfloat16* b3[2];
+  __hip_bfloat16* out[2];
+  __hip_bfloat16* h1[2];
+  __hip_bfloat16* h2[2];
+  int H[2];
+  int A2[2];
+};
+
+template <typename TX, int R>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd2_kernel(
+    const TX* __restrict__ x, const AC2Fwd args,
+    __hip_bfloat16* __restrict__ xb_out, const int N, const int O) {
+  const int net = blockIdx.y;
+  mlp3_mfma_fwd_impl<TX, R>(x, args.w1[net], args.b1[net], args.w2[net],
+                            args.b2[net], args.w3[net], args.b3[net],
+                            args.out[net], args.h1[net], args.h2[net],
+                            net == 0 ? xb_out : nullptr, N, O, args.H[net],
+                            args.A2[net]);
+}
+
 template <typename TD, int R>
-__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
+__device__ void mlp3_mfma_bwd_impl(
     const TD* __restrict__ dout,            // [N, A2]
     const __hip_bfloat16* __restrict__ h1,  // [N, H]
     const __hip_bfloat16* __restrict__ h2,  // [N, H]
@@ -494,6 +539,38 @@ __global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
   }
 }
 
+template <typename TD, int R>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd_kernel(
+    const TD* __restrict__ dout, const __hip_bfloat16* __restrict__ h1,
+    const __hip_bfloat16* __restrict__ h2,
+    const __hip_bfloat16* __restrict__ w2,
+    const __hip_bfloat16* __restrict__ w3, __hip_bfloat16* __restrict__ dh1,
+    __hip_bfloat16* __restrict__ dh2, const int N, const int H,
+    const int A2) {
+  mlp3_mfma_bwd_impl<TD, R>(dout, h1, h2, w2, w3, dh1, dh2, N, H, A2);
+}
+
+struct AC2Bwd {
+  const __hip_bfloat16* dout[2];
+  const __hip_bfloat16* h1[2];
+  const __hip_bfloat16* h2[2];
+  const __hip_bfloat16* w2[2];
+  const __hip_bfloat16* w3[2];
+  __hip_bfloat16* dh1[2];
+  __hip_bfloat16* dh2[2];
+  int H[2];
+  int A2[2];
+};
+
+template <int R>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd2_kernel(
+    const AC2Bwd args, const int N) {
+  const int net = blockIdx.y;
+  mlp3_mfma_bwd_impl<__hip_bfloat16, R>(
+      args.dout[net], args.h1[net], args.h2[net], args.w2[net], args.w3[net],
+      args.dh1[net], args.dh2[net], N, args.H[net], args.A2[net]);
+}
+
 }  // namespace
 
 extern "C" int mlp3_mfma_lds_bytes(int O, int H, int A2) {
@@ -561,4 +638,89 @@ extern "C" void launch_mlp3_mfma_bwd(const void* dout, int d_is_f32,
     else M3_LAUNCH_BWD(__hip_bfloat16, 32);
   }
 #undef M3_LAUNCH_BWD
+}
+
+extern "C" void launch_mlp3_mfma_fwd2(const void* x, int x_is_f32,
+                                      const void* const* w,  // 12 ptrs
+                                      void* const* o,        // 6 ptrs
+                                      void* xb, long N, int O, const int* H,
+                                      const int* A2, void* stream) {
+  AC2Fwd a;
+  for (int n = 0; n < 2; ++n) {
+    a.w1[n] = (const __hip_bfloat16*)w[n * 6 + 0];
+    a.b1[n] = (const __hip_bfloat16*)w[n * 6 + 1];
+    a.w2[n] = (const __hip_bfloat16*)w[n * 6 + 2];
+    a.b2[n] = (const __hip_bfloat16*)w[n * 6 + 3];
+    a.w3[n] = (const __hip_bfloat16*)w[n * 6 + 4];
+    a.b3[n] = (const __hip_bfloat16*)w[n * 6 + 5];
+    a.out[n] = (__hip_bfloat16*)o[n * 3 + 0];
+    a.h1[n] = (__hip_bfloat16*)o[n * 3 + 1];
+    a.h2[n] = (__hip_bfloat16*)o[n * 3 + 2];
+    a.H[n] = H[n];
+    a.A2[n] = A2[n];
+  }
+  const int R = N >= M3_ROWS_SWITCH ? 64 : 32;
+  const int blocks = (int)((N + R - 1) / R);
+  int lds = 0;
+  for (int n = 0; n < 2; ++n) {
+    const int Op = (O + 31) & ~31, Hp = (H[n] + 31) & ~31;
+    const int lx = Op + 8, lh = Hp + 8, A2p = (A2[n] + 15) & ~15;
+    const int b = 2 * (R * lx + H[n] * lx + 2 * R * lh + H[n] * lh + A2p * lh);
+    if (b > lds) lds = b;
+  }
+  dim3 grid(blocks, 2);
+  if (x_is_f32) {
+    if (R == 64)
+      hipLaunchKernelGGL((mlp3_mfma_fwd2_kernel<float, 64>), grid,
+                         dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                         (const float*)x, a, (__hip_bfloat16*)xb, (int)N, O);
+    else
+      hipLaunchKernelGGL((mlp3_mfma_fwd2_kernel<float, 32>), grid,
+                         dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                         (const float*)x, a, (__hip_bfloat16*)xb, (int)N, O);
+  } else {
+    if (R == 64)
+      hipLaunchKernelGGL((mlp3_mfma_fwd2_kernel<__hip_bfloat16, 64>), grid,
+                         dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                         (const __hip_bfloat16*)x, a, (__hip_bfloat16*)xb,
+                         (int)N, O);
+    else
+      hipLaunchKernelGGL((mlp3_mfma_fwd2_kernel<__hip_bfloat16, 32>), grid,
+                         dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                         (const __hip_bfloat16*)x, a, (__hip_bfloat16*)xb,
+                         (int)N, O);
+  }
+}
+
+extern "C" void launch_mlp3_mfma_bwd2(const void* const* dptr,  // 10 ptrs
+                                      void* const* dh,          // 4 ptrs
+                                      long N, const int* H, const int* A2,
+                                      void* stream) {
+  AC2Bwd a;
+  for (int n = 0; n < 2; ++n) {
+    a.dout[n] = (const __hip_bfloat16*)dptr[n * 5 + 0];
+    a.h1[n] = (const __hip_bfloat16*)dptr[n * 5 + 1];
+    a.h2[n] = (const __hip_bfloat16*)dptr[n * 5 + 2];
+    a.w2[n] = (const __hip_bfloat16*)dptr[n * 5 + 3];
+    a.w3[n] = (const __hip_bfloat16*)dptr[n * 5 + 4];
+    a.dh1[n] = (__hip_bfloat16*)dh[n * 2 + 0];
+    a.dh2[n] = (__hip_bfloat16*)dh[n * 2 + 1];
+    a.H[n] = H[n];
+    a.A2[n] = A2[n];
+  }
+  const int R = N >= M3_ROWS_SWITCH ? 64 : 32;
+  const int blocks = (int)((N + R - 1) / R);
+  int lds = 0;
+  for (int n = 0; n < 2; ++n) {
+    const int ld = ((A2[n] + 31) & ~31) + 8, lh = ((H[n] + 31) & ~31) + 8;
+    const int b = 2 * (R * ld + H[n] * ld + 2 * R * lh + H[n] * lh);
+    if (b > lds) lds = b;
+  }
+  dim3 grid(blocks, 2);
+  if (R == 64)
+    hipLaunchKernelGGL((mlp3_mfma_bwd2_kernel<64>), grid, dim3(MLP_THREADS),
+                       lds, (hipStream_t)stream, a, (int)N);
+  else
+    hipLaunchKernelGGL((mlp3_mfma_bwd2_kernel<32>), grid, dim3(MLP_THREADS),
+                       lds, (hipStream_t)stream, a, (int)N);
 }

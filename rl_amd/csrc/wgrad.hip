@@ -409,14 +409,14 @@ extern "C" void launch_wgrad_splitk(const void* dy, const void* x, float* dw,
 // ---------------------------------------------------------------------------
 
 struct Wg3Args {
-  const __hip_bfloat16* dy[3];
-  const __hip_bfloat16* x[3];
-  float* part[3];       // [slabs, 64*64] each
-  float* bias_part[3];  // [slabs, 64] each
-  float* dw[3];
-  float* db[3];
-  int N[3];
-  int M[3];
+  const __hip_bfloat16* dy[6];
+  const __hip_bfloat16* x[6];
+  float* part[6];       // [slabs, 64*64] each
+  float* bias_part[6];  // [slabs, 64] each
+  float* dw[6];
+  float* db[6];
+  int N[6];
+  int M[6];
 };
 
 namespace {
@@ -560,10 +560,10 @@ extern "C" int wgrad3_slab_count(long K) {
 extern "C" void launch_wgrad3(const void* const* dy, const void* const* x,
                               float* const* part, float* const* bias_part,
                               float* const* dw, float* const* db,
-                              const int* N, const int* M, long K,
-                              void* stream) {
+                              const int* N, const int* M, int n_layers,
+                              long K, void* stream) {
   Wg3Args a;
-  for (int l = 0; l < 3; ++l) {
+  for (int l = 0; l < n_layers; ++l) {
     a.dy[l] = (const __hip_bfloat16*)dy[l];
     a.x[l] = (const __hip_bfloat16*)x[l];
     a.part[l] = part[l];
@@ -578,9 +578,10 @@ extern "C" void launch_wgrad3(const void* const* dy, const void* const* x,
   k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
   if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
   const int slabs = (int)((K + k_slab - 1) / k_slab);
-  hipLaunchKernelGGL(wgrad3_mfma_kernel, dim3(slabs, 3), dim3(WG_THREADS), 0,
-                     (hipStream_t)stream, *args, K, (int)k_slab);
-  const int blocks = 512;  // 3 * 512 WGs cover the chip for the reduce
-  hipLaunchKernelGGL(wgrad3_reduce_kernel, dim3(blocks, 3), dim3(256), 0,
-                     (hipStream_t)stream, *args, slabs);
+  hipLaunchKernelGGL(wgrad3_mfma_kernel, dim3(slabs, n_layers),
+                     dim3(WG_THREADS), 0, (hipStream_t)stream, *args, K,
+                     (int)k_slab);
+  const int blocks = 512;  // n * 512 WGs cover the chip for the reduce
+  hipLaunchKernelGGL(wgrad3_reduce_kernel, dim3(blocks, n_layers), dim3(256),
+                     0, (hipStream_t)stream, *args, slabs);
 }

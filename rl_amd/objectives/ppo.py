@@ -284,7 +284,26 @@ class ClipPPOLoss(PPOLoss):
                         "obs_key": tdm.in_keys[0],
                         "sp_bias": float(ext._inv_softplus_bias),
                         "scale_lb": float(ext.scale_lb),
+                        "ac_pair": None,
                     }
+                    # actor+critic dual-kernel pair: both bodies are
+                    # FusedMLP3 over the same observation key
+                    try:
+                        from ..ops import FusedMLP3
+
+                        crit = self.critic_network
+                        body = info["body"]
+                        if (
+                            len(body) == 1
+                            and isinstance(body[0], FusedMLP3)
+                            and isinstance(crit, TensorDictModule)
+                            and list(crit.in_keys) == [info["obs_key"]]
+                            and isinstance(crit.module, FusedMLP3)
+                            and crit.module.lin3.out_features == 1
+                        ):
+                            info["ac_pair"] = (body[0], crit.module)
+                    except Exception:
+                        pass
         except Exception:
             info = False
         self.__dict__["_mega_info"] = info
@@ -320,9 +339,32 @@ class ClipPPOLoss(PPOLoss):
         N = action.numel() // A
         if advantage.numel() != N or prev_lp.numel() != N:
             return None
-        head = obs
-        for m in info["body"]:
-            head = m(head)
+        value = None
+        ac = info.get("ac_pair")
+        if (
+            ac is not None
+            and self.clip_value is None
+            and self.loss_critic_type in ("smooth_l1", "huber")
+            and self.critic_coeff is not None
+            and self.critic_coeff > 0
+        ):
+            target = td.get(self.tensor_keys.value_target, None)
+            O = obs.shape[-1]
+            if (
+                target is not None
+                and target.dtype == torch.float32
+                and target.numel() == N
+                and ops.actor_critic_mlp3_ok(ac[0], ac[1], O)
+                and getattr(ac[0].lin1, "_bf16_cache", False)
+                and getattr(ac[1].lin1, "_bf16_cache", False)
+            ):
+                head, value = ops.actor_critic_mlp3(
+                    obs.reshape(N, O), ac[0], ac[1]
+                )
+        if value is None:
+            head = obs
+            for m in info["body"]:
+                head = m(head)
         if head.dtype not in (torch.float32, torch.bfloat16):
             return None
         if head.shape[-1] != 2 * A:
@@ -342,7 +384,14 @@ class ClipPPOLoss(PPOLoss):
             entropy_coeff=float(self.entropy_coeff),
             normalize=normalize,
         )
-        return loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act
+        loss_critic = None
+        if value is not None:
+            loss_critic = ops.smooth_l1_mean(
+                value, td.get(self.tensor_keys.value_target),
+                float(self.critic_coeff),
+            )
+        return (loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act,
+                loss_critic)
 
     def _loss_critic_side_stream(self, td: TensorDictBase) -> torch.Tensor:
         """Critic loss forward on a side HIP stream: the critic chain
@@ -387,7 +436,8 @@ class ClipPPOLoss(PPOLoss):
         normalize = self.normalize_advantage and advantage.numel() > 1
         mega = self._mega_head_loss(td, advantage, normalize)
         if mega is not None:
-            loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act = mega
+            (loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act,
+             loss_critic) = mega
             out = TensorDict(
                 {
                     "loss_objective": loss_obj,
@@ -401,7 +451,9 @@ class ClipPPOLoss(PPOLoss):
                 },
                 batch_size=[],
             )
-            if self.critic_coeff is not None and self.critic_coeff > 0:
+            if loss_critic is not None:
+                out.set("loss_critic", loss_critic)
+            elif self.critic_coeff is not None and self.critic_coeff > 0:
                 out.set("loss_critic", self._loss_critic_side_stream(td))
             return out
         log_weight, dist, _ = self._log_weight(td)

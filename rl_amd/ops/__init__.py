@@ -965,6 +965,84 @@ class FusedMLP3(torch.nn.Module):
         return self.eager(x)
 
 
+class _FusedACFn(torch.autograd.Function):
+    """Actor + critic whole-MLP forward/backward fused across networks
+    (csrc/fused_mlp.hip fwd2/bwd2 + the 6-layer batched wgrad): both
+    nets read the SAME input rows, so one launch each way covers the
+    entire minibatch model compute of a PPO update."""
+
+    @staticmethod
+    def forward(ctx, x, aw1b, ab1b, aw2b, ab2b, aw3b, ab3b,
+                cw1b, cb1b, cw2b, cb2b, cw3b, cb3b,
+                aw1, ab1, aw2, ab2, aw3, ab3,
+                cw1, cb1, cw2, cb2, cw3, cb3):
+        x = x.contiguous()
+        head, a_h1, a_h2, value, c_h1, c_h2, xb = _C.mlp3_mfma_fwd2(
+            x,
+            [aw1b, ab1b, aw2b, ab2b, aw3b, ab3b],
+            [cw1b, cb1b, cw2b, cb2b, cw3b, cb3b],
+        )
+        ctx.save_for_backward(xb, a_h1, a_h2, aw2b, aw3b, c_h1, c_h2, cw2b,
+                              cw3b)
+        return head, value
+
+    @staticmethod
+    def backward(ctx, dhead, dvalue):
+        (xb, a_h1, a_h2, aw2b, aw3b, c_h1, c_h2, cw2b,
+         cw3b) = ctx.saved_tensors
+        dhead = dhead.contiguous()
+        dvalue = dvalue.contiguous()
+        if dhead.dtype != torch.bfloat16:
+            dhead = dhead.to(torch.bfloat16)
+        if dvalue.dtype != torch.bfloat16:
+            dvalue = dvalue.to(torch.bfloat16)
+        a_dh1, a_dh2, c_dh1, c_dh2 = _C.mlp3_mfma_bwd2(
+            dhead, a_h1, a_h2, aw2b, aw3b, dvalue, c_h1, c_h2, cw2b, cw3b
+        )
+        (adw3, adb3, adw2, adb2, adw1, adb1, cdw3, cdb3, cdw2, cdb2, cdw1,
+         cdb1) = _C.wgrad_splitk_batch(
+            [dhead, a_dh2, a_dh1, dvalue, c_dh2, c_dh1],
+            [a_h2, a_h1, xb, c_h2, c_h1, xb],
+        )
+        return (None,) * 13 + (adw1, adb1, adw2, adb2, adw3, adb3,
+                               cdw1, cdb1, cdw2, cdb2, cdw3, cdb3)
+
+
+def actor_critic_mlp3(x, actor_fused, critic_fused):
+    """Run two :class:`FusedMLP3` networks (same input) as ONE launch
+    each way.  Returns ``(actor_head, critic_out)``."""
+    _require_ext()
+    a, c = actor_fused, critic_fused
+    return _FusedACFn.apply(
+        x,
+        a.lin1.weight_bf16, a.lin1.bias_bf16, a.lin2.weight_bf16,
+        a.lin2.bias_bf16, a.lin3.weight_bf16, a.lin3.bias_bf16,
+        c.lin1.weight_bf16, c.lin1.bias_bf16, c.lin2.weight_bf16,
+        c.lin2.bias_bf16, c.lin3.weight_bf16, c.lin3.bias_bf16,
+        a.lin1.weight, a.lin1.bias, a.lin2.weight, a.lin2.bias,
+        a.lin3.weight, a.lin3.bias,
+        c.lin1.weight, c.lin1.bias, c.lin2.weight, c.lin2.bias,
+        c.lin3.weight, c.lin3.bias,
+    )
+
+
+def actor_critic_mlp3_ok(actor_fused, critic_fused, in_features: int) -> bool:
+    """Shape eligibility for the dual-network kernels."""
+    if not HAS_HIP_EXT:
+        return False
+    try:
+        return bool(
+            _C.mlp3_mfma_ok(in_features, actor_fused.lin1.out_features,
+                            actor_fused.lin3.out_features)
+            and _C.mlp3_mfma_ok(in_features, critic_fused.lin1.out_features,
+                                critic_fused.lin3.out_features)
+            and actor_fused.lin1.in_features == in_features
+            and critic_fused.lin1.in_features == in_features
+        )
+    except Exception:
+        return False
+
+
 def fuse_mlp3(module: torch.nn.Module) -> torch.nn.Module:
     """Wrap a Sequential [SplitKLinear, Tanh, SplitKLinear, Tanh,
     SplitKLinear] (e.g. rl_amd MLP internals after
