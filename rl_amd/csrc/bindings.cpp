@@ -83,7 +83,9 @@ void launch_ppo_head_bwd(const void*, const float*, const float*,
                          void*);
 int mlp3_mfma_lds_bytes(int, int, int);
 int wgrad3_slab_count(long);
-void launch_grad_clip_coef(const void*, int, float, float*, float*, void*);
+int wgrad3_slab_count_n(long, int);
+void launch_grad_clip_coef(const void*, int, float, float*, float*, int,
+                           void*);
 void launch_multi_gather(const void*, int, const long*, long, void*);
 void launch_multi_shuffle(const void*, int, const int*, long, void*);
 void launch_wgrad3(const void* const*, const void* const*, float* const*,
@@ -723,7 +725,8 @@ torch::Tensor smooth_l1_bwd(torch::Tensor v, torch::Tensor t,
 // the clamped scale coefficient (torch's clip_grad_norm_ runs ~10
 // launches for the same).  Apply with torch._foreach_mul_(grads, coef).
 torch::Tensor fused_grad_clip_coef(std::vector<torch::Tensor> grads,
-                                   double max_norm) {
+                                   double max_norm, bool inverse,
+                                   torch::Tensor out) {
   struct {
     const float* g[32];
     int len[32];
@@ -737,12 +740,12 @@ torch::Tensor fused_grad_clip_coef(std::vector<torch::Tensor> grads,
     args.g[i] = grads[i].data_ptr<float>();
     args.len[i] = (int)grads[i].numel();
   }
-  auto coef = torch::empty({}, grads[0].options());
+  auto coef = out.numel() ? out : torch::empty({}, grads[0].options());
   auto part = torch::empty({32}, grads[0].options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_grad_clip_coef(&args, (int)grads.size(), (float)max_norm,
                         part.data_ptr<float>(), coef.data_ptr<float>(),
-                        (void*)stream);
+                        inverse ? 1 : 0, (void*)stream);
   return coef;
 }
 
@@ -826,7 +829,7 @@ std::vector<torch::Tensor> wgrad_splitk_batch(
   torch::Tensor* dys = dys_v.data();
   torch::Tensor* xs = xs_v.data();
   const long K = dys[0].size(0);
-  const int slabs = wgrad3_slab_count(K);
+  const int slabs = wgrad3_slab_count_n(K, n_layers);
   const void* dyp[6];
   const void* xp[6];
   float *partp[6], *biasp[6], *dwp[6], *dbp[6];

@@ -721,13 +721,20 @@ __global__ void grad_clip_partials_k(const ClipArgs args, const int nt,
 
 __global__ void grad_clip_finalize_k(const float* __restrict__ part,
                                      const float max_norm,
-                                     float* __restrict__ coef) {
+                                     float* __restrict__ coef,
+                                     const int inverse) {
   float ss = (threadIdx.x < CLIP_BLOCKS) ? part[threadIdx.x] : 0.f;
   __shared__ float smem[8];
   const float tot = block_sum(ss, smem);
   if (threadIdx.x == 0) {
-    const float c = max_norm / (sqrtf(tot) + 1e-6f);
-    coef[0] = c < 1.f ? c : 1.f;
+    if (inverse) {
+      // fused-Adam grad_scale form: grads are DIVIDED by this
+      const float c = (sqrtf(tot) + 1e-6f) / max_norm;
+      coef[0] = c > 1.f ? c : 1.f;
+    } else {
+      const float c = max_norm / (sqrtf(tot) + 1e-6f);
+      coef[0] = c < 1.f ? c : 1.f;
+    }
   }
 }
 
@@ -788,12 +795,13 @@ __global__ void multi_shuffle_k(const GatherArgs args,
 
 extern "C" void launch_grad_clip_coef(const void* args, int nt,
                                       float max_norm, float* part,
-                                      float* coef, void* stream) {
+                                      float* coef, int inverse,
+                                      void* stream) {
   hipLaunchKernelGGL(grad_clip_partials_k, dim3(CLIP_BLOCKS),
                      dim3(LP_THREADS), 0, (hipStream_t)stream,
                      *(const ClipArgs*)args, nt, part);
   hipLaunchKernelGGL(grad_clip_finalize_k, dim3(1), dim3(64), 0,
-                     (hipStream_t)stream, part, max_norm, coef);
+                     (hipStream_t)stream, part, max_norm, coef, inverse);
 }
 
 extern "C" void launch_multi_gather(const void* args, int nt,

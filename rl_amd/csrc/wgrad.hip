@@ -549,13 +549,18 @@ __global__ void wgrad3_reduce_kernel(const Wg3Args args, int slabs) {
 
 }  // namespace
 
-extern "C" int wgrad3_slab_count(long K) {
-  // ~256 slabs: x3 layers in grid.y
-  long k_slab = (K + 255) / 256;
+extern "C" int wgrad3_slab_count_n(long K, int n_layers) {
+  // target ~768 workgroups TOTAL across the layer axis: partials
+  // traffic (written + re-read by the reduce) scales with slabs, so
+  // more layers per launch means fewer slabs each
+  long target = 768 / (n_layers > 0 ? n_layers : 1);
+  long k_slab = (K + target - 1) / target;
   k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
   if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
   return (int)((K + k_slab - 1) / k_slab);
 }
+
+extern "C" int wgrad3_slab_count(long K) { return wgrad3_slab_count_n(K, 3); }
 
 extern "C" void launch_wgrad3(const void* const* dy, const void* const* x,
                               float* const* part, float* const* bias_part,
@@ -574,7 +579,8 @@ extern "C" void launch_wgrad3(const void* const* dy, const void* const* x,
     a.M[l] = M[l];
   }
   const Wg3Args* args = &a;
-  long k_slab = (K + 255) / 256;
+  long target = 768 / (n_layers > 0 ? n_layers : 1);
+  long k_slab = (K + target - 1) / target;
   k_slab = ((k_slab + ROW_CHUNK - 1) / ROW_CHUNK) * ROW_CHUNK;
   if (k_slab < ROW_CHUNK) k_slab = ROW_CHUNK;
   const int slabs = (int)((K + k_slab - 1) / k_slab);

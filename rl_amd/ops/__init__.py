@@ -745,8 +745,28 @@ def fused_grad_clip_(parameters, max_norm: float) -> bool:
     for g in grads:
         if not (g.is_cuda and g.dtype == torch.float32 and g.is_contiguous()):
             return False
-    coef = _C.fused_grad_clip_coef(grads, float(max_norm))
+    coef = _C.fused_grad_clip_coef(
+        grads, float(max_norm), False, torch.empty(0)
+    )
     torch._foreach_mul_(grads, coef)
+    return True
+
+
+def fused_grad_clip_scale_(parameters, max_norm: float,
+                           out_scale: torch.Tensor) -> bool:
+    """Write the INVERSE clip coefficient ``max(1, norm/max_norm)``
+    into ``out_scale`` (a persistent 0-d fp32 cuda tensor) — exactly
+    the ``grad_scale`` divisor torch's fused Adam consumes, so the
+    clip costs two launches and NO gradient-multiply at all."""
+    if not HAS_HIP_EXT:
+        return False
+    grads = [p.grad for p in parameters if p.grad is not None]
+    if not grads or len(grads) > 32:
+        return False
+    for g in grads:
+        if not (g.is_cuda and g.dtype == torch.float32 and g.is_contiguous()):
+            return False
+    _C.fused_grad_clip_coef(grads, float(max_norm), True, out_scale)
     return True
 
 

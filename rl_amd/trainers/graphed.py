@@ -117,6 +117,19 @@ class GraphedPPO:
             self._distributed = True
 
         self._seed_one = None
+        self._clip_scale = None
+        if (
+            self._cuda
+            and max_grad_norm
+            and all(g.get("fused") for g in optimizer.param_groups)
+            and isinstance(optimizer, torch.optim.Adam)
+        ):
+            try:
+                dev = next(p.device for p in self._params)
+                self._clip_scale = torch.ones((), device=dev)
+                optimizer.grad_scale = self._clip_scale
+            except Exception:
+                self._clip_scale = None
         self._initialized = False
         self._step_fn: Optional[Callable[[], None]] = None
         self._full_graph = False
@@ -212,11 +225,15 @@ class GraphedPPO:
         if self.reducer is not None:
             self.reducer.finalize()
         if self.max_grad_norm:
-            # single-kernel global-norm clip (csrc/loss_ops.hip);
-            # torch's foreach chain is ~10 launches for the same
             from .. import ops
 
-            if not ops.fused_grad_clip_(self._params, self.max_grad_norm):
+            if self._clip_scale is not None:
+                # the clip coefficient rides into the fused Adam as its
+                # grad_scale divisor: no gradient multiply at all
+                ops.fused_grad_clip_scale_(
+                    self._params, self.max_grad_norm, self._clip_scale
+                )
+            elif not ops.fused_grad_clip_(self._params, self.max_grad_norm):
                 torch.nn.utils.clip_grad_norm_(self._params, self.max_grad_norm)
         self.optimizer.step()
         if self.post_optim_hook is not None:

@@ -864,3 +864,29 @@ class TestActorCriticFused:
             assert torch.allclose(g1, p.grad, atol=1e-4, rtol=1e-3), (
                 (g1 - p.grad).abs().max()
             )
+
+
+@pytest.mark.gpu
+def test_grad_scale_clip_equals_explicit_clip():
+    """GraphedPPO's clip-via-grad_scale: fused Adam with grad_scale =
+    max(1, norm/max_norm) must match clip_grad_norm_ + step."""
+    from rl_amd import ops
+
+    torch.manual_seed(0)
+    p1 = torch.randn(100, device="cuda", requires_grad=True)
+    p2 = p1.detach().clone().requires_grad_()
+    g = torch.randn(100, device="cuda") * 3
+    max_norm = 1.0
+
+    o1 = torch.optim.Adam([p1], lr=1e-2, fused=True)
+    p1.grad = g.clone()
+    scale = torch.ones((), device="cuda")
+    o1.grad_scale = scale
+    assert ops.fused_grad_clip_scale_([p1], max_norm, scale)
+    o1.step()
+
+    o2 = torch.optim.Adam([p2], lr=1e-2, fused=True)
+    p2.grad = g.clone()
+    torch.nn.utils.clip_grad_norm_([p2], max_norm)
+    o2.step()
+    assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
