@@ -73,14 +73,16 @@ void launch_smooth_l1_fwd(const void*, const float*, float*, float*, float,
 void launch_smooth_l1_bwd(const void*, const float*, const float*, void*,
                           float, long, int, void*);
 void launch_ppo_head_fwd(const void*, const float*, const float*,
-                         const float*, const float*, const float*, float*,
-                         float* const*, float, float, float, float, float,
-                         long, int, int, void*);
+                         const float*, const float*, const float*,
+                         const void*, const float*, float*, float* const*,
+                         float, float, float, float, float, float, long, int,
+                         int, void*);
 void launch_ppo_head_bwd(const void*, const float*, const float*,
                          const float*, const float*, const float*,
-                         const float*, const float*, const float*, void*,
-                         float, float, float, float, float, long, int, int,
-                         void*);
+                         const void*, const float*, const float*,
+                         const float*, const float*, const float*,
+                         const float*, void*, void*, float, float, float,
+                         float, float, float, long, int, int, void*);
 int mlp3_mfma_lds_bytes(int, int, int);
 int wgrad3_slab_count(long);
 int wgrad3_slab_count_n(long, int);
@@ -1009,9 +1011,12 @@ std::vector<torch::Tensor> mlp3_mfma_bwd(torch::Tensor dout, torch::Tensor h1,
 std::vector<torch::Tensor> ppo_head_fwd(torch::Tensor head,
                                         torch::Tensor action,
                                         torch::Tensor eps, torch::Tensor prev,
-                                        torch::Tensor adv, double sp_bias,
-                                        double lb, double lo, double hi,
-                                        double ent_coeff, bool normalize) {
+                                        torch::Tensor adv,
+                                        torch::Tensor value,
+                                        torch::Tensor vtarget,
+                                        double sp_bias, double lb, double lo,
+                                        double hi, double ent_coeff,
+                                        double crit_scale, bool normalize) {
   TORCH_CHECK(head.is_cuda() && head.is_contiguous(), "head cuda contiguous");
   const bool bf16 = head.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(bf16 || head.scalar_type() == torch::kFloat32,
@@ -1030,11 +1035,19 @@ std::vector<torch::Tensor> ppo_head_fwd(torch::Tensor head,
                   eps.numel() == N * A && prev.numel() == N &&
                   adv.numel() == N,
               "shape mismatch");
+  const bool has_crit = value.numel() > 0;
+  if (has_crit) {
+    TORCH_CHECK(value.numel() == N && vtarget.numel() == N &&
+                    value.scalar_type() == head.scalar_type() &&
+                    vtarget.scalar_type() == torch::kFloat32 &&
+                    value.is_contiguous() && vtarget.is_contiguous(),
+                "ppo_head_fwd: critic inputs");
+  }
   auto opt = action.options();
-  auto part = torch::empty({256 * 5}, opt);
+  auto part = torch::empty({256 * 6}, opt);
   std::vector<torch::Tensor> outs;
-  float* outp[6];
-  for (int i = 0; i < 6; ++i) {
+  float* outp[8];
+  for (int i = 0; i < 8; ++i) {
     outs.push_back(torch::empty({}, opt));
     outp[i] = outs.back().data_ptr<float>();
   }
@@ -1047,34 +1060,45 @@ std::vector<torch::Tensor> ppo_head_fwd(torch::Tensor head,
                       eps.data_ptr<float>(), prev.data_ptr<float>(),
                       adv.data_ptr<float>(),
                       normalize ? stats.data_ptr<float>() : nullptr,
+                      has_crit ? value.data_ptr() : nullptr,
+                      has_crit ? vtarget.data_ptr<float>() : nullptr,
                       part.data_ptr<float>(), outp, (float)sp_bias,
-                      (float)lb, (float)lo, (float)hi, (float)ent_coeff, N,
-                      A, bf16 ? 1 : 0, (void*)stream);
+                      (float)lb, (float)lo, (float)hi, (float)ent_coeff,
+                      (float)crit_scale, N, A, bf16 ? 1 : 0, (void*)stream);
   outs.push_back(stats);
-  return outs;
+  return outs;  // 8 scalars + stats
 }
 
-torch::Tensor ppo_head_bwd(torch::Tensor head, torch::Tensor action,
-                           torch::Tensor eps, torch::Tensor prev,
-                           torch::Tensor adv, torch::Tensor stats,
-                           torch::Tensor gobj, torch::Tensor gent,
-                           torch::Tensor gact, double sp_bias, double lb,
-                           double lo, double hi, double ent_coeff) {
+std::vector<torch::Tensor> ppo_head_bwd(
+    torch::Tensor head, torch::Tensor action, torch::Tensor eps,
+    torch::Tensor prev, torch::Tensor adv, torch::Tensor stats,
+    torch::Tensor value, torch::Tensor vtarget, torch::Tensor gobj,
+    torch::Tensor gent, torch::Tensor gact, torch::Tensor gcrit,
+    torch::Tensor gtot, double sp_bias, double lb, double lo, double hi,
+    double ent_coeff, double crit_scale) {
   const bool bf16 = head.scalar_type() == torch::kBFloat16;
   const long N = head.size(0);
   const int A = (int)(head.size(1) / 2);
+  const bool has_crit = value.numel() > 0;
   auto dhead = torch::empty_like(head);
+  auto dvalue = has_crit ? torch::empty_like(value)
+                         : torch::empty({0}, head.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_ppo_head_bwd(
       head.data_ptr(), action.data_ptr<float>(), eps.data_ptr<float>(),
       prev.data_ptr<float>(), adv.data_ptr<float>(),
       stats.numel() ? stats.data_ptr<float>() : nullptr,
+      has_crit ? value.data_ptr() : nullptr,
+      has_crit ? vtarget.data_ptr<float>() : nullptr,
       gobj.numel() ? gobj.data_ptr<float>() : nullptr,
       gent.numel() ? gent.data_ptr<float>() : nullptr,
-      gact.numel() ? gact.data_ptr<float>() : nullptr, dhead.data_ptr(),
-      (float)sp_bias, (float)lb, (float)lo, (float)hi, (float)ent_coeff, N, A,
+      gact.numel() ? gact.data_ptr<float>() : nullptr,
+      gcrit.numel() ? gcrit.data_ptr<float>() : nullptr,
+      gtot.numel() ? gtot.data_ptr<float>() : nullptr, dhead.data_ptr(),
+      has_crit ? dvalue.data_ptr() : nullptr, (float)sp_bias, (float)lb,
+      (float)lo, (float)hi, (float)ent_coeff, (float)crit_scale, N, A,
       bf16 ? 1 : 0, (void*)stream);
-  return dhead;
+  return {dhead, dvalue};
 }
 #endif  // RL_AMD_WITH_HIP
 

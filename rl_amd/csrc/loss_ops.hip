@@ -473,15 +473,22 @@ __global__ void ppo_head_fwd_partials_k(
     const float* __restrict__ prev,   // [N] behavior log-prob
     const float* __restrict__ adv,    // [N]
     const float* __restrict__ stats,  // nullable (mu, 1/sigma)
-    float* __restrict__ part,         // [nWG, 5]
+    const TV* __restrict__ value,     // nullable [N] critic output
+    const float* __restrict__ vtarget,  // nullable [N]
+    float* __restrict__ part,         // [nWG, 6]
     const float sp_bias, const float lb, const float lo, const float hi,
     const long N, const int A) {
   const float mu = stats ? stats[0] : 0.f;
   const float isd = stats ? stats[1] : 1.f;
-  float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f;
+  float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f, sv = 0.f;
   for (long n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
        n += (long)gridDim.x * blockDim.x) {
     float lp = 0.f, ent = 0.f;
+    if (value) {
+      const float z = (float)value[n] - vtarget[n];
+      const float az = fabsf(z);
+      sv += (az < 1.f) ? 0.5f * z * z : az - 0.5f;
+    }
     for (int a = 0; a < A; ++a) {
       const float loc = (float)head[n * 2 * A + a];
       const float spre = (float)head[n * 2 * A + A + a] + sp_bias;
@@ -509,37 +516,44 @@ __global__ void ppo_head_fwd_partials_k(
   __shared__ float smem[8];
   float t;
   t = block_sum(sg, smem);
-  if (threadIdx.x == 0) part[blockIdx.x * 5] = t;
+  if (threadIdx.x == 0) part[blockIdx.x * 6] = t;
   __syncthreads();
   t = block_sum(sr, smem);
-  if (threadIdx.x == 0) part[blockIdx.x * 5 + 1] = t;
+  if (threadIdx.x == 0) part[blockIdx.x * 6 + 1] = t;
   __syncthreads();
   t = block_sum(sr2, smem);
-  if (threadIdx.x == 0) part[blockIdx.x * 5 + 2] = t;
+  if (threadIdx.x == 0) part[blockIdx.x * 6 + 2] = t;
   __syncthreads();
   t = block_sum(sc, smem);
-  if (threadIdx.x == 0) part[blockIdx.x * 5 + 3] = t;
+  if (threadIdx.x == 0) part[blockIdx.x * 6 + 3] = t;
   __syncthreads();
   t = block_sum(se, smem);
-  if (threadIdx.x == 0) part[blockIdx.x * 5 + 4] = t;
+  if (threadIdx.x == 0) part[blockIdx.x * 6 + 4] = t;
+  __syncthreads();
+  t = block_sum(sv, smem);
+  if (threadIdx.x == 0) part[blockIdx.x * 6 + 5] = t;
 }
 
 __global__ void ppo_head_finalize_k(const float* __restrict__ part,
                                     const int nwg, const long N,
                                     const float ent_coeff,
+                                    const float crit_scale,
                                     float* __restrict__ o_loss,
                                     float* __restrict__ o_ess,
                                     float* __restrict__ o_cf,
                                     float* __restrict__ o_ent,
                                     float* __restrict__ o_lent,
-                                    float* __restrict__ o_act) {
-  float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f;
+                                    float* __restrict__ o_act,
+                                    float* __restrict__ o_crit,
+                                    float* __restrict__ o_total) {
+  float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f, sv = 0.f;
   for (int i = threadIdx.x; i < nwg; i += blockDim.x) {
-    sg += part[5 * i];
-    sr += part[5 * i + 1];
-    sr2 += part[5 * i + 2];
-    sc += part[5 * i + 3];
-    se += part[5 * i + 4];
+    sg += part[6 * i];
+    sr += part[6 * i + 1];
+    sr2 += part[6 * i + 2];
+    sc += part[6 * i + 3];
+    se += part[6 * i + 4];
+    sv += part[6 * i + 5];
   }
   __shared__ float smem[8];
   float t;
@@ -558,12 +572,21 @@ __global__ void ppo_head_finalize_k(const float* __restrict__ part,
   if (threadIdx.x == 0) *o_cf = t / (float)N;  // clip_fraction
   __syncthreads();
   t = block_sum(se, smem);
+  if (threadIdx.x == 0) smem[5] = t;
+  __syncthreads();
+  const float tse = smem[5];
+  t = block_sum(sv, smem);
   if (threadIdx.x == 0) {
-    *o_ent = t / (float)N;                   // entropy (mean)
-    const float lent = -ent_coeff * t / (float)N;
+    *o_ent = tse / (float)N;                 // entropy (mean)
+    const float lent = -ent_coeff * tse / (float)N;
     *o_lent = lent;                          // loss_entropy
-    *o_act = *o_loss + lent;  // pre-summed actor loss (saves the eager
-                              // add + its backward per minibatch)
+    const float lact = *o_loss + lent;       // pre-summed actor loss
+    *o_act = lact;
+    if (o_crit != nullptr) {
+      const float lcrit = crit_scale * t / (float)N;
+      *o_crit = lcrit;                       // scaled critic loss
+      *o_total = lact + lcrit;               // whole minibatch loss
+    }
   }
 }
 
@@ -572,20 +595,31 @@ __global__ void ppo_head_bwd_k(
     const TV* __restrict__ head, const float* __restrict__ action,
     const float* __restrict__ eps, const float* __restrict__ prev,
     const float* __restrict__ adv, const float* __restrict__ stats,
+    const TV* __restrict__ value,    // nullable [N]
+    const float* __restrict__ vtarget,
     const float* __restrict__ gobj,  // 0-d upstream grad of loss_objective
     const float* __restrict__ gent,  // nullable: grad of loss_entropy
     const float* __restrict__ gact,  // nullable: grad of the pre-summed
                                      // actor loss (adds to both)
-    TV* __restrict__ dhead, const float sp_bias, const float lb,
-    const float lo, const float hi, const float ent_coeff, const long N,
-    const int A) {
+    const float* __restrict__ gcrit,  // nullable: grad of loss_critic
+    const float* __restrict__ gtot,   // nullable: grad of the total
+    TV* __restrict__ dhead, TV* __restrict__ dvalue, const float sp_bias,
+    const float lb, const float lo, const float hi, const float ent_coeff,
+    const float crit_scale, const long N, const int A) {
   const float mu = stats ? stats[0] : 0.f;
   const float isd = stats ? stats[1] : 1.f;
-  const float ga = gact ? gact[0] : 0.f;
+  const float gt = gtot ? gtot[0] : 0.f;
+  const float ga = (gact ? gact[0] : 0.f) + gt;
   const float g1s = -(ga + (gobj ? gobj[0] : 0.f)) / (float)N;
   const float ges = -ent_coeff * (ga + (gent ? gent[0] : 0.f)) / (float)N;
+  const float gvs =
+      crit_scale * ((gcrit ? gcrit[0] : 0.f) + gt) / (float)N;
   for (long n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
        n += (long)gridDim.x * blockDim.x) {
+    if (value) {
+      const float z = (float)value[n] - vtarget[n];
+      dvalue[n] = (TV)(gvs * fminf(fmaxf(z, -1.f), 1.f));
+    }
     // pass 1: recompute lp -> per-row dlw
     float lp = 0.f;
     for (int a = 0; a < A; ++a) {
@@ -631,47 +665,59 @@ __global__ void ppo_head_bwd_k(
 extern "C" void launch_ppo_head_fwd(const void* head, const float* action,
                                     const float* eps, const float* prev,
                                     const float* adv, const float* stats,
+                                    const void* value, const float* vtarget,
                                     float* part, float* const* outs,
                                     float sp_bias, float lb, float lo,
-                                    float hi, float ent_coeff, long N, int A,
+                                    float hi, float ent_coeff,
+                                    float crit_scale, long N, int A,
                                     int head_is_bf16, void* stream) {
   const int blocks = red_blocks(N);
   if (head_is_bf16)
     hipLaunchKernelGGL(ppo_head_fwd_partials_k<__hip_bfloat16>, dim3(blocks),
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)head, action, eps, prev, adv,
-                       stats, part, sp_bias, lb, lo, hi, N, A);
+                       stats, (const __hip_bfloat16*)value, vtarget, part,
+                       sp_bias, lb, lo, hi, N, A);
   else
     hipLaunchKernelGGL(ppo_head_fwd_partials_k<float>, dim3(blocks),
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const float*)head, action, eps, prev, adv, stats,
-                       part, sp_bias, lb, lo, hi, N, A);
+                       (const float*)value, vtarget, part, sp_bias, lb, lo,
+                       hi, N, A);
   hipLaunchKernelGGL(ppo_head_finalize_k, dim3(1), dim3(LP_THREADS), 0,
                      (hipStream_t)stream, part, blocks, N, ent_coeff,
-                     outs[0], outs[1], outs[2], outs[3], outs[4],
-                     outs[5]);
+                     crit_scale, outs[0], outs[1], outs[2], outs[3], outs[4],
+                     outs[5], value ? outs[6] : nullptr,
+                     value ? outs[7] : nullptr);
 }
 
 extern "C" void launch_ppo_head_bwd(const void* head, const float* action,
                                     const float* eps, const float* prev,
                                     const float* adv, const float* stats,
+                                    const void* value, const float* vtarget,
                                     const float* gobj, const float* gent,
-                                    const float* gact, void* dhead,
-                                    float sp_bias, float lb, float lo,
-                                    float hi, float ent_coeff, long N, int A,
+                                    const float* gact, const float* gcrit,
+                                    const float* gtot, void* dhead,
+                                    void* dvalue, float sp_bias, float lb,
+                                    float lo, float hi, float ent_coeff,
+                                    float crit_scale, long N, int A,
                                     int head_is_bf16, void* stream) {
   const int blocks = red_blocks(N);
   if (head_is_bf16)
     hipLaunchKernelGGL(ppo_head_bwd_k<__hip_bfloat16>, dim3(blocks),
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)head, action, eps, prev, adv,
-                       stats, gobj, gent, gact, (__hip_bfloat16*)dhead,
-                       sp_bias, lb, lo, hi, ent_coeff, N, A);
+                       stats, (const __hip_bfloat16*)value, vtarget, gobj,
+                       gent, gact, gcrit, gtot, (__hip_bfloat16*)dhead,
+                       (__hip_bfloat16*)dvalue, sp_bias, lb, lo, hi,
+                       ent_coeff, crit_scale, N, A);
   else
     hipLaunchKernelGGL(ppo_head_bwd_k<float>, dim3(blocks), dim3(LP_THREADS),
                        0, (hipStream_t)stream, (const float*)head, action,
-                       eps, prev, adv, stats, gobj, gent, gact,
-                       (float*)dhead, sp_bias, lb, lo, hi, ent_coeff, N, A);
+                       eps, prev, adv, stats, (const float*)value, vtarget,
+                       gobj, gent, gact, gcrit, gtot, (float*)dhead,
+                       (float*)dvalue, sp_bias, lb, lo, hi, ent_coeff,
+                       crit_scale, N, A);
 }
 
 // ---------------------------------------------------------------------------

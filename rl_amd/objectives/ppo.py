@@ -371,7 +371,8 @@ class ClipPPOLoss(PPOLoss):
             return None
         eps = torch.randn(N, A, device=action.device, dtype=torch.float32)
         lo, hi = self._clip_bounds
-        loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act = ops.ppo_head_loss(
+        (loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act, loss_crit,
+         loss_total) = ops.ppo_head_loss(
             head.reshape(N, 2 * A),
             action.reshape(N, A).float(),
             prev_lp.reshape(N),
@@ -383,15 +384,15 @@ class ClipPPOLoss(PPOLoss):
             hi=hi,
             entropy_coeff=float(self.entropy_coeff),
             normalize=normalize,
+            value=value,
+            value_target=(td.get(self.tensor_keys.value_target)
+                          if value is not None else None),
+            critic_scale=float(self.critic_coeff or 0.0),
         )
-        loss_critic = None
-        if value is not None:
-            loss_critic = ops.smooth_l1_mean(
-                value, td.get(self.tensor_keys.value_target),
-                float(self.critic_coeff),
-            )
+        if value is None:
+            loss_crit = loss_total = None
         return (loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act,
-                loss_critic)
+                loss_crit, loss_total)
 
     def _loss_critic_side_stream(self, td: TensorDictBase) -> torch.Tensor:
         """Critic loss forward on a side HIP stream: the critic chain
@@ -437,7 +438,7 @@ class ClipPPOLoss(PPOLoss):
         mega = self._mega_head_loss(td, advantage, normalize)
         if mega is not None:
             (loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act,
-             loss_critic) = mega
+             loss_critic, loss_total) = mega
             out = TensorDict(
                 {
                     "loss_objective": loss_obj,
@@ -453,6 +454,9 @@ class ClipPPOLoss(PPOLoss):
             )
             if loss_critic is not None:
                 out.set("loss_critic", loss_critic)
+                # kernel-side total (actor + scaled critic): trainers
+                # that recognize it skip every eager add
+                out.set("_loss_total", loss_total)
             elif self.critic_coeff is not None and self.critic_coeff > 0:
                 out.set("loss_critic", self._loss_critic_side_stream(td))
             return out

@@ -698,38 +698,49 @@ class _PPOHeadLossFn(torch.autograd.Function):
     under autocast, so the MLP backward needs no casts)."""
 
     @staticmethod
-    def forward(ctx, head, action, prev_lp, adv, eps, sp_bias, lb, lo, hi,
-                ent_coeff, normalize):
+    def forward(ctx, head, action, prev_lp, adv, eps, value, vtarget,
+                sp_bias, lb, lo, hi, ent_coeff, crit_scale, normalize):
         ctx.set_materialize_grads(False)
         head = head.contiguous()
         action = action.contiguous().detach()
         prev_lp = prev_lp.contiguous().detach()
         adv = adv.contiguous().detach()
         eps = eps.contiguous()
-        (loss_obj, ess, clip_frac, ent_mean, loss_ent, loss_act,
-         stats) = _C.ppo_head_fwd(
-            head, action, eps, prev_lp, adv, sp_bias, lb, lo, hi, ent_coeff,
-            normalize,
+        empty = head.new_empty(0)
+        has_crit = value is not None
+        v = value.contiguous().reshape(-1) if has_crit else empty
+        vt = (vtarget.contiguous().reshape(-1).detach()
+              if has_crit else head.new_empty(0, dtype=torch.float32))
+        (loss_obj, ess, clip_frac, ent_mean, loss_ent, loss_act, loss_crit,
+         loss_total, stats) = _C.ppo_head_fwd(
+            head, action, eps, prev_lp, adv, v, vt, sp_bias, lb, lo, hi,
+            ent_coeff, crit_scale, normalize,
         )
-        ctx.save_for_backward(head, action, eps, prev_lp, adv, stats)
-        ctx.cfg = (sp_bias, lb, lo, hi, ent_coeff)
+        ctx.save_for_backward(head, action, eps, prev_lp, adv, stats, v, vt)
+        ctx.cfg = (sp_bias, lb, lo, hi, ent_coeff, crit_scale)
+        ctx.has_crit = has_crit
+        ctx.v_shape = value.shape if has_crit else None
         ctx.mark_non_differentiable(ess, clip_frac, ent_mean)
-        return loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act
+        return (loss_obj, loss_ent, ent_mean, ess, clip_frac, loss_act,
+                loss_crit, loss_total)
 
     @staticmethod
-    def backward(ctx, g_obj, g_ent, g_em, g_ess, g_cf, g_act):
-        head, action, eps, prev_lp, adv, stats = ctx.saved_tensors
-        sp_bias, lb, lo, hi, ent_coeff = ctx.cfg
+    def backward(ctx, g_obj, g_ent, g_em, g_ess, g_cf, g_act, g_crit, g_tot):
+        (head, action, eps, prev_lp, adv, stats, v, vt) = ctx.saved_tensors
+        sp_bias, lb, lo, hi, ent_coeff, crit_scale = ctx.cfg
         empty = head.new_empty(0, dtype=torch.float32)
         gobj = g_obj.contiguous() if g_obj is not None else empty
         gent = g_ent.contiguous() if g_ent is not None else empty
         gact = g_act.contiguous() if g_act is not None else empty
-        dhead = _C.ppo_head_bwd(
-            head, action, eps, prev_lp, adv, stats, gobj, gent, gact,
-            sp_bias, lb, lo, hi, ent_coeff,
+        gcrit = g_crit.contiguous() if g_crit is not None else empty
+        gtot = g_tot.contiguous() if g_tot is not None else empty
+        dhead, dvalue = _C.ppo_head_bwd(
+            head, action, eps, prev_lp, adv, stats, v, vt, gobj, gent, gact,
+            gcrit, gtot, sp_bias, lb, lo, hi, ent_coeff, crit_scale,
         )
-        return (dhead, None, None, None, None, None, None, None, None, None,
-                None)
+        dv = dvalue.reshape(ctx.v_shape) if ctx.has_crit else None
+        return (dhead, None, None, None, None, dv, None, None, None, None,
+                None, None, None, None)
 
 
 def fused_grad_clip_(parameters, max_norm: float) -> bool:
@@ -846,18 +857,21 @@ def multi_shuffle_td(flat_td, keys: torch.Tensor):
 
 
 def ppo_head_loss(head, action, prev_log_prob, advantage, eps, *, sp_bias,
-                  scale_lb, lo, hi, entropy_coeff, normalize):
-    """Fused ClipPPO actor losses straight from the raw policy-head
-    output (``[N, 2A]`` = loc | pre-softplus scale).  Returns
+                  scale_lb, lo, hi, entropy_coeff, normalize, value=None,
+                  value_target=None, critic_scale=1.0):
+    """Fused ClipPPO losses straight from the raw policy-head output
+    (``[N, 2A]`` = loc | pre-softplus scale).  Returns
     ``(loss_objective, loss_entropy, entropy_mean, ESS_per_sample,
-    clip_fraction, loss_actor)`` where ``loss_actor`` is the pre-summed
-    ``loss_objective + loss_entropy``; the two losses and the sum carry
-    gradient to ``head``."""
+    clip_fraction, loss_actor, loss_critic, loss_total)``.  When
+    ``value``/``value_target`` are given, the scaled smooth-L1 critic
+    loss and the whole-minibatch total ride in the SAME kernels (the
+    backward emits d(head) and d(value) in one pass); otherwise the
+    last two outputs are undefined scalars."""
     _require_ext()
     return _PPOHeadLossFn.apply(
-        head, action, prev_log_prob, advantage, eps, float(sp_bias),
-        float(scale_lb), float(lo), float(hi), float(entropy_coeff),
-        bool(normalize),
+        head, action, prev_log_prob, advantage, eps, value, value_target,
+        float(sp_bias), float(scale_lb), float(lo), float(hi),
+        float(entropy_coeff), float(critic_scale), bool(normalize),
     )
 
 

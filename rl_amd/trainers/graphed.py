@@ -145,9 +145,22 @@ class GraphedPPO:
 
     # ------------------------------------------------------------------ #
     def _total_loss(self, out: TensorDictBase) -> torch.Tensor:
-        # "_loss_actor" is the kernel-side pre-summed
-        # loss_objective + loss_entropy (ClipPPOLoss mega path)
-        presummed = out.get("_loss_actor", None) if hasattr(out, "get") else None
+        # kernel-side pre-summed losses from the ClipPPOLoss mega path:
+        # "_loss_total" = objective + entropy + scaled critic;
+        # "_loss_actor" = objective + entropy
+        if hasattr(out, "get"):
+            full = out.get("_loss_total", None)
+            if full is not None:
+                skip = {"loss_objective", "loss_entropy", "loss_critic"}
+                total = full
+                for k in out.keys():
+                    if (isinstance(k, str) and k.startswith("loss_")
+                            and k not in skip):
+                        total = total + out.get(k)
+                return total
+            presummed = out.get("_loss_actor", None)
+        else:
+            presummed = None
         skip = {"loss_objective", "loss_entropy"} if presummed is not None else set()
         total = presummed
         for k in out.keys():

@@ -780,7 +780,7 @@ class TestPPOHeadMega:
         lo, hi = math.log1p(-0.2), math.log1p(0.2)
         o_f = ops.ppo_head_loss(head, action, prev, adv, eps, sp_bias=bias,
                                 scale_lb=lb, lo=lo, hi=hi, entropy_coeff=0.01,
-                                normalize=normalize)
+                                normalize=normalize)[:6]
         head2 = head.detach().clone().requires_grad_()
         o_e = self._eager(head2, action, prev, adv, eps, bias, lb, lo, hi,
                           0.01, normalize)
@@ -810,7 +810,7 @@ class TestPPOHeadMega:
         o = ops.ppo_head_loss(head, action, prev, adv, eps,
                               sp_bias=0.5413248546129181, scale_lb=1e-4,
                               lo=lo, hi=hi, entropy_coeff=0.01,
-                              normalize=True)
+                              normalize=True)[:6]
         (g,) = torch.autograd.grad(o[0], head)
         head2 = head.detach().clone().requires_grad_()
         o_e = TestPPOHeadMega._eager(self, head2, action, prev, adv, eps,
@@ -890,3 +890,37 @@ def test_grad_scale_clip_equals_explicit_clip():
     torch.nn.utils.clip_grad_norm_([p2], max_norm)
     o2.step()
     assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
+
+
+@pytest.mark.gpu
+def test_ppo_head_with_critic_fused():
+    """Critic smooth-L1 riding in the head kernels: outputs and BOTH
+    gradients (d head, d value) match the separate eager chain."""
+    from rl_amd import ops
+
+    torch.manual_seed(3)
+    N, A = 1537, 6
+    head = (0.4 * torch.randn(N, 2 * A, device="cuda")).to(torch.bfloat16).requires_grad_()
+    value = torch.randn(N, device="cuda").to(torch.bfloat16).requires_grad_()
+    target = torch.randn(N, device="cuda")
+    action = torch.rand(N, A, device="cuda") * 1.6 - 0.8
+    prev = torch.randn(N, device="cuda") * 0.1
+    adv = torch.randn(N, device="cuda")
+    eps = torch.randn(N, A, device="cuda")
+    lo, hi = math.log1p(-0.2), math.log1p(0.2)
+    kw = dict(sp_bias=0.5413248546129181, scale_lb=1e-4, lo=lo, hi=hi,
+              entropy_coeff=0.01, normalize=True)
+    o = ops.ppo_head_loss(head, action, prev, adv, eps, value=value,
+                          value_target=target, critic_scale=0.5, **kw)
+    l_obj, l_ent, _, _, _, l_act, l_crit, l_tot = o
+    # reference: separate fused paths on cloned leaves
+    head2 = head.detach().clone().requires_grad_()
+    value2 = value.detach().clone().requires_grad_()
+    o2 = ops.ppo_head_loss(head2, action, prev, adv, eps, **kw)
+    crit_ref = ops.smooth_l1_mean(value2, target, 0.5)
+    assert torch.allclose(l_crit, crit_ref, atol=1e-5)
+    assert torch.allclose(l_tot, o2[5] + crit_ref, atol=1e-5)
+    l_tot.backward()
+    (o2[5] + crit_ref).backward()
+    assert torch.allclose(head.grad.float(), head2.grad.float(), atol=1e-5)
+    assert torch.allclose(value.grad.float(), value2.grad.float(), atol=1e-5)
