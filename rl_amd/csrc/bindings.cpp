@@ -98,6 +98,9 @@ void launch_mlp3_mfma_fwd2(const void*, int, const void* const*,
                            const int*, void*);
 void launch_mlp3_mfma_bwd2(const void* const*, void* const*, long,
                            const int*, const int*, void*);
+void launch_mlp3_mfma_fwdpair(const void*, const void*, int,
+                              const void* const*, void*, void*, void*, long,
+                              int, int, int, void*);
 void launch_mlp3_mfma_fwd(const void*, int, const void*, const void*,
                           const void*, const void*, const void*, const void*,
                           void*, void*, void*, void*, long, int, int, int,
@@ -1005,6 +1008,35 @@ std::vector<torch::Tensor> mlp3_mfma_bwd(torch::Tensor dout, torch::Tensor h1,
   return {dh1, dh2};
 }
 
+// Same-network pair forward (GAE value/next_value in one launch; no
+// gradients, scratch activations discarded).
+std::vector<torch::Tensor> mlp3_mfma_fwdpair(torch::Tensor x0,
+                                             torch::Tensor x1,
+                                             std::vector<torch::Tensor> w) {
+  TORCH_CHECK(x0.is_cuda() && x0.is_contiguous() && x1.is_contiguous(),
+              "inputs cuda contiguous");
+  TORCH_CHECK(x0.scalar_type() == x1.scalar_type() &&
+                  x0.sizes() == x1.sizes(),
+              "pair inputs must match");
+  TORCH_CHECK(w.size() == 6, "6 weight tensors");
+  const bool xf32 = x0.scalar_type() == torch::kFloat32;
+  const long N = x0.size(0), O = x0.size(1);
+  const long H = w[0].size(0), A2 = w[4].size(0);
+  TORCH_CHECK(mlp3_mfma_ok(O, H, A2), "unsupported dims");
+  const void* wp[6];
+  for (int i = 0; i < 6; ++i) wp[i] = w[i].data_ptr();
+  auto bopt = w[0].options();
+  auto out0 = torch::empty({N, A2}, bopt);
+  auto out1 = torch::empty({N, A2}, bopt);
+  auto scratch = torch::empty({4 * N * H}, bopt);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_mlp3_mfma_fwdpair(x0.data_ptr(), x1.data_ptr(), xf32 ? 1 : 0, wp,
+                           out0.data_ptr(), out1.data_ptr(),
+                           scratch.data_ptr(), N, (int)O, (int)H, (int)A2,
+                           (void*)stream);
+  return {out0, out1};
+}
+
 // Mega-fused TanhNormal head loss: raw actor-head output [N, 2A] ->
 // (out[5] = {loss_objective, ESS/N, clip_fraction, entropy_mean,
 // loss_entropy}, stats).  See csrc/loss_ops.hip.
@@ -1161,6 +1193,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dual-network MFMA MLP forward (HIP)");
   m.def("mlp3_mfma_bwd2", &mlp3_mfma_bwd2,
         "dual-network MFMA dgrad chain (HIP)");
+  m.def("mlp3_mfma_fwdpair", &mlp3_mfma_fwdpair,
+        "same-network pair forward (HIP)");
   m.def("fused_grad_clip_coef", &fused_grad_clip_coef,
         "single-kernel global grad-norm clip coefficient (HIP)");
   m.def("multi_gather", &multi_gather,

@@ -724,3 +724,56 @@ extern "C" void launch_mlp3_mfma_bwd2(const void* const* dptr,  // 10 ptrs
     hipLaunchKernelGGL((mlp3_mfma_bwd2_kernel<32>), grid, dim3(MLP_THREADS),
                        lds, (hipStream_t)stream, a, (int)N);
 }
+
+// Same-network PAIR forward (no gradients): grid.y picks which of two
+// inputs to run — GAE's value / next_value evals in ONE launch.
+template <typename TX, int R>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwdpair_kernel(
+    const TX* __restrict__ x0, const TX* __restrict__ x1,
+    const __hip_bfloat16* __restrict__ w1,
+    const __hip_bfloat16* __restrict__ b1,
+    const __hip_bfloat16* __restrict__ w2,
+    const __hip_bfloat16* __restrict__ b2,
+    const __hip_bfloat16* __restrict__ w3,
+    const __hip_bfloat16* __restrict__ b3, __hip_bfloat16* __restrict__ out0,
+    __hip_bfloat16* __restrict__ out1, __hip_bfloat16* __restrict__ scratch,
+    const int N, const int O, const int H, const int A2) {
+  const int net = blockIdx.y;
+  __hip_bfloat16* h1 = scratch + (size_t)net * 2 * N * H;
+  __hip_bfloat16* h2 = h1 + (size_t)N * H;
+  mlp3_mfma_fwd_impl<TX, R>(net ? x1 : x0, w1, b1, w2, b2, w3, b3,
+                            net ? out1 : out0, h1, h2, nullptr, N, O, H, A2);
+}
+
+extern "C" void launch_mlp3_mfma_fwdpair(const void* x0, const void* x1,
+                                         int x_is_f32, const void* const* w,
+                                         void* out0, void* out1,
+                                         void* scratch, long N, int O, int H,
+                                         int A2, void* stream) {
+  const int R = N >= M3_ROWS_SWITCH ? 64 : 32;
+  const int blocks = (int)((N + R - 1) / R);
+  const int Op = (O + 31) & ~31, Hp = (H + 31) & ~31;
+  const int lx = Op + 8, lh = Hp + 8, A2p = (A2 + 15) & ~15;
+  const int lds = 2 * (R * lx + H * lx + 2 * R * lh + H * lh + A2p * lh);
+  dim3 grid(blocks, 2);
+#define M3_LAUNCH_PAIR(TX, RR)                                              \
+  hipLaunchKernelGGL((mlp3_mfma_fwdpair_kernel<TX, RR>), grid,              \
+                     dim3(MLP_THREADS), lds, (hipStream_t)stream,           \
+                     (const TX*)x0, (const TX*)x1,                          \
+                     (const __hip_bfloat16*)w[0],                           \
+                     (const __hip_bfloat16*)w[1],                           \
+                     (const __hip_bfloat16*)w[2],                           \
+                     (const __hip_bfloat16*)w[3],                           \
+                     (const __hip_bfloat16*)w[4],                           \
+                     (const __hip_bfloat16*)w[5], (__hip_bfloat16*)out0,    \
+                     (__hip_bfloat16*)out1, (__hip_bfloat16*)scratch,       \
+                     (int)N, O, H, A2)
+  if (x_is_f32) {
+    if (R == 64) M3_LAUNCH_PAIR(float, 64);
+    else M3_LAUNCH_PAIR(float, 32);
+  } else {
+    if (R == 64) M3_LAUNCH_PAIR(__hip_bfloat16, 64);
+    else M3_LAUNCH_PAIR(__hip_bfloat16, 32);
+  }
+#undef M3_LAUNCH_PAIR
+}

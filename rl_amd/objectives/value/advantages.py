@@ -111,14 +111,57 @@ class ValueEstimatorBase(TensorDictModuleBase):
                     next_value = next_value.clone()
                     next_value[idx] = sub.get(value_key)
         else:
-            with ctx:
-                td_root = self.value_network(td)
-                value = td_root.get(value_key)
-                nxt = self.value_network(nxt.clone(False))
-                next_value = nxt.get(value_key)
+            paired = None
+            if not self.differentiable:
+                paired = self._paired_value_eval(td, nxt, value_key)
+            if paired is not None:
+                value, next_value = paired
+            else:
+                with ctx:
+                    td_root = self.value_network(td)
+                    value = td_root.get(value_key)
+                    nxt = self.value_network(nxt.clone(False))
+                    next_value = nxt.get(value_key)
         td.set(value_key, value)
         td.get("next").set(value_key, next_value)
         return value, next_value
+
+    def _paired_value_eval(self, td, nxt, value_key):
+        """value(obs) and value(next_obs) in ONE kernel launch when the
+        critic is a FusedMLP3 over a single observation key
+        (csrc/fused_mlp.hip fwdpair); None = take the eager path."""
+        try:
+            from ... import ops
+
+            if not ops.HAS_HIP_EXT:
+                return None
+            from ...ops import FusedMLP3
+
+            net = self.value_network
+            mod = getattr(net, "module", None)
+            in_keys = list(getattr(net, "in_keys", []))
+            if not (isinstance(mod, FusedMLP3) and len(in_keys) == 1):
+                return None
+            obs = td.get(in_keys[0], None)
+            nobs = nxt.get(in_keys[0], None)
+            if obs is None or nobs is None or not obs.is_cuda:
+                return None
+            if obs.dtype not in (torch.float32, torch.bfloat16):
+                return None
+            if not getattr(mod.lin1, "_bf16_cache", False):
+                return None
+            O = obs.shape[-1]
+            if mod.lin1.in_features != O:
+                return None
+            lead = obs.shape[:-1]
+            with torch.no_grad():
+                v0, v1 = ops.value_pair_eval(
+                    mod, obs.reshape(-1, O), nobs.reshape(-1, O)
+                )
+            A2 = v0.shape[-1]
+            return v0.reshape(*lead, A2), v1.reshape(*lead, A2)
+        except Exception:
+            return None
 
     def _get_done_terminated_reward(self, td: TensorDictBase):
         reward = td.get(unravel_key(self.tensor_keys.reward))

@@ -1060,6 +1060,17 @@ def actor_critic_mlp3(x, actor_fused, critic_fused):
     )
 
 
+def value_pair_eval(fused, x0, x1):
+    """No-grad evaluation of one :class:`FusedMLP3` on TWO equal-shaped
+    inputs in a single launch (csrc/fused_mlp.hip fwdpair) — GAE's
+    value/next_value critic calls.  Returns ``(y0, y1)``."""
+    _require_ext()
+    w = [fused.lin1.weight_bf16, fused.lin1.bias_bf16,
+         fused.lin2.weight_bf16, fused.lin2.bias_bf16,
+         fused.lin3.weight_bf16, fused.lin3.bias_bf16]
+    return _C.mlp3_mfma_fwdpair(x0.contiguous(), x1.contiguous(), w)
+
+
 def actor_critic_mlp3_ok(actor_fused, critic_fused, in_features: int) -> bool:
     """Shape eligibility for the dual-network kernels."""
     if not HAS_HIP_EXT:

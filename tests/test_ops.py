@@ -924,3 +924,25 @@ def test_ppo_head_with_critic_fused():
     (o2[5] + crit_ref).backward()
     assert torch.allclose(head.grad.float(), head2.grad.float(), atol=1e-5)
     assert torch.allclose(value.grad.float(), value2.grad.float(), atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_value_pair_eval_matches_two_calls():
+    from rl_amd import ops
+    from rl_amd.ops import convert_linears_to_splitk, enable_splitk_bf16_cache, fuse_mlp3
+
+    torch.manual_seed(0)
+    net = torch.nn.Sequential(
+        torch.nn.Linear(17, 64), torch.nn.Tanh(),
+        torch.nn.Linear(64, 64), torch.nn.Tanh(),
+        torch.nn.Linear(64, 1),
+    ).cuda()
+    convert_linears_to_splitk(net)
+    enable_splitk_bf16_cache(net)
+    fused = fuse_mlp3(net)
+    x0 = torch.randn(3000, 17, device="cuda")
+    x1 = torch.randn(3000, 17, device="cuda")
+    y0, y1 = ops.value_pair_eval(fused, x0, x1)
+    with torch.no_grad():
+        r0, r1 = fused(x0), fused(x1)
+    assert torch.equal(y0, r0) and torch.equal(y1, r1)
