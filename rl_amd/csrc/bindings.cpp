@@ -41,6 +41,13 @@ int lstm_fused_lds_bytes(int);
 int wgrad_slab_count(long);
 int mlp3_lds_bytes(int, int, int);
 int fused_rollout_lds_bytes(int, int, int, int);
+int fused_rollout_mfma_ok(int, int, int, int);
+void launch_fused_rollout_mfma(float*, float*, const void*, const void*,
+                               const void*, const void*, const void*,
+                               const void*, const float*, const float*,
+                               const float*, const float*, float*, float*,
+                               float*, float*, float*, bool*, int, int, int,
+                               int, int, int, float, float, float, void*);
 void launch_fused_rollout(float*, float*, const float*, const float*,
                           const float*, const float*, const float*,
                           const float*, const float*, const float*,
@@ -507,7 +514,8 @@ void fused_rollout(torch::Tensor state, torch::Tensor step_ct,
                    torch::Tensor st_act, torch::Tensor st_logp,
                    torch::Tensor st_nobs, torch::Tensor st_rew,
                    torch::Tensor st_done, double max_steps,
-                   double inv_softplus_bias, double scale_lb) {
+                   double inv_softplus_bias, double scale_lb,
+                   std::vector<torch::Tensor> bf16_weights) {
   TORCH_CHECK(state.is_cuda() && state.scalar_type() == torch::kFloat32,
               "fused_rollout: fp32 cuda state");
   const int Bn = (int)state.size(0), S = (int)state.size(1);
@@ -521,6 +529,24 @@ void fused_rollout(torch::Tensor state, torch::Tensor step_ct,
                   st_rew.is_contiguous() && st_done.is_contiguous(),
               "store tensors must be contiguous [B, T, ...]");
   auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (bf16_weights.size() == 6 &&
+      fused_rollout_mfma_ok(S, H1, H2, Aact)) {
+    for (auto& t : bf16_weights)
+      TORCH_CHECK(t.scalar_type() == torch::kBFloat16 && t.is_contiguous(),
+                  "fused_rollout: bf16 cache tensors");
+    launch_fused_rollout_mfma(
+        state.data_ptr<float>(), step_ct.data_ptr<float>(),
+        bf16_weights[0].data_ptr(), bf16_weights[1].data_ptr(),
+        bf16_weights[2].data_ptr(), bf16_weights[3].data_ptr(),
+        bf16_weights[4].data_ptr(), bf16_weights[5].data_ptr(),
+        A.data_ptr<float>(), B.data_ptr<float>(), eps.data_ptr<float>(),
+        noise.data_ptr<float>(), st_obs.data_ptr<float>(),
+        st_act.data_ptr<float>(), st_logp.data_ptr<float>(),
+        st_nobs.data_ptr<float>(), st_rew.data_ptr<float>(),
+        st_done.data_ptr<bool>(), Bn, S, H1, H2, Aact, T, (float)max_steps,
+        (float)inv_softplus_bias, (float)scale_lb, (void*)stream);
+    return;
+  }
   launch_fused_rollout(
       state.data_ptr<float>(), step_ct.data_ptr<float>(),
       w1.data_ptr<float>(), b1.data_ptr<float>(), w2.data_ptr<float>(),

@@ -23,6 +23,7 @@
 // Numerics match the fused_actor + synthetic_env_step pair bit-for-bit
 // (same math, same order); validated in tests/test_ops.py.
 
+#include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 
 #define RO_THREADS 256
@@ -232,4 +233,267 @@ extern "C" void launch_fused_rollout(
                      w3, b3, Amat, Bmat, eps, noise, st_obs, st_act, st_logp,
                      st_nobs, st_rew, st_done, B, S, H1, H2, Aact, T,
                      max_steps, inv_softplus_bias, scale_lb);
+}
+
+
+// ---------------------------------------------------------------------------
+// MFMA rollout variant: the policy MLP on the matrix cores.
+//
+// Same loop/store/env semantics as fused_rollout_kernel, but the three
+// policy GEMMs use v_mfma_f32_16x16x32_bf16 against the actor's bf16
+// weight caches (the SAME caches the update phase computes with, so
+// rollout-time log-probs and epoch-1 recomputed log-probs now agree to
+// the kernel's fp32 reductions).  16 env rows per workgroup (the MFMA
+// tile height); 4 waves split the H/16 output col tiles.  The env
+// transition stays fp32 VALU (S ~ 17: not GEMM-shaped).
+// Requirements (launcher-guarded): H1 == H2, H % 32 == 0, 2A <= 16.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+using ro_bfrag = __attribute__((ext_vector_type(8))) short;
+using ro_ffrag = __attribute__((ext_vector_type(4))) float;
+
+__device__ __forceinline__ void ro_gemm_tile(
+    const __hip_bfloat16* s_a, int lda, const __hip_bfloat16* s_w, int ldw,
+    int j0, int Kp, int lane, ro_ffrag* acc) {
+  const int row = lane & 15;
+  const int koff = 8 * (lane >> 4);
+  for (int kc = 0; kc < Kp; kc += 32) {
+    const ro_bfrag a = *reinterpret_cast<const ro_bfrag*>(
+        &s_a[(size_t)row * lda + kc + koff]);
+    const ro_bfrag b = *reinterpret_cast<const ro_bfrag*>(
+        &s_w[(size_t)(j0 + row) * ldw + kc + koff]);
+    *acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, *acc, 0, 0, 0);
+  }
+}
+
+__global__ void __launch_bounds__(RO_THREADS) fused_rollout_mfma_kernel(
+    float* __restrict__ state, float* __restrict__ step_ct,
+    const __hip_bfloat16* __restrict__ w1, const __hip_bfloat16* __restrict__ b1,
+    const __hip_bfloat16* __restrict__ w2, const __hip_bfloat16* __restrict__ b2,
+    const __hip_bfloat16* __restrict__ w3, const __hip_bfloat16* __restrict__ b3,
+    const float* __restrict__ Amat, const float* __restrict__ Bmat,
+    const float* __restrict__ eps, const float* __restrict__ noise,
+    float* __restrict__ st_obs, float* __restrict__ st_act,
+    float* __restrict__ st_logp, float* __restrict__ st_nobs,
+    float* __restrict__ st_rew, bool* __restrict__ st_done, const int B,
+    const int S, const int H, const int Aact, const int T,
+    const float max_steps, const float inv_softplus_bias,
+    const float scale_lb) {
+  extern __shared__ char smem_raw[];
+  const int Sp = (S + 31) & ~31;
+  const int lx = Sp + 8, lh = H + 8;
+  const int A2 = 2 * Aact;
+  const int as = S | 1;
+  char* ptr = smem_raw;
+  auto alloc = [&](size_t bytes) {
+    char* r = ptr;
+    ptr += (bytes + 15) & ~size_t(15);
+    return r;
+  };
+  __hip_bfloat16* s_w1 = (__hip_bfloat16*)alloc((size_t)H * lx * 2);
+  __hip_bfloat16* s_w2 = (__hip_bfloat16*)alloc((size_t)H * lh * 2);
+  __hip_bfloat16* s_w3 = (__hip_bfloat16*)alloc((size_t)16 * lh * 2);
+  float* s_bias = (float*)alloc((size_t)(2 * H + 16) * 4);
+  float* s_A = (float*)alloc((size_t)S * as * 4);
+  float* s_B = (float*)alloc((size_t)Aact * as * 4);
+  float* s_state = (float*)alloc((size_t)16 * S * 4);
+  float* s_next = (float*)alloc((size_t)16 * S * 4);
+  __hip_bfloat16* s_sbf = (__hip_bfloat16*)alloc((size_t)16 * lx * 2);
+  __hip_bfloat16* s_h1 = (__hip_bfloat16*)alloc((size_t)16 * lh * 2);
+  __hip_bfloat16* s_h2 = (__hip_bfloat16*)alloc((size_t)16 * lh * 2);
+  float* s_head = (float*)alloc((size_t)16 * 16 * 4);
+  float* s_act = (float*)alloc((size_t)16 * Aact * 4);
+  float* s_lp = (float*)alloc((size_t)16 * Aact * 4);
+  float* s_ct = (float*)alloc((size_t)16 * 4);
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int row0 = blockIdx.x * 16;
+  const int rows = min(16, B - row0);
+
+  for (int i = tid; i < H * lx; i += RO_THREADS) {
+    const int j = i / lx, k = i % lx;
+    s_w1[i] = (k < S) ? w1[(size_t)j * S + k] : __hip_bfloat16(0.f);
+  }
+  for (int i = tid; i < H * lh; i += RO_THREADS) {
+    const int j = i / lh, k = i % lh;
+    s_w2[i] = (k < H) ? w2[(size_t)j * H + k] : __hip_bfloat16(0.f);
+  }
+  for (int i = tid; i < 16 * lh; i += RO_THREADS) {
+    const int j = i / lh, k = i % lh;
+    s_w3[i] = (j < A2 && k < H) ? w3[(size_t)j * H + k] : __hip_bfloat16(0.f);
+  }
+  for (int i = tid; i < H; i += RO_THREADS) {
+    s_bias[i] = __bfloat162float(b1[i]);
+    s_bias[H + i] = __bfloat162float(b2[i]);
+  }
+  for (int i = tid; i < A2; i += RO_THREADS)
+    s_bias[2 * H + i] = __bfloat162float(b3[i]);
+  for (int i = tid; i < S * S; i += RO_THREADS)
+    s_A[(i / S) * as + i % S] = Amat[i];
+  for (int i = tid; i < Aact * S; i += RO_THREADS)
+    s_B[(i / S) * as + i % S] = Bmat[i];
+  for (int i = tid; i < rows * S; i += RO_THREADS)
+    s_state[i] = state[(size_t)(row0 + i / S) * S + i % S];
+  for (int r = tid; r < rows; r += RO_THREADS) s_ct[r] = step_ct[row0 + r];
+  __syncthreads();
+
+  const float LOG_SQRT_2PI = 0.9189385332046727f;
+  const float LOG2 = 0.6931471805599453f;
+  const float lim = 1.0f - 1.1920929e-7f;
+  const int erow = (lane >> 4) * 4;
+  const int ecol = lane & 15;
+
+  for (int t = 0; t < T; ++t) {
+    // store obs + restage the bf16 policy input (zero pads)
+    for (int i = tid; i < 16 * lx; i += RO_THREADS) {
+      const int r = i / lx, k = i % lx;
+      s_sbf[i] = (r < rows && k < S) ? __hip_bfloat16(s_state[r * S + k])
+                                     : __hip_bfloat16(0.f);
+    }
+    for (int i = tid; i < rows * S; i += RO_THREADS)
+      st_obs[((size_t)(row0 + i / S) * T + t) * S + i % S] =
+          s_state[(i / S) * S + i % S];
+    __syncthreads();
+    // layer 1
+    for (int ct = wave; ct < H / 16; ct += 4) {
+      ro_ffrag acc = {};
+      ro_gemm_tile(s_sbf, lx, s_w1, lx, ct * 16, Sp, lane, &acc);
+      const int col = ct * 16 + ecol;
+      const float bias = s_bias[col];
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        s_h1[(size_t)(erow + q) * lh + col] =
+            __hip_bfloat16(tanhf(acc[q] + bias));
+    }
+    __syncthreads();
+    // layer 2
+    for (int ct = wave; ct < H / 16; ct += 4) {
+      ro_ffrag acc = {};
+      ro_gemm_tile(s_h1, lh, s_w2, lh, ct * 16, H, lane, &acc);
+      const int col = ct * 16 + ecol;
+      const float bias = s_bias[H + col];
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        s_h2[(size_t)(erow + q) * lh + col] =
+            __hip_bfloat16(tanhf(acc[q] + bias));
+    }
+    __syncthreads();
+    // heads (one 16x16 tile; wave 0)
+    if (wave == 0) {
+      ro_ffrag acc = {};
+      ro_gemm_tile(s_h2, lh, s_w3, lh, 0, H, lane, &acc);
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        s_head[(size_t)(erow + q) * 16 + ecol] =
+            acc[q] + (ecol < A2 ? s_bias[2 * H + ecol] : 0.f);
+    }
+    __syncthreads();
+    // sample + per-element log-prob
+    for (int i = tid; i < rows * Aact; i += RO_THREADS) {
+      const int r = i / Aact, a = i % Aact;
+      const float loc = s_head[r * 16 + a];
+      float scale = softplusf_(s_head[r * 16 + Aact + a] + inv_softplus_bias);
+      scale = fmaxf(scale, scale_lb);
+      const float e = eps[((size_t)t * B + row0 + r) * Aact + a];
+      const float u = loc + scale * e;
+      float act = tanhf(u);
+      act = fminf(fmaxf(act, -lim), lim);
+      s_act[r * Aact + a] = act;
+      st_act[((size_t)(row0 + r) * T + t) * Aact + a] = act;
+      s_lp[r * Aact + a] =
+          -0.5f * e * e - __logf(scale) - LOG_SQRT_2PI
+          - 2.0f * (LOG2 - u - softplusf_(-2.0f * u));
+    }
+    __syncthreads();
+    for (int r = tid; r < rows; r += RO_THREADS) {
+      float sum = 0.f;
+      for (int a = 0; a < Aact; ++a) sum += s_lp[r * Aact + a];
+      st_logp[(size_t)(row0 + r) * T + t] = sum;
+    }
+    // env transition (fp32, same math/order as the VALU kernel)
+    for (int i = tid; i < rows * S; i += RO_THREADS) {
+      const int r = i / S, j = i % S;
+      float acc = 0.f;
+      const float* sr = &s_state[r * S];
+#pragma unroll 4
+      for (int k = 0; k < S; ++k) acc += sr[k] * s_A[k * as + j];
+      const float* ar = &s_act[r * Aact];
+#pragma unroll
+      for (int k = 0; k < Aact; ++k) acc += ar[k] * s_B[k * as + j];
+      s_next[r * S + j] = tanhf(acc);
+    }
+    __syncthreads();
+    for (int i = tid; i < rows * S; i += RO_THREADS) {
+      const int r = i / S, j = i % S;
+      const bool trunc = (s_ct[r] + 1.f) >= max_steps;
+      const float ns = s_next[r * S + j];
+      st_nobs[((size_t)(row0 + r) * T + t) * S + j] = ns;
+      const float carry =
+          trunc ? noise[((size_t)t * B + row0 + r) * S + j] : ns;
+      s_next[r * S + j] = carry;  // committed after the sync below
+      if (j == 0) {
+        float ctrl = 0.f;
+#pragma unroll
+        for (int k = 0; k < Aact; ++k)
+          ctrl += s_act[r * Aact + k] * s_act[r * Aact + k];
+        st_rew[(size_t)(row0 + r) * T + t] = ns - 0.1f * ctrl;
+        st_done[(size_t)(row0 + r) * T + t] = trunc;
+      }
+    }
+    __syncthreads();
+    for (int i = tid; i < rows * S; i += RO_THREADS)
+      s_state[i] = s_next[i];
+    for (int r = tid; r < rows; r += RO_THREADS) {
+      const bool trunc = (s_ct[r] + 1.f) >= max_steps;
+      s_ct[r] = trunc ? 0.f : s_ct[r] + 1.f;
+    }
+    __syncthreads();
+  }
+
+  for (int i = tid; i < rows * S; i += RO_THREADS)
+    state[(size_t)(row0 + i / S) * S + i % S] = s_state[i];
+  for (int r = tid; r < rows; r += RO_THREADS) step_ct[row0 + r] = s_ct[r];
+}
+
+}  // namespace
+
+extern "C" int fused_rollout_mfma_ok(int S, int H1, int H2, int Aact) {
+  if (H1 != H2 || H1 % 32 != 0 || 2 * Aact > 16) return 0;
+  const int Sp = (S + 31) & ~31;
+  const int lx = Sp + 8, lh = H1 + 8, as = S | 1;
+  const size_t lds =
+      (size_t)H1 * lx * 2 + (size_t)H1 * lh * 2 + 16 * lh * 2 +
+      (2 * H1 + 16) * 4 + (size_t)S * as * 4 + (size_t)Aact * as * 4 +
+      16 * S * 8 + 16 * lx * 2 + 2 * 16 * lh * 2 + 16 * 16 * 4 +
+      16 * Aact * 8 + 64 + 16 * 16;  // + alignment slack
+  return lds <= 160 * 1024;
+}
+
+extern "C" void launch_fused_rollout_mfma(
+    float* state, float* step_ct, const void* w1, const void* b1,
+    const void* w2, const void* b2, const void* w3, const void* b3,
+    const float* Amat, const float* Bmat, const float* eps,
+    const float* noise, float* st_obs, float* st_act, float* st_logp,
+    float* st_nobs, float* st_rew, bool* st_done, int B, int S, int H1,
+    int H2, int Aact, int T, float max_steps, float inv_softplus_bias,
+    float scale_lb, void* stream) {
+  const int blocks = (B + 15) / 16;
+  const int Sp = (S + 31) & ~31;
+  const int lx = Sp + 8, lh = H1 + 8, as = S | 1;
+  const size_t lds =
+      (size_t)H1 * lx * 2 + (size_t)H1 * lh * 2 + 16 * lh * 2 +
+      (2 * H1 + 16) * 4 + (size_t)S * as * 4 + (size_t)Aact * as * 4 +
+      16 * S * 8 + 16 * lx * 2 + 2 * 16 * lh * 2 + 16 * 16 * 4 +
+      16 * Aact * 8 + 64 + 16 * 16;
+  hipLaunchKernelGGL(fused_rollout_mfma_kernel, dim3(blocks),
+                     dim3(RO_THREADS), (int)lds, (hipStream_t)stream, state,
+                     step_ct, (const __hip_bfloat16*)w1,
+                     (const __hip_bfloat16*)b1, (const __hip_bfloat16*)w2,
+                     (const __hip_bfloat16*)b2, (const __hip_bfloat16*)w3,
+                     (const __hip_bfloat16*)b3, Amat, Bmat, eps, noise,
+                     st_obs, st_act, st_logp, st_nobs, st_rew, st_done, B, S,
+                     H1, Aact, T, max_steps, inv_softplus_bias, scale_lb);
 }

@@ -256,6 +256,20 @@ class SyntheticMuJoCoEnv(EnvBase):
         w1, w2, w3 = policy.linears
         eps_all = torch.randn(T, B, self.act_dim, device=self.device)
         noise_all = torch.randn(T, B, self.obs_dim, device=self.device) * 0.1
+        # bf16 weight caches (when the actor has them): the MFMA rollout
+        # variant computes the policy on the matrix cores with the SAME
+        # weights the update phase uses
+        bf16_w = []
+        import os
+
+        if os.environ.get("RL_AMD_ROLLOUT_MFMA", "1") != "0":
+            for lin in (w1, w2, w3):
+                wb = getattr(lin, "weight_bf16", None)
+                bb = getattr(lin, "bias_bf16", None)
+                if wb is None or bb is None:
+                    bf16_w = []
+                    break
+                bf16_w += [wb, bb]
         _C.fused_rollout(
             self._state,
             self._t.reshape(-1),
@@ -273,6 +287,7 @@ class SyntheticMuJoCoEnv(EnvBase):
             float(self.max_steps),
             policy.inv_softplus_bias,
             policy.scale_lb,
+            bf16_w,
         )
 
 

@@ -586,7 +586,8 @@ class TestFusedRollout:
         _C.fused_rollout(env1._state, env1._t.reshape(-1), w1, b1, w2, b2,
                          w3, b3, env1.A, env1.B, eps, noise,
                          st1["obs"], st1["act"], st1["lp"], st1["nobs"],
-                         st1["rew"], st1["done"], float(env1.max_steps), isb, 1e-4)
+                         st1["rew"], st1["done"], float(env1.max_steps), isb, 1e-4,
+                         [])
         for t in range(T):
             _C.fused_actor_into(env2._state, w1, b1, w2, b2, w3, b3, eps[t],
                                 st2["act"][:, t], st2["lp"][:, t], isb, 1e-4)
@@ -946,3 +947,68 @@ def test_value_pair_eval_matches_two_calls():
     with torch.no_grad():
         r0, r1 = fused(x0), fused(x1)
     assert torch.equal(y0, r0) and torch.equal(y1, r1)
+
+
+@pytest.mark.gpu
+def test_fused_rollout_mfma_vs_reference():
+    """MFMA rollout variant (policy on the matrix cores with bf16
+    weight caches): every step's action/log-prob must match a torch
+    reference computed from the STORED per-step observation (so state
+    drift cannot compound), and the fp32 env transition must match
+    exactly."""
+    import rl_amd._C as _C
+    from rl_amd.envs.custom.synthetic import HalfCheetahVec
+
+    torch.manual_seed(0)
+    B, T, H = 256, 6, 64
+    env = HalfCheetahVec(batch_size=[B], device="cuda")
+    env.reset()
+    S, A = env.obs_dim, env.act_dim
+    w1 = torch.randn(H, S, device="cuda") * 0.2
+    b1 = torch.randn(H, device="cuda") * 0.1
+    w2 = torch.randn(H, H, device="cuda") * 0.2
+    b2 = torch.randn(H, device="cuda") * 0.1
+    w3 = torch.randn(2 * A, H, device="cuda") * 0.2
+    b3 = torch.randn(2 * A, device="cuda") * 0.1
+    bf = [t.to(torch.bfloat16) for t in (w1, b1, w2, b2, w3, b3)]
+    eps = torch.randn(T, B, A, device="cuda")
+    noise = torch.randn(T, B, S, device="cuda") * 0.1
+    st = dict(
+        obs=torch.zeros(B, T, S, device="cuda"),
+        act=torch.zeros(B, T, A, device="cuda"),
+        lp=torch.zeros(B, T, device="cuda"),
+        nobs=torch.zeros(B, T, S, device="cuda"),
+        rew=torch.zeros(B, T, 1, device="cuda"),
+        done=torch.zeros(B, T, 1, dtype=torch.bool, device="cuda"),
+    )
+    isb = 0.5413248546129181
+    _C.fused_rollout(env._state, env._t.reshape(-1), w1, b1, w2, b2, w3, b3,
+                     env.A, env.B, eps, noise, st["obs"], st["act"],
+                     st["lp"], st["nobs"], st["rew"], st["done"],
+                     float(env.max_steps), isb, 1e-4, bf)
+    for t in range(T):
+        obs = st["obs"][:, t]
+        h1 = torch.tanh(
+            obs.to(torch.bfloat16).float() @ bf[0].float().t() + bf[1].float()
+        ).to(torch.bfloat16)
+        h2 = torch.tanh(h1.float() @ bf[2].float().t() + bf[3].float()).to(
+            torch.bfloat16
+        )
+        head = h2.float() @ bf[4].float().t() + bf[5].float()
+        loc, sraw = head.chunk(2, -1)
+        scale = torch.nn.functional.softplus(sraw + isb).clamp_min(1e-4)
+        u = loc + scale * eps[t]
+        act_ref = torch.tanh(u).clamp(-1 + 1.2e-7, 1 - 1.2e-7)
+        assert torch.allclose(st["act"][:, t], act_ref, atol=3e-3), (
+            t, (st["act"][:, t] - act_ref).abs().max()
+        )
+        lp_ref = (
+            -0.5 * eps[t] ** 2 - scale.log() - 0.5 * math.log(2 * math.pi)
+            - 2.0 * (math.log(2.0) - u - torch.nn.functional.softplus(-2 * u))
+        ).sum(-1)
+        assert torch.allclose(st["lp"][:, t], lp_ref, atol=2e-2, rtol=1e-3), (
+            t, (st["lp"][:, t] - lp_ref).abs().max()
+        )
+        # env transition is fp32 on the STORED action: tight tolerance
+        nobs_ref = torch.tanh(obs @ env.A + st["act"][:, t] @ env.B)
+        assert torch.allclose(st["nobs"][:, t], nobs_ref, atol=1e-5)
