@@ -51,7 +51,10 @@ def main():
     # bf16 compute: CDNA4 has no fp32 MFMA — fp32 GEMMs of the 256x256
     # MLPs ran single-workgroup at 80 us each (46% of the step in the
     # SAC rocprof); autocast moves them to the matrix cores
-    autocast = torch.autocast("cuda", dtype=torch.bfloat16, enabled=cuda,
+    import os as _os
+
+    _ac_on = cuda and _os.environ.get("RL_AMD_SAC_AUTOCAST", "1") != "0"
+    autocast = torch.autocast("cuda", dtype=torch.bfloat16, enabled=_ac_on,
                               cache_enabled=False)
 
     env = HumanoidVec(batch_size=[args.envs], device=device)
@@ -73,11 +76,30 @@ def main():
         MLP(in_features=obs_dim + act_dim, out_features=1, num_cells=[256, 256], device=device),
         in_keys=["observation", "action"],
     )
+    if cuda:
+        from rl_amd.ops import HAS_HIP_EXT, convert_linears_to_splitk
+
+        if HAS_HIP_EXT:
+            # split-K wgrad Linears BEFORE the loss builds its ensemble
+            # and target deep-copies, so every copy shares the class
+            convert_linears_to_splitk(actor)
+            convert_linears_to_splitk(qnet)
     loss = SACLoss(actor, qnet, num_qvalue_nets=2)
     loss.make_value_estimator()
     loss = loss.to(device)
+    if cuda:
+        from rl_amd.ops import HAS_HIP_EXT, enable_splitk_bf16_cache
+
+        if HAS_HIP_EXT:
+            # one bf16 weight cast per STEP for online, ensemble and
+            # target nets (autocast re-casts every call under capture)
+            enable_splitk_bf16_cache(loss)
     use_graph = bool(args.graph and cuda)
-    optim = torch.optim.Adam(loss.parameters(), lr=3e-4, capturable=use_graph)
+    try:
+        optim = torch.optim.Adam(loss.parameters(), lr=3e-4,
+                                 capturable=use_graph, fused=cuda)
+    except Exception:
+        optim = torch.optim.Adam(loss.parameters(), lr=3e-4, capturable=use_graph)
     updater = SoftUpdate(loss, tau=0.005)
 
     rb = TensorDictPrioritizedReplayBuffer(
@@ -112,10 +134,17 @@ def main():
         with autocast:
             out = loss(batch)
         total = out.get("loss_actor") + out.get("loss_qvalue") + out.get("loss_alpha")
-        optim.zero_grad(set_to_none=not use_graph)
+        # set_to_none even under capture: grads live in the graph pool
+        # (stable across replays) and the zero-fill + accumulate-add per
+        # parameter disappears (measured on the PPO step, r2)
+        optim.zero_grad(set_to_none=True)
         total.backward()
         optim.step()
         updater.step()
+        if cuda:
+            from rl_amd.ops import refresh_splitk_caches
+
+            refresh_splitk_caches(loss)
 
     # hipGraph-captured update: the SAC update at batch 256 is pure launch
     # overhead (hundreds of ~5us kernels); a single graph replay removes

@@ -87,13 +87,17 @@ class _EnsembleModule(nn.Module):
                     return False
                 layers = [sub for sub in m.modules() if isinstance(sub, _FAST)]
                 per_copy.append(layers)
+            def _kind(l):
+                # subclasses (e.g. ops.SplitKLinear) batch as their base
+                return "Linear" if isinstance(l, nn.Linear) else type(l).__name__
+
             sig = [
-                (type(l).__name__,
+                (_kind(l),
                  getattr(l, "in_features", None), getattr(l, "out_features", None))
                 for l in per_copy[0]
             ]
             ok = all(
-                [(type(l).__name__, getattr(l, "in_features", None),
+                [(_kind(l), getattr(l, "in_features", None),
                   getattr(l, "out_features", None)) for l in layers] == sig
                 for layers in per_copy[1:]
             )
@@ -106,6 +110,44 @@ class _EnsembleModule(nn.Module):
             self._batched = False
         return self._batched
 
+    def enable_bf16_cache(self):
+        """Pre-stack the per-copy Linear weights as bf16 buffers —
+        refreshed once per optimizer step (``refresh_bf16_cache_``)
+        instead of a stack + autocast cast per forward call."""
+        batched = self._try_batched_mlp()
+        if not batched:
+            return self
+        per_copy, sig, _, _ = batched
+        for li, (kind, _, _) in enumerate(sig):
+            if kind != "Linear":
+                continue
+            W = torch.stack(
+                [per_copy[n][li].weight.detach() for n in range(self.num_copies)]
+            ).to(torch.bfloat16)
+            b = torch.stack(
+                [per_copy[n][li].bias.detach() for n in range(self.num_copies)]
+            ).to(torch.bfloat16)
+            self.register_buffer(f"_wstack_bf16_{li}", W)
+            self.register_buffer(f"_bstack_bf16_{li}", b)
+        self._stack_cache = True
+        return self
+
+    def refresh_bf16_cache_(self):
+        if not getattr(self, "_stack_cache", False):
+            return
+        per_copy, sig, _, _ = self._try_batched_mlp()
+        for li, (kind, _, _) in enumerate(sig):
+            if kind != "Linear":
+                continue
+            getattr(self, f"_wstack_bf16_{li}").copy_(
+                torch.stack([per_copy[n][li].weight.detach()
+                             for n in range(self.num_copies)])
+            )
+            getattr(self, f"_bstack_bf16_{li}").copy_(
+                torch.stack([per_copy[n][li].bias.detach()
+                             for n in range(self.num_copies)])
+            )
+
     def forward(self, td: TensorDictBase) -> TensorDictBase:
         """Run all copies on the same input; stack outputs along dim 0."""
         from ..tensordict import stack as td_stack
@@ -113,15 +155,31 @@ class _EnsembleModule(nn.Module):
         batched = self._try_batched_mlp()
         if batched:
             per_copy, sig, in_keys, out_keys = batched
+            # the cache is a buffer: it would SWALLOW gradients, so it
+            # only serves no-grad forwards (target-net evaluations)
+            cached = (
+                getattr(self, "_stack_cache", False)
+                and torch.is_autocast_enabled()
+                and not torch.is_grad_enabled()
+            )
             x = torch.cat([td.get(k) for k in in_keys], dim=-1)
             lead = x.shape[:-1]
             h = x.reshape(1, -1, x.shape[-1]).expand(self.num_copies, -1, x.shape[-1])
             li = 0
             for kind, fin, fout in sig:
                 if kind == "Linear":
-                    W = torch.stack([per_copy[n][li].weight for n in range(self.num_copies)])
-                    bias = torch.stack([per_copy[n][li].bias for n in range(self.num_copies)])
-                    h = torch.baddbmm(bias.unsqueeze(1), h, W.transpose(-2, -1))
+                    if cached:
+                        W = getattr(self, f"_wstack_bf16_{li}")
+                        bias = getattr(self, f"_bstack_bf16_{li}")
+                        with torch.autocast("cuda", enabled=False):
+                            h = torch.baddbmm(
+                                bias.unsqueeze(1), h.to(torch.bfloat16),
+                                W.transpose(-2, -1),
+                            )
+                    else:
+                        W = torch.stack([per_copy[n][li].weight for n in range(self.num_copies)])
+                        bias = torch.stack([per_copy[n][li].bias for n in range(self.num_copies)])
+                        h = torch.baddbmm(bias.unsqueeze(1), h, W.transpose(-2, -1))
                 else:
                     h = getattr(torch, kind.lower())(h) if kind == "Tanh" else getattr(torch.nn.functional, kind.lower())(h)
                 li += 1

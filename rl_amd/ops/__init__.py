@@ -535,6 +535,9 @@ def enable_splitk_bf16_cache(module: torch.nn.Module) -> None:
     for m in module.modules():
         if isinstance(m, SplitKLinear):
             m.enable_bf16_cache()
+        elif hasattr(m, "enable_bf16_cache") and not isinstance(m, SplitKLinear):
+            # e.g. loss ensembles with their own stacked-weight caches
+            m.enable_bf16_cache()
 
 
 def refresh_splitk_caches(*modules: torch.nn.Module) -> None:
@@ -554,6 +557,10 @@ def refresh_splitk_caches(*modules: torch.nn.Module) -> None:
                     srcs.append(m.bias.detach())
     if dsts:
         torch._foreach_copy_(dsts, srcs)
+    for module in modules:
+        for m in module.modules():
+            if hasattr(m, "refresh_bf16_cache_") and not isinstance(m, SplitKLinear):
+                m.refresh_bf16_cache_()
 
 
 def convert_linears_to_splitk(module: torch.nn.Module) -> torch.nn.Module:
