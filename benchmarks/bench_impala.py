@@ -127,8 +127,14 @@ def main():
     # behavior policy: slightly stale copy of the learner (IMPALA lag)
     behavior = ImpalaNet(A, device=device)
     behavior.load_state_dict(net.state_dict())
-    optim = torch.optim.Adam(net.parameters(), lr=6e-4)
+    try:
+        optim = torch.optim.Adam(net.parameters(), lr=6e-4, fused=cuda)
+    except Exception:
+        optim = torch.optim.Adam(net.parameters(), lr=6e-4)
     gamma = 0.99
+
+    _b_params = [p.detach() for p in behavior.parameters()] + list(behavior.buffers())
+    _n_params = [p.detach() for p in net.parameters()] + list(net.buffers())
 
     obs = env.reset()
     # static rollout buffers: reused across iterations so the whole
@@ -214,7 +220,8 @@ def main():
                 g.copy_(flat_g[i : i + g.numel()].reshape(g.shape))
                 i += g.numel()
         optim.step()
-        behavior.load_state_dict(net.state_dict())
+        # one foreach copy instead of load_state_dict's per-tensor loop
+        torch._foreach_copy_(_b_params, _n_params)
 
     for _ in range(args.warmup):
         one_iteration()
