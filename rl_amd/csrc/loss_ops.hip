@@ -461,11 +461,43 @@ extern "C" void launch_smooth_l1_bwd(const void* v, const float* t,
 
 namespace {
 
+// fast-math variants for the bf16 head instantiation: the head is
+// already bf16-rounded (~3 decimal digits), so ~1-ulp-of-fp32
+// intrinsic error is invisible; the fp32 instantiation keeps the
+// precise libm calls for the oracle tests.
+template <bool FAST>
+__device__ __forceinline__ float ph_exp(float x) {
+  return FAST ? __expf(x) : expf(x);
+}
+template <bool FAST>
+__device__ __forceinline__ float ph_log(float x) {
+  return FAST ? __logf(x) : __logf(x);
+}
+template <bool FAST>
+__device__ __forceinline__ float ph_tanh(float x) {
+  if (FAST) {
+    const float cx = fminf(fmaxf(x, -15.f), 15.f);
+    const float t = __expf(2.f * cx);
+    return (t - 1.f) / (t + 1.f);
+  }
+  return tanhf(x);
+}
+template <bool FAST>
+__device__ __forceinline__ float ph_atanh(float y) {
+  if (FAST) return 0.5f * __logf((1.f + y) / (1.f - y));
+  return atanhf(y);
+}
+template <bool FAST>
+__device__ __forceinline__ float ph_softplus(float x) {
+  if (x > 20.f) return x;
+  return FAST ? __logf(1.f + __expf(x)) : log1pf(__expf(x));
+}
+
 #define PH_LOG_SQRT_2PI 0.9189385332046727f
 #define PH_LOG2 0.6931471805599453f
 #define PH_ATANH_LIM (1.0f - 1.1920929e-7f)
 
-template <typename TV>
+template <typename TV, bool FAST>
 __global__ void ppo_head_fwd_partials_k(
     const TV* __restrict__ head,      // [N, 2A]: loc | raw scale
     const float* __restrict__ action, // [N, A]
@@ -492,24 +524,24 @@ __global__ void ppo_head_fwd_partials_k(
     for (int a = 0; a < A; ++a) {
       const float loc = (float)head[n * 2 * A + a];
       const float spre = (float)head[n * 2 * A + A + a] + sp_bias;
-      const float s = fmaxf(softplusf(spre), lb);
+      const float s = fmaxf(ph_softplus<FAST>(spre), lb);
       const float y = fminf(fmaxf(action[n * A + a], -PH_ATANH_LIM),
                             PH_ATANH_LIM);
-      const float u = atanhf(y);
+      const float u = ph_atanh<FAST>(y);
       const float z = (u - loc) / s;
       lp += -0.5f * z * z - __logf(s) - PH_LOG_SQRT_2PI
-            - 2.0f * (PH_LOG2 - u - softplusf(-2.0f * u));
+            - 2.0f * (PH_LOG2 - u - ph_softplus<FAST>(-2.0f * u));
       const float e = eps[n * A + a];
-      const float x = tanhf(loc + s * e);
+      const float x = ph_tanh<FAST>(loc + s * e);
       ent += 0.5f * e * e + __logf(s) + PH_LOG_SQRT_2PI + log1pf(-x * x);
     }
     const float w = lp - prev[n];
     const float an = (adv[n] - mu) * isd;
-    const float r = expf(w);
-    const float rc = expf(fminf(fmaxf(w, lo), hi));
+    const float r = ph_exp<FAST>(w);
+    const float rc = ph_exp<FAST>(fminf(fmaxf(w, lo), hi));
     sg += fminf(r * an, rc * an);
     sr += r;
-    sr2 += expf(2.0f * w);
+    sr2 += ph_exp<FAST>(2.0f * w);
     sc += (rc != r) ? 1.f : 0.f;
     se += ent;
   }
@@ -590,7 +622,7 @@ __global__ void ppo_head_finalize_k(const float* __restrict__ part,
   }
 }
 
-template <typename TV>
+template <typename TV, bool FAST>
 __global__ void ppo_head_bwd_k(
     const TV* __restrict__ head, const float* __restrict__ action,
     const float* __restrict__ eps, const float* __restrict__ prev,
@@ -625,31 +657,31 @@ __global__ void ppo_head_bwd_k(
     for (int a = 0; a < A; ++a) {
       const float loc = (float)head[n * 2 * A + a];
       const float s = fmaxf(
-          softplusf((float)head[n * 2 * A + A + a] + sp_bias), lb);
+          ph_softplus<FAST>((float)head[n * 2 * A + A + a] + sp_bias), lb);
       const float y = fminf(fmaxf(action[n * A + a], -PH_ATANH_LIM),
                             PH_ATANH_LIM);
-      const float u = atanhf(y);
+      const float u = ph_atanh<FAST>(y);
       const float z = (u - loc) / s;
       lp += -0.5f * z * z - __logf(s) - PH_LOG_SQRT_2PI
-            - 2.0f * (PH_LOG2 - u - softplusf(-2.0f * u));
+            - 2.0f * (PH_LOG2 - u - ph_softplus<FAST>(-2.0f * u));
     }
     const float w = lp - prev[n];
     const float an = (adv[n] - mu) * isd;
-    const float r = expf(w);
-    const float rc = expf(fminf(fmaxf(w, lo), hi));
+    const float r = ph_exp<FAST>(w);
+    const float rc = ph_exp<FAST>(fminf(fmaxf(w, lo), hi));
     const float dlw = (r * an <= rc * an) ? g1s * an * r : 0.f;
     // pass 2: analytic d(head)
     for (int a = 0; a < A; ++a) {
       const float loc = (float)head[n * 2 * A + a];
       const float spre = (float)head[n * 2 * A + A + a] + sp_bias;
-      const float s0 = softplusf(spre);
+      const float s0 = ph_softplus<FAST>(spre);
       const float s = fmaxf(s0, lb);
       const float y = fminf(fmaxf(action[n * A + a], -PH_ATANH_LIM),
                             PH_ATANH_LIM);
-      const float u = atanhf(y);
+      const float u = ph_atanh<FAST>(y);
       const float z = (u - loc) / s;
       const float e = eps[n * A + a];
-      const float x = tanhf(loc + s * e);
+      const float x = ph_tanh<FAST>(loc + s * e);
       const float dloc = dlw * z / s + ges * (-2.0f * x);
       float ds = dlw * (z * z - 1.0f) / s + ges * (1.0f / s - 2.0f * x * e);
       ds = (s0 >= lb) ? ds : 0.f;  // clamp_min backward
@@ -673,13 +705,13 @@ extern "C" void launch_ppo_head_fwd(const void* head, const float* action,
                                     int head_is_bf16, void* stream) {
   const int blocks = red_blocks(N);
   if (head_is_bf16)
-    hipLaunchKernelGGL(ppo_head_fwd_partials_k<__hip_bfloat16>, dim3(blocks),
+    hipLaunchKernelGGL((ppo_head_fwd_partials_k<__hip_bfloat16, true>), dim3(blocks),
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)head, action, eps, prev, adv,
                        stats, (const __hip_bfloat16*)value, vtarget, part,
                        sp_bias, lb, lo, hi, N, A);
   else
-    hipLaunchKernelGGL(ppo_head_fwd_partials_k<float>, dim3(blocks),
+    hipLaunchKernelGGL((ppo_head_fwd_partials_k<float, false>), dim3(blocks),
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const float*)head, action, eps, prev, adv, stats,
                        (const float*)value, vtarget, part, sp_bias, lb, lo,
@@ -704,7 +736,7 @@ extern "C" void launch_ppo_head_bwd(const void* head, const float* action,
                                     int head_is_bf16, void* stream) {
   const int blocks = red_blocks(N);
   if (head_is_bf16)
-    hipLaunchKernelGGL(ppo_head_bwd_k<__hip_bfloat16>, dim3(blocks),
+    hipLaunchKernelGGL((ppo_head_bwd_k<__hip_bfloat16, true>), dim3(blocks),
                        dim3(LP_THREADS), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)head, action, eps, prev, adv,
                        stats, (const __hip_bfloat16*)value, vtarget, gobj,
@@ -712,7 +744,7 @@ extern "C" void launch_ppo_head_bwd(const void* head, const float* action,
                        (__hip_bfloat16*)dvalue, sp_bias, lb, lo, hi,
                        ent_coeff, crit_scale, N, A);
   else
-    hipLaunchKernelGGL(ppo_head_bwd_k<float>, dim3(blocks), dim3(LP_THREADS),
+    hipLaunchKernelGGL((ppo_head_bwd_k<float, false>), dim3(blocks), dim3(LP_THREADS),
                        0, (hipStream_t)stream, (const float*)head, action,
                        eps, prev, adv, stats, (const float*)value, vtarget,
                        gobj, gent, gact, gcrit, gtot, (float*)dhead,
