@@ -254,6 +254,16 @@ namespace {
 using ro_bfrag = __attribute__((ext_vector_type(8))) short;
 using ro_ffrag = __attribute__((ext_vector_type(4))) float;
 
+__device__ __forceinline__ float ro_fast_tanh(float x) {
+  const float cx = fminf(fmaxf(x, -15.f), 15.f);
+  const float t = __expf(2.f * cx);
+  return (t - 1.f) / (t + 1.f);
+}
+__device__ __forceinline__ float ro_fast_softplus(float x) {
+  if (x > 20.f) return x;
+  return __logf(1.f + __expf(x));
+}
+
 __device__ __forceinline__ void ro_gemm_tile(
     const __hip_bfloat16* s_a, int lda, const __hip_bfloat16* s_w, int ldw,
     int j0, int Kp, int lane, ro_ffrag* acc) {
@@ -366,7 +376,7 @@ __global__ void __launch_bounds__(RO_THREADS) fused_rollout_mfma_kernel(
 #pragma unroll
       for (int q = 0; q < 4; ++q)
         s_h1[(size_t)(erow + q) * lh + col] =
-            __hip_bfloat16(tanhf(acc[q] + bias));
+            __hip_bfloat16(ro_fast_tanh(acc[q] + bias));
     }
     __syncthreads();
     // layer 2
@@ -378,7 +388,7 @@ __global__ void __launch_bounds__(RO_THREADS) fused_rollout_mfma_kernel(
 #pragma unroll
       for (int q = 0; q < 4; ++q)
         s_h2[(size_t)(erow + q) * lh + col] =
-            __hip_bfloat16(tanhf(acc[q] + bias));
+            __hip_bfloat16(ro_fast_tanh(acc[q] + bias));
     }
     __syncthreads();
     // heads (one 16x16 tile; wave 0)
@@ -395,17 +405,18 @@ __global__ void __launch_bounds__(RO_THREADS) fused_rollout_mfma_kernel(
     for (int i = tid; i < rows * Aact; i += RO_THREADS) {
       const int r = i / Aact, a = i % Aact;
       const float loc = s_head[r * 16 + a];
-      float scale = softplusf_(s_head[r * 16 + Aact + a] + inv_softplus_bias);
+      float scale =
+          ro_fast_softplus(s_head[r * 16 + Aact + a] + inv_softplus_bias);
       scale = fmaxf(scale, scale_lb);
       const float e = eps[((size_t)t * B + row0 + r) * Aact + a];
       const float u = loc + scale * e;
-      float act = tanhf(u);
+      float act = ro_fast_tanh(u);
       act = fminf(fmaxf(act, -lim), lim);
       s_act[r * Aact + a] = act;
       st_act[((size_t)(row0 + r) * T + t) * Aact + a] = act;
       s_lp[r * Aact + a] =
           -0.5f * e * e - __logf(scale) - LOG_SQRT_2PI
-          - 2.0f * (LOG2 - u - softplusf_(-2.0f * u));
+          - 2.0f * (LOG2 - u - ro_fast_softplus(-2.0f * u));
     }
     __syncthreads();
     for (int r = tid; r < rows; r += RO_THREADS) {
