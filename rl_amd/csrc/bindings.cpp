@@ -108,6 +108,21 @@ void launch_mlp3_mfma_bwd2(const void* const*, void* const*, long,
 void launch_mlp3_mfma_fwdpair(const void*, const void*, int,
                               const void* const*, void*, void*, void*, long,
                               int, int, int, void*);
+void launch_mlp3_mfma_fwd2_loss(const void*, int, const void* const*,
+                                void* const*, void*, const float*,
+                                const float*, const float*, const float*,
+                                const float*, const float*, float*,
+                                float* const*, float, float, float, float,
+                                float, float, long, int, const int*,
+                                const int*, int, void*);
+void launch_mlp3_mfma_bwd2_loss(const void* const*, void* const*,
+                                const float*, const float*, const float*,
+                                const float*, const float*, const float*,
+                                const float*, const float*, const float*,
+                                const float*, const float*, void*, void*,
+                                const void*, const void*, float, float,
+                                float, float, float, float, long,
+                                const int*, const int*, int, void*);
 void launch_mlp3_mfma_fwd(const void*, int, const void*, const void*,
                           const void*, const void*, const void*, const void*,
                           void*, void*, void*, void*, long, int, int, int,
@@ -1063,6 +1078,118 @@ std::vector<torch::Tensor> mlp3_mfma_fwdpair(torch::Tensor x0,
   return {out0, out1};
 }
 
+// Whole-minibatch fused forward: actor+critic MLPs AND every PPO loss
+// scalar in one launch + one finalize (csrc/fused_mlp.hip).
+std::vector<torch::Tensor> acloss_fwd(
+    torch::Tensor x, std::vector<torch::Tensor> aw,
+    std::vector<torch::Tensor> cw, torch::Tensor action, torch::Tensor eps,
+    torch::Tensor prev, torch::Tensor adv, torch::Tensor vtarget,
+    double sp_bias, double lb, double lo, double hi, double ent_coeff,
+    double crit_scale, bool normalize) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x cuda contiguous");
+  const bool xf32 = x.scalar_type() == torch::kFloat32;
+  const long N = x.size(0), O = x.size(1);
+  int H[2] = {(int)aw[0].size(0), (int)cw[0].size(0)};
+  int A2[2] = {(int)aw[4].size(0), (int)cw[4].size(0)};
+  const int Aact = A2[0] / 2;
+  TORCH_CHECK(A2[1] == 1 && A2[0] == 2 * Aact, "actor 2A, critic 1");
+  TORCH_CHECK(mlp3_mfma_ok(O, H[0], A2[0]) && mlp3_mfma_ok(O, H[1], A2[1]),
+              "unsupported dims");
+  TORCH_CHECK(action.is_contiguous() && eps.is_contiguous() &&
+                  prev.is_contiguous() && adv.is_contiguous() &&
+                  vtarget.is_contiguous(),
+              "loss inputs contiguous");
+  auto bopt = aw[0].options();
+  auto fopt = action.options();
+  const void* wp[12];
+  void* op[6];
+  std::vector<torch::Tensor> res;
+  for (int n = 0; n < 2; ++n) {
+    auto& w = n == 0 ? aw : cw;
+    for (int i = 0; i < 6; ++i) wp[n * 6 + i] = w[i].data_ptr();
+    auto outn = torch::empty({N, A2[n]}, bopt);
+    auto h1n = torch::empty({N, H[n]}, bopt);
+    auto h2n = torch::empty({N, H[n]}, bopt);
+    op[n * 3 + 0] = outn.data_ptr();
+    op[n * 3 + 1] = h1n.data_ptr();
+    op[n * 3 + 2] = h2n.data_ptr();
+    res.push_back(outn);
+    res.push_back(h1n);
+    res.push_back(h2n);
+  }
+  auto xb = xf32 ? torch::empty({N, O}, bopt) : x;
+  const int R = N >= 32768 ? 64 : 32;
+  const int blocks = (int)((N + R - 1) / R);
+  auto part = torch::empty({(long)blocks * 7 + 512}, fopt);
+  float* outp[8];
+  std::vector<torch::Tensor> scalars;
+  for (int i = 0; i < 8; ++i) {
+    scalars.push_back(torch::empty({}, fopt));
+    outp[i] = scalars.back().data_ptr<float>();
+  }
+  auto stats = normalize ? torch::empty({2}, fopt) : torch::empty({0}, fopt);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (normalize)
+    launch_adv_stats(adv.data_ptr<float>(), part.data_ptr<float>(),
+                     stats.data_ptr<float>(), N, (void*)stream);
+  launch_mlp3_mfma_fwd2_loss(
+      x.data_ptr(), xf32 ? 1 : 0, wp, op, xb.data_ptr(),
+      action.data_ptr<float>(), eps.data_ptr<float>(),
+      prev.data_ptr<float>(), adv.data_ptr<float>(),
+      normalize ? stats.data_ptr<float>() : nullptr,
+      vtarget.data_ptr<float>(), part.data_ptr<float>(), outp,
+      (float)sp_bias, (float)lb, (float)lo, (float)hi, (float)ent_coeff,
+      (float)crit_scale, N, (int)O, H, A2, Aact, (void*)stream);
+  res.push_back(xb);
+  for (auto& t : scalars) res.push_back(t);
+  res.push_back(stats);
+  // head, a_h1, a_h2, value, c_h1, c_h2, xb, 8 scalars, stats
+  return res;
+}
+
+std::vector<torch::Tensor> acloss_bwd(
+    torch::Tensor head, torch::Tensor value, torch::Tensor a_h1,
+    torch::Tensor a_h2, torch::Tensor a_w2, torch::Tensor a_w3,
+    torch::Tensor c_h1, torch::Tensor c_h2, torch::Tensor c_w2,
+    torch::Tensor c_w3, torch::Tensor action, torch::Tensor eps,
+    torch::Tensor prev, torch::Tensor adv, torch::Tensor stats,
+    torch::Tensor vtarget, torch::Tensor gobj, torch::Tensor gent,
+    torch::Tensor gact, torch::Tensor gcrit, torch::Tensor gtot,
+    double sp_bias, double lb, double lo, double hi, double ent_coeff,
+    double crit_scale) {
+  const long N = head.size(0);
+  int H[2] = {(int)a_h1.size(1), (int)c_h1.size(1)};
+  int A2[2] = {(int)head.size(1), 1};
+  const int Aact = A2[0] / 2;
+  auto bopt = head.options();
+  auto dhead = torch::empty_like(head);
+  auto dvalue = torch::empty_like(value);
+  const void* hw[8] = {a_h1.data_ptr(), a_h2.data_ptr(), a_w2.data_ptr(),
+                       a_w3.data_ptr(), c_h1.data_ptr(), c_h2.data_ptr(),
+                       c_w2.data_ptr(), c_w3.data_ptr()};
+  auto a_dh1 = torch::empty({N, H[0]}, bopt);
+  auto a_dh2 = torch::empty({N, H[0]}, bopt);
+  auto c_dh1 = torch::empty({N, H[1]}, bopt);
+  auto c_dh2 = torch::empty({N, H[1]}, bopt);
+  void* dh[4] = {a_dh1.data_ptr(), a_dh2.data_ptr(), c_dh1.data_ptr(),
+                 c_dh2.data_ptr()};
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_mlp3_mfma_bwd2_loss(
+      hw, dh, action.data_ptr<float>(), eps.data_ptr<float>(),
+      prev.data_ptr<float>(), adv.data_ptr<float>(),
+      stats.numel() ? stats.data_ptr<float>() : nullptr,
+      vtarget.data_ptr<float>(),
+      gobj.numel() ? gobj.data_ptr<float>() : nullptr,
+      gent.numel() ? gent.data_ptr<float>() : nullptr,
+      gact.numel() ? gact.data_ptr<float>() : nullptr,
+      gcrit.numel() ? gcrit.data_ptr<float>() : nullptr,
+      gtot.numel() ? gtot.data_ptr<float>() : nullptr, dhead.data_ptr(),
+      dvalue.data_ptr(), head.data_ptr(), value.data_ptr(), (float)sp_bias,
+      (float)lb, (float)lo, (float)hi, (float)ent_coeff, (float)crit_scale,
+      N, H, A2, Aact, (void*)stream);
+  return {dhead, dvalue, a_dh1, a_dh2, c_dh1, c_dh2};
+}
+
 // Mega-fused TanhNormal head loss: raw actor-head output [N, 2A] ->
 // (out[5] = {loss_objective, ESS/N, clip_fraction, entropy_mean,
 // loss_entropy}, stats).  See csrc/loss_ops.hip.
@@ -1221,6 +1348,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dual-network MFMA dgrad chain (HIP)");
   m.def("mlp3_mfma_fwdpair", &mlp3_mfma_fwdpair,
         "same-network pair forward (HIP)");
+  m.def("acloss_fwd", &acloss_fwd,
+        "actor+critic MLPs + every PPO loss scalar in one launch (HIP)");
+  m.def("acloss_bwd", &acloss_bwd,
+        "merged d(head)/d(value) + dgrad chains in one launch (HIP)");
   m.def("fused_grad_clip_coef", &fused_grad_clip_coef,
         "single-kernel global grad-norm clip coefficient (HIP)");
   m.def("multi_gather", &multi_gather,

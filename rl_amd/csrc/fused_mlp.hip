@@ -777,3 +777,440 @@ extern "C" void launch_mlp3_mfma_fwdpair(const void* x0, const void* x1,
   }
 #undef M3_LAUNCH_PAIR
 }
+
+// ---------------------------------------------------------------------------
+// Dual-network kernels WITH the PPO loss fused in (the whole-minibatch
+// endgame): grid.y = 0 runs the actor MLP and emits the clipped-PPO /
+// entropy per-workgroup partial sums straight from its LDS-staged head
+// rows; grid.y = 1 runs the critic and emits the smooth-L1 partial.
+// One finalize writes the 8 loss scalars.  The backward twin computes
+// d(head)/d(value) per row IN the dgrad kernel (writing them to global
+// for the batched wgrad) before running the dgrad chain.  Loss math is
+// copied 1:1 from csrc/loss_ops.hip (fast-math forms: both nets are
+// bf16-quantized).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+__device__ __forceinline__ float ac_exp(float x) { return __expf(x); }
+__device__ __forceinline__ float ac_log(float x) { return __logf(x); }
+__device__ __forceinline__ float ac_tanh(float x) {
+  const float cx = fminf(fmaxf(x, -15.f), 15.f);
+  const float t = __expf(2.f * cx);
+  return (t - 1.f) / (t + 1.f);
+}
+__device__ __forceinline__ float ac_atanh(float y) {
+  return 0.5f * __logf((1.f + y) / (1.f - y));
+}
+__device__ __forceinline__ float ac_softplus(float x) {
+  if (x > 20.f) return x;
+  return __logf(1.f + __expf(x));
+}
+
+#define AC_LOG_SQRT_2PI 0.9189385332046727f
+#define AC_LOG2 0.6931471805599453f
+#define AC_ATANH_LIM (1.0f - 1.1920929e-7f)
+
+struct ACLossArgs {
+  const float* action;   // [N, A]
+  const float* eps;      // [N, A]
+  const float* prev;     // [N]
+  const float* adv;      // [N]
+  const float* stats;    // nullable (mu, 1/sigma)
+  const float* vtarget;  // [N]
+  float* part;           // [nWG, 7]: 0..4 actor sums (WG y=0), 5 crit,
+                         // 6 unused
+  float sp_bias, lb, lo, hi;
+  int Aact;
+};
+
+__device__ __forceinline__ float ac_block_sum(float v, float* smem) {
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) smem[wave] = v;
+  __syncthreads();
+  if (wave == 0) {
+    v = (lane < (int)(blockDim.x >> 6)) ? smem[lane] : 0.f;
+    for (int off = 2; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  }
+  return v;
+}
+
+template <typename TX, int R>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_fwd2_loss_kernel(
+    const TX* __restrict__ x, const AC2Fwd nets, const ACLossArgs loss,
+    __hip_bfloat16* __restrict__ xb_out, const int N, const int O) {
+  const int net = blockIdx.y;
+  mlp3_mfma_fwd_impl<TX, R>(x, nets.w1[net], nets.b1[net], nets.w2[net],
+                            nets.b2[net], nets.w3[net], nets.b3[net],
+                            nets.out[net], nets.h1[net], nets.h2[net],
+                            net == 0 ? xb_out : nullptr, N, O, nets.H[net],
+                            nets.A2[net]);
+  // the head/value rows this WG just produced are still LDS-staged in
+  // the impl's s_h1 region (same extern-smem layout math as the impl)
+  const int H = nets.H[net];
+  const int Op = m3_pad32(O), Hp = m3_pad32(H);
+  const int lx = Op + 8, lh = Hp + 8;
+  extern __shared__ __hip_bfloat16 smem[];
+  const __hip_bfloat16* s_head = smem + (size_t)R * lx + (size_t)H * lx;
+  const long row0 = (long)blockIdx.x * R;
+  const int rows = (int)min((long)R, (long)N - row0);
+  const int tid = threadIdx.x;
+  __shared__ float red[8];
+  const int A = loss.Aact;
+  if (net == 1) {
+    // critic: smooth-L1 partial over this WG's rows
+    float sv = 0.f;
+    for (int r = tid; r < rows; r += MLP_THREADS) {
+      const float z =
+          __bfloat162float(s_head[(size_t)r * lh]) - loss.vtarget[row0 + r];
+      const float az = fabsf(z);
+      sv += (az < 1.f) ? 0.5f * z * z : az - 0.5f;
+    }
+    const float t = ac_block_sum(sv, red);
+    if (tid == 0) loss.part[(size_t)blockIdx.x * 7 + 5] = t;
+    return;
+  }
+  // actor: clipped surrogate + entropy partials from the staged head
+  const float mu = loss.stats ? loss.stats[0] : 0.f;
+  const float isd = loss.stats ? loss.stats[1] : 1.f;
+  float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f;
+  for (int r = tid; r < rows; r += MLP_THREADS) {
+    const long n = row0 + r;
+    float lp = 0.f, ent = 0.f;
+    for (int a = 0; a < A; ++a) {
+      const float loc = __bfloat162float(s_head[(size_t)r * lh + a]);
+      const float spre =
+          __bfloat162float(s_head[(size_t)r * lh + A + a]) + loss.sp_bias;
+      const float sv_ = fmaxf(ac_softplus(spre), loss.lb);
+      const float y = fminf(fmaxf(loss.action[n * A + a], -AC_ATANH_LIM),
+                            AC_ATANH_LIM);
+      const float u = ac_atanh(y);
+      const float z = (u - loc) / sv_;
+      lp += -0.5f * z * z - ac_log(sv_) - AC_LOG_SQRT_2PI
+            - 2.0f * (AC_LOG2 - u - ac_softplus(-2.0f * u));
+      const float e = loss.eps[n * A + a];
+      const float xx = ac_tanh(loc + sv_ * e);
+      ent += 0.5f * e * e + ac_log(sv_) + AC_LOG_SQRT_2PI + log1pf(-xx * xx);
+    }
+    const float w = lp - loss.prev[n];
+    const float an = (loss.adv[n] - mu) * isd;
+    const float rr = ac_exp(w);
+    const float rc = ac_exp(fminf(fmaxf(w, loss.lo), loss.hi));
+    sg += fminf(rr * an, rc * an);
+    sr += rr;
+    sr2 += ac_exp(2.0f * w);
+    sc += (rc != rr) ? 1.f : 0.f;
+    se += ent;
+  }
+  float t;
+  t = ac_block_sum(sg, red);
+  if (tid == 0) loss.part[(size_t)blockIdx.x * 7 + 0] = t;
+  __syncthreads();
+  t = ac_block_sum(sr, red);
+  if (tid == 0) loss.part[(size_t)blockIdx.x * 7 + 1] = t;
+  __syncthreads();
+  t = ac_block_sum(sr2, red);
+  if (tid == 0) loss.part[(size_t)blockIdx.x * 7 + 2] = t;
+  __syncthreads();
+  t = ac_block_sum(sc, red);
+  if (tid == 0) loss.part[(size_t)blockIdx.x * 7 + 3] = t;
+  __syncthreads();
+  t = ac_block_sum(se, red);
+  if (tid == 0) loss.part[(size_t)blockIdx.x * 7 + 4] = t;
+}
+
+struct ACOuts {
+  float* o[8];
+};
+
+__global__ void acloss_finalize_k(const float* __restrict__ part,
+                                  const int nwg, const long N,
+                                  const float ent_coeff,
+                                  const float crit_scale,
+                                  const ACOuts outs_) {
+  float* const* outs = outs_.o;
+  float sg = 0.f, sr = 0.f, sr2 = 0.f, sc = 0.f, se = 0.f, sv = 0.f;
+  for (int i = threadIdx.x; i < nwg; i += blockDim.x) {
+    sg += part[7 * i];
+    sr += part[7 * i + 1];
+    sr2 += part[7 * i + 2];
+    sc += part[7 * i + 3];
+    se += part[7 * i + 4];
+    sv += part[7 * i + 5];
+  }
+  __shared__ float smem[8];
+  float t;
+  t = ac_block_sum(sg, smem);
+  if (threadIdx.x == 0) smem[4] = t;
+  __syncthreads();
+  const float tsg = smem[4];
+  t = ac_block_sum(sr, smem);
+  if (threadIdx.x == 0) smem[4] = t;
+  __syncthreads();
+  const float tsr = smem[4];
+  t = ac_block_sum(sr2, smem);
+  if (threadIdx.x == 0) smem[4] = t;
+  __syncthreads();
+  const float tsr2 = smem[4];
+  t = ac_block_sum(sc, smem);
+  if (threadIdx.x == 0) smem[4] = t;
+  __syncthreads();
+  const float tsc = smem[4];
+  t = ac_block_sum(se, smem);
+  if (threadIdx.x == 0) smem[4] = t;
+  __syncthreads();
+  const float tse = smem[4];
+  t = ac_block_sum(sv, smem);
+  if (threadIdx.x == 0) {
+    const float loss_obj = -tsg / (float)N;
+    *outs[0] = loss_obj;
+    *outs[1] = tsr * tsr / fmaxf(tsr2, 1e-12f) / (float)N;  // ESS/N
+    *outs[2] = tsc / (float)N;                              // clip_fraction
+    *outs[3] = tse / (float)N;                              // entropy
+    const float lent = -ent_coeff * tse / (float)N;
+    *outs[4] = lent;                                        // loss_entropy
+    const float lact = loss_obj + lent;
+    *outs[5] = lact;                                        // loss_actor
+    const float lcrit = crit_scale * t / (float)N;
+    *outs[6] = lcrit;                                       // loss_critic
+    *outs[7] = lact + lcrit;                                // total
+  }
+}
+
+struct ACLossBwdArgs {
+  const float* action;
+  const float* eps;
+  const float* prev;
+  const float* adv;
+  const float* stats;
+  const float* vtarget;
+  const float* gobj;   // nullable 0-d grads
+  const float* gent;
+  const float* gact;
+  const float* gcrit;
+  const float* gtot;
+  float sp_bias, lb, lo, hi, ent_coeff, crit_scale;
+  int Aact;
+};
+
+// dhead/dvalue computed per row into the GLOBAL dout buffers (the
+// batched wgrad reads them), then the standard dgrad chain runs.
+template <int R>
+__global__ void __launch_bounds__(MLP_THREADS) mlp3_mfma_bwd2_loss_kernel(
+    const AC2Bwd nets, const ACLossBwdArgs a,
+    __hip_bfloat16* __restrict__ dhead_out,   // [N, A2_actor]
+    __hip_bfloat16* __restrict__ dvalue_out,  // [N, 1]
+    const __hip_bfloat16* __restrict__ head,  // [N, A2_actor]
+    const __hip_bfloat16* __restrict__ value, // [N, 1]
+    const int N) {
+  const int net = blockIdx.y;
+  const long row0 = (long)blockIdx.x * R;
+  const int rows = (int)min((long)R, (long)N - row0);
+  const int tid = threadIdx.x;
+  const int A = a.Aact;
+  const float gt = a.gtot ? a.gtot[0] : 0.f;
+  if (net == 1) {
+    const float gvs = a.crit_scale *
+                      ((a.gcrit ? a.gcrit[0] : 0.f) + gt) / (float)N;
+    for (int r = tid; r < rows; r += MLP_THREADS) {
+      const long n = row0 + r;
+      const float z = __bfloat162float(value[n]) - a.vtarget[n];
+      dvalue_out[n] = __hip_bfloat16(gvs * fminf(fmaxf(z, -1.f), 1.f));
+    }
+  } else {
+    const float mu = a.stats ? a.stats[0] : 0.f;
+    const float isd = a.stats ? a.stats[1] : 1.f;
+    const float ga = (a.gact ? a.gact[0] : 0.f) + gt;
+    const float g1s = -(ga + (a.gobj ? a.gobj[0] : 0.f)) / (float)N;
+    const float ges =
+        -a.ent_coeff * (ga + (a.gent ? a.gent[0] : 0.f)) / (float)N;
+    for (int r = tid; r < rows; r += MLP_THREADS) {
+      const long n = row0 + r;
+      float lp = 0.f;
+      for (int q = 0; q < A; ++q) {
+        const float loc = __bfloat162float(head[n * 2 * A + q]);
+        const float sv_ = fmaxf(
+            ac_softplus(__bfloat162float(head[n * 2 * A + A + q]) +
+                        a.sp_bias),
+            a.lb);
+        const float y = fminf(fmaxf(a.action[n * A + q], -AC_ATANH_LIM),
+                              AC_ATANH_LIM);
+        const float u = ac_atanh(y);
+        const float z = (u - loc) / sv_;
+        lp += -0.5f * z * z - ac_log(sv_) - AC_LOG_SQRT_2PI
+              - 2.0f * (AC_LOG2 - u - ac_softplus(-2.0f * u));
+      }
+      const float w = lp - a.prev[n];
+      const float an = (a.adv[n] - mu) * isd;
+      const float rr = ac_exp(w);
+      const float rc = ac_exp(fminf(fmaxf(w, a.lo), a.hi));
+      const float dlw = (rr * an <= rc * an) ? g1s * an * rr : 0.f;
+      for (int q = 0; q < A; ++q) {
+        const float loc = __bfloat162float(head[n * 2 * A + q]);
+        const float spre =
+            __bfloat162float(head[n * 2 * A + A + q]) + a.sp_bias;
+        const float s0 = ac_softplus(spre);
+        const float sv_ = fmaxf(s0, a.lb);
+        const float y = fminf(fmaxf(a.action[n * A + q], -AC_ATANH_LIM),
+                              AC_ATANH_LIM);
+        const float u = ac_atanh(y);
+        const float z = (u - loc) / sv_;
+        const float e = a.eps[n * A + q];
+        const float xx = ac_tanh(loc + sv_ * e);
+        const float dloc = dlw * z / sv_ + ges * (-2.0f * xx);
+        float ds = dlw * (z * z - 1.0f) / sv_ +
+                   ges * (1.0f / sv_ - 2.0f * xx * e);
+        ds = (s0 >= a.lb) ? ds : 0.f;
+        const float sig = 1.0f / (1.0f + __expf(-spre));
+        dhead_out[n * 2 * A + q] = __hip_bfloat16(dloc);
+        dhead_out[n * 2 * A + A + q] = __hip_bfloat16(ds * sig);
+      }
+    }
+  }
+  __syncthreads();  // the WG's own global dout rows are what it stages
+  mlp3_mfma_bwd_impl<__hip_bfloat16, R>(
+      net == 0 ? dhead_out : dvalue_out, nets.h1[net], nets.h2[net],
+      nets.w2[net], nets.w3[net], nets.dh1[net], nets.dh2[net], N,
+      nets.H[net], nets.A2[net]);
+}
+
+}  // namespace
+
+extern "C" void launch_mlp3_mfma_fwd2_loss(
+    const void* x, int x_is_f32, const void* const* w, void* const* o,
+    void* xb, const float* action, const float* eps, const float* prev,
+    const float* adv, const float* stats, const float* vtarget, float* part,
+    float* const* outs, float sp_bias, float lb, float lo, float hi,
+    float ent_coeff, float crit_scale, long N, int O, const int* H,
+    const int* A2, int Aact, void* stream) {
+  AC2Fwd nets;
+  for (int n = 0; n < 2; ++n) {
+    nets.w1[n] = (const __hip_bfloat16*)w[n * 6 + 0];
+    nets.b1[n] = (const __hip_bfloat16*)w[n * 6 + 1];
+    nets.w2[n] = (const __hip_bfloat16*)w[n * 6 + 2];
+    nets.b2[n] = (const __hip_bfloat16*)w[n * 6 + 3];
+    nets.w3[n] = (const __hip_bfloat16*)w[n * 6 + 4];
+    nets.b3[n] = (const __hip_bfloat16*)w[n * 6 + 5];
+    nets.out[n] = (__hip_bfloat16*)o[n * 3 + 0];
+    nets.h1[n] = (__hip_bfloat16*)o[n * 3 + 1];
+    nets.h2[n] = (__hip_bfloat16*)o[n * 3 + 2];
+    nets.H[n] = H[n];
+    nets.A2[n] = A2[n];
+  }
+  ACLossArgs la;
+  la.action = action;
+  la.eps = eps;
+  la.prev = prev;
+  la.adv = adv;
+  la.stats = stats;
+  la.vtarget = vtarget;
+  la.part = part;
+  la.sp_bias = sp_bias;
+  la.lb = lb;
+  la.lo = lo;
+  la.hi = hi;
+  la.Aact = Aact;
+  const int R = N >= M3_ROWS_SWITCH ? 64 : 32;
+  const int blocks = (int)((N + R - 1) / R);
+  int lds = 0;
+  for (int n = 0; n < 2; ++n) {
+    const int Op = (O + 31) & ~31, Hp = (H[n] + 31) & ~31;
+    const int lx = Op + 8, lh = Hp + 8, A2p = (A2[n] + 15) & ~15;
+    const int b = 2 * (R * lx + H[n] * lx + 2 * R * lh + H[n] * lh + A2p * lh);
+    if (b > lds) lds = b;
+  }
+  dim3 grid(blocks, 2);
+  if (x_is_f32) {
+    if (R == 64)
+      hipLaunchKernelGGL((mlp3_mfma_fwd2_loss_kernel<float, 64>), grid,
+                         dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                         (const float*)x, nets, la, (__hip_bfloat16*)xb,
+                         (int)N, O);
+    else
+      hipLaunchKernelGGL((mlp3_mfma_fwd2_loss_kernel<float, 32>), grid,
+                         dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                         (const float*)x, nets, la, (__hip_bfloat16*)xb,
+                         (int)N, O);
+  } else {
+    if (R == 64)
+      hipLaunchKernelGGL((mlp3_mfma_fwd2_loss_kernel<__hip_bfloat16, 64>),
+                         grid, dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                         (const __hip_bfloat16*)x, nets, la,
+                         (__hip_bfloat16*)xb, (int)N, O);
+    else
+      hipLaunchKernelGGL((mlp3_mfma_fwd2_loss_kernel<__hip_bfloat16, 32>),
+                         grid, dim3(MLP_THREADS), lds, (hipStream_t)stream,
+                         (const __hip_bfloat16*)x, nets, la,
+                         (__hip_bfloat16*)xb, (int)N, O);
+  }
+  ACOuts ao;
+  for (int i = 0; i < 8; ++i) ao.o[i] = outs[i];
+  hipLaunchKernelGGL(acloss_finalize_k, dim3(1), dim3(MLP_THREADS), 0,
+                     (hipStream_t)stream, part, blocks, N, ent_coeff,
+                     crit_scale, ao);
+}
+
+extern "C" void launch_mlp3_mfma_bwd2_loss(
+    const void* const* hw,  // 8: a_h1,a_h2,a_w2,a_w3,c_h1,c_h2,c_w2,c_w3
+    void* const* dh,        // 4: a_dh1,a_dh2,c_dh1,c_dh2
+    const float* action, const float* eps, const float* prev,
+    const float* adv, const float* stats, const float* vtarget,
+    const float* gobj, const float* gent, const float* gact,
+    const float* gcrit, const float* gtot, void* dhead, void* dvalue,
+    const void* head, const void* value, float sp_bias, float lb, float lo,
+    float hi, float ent_coeff, float crit_scale, long N, const int* H,
+    const int* A2, int Aact, void* stream) {
+  AC2Bwd nets;
+  for (int n = 0; n < 2; ++n) {
+    nets.dout[n] = nullptr;
+    nets.h1[n] = (const __hip_bfloat16*)hw[n * 4 + 0];
+    nets.h2[n] = (const __hip_bfloat16*)hw[n * 4 + 1];
+    nets.w2[n] = (const __hip_bfloat16*)hw[n * 4 + 2];
+    nets.w3[n] = (const __hip_bfloat16*)hw[n * 4 + 3];
+    nets.dh1[n] = (__hip_bfloat16*)dh[n * 2 + 0];
+    nets.dh2[n] = (__hip_bfloat16*)dh[n * 2 + 1];
+    nets.H[n] = H[n];
+    nets.A2[n] = A2[n];
+  }
+  ACLossBwdArgs a;
+  a.action = action;
+  a.eps = eps;
+  a.prev = prev;
+  a.adv = adv;
+  a.stats = stats;
+  a.vtarget = vtarget;
+  a.gobj = gobj;
+  a.gent = gent;
+  a.gact = gact;
+  a.gcrit = gcrit;
+  a.gtot = gtot;
+  a.sp_bias = sp_bias;
+  a.lb = lb;
+  a.lo = lo;
+  a.hi = hi;
+  a.ent_coeff = ent_coeff;
+  a.crit_scale = crit_scale;
+  a.Aact = Aact;
+  const int R = N >= M3_ROWS_SWITCH ? 64 : 32;
+  const int blocks = (int)((N + R - 1) / R);
+  int lds = 0;
+  for (int n = 0; n < 2; ++n) {
+    const int ld = ((A2[n] + 31) & ~31) + 8, lh = ((H[n] + 31) & ~31) + 8;
+    const int b = 2 * (R * ld + H[n] * ld + 2 * R * lh + H[n] * lh);
+    if (b > lds) lds = b;
+  }
+  dim3 grid(blocks, 2);
+  if (R == 64)
+    hipLaunchKernelGGL((mlp3_mfma_bwd2_loss_kernel<64>), grid,
+                       dim3(MLP_THREADS), lds, (hipStream_t)stream, nets, a,
+                       (__hip_bfloat16*)dhead, (__hip_bfloat16*)dvalue,
+                       (const __hip_bfloat16*)head,
+                       (const __hip_bfloat16*)value, (int)N);
+  else
+    hipLaunchKernelGGL((mlp3_mfma_bwd2_loss_kernel<32>), grid,
+                       dim3(MLP_THREADS), lds, (hipStream_t)stream, nets, a,
+                       (__hip_bfloat16*)dhead, (__hip_bfloat16*)dvalue,
+                       (const __hip_bfloat16*)head,
+                       (const __hip_bfloat16*)value, (int)N);
+}

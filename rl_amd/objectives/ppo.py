@@ -7,6 +7,7 @@ from __future__ import annotations
 
 import dataclasses
 import math
+import os
 from typing import Optional, Union
 
 import torch
@@ -358,6 +359,30 @@ class ClipPPOLoss(PPOLoss):
                 and getattr(ac[0].lin1, "_bf16_cache", False)
                 and getattr(ac[1].lin1, "_bf16_cache", False)
             ):
+                A = action.shape[-1]
+                if (
+                    ac[0].lin3.out_features == 2 * A
+                    and os.environ.get("RL_AMD_MERGED_LOSS", "1") != "0"
+                ):
+                    # the fully-merged path: MLPs AND loss in one
+                    # launch pair (csrc/fused_mlp.hip acloss kernels)
+                    eps = torch.randn(N, A, device=action.device,
+                                      dtype=torch.float32)
+                    lo, hi = self._clip_bounds
+                    (loss_obj, loss_ent, ent_mean, ess, clip_frac,
+                     loss_act, loss_crit, loss_total) = ops.actor_critic_loss(
+                        obs.reshape(N, O), ac[0], ac[1],
+                        action.reshape(N, A).float(),
+                        prev_lp.reshape(N), advantage.reshape(N),
+                        target.reshape(N), eps,
+                        sp_bias=info["sp_bias"], scale_lb=info["scale_lb"],
+                        lo=lo, hi=hi,
+                        entropy_coeff=float(self.entropy_coeff),
+                        critic_scale=float(self.critic_coeff),
+                        normalize=normalize,
+                    )
+                    return (loss_obj, loss_ent, ent_mean, ess, clip_frac,
+                            loss_act, loss_crit, loss_total)
                 head, value = ops.actor_critic_mlp3(
                     obs.reshape(N, O), ac[0], ac[1]
                 )

@@ -1078,6 +1078,90 @@ def value_pair_eval(fused, x0, x1):
     return _C.mlp3_mfma_fwdpair(x0.contiguous(), x1.contiguous(), w)
 
 
+class _ACLossFn(torch.autograd.Function):
+    """THE whole-minibatch Function: actor+critic MFMA MLPs and every
+    ClipPPO loss scalar in one forward launch (+1-WG finalize), and one
+    backward launch computing d(head)/d(value) inline before the dgrad
+    chains, followed by the 6-layer batched wgrad
+    (csrc/fused_mlp.hip acloss kernels)."""
+
+    @staticmethod
+    def forward(ctx, x, aw1b, ab1b, aw2b, ab2b, aw3b, ab3b,
+                cw1b, cb1b, cw2b, cb2b, cw3b, cb3b,
+                aw1, ab1, aw2, ab2, aw3, ab3,
+                cw1, cb1, cw2, cb2, cw3, cb3,
+                action, prev_lp, adv, vtarget, eps,
+                sp_bias, lb, lo, hi, ent_coeff, crit_scale, normalize):
+        ctx.set_materialize_grads(False)
+        x = x.contiguous()
+        action = action.contiguous().detach()
+        prev_lp = prev_lp.contiguous().detach()
+        adv = adv.contiguous().detach()
+        vtarget = vtarget.contiguous().detach()
+        eps = eps.contiguous()
+        (head, a_h1, a_h2, value, c_h1, c_h2, xb,
+         loss_obj, ess, cf, ent, lent, lact, lcrit, ltotal,
+         stats) = _C.acloss_fwd(
+            x, [aw1b, ab1b, aw2b, ab2b, aw3b, ab3b],
+            [cw1b, cb1b, cw2b, cb2b, cw3b, cb3b],
+            action, eps, prev_lp, adv, vtarget,
+            sp_bias, lb, lo, hi, ent_coeff, crit_scale, normalize,
+        )
+        ctx.save_for_backward(head, value, a_h1, a_h2, aw2b, aw3b, c_h1,
+                              c_h2, cw2b, cw3b, action, eps, prev_lp, adv,
+                              stats, vtarget, xb)
+        ctx.cfg = (sp_bias, lb, lo, hi, ent_coeff, crit_scale)
+        ctx.mark_non_differentiable(ess, cf, ent)
+        return loss_obj, lent, ent, ess, cf, lact, lcrit, ltotal
+
+    @staticmethod
+    def backward(ctx, g_obj, g_ent, g_em, g_ess, g_cf, g_act, g_crit, g_tot):
+        (head, value, a_h1, a_h2, aw2b, aw3b, c_h1, c_h2, cw2b, cw3b,
+         action, eps, prev_lp, adv, stats, vtarget, xb) = ctx.saved_tensors
+        sp_bias, lb, lo, hi, ent_coeff, crit_scale = ctx.cfg
+        empty = head.new_empty(0, dtype=torch.float32)
+        g = lambda t: t.contiguous() if t is not None else empty
+        dhead, dvalue, a_dh1, a_dh2, c_dh1, c_dh2 = _C.acloss_bwd(
+            head, value, a_h1, a_h2, aw2b, aw3b, c_h1, c_h2, cw2b, cw3b,
+            action, eps, prev_lp, adv, stats, vtarget,
+            g(g_obj), g(g_ent), g(g_act), g(g_crit), g(g_tot),
+            sp_bias, lb, lo, hi, ent_coeff, crit_scale,
+        )
+        (adw3, adb3, adw2, adb2, adw1, adb1, cdw3, cdb3, cdw2, cdb2, cdw1,
+         cdb1) = _C.wgrad_splitk_batch(
+            [dhead, a_dh2, a_dh1, dvalue, c_dh2, c_dh1],
+            [a_h2, a_h1, xb, c_h2, c_h1, xb],
+        )
+        return ((None,) * 13
+                + (adw1, adb1, adw2, adb2, adw3, adb3,
+                   cdw1, cdb1, cdw2, cdb2, cdw3, cdb3)
+                + (None,) * 12)
+
+
+def actor_critic_loss(x, actor_fused, critic_fused, action, prev_log_prob,
+                      advantage, value_target, eps, *, sp_bias, scale_lb,
+                      lo, hi, entropy_coeff, critic_scale, normalize):
+    """Run the fully-merged actor+critic+loss Function.  Returns
+    ``(loss_objective, loss_entropy, entropy, ESS, clip_fraction,
+    loss_actor, loss_critic, loss_total)``."""
+    _require_ext()
+    a, c = actor_fused, critic_fused
+    return _ACLossFn.apply(
+        x,
+        a.lin1.weight_bf16, a.lin1.bias_bf16, a.lin2.weight_bf16,
+        a.lin2.bias_bf16, a.lin3.weight_bf16, a.lin3.bias_bf16,
+        c.lin1.weight_bf16, c.lin1.bias_bf16, c.lin2.weight_bf16,
+        c.lin2.bias_bf16, c.lin3.weight_bf16, c.lin3.bias_bf16,
+        a.lin1.weight, a.lin1.bias, a.lin2.weight, a.lin2.bias,
+        a.lin3.weight, a.lin3.bias,
+        c.lin1.weight, c.lin1.bias, c.lin2.weight, c.lin2.bias,
+        c.lin3.weight, c.lin3.bias,
+        action, prev_log_prob, advantage, value_target, eps,
+        float(sp_bias), float(scale_lb), float(lo), float(hi),
+        float(entropy_coeff), float(critic_scale), bool(normalize),
+    )
+
+
 def actor_critic_mlp3_ok(actor_fused, critic_fused, in_features: int) -> bool:
     """Shape eligibility for the dual-network kernels."""
     if not HAS_HIP_EXT:

@@ -1012,3 +1012,81 @@ def test_fused_rollout_mfma_vs_reference():
         # env transition is fp32 on the STORED action: tight tolerance
         nobs_ref = torch.tanh(obs @ env.A + st["act"][:, t] @ env.B)
         assert torch.allclose(st["nobs"][:, t], nobs_ref, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_merged_acloss_matches_split_path():
+    """The whole-minibatch merged Function (acloss kernels) vs the
+    split fwd2 + head-loss path: all 8 scalars and every master-weight
+    gradient."""
+    import os
+
+    from rl_amd import ops
+    from rl_amd.modules import MLP, NormalParamExtractor, ProbabilisticActor, TanhNormal, ValueOperator
+    from rl_amd.objectives import ClipPPOLoss
+    from rl_amd.data.tensor_specs import Bounded
+    from rl_amd.tensordict import TensorDict, TensorDictModule
+    from rl_amd.ops import convert_linears_to_splitk, enable_splitk_bf16_cache, fuse_mlp3
+
+    def build(seed):
+        torch.manual_seed(seed)
+        dev = "cuda"
+        actor = ProbabilisticActor(
+            TensorDictModule(
+                torch.nn.Sequential(
+                    MLP(in_features=17, out_features=12, num_cells=[64, 64], device=dev),
+                    NormalParamExtractor(),
+                ),
+                in_keys=["observation"], out_keys=["loc", "scale"],
+            ),
+            in_keys=["loc", "scale"], distribution_class=TanhNormal,
+            return_log_prob=True,
+            spec=Bounded(-1.0, 1.0, shape=(6,), device=dev),
+        )
+        critic = ValueOperator(
+            MLP(in_features=17, out_features=1, num_cells=[64, 64], device=dev),
+            in_keys=["observation"],
+        )
+        convert_linears_to_splitk(actor)
+        convert_linears_to_splitk(critic)
+        enable_splitk_bf16_cache(actor)
+        enable_splitk_bf16_cache(critic)
+        actor.module[0].module[0] = fuse_mlp3(actor.module[0].module[0])
+        critic.module = fuse_mlp3(critic.module)
+        return ClipPPOLoss(actor, critic, normalize_advantage=True,
+                           critic_coeff=0.5).to(dev)
+
+    N = 2048
+    torch.manual_seed(7)
+    td = TensorDict(
+        {
+            "observation": torch.randn(N, 17, device="cuda"),
+            "action": torch.rand(N, 6, device="cuda") * 1.6 - 0.8,
+            "sample_log_prob": torch.randn(N, device="cuda") * 0.1,
+            "advantage": torch.randn(N, 1, device="cuda"),
+            "value_target": torch.randn(N, 1, device="cuda"),
+        },
+        batch_size=[N],
+    )
+    outs, grads = {}, {}
+    for mode in ("1", "0"):
+        os.environ["RL_AMD_MERGED_LOSS"] = mode
+        loss_mod = build(3)
+        torch.manual_seed(11)  # same eps draw
+        out = loss_mod(td.clone())
+        total = out.get("_loss_total", None)
+        if total is None:
+            total = (out.get("loss_objective") + out.get("loss_entropy")
+                     + out.get("loss_critic"))
+        g = torch.autograd.grad(total, list(loss_mod.parameters()))
+        outs[mode] = {k: out.get(k).detach() for k in
+                      ("loss_objective", "loss_entropy", "loss_critic",
+                       "ESS", "clip_fraction", "entropy")}
+        grads[mode] = g
+    os.environ.pop("RL_AMD_MERGED_LOSS", None)
+    for k in outs["1"]:
+        assert torch.allclose(outs["1"][k], outs["0"][k], atol=2e-4, rtol=1e-3), (
+            k, outs["1"][k], outs["0"][k]
+        )
+    for g1, g0 in zip(grads["1"], grads["0"]):
+        assert torch.allclose(g1, g0, atol=2e-3, rtol=1e-2), (g1 - g0).abs().max()
