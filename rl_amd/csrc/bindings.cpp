@@ -71,6 +71,7 @@ void launch_tanh_normal_entropy_bwd(const float*, const float*, const float*,
                                     const float*, float*, float*, int, int,
                                     void*);
 void launch_adv_stats(const float*, float*, float*, long, void*);
+void launch_adv_stats_batch(const float*, float*, float*, long, int, void*);
 void launch_ppo_clip_fwd(const float*, const float*, const float*, float*,
                          float*, float*, float*, float, float, long, void*);
 void launch_ppo_clip_bwd(const float*, const float*, const float*,
@@ -1078,14 +1079,30 @@ std::vector<torch::Tensor> mlp3_mfma_fwdpair(torch::Tensor x0,
   return {out0, out1};
 }
 
+// Per-minibatch advantage mean/std in one launch pair (grid.y = slice)
+torch::Tensor adv_stats_batch(torch::Tensor adv, long n_mb) {
+  TORCH_CHECK(adv.is_cuda() && adv.is_contiguous() &&
+                  adv.scalar_type() == torch::kFloat32,
+              "adv fp32 cuda contiguous");
+  const long n = adv.numel();
+  TORCH_CHECK(n % n_mb == 0, "adv not divisible into minibatches");
+  auto part = torch::empty({n_mb * 512}, adv.options());
+  auto stats = torch::empty({n_mb, 2}, adv.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_adv_stats_batch(adv.data_ptr<float>(), part.data_ptr<float>(),
+                         stats.data_ptr<float>(), n / n_mb, (int)n_mb,
+                         (void*)stream);
+  return stats;
+}
+
 // Whole-minibatch fused forward: actor+critic MLPs AND every PPO loss
 // scalar in one launch + one finalize (csrc/fused_mlp.hip).
 std::vector<torch::Tensor> acloss_fwd(
     torch::Tensor x, std::vector<torch::Tensor> aw,
     std::vector<torch::Tensor> cw, torch::Tensor action, torch::Tensor eps,
     torch::Tensor prev, torch::Tensor adv, torch::Tensor vtarget,
-    double sp_bias, double lb, double lo, double hi, double ent_coeff,
-    double crit_scale, bool normalize) {
+    torch::Tensor stats_in, double sp_bias, double lb, double lo, double hi,
+    double ent_coeff, double crit_scale, bool normalize) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x cuda contiguous");
   const bool xf32 = x.scalar_type() == torch::kFloat32;
   const long N = x.size(0), O = x.size(1);
@@ -1127,9 +1144,13 @@ std::vector<torch::Tensor> acloss_fwd(
     scalars.push_back(torch::empty({}, fopt));
     outp[i] = scalars.back().data_ptr<float>();
   }
-  auto stats = normalize ? torch::empty({2}, fopt) : torch::empty({0}, fopt);
+  const bool ext_stats = stats_in.numel() == 2;
+  auto stats = ext_stats
+                   ? stats_in
+                   : (normalize ? torch::empty({2}, fopt)
+                                : torch::empty({0}, fopt));
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  if (normalize)
+  if (normalize && !ext_stats)
     launch_adv_stats(adv.data_ptr<float>(), part.data_ptr<float>(),
                      stats.data_ptr<float>(), N, (void*)stream);
   launch_mlp3_mfma_fwd2_loss(
@@ -1348,6 +1369,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dual-network MFMA dgrad chain (HIP)");
   m.def("mlp3_mfma_fwdpair", &mlp3_mfma_fwdpair,
         "same-network pair forward (HIP)");
+  m.def("adv_stats_batch", &adv_stats_batch,
+        "per-minibatch advantage mean/std, one launch pair (HIP)");
   m.def("acloss_fwd", &acloss_fwd,
         "actor+critic MLPs + every PPO loss scalar in one launch (HIP)");
   m.def("acloss_bwd", &acloss_bwd,

@@ -237,6 +237,52 @@ __global__ void adv_stats_finalize_k(const float* __restrict__ part,
   }
 }
 
+// per-minibatch advantage stats in ONE launch pair: grid.y = slice
+__global__ void adv_stats_partials_mb_k(const float* __restrict__ adv,
+                                        float* __restrict__ part,
+                                        const long chunk) {
+  adv += (size_t)blockIdx.y * chunk;
+  part += (size_t)blockIdx.y * 512;
+  float s = 0.f, ss = 0.f;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < chunk;
+       i += (long)gridDim.x * blockDim.x) {
+    const float a = adv[i];
+    s += a;
+    ss += a * a;
+  }
+  __shared__ float smem[8];
+  const float ts = block_sum(s, smem);
+  __syncthreads();
+  const float tss = block_sum(ss, smem);
+  if (threadIdx.x == 0) {
+    part[blockIdx.x * 2] = ts;
+    part[blockIdx.x * 2 + 1] = tss;
+  }
+}
+
+__global__ void adv_stats_finalize_mb_k(const float* __restrict__ part,
+                                        const int nwg, const long chunk,
+                                        float* __restrict__ stats) {
+  part += (size_t)blockIdx.y * 512;
+  stats += (size_t)blockIdx.y * 2;
+  float s = 0.f, ss = 0.f;
+  for (int i = threadIdx.x; i < nwg; i += blockDim.x) {
+    s += part[2 * i];
+    ss += part[2 * i + 1];
+  }
+  __shared__ float smem[8];
+  const float ts = block_sum(s, smem);
+  __syncthreads();
+  const float tss = block_sum(ss, smem);
+  if (threadIdx.x == 0) {
+    const float mu = ts / (float)chunk;
+    const float var =
+        fmaxf((tss - ts * ts / (float)chunk) / (float)(chunk - 1), 0.f);
+    stats[0] = mu;
+    stats[1] = 1.0f / fmaxf(sqrtf(var), 1e-6f);
+  }
+}
+
 __global__ void ppo_clip_fwd_partials_k(const float* __restrict__ lw,
                                         const float* __restrict__ adv,
                                         const float* __restrict__ stats,
@@ -381,6 +427,18 @@ extern "C" void launch_adv_stats(const float* adv, float* part, float* stats,
                      (hipStream_t)stream, adv, part, N);
   hipLaunchKernelGGL(adv_stats_finalize_k, dim3(1), dim3(LP_THREADS), 0,
                      (hipStream_t)stream, part, blocks, N, stats);
+}
+
+extern "C" void launch_adv_stats_batch(const float* adv, float* part,
+                                       float* stats, long chunk, int n_mb,
+                                       void* stream) {
+  const int blocks = red_blocks(chunk);
+  hipLaunchKernelGGL(adv_stats_partials_mb_k, dim3(blocks, n_mb),
+                     dim3(LP_THREADS), 0, (hipStream_t)stream, adv, part,
+                     chunk);
+  hipLaunchKernelGGL(adv_stats_finalize_mb_k, dim3(1, n_mb),
+                     dim3(LP_THREADS), 0, (hipStream_t)stream, part, blocks,
+                     chunk, stats);
 }
 
 extern "C" void launch_ppo_clip_fwd(const float* lw, const float* adv,

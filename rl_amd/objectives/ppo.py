@@ -380,6 +380,10 @@ class ClipPPOLoss(PPOLoss):
                         entropy_coeff=float(self.entropy_coeff),
                         critic_scale=float(self.critic_coeff),
                         normalize=normalize,
+                        # trainer-precomputed per-minibatch stats (one
+                        # batched launch per epoch instead of two per
+                        # minibatch) — see GraphedPPO._update_phase
+                        stats_in=self.__dict__.get("_mega_stats"),
                     )
                     return (loss_obj, loss_ent, ent_mean, ess, clip_frac,
                             loss_act, loss_crit, loss_total)

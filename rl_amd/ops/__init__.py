@@ -1090,7 +1090,7 @@ class _ACLossFn(torch.autograd.Function):
                 cw1b, cb1b, cw2b, cb2b, cw3b, cb3b,
                 aw1, ab1, aw2, ab2, aw3, ab3,
                 cw1, cb1, cw2, cb2, cw3, cb3,
-                action, prev_lp, adv, vtarget, eps,
+                action, prev_lp, adv, vtarget, eps, stats_in,
                 sp_bias, lb, lo, hi, ent_coeff, crit_scale, normalize):
         ctx.set_materialize_grads(False)
         x = x.contiguous()
@@ -1099,12 +1099,14 @@ class _ACLossFn(torch.autograd.Function):
         adv = adv.contiguous().detach()
         vtarget = vtarget.contiguous().detach()
         eps = eps.contiguous()
+        if stats_in is None:
+            stats_in = x.new_empty(0, dtype=torch.float32)
         (head, a_h1, a_h2, value, c_h1, c_h2, xb,
          loss_obj, ess, cf, ent, lent, lact, lcrit, ltotal,
          stats) = _C.acloss_fwd(
             x, [aw1b, ab1b, aw2b, ab2b, aw3b, ab3b],
             [cw1b, cb1b, cw2b, cb2b, cw3b, cb3b],
-            action, eps, prev_lp, adv, vtarget,
+            action, eps, prev_lp, adv, vtarget, stats_in.contiguous(),
             sp_bias, lb, lo, hi, ent_coeff, crit_scale, normalize,
         )
         ctx.save_for_backward(head, value, a_h1, a_h2, aw2b, aw3b, c_h1,
@@ -1135,12 +1137,13 @@ class _ACLossFn(torch.autograd.Function):
         return ((None,) * 13
                 + (adw1, adb1, adw2, adb2, adw3, adb3,
                    cdw1, cdb1, cdw2, cdb2, cdw3, cdb3)
-                + (None,) * 12)
+                + (None,) * 13)
 
 
 def actor_critic_loss(x, actor_fused, critic_fused, action, prev_log_prob,
                       advantage, value_target, eps, *, sp_bias, scale_lb,
-                      lo, hi, entropy_coeff, critic_scale, normalize):
+                      lo, hi, entropy_coeff, critic_scale, normalize,
+                      stats_in=None):
     """Run the fully-merged actor+critic+loss Function.  Returns
     ``(loss_objective, loss_entropy, entropy, ESS, clip_fraction,
     loss_actor, loss_critic, loss_total)``."""
@@ -1156,10 +1159,17 @@ def actor_critic_loss(x, actor_fused, critic_fused, action, prev_log_prob,
         a.lin3.weight, a.lin3.bias,
         c.lin1.weight, c.lin1.bias, c.lin2.weight, c.lin2.bias,
         c.lin3.weight, c.lin3.bias,
-        action, prev_log_prob, advantage, value_target, eps,
+        action, prev_log_prob, advantage, value_target, eps, stats_in,
         float(sp_bias), float(scale_lb), float(lo), float(hi),
         float(entropy_coeff), float(critic_scale), bool(normalize),
     )
+
+
+def adv_stats_batch(adv_flat: torch.Tensor, n_mb: int) -> torch.Tensor:
+    """Per-minibatch advantage (mean, 1/std) for ``n_mb`` contiguous
+    slices of ``adv_flat`` in ONE launch pair.  Returns [n_mb, 2]."""
+    _require_ext()
+    return _C.adv_stats_batch(adv_flat.contiguous(), int(n_mb))
 
 
 def actor_critic_mlp3_ok(actor_fused, critic_fused, in_features: int) -> bool:

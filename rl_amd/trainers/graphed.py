@@ -317,8 +317,33 @@ class GraphedPPO:
                 shuffled = ops.multi_gather_td(flat, perm) if on_gpu else None
                 if shuffled is None:
                     shuffled = flat[perm]
+            # one batched launch computes every minibatch's advantage
+            # normalization stats (the loss would otherwise launch a
+            # stats pair per minibatch).  Not in mb-graph mode: the
+            # capture would bake the first epoch's stats addresses.
+            stats_all = None
+            if (
+                on_gpu
+                and not self._distributed
+                and getattr(self.loss_module, "normalize_advantage", False)
+                and ops.HAS_HIP_EXT
+                and n % self.minibatches == 0
+            ):
+                try:
+                    advk = self.loss_module.tensor_keys.advantage
+                    adv_t = shuffled.get(advk, None)
+                    if adv_t is not None and adv_t.dtype == torch.float32:
+                        stats_all = ops.adv_stats_batch(
+                            adv_t.reshape(-1), self.minibatches
+                        )
+                except Exception:
+                    stats_all = None
             for i in range(self.minibatches):
+                if stats_all is not None:
+                    self.loss_module.__dict__["_mega_stats"] = stats_all[i]
                 self._run_minibatch(shuffled[i * mb : (i + 1) * mb])
+            if stats_all is not None:
+                self.loss_module.__dict__.pop("_mega_stats", None)
 
     def _one_iter_inline(self) -> None:
         batch = self.collector.rollout_inline()
