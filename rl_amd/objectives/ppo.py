@@ -366,8 +366,10 @@ class ClipPPOLoss(PPOLoss):
                 ):
                     # the fully-merged path: MLPs AND loss in one
                     # launch pair (csrc/fused_mlp.hip acloss kernels)
-                    eps = torch.randn(N, A, device=action.device,
-                                      dtype=torch.float32)
+                    eps = self.__dict__.get("_mega_eps")
+                    if eps is None or eps.shape != (N, A):
+                        eps = torch.randn(N, A, device=action.device,
+                                          dtype=torch.float32)
                     lo, hi = self._clip_bounds
                     (loss_obj, loss_ent, ent_mean, ess, clip_frac,
                      loss_act, loss_crit, loss_total) = ops.actor_critic_loss(

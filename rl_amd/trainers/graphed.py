@@ -322,6 +322,7 @@ class GraphedPPO:
             # stats pair per minibatch).  Not in mb-graph mode: the
             # capture would bake the first epoch's stats addresses.
             stats_all = None
+            eps_all = None
             if (
                 on_gpu
                 and not self._distributed
@@ -336,14 +337,26 @@ class GraphedPPO:
                         stats_all = ops.adv_stats_batch(
                             adv_t.reshape(-1), self.minibatches
                         )
+                    # one philox draw for every minibatch's entropy eps
+                    ak = self.loss_module.tensor_keys.action
+                    act_t = shuffled.get(ak, None)
+                    if act_t is not None:
+                        eps_all = torch.randn(
+                            self.minibatches, mb, act_t.shape[-1],
+                            device=device, dtype=torch.float32,
+                        )
                 except Exception:
-                    stats_all = None
+                    stats_all = eps_all = None
             for i in range(self.minibatches):
                 if stats_all is not None:
                     self.loss_module.__dict__["_mega_stats"] = stats_all[i]
+                if eps_all is not None:
+                    self.loss_module.__dict__["_mega_eps"] = eps_all[i]
                 self._run_minibatch(shuffled[i * mb : (i + 1) * mb])
             if stats_all is not None:
                 self.loss_module.__dict__.pop("_mega_stats", None)
+            if eps_all is not None:
+                self.loss_module.__dict__.pop("_mega_eps", None)
 
     def _one_iter_inline(self) -> None:
         batch = self.collector.rollout_inline()
