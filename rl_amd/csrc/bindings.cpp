@@ -100,7 +100,9 @@ void launch_multi_gather(const void*, int, const long*, long, void*);
 void launch_multi_shuffle(const void*, int, const int*, long, void*);
 void launch_wgrad3(const void* const*, const void* const*, float* const*,
                    float* const*, float* const*, float* const*, const int*,
-                   const int*, int, long, void*);
+                   const int*, int, long, float*, void*);
+void launch_wgrad_clip_finalize(const float*, int, float, float*, int,
+                                void*);
 void launch_mlp3_mfma_fwd2(const void*, int, const void* const*,
                            void* const*, void*, long, int, const int*,
                            const int*, void*);
@@ -867,8 +869,9 @@ std::vector<torch::Tensor> multi_shuffle(torch::Tensor keys,
 // and a shared K are required (the PPO/critic MLP shape).
 bool mlp3_mfma_ok(long O, long H, long A2);
 
-std::vector<torch::Tensor> wgrad_splitk_batch(
-    std::vector<torch::Tensor> dys_v, std::vector<torch::Tensor> xs_v) {
+std::vector<torch::Tensor> wgrad_splitk_batch_sq(
+    std::vector<torch::Tensor> dys_v, std::vector<torch::Tensor> xs_v,
+    torch::Tensor sq_part) {
   const int n_layers = (int)dys_v.size();
   TORCH_CHECK(n_layers >= 1 && n_layers <= 6 &&
                   xs_v.size() == dys_v.size(),
@@ -911,8 +914,25 @@ std::vector<torch::Tensor> wgrad_splitk_batch(
   }
   auto stream = c10::hip::getCurrentHIPStream().stream();
   launch_wgrad3(dyp, xp, partp, biasp, dwp, dbp, N, M, n_layers, K,
+                sq_part.numel() ? sq_part.data_ptr<float>() : nullptr,
                 (void*)stream);
   return out;
+}
+
+std::vector<torch::Tensor> wgrad_splitk_batch(
+    std::vector<torch::Tensor> dys_v, std::vector<torch::Tensor> xs_v) {
+  return wgrad_splitk_batch_sq(dys_v, xs_v, torch::Tensor());
+}
+
+// grad-clip coefficient from the wgrad reduce's [n_layers*512] sq
+// partials: call after wgrad_splitk_batch_sq covered EVERY gradient.
+void wgrad_clip_finalize(torch::Tensor sq_part, double max_norm,
+                         torch::Tensor coef, bool inverse) {
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  launch_wgrad_clip_finalize(sq_part.data_ptr<float>(),
+                             (int)sq_part.numel(), (float)max_norm,
+                             coef.data_ptr<float>(), inverse ? 1 : 0,
+                             (void*)stream);
 }
 
 std::vector<torch::Tensor> wgrad_splitk3(torch::Tensor dy0, torch::Tensor x0,
@@ -1363,6 +1383,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "batched 3-layer split-K wgrad (HIP)");
   m.def("wgrad_splitk_batch", &wgrad_splitk_batch,
         "batched 1..6-layer split-K wgrad (HIP)");
+  m.def("wgrad_splitk_batch_sq", &wgrad_splitk_batch_sq,
+        "batched wgrad emitting grad-sumsq partials (HIP)");
+  m.def("wgrad_clip_finalize", &wgrad_clip_finalize,
+        "clip coefficient from wgrad sq partials (HIP)");
   m.def("mlp3_mfma_fwd2", &mlp3_mfma_fwd2,
         "dual-network MFMA MLP forward (HIP)");
   m.def("mlp3_mfma_bwd2", &mlp3_mfma_bwd2,

@@ -519,7 +519,8 @@ __global__ void __launch_bounds__(WG_THREADS) wgrad3_mfma_kernel(
   if (lr == 0) args.bias_part[l][(long)blockIdx.x * 64 + lc] = s_bias[lc];
 }
 
-__global__ void wgrad3_reduce_kernel(const Wg3Args args, int slabs) {
+__global__ void wgrad3_reduce_kernel(const Wg3Args args, int slabs,
+                                     float* __restrict__ sq_part) {
   const int l = blockIdx.y;
   const int N = args.N[l], M = args.M[l];
   const float* __restrict__ part = args.part[l];
@@ -527,6 +528,7 @@ __global__ void wgrad3_reduce_kernel(const Wg3Args args, int slabs) {
   const long waves = ((long)gridDim.x * blockDim.x) >> 6;
   const long wave_id = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int lane = threadIdx.x & 63;
+  float sq = 0.f;  // this block's sum of squared gradient elements
   for (long e = wave_id; e < NM; e += waves) {
     const int n = (int)(e / M), m = (int)(e % M);
     const long off = (long)n * 64 + m;
@@ -535,7 +537,10 @@ __global__ void wgrad3_reduce_kernel(const Wg3Args args, int slabs) {
       acc += part[(long)s * (64 * 64) + off];
 #pragma unroll
     for (int d = 32; d > 0; d >>= 1) acc += __shfl_down(acc, d);
-    if (lane == 0) args.dw[l][e] = acc;
+    if (lane == 0) {
+      args.dw[l][e] = acc;
+      sq += acc * acc;
+    }
   }
   for (long n = wave_id; n < N; n += waves) {
     float acc = 0.f;
@@ -543,7 +548,50 @@ __global__ void wgrad3_reduce_kernel(const Wg3Args args, int slabs) {
       acc += args.bias_part[l][(long)s * 64 + n];
 #pragma unroll
     for (int d = 32; d > 0; d >>= 1) acc += __shfl_down(acc, d);
-    if (lane == 0) args.db[l][n] = acc;
+    if (lane == 0) {
+      args.db[l][n] = acc;
+      sq += acc * acc;
+    }
+  }
+  if (sq_part != nullptr) {
+    // block-reduce the per-thread sq partials (nonzero at lane 0s)
+    __shared__ float smem[8];
+    for (int off = 32; off > 0; off >>= 1) sq += __shfl_down(sq, off, 64);
+    const int wv = threadIdx.x >> 6;
+    if (lane == 0) smem[wv] = sq;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float t = 0.f;
+      for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += smem[w];
+      sq_part[(long)blockIdx.y * gridDim.x + blockIdx.x] = t;
+    }
+  }
+}
+
+// grad-clip coefficient straight from the wgrad reduce's sq partials:
+// ONE tiny kernel instead of re-reading every gradient tensor.
+__global__ void wgrad_clip_finalize_k(const float* __restrict__ sq_part,
+                                      const int n_part,
+                                      const float max_norm,
+                                      float* __restrict__ coef,
+                                      const int inverse) {
+  float s = 0.f;
+  for (int i = threadIdx.x; i < n_part; i += blockDim.x) s += sq_part[i];
+  __shared__ float smem[8];
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+  const int lane = threadIdx.x & 63, wv = threadIdx.x >> 6;
+  if (lane == 0) smem[wv] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += smem[w];
+    if (inverse) {
+      const float c = (sqrtf(t) + 1e-6f) / max_norm;
+      coef[0] = c > 1.f ? c : 1.f;
+    } else {
+      const float c = max_norm / (sqrtf(t) + 1e-6f);
+      coef[0] = c < 1.f ? c : 1.f;
+    }
   }
 }
 
@@ -566,7 +614,7 @@ extern "C" void launch_wgrad3(const void* const* dy, const void* const* x,
                               float* const* part, float* const* bias_part,
                               float* const* dw, float* const* db,
                               const int* N, const int* M, int n_layers,
-                              long K, void* stream) {
+                              long K, float* sq_part, void* stream) {
   Wg3Args a;
   for (int l = 0; l < n_layers; ++l) {
     a.dy[l] = (const __hip_bfloat16*)dy[l];
@@ -589,5 +637,13 @@ extern "C" void launch_wgrad3(const void* const* dy, const void* const* x,
                      (int)k_slab);
   const int blocks = 512;  // n * 512 WGs cover the chip for the reduce
   hipLaunchKernelGGL(wgrad3_reduce_kernel, dim3(blocks, n_layers), dim3(256),
-                     0, (hipStream_t)stream, *args, slabs);
+                     0, (hipStream_t)stream, *args, slabs, sq_part);
+}
+
+extern "C" void launch_wgrad_clip_finalize(const float* sq_part, int n_part,
+                                           float max_norm, float* coef,
+                                           int inverse, void* stream) {
+  hipLaunchKernelGGL(wgrad_clip_finalize_k, dim3(1), dim3(256), 0,
+                     (hipStream_t)stream, sq_part, n_part, max_norm, coef,
+                     inverse);
 }

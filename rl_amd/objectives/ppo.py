@@ -386,7 +386,12 @@ class ClipPPOLoss(PPOLoss):
                         # batched launch per epoch instead of two per
                         # minibatch) — see GraphedPPO._update_phase
                         stats_in=self.__dict__.get("_mega_stats"),
+                        gradsq_out=self.__dict__.get("_gradsq_part"),
                     )
+                    if self.__dict__.get("_gradsq_part") is not None:
+                        # signals the trainer that this step's sq
+                        # partials WILL be written by the backward
+                        self.__dict__["_gradsq_armed"] = True
                     return (loss_obj, loss_ent, ent_mean, ess, clip_frac,
                             loss_act, loss_crit, loss_total)
                 head, value = ops.actor_critic_mlp3(

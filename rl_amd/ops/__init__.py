@@ -1090,7 +1090,7 @@ class _ACLossFn(torch.autograd.Function):
                 cw1b, cb1b, cw2b, cb2b, cw3b, cb3b,
                 aw1, ab1, aw2, ab2, aw3, ab3,
                 cw1, cb1, cw2, cb2, cw3, cb3,
-                action, prev_lp, adv, vtarget, eps, stats_in,
+                action, prev_lp, adv, vtarget, eps, stats_in, gradsq_out,
                 sp_bias, lb, lo, hi, ent_coeff, crit_scale, normalize):
         ctx.set_materialize_grads(False)
         x = x.contiguous()
@@ -1112,6 +1112,7 @@ class _ACLossFn(torch.autograd.Function):
         ctx.save_for_backward(head, value, a_h1, a_h2, aw2b, aw3b, c_h1,
                               c_h2, cw2b, cw3b, action, eps, prev_lp, adv,
                               stats, vtarget, xb)
+        ctx.gradsq_out = gradsq_out
         ctx.cfg = (sp_bias, lb, lo, hi, ent_coeff, crit_scale)
         ctx.mark_non_differentiable(ess, cf, ent)
         return loss_obj, lent, ent, ess, cf, lact, lcrit, ltotal
@@ -1129,21 +1130,26 @@ class _ACLossFn(torch.autograd.Function):
             g(g_obj), g(g_ent), g(g_act), g(g_crit), g(g_tot),
             sp_bias, lb, lo, hi, ent_coeff, crit_scale,
         )
-        (adw3, adb3, adw2, adb2, adw1, adb1, cdw3, cdb3, cdw2, cdb2, cdw1,
-         cdb1) = _C.wgrad_splitk_batch(
-            [dhead, a_dh2, a_dh1, dvalue, c_dh2, c_dh1],
-            [a_h2, a_h1, xb, c_h2, c_h1, xb],
-        )
+        dys = [dhead, a_dh2, a_dh1, dvalue, c_dh2, c_dh1]
+        xs = [a_h2, a_h1, xb, c_h2, c_h1, xb]
+        if ctx.gradsq_out is not None:
+            # grad-sumsq partials ride in the reduce: the trainer turns
+            # them into the clip coefficient with one tiny kernel
+            (adw3, adb3, adw2, adb2, adw1, adb1, cdw3, cdb3, cdw2, cdb2,
+             cdw1, cdb1) = _C.wgrad_splitk_batch_sq(dys, xs, ctx.gradsq_out)
+        else:
+            (adw3, adb3, adw2, adb2, adw1, adb1, cdw3, cdb3, cdw2, cdb2,
+             cdw1, cdb1) = _C.wgrad_splitk_batch(dys, xs)
         return ((None,) * 13
                 + (adw1, adb1, adw2, adb2, adw3, adb3,
                    cdw1, cdb1, cdw2, cdb2, cdw3, cdb3)
-                + (None,) * 13)
+                + (None,) * 14)
 
 
 def actor_critic_loss(x, actor_fused, critic_fused, action, prev_log_prob,
                       advantage, value_target, eps, *, sp_bias, scale_lb,
                       lo, hi, entropy_coeff, critic_scale, normalize,
-                      stats_in=None):
+                      stats_in=None, gradsq_out=None):
     """Run the fully-merged actor+critic+loss Function.  Returns
     ``(loss_objective, loss_entropy, entropy, ESS, clip_fraction,
     loss_actor, loss_critic, loss_total)``."""
@@ -1160,6 +1166,7 @@ def actor_critic_loss(x, actor_fused, critic_fused, action, prev_log_prob,
         c.lin1.weight, c.lin1.bias, c.lin2.weight, c.lin2.bias,
         c.lin3.weight, c.lin3.bias,
         action, prev_log_prob, advantage, value_target, eps, stats_in,
+        gradsq_out,
         float(sp_bias), float(scale_lb), float(lo), float(hi),
         float(entropy_coeff), float(critic_scale), bool(normalize),
     )

@@ -118,6 +118,7 @@ class GraphedPPO:
 
         self._seed_one = None
         self._clip_scale = None
+        self._gradsq_part = None
         if (
             self._cuda
             and max_grad_norm
@@ -240,7 +241,19 @@ class GraphedPPO:
         if self.max_grad_norm:
             from .. import ops
 
-            if self._clip_scale is not None:
+            if (
+                self._clip_scale is not None
+                and self.loss_module.__dict__.pop("_gradsq_armed", False)
+            ):
+                # the merged-loss backward already left grad-sumsq
+                # partials in _gradsq_part: one tiny finalize kernel
+                from .. import _C
+
+                _C.wgrad_clip_finalize(
+                    self._gradsq_part, self.max_grad_norm,
+                    self._clip_scale, True,
+                )
+            elif self._clip_scale is not None:
                 # the clip coefficient rides into the fused Adam as its
                 # grad_scale divisor: no gradient multiply at all
                 ops.fused_grad_clip_scale_(
@@ -336,6 +349,21 @@ class GraphedPPO:
                     if adv_t is not None and adv_t.dtype == torch.float32:
                         stats_all = ops.adv_stats_batch(
                             adv_t.reshape(-1), self.minibatches
+                        )
+                    if (
+                        self._clip_scale is not None
+                        and self.reducer is None
+                        and self._gradsq_part is None
+                        # the sq partials only cover the merged-loss
+                        # Function's 12 gradients (2 nets x 3 linears)
+                        and len(self._params) == 12
+                    ):
+                        self._gradsq_part = torch.empty(
+                            6 * 512, device=device
+                        )
+                    if self._gradsq_part is not None:
+                        self.loss_module.__dict__["_gradsq_part"] = (
+                            self._gradsq_part
                         )
                     # one philox draw for every minibatch's entropy eps
                     ak = self.loss_module.tensor_keys.action

@@ -1090,3 +1090,49 @@ def test_merged_acloss_matches_split_path():
         )
     for g1, g0 in zip(grads["1"], grads["0"]):
         assert torch.allclose(g1, g0, atol=2e-3, rtol=1e-2), (g1 - g0).abs().max()
+
+
+@pytest.mark.gpu
+def test_wgrad_sq_partials_clip_coefficient():
+    """Clip coefficient from the wgrad reduce's sq partials must equal
+    the true global gradient norm's coefficient."""
+    import rl_amd._C as _C
+    from rl_amd import ops
+    from rl_amd.ops import convert_linears_to_splitk, enable_splitk_bf16_cache, fuse_mlp3
+
+    torch.manual_seed(5)
+
+    def mk(a2, seed):
+        torch.manual_seed(seed)
+        net = torch.nn.Sequential(
+            torch.nn.Linear(17, 64), torch.nn.Tanh(),
+            torch.nn.Linear(64, 64), torch.nn.Tanh(),
+            torch.nn.Linear(64, a2),
+        ).cuda()
+        convert_linears_to_splitk(net)
+        enable_splitk_bf16_cache(net)
+        return fuse_mlp3(net)
+
+    actor, critic = mk(12, 0), mk(1, 1)
+    N = 2048
+    x = torch.randn(N, 17, device="cuda")
+    action = torch.rand(N, 6, device="cuda") * 1.6 - 0.8
+    prev = torch.randn(N, device="cuda") * 0.1
+    adv = torch.randn(N, device="cuda")
+    vt = torch.randn(N, device="cuda")
+    eps = torch.randn(N, 6, device="cuda")
+    sq = torch.empty(6 * 512, device="cuda")
+    out = ops.actor_critic_loss(
+        x, actor, critic, action, prev, adv, vt, eps,
+        sp_bias=0.5413248546129181, scale_lb=1e-4,
+        lo=math.log1p(-0.2), hi=math.log1p(0.2),
+        entropy_coeff=0.01, critic_scale=0.5, normalize=True,
+        gradsq_out=sq,
+    )
+    out[7].backward()  # loss_total
+    params = list(actor.parameters()) + list(critic.parameters())
+    true_norm = torch.sqrt(sum(p.grad.float().pow(2).sum() for p in params))
+    coef = torch.empty((), device="cuda")
+    _C.wgrad_clip_finalize(sq, 1.0, coef, True)
+    expect = torch.clamp((true_norm + 1e-6) / 1.0, min=1.0)
+    assert torch.allclose(coef, expect, rtol=1e-4), (coef, expect)
