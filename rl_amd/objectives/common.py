@@ -148,19 +148,26 @@ class _EnsembleModule(nn.Module):
                              for n in range(self.num_copies)])
             )
 
-    def forward(self, td: TensorDictBase) -> TensorDictBase:
-        """Run all copies on the same input; stack outputs along dim 0."""
+    def forward(self, td: TensorDictBase, detach_params: bool = False) -> TensorDictBase:
+        """Run all copies on the same input; stack outputs along dim 0.
+
+        ``detach_params=True`` evaluates with the weights treated as
+        constants while gradients still flow through the INPUT (the
+        reference's ``params.detach()`` actor-loss semantics) — served
+        by the stacked-bf16 buffers when present, so no per-call weight
+        stack/cast happens at all."""
         from ..tensordict import stack as td_stack
 
         batched = self._try_batched_mlp()
         if batched:
             per_copy, sig, in_keys, out_keys = batched
-            # the cache is a buffer: it would SWALLOW gradients, so it
-            # only serves no-grad forwards (target-net evaluations)
+            # the cache is a buffer: it would SWALLOW weight gradients,
+            # so it serves no-grad forwards (target nets) and
+            # detach_params forwards (actor-loss q evaluations)
             cached = (
                 getattr(self, "_stack_cache", False)
                 and torch.is_autocast_enabled()
-                and not torch.is_grad_enabled()
+                and (not torch.is_grad_enabled() or detach_params)
             )
             x = torch.cat([td.get(k) for k in in_keys], dim=-1)
             lead = x.shape[:-1]
@@ -179,6 +186,9 @@ class _EnsembleModule(nn.Module):
                     else:
                         W = torch.stack([per_copy[n][li].weight for n in range(self.num_copies)])
                         bias = torch.stack([per_copy[n][li].bias for n in range(self.num_copies)])
+                        if detach_params:
+                            W = W.detach()
+                            bias = bias.detach()
                         h = torch.baddbmm(bias.unsqueeze(1), h, W.transpose(-2, -1))
                 else:
                     h = getattr(torch, kind.lower())(h) if kind == "Tanh" else getattr(torch.nn.functional, kind.lower())(h)
@@ -188,6 +198,18 @@ class _EnsembleModule(nn.Module):
             res = res.clone(False)
             res.set(out_keys[0], out)
             return res
+        if detach_params:
+            # exact per-copy path with constant weights: evaluate under
+            # a fresh graph from detached parameter clones
+            import copy as _copy
+
+            outs = []
+            for m in self.modules_list:
+                frozen = _copy.deepcopy(m)
+                for pp in frozen.parameters():
+                    pp.requires_grad_(False)
+                outs.append(frozen(td.clone(False)))
+            return td_stack(outs, 0)
         outs = [m(td.clone(False)) for m in self.modules_list]
         return td_stack(outs, 0)
 

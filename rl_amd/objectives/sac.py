@@ -127,9 +127,18 @@ class SACLoss(LossModule):
             return x.sum()
         return x
 
-    def _qvalues(self, net, td: TensorDictBase) -> torch.Tensor:
-        """[num_nets, *batch, 1] state-action values."""
-        out = net(td)
+    def _qvalues(self, net, td: TensorDictBase, detach_params: bool = False) -> torch.Tensor:
+        """[num_nets, *batch, 1] state-action values.
+
+        ``detach_params`` mirrors the reference's
+        ``qvalue_network_params.detach()`` for the actor loss: the
+        gradient flows through the ACTION, not the Q weights."""
+        from .common import _EnsembleModule
+
+        if detach_params and isinstance(net, _EnsembleModule):
+            out = net(td, detach_params=True)
+        else:
+            out = net(td)
         return out.get(self.tensor_keys.state_action_value)
 
     def _actor_loss(self, td: TensorDictBase):
@@ -138,7 +147,7 @@ class SACLoss(LossModule):
         action = dist.rsample()
         log_prob = dist.log_prob(action)
         d.set(self.tensor_keys.action, action)
-        q = self._qvalues(self.qvalue_network, d.clone(False))
+        q = self._qvalues(self.qvalue_network, d.clone(False), detach_params=True)
         min_q = q.min(0).values.squeeze(-1)
         loss = self.alpha * log_prob - min_q
         return loss, log_prob.detach()
