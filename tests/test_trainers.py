@@ -398,3 +398,72 @@ class TestAsyncCollection:
         tr.train()
         assert tr.collected_frames >= 96
         assert tr._optim_count > 0
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_graphed_ppo_multi_epoch_gpu():
+    """GraphedPPO with epochs=2 on the GPU fast path: the epoch loop
+    re-shuffles (Feistel), recomputes the batched advantage stats and
+    eps draws, and replays cleanly under full-step capture."""
+    import torch
+
+    from rl_amd.collectors import Collector
+    from rl_amd.envs.custom.synthetic import HalfCheetahVec
+    from rl_amd.modules import MLP, NormalParamExtractor, ProbabilisticActor, TanhNormal, ValueOperator
+    from rl_amd.objectives import ClipPPOLoss
+    from rl_amd.objectives.value.advantages import GAE
+    from rl_amd.ops import (
+        convert_linears_to_splitk,
+        enable_splitk_bf16_cache,
+        fuse_mlp3,
+        refresh_splitk_caches,
+    )
+    from rl_amd.tensordict import TensorDictModule
+    from rl_amd.trainers import GraphedPPO
+
+    torch.manual_seed(0)
+    dev = "cuda"
+    env = HalfCheetahVec(batch_size=[256], device=dev)
+    actor = ProbabilisticActor(
+        TensorDictModule(
+            torch.nn.Sequential(
+                MLP(in_features=17, out_features=12, num_cells=[64, 64], device=dev),
+                NormalParamExtractor(),
+            ),
+            in_keys=["observation"], out_keys=["loc", "scale"],
+        ),
+        in_keys=["loc", "scale"], distribution_class=TanhNormal,
+        return_log_prob=True,
+    )
+    critic = ValueOperator(
+        MLP(in_features=17, out_features=1, num_cells=[64, 64], device=dev),
+        in_keys=["observation"],
+    )
+    convert_linears_to_splitk(actor)
+    convert_linears_to_splitk(critic)
+    enable_splitk_bf16_cache(actor)
+    enable_splitk_bf16_cache(critic)
+    actor.module[0].module[0] = fuse_mlp3(actor.module[0].module[0])
+    critic.module = fuse_mlp3(critic.module)
+    loss = ClipPPOLoss(actor, critic, normalize_advantage=True, critic_coeff=0.5)
+    gae = GAE(gamma=0.99, lmbda=0.95, value_network=critic, vectorized=True)
+    optim = torch.optim.Adam(
+        list(actor.parameters()) + list(critic.parameters()), lr=3e-4,
+        capturable=True, fused=True,
+    )
+    col = Collector(env, actor, frames_per_batch=256 * 8, total_frames=-1)
+    runner = GraphedPPO(
+        col, gae, loss, optim, minibatches=2, epochs=2,
+        post_optim_hook=lambda: refresh_splitk_caches(actor, critic),
+    )
+    p0 = [p.detach().clone() for p in actor.parameters()]
+    for _ in range(3):
+        runner.step()
+    torch.cuda.synchronize()
+    moved = any(not torch.equal(a, b.detach())
+                for a, b in zip(p0, actor.parameters()))
+    assert moved
+    for p in actor.parameters():
+        assert torch.isfinite(p).all()
+    col.shutdown()
