@@ -1136,3 +1136,46 @@ def test_wgrad_sq_partials_clip_coefficient():
     _C.wgrad_clip_finalize(sq, 1.0, coef, True)
     expect = torch.clamp((true_norm + 1e-6) / 1.0, min=1.0)
     assert torch.allclose(coef, expect, rtol=1e-4), (coef, expect)
+
+
+@pytest.mark.gpu
+def test_multi_shuffle_is_consistent_permutation():
+    """Feistel shuffle: a real permutation of [0, n), identical across
+    tensors, fresh for fresh keys."""
+    from rl_amd import ops
+    from rl_amd.tensordict import TensorDict
+
+    for n in (4096, 5000):  # power of two + cycle-walking case
+        idx = torch.arange(n, device="cuda", dtype=torch.float32)
+        td = TensorDict(
+            {"a": idx.clone(), "b": torch.stack([idx, idx * 2], -1).contiguous()},
+            batch_size=[n],
+        )
+        keys = torch.randint(-2**31, 2**31 - 1, (4,), device="cuda", dtype=torch.int32)
+        out = ops.multi_shuffle_td(td, keys)
+        a = out.get("a")
+        # permutation property
+        assert torch.equal(torch.sort(a).values, idx)
+        assert not torch.equal(a, idx)  # actually shuffled
+        # same permutation applied to every tensor
+        assert torch.equal(out.get("b")[:, 0], a)
+        assert torch.equal(out.get("b")[:, 1], a * 2)
+        # fresh keys -> different permutation
+        keys2 = keys + 12345
+        out2 = ops.multi_shuffle_td(td, keys2)
+        assert not torch.equal(out2.get("a"), a)
+
+
+@pytest.mark.gpu
+def test_adv_stats_batch_matches_torch():
+    from rl_amd import ops
+
+    torch.manual_seed(0)
+    n_mb, mb = 4, 16384
+    adv = torch.randn(n_mb * mb, device="cuda") * 3 + 1
+    stats = ops.adv_stats_batch(adv, n_mb)
+    for i in range(n_mb):
+        sl = adv[i * mb : (i + 1) * mb]
+        assert torch.allclose(stats[i, 0], sl.mean(), atol=1e-4)
+        assert torch.allclose(stats[i, 1], 1.0 / sl.std().clamp_min(1e-6),
+                              rtol=1e-4)
