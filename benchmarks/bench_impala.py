@@ -128,7 +128,8 @@ def main():
     behavior = ImpalaNet(A, device=device)
     behavior.load_state_dict(net.state_dict())
     try:
-        optim = torch.optim.Adam(net.parameters(), lr=6e-4, fused=cuda)
+        optim = torch.optim.Adam(net.parameters(), lr=6e-4, fused=cuda,
+                                 capturable=bool(args.graph and cuda and not distributed))
     except Exception:
         optim = torch.optim.Adam(net.parameters(), lr=6e-4)
     gamma = 0.99
@@ -178,8 +179,7 @@ def main():
             traceback.print_exc()
             args.graph = False
 
-    def one_iteration():
-        rollout()
+    def learner_body():
         # learner: recompute pi under current weights over the unroll
         flat = frames.reshape(B * T, *env.frame_shape)
         with autocast:
@@ -222,6 +222,32 @@ def main():
         optim.step()
         # one foreach copy instead of load_state_dict's per-tensor loop
         torch._foreach_copy_(_b_params, _n_params)
+
+    learner = learner_body
+    if args.graph and cuda and not distributed:
+        # the learner reads only the static rollout buffers + carried
+        # obs: capture it as a second graph (the update itself is ~100s
+        # of small conv/glue launches at B*T = 10240)
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    learner_body()
+            torch.cuda.current_stream().wait_stream(side)
+            gl = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(gl):
+                learner_body()
+            learner = gl.replay
+        except Exception:
+            import traceback
+
+            traceback.print_exc()
+            learner = learner_body
+
+    def one_iteration():
+        rollout()
+        learner()
 
     for _ in range(args.warmup):
         one_iteration()
