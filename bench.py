@@ -282,6 +282,14 @@ def run_ppo(args):
                 "full_step_graph": runner.full_graph,
                 "minibatch_graph": runner.minibatch_graph,
                 "overlapped_comm": bool(runner.reducer is not None),
+                "merged_loss_kernels": bool(
+                    cuda
+                    and os.environ.get("RL_AMD_MERGED_LOSS", "1") != "0"
+                ),
+                "rollout_mfma": bool(
+                    cuda
+                    and os.environ.get("RL_AMD_ROLLOUT_MFMA", "1") != "0"
+                ),
             },
         }
         print(json.dumps(result))
