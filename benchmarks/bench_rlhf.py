@@ -49,8 +49,10 @@ def main():
     p.add_argument("--batch", type=int, default=256, help="prompts per rank")
     p.add_argument("--prompt-len", type=int, default=64)
     p.add_argument("--gen-len", type=int, default=32)
-    p.add_argument("--decode-mode", choices=["generate", "static", "graph"],
-                   default="generate",
+    p.add_argument("--decode-mode",
+                   choices=["generate", "manual", "graphdec", "static",
+                            "graph"],
+                   default="graphdec",
                    help="HF generate (default) / eager static-KV loop / "
                         "hipGraph-captured loop.  static and graph SEGFAULT on "
                         "this stack (ROCm 7.2 + transformers 5.15 GPT2 "
@@ -159,10 +161,46 @@ def main():
                 traceback.print_exc()
                 decode_graph = None
 
+    graphdec = None
+    if args.decode_mode == "graphdec" and cuda:
+        from rl_amd.modules.llm.decode import GraphedGPT2Decoder
+
+        graphdec = GraphedGPT2Decoder(policy, B, P + G, device)
+        # warm + capture once (needs a prefilled state)
+        graphdec.prefill(torch.randint(0, V, (B, P), device=device))
+        if not graphdec.capture():
+            print("graphdec capture failed; eager decode body stays",
+                  file=sys.stderr)
+
+    def manual_rollout(prompts):
+        """Hand-rolled sampling loop over the ordinary dynamic KV cache
+        (the same cache `generate` uses — StaticCache segfaults on this
+        stack): one prefill + G-1 single-token forwards with none of
+        generate's ~6 ms/token host-side logic.  Identical sampling
+        distribution (do_sample, top_k=0)."""
+        with torch.no_grad(), autocast:
+            out = policy(input_ids=prompts, use_cache=True)
+            past = out.past_key_values
+            logits = out.logits[:, -1].float()
+            toks = []
+            for i in range(G):
+                nxt = torch.multinomial(torch.softmax(logits, -1), 1)
+                toks.append(nxt)
+                if i + 1 < G:
+                    out = policy(input_ids=nxt, past_key_values=past,
+                                 use_cache=True)
+                    past = out.past_key_values
+                    logits = out.logits[:, -1].float()
+        return torch.cat([prompts] + toks, 1)
+
     def one_iteration():
         prompts = torch.randint(0, V, (B, P), device=device)
         mask = torch.ones_like(prompts)
-        if use_static:
+        if graphdec is not None:
+            gen = graphdec.rollout(prompts, G)
+        elif args.decode_mode == "manual":
+            gen = manual_rollout(prompts)
+        elif use_static:
             with torch.no_grad():
                 prefill(prompts)
                 if decode_graph is not None:
@@ -216,6 +254,8 @@ def main():
                 g.copy_(flat[i : i + g.numel()].reshape(g.shape))
                 i += g.numel()
         optim.step()
+        if graphdec is not None:
+            graphdec.refresh_weights()
 
     for _ in range(args.warmup):
         one_iteration()
