@@ -1179,3 +1179,48 @@ def test_adv_stats_batch_matches_torch():
         assert torch.allclose(stats[i, 0], sl.mean(), atol=1e-4)
         assert torch.allclose(stats[i, 1], 1.0 / sl.std().clamp_min(1e-6),
                               rtol=1e-4)
+
+
+@pytest.mark.gpu
+def test_gae_paired_value_eval_matches_eager_critic():
+    """GAE through a cached FusedMLP3 critic (ONE paired launch for
+    value/next_value) must match GAE through the identical eager
+    critic."""
+    import copy
+
+    from rl_amd.modules import MLP, ValueOperator
+    from rl_amd.objectives.value.advantages import GAE
+    from rl_amd.ops import convert_linears_to_splitk, enable_splitk_bf16_cache, fuse_mlp3
+    from rl_amd.tensordict import TensorDict
+
+    torch.manual_seed(0)
+    dev = "cuda"
+    critic = ValueOperator(
+        MLP(in_features=17, out_features=1, num_cells=[64, 64], device=dev),
+        in_keys=["observation"],
+    )
+    critic_ref = copy.deepcopy(critic)
+    convert_linears_to_splitk(critic)
+    enable_splitk_bf16_cache(critic)
+    critic.module = fuse_mlp3(critic.module)
+    B, T = 32, 16
+    td = TensorDict(
+        {
+            "observation": torch.randn(B, T, 17, device=dev),
+            "next": {
+                "observation": torch.randn(B, T, 17, device=dev),
+                "reward": torch.randn(B, T, 1, device=dev),
+                "done": torch.rand(B, T, 1, device=dev) < 0.05,
+                "terminated": torch.rand(B, T, 1, device=dev) < 0.02,
+            },
+        },
+        batch_size=[B, T],
+    )
+    g1 = GAE(gamma=0.99, lmbda=0.95, value_network=critic, vectorized=True)
+    g2 = GAE(gamma=0.99, lmbda=0.95, value_network=critic_ref, vectorized=True)
+    with torch.no_grad():
+        a1 = g1(td.clone()).get("advantage")
+        with torch.autocast("cuda", dtype=torch.bfloat16, cache_enabled=False):
+            a2 = g2(td.clone()).get("advantage")
+    # identical weights; paired path runs bf16 caches vs eager autocast
+    assert torch.allclose(a1, a2, atol=3e-2, rtol=1e-2), (a1 - a2).abs().max()

@@ -467,3 +467,24 @@ def test_graphed_ppo_multi_epoch_gpu():
     for p in actor.parameters():
         assert torch.isfinite(p).all()
     col.shutdown()
+
+
+def test_total_loss_prefers_kernel_presums():
+    """GraphedPPO._total_loss: _loss_total wins; _loss_actor skips the
+    component keys; plain sums otherwise."""
+    import torch
+
+    from rl_amd.tensordict import TensorDict
+    from rl_amd.trainers.graphed import GraphedPPO
+
+    t = lambda v: torch.tensor(float(v))
+    self = type("S", (), {"_total_loss": GraphedPPO._total_loss})()
+    out = TensorDict({"loss_objective": t(1), "loss_entropy": t(2),
+                      "loss_critic": t(3)}, batch_size=[])
+    assert float(GraphedPPO._total_loss(self, out)) == 6.0
+    out.set("_loss_actor", t(3))  # pre-summed obj+ent
+    assert float(GraphedPPO._total_loss(self, out)) == 6.0
+    out.set("_loss_total", t(6))
+    assert float(GraphedPPO._total_loss(self, out)) == 6.0
+    out.set("loss_extra", t(4))  # extra loss keys still accumulate
+    assert float(GraphedPPO._total_loss(self, out)) == 10.0
