@@ -107,7 +107,8 @@ PIXEL_TRANSFORMS = [
 
 
 @pytest.mark.parametrize(
-    "transform", VEC_TRANSFORMS, ids=lambda t: type(t).__name__ + str(id(t) % 97)
+    "transform", VEC_TRANSFORMS,
+    ids=[type(t).__name__ + str(i) for i, t in enumerate(VEC_TRANSFORMS)]
 )
 def test_vec_transform_spec_contract(transform):
     env = TransformedEnv(make_vec_env(), transform)
@@ -115,7 +116,8 @@ def test_vec_transform_spec_contract(transform):
 
 
 @pytest.mark.parametrize(
-    "transform", PIXEL_TRANSFORMS, ids=lambda t: type(t).__name__ + str(id(t) % 97)
+    "transform", PIXEL_TRANSFORMS,
+    ids=[type(t).__name__ + str(i) for i, t in enumerate(PIXEL_TRANSFORMS)]
 )
 def test_pixel_transform_spec_contract(transform):
     env = TransformedEnv(PixelEnv(batch_size=[2]), transform)
