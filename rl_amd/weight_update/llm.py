@@ -202,7 +202,13 @@ class LLMDoubleBufferWeightSyncScheme(WeightSyncScheme):
         if name is None or version == self._seen_version:
             return False
         arr = np.memmap(os.path.join(self.path, name), dtype=np.uint8, mode="r")
-        buf = torch.from_numpy(np.asarray(arr))
+        # the reader only COPIES from this view (torch warns about
+        # non-writable arrays; writing would indeed be UB — we never do)
+        import warnings
+
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore", UserWarning)
+            buf = torch.from_numpy(np.asarray(arr))
         sd = {
             k: v
             for k, v in model.state_dict().items()
