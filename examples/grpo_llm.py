@@ -1,43 +1,74 @@
-"""GRPO on a tiny random-init LM — the RLHF vertical end to end.
+"""GRPO RLHF loop on a tiny random-init LM (reference
+sota-implementations/grpo shape): PromptDataset → ChatEnv →
+LLMCollector (sampling generation) → MCAdvantage (per-group whitening)
+→ GRPOLoss (+ optional KL-to-reference) → Adam.
 
-Reference analog: pytorch/rl sota-implementations/grpo.  Pipeline:
-PromptDataset → ChatEnv (length reward) → LLMCollector (generation) →
-MCAdvantage group baseline → GRPOLoss (+ KL to reference) → Adam.
-Runs on CPU in under a minute; swap in a transformers checkpoint and a
-real reward model for production use.
+Runs anywhere (CPU ok; tiny model).  For production GPT-2 decode on
+MI355X see rl_amd.modules.llm.decode.GraphedGPT2Decoder (one hipGraph
+replay per generated token — the RLHF bench's rollout engine).
+
+Run: python examples/grpo_llm.py [--iters 4]
 """
-import sys, os
-sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from __future__ import annotations
+
+import argparse
+import os
+import sys
 
 import torch
 
-from rl_amd.collectors import LLMCollector
-from rl_amd.data.llm import PromptDataset
-from rl_amd.envs import ChatEnv
-from rl_amd.modules import TransformersWrapper
-from rl_amd.objectives import GRPOLoss, MCAdvantage
-from rl_amd.testing.llm_mocks import ByteTokenizer, make_tiny_lm
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
-def main(total_turns: int = 16, group: int = 4):
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--iters", type=int, default=4)
+    args = p.parse_args()
+
+    try:
+        import transformers  # noqa: F401
+    except ImportError:
+        print("transformers not installed; skipping")
+        return
+
+    from rl_amd.data.llm import PromptDataset
+    from rl_amd.envs.llm import ChatEnv
+    from rl_amd.collectors.llm import LLMCollector
+    from rl_amd.modules.llm import TransformersWrapper
+    from rl_amd.objectives import GRPOLoss, MCAdvantage
+
     torch.manual_seed(0)
-    lm, tok = make_tiny_lm(), ByteTokenizer()
-    prompts = PromptDataset(["2+2?", "3+3?", "capital of France?", "9*9?"], repeat=True)
-    env = ChatEnv(iter(prompts), reward_fn=lambda h: float(len(h.last_content)))
-    policy = TransformersWrapper(lm, tokenizer=tok, generate=True, max_new_tokens=6)
-    col = LLMCollector(env, policy, dialog_turns_per_batch=group, total_dialog_turns=total_turns)
+    # tiny random-init LM + byte-level tokenizer (no network needed);
+    # swap in GPT2LMHeadModel + AutoTokenizer for the real thing
+    from rl_amd.testing.llm_mocks import ByteTokenizer, make_tiny_lm
+
+    lm = make_tiny_lm()
+    tok = ByteTokenizer()
+    prompts = PromptDataset(
+        ["2+2?", "3+3?", "4+4?", "5+5?", "6+6?", "7+7?"], repeat=True
+    )
+    env = ChatEnv(iter(prompts),
+                  reward_fn=lambda h: float(len(h.last_content)))
+    policy = TransformersWrapper(lm, tokenizer=tok, generate=True,
+                                 max_new_tokens=6)
     actor = TransformersWrapper(lm, tokenizer=tok, generate=False)
+    col = LLMCollector(env, policy, dialog_turns_per_batch=4,
+                       total_dialog_turns=4 * args.iters)
+    adv = MCAdvantage(grpo_size=2)
     loss_mod = GRPOLoss(actor)
     optim = torch.optim.Adam(lm.parameters(), lr=1e-4)
+
     for i, batch in enumerate(col):
-        batch = batch.reshape(-1)
-        MCAdvantage(grpo_size=group)(batch)
+        adv(batch)
         out = loss_mod(batch)
+        total = out.get("loss_objective")
         optim.zero_grad()
-        out.get("loss_objective").backward()
+        total.backward()
         optim.step()
         r = batch.get(("next", "reward")).float().mean().item()
-        print(f"update {i}: mean reward {r:.2f}, loss {out.get('loss_objective').item():.4f}")
+        print(f"iter {i}: loss {float(total):.4f} mean reward {r:.2f}")
+    col.shutdown()
+    print("done")
 
 
 if __name__ == "__main__":
