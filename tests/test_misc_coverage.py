@@ -377,3 +377,40 @@ def test_splitk_refresh_foreach_matches_per_layer():
     refresh_splitk_caches(mod)
     assert torch.equal(lin1.weight_bf16, lin1.weight.detach().to(torch.bfloat16))
     assert torch.equal(lin2.bias_bf16, lin2.bias.detach().to(torch.bfloat16))
+
+
+class TestFusedModelCheckpoint:
+    def test_fused_mlp3_state_dict_roundtrip(self):
+        """state_dict through SplitK + bf16 cache + FusedMLP3 wrappers:
+        loads cleanly and preserves the shared-weight aliasing the fast
+        path depends on."""
+        import torch
+
+        from rl_amd.ops import (
+            convert_linears_to_splitk,
+            enable_splitk_bf16_cache,
+            fuse_mlp3,
+        )
+
+        def build(seed):
+            torch.manual_seed(seed)
+            net = torch.nn.Sequential(
+                torch.nn.Linear(7, 32), torch.nn.Tanh(),
+                torch.nn.Linear(32, 32), torch.nn.Tanh(),
+                torch.nn.Linear(32, 4),
+            )
+            convert_linears_to_splitk(net)
+            enable_splitk_bf16_cache(net)
+            return fuse_mlp3(net)
+
+        a, b = build(0), build(1)
+        missing, unexpected = b.load_state_dict(a.state_dict())
+        assert not missing and not unexpected
+        for (k, p1), (_, p2) in zip(a.state_dict().items(),
+                                    b.state_dict().items()):
+            assert torch.equal(p1, p2), k
+        # the fused fast path reads lin*, the eager fallback reads
+        # eager.* — they must stay the SAME tensors after loading
+        assert b.lin1.weight is b.eager[0].weight
+        x = torch.randn(5, 7)
+        assert torch.allclose(a.eager(x), b.eager(x))
