@@ -414,3 +414,27 @@ class TestFusedModelCheckpoint:
         assert b.lin1.weight is b.eager[0].weight
         x = torch.randn(5, 7)
         assert torch.allclose(a.eager(x), b.eager(x))
+
+
+class TestExamplesSmoke:
+    """The examples README section claims every script runs here —
+    enforce a fast subset as subprocesses."""
+
+    @pytest.mark.parametrize("script,args", [
+        ("offline_iql.py", ["--steps", "20"]),
+        ("decision_transformer.py", ["--steps", "10"]),
+        ("grpo_llm.py", ["--iters", "1"]),
+    ])
+    @pytest.mark.timeout(300)
+    def test_example_runs(self, script, args):
+        import os
+        import subprocess
+        import sys
+
+        root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        r = subprocess.run(
+            [sys.executable, os.path.join(root, "examples", script)] + args,
+            capture_output=True, text=True, timeout=280, cwd=root,
+        )
+        assert r.returncode == 0, r.stderr[-800:]
+        assert "done" in r.stdout
